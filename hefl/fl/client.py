@@ -1,0 +1,103 @@
+"""Per-client local training.
+
+Replaces the reference's `train_clients` (FLPyfhelin.py:179-198). Deliberate
+behavioral delta (SURVEY.md section 7 quirks catalog): each client trains an
+INDEPENDENT model initialized from the global weights each round — the
+reference's shared-model-across-clients bug (FLPyfhelin.py:180, the :194
+reload is commented out) is not reproduced.
+"""
+from __future__ import annotations
+
+import time
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional
+
+import torch
+
+from ..config import RunConfig
+from ..data.synthetic import ClientLoader, SyntheticMedicalImages
+from ..data.shard import shard_indices
+from ..models import build_model
+from ..ops.adam import FusedAdam
+from ..ops.functional import softmax_xent
+from .weights import flat_params, load_flat_params
+
+
+@dataclass
+class RoundStats:
+    train_loss: float = 0.0
+    train_acc: float = 0.0
+    samples: int = 0
+    steps: int = 0
+    seconds: float = 0.0
+    phase_seconds: Dict[str, float] = field(default_factory=dict)
+
+
+class LocalClient:
+    """One FL client: a model + optimizer + data shard on one device."""
+
+    def __init__(self, cfg: RunConfig, client_id: int, device: str = "cpu"):
+        self.cfg = cfg
+        self.client_id = client_id
+        self.device = torch.device(device)
+        self.model = build_model(cfg.model, seed=cfg.fl.seed).to(self.device)
+        t = cfg.train
+        self.opt = FusedAdam(self.model.parameters(), lr=t.lr, decay=t.lr_decay,
+                             beta1=t.beta1, beta2=t.beta2, eps=t.eps)
+        self.compute_dtype = (torch.bfloat16 if (self.device.type == "cuda"
+                              and t.dtype == "bf16") else torch.float32)
+        # Data: the full federated dataset is conceptually
+        # n_clients * samples_per_client samples; this client materializes its
+        # contiguous shard (reference sharding semantics, FLPyfhelin.py:75-78).
+        n_total = cfg.fl.n_clients * cfg.fl.samples_per_client
+        self.dataset = SyntheticMedicalImages(
+            n_total, cfg.model.in_shape, cfg.model.n_classes,
+            seed=cfg.fl.seed, device=device)
+        idx = shard_indices(n_total, client_id, cfg.fl.n_clients)
+        self.loader = ClientLoader(self.dataset, idx, t.batch_size,
+                                   seed=cfg.fl.seed + client_id)
+
+    def train_step(self, x: torch.Tensor, y: torch.Tensor):
+        x = x.to(self.compute_dtype)
+        logits = self.model(x)
+        loss = softmax_xent(logits, y)
+        self.opt.zero_grad()
+        loss.backward()
+        self.opt.step()
+        return loss, logits
+
+    def local_train(self, epochs: Optional[int] = None) -> RoundStats:
+        epochs = self.cfg.train.local_epochs if epochs is None else epochs
+        stats = RoundStats()
+        t0 = time.perf_counter()
+        for _ in range(epochs):
+            for x, y in self.loader:
+                loss, logits = self.train_step(x, y)
+                stats.steps += 1
+                stats.samples += y.numel()
+                stats.train_loss += float(loss.detach())
+                stats.train_acc += float((logits.detach().float().argmax(-1) == y)
+                                         .float().sum())
+        stats.seconds = time.perf_counter() - t0
+        if stats.steps:
+            stats.train_loss /= stats.steps
+            stats.train_acc /= max(stats.samples, 1)
+        return stats
+
+    def get_weights(self) -> torch.Tensor:
+        return flat_params(self.model)
+
+    def set_weights(self, vec: torch.Tensor) -> None:
+        load_flat_params(self.model, vec.to(self.device))
+
+    @torch.no_grad()
+    def evaluate(self, dataset: SyntheticMedicalImages, indices: torch.Tensor,
+                 batch_size: int = 64):
+        """Return (y_true, y_pred) over the given samples."""
+        preds, trues = [], []
+        for i in range(0, indices.numel(), batch_size):
+            x, y = dataset.batch(indices[i:i + batch_size])
+            logits = self.model(x.to(self.compute_dtype))
+            preds.append(logits.float().argmax(-1).cpu())
+            trues.append(y.cpu())
+        return torch.cat(trues), torch.cat(preds)

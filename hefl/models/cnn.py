@@ -1,0 +1,91 @@
+"""CNN model families (NHWC, valid padding) for the BASELINE.json configs.
+
+- RefCNN6: the reference's exact architecture — 6x[Conv3x3 valid + ReLU +
+  MaxPool2x2] with filters 32,32,32,64,64,128 -> Flatten(512) -> Dense 128
+  ReLU -> Dense 64 ReLU -> Dense n_classes. 222,722 params at 256x256x3 / 2
+  classes (FLPyfhelin.py:118-141; param count verified in tests).
+- CNN2: config #2 headline model (2-conv on 28x28).
+- LeNet5: config #3 (32x32x3).
+- CNN4: config #4 (4-conv on 224x224).
+
+Models output LOGITS; softmax lives in the fused softmax-CE loss (the
+reference's softmax head + categorical cross-entropy, FLPyfhelin.py:138,141).
+"""
+from __future__ import annotations
+
+import torch
+import torch.nn as nn
+
+from ..config import ModelConfig
+from ..ops.modules import Conv2dValid, Dense, Flatten, MaxPool2x2
+
+
+class _SeqCNN(nn.Module):
+    """Conv/pool trunk + dense head, NHWC."""
+
+    def __init__(self, in_shape, conv_filters, conv_k, dense_units, n_classes,
+                 seed: int = 0):
+        super().__init__()
+        gen = torch.Generator().manual_seed(seed)
+        H, W, C = in_shape
+        layers = []
+        cin = C
+        for f in conv_filters:
+            conv = Conv2dValid(cin, f, k=conv_k, relu=True, gen=gen)
+            layers += [conv, MaxPool2x2()]
+            H, W = conv.out_hw(H, W)
+            H, W = H // 2, W // 2
+            cin = f
+        self.trunk = nn.ModuleList(layers)
+        self.flatten = Flatten()
+        feat = H * W * cin
+        self.feat_dim = feat
+        head = []
+        din = feat
+        for u in dense_units:
+            head.append(Dense(din, u, relu=True, gen=gen))
+            din = u
+        head.append(Dense(din, n_classes, relu=False, gen=gen))
+        self.head = nn.ModuleList(head)
+
+    def forward(self, x):
+        for m in self.trunk:
+            x = m(x)
+        x = self.flatten(x)
+        for m in self.head:
+            x = m(x)
+        return x
+
+    def n_params(self) -> int:
+        return sum(p.numel() for p in self.parameters())
+
+
+class RefCNN6(_SeqCNN):
+    def __init__(self, in_shape=(256, 256, 3), n_classes=2, seed=0):
+        super().__init__(in_shape, (32, 32, 32, 64, 64, 128), 3, (128, 64),
+                         n_classes, seed)
+
+
+class CNN2(_SeqCNN):
+    def __init__(self, in_shape=(28, 28, 1), n_classes=10, seed=0):
+        super().__init__(in_shape, (16, 32), 3, (64,), n_classes, seed)
+
+
+class CNN4(_SeqCNN):
+    def __init__(self, in_shape=(224, 224, 1), n_classes=2, seed=0):
+        super().__init__(in_shape, (32, 64, 64, 128), 3, (128,), n_classes, seed)
+
+
+class LeNet5(_SeqCNN):
+    def __init__(self, in_shape=(32, 32, 3), n_classes=10, seed=0):
+        super().__init__(in_shape, (6, 16), 5, (120, 84), n_classes, seed)
+
+
+def build_model(cfg: ModelConfig, seed: int = 0) -> _SeqCNN:
+    cls = {"cnn2": CNN2, "refcnn6": RefCNN6, "cnn4": CNN4, "lenet5": LeNet5}
+    if cfg.name == "resnet18":
+        from .resnet import ResNet18
+        return ResNet18(cfg.in_shape, cfg.n_classes, seed=seed)
+    if cfg.name not in cls:
+        raise KeyError(f"unknown model {cfg.name}")
+    return cls[cfg.name](cfg.in_shape, cfg.n_classes, seed=seed)

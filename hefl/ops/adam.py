@@ -1,0 +1,65 @@
+"""Adam with Keras-style lr decay (lr_t = lr / (1 + decay * t)).
+
+Matches the reference's optimizer: Adam(learning_rate=INIT_LR,
+decay=INIT_LR/10) with epsilon=1e-7 (FLPyfhelin.py:140; Keras defaults).
+GPU path is one fused HIP kernel per parameter tensor (m, v update, bias
+correction, decayed lr, parameter write) — replacing TF's Adam kernels
+(Adam row).
+"""
+from __future__ import annotations
+
+import math
+from typing import Iterable
+
+import torch
+
+import hefl
+
+
+class FusedAdam:
+    def __init__(self, params: Iterable[torch.nn.Parameter], lr: float = 1e-3,
+                 decay: float = 1e-4, beta1: float = 0.9, beta2: float = 0.999,
+                 eps: float = 1e-7):
+        self.params = [p for p in params if p.requires_grad]
+        self.lr, self.decay = lr, decay
+        self.beta1, self.beta2, self.eps = beta1, beta2, eps
+        self.step_count = 0
+        self.m = [torch.zeros_like(p, dtype=torch.float32) for p in self.params]
+        self.v = [torch.zeros_like(p, dtype=torch.float32) for p in self.params]
+
+    @torch.no_grad()
+    def step(self):
+        self.step_count += 1
+        t = self.step_count
+        lr_t = self.lr / (1.0 + self.decay * (t - 1))  # Keras decay schedule
+        bc1 = 1.0 - self.beta1 ** t
+        bc2 = 1.0 - self.beta2 ** t
+        for p, m, v in zip(self.params, self.m, self.v):
+            if p.grad is None:
+                continue
+            g = p.grad
+            if p.is_cuda:
+                hefl.load_extension().fused_adam(
+                    p.data, g, m, v, lr_t, self.beta1, self.beta2, self.eps, bc1, bc2)
+            else:
+                g = g.float()
+                m.mul_(self.beta1).add_(g, alpha=1 - self.beta1)
+                v.mul_(self.beta2).addcmul_(g, g, value=1 - self.beta2)
+                mh = m / bc1
+                vh = v / bc2
+                p.data.add_(-lr_t * mh / (vh.sqrt() + self.eps))
+
+    def zero_grad(self):
+        for p in self.params:
+            p.grad = None
+
+    def state_dict(self):
+        return {"step": self.step_count, "m": self.m, "v": self.v,
+                "lr": self.lr, "decay": self.decay}
+
+    def load_state_dict(self, sd):
+        self.step_count = sd["step"]
+        for dst, src in zip(self.m, sd["m"]):
+            dst.copy_(src)
+        for dst, src in zip(self.v, sd["v"]):
+            dst.copy_(src)

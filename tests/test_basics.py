@@ -1,0 +1,82 @@
+"""Config / data / model structure tests (CPU)."""
+import pytest
+import torch
+
+from hefl.config import HEConfig, preset
+from hefl.data.shard import shard_indices
+from hefl.data.synthetic import ClientLoader, SyntheticMedicalImages
+from hefl.models import CNN2, LeNet5, RefCNN6, build_model
+
+
+def test_presets():
+    for name in ["config1", "config2", "config3", "config4", "config5", "reference"]:
+        cfg = preset(name)
+        assert cfg.he.m & (cfg.he.m - 1) == 0
+    assert preset("config2").he.m == 2 ** 13
+    assert preset("config3").he.m == 2 ** 14
+    assert preset("config5").he.m == 2 ** 15
+
+
+def test_he_config_validation():
+    with pytest.raises(ValueError):
+        HEConfig(m=1000)
+    with pytest.raises(ValueError):
+        HEConfig(q_bits=(62,))  # violates lazy-allreduce bound
+
+
+def test_shard_contiguous_equal():
+    # reference semantics: ratio = len // n, shard i = [i*ratio, (i+1)*ratio)
+    idx0 = shard_indices(1601, 0, 2)
+    idx1 = shard_indices(1601, 1, 2)
+    assert idx0.numel() == idx1.numel() == 800
+    assert idx0[0] == 0 and idx0[-1] == 799
+    assert idx1[0] == 800 and idx1[-1] == 1599  # tail sample 1600 dropped
+    assert not set(idx0.tolist()) & set(idx1.tolist())
+
+
+def test_synthetic_deterministic():
+    ds1 = SyntheticMedicalImages(64, (28, 28, 1), 10, seed=7)
+    ds2 = SyntheticMedicalImages(64, (28, 28, 1), 10, seed=7)
+    idx = torch.arange(8)
+    x1, y1 = ds1.batch(idx)
+    x2, y2 = ds2.batch(idx)
+    assert torch.equal(x1, x2) and torch.equal(y1, y2)
+    assert x1.shape == (8, 28, 28, 1)
+    assert float(x1.min()) >= 0 and float(x1.max()) <= 1
+
+
+def test_loader_epochs():
+    ds = SyntheticMedicalImages(100, (28, 28, 1), 10, seed=1)
+    loader = ClientLoader(ds, torch.arange(70), batch_size=32, seed=3)
+    batches = list(loader)
+    assert len(batches) == 3  # 32 + 32 + 6
+    assert sum(b[1].numel() for b in batches) == 70
+
+
+def test_refcnn6_param_count():
+    m = RefCNN6((256, 256, 3), 2)
+    # Reference: 222,722 params (model factory row)
+    assert m.n_params() == 222722
+    assert m.feat_dim == 512
+
+
+def test_model_shapes():
+    m = CNN2((28, 28, 1), 10)
+    x = torch.rand(4, 28, 28, 1)
+    out = m(x)
+    assert out.shape == (4, 10)
+
+    m = LeNet5((32, 32, 3), 10)
+    out = m(torch.rand(2, 32, 32, 3))
+    assert out.shape == (2, 10)
+
+    m = build_model(preset("config4").model)
+    out = m(torch.rand(1, 224, 224, 1))
+    assert out.shape == (1, 2)
+
+
+def test_model_deterministic_init():
+    a = CNN2(seed=5)
+    b = CNN2(seed=5)
+    for pa, pb in zip(a.parameters(), b.parameters()):
+        assert torch.equal(pa, pb)
