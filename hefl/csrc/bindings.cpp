@@ -1,0 +1,65 @@
+// hefl._C — torch extension binding the gfx950 HIP kernels:
+// CKKS NTT/pointwise (ntt.hip) + CNN training ops (cnn.hip).
+#include <torch/extension.h>
+
+#include <vector>
+
+// ntt.hip
+void ntt_batch(torch::Tensor x, torch::Tensor w, torch::Tensor wsh, int64_t q);
+void intt_batch(torch::Tensor x, torch::Tensor winv, torch::Tensor winvsh,
+                int64_t q, int64_t ninv, int64_t ninvsh);
+torch::Tensor modmul(torch::Tensor a, torch::Tensor b, int64_t q);
+torch::Tensor modmul_scalar(torch::Tensor a, int64_t s, int64_t q);
+torch::Tensor modadd(torch::Tensor a, torch::Tensor b, int64_t q);
+torch::Tensor modsub(torch::Tensor a, torch::Tensor b, int64_t q);
+void modreduce_(torch::Tensor x, torch::Tensor qs);
+
+// cnn.hip
+torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
+                         int64_t stride, bool relu, int64_t pad);
+torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor w, int64_t stride,
+                           int64_t H, int64_t W, int64_t pad);
+torch::Tensor conv2d_wgrad(torch::Tensor dy, torch::Tensor x, int64_t stride,
+                           int64_t R, int64_t S, int64_t pad);
+torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
+                         bool relu);
+torch::Tensor linear_dgrad(torch::Tensor dy, torch::Tensor w);
+torch::Tensor linear_wgrad(torch::Tensor dy, torch::Tensor x);
+std::vector<torch::Tensor> maxpool2x2_fwd(torch::Tensor x);
+torch::Tensor maxpool2x2_bwd(torch::Tensor dy, torch::Tensor idx, int64_t H,
+                             int64_t W);
+std::vector<torch::Tensor> softmax_xent_fwd(torch::Tensor logits,
+                                            torch::Tensor labels);
+torch::Tensor softmax_xent_bwd(torch::Tensor probs, torch::Tensor labels,
+                               double scale);
+void fused_adam(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+                torch::Tensor v, double lr, double b1, double b2, double eps,
+                double bc1, double bc2);
+torch::Tensor relu_bwd(torch::Tensor dy, torch::Tensor y);
+torch::Tensor bias_grad(torch::Tensor dy);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.doc() = "hefl gfx950 (MI355X/CDNA4) HIP kernels";
+    // HE / CKKS
+    m.def("ntt_batch", &ntt_batch, "in-place forward negacyclic NTT [rows, n]");
+    m.def("intt_batch", &intt_batch, "in-place inverse negacyclic NTT");
+    m.def("modmul", &modmul, "pointwise Barrett modmul");
+    m.def("modmul_scalar", &modmul_scalar, "pointwise Shoup scalar modmul");
+    m.def("modadd", &modadd);
+    m.def("modsub", &modsub);
+    m.def("modreduce_", &modreduce_, "in-place per-limb reduction after lazy sum");
+    // CNN
+    m.def("conv2d_fwd", &conv2d_fwd);
+    m.def("conv2d_dgrad", &conv2d_dgrad);
+    m.def("conv2d_wgrad", &conv2d_wgrad);
+    m.def("linear_fwd", &linear_fwd);
+    m.def("linear_dgrad", &linear_dgrad);
+    m.def("linear_wgrad", &linear_wgrad);
+    m.def("maxpool2x2_fwd", &maxpool2x2_fwd);
+    m.def("maxpool2x2_bwd", &maxpool2x2_bwd);
+    m.def("softmax_xent_fwd", &softmax_xent_fwd);
+    m.def("softmax_xent_bwd", &softmax_xent_bwd);
+    m.def("fused_adam", &fused_adam);
+    m.def("relu_bwd", &relu_bwd);
+    m.def("bias_grad", &bias_grad);
+}

@@ -1,0 +1,750 @@
+// CNN training kernels for gfx950 (CDNA4): MFMA implicit-GEMM conv
+// fwd/dgrad/wgrad, MFMA linear fwd/dgrad/wgrad, maxpool 2x2, fused
+// softmax-CE, fused Adam, bias grad, ReLU bwd.
+//
+// Replaces the TF/Keras native surface the reference leans on for
+// model.fit/predict (reference model at
+// FLPyfhelin.py:118-141, Adam at :140, loss at :141).
+//
+// Layout: NHWC activations (bf16), [K,R,S,C] weights (bf16 on device,
+// fp32 master copy in Python), fp32 accumulation via
+// v_mfma_f32_16x16x32_bf16 (one 16x16 output fragment per wave,
+// 4 waves per workgroup covering 64 output rows).
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+#include <algorithm>
+
+#define CHECK_GPU(x) TORCH_CHECK((x).is_cuda(), #x " must be on GPU")
+
+namespace {
+
+using bf16_t = __bf16;
+typedef bf16_t bf16x8 __attribute__((ext_vector_type(8)));
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+__device__ __forceinline__ float bf2f(unsigned short u) {
+    union { unsigned int i; float f; } v;
+    v.i = ((unsigned int)u) << 16;
+    return v.f;
+}
+
+__device__ __forceinline__ unsigned short f2bf(float f) {
+    union { float f; unsigned int i; } v;
+    v.f = f;
+    unsigned int x = v.i;
+    unsigned int lsb = (x >> 16) & 1;           // round-to-nearest-even
+    x += 0x7fffu + lsb;
+    if ((v.i & 0x7fffffffu) > 0x7f800000u) return 0x7fc0u;  // NaN
+    return (unsigned short)(x >> 16);
+}
+
+// ---------------------------------------------------------------------------
+// Shared MFMA tile machinery: BM=64 (4 waves stacked on M), BN=16, BK=32.
+// LDS images As[64][32] and Bs[16][32] (B stored col-major-as-rows so each
+// lane's 8 k-consecutive bf16 are one 16-B read).
+// Fragment maps for v_mfma_f32_16x16x32_bf16:
+//   A: row = lane&15, k = (lane>>4)*8 + j   (j = 0..7)
+//   B: col = lane&15, k = (lane>>4)*8 + j
+//   D: col = lane&15, row = (lane>>4)*4 + r (r = 0..3)
+// ---------------------------------------------------------------------------
+
+constexpr int BM = 64;
+constexpr int BN = 16;
+constexpr int BK = 32;
+constexpr int TPB = 256;  // 4 waves
+
+struct TileSmem {
+    unsigned short As[BM][BK];
+    unsigned short Bs[BN][BK];
+};
+
+__device__ __forceinline__ f32x4 tile_mfma_step(const TileSmem& sm, int wave,
+                                                int lane, f32x4 acc) {
+    const int half = lane >> 4;          // 0..3 (k-slice of 8)
+    const int sub = lane & 15;
+    bf16x8 a = *reinterpret_cast<const bf16x8*>(&sm.As[wave * 16 + sub][half * 8]);
+    bf16x8 b = *reinterpret_cast<const bf16x8*>(&sm.Bs[sub][half * 8]);
+    return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+}
+
+// ---------------------------------------------------------------------------
+// Generic GEMM: C[M,N] = A[M,K] x B + bias, optional relu.
+// AT: A stored [K,M] (read transposed); BT: B stored [N,K].
+// OUT_BF16: write bf16, else f32.
+// ---------------------------------------------------------------------------
+
+template <bool AT, bool BT, bool OUT_BF16, bool ACCUM>
+__global__ void __launch_bounds__(TPB)
+gemm_kernel(const unsigned short* __restrict__ A,
+            const unsigned short* __restrict__ B, void* __restrict__ C,
+            const float* __restrict__ bias, int M, int N, int K, int relu) {
+    __shared__ TileSmem sm;
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+    const int m0 = blockIdx.x * BM;
+    const int n0 = blockIdx.y * BN;
+
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    for (int k0 = 0; k0 < K; k0 += BK) {
+        // stage A: 64x32 = 2048 elems, 8 per thread
+        for (int i = tid; i < BM * BK / 8; i += TPB) {
+            int row = i / (BK / 8);
+            int kc = (i % (BK / 8)) * 8;
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                int m = m0 + row, k = k0 + kc + j;
+                unsigned short v = 0;
+                if (m < M && k < K)
+                    v = AT ? A[(int64_t)k * M + m] : A[(int64_t)m * K + k];
+                sm.As[row][kc + j] = v;
+            }
+        }
+        // stage B: 16x32 = 512 elems
+        for (int i = tid; i < BN * BK / 8; i += TPB) {
+            int col = i / (BK / 8);
+            int kc = (i % (BK / 8)) * 8;
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                int nn = n0 + col, k = k0 + kc + j;
+                unsigned short v = 0;
+                if (nn < N && k < K)
+                    v = BT ? B[(int64_t)nn * K + k] : B[(int64_t)k * N + nn];
+                sm.Bs[col][kc + j] = v;
+            }
+        }
+        __syncthreads();
+        acc = tile_mfma_step(sm, wave, lane, acc);
+        __syncthreads();
+    }
+    // epilogue
+    const int col = n0 + (lane & 15);
+    if (col >= N) return;
+    float bv = bias ? bias[col] : 0.f;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        int row = m0 + wave * 16 + (lane >> 4) * 4 + r;
+        if (row >= M) continue;
+        float v = acc[r] + bv;
+        if (relu) v = v > 0.f ? v : 0.f;
+        if (OUT_BF16) {
+            reinterpret_cast<unsigned short*>(C)[(int64_t)row * N + col] = f2bf(v);
+        } else if (ACCUM) {
+            atomicAdd(reinterpret_cast<float*>(C) + (int64_t)row * N + col, v);
+        } else {
+            reinterpret_cast<float*>(C)[(int64_t)row * N + col] = v;
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
+// Conv2d forward, implicit GEMM: M = N*OH*OW, N-dim = Kout, K-dim = R*S*C.
+// x [N,H,W,C] bf16; w [Kout,R,S,C] bf16; y [N,OH,OW,Kout] bf16.
+// ---------------------------------------------------------------------------
+
+struct ConvShape {
+    int N, H, W, C, Kout, R, S, OH, OW, stride, pad;
+};
+
+__global__ void __launch_bounds__(TPB)
+conv_fwd_kernel(const unsigned short* __restrict__ x,
+                const unsigned short* __restrict__ w,
+                const float* __restrict__ bias, unsigned short* __restrict__ y,
+                ConvShape s, int relu) {
+    __shared__ TileSmem sm;
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+    const int m0 = blockIdx.x * BM;
+    const int n0 = blockIdx.y * BN;
+    const int M = s.N * s.OH * s.OW;
+    const int KK = s.R * s.S * s.C;
+
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    for (int k0 = 0; k0 < KK; k0 += BK) {
+        for (int i = tid; i < BM * BK / 8; i += TPB) {
+            int row = i / (BK / 8);
+            int kc = (i % (BK / 8)) * 8;
+            int m = m0 + row;
+            int n_ = m / (s.OH * s.OW);
+            int rem = m % (s.OH * s.OW);
+            int oh = rem / s.OW, ow = rem % s.OW;
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                int k = k0 + kc + j;
+                unsigned short v = 0;
+                if (m < M && k < KK) {
+                    int c = k % s.C;
+                    int rs = k / s.C;        // k-dim ordered (r, s, c)
+                    int r = rs / s.S, ss = rs % s.S;
+                    int ih = oh * s.stride + r - s.pad;
+                    int iw = ow * s.stride + ss - s.pad;
+                    if (ih >= 0 && ih < s.H && iw >= 0 && iw < s.W)
+                        v = x[(((int64_t)n_ * s.H + ih) * s.W + iw) * s.C + c];
+                }
+                sm.As[row][kc + j] = v;
+            }
+        }
+        for (int i = tid; i < BN * BK / 8; i += TPB) {
+            int col = i / (BK / 8);
+            int kc = (i % (BK / 8)) * 8;
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                int ko = n0 + col, k = k0 + kc + j;
+                unsigned short v = 0;
+                if (ko < s.Kout && k < KK) {
+                    int c = k % s.C;
+                    int rs = k / s.C;
+                    v = w[((int64_t)ko * s.R * s.S + rs) * s.C + c];
+                }
+                sm.Bs[col][kc + j] = v;
+            }
+        }
+        __syncthreads();
+        acc = tile_mfma_step(sm, wave, lane, acc);
+        __syncthreads();
+    }
+    const int col = n0 + (lane & 15);
+    if (col >= s.Kout) return;
+    float bv = bias ? bias[col] : 0.f;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        int row = m0 + wave * 16 + (lane >> 4) * 4 + r;
+        if (row >= M) continue;
+        float v = acc[r] + bv;
+        if (relu) v = v > 0.f ? v : 0.f;
+        y[(int64_t)row * s.Kout + col] = f2bf(v);
+    }
+}
+
+// Conv2d dgrad: dx[n,ih,iw,c] = sum_{ko,r,s} dy[n,oh,ow,ko] * w[ko,r,s,c]
+// where oh = (ih + pad - r)/stride when divisible. Implicit GEMM:
+// M = N*H*W, N-dim = C, K-dim = Kout*R*S, B = w viewed [Kout*R*S, C].
+__global__ void __launch_bounds__(TPB)
+conv_dgrad_kernel(const unsigned short* __restrict__ dy,
+                  const unsigned short* __restrict__ w,
+                  unsigned short* __restrict__ dx, ConvShape s) {
+    __shared__ TileSmem sm;
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+    const int m0 = blockIdx.x * BM;
+    const int n0 = blockIdx.y * BN;
+    const int M = s.N * s.H * s.W;
+    const int KK = s.Kout * s.R * s.S;
+
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    for (int k0 = 0; k0 < KK; k0 += BK) {
+        for (int i = tid; i < BM * BK / 8; i += TPB) {
+            int row = i / (BK / 8);
+            int kc = (i % (BK / 8)) * 8;
+            int m = m0 + row;
+            int n_ = m / (s.H * s.W);
+            int rem = m % (s.H * s.W);
+            int ih = rem / s.W, iw = rem % s.W;
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                int k = k0 + kc + j;
+                unsigned short v = 0;
+                if (m < M && k < KK) {
+                    int rs = k % (s.R * s.S);
+                    int ko = k / (s.R * s.S);
+                    int r = rs / s.S, ss = rs % s.S;
+                    int oh_num = ih + s.pad - r, ow_num = iw + s.pad - ss;
+                    if (oh_num >= 0 && ow_num >= 0 && oh_num % s.stride == 0 &&
+                        ow_num % s.stride == 0) {
+                        int oh = oh_num / s.stride, ow = ow_num / s.stride;
+                        if (oh < s.OH && ow < s.OW)
+                            v = dy[(((int64_t)n_ * s.OH + oh) * s.OW + ow) *
+                                       s.Kout + ko];
+                    }
+                }
+                sm.As[row][kc + j] = v;
+            }
+        }
+        for (int i = tid; i < BN * BK / 8; i += TPB) {
+            int col = i / (BK / 8);
+            int kc = (i % (BK / 8)) * 8;
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                int c = n0 + col, k = k0 + kc + j;
+                unsigned short v = 0;
+                if (c < s.C && k < KK)
+                    v = w[(int64_t)k * s.C + c];  // w flat [(ko,r,s), c]
+                sm.Bs[col][kc + j] = v;
+            }
+        }
+        __syncthreads();
+        acc = tile_mfma_step(sm, wave, lane, acc);
+        __syncthreads();
+    }
+    const int col = n0 + (lane & 15);
+    if (col >= s.C) return;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        int row = m0 + wave * 16 + (lane >> 4) * 4 + r;
+        if (row >= M) continue;
+        dx[(int64_t)row * s.C + col] = f2bf(acc[r]);
+    }
+}
+
+// Conv2d wgrad: dw[ko, r, s, c] = sum_pix dy[pix, ko] * x[pix->(ih,iw), c].
+// GEMM: M = Kout, N-dim = R*S*C, K-dim = N*OH*OW; split-K over grid.z with
+// fp32 atomics into dw (zeroed by the host wrapper).
+__global__ void __launch_bounds__(TPB)
+conv_wgrad_kernel(const unsigned short* __restrict__ dy,
+                  const unsigned short* __restrict__ x,
+                  float* __restrict__ dw, ConvShape s, int k_chunks) {
+    __shared__ TileSmem sm;
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+    const int m0 = blockIdx.x * BM;   // over Kout
+    const int n0 = blockIdx.y * BN;   // over RSC
+    const int M = s.Kout;
+    const int NN = s.R * s.S * s.C;
+    const int KK = s.N * s.OH * s.OW;
+    const int chunk = (KK + k_chunks - 1) / k_chunks;
+    const int kbeg = blockIdx.z * chunk;
+    const int kend = min(kbeg + chunk, KK);
+
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    for (int k0 = kbeg; k0 < kend; k0 += BK) {
+        for (int i = tid; i < BM * BK / 8; i += TPB) {
+            int row = i / (BK / 8);
+            int kc = (i % (BK / 8)) * 8;
+            int ko = m0 + row;
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                int k = k0 + kc + j;  // pixel index
+                unsigned short v = 0;
+                if (ko < M && k < kend)
+                    v = dy[(int64_t)k * s.Kout + ko];
+                sm.As[row][kc + j] = v;
+            }
+        }
+        for (int i = tid; i < BN * BK / 8; i += TPB) {
+            int col = i / (BK / 8);
+            int kc = (i % (BK / 8)) * 8;
+            int nn = n0 + col;
+            int c = nn % s.C;
+            int rs = nn / s.C;
+            int r = rs / s.S, ss = rs % s.S;
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                int k = k0 + kc + j;
+                unsigned short v = 0;
+                if (nn < NN && k < kend) {
+                    int n_ = k / (s.OH * s.OW);
+                    int rem = k % (s.OH * s.OW);
+                    int oh = rem / s.OW, ow = rem % s.OW;
+                    int ih = oh * s.stride + r - s.pad;
+                    int iw = ow * s.stride + ss - s.pad;
+                    if (ih >= 0 && ih < s.H && iw >= 0 && iw < s.W)
+                        v = x[(((int64_t)n_ * s.H + ih) * s.W + iw) * s.C + c];
+                }
+                sm.Bs[col][kc + j] = v;
+            }
+        }
+        __syncthreads();
+        acc = tile_mfma_step(sm, wave, lane, acc);
+        __syncthreads();
+    }
+    const int col = n0 + (lane & 15);
+    if (col >= NN) return;
+    // dw layout [Kout, R, S, C] flat = [ko][rsc] with rsc ordered (r,s,c):
+    // col indexes (r,s,c) in that same order.
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        int ko = m0 + wave * 16 + (lane >> 4) * 4 + r;
+        if (ko >= M) continue;
+        if (k_chunks > 1)
+            atomicAdd(dw + (int64_t)ko * NN + col, acc[r]);
+        else
+            dw[(int64_t)ko * NN + col] = acc[r];
+    }
+}
+
+// ---------------------------------------------------------------------------
+// MaxPool 2x2 stride 2 (NHWC bf16), argmax corner saved for backward.
+// ---------------------------------------------------------------------------
+
+__global__ void maxpool_fwd_kernel(const unsigned short* __restrict__ x,
+                                   unsigned short* __restrict__ y,
+                                   uint8_t* __restrict__ idx, int N, int H,
+                                   int W, int C, int OH, int OW) {
+    int64_t total = (int64_t)N * OH * OW * C;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        int c = i % C;
+        int64_t t = i / C;
+        int ow = t % OW;
+        t /= OW;
+        int oh = t % OH;
+        int n = t / OH;
+        int ih = oh * 2, iw = ow * 2;
+        float best = -3.4e38f;
+        int bi = 0;
+#pragma unroll
+        for (int d = 0; d < 4; ++d) {
+            int dh = d >> 1, dw_ = d & 1;
+            float v = bf2f(x[(((int64_t)n * H + ih + dh) * W + iw + dw_) * C + c]);
+            if (v > best) { best = v; bi = d; }
+        }
+        y[i] = f2bf(best);
+        idx[i] = (uint8_t)bi;
+    }
+}
+
+__global__ void maxpool_bwd_kernel(const unsigned short* __restrict__ dy,
+                                   const uint8_t* __restrict__ idx,
+                                   unsigned short* __restrict__ dx, int N,
+                                   int H, int W, int C, int OH, int OW) {
+    int64_t total = (int64_t)N * OH * OW * C;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        int c = i % C;
+        int64_t t = i / C;
+        int ow = t % OW;
+        t /= OW;
+        int oh = t % OH;
+        int n = t / OH;
+        int d = idx[i];
+        int ih = oh * 2 + (d >> 1), iw = ow * 2 + (d & 1);
+        dx[(((int64_t)n * H + ih) * W + iw) * C + c] = dy[i];
+    }
+}
+
+// ---------------------------------------------------------------------------
+// Fused softmax + categorical cross-entropy (mean), one wave per row.
+// logits bf16 [M, C]; probs f32 out; loss = sum(-log p[label]) / M.
+// ---------------------------------------------------------------------------
+
+__global__ void softmax_xent_fwd_kernel(const unsigned short* __restrict__ logits,
+                                        const int64_t* __restrict__ labels,
+                                        float* __restrict__ probs,
+                                        float* __restrict__ loss, int M, int C) {
+    const int row = blockIdx.x * (blockDim.x / 64) + (threadIdx.x >> 6);
+    const int lane = threadIdx.x & 63;
+    if (row >= M) return;
+    const unsigned short* lr = logits + (int64_t)row * C;
+    float mx = -3.4e38f;
+    for (int c = lane; c < C; c += 64) mx = fmaxf(mx, bf2f(lr[c]));
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+        mx = fmaxf(mx, __shfl_xor(mx, off, 64));
+    float sum = 0.f;
+    for (int c = lane; c < C; c += 64) sum += __expf(bf2f(lr[c]) - mx);
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) sum += __shfl_xor(sum, off, 64);
+    float inv = 1.f / sum;
+    for (int c = lane; c < C; c += 64)
+        probs[(int64_t)row * C + c] = __expf(bf2f(lr[c]) - mx) * inv;
+    if (lane == 0) {
+        int64_t lab = labels[row];
+        float p = __expf(bf2f(lr[lab]) - mx) * inv;
+        atomicAdd(loss, -__logf(fmaxf(p, 1e-30f)) / M);
+    }
+}
+
+__global__ void softmax_xent_bwd_kernel(const float* __restrict__ probs,
+                                        const int64_t* __restrict__ labels,
+                                        float* __restrict__ dlogits, int64_t M,
+                                        int C, float scale) {
+    int64_t total = M * C;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        int64_t row = i / C;
+        int c = i % C;
+        float g = probs[i] - (labels[row] == c ? 1.f : 0.f);
+        dlogits[i] = g * scale;
+    }
+}
+
+// ---------------------------------------------------------------------------
+// Fused Adam (fp32 params/grads/states, Keras-decay lr passed pre-computed).
+// ---------------------------------------------------------------------------
+
+__global__ void fused_adam_kernel(float* __restrict__ p,
+                                  const float* __restrict__ g,
+                                  float* __restrict__ m, float* __restrict__ v,
+                                  int64_t total, float lr, float b1, float b2,
+                                  float eps, float bc1, float bc2) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        float gi = g[i];
+        float mi = b1 * m[i] + (1.f - b1) * gi;
+        float vi = b2 * v[i] + (1.f - b2) * gi * gi;
+        m[i] = mi;
+        v[i] = vi;
+        p[i] -= lr * (mi / bc1) / (sqrtf(vi / bc2) + eps);
+    }
+}
+
+// ---------------------------------------------------------------------------
+// ReLU backward + bias grad
+// ---------------------------------------------------------------------------
+
+__global__ void relu_bwd_kernel(const unsigned short* __restrict__ dy,
+                                const unsigned short* __restrict__ y,
+                                unsigned short* __restrict__ dx, int64_t total) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        dx[i] = (y[i] & 0x7fffu) != 0 && !(y[i] & 0x8000u) ? dy[i] : 0;
+    }
+}
+
+// db[k] = sum over rows of dy[., k]; dy bf16 [M, K]; one block per k.
+__global__ void bias_grad_kernel(const unsigned short* __restrict__ dy,
+                                 float* __restrict__ db, int64_t M, int K) {
+    __shared__ float red[256];
+    const int k = blockIdx.x;
+    float acc = 0.f;
+    for (int64_t r = threadIdx.x; r < M; r += blockDim.x)
+        acc += bf2f(dy[r * K + k]);
+    red[threadIdx.x] = acc;
+    __syncthreads();
+    for (int off = 128; off > 0; off >>= 1) {
+        if (threadIdx.x < off) red[threadIdx.x] += red[threadIdx.x + off];
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) db[k] = red[0];
+}
+
+inline int ceildiv(int64_t a, int64_t b) { return (int)((a + b - 1) / b); }
+
+inline ConvShape make_shape(const torch::Tensor& x, const torch::Tensor& w,
+                            int stride, int pad) {
+    ConvShape s;
+    s.N = (int)x.size(0);
+    s.H = (int)x.size(1);
+    s.W = (int)x.size(2);
+    s.C = (int)x.size(3);
+    s.Kout = (int)w.size(0);
+    s.R = (int)w.size(1);
+    s.S = (int)w.size(2);
+    s.stride = stride;
+    s.pad = pad;
+    s.OH = (s.H + 2 * pad - s.R) / stride + 1;
+    s.OW = (s.W + 2 * pad - s.S) / stride + 1;
+    return s;
+}
+
+const unsigned short* bf_ptr(const torch::Tensor& t) {
+    return reinterpret_cast<const unsigned short*>(t.data_ptr<at::BFloat16>());
+}
+unsigned short* bf_ptr_mut(torch::Tensor& t) {
+    return reinterpret_cast<unsigned short*>(t.data_ptr<at::BFloat16>());
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// Host wrappers
+// ---------------------------------------------------------------------------
+
+torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
+                         int64_t stride, bool relu, int64_t pad) {
+    CHECK_GPU(x);
+    TORCH_CHECK(x.is_contiguous() && w.is_contiguous());
+    TORCH_CHECK(x.dtype() == torch::kBFloat16 && w.dtype() == torch::kBFloat16);
+    ConvShape s = make_shape(x, w, (int)stride, (int)pad);
+    auto y = torch::empty({s.N, s.OH, s.OW, s.Kout}, x.options());
+    const int M = s.N * s.OH * s.OW;
+    dim3 grid(ceildiv(M, BM), ceildiv(s.Kout, BN));
+    const float* bias = b.numel() ? b.data_ptr<float>() : nullptr;
+    hipLaunchKernelGGL(conv_fwd_kernel, grid, dim3(TPB), 0,
+                       at::cuda::getCurrentCUDAStream(), bf_ptr(x), bf_ptr(w),
+                       bias, bf_ptr_mut(y), s, relu ? 1 : 0);
+    return y;
+}
+
+torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor w, int64_t stride,
+                           int64_t H, int64_t W, int64_t pad) {
+    CHECK_GPU(dy);
+    TORCH_CHECK(dy.is_contiguous() && w.is_contiguous());
+    const int N = (int)dy.size(0);
+    const int C = (int)w.size(3);
+    ConvShape s;
+    s.N = N; s.H = (int)H; s.W = (int)W; s.C = C;
+    s.Kout = (int)w.size(0); s.R = (int)w.size(1); s.S = (int)w.size(2);
+    s.stride = (int)stride; s.pad = (int)pad;
+    s.OH = (int)dy.size(1); s.OW = (int)dy.size(2);
+    auto dx = torch::empty({N, (int64_t)H, (int64_t)W, C}, dy.options());
+    const int M = N * (int)H * (int)W;
+    dim3 grid(ceildiv(M, BM), ceildiv(C, BN));
+    hipLaunchKernelGGL(conv_dgrad_kernel, grid, dim3(TPB), 0,
+                       at::cuda::getCurrentCUDAStream(), bf_ptr(dy), bf_ptr(w),
+                       bf_ptr_mut(dx), s);
+    return dx;
+}
+
+torch::Tensor conv2d_wgrad(torch::Tensor dy, torch::Tensor x, int64_t stride,
+                           int64_t R, int64_t S, int64_t pad) {
+    CHECK_GPU(dy);
+    TORCH_CHECK(dy.is_contiguous() && x.is_contiguous());
+    ConvShape s;
+    s.N = (int)x.size(0); s.H = (int)x.size(1); s.W = (int)x.size(2);
+    s.C = (int)x.size(3);
+    s.Kout = (int)dy.size(3); s.R = (int)R; s.S = (int)S;
+    s.stride = (int)stride; s.pad = (int)pad;
+    s.OH = (int)dy.size(1); s.OW = (int)dy.size(2);
+    const int NN = s.R * s.S * s.C;
+    const int KK = s.N * s.OH * s.OW;
+    // split-K to fill the chip: target >= 512 blocks
+    int tiles = ceildiv(s.Kout, BM) * ceildiv(NN, BN);
+    int k_chunks = std::max(1, std::min(ceildiv(KK, BK * 8), 512 / std::max(tiles, 1)));
+    auto dw = k_chunks > 1
+                  ? torch::zeros({s.Kout, R, S, s.C},
+                                 x.options().dtype(torch::kFloat32))
+                  : torch::empty({s.Kout, R, S, s.C},
+                                 x.options().dtype(torch::kFloat32));
+    dim3 grid(ceildiv(s.Kout, BM), ceildiv(NN, BN), k_chunks);
+    hipLaunchKernelGGL(conv_wgrad_kernel, grid, dim3(TPB), 0,
+                       at::cuda::getCurrentCUDAStream(), bf_ptr(dy), bf_ptr(x),
+                       dw.data_ptr<float>(), s, k_chunks);
+    return dw;
+}
+
+torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
+                         bool relu) {
+    CHECK_GPU(x);
+    TORCH_CHECK(x.is_contiguous() && w.is_contiguous());
+    const int M = (int)x.size(0), K = (int)x.size(1), N = (int)w.size(0);
+    auto y = torch::empty({M, N}, x.options());
+    dim3 grid(ceildiv(M, BM), ceildiv(N, BN));
+    const float* bias = b.numel() ? b.data_ptr<float>() : nullptr;
+    hipLaunchKernelGGL((gemm_kernel<false, true, true, false>), grid, dim3(TPB),
+                       0, at::cuda::getCurrentCUDAStream(), bf_ptr(x), bf_ptr(w),
+                       bf_ptr_mut(y), bias, M, N, K, relu ? 1 : 0);
+    return y;
+}
+
+torch::Tensor linear_dgrad(torch::Tensor dy, torch::Tensor w) {
+    CHECK_GPU(dy);
+    const int M = (int)dy.size(0), N = (int)dy.size(1), K = (int)w.size(1);
+    auto dx = torch::empty({M, K}, dy.options());
+    dim3 grid(ceildiv(M, BM), ceildiv(K, BN));
+    // dx[M,K] = dy[M,N] @ w[N,K]: A = dy (plain), B = w (plain [N->K])
+    hipLaunchKernelGGL((gemm_kernel<false, false, true, false>), grid, dim3(TPB),
+                       0, at::cuda::getCurrentCUDAStream(), bf_ptr(dy), bf_ptr(w),
+                       bf_ptr_mut(dx), nullptr, M, K, N, 0);
+    return dx;
+}
+
+torch::Tensor linear_wgrad(torch::Tensor dy, torch::Tensor x) {
+    CHECK_GPU(dy);
+    const int M = (int)dy.size(0), N = (int)dy.size(1), K = (int)x.size(1);
+    auto dw = torch::empty({N, K}, dy.options().dtype(torch::kFloat32));
+    dim3 grid(ceildiv(N, BM), ceildiv(K, BN));
+    // dw[N,K] = dy^T[N,M] @ x[M,K]: A transposed (stored [M,N]), B plain
+    hipLaunchKernelGGL((gemm_kernel<true, false, false, false>), grid, dim3(TPB),
+                       0, at::cuda::getCurrentCUDAStream(), bf_ptr(dy), bf_ptr(x),
+                       dw.data_ptr<float>(), nullptr, N, K, M, 0);
+    return dw;
+}
+
+std::vector<torch::Tensor> maxpool2x2_fwd(torch::Tensor x) {
+    CHECK_GPU(x);
+    TORCH_CHECK(x.is_contiguous());
+    const int N = (int)x.size(0), H = (int)x.size(1), W = (int)x.size(2),
+              C = (int)x.size(3);
+    const int OH = H / 2, OW = W / 2;
+    auto y = torch::empty({N, OH, OW, C}, x.options());
+    auto idx = torch::empty({N, OH, OW, C}, x.options().dtype(torch::kUInt8));
+    int64_t total = (int64_t)N * OH * OW * C;
+    int blocks = std::min<int64_t>(ceildiv(total, 256), 2048);
+    hipLaunchKernelGGL(maxpool_fwd_kernel, dim3(blocks), dim3(256), 0,
+                       at::cuda::getCurrentCUDAStream(), bf_ptr(x),
+                       bf_ptr_mut(y), idx.data_ptr<uint8_t>(), N, H, W, C, OH,
+                       OW);
+    return {y, idx};
+}
+
+torch::Tensor maxpool2x2_bwd(torch::Tensor dy, torch::Tensor idx, int64_t H,
+                             int64_t W) {
+    CHECK_GPU(dy);
+    const int N = (int)dy.size(0), OH = (int)dy.size(1), OW = (int)dy.size(2),
+              C = (int)dy.size(3);
+    auto dx = torch::zeros({N, H, W, C}, dy.options());
+    int64_t total = (int64_t)N * OH * OW * C;
+    int blocks = std::min<int64_t>(ceildiv(total, 256), 2048);
+    hipLaunchKernelGGL(maxpool_bwd_kernel, dim3(blocks), dim3(256), 0,
+                       at::cuda::getCurrentCUDAStream(), bf_ptr(dy),
+                       idx.data_ptr<uint8_t>(), bf_ptr_mut(dx), N, (int)H,
+                       (int)W, C, OH, OW);
+    return dx;
+}
+
+std::vector<torch::Tensor> softmax_xent_fwd(torch::Tensor logits,
+                                            torch::Tensor labels) {
+    CHECK_GPU(logits);
+    TORCH_CHECK(logits.is_contiguous());
+    const int M = (int)logits.size(0), C = (int)logits.size(1);
+    auto probs = torch::empty({M, C}, logits.options().dtype(torch::kFloat32));
+    auto loss = torch::zeros({}, logits.options().dtype(torch::kFloat32));
+    const int waves_per_block = 4;
+    int blocks = ceildiv(M, waves_per_block);
+    hipLaunchKernelGGL(softmax_xent_fwd_kernel, dim3(blocks),
+                       dim3(64 * waves_per_block), 0,
+                       at::cuda::getCurrentCUDAStream(), bf_ptr(logits),
+                       labels.data_ptr<int64_t>(), probs.data_ptr<float>(),
+                       loss.data_ptr<float>(), M, C);
+    return {loss, probs};
+}
+
+torch::Tensor softmax_xent_bwd(torch::Tensor probs, torch::Tensor labels,
+                               double scale) {
+    CHECK_GPU(probs);
+    const int64_t M = probs.size(0);
+    const int C = (int)probs.size(1);
+    auto dlogits = torch::empty_like(probs);
+    int64_t total = M * C;
+    int blocks = std::min<int64_t>(ceildiv(total, 256), 2048);
+    hipLaunchKernelGGL(softmax_xent_bwd_kernel, dim3(blocks), dim3(256), 0,
+                       at::cuda::getCurrentCUDAStream(), probs.data_ptr<float>(),
+                       labels.data_ptr<int64_t>(), dlogits.data_ptr<float>(), M,
+                       C, (float)scale);
+    return dlogits;
+}
+
+void fused_adam(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+                torch::Tensor v, double lr, double b1, double b2, double eps,
+                double bc1, double bc2) {
+    CHECK_GPU(p);
+    auto gc = g.contiguous();
+    int64_t total = p.numel();
+    int blocks = std::min<int64_t>(ceildiv(total, 256), 2048);
+    hipLaunchKernelGGL(fused_adam_kernel, dim3(blocks), dim3(256), 0,
+                       at::cuda::getCurrentCUDAStream(), p.data_ptr<float>(),
+                       gc.data_ptr<float>(), m.data_ptr<float>(),
+                       v.data_ptr<float>(), total, (float)lr, (float)b1,
+                       (float)b2, (float)eps, (float)bc1, (float)bc2);
+}
+
+torch::Tensor relu_bwd(torch::Tensor dy, torch::Tensor y) {
+    CHECK_GPU(dy);
+    auto dx = torch::empty_like(dy);
+    int64_t total = dy.numel();
+    int blocks = std::min<int64_t>(ceildiv(total, 256), 2048);
+    hipLaunchKernelGGL(relu_bwd_kernel, dim3(blocks), dim3(256), 0,
+                       at::cuda::getCurrentCUDAStream(), bf_ptr(dy), bf_ptr(y),
+                       bf_ptr_mut(dx), total);
+    return dx;
+}
+
+torch::Tensor bias_grad(torch::Tensor dy) {
+    CHECK_GPU(dy);
+    auto dyc = dy.contiguous();
+    const int K = (int)dyc.size(-1);
+    const int64_t M = dyc.numel() / K;
+    auto db = torch::empty({K}, dyc.options().dtype(torch::kFloat32));
+    hipLaunchKernelGGL(bias_grad_kernel, dim3(K), dim3(256), 0,
+                       at::cuda::getCurrentCUDAStream(), bf_ptr(dyc),
+                       db.data_ptr<float>(), M, K);
+    return db;
+}

@@ -1,0 +1,358 @@
+// Batched negacyclic NTT / INTT over RNS limbs + pointwise modular ops —
+// the keystone HIP kernels of the CKKS layer (NTT row).
+//
+// Replaces SEAL's NTT inside every encrypt/mult the reference reaches via
+// Pyfhel (FLPyfhelin.py:217,295,381,385). Layout matches the CPU oracle
+// hefl/he/ntt_cpu.py exactly: Cooley-Tukey forward / Gentleman-Sande inverse
+// with merged psi powers in bit-reversed table order (tables built in
+// Python, uploaded once per context).
+//
+// Structure per row of length n (n = 2^6 .. 2^15):
+//  * stages whose butterfly span exceeds NBLK run in a strided global-memory
+//    kernel (only n=2^15 needs any on MI355X with NBLK=8192);
+//  * the remaining stages run inside one workgroup with the whole block
+//    resident in LDS (NBLK * 8 B <= 64 KiB of the CU's 160 KiB), one
+//    __syncthreads between stages. Twiddles come from global memory — they
+//    are shared by every row of the batch, so they sit in the XCD L2s.
+//
+// Batch dimension (ciphertexts x limbs) is gridDim.y — HE workloads here
+// launch hundreds of rows, comfortably above the 256-CU fill point.
+
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+#include <algorithm>
+
+#include "modmath.h"
+
+#define CHECK_CUDA_OK(x) TORCH_CHECK((x).is_cuda(), #x " must be on GPU")
+
+namespace {
+
+constexpr int kThreads = 256;
+constexpr int kNblkMax = 8192;  // 64 KiB int64 in LDS
+
+// ---------------------------------------------------------------------------
+// Forward NTT
+// ---------------------------------------------------------------------------
+
+// One global-memory stage: group count m, butterfly stride t = n/(2m) > NBLK/2.
+__global__ void ntt_global_stage_kernel(int64_t* __restrict__ x,
+                                        const int64_t* __restrict__ w,
+                                        const int64_t* __restrict__ wsh,
+                                        uint64_t q, int n, int m) {
+    const int64_t nhalf = n >> 1;
+    const int64_t row = blockIdx.y;
+    int64_t base_off = row * (int64_t)n;
+    const uint32_t t = (uint32_t)(n / (2 * m));
+    for (int64_t k = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; k < nhalf;
+         k += (int64_t)gridDim.x * blockDim.x) {
+        uint32_t j = (uint32_t)(k / t);
+        uint32_t pos = (uint32_t)(k % t);
+        int64_t i0 = base_off + (int64_t)j * 2 * t + pos;
+        uint64_t W = (uint64_t)w[m + j];
+        uint64_t Wsh = (uint64_t)wsh[m + j];
+        uint64_t U = (uint64_t)x[i0];
+        uint64_t V = mulmod_shoup((uint64_t)x[i0 + t], W, Wsh, q);
+        x[i0] = (int64_t)addmod_u64(U, V, q);
+        x[i0 + t] = (int64_t)submod_u64(U, V, q);
+    }
+}
+
+// LDS phase: all stages with butterfly span <= nblk. blockIdx.x = block
+// within row, blockIdx.y = row. m_start = n / nblk (global group count at
+// the first LDS stage).
+__global__ void __launch_bounds__(kThreads)
+ntt_lds_kernel(int64_t* __restrict__ x, const int64_t* __restrict__ w,
+               const int64_t* __restrict__ wsh, uint64_t q, int n, int nblk) {
+    extern __shared__ __attribute__((aligned(16))) int64_t smem[];
+    const int tid = threadIdx.x;
+    const int blk = blockIdx.x;
+    int64_t* xr = x + (int64_t)blockIdx.y * n + (int64_t)blk * nblk;
+    for (int i = tid; i < nblk; i += kThreads) smem[i] = xr[i];
+    __syncthreads();
+    const int nb2 = nblk >> 1;
+    for (int m = n / nblk; m < n; m <<= 1) {
+        const uint32_t t = (uint32_t)(n / (2 * m));  // <= nblk/2 here
+        for (int lb = tid; lb < nb2; lb += kThreads) {
+            uint32_t jloc = (uint32_t)lb / t;
+            uint32_t pos = (uint32_t)lb % t;
+            uint32_t base = jloc * 2 * t + pos;
+            uint32_t jglob = (uint32_t)blk * ((uint32_t)nb2 / t) + jloc;
+            uint64_t W = (uint64_t)w[m + jglob];
+            uint64_t Wsh = (uint64_t)wsh[m + jglob];
+            uint64_t U = (uint64_t)smem[base];
+            uint64_t V = mulmod_shoup((uint64_t)smem[base + t], W, Wsh, q);
+            smem[base] = (int64_t)addmod_u64(U, V, q);
+            smem[base + t] = (int64_t)submod_u64(U, V, q);
+        }
+        __syncthreads();
+    }
+    for (int i = tid; i < nblk; i += kThreads) xr[i] = smem[i];
+}
+
+// ---------------------------------------------------------------------------
+// Inverse NTT (Gentleman-Sande). LDS phase first (small strides), then
+// global stages, final scaling by n^-1 folded into the last stage.
+// ---------------------------------------------------------------------------
+
+__global__ void __launch_bounds__(kThreads)
+intt_lds_kernel(int64_t* __restrict__ x, const int64_t* __restrict__ winv,
+                const int64_t* __restrict__ winvsh, uint64_t q, int n,
+                int nblk, uint64_t ninv, uint64_t ninvsh, int scale_here) {
+    extern __shared__ __attribute__((aligned(16))) int64_t smem[];
+    const int tid = threadIdx.x;
+    const int blk = blockIdx.x;
+    int64_t* xr = x + (int64_t)blockIdx.y * n + (int64_t)blk * nblk;
+    for (int i = tid; i < nblk; i += kThreads) smem[i] = xr[i];
+    __syncthreads();
+    const int nb2 = nblk >> 1;
+    // iterate m = n, n/2, ..., down while 2t <= nblk i.e. m >= 2n/nblk
+    for (int m = n; m >= 2 * (n / nblk); m >>= 1) {
+        const int h = m >> 1;
+        const uint32_t t = (uint32_t)(n / m);
+        for (int lb = tid; lb < nb2; lb += kThreads) {
+            uint32_t jloc = (uint32_t)lb / t;
+            uint32_t pos = (uint32_t)lb % t;
+            uint32_t base = jloc * 2 * t + pos;
+            uint32_t jglob = (uint32_t)blk * ((uint32_t)nb2 / t) + jloc;
+            uint64_t S = (uint64_t)winv[h + jglob];
+            uint64_t Ssh = (uint64_t)winvsh[h + jglob];
+            uint64_t U = (uint64_t)smem[base];
+            uint64_t V = (uint64_t)smem[base + t];
+            smem[base] = (int64_t)addmod_u64(U, V, q);
+            smem[base + t] = (int64_t)mulmod_shoup(submod_u64(U, V, q), S, Ssh, q);
+        }
+        __syncthreads();
+    }
+    if (scale_here) {
+        for (int i = tid; i < nblk; i += kThreads)
+            smem[i] = (int64_t)mulmod_shoup((uint64_t)smem[i], ninv, ninvsh, q);
+        __syncthreads();
+    }
+    for (int i = tid; i < nblk; i += kThreads) xr[i] = smem[i];
+}
+
+__global__ void intt_global_stage_kernel(int64_t* __restrict__ x,
+                                         const int64_t* __restrict__ winv,
+                                         const int64_t* __restrict__ winvsh,
+                                         uint64_t q, int n, int m,
+                                         uint64_t ninv, uint64_t ninvsh,
+                                         int scale_here) {
+    const int64_t nhalf = n >> 1;
+    const int64_t row = blockIdx.y;
+    int64_t base_off = row * (int64_t)n;
+    const int h = m >> 1;
+    const uint32_t t = (uint32_t)(n / m);
+    for (int64_t k = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; k < nhalf;
+         k += (int64_t)gridDim.x * blockDim.x) {
+        uint32_t j = (uint32_t)(k / t);
+        uint32_t pos = (uint32_t)(k % t);
+        int64_t i0 = base_off + (int64_t)j * 2 * t + pos;
+        uint64_t S = (uint64_t)winv[h + j];
+        uint64_t Ssh = (uint64_t)winvsh[h + j];
+        uint64_t U = (uint64_t)x[i0];
+        uint64_t V = (uint64_t)x[i0 + t];
+        uint64_t a = addmod_u64(U, V, q);
+        uint64_t b = mulmod_shoup(submod_u64(U, V, q), S, Ssh, q);
+        if (scale_here) {
+            a = mulmod_shoup(a, ninv, ninvsh, q);
+            b = mulmod_shoup(b, ninv, ninvsh, q);
+        }
+        x[i0] = (int64_t)a;
+        x[i0 + t] = (int64_t)b;
+    }
+}
+
+// ---------------------------------------------------------------------------
+// Pointwise modular ops
+// ---------------------------------------------------------------------------
+
+__global__ void modmul_kernel(const int64_t* __restrict__ a,
+                              const int64_t* __restrict__ b,
+                              int64_t* __restrict__ out, int64_t total,
+                              int64_t b_numel, uint64_t q, uint64_t r0,
+                              uint64_t r1) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        out[i] = (int64_t)mulmod_barrett((uint64_t)a[i],
+                                         (uint64_t)b[i % b_numel], q, r0, r1);
+    }
+}
+
+__global__ void modmul_scalar_kernel(const int64_t* __restrict__ a,
+                                     int64_t* __restrict__ out, int64_t total,
+                                     uint64_t s, uint64_t ssh, uint64_t q) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        out[i] = (int64_t)mulmod_shoup((uint64_t)a[i], s, ssh, q);
+    }
+}
+
+__global__ void modadd_kernel(const int64_t* __restrict__ a,
+                              const int64_t* __restrict__ b,
+                              int64_t* __restrict__ out, int64_t total,
+                              uint64_t q) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        out[i] = (int64_t)addmod_u64((uint64_t)a[i], (uint64_t)b[i], q);
+    }
+}
+
+__global__ void modsub_kernel(const int64_t* __restrict__ a,
+                              const int64_t* __restrict__ b,
+                              int64_t* __restrict__ out, int64_t total,
+                              uint64_t q) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        out[i] = (int64_t)submod_u64((uint64_t)a[i], (uint64_t)b[i], q);
+    }
+}
+
+// Reduce lazily-summed int64 values (< 8q, q < 2^60 so no wrap) back to
+// [0, q) — the step right after the RCCL all-reduce on raw coefficients.
+// x is [..., L, n]; limb selects q via qs[]. In place.
+__global__ void modreduce_kernel(int64_t* __restrict__ x,
+                                 const int64_t* __restrict__ qs, int64_t total,
+                                 int64_t limb_stride, int L) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        uint64_t q = (uint64_t)qs[(i / limb_stride) % L];
+        uint64_t v = (uint64_t)x[i];
+        // v < 8q < 2^63: at most 3 conditional halving steps then compare
+        v %= q;
+        x[i] = (int64_t)v;
+    }
+}
+
+int64_t shoup_of(uint64_t w, uint64_t q) {
+    unsigned __int128 t = ((unsigned __int128)w) << 64;
+    return (int64_t)(uint64_t)(t / q);
+}
+
+inline dim3 rows_grid(int blocks_per_row, int64_t rows) {
+    TORCH_CHECK(rows <= 65535, "batch too large for gridDim.y");
+    return dim3((unsigned)blocks_per_row, (unsigned)rows, 1);
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// Host entry points (registered in bindings.cpp)
+// ---------------------------------------------------------------------------
+
+// In-place forward NTT over x [rows, n] for a single limb prime q.
+void ntt_batch(torch::Tensor x, torch::Tensor w, torch::Tensor wsh, int64_t q) {
+    CHECK_CUDA_OK(x);
+    TORCH_CHECK(x.is_contiguous() && x.dtype() == torch::kInt64);
+    const int n = (int)x.size(-1);
+    const int64_t rows = x.numel() / n;
+    const int nblk = n < kNblkMax ? n : kNblkMax;
+    auto stream = at::cuda::getCurrentCUDAStream();
+    // global stages: m = 1 .. n/nblk/2... while span > nblk (t >= nblk/... )
+    for (int m = 1; m < n / nblk; m <<= 1) {
+        int blocks = (int)std::min<int64_t>((n / 2 + kThreads - 1) / kThreads, 1024);
+        hipLaunchKernelGGL(ntt_global_stage_kernel, rows_grid(blocks, rows),
+                           dim3(kThreads), 0, stream,
+                           x.data_ptr<int64_t>(), w.data_ptr<int64_t>(),
+                           wsh.data_ptr<int64_t>(), (uint64_t)q, n, m);
+    }
+    hipLaunchKernelGGL(ntt_lds_kernel, rows_grid(n / nblk, rows), dim3(kThreads),
+                       nblk * sizeof(int64_t), stream, x.data_ptr<int64_t>(),
+                       w.data_ptr<int64_t>(), wsh.data_ptr<int64_t>(),
+                       (uint64_t)q, n, nblk);
+}
+
+// In-place inverse NTT over x [rows, n].
+void intt_batch(torch::Tensor x, torch::Tensor winv, torch::Tensor winvsh,
+                int64_t q, int64_t ninv, int64_t ninvsh) {
+    CHECK_CUDA_OK(x);
+    TORCH_CHECK(x.is_contiguous() && x.dtype() == torch::kInt64);
+    const int n = (int)x.size(-1);
+    const int64_t rows = x.numel() / n;
+    const int nblk = n < kNblkMax ? n : kNblkMax;
+    auto stream = at::cuda::getCurrentCUDAStream();
+    const bool has_global = nblk < n;
+    hipLaunchKernelGGL(intt_lds_kernel, rows_grid(n / nblk, rows), dim3(kThreads),
+                       nblk * sizeof(int64_t), stream, x.data_ptr<int64_t>(),
+                       winv.data_ptr<int64_t>(), winvsh.data_ptr<int64_t>(),
+                       (uint64_t)q, n, nblk, (uint64_t)ninv, (uint64_t)ninvsh,
+                       has_global ? 0 : 1);
+    // remaining stages: m = n/nblk (after LDS ran m = n .. 2n/nblk) down to 2
+    for (int m = n / nblk; m >= 2; m >>= 1) {
+        int blocks = (int)std::min<int64_t>((n / 2 + kThreads - 1) / kThreads, 1024);
+        hipLaunchKernelGGL(intt_global_stage_kernel, rows_grid(blocks, rows),
+                           dim3(kThreads), 0, stream, x.data_ptr<int64_t>(),
+                           winv.data_ptr<int64_t>(), winvsh.data_ptr<int64_t>(),
+                           (uint64_t)q, n, m, (uint64_t)ninv, (uint64_t)ninvsh,
+                           m == 2 ? 1 : 0);
+    }
+}
+
+torch::Tensor modmul(torch::Tensor a, torch::Tensor b, int64_t q) {
+    CHECK_CUDA_OK(a);
+    TORCH_CHECK(a.is_contiguous() && b.is_contiguous());
+    TORCH_CHECK(a.numel() % b.numel() == 0, "b must tile a");
+    auto out = torch::empty_like(a);
+    uint64_t r0, r1;
+    barrett_ratio((uint64_t)q, &r0, &r1);
+    int64_t total = a.numel();
+    int blocks = (int)std::min<int64_t>((total + kThreads - 1) / kThreads, 2048);
+    hipLaunchKernelGGL(modmul_kernel, dim3(blocks), dim3(kThreads), 0,
+                       at::cuda::getCurrentCUDAStream(), a.data_ptr<int64_t>(),
+                       b.data_ptr<int64_t>(), out.data_ptr<int64_t>(), total,
+                       b.numel(), (uint64_t)q, r0, r1);
+    return out;
+}
+
+torch::Tensor modmul_scalar(torch::Tensor a, int64_t s, int64_t q) {
+    CHECK_CUDA_OK(a);
+    TORCH_CHECK(a.is_contiguous());
+    auto out = torch::empty_like(a);
+    int64_t total = a.numel();
+    int blocks = (int)std::min<int64_t>((total + kThreads - 1) / kThreads, 2048);
+    hipLaunchKernelGGL(modmul_scalar_kernel, dim3(blocks), dim3(kThreads), 0,
+                       at::cuda::getCurrentCUDAStream(), a.data_ptr<int64_t>(),
+                       out.data_ptr<int64_t>(), total, (uint64_t)s,
+                       (uint64_t)shoup_of((uint64_t)s, (uint64_t)q), (uint64_t)q);
+    return out;
+}
+
+torch::Tensor modadd(torch::Tensor a, torch::Tensor b, int64_t q) {
+    CHECK_CUDA_OK(a);
+    auto out = torch::empty_like(a);
+    int64_t total = a.numel();
+    int blocks = (int)std::min<int64_t>((total + kThreads - 1) / kThreads, 2048);
+    hipLaunchKernelGGL(modadd_kernel, dim3(blocks), dim3(kThreads), 0,
+                       at::cuda::getCurrentCUDAStream(), a.data_ptr<int64_t>(),
+                       b.data_ptr<int64_t>(), out.data_ptr<int64_t>(), total,
+                       (uint64_t)q);
+    return out;
+}
+
+torch::Tensor modsub(torch::Tensor a, torch::Tensor b, int64_t q) {
+    CHECK_CUDA_OK(a);
+    auto out = torch::empty_like(a);
+    int64_t total = a.numel();
+    int blocks = (int)std::min<int64_t>((total + kThreads - 1) / kThreads, 2048);
+    hipLaunchKernelGGL(modsub_kernel, dim3(blocks), dim3(kThreads), 0,
+                       at::cuda::getCurrentCUDAStream(), a.data_ptr<int64_t>(),
+                       b.data_ptr<int64_t>(), out.data_ptr<int64_t>(), total,
+                       (uint64_t)q);
+    return out;
+}
+
+// In-place: reduce x [..., L, n] mod per-limb primes qs [L].
+void modreduce_(torch::Tensor x, torch::Tensor qs) {
+    CHECK_CUDA_OK(x);
+    TORCH_CHECK(x.is_contiguous() && qs.is_contiguous());
+    const int L = (int)qs.numel();
+    const int64_t n = x.size(-1);
+    TORCH_CHECK(x.size(-2) == L, "limb dim mismatch");
+    int64_t total = x.numel();
+    int blocks = (int)std::min<int64_t>((total + kThreads - 1) / kThreads, 2048);
+    hipLaunchKernelGGL(modreduce_kernel, dim3(blocks), dim3(kThreads), 0,
+                       at::cuda::getCurrentCUDAStream(), x.data_ptr<int64_t>(),
+                       qs.data_ptr<int64_t>(), total, n, L);
+}

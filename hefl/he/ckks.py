@@ -186,6 +186,12 @@ class CKKSContext:
             self.backend = CpuBackend(self.all_primes, self.n)
         self._cpu_rng = np.random.default_rng(cfg.seed)
 
+    def reseed(self, salt: int) -> None:
+        """Re-seed encryption randomness (e.g. per FL rank) after a shared-seed
+        keygen, so ciphertext noise is independent across clients."""
+        seed = None if self.cfg.seed is None else self.cfg.seed + 1000003 * (salt + 1)
+        self._cpu_rng = np.random.default_rng(seed)
+
     # ----- helpers -----
     def _q(self, limb: int) -> int:
         return self.all_primes[limb]  # index L = the special prime
@@ -259,34 +265,43 @@ class CKKSContext:
         nlimbs = self.L if nlimbs is None else nlimbs
         coeffs = self.encoder.encode(np.asarray(vals, dtype=np.float64), scale)
         out = []
-        for i in range(nlimbs):
-            q = self._q(i)
-            ci = np.mod(coeffs, q)  # object -> [0, q)
-            ci_t = torch.from_numpy(ci.astype(np.int64)).to(self.device)
-            out.append(self.backend.ntt(ci_t, i))
+        if coeffs.dtype == np.int64:
+            # fast path: one upload, per-limb remainder + NTT on device
+            ct = torch.from_numpy(coeffs).to(self.device)
+            for i in range(nlimbs):
+                out.append(self.backend.ntt(torch.remainder(ct, self._q(i)), i))
+        else:
+            for i in range(nlimbs):
+                q = self._q(i)
+                ci = np.mod(coeffs, q)  # object -> [0, q)
+                ci_t = torch.from_numpy(ci.astype(np.int64)).to(self.device)
+                out.append(self.backend.ntt(ci_t, i))
         return Plaintext(torch.stack(out, dim=-2), scale)
 
     def decode(self, pt: Plaintext, k: int) -> np.ndarray:
         data = pt.data
         nlimbs = data.shape[-2]
+        if nlimbs == 1:
+            # fast path (the FedAvg pipeline decrypts at level 1): center on
+            # device, f64 is exact for plaintext magnitudes << 2^53
+            q = self._q(0)
+            c = self.backend.ntt(data[..., 0, :].contiguous(), 0, inverse=True)
+            cent = torch.where(c > q // 2, c - q, c)
+            return self.encoder.decode(cent.cpu().numpy(), pt.scale, k)
+        # exact big-int CRT path (multi-limb decrypts: tests / mul chains)
         coeff_limbs = []
         for i in range(nlimbs):
-            c = self.backend.ntt(data[..., i, :], i, inverse=True)
+            c = self.backend.ntt(data[..., i, :].contiguous(), i, inverse=True)
             coeff_limbs.append(c.cpu().numpy().astype(object))
-        if nlimbs == 1:
-            q = self._q(0)
-            x = coeff_limbs[0]
-            centered = np.where(x > q // 2, x - q, x)
-        else:
-            qs = [self._q(i) for i in range(nlimbs)]
-            Q = math.prod(qs)
-            x = 0
-            for i, q in enumerate(qs):
-                Qi = Q // q
-                hi = pow(Qi % q, -1, q)
-                x = x + coeff_limbs[i] * ((Qi * hi) % Q)
-            x = np.mod(x, Q)
-            centered = np.where(x > Q // 2, x - Q, x)
+        qs = [self._q(i) for i in range(nlimbs)]
+        Q = math.prod(qs)
+        x = 0
+        for i, q in enumerate(qs):
+            Qi = Q // q
+            hi = pow(Qi % q, -1, q)
+            x = x + coeff_limbs[i] * ((Qi * hi) % Q)
+        x = np.mod(x, Q)
+        centered = np.where(x > Q // 2, x - Q, x)
         return self.encoder.decode(centered, pt.scale, k)
 
     # ----- encrypt / decrypt (single ct or batched [..., 2, L, n]) -----

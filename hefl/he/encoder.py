@@ -81,10 +81,13 @@ class Encoder:
             vals = np.concatenate(
                 [vals, np.zeros(vals.shape[:-1] + (pad,))], axis=-1)
         z = self.fft_special_inv(vals)
-        re = np.round(z.real * scale).astype(object)
-        im = np.round(z.imag * scale).astype(object)
-        coeffs = np.concatenate([re, im], axis=-1)  # [..., n]
-        # round() on float64 gives floats; convert to exact ints
+        re = np.round(z.real * scale)
+        im = np.round(z.imag * scale)
+        peak = max(np.abs(re).max(initial=0.0), np.abs(im).max(initial=0.0))
+        if peak < 2.0 ** 52:
+            # int64 fast path (exact: f64 integers below 2^53)
+            return np.concatenate([re, im], axis=-1).astype(np.int64)
+        coeffs = np.concatenate([re.astype(object), im.astype(object)], axis=-1)
         flat = coeffs.reshape(-1)
         for i in range(flat.shape[0]):
             flat[i] = int(flat[i])
@@ -92,7 +95,7 @@ class Encoder:
 
     def decode(self, coeffs: np.ndarray, scale: float, k: int) -> np.ndarray:
         """Centered integer coeffs [..., n] -> real slot values [..., k]."""
-        c = np.asarray(coeffs, dtype=object)
+        c = np.asarray(coeffs)
         half = self.slots
         cf = c.astype(np.float64)
         z = cf[..., :half] + 1j * cf[..., half:]

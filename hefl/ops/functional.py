@@ -33,23 +33,24 @@ def _gpu_dtype(t: torch.Tensor) -> torch.Tensor:
 
 class _Conv2dFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, w, b, stride: int, relu: bool):
+    def forward(ctx, x, w, b, stride: int, relu: bool, pad: int):
         # x: [N,H,W,C] (bf16 on GPU / f32 on CPU); w: [K,R,S,C] f32 param;
         # b: [K] f32 or None
         ctx.stride = stride
         ctx.relu = relu
+        ctx.pad = pad
         ctx.has_bias = b is not None
         if x.is_cuda:
             wb = _gpu_dtype(w.detach())
             bb = b.detach().float() if b is not None else torch.empty(0, device=x.device)
-            y = _C().conv2d_fwd(x, wb, bb, stride, relu)
+            y = _C().conv2d_fwd(x.contiguous(), wb.contiguous(), bb, stride, relu, pad)
             ctx.save_for_backward(x, wb, y)
         else:
             xf = x.float()
             # torch reference: NHWC -> NCHW
             xn = xf.permute(0, 3, 1, 2)
             wn = w.permute(0, 3, 1, 2)  # [K,C,R,S]
-            yn = F.conv2d(xn, wn, b, stride=stride)
+            yn = F.conv2d(xn, wn, b, stride=stride, padding=pad)
             y = yn.permute(0, 2, 3, 1).contiguous()
             if relu:
                 y = F.relu(y)
@@ -59,14 +60,15 @@ class _Conv2dFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         x, w, y = ctx.saved_tensors
-        stride, relu = ctx.stride, ctx.relu
+        stride, relu, pad = ctx.stride, ctx.relu, ctx.pad
         if dy.is_cuda:
             dy = dy.contiguous()
             if relu:
                 dy = _C().relu_bwd(dy, y)
-            dx = _C().conv2d_dgrad(dy, w, stride, x.shape[1], x.shape[2]) \
+            dx = _C().conv2d_dgrad(dy, w, stride, x.shape[1], x.shape[2], pad) \
                 if ctx.needs_input_grad[0] else None
-            dw = _C().conv2d_wgrad(dy, x, stride).to(torch.float32)
+            dw = _C().conv2d_wgrad(dy, x.contiguous(), stride, w.shape[1],
+                                   w.shape[2], pad)
             db = _C().bias_grad(dy) if ctx.has_bias else None
         else:
             dy = dy.float()
@@ -77,16 +79,18 @@ class _Conv2dFn(torch.autograd.Function):
             dyn = dy.permute(0, 3, 1, 2)
             dx = None
             if ctx.needs_input_grad[0]:
-                dxn = torch.nn.grad.conv2d_input(xn.shape, wn, dyn, stride=stride)
+                dxn = torch.nn.grad.conv2d_input(xn.shape, wn, dyn, stride=stride,
+                                                 padding=pad)
                 dx = dxn.permute(0, 2, 3, 1).contiguous()
-            dwn = torch.nn.grad.conv2d_weight(xn, wn.shape, dyn, stride=stride)
+            dwn = torch.nn.grad.conv2d_weight(xn, wn.shape, dyn, stride=stride,
+                                              padding=pad)
             dw = dwn.permute(0, 2, 3, 1).contiguous()  # [K,R,S,C]
             db = dy.sum(dim=(0, 1, 2)) if ctx.has_bias else None
-        return dx, dw, db, None, None
+        return dx, dw, db, None, None, None
 
 
-def conv2d(x, w, b=None, stride: int = 1, relu: bool = False):
-    return _Conv2dFn.apply(x, w, b, stride, relu)
+def conv2d(x, w, b=None, stride: int = 1, relu: bool = False, pad: int = 0):
+    return _Conv2dFn.apply(x, w, b, stride, relu, pad)
 
 
 # ---------------------------------------------------------------------------

@@ -24,19 +24,22 @@ class Conv2dValid(nn.Module):
     """Conv2d, NHWC, 'valid' padding, square kernel/stride, optional fused ReLU."""
 
     def __init__(self, cin: int, cout: int, k: int = 3, stride: int = 1,
-                 relu: bool = True, bias: bool = True, gen=None):
+                 relu: bool = True, bias: bool = True, pad: int = 0, gen=None):
         super().__init__()
         self.cin, self.cout, self.k, self.stride, self.relu = cin, cout, k, stride, relu
+        self.pad = pad
         fan_in, fan_out = cin * k * k, cout * k * k
         w = _glorot_uniform((cout, k, k, cin), fan_in, fan_out, gen)  # [K,R,S,C]
         self.weight = nn.Parameter(w)
         self.bias = nn.Parameter(torch.zeros(cout)) if bias else None
 
     def out_hw(self, h: int, w: int):
-        return ((h - self.k) // self.stride + 1, (w - self.k) // self.stride + 1)
+        return ((h + 2 * self.pad - self.k) // self.stride + 1,
+                (w + 2 * self.pad - self.k) // self.stride + 1)
 
     def forward(self, x):
-        return Fx.conv2d(x, self.weight, self.bias, self.stride, self.relu)
+        return Fx.conv2d(x, self.weight, self.bias, self.stride, self.relu,
+                         self.pad)
 
 
 class MaxPool2x2(nn.Module):

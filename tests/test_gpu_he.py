@@ -1,0 +1,121 @@
+"""GPU numerics tests: HIP CKKS kernels vs the exact CPU oracle.
+
+Every test compares hefl._C kernels against the big-integer CPU reference
+(hefl/he/ntt_cpu.py) or the CPU CKKSContext on identical inputs.
+"""
+import numpy as np
+import pytest
+import torch
+
+import hefl
+from hefl.config import HEConfig
+from hefl.he.ckks import CKKSContext, CtxtTensor
+from hefl.he.ntt_cpu import NttTables, fwd_ntt, inv_ntt
+from hefl.he.primes import gen_prime_chain
+
+pytestmark = pytest.mark.gpu
+
+
+def _tables_to_gpu(q, n):
+    tb = NttTables(q, n)
+    w = torch.from_numpy(tb.w.astype(np.int64)).cuda()
+    winv = torch.from_numpy(tb.winv.astype(np.int64)).cuda()
+
+    def shoup(v):
+        return torch.from_numpy(
+            ((v.cpu().numpy().astype(object) << 64) // q)
+            .astype(np.uint64).astype(np.int64)).cuda()
+
+    ninv = tb.n_inv
+    ninvsh = (ninv << 64) // q
+    return tb, w, shoup(w), winv, shoup(winv), ninv, int(np.int64(np.uint64(ninvsh)))
+
+
+@pytest.mark.parametrize("n", [64, 1024, 8192, 16384, 32768])
+def test_ntt_matches_cpu_oracle(n):
+    C = hefl.load_extension()
+    q = gen_prime_chain(n, (60,))[0]
+    tb, w, wsh, winv, winvsh, ninv, ninvsh = _tables_to_gpu(q, n)
+    rng = np.random.default_rng(0)
+    rows = 3
+    a = rng.integers(0, q, size=(rows, n), dtype=np.int64)
+    ref = fwd_ntt(a.astype(object), tb).astype(np.int64)
+
+    x = torch.from_numpy(a).cuda()
+    C.ntt_batch(x, w, wsh, q)
+    assert np.array_equal(x.cpu().numpy(), ref), "forward NTT mismatch"
+
+    C.intt_batch(x, winv, winvsh, q, ninv, ninvsh)
+    assert np.array_equal(x.cpu().numpy(), a), "inverse NTT roundtrip mismatch"
+
+
+def test_pointwise_modops():
+    C = hefl.load_extension()
+    q = gen_prime_chain(1024, (60,))[0]
+    rng = np.random.default_rng(1)
+    a = rng.integers(0, q, size=(4, 1024), dtype=np.int64)
+    b = rng.integers(0, q, size=(4, 1024), dtype=np.int64)
+    ag, bg = torch.from_numpy(a).cuda(), torch.from_numpy(b).cuda()
+
+    mm = C.modmul(ag, bg, q).cpu().numpy()
+    ref = (a.astype(object) * b.astype(object)) % q
+    assert np.array_equal(mm, ref.astype(np.int64))
+
+    s = int(rng.integers(1, q))
+    ms = C.modmul_scalar(ag, s, q).cpu().numpy()
+    assert np.array_equal(ms, ((a.astype(object) * s) % q).astype(np.int64))
+
+    ad = C.modadd(ag, bg, q).cpu().numpy()
+    assert np.array_equal(ad, ((a.astype(object) + b) % q).astype(np.int64))
+
+    su = C.modsub(ag, bg, q).cpu().numpy()
+    assert np.array_equal(su, ((a.astype(object) - b) % q).astype(np.int64))
+
+
+def test_modreduce_lazy_sum():
+    C = hefl.load_extension()
+    qs_list = gen_prime_chain(256, (60, 40))
+    rng = np.random.default_rng(2)
+    # 8-client lazy sums: values up to 8 * q < 2^63
+    x = np.stack([rng.integers(0, 8 * q, size=(5, 256), dtype=np.int64)
+                  for q in qs_list], axis=1)  # [5, L, 256]
+    ref = np.stack([x[:, i].astype(object) % q
+                    for i, q in enumerate(qs_list)], axis=1).astype(np.int64)
+    xg = torch.from_numpy(x.copy()).cuda()
+    C.modreduce_(xg, torch.tensor(qs_list, dtype=torch.int64).cuda())
+    assert np.array_equal(xg.cpu().numpy(), ref)
+
+
+def test_gpu_context_encrypt_decrypt_matches_cpu():
+    """GPU CKKSContext must round-trip and agree with CPU on decode."""
+    cfg = HEConfig(m=8192, scale_bits=40, q_bits=(60, 40), seed=7)
+    gpu = CKKSContext(cfg, device="cuda")
+    kp = gpu.keygen()
+    vec = torch.randn(10000)
+    ct = gpu.encrypt_tensor(vec, kp.pk)
+    assert ct.data.is_cuda
+    back = gpu.decrypt_tensor(ct, kp.sk)
+    assert (back - vec).abs().max().item() < 1e-3
+
+    # homomorphic FedAvg semantics on GPU
+    ct2 = gpu.encrypt_tensor(vec * 3, kp.pk)
+    lazy = CtxtTensor(ct.data + ct2.data, ct.scale, ct.count)
+    gpu.modreduce_tensor_(lazy)
+    avg = gpu.rescale_tensor(gpu.mul_scalar_tensor(lazy, 0.5))
+    out = gpu.decrypt_tensor(avg, kp.sk)
+    assert (out - 2 * vec).abs().max().item() < 1e-2
+
+
+def test_gpu_vs_cpu_identical_ntt_path():
+    """Same seed => GPU and CPU contexts produce identical ciphertext ints."""
+    cfg = HEConfig(m=1024, scale_bits=30, q_bits=(50, 30), seed=3)
+    cpu = CKKSContext(cfg, device="cpu")
+    gpu = CKKSContext(cfg, device="cuda")
+    kp_c = cpu.keygen()
+    kp_g = gpu.keygen()
+    assert torch.equal(kp_c.sk, kp_g.sk.cpu())
+    assert torch.equal(kp_c.pk, kp_g.pk.cpu())
+    v = np.linspace(-1, 1, cfg.m // 2)
+    pt_c = cpu.encode(v)
+    pt_g = gpu.encode(v)
+    assert torch.equal(pt_c.data, pt_g.data.cpu())

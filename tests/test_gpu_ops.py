@@ -1,0 +1,175 @@
+"""GPU numerics tests: HIP CNN kernels vs plain PyTorch fp32 CPU references.
+
+bf16 compute => tolerances sized to bf16 rounding (rel ~1e-2 on
+accumulated results).
+"""
+import pytest
+import torch
+import torch.nn.functional as F
+
+from hefl.ops import functional as Fx
+from hefl.ops.adam import FusedAdam
+
+pytestmark = pytest.mark.gpu
+
+
+def _close(a, b, atol, rtol=2e-2):
+    a = a.detach().float().cpu()
+    b = b.detach().float().cpu()
+    ok = torch.isclose(a, b, atol=atol, rtol=rtol)
+    frac = ok.float().mean().item()
+    assert frac > 0.999, (
+        f"mismatch {1-frac:.4%}, max abs err "
+        f"{(a-b).abs().max().item():.4g} vs scale {b.abs().max().item():.4g}")
+
+
+@pytest.mark.parametrize("shape", [
+    # (N, H, W, C, K, ksize, stride, pad)
+    (4, 28, 28, 1, 16, 3, 1, 0),     # cnn2 conv1
+    (4, 13, 13, 16, 32, 3, 1, 0),    # cnn2 conv2
+    (2, 32, 32, 3, 6, 5, 1, 0),      # lenet5 conv1
+    (2, 16, 16, 32, 64, 3, 2, 1),    # resnet-style strided+padded
+])
+def test_conv2d_fwd_bwd(shape):
+    N, H, W, C, K, ks, stride, pad = shape
+    torch.manual_seed(0)
+    x = torch.randn(N, H, W, C)
+    w = torch.randn(K, ks, ks, C) * 0.2
+    b = torch.randn(K) * 0.1
+
+    # CPU fp32 reference through the same autograd op
+    xc = x.clone().requires_grad_(True)
+    wc = w.clone().requires_grad_(True)
+    bc = b.clone().requires_grad_(True)
+    yc = Fx.conv2d(xc, wc, bc, stride=stride, relu=True, pad=pad)
+    g = torch.randn_like(yc)
+    yc.backward(g)
+
+    xg = x.to("cuda", torch.bfloat16).requires_grad_(True)
+    wg = w.cuda().requires_grad_(True)
+    bg = b.cuda().requires_grad_(True)
+    yg = Fx.conv2d(xg, wg, bg, stride=stride, relu=True, pad=pad)
+    yg.backward(g.to("cuda", torch.bfloat16))
+
+    _close(yg, yc, atol=0.05)
+    _close(wg.grad, wc.grad, atol=0.3)
+    _close(bg.grad, bc.grad, atol=0.3)
+    _close(xg.grad, xc.grad, atol=0.1)
+
+
+def test_linear_fwd_bwd():
+    torch.manual_seed(1)
+    M, K, N = 32, 800, 64
+    x = torch.randn(M, K)
+    w = torch.randn(N, K) * 0.05
+    b = torch.randn(N) * 0.1
+    xc = x.clone().requires_grad_(True)
+    wc = w.clone().requires_grad_(True)
+    bc = b.clone().requires_grad_(True)
+    yc = Fx.linear(xc, wc, bc, relu=True)
+    g = torch.randn_like(yc)
+    yc.backward(g)
+
+    xg = x.to("cuda", torch.bfloat16).requires_grad_(True)
+    wg = w.cuda().requires_grad_(True)
+    bg = b.cuda().requires_grad_(True)
+    yg = Fx.linear(xg, wg, bg, relu=True)
+    yg.backward(g.to("cuda", torch.bfloat16))
+    _close(yg, yc, atol=0.1)
+    _close(wg.grad, wc.grad, atol=0.3)
+    _close(xg.grad, xc.grad, atol=0.1)
+    _close(bg.grad, bc.grad, atol=0.3)
+
+
+def test_maxpool_gpu():
+    torch.manual_seed(2)
+    x = torch.randn(3, 14, 14, 8)
+    xc = x.clone().requires_grad_(True)
+    yc = Fx.maxpool2x2(xc)
+    g = torch.randn_like(yc)
+    yc.backward(g)
+
+    xg = x.to("cuda", torch.bfloat16).requires_grad_(True)
+    yg = Fx.maxpool2x2(xg)
+    yg.backward(g.to("cuda", torch.bfloat16))
+    _close(yg, yc, atol=0.02)
+    _close(xg.grad, xc.grad, atol=0.02)
+
+
+def test_softmax_xent_gpu():
+    torch.manual_seed(3)
+    M, C = 64, 10
+    logits = torch.randn(M, C) * 3
+    labels = torch.randint(0, C, (M,))
+    lc = logits.clone().requires_grad_(True)
+    loss_c = Fx.softmax_xent(lc, labels)
+    loss_c.backward()
+
+    lg = logits.to("cuda", torch.bfloat16).requires_grad_(True)
+    loss_g = Fx.softmax_xent(lg, labels.cuda())
+    loss_g.backward()
+    assert abs(loss_g.item() - loss_c.item()) < 0.02
+    _close(lg.grad, lc.grad, atol=0.01)
+
+
+def test_fused_adam_gpu():
+    torch.manual_seed(4)
+    p0 = torch.randn(1000)
+    g0 = torch.randn(1000)
+
+    pc = torch.nn.Parameter(p0.clone())
+    oc = FusedAdam([pc], lr=1e-3, decay=1e-4)
+    pg = torch.nn.Parameter(p0.clone().cuda())
+    og = FusedAdam([pg], lr=1e-3, decay=1e-4)
+    for _ in range(5):
+        pc.grad = g0.clone()
+        pg.grad = g0.clone().cuda()
+        oc.step()
+        og.step()
+    _close(pg.data, pc.data, atol=1e-5, rtol=1e-4)
+
+
+def test_model_step_gpu_matches_cpu_direction():
+    """One full fwd+bwd+Adam step of CNN2 on GPU: loss decreases over steps
+    and weights stay close to the CPU fp32 trajectory after 1 step."""
+    from hefl.models import CNN2
+    from hefl.ops.functional import softmax_xent
+
+    torch.manual_seed(5)
+    x = torch.randn(16, 28, 28, 1)
+    y = torch.randint(0, 10, (16,))
+
+    mc = CNN2((28, 28, 1), 10, seed=0)
+    oc = FusedAdam(mc.parameters(), lr=1e-3)
+    lc = softmax_xent(mc(x), y)
+    lc.backward()
+    oc.step()
+
+    mg = CNN2((28, 28, 1), 10, seed=0).cuda()
+    og = FusedAdam(mg.parameters(), lr=1e-3)
+    lg = softmax_xent(mg(x.to("cuda", torch.bfloat16)), y.cuda())
+    lg.backward()
+    og.step()
+
+    assert abs(lg.item() - lc.item()) < 0.05
+    for p_c, p_g in zip(mc.parameters(), mg.parameters()):
+        _close(p_g, p_c, atol=5e-4, rtol=1e-2)
+
+
+def test_training_reduces_loss_gpu():
+    from hefl.models import CNN2
+    from hefl.ops.functional import softmax_xent
+
+    torch.manual_seed(6)
+    m = CNN2((28, 28, 1), 10, seed=1).cuda()
+    opt = FusedAdam(m.parameters(), lr=2e-3)
+    x = torch.randn(64, 28, 28, 1, device="cuda", dtype=torch.bfloat16)
+    y = torch.randint(0, 10, (64,), device="cuda")
+    losses = []
+    for _ in range(30):
+        loss = softmax_xent(m(x), y)
+        opt.zero_grad()
+        loss.backward()
+        opt.step()
+        losses.append(loss.item())
+    assert losses[-1] < losses[0] * 0.5, losses[::5]
