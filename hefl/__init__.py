@@ -35,8 +35,10 @@ def load_extension():
     if _C is not None:
         return _C
     try:
-        from . import _C as _Cmod  # built in-tree by setup.py build_ext --inplace
-        _C = _Cmod
+        # NOTE: must be importlib, not `from . import _C` — the package-level
+        # `_C = None` attribute above would shadow the submodule import.
+        import importlib
+        _C = importlib.import_module("hefl._C")
     except ImportError as e:  # pragma: no cover
         _C_IMPORT_ERROR = e
         raise RuntimeError(
