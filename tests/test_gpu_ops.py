@@ -13,14 +13,15 @@ from hefl.ops.adam import FusedAdam
 pytestmark = pytest.mark.gpu
 
 
-def _close(a, b, atol, rtol=2e-2):
+def _close(a, b, atol=None, rel=3e-2):
+    """Max-norm check scaled to the reference tensor: bf16-rounded inputs
+    accumulated in fp32 stay within a few percent of the fp32 reference."""
     a = a.detach().float().cpu()
     b = b.detach().float().cpu()
-    ok = torch.isclose(a, b, atol=atol, rtol=rtol)
-    frac = ok.float().mean().item()
-    assert frac > 0.999, (
-        f"mismatch {1-frac:.4%}, max abs err "
-        f"{(a-b).abs().max().item():.4g} vs scale {b.abs().max().item():.4g}")
+    scale = b.abs().max().item() + 1e-8
+    err = (a - b).abs().max().item()
+    bound = max(atol if atol is not None else 0.0, rel * scale)
+    assert err < bound, f"max abs err {err:.4g} vs scale {scale:.4g} (bound {bound:.4g})"
 
 
 @pytest.mark.parametrize("shape", [
@@ -51,10 +52,10 @@ def test_conv2d_fwd_bwd(shape):
     yg = Fx.conv2d(xg, wg, bg, stride=stride, relu=True, pad=pad)
     yg.backward(g.to("cuda", torch.bfloat16))
 
-    _close(yg, yc, atol=0.05)
-    _close(wg.grad, wc.grad, atol=0.3)
-    _close(bg.grad, bc.grad, atol=0.3)
-    _close(xg.grad, xc.grad, atol=0.1)
+    _close(yg, yc)
+    _close(wg.grad, wc.grad, rel=5e-2)
+    _close(bg.grad, bc.grad, rel=5e-2)
+    _close(xg.grad, xc.grad, rel=5e-2)
 
 
 def test_linear_fwd_bwd():
@@ -75,10 +76,10 @@ def test_linear_fwd_bwd():
     bg = b.cuda().requires_grad_(True)
     yg = Fx.linear(xg, wg, bg, relu=True)
     yg.backward(g.to("cuda", torch.bfloat16))
-    _close(yg, yc, atol=0.1)
-    _close(wg.grad, wc.grad, atol=0.3)
-    _close(xg.grad, xc.grad, atol=0.1)
-    _close(bg.grad, bc.grad, atol=0.3)
+    _close(yg, yc)
+    _close(wg.grad, wc.grad, rel=5e-2)
+    _close(xg.grad, xc.grad, rel=5e-2)
+    _close(bg.grad, bc.grad, rel=5e-2)
 
 
 def test_maxpool_gpu():
@@ -92,8 +93,8 @@ def test_maxpool_gpu():
     xg = x.to("cuda", torch.bfloat16).requires_grad_(True)
     yg = Fx.maxpool2x2(xg)
     yg.backward(g.to("cuda", torch.bfloat16))
-    _close(yg, yc, atol=0.02)
-    _close(xg.grad, xc.grad, atol=0.02)
+    _close(yg, yc)
+    _close(xg.grad, xc.grad)
 
 
 def test_softmax_xent_gpu():
@@ -109,7 +110,7 @@ def test_softmax_xent_gpu():
     loss_g = Fx.softmax_xent(lg, labels.cuda())
     loss_g.backward()
     assert abs(loss_g.item() - loss_c.item()) < 0.02
-    _close(lg.grad, lc.grad, atol=0.01)
+    _close(lg.grad, lc.grad)
 
 
 def test_fused_adam_gpu():
@@ -126,7 +127,7 @@ def test_fused_adam_gpu():
         pg.grad = g0.clone().cuda()
         oc.step()
         og.step()
-    _close(pg.data, pc.data, atol=1e-5, rtol=1e-4)
+    _close(pg.data, pc.data, rel=1e-4)
 
 
 def test_model_step_gpu_matches_cpu_direction():
@@ -153,7 +154,7 @@ def test_model_step_gpu_matches_cpu_direction():
 
     assert abs(lg.item() - lc.item()) < 0.05
     for p_c, p_g in zip(mc.parameters(), mg.parameters()):
-        _close(p_g, p_c, atol=5e-4, rtol=1e-2)
+        _close(p_g, p_c, atol=3e-3)
 
 
 def test_training_reduces_loss_gpu():
