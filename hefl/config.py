@@ -61,6 +61,7 @@ class TrainConfig:
     beta2: float = 0.999
     eps: float = 1e-7                     # Keras Adam default epsilon
     dtype: str = "bf16"                   # compute dtype on GPU; fp32 on CPU
+    hip_graphs: bool = True               # capture the train step in a hipGraph (GPU)
 
 
 @dataclass

@@ -35,6 +35,11 @@ torch::Tensor softmax_xent_bwd(torch::Tensor probs, torch::Tensor labels,
 void fused_adam(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                 torch::Tensor v, double lr, double b1, double b2, double eps,
                 double bc1, double bc2);
+void adam_prep(torch::Tensor step, torch::Tensor sched, double lr,
+               double decay, double b1, double b2);
+void fused_adam_sched(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+                      torch::Tensor v, torch::Tensor sched, double b1,
+                      double b2, double eps);
 torch::Tensor relu_bwd(torch::Tensor dy, torch::Tensor y);
 torch::Tensor bias_grad(torch::Tensor dy);
 
@@ -60,6 +65,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("softmax_xent_fwd", &softmax_xent_fwd);
     m.def("softmax_xent_bwd", &softmax_xent_bwd);
     m.def("fused_adam", &fused_adam);
+    m.def("adam_prep", &adam_prep);
+    m.def("fused_adam_sched", &fused_adam_sched);
     m.def("relu_bwd", &relu_bwd);
     m.def("bias_grad", &bias_grad);
 }

@@ -485,6 +485,38 @@ __global__ void fused_adam_kernel(float* __restrict__ p,
     }
 }
 
+// hipGraph-capturable Adam: the per-step schedule {lr_t, bc1, bc2} lives in a
+// device buffer advanced by adam_prep_kernel INSIDE the graph — replays need
+// no host-side scalar updates.
+__global__ void adam_prep_kernel(int64_t* __restrict__ step,
+                                 float* __restrict__ sched, float lr,
+                                 float decay, float b1, float b2) {
+    if (threadIdx.x == 0 && blockIdx.x == 0) {
+        int64_t t = ++step[0];
+        sched[0] = lr / (1.f + decay * (float)(t - 1));  // Keras decay schedule
+        sched[1] = 1.f - powf(b1, (float)t);
+        sched[2] = 1.f - powf(b2, (float)t);
+    }
+}
+
+__global__ void fused_adam_sched_kernel(float* __restrict__ p,
+                                        const float* __restrict__ g,
+                                        float* __restrict__ m,
+                                        float* __restrict__ v, int64_t total,
+                                        const float* __restrict__ sched,
+                                        float b1, float b2, float eps) {
+    const float lr = sched[0], bc1 = sched[1], bc2 = sched[2];
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        float gi = g[i];
+        float mi = b1 * m[i] + (1.f - b1) * gi;
+        float vi = b2 * v[i] + (1.f - b2) * gi * gi;
+        m[i] = mi;
+        v[i] = vi;
+        p[i] -= lr * (mi / bc1) / (sqrtf(vi / bc2) + eps);
+    }
+}
+
 // ---------------------------------------------------------------------------
 // ReLU backward + bias grad
 // ---------------------------------------------------------------------------
@@ -724,6 +756,29 @@ void fused_adam(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                        gc.data_ptr<float>(), m.data_ptr<float>(),
                        v.data_ptr<float>(), total, (float)lr, (float)b1,
                        (float)b2, (float)eps, (float)bc1, (float)bc2);
+}
+
+void adam_prep(torch::Tensor step, torch::Tensor sched, double lr,
+               double decay, double b1, double b2) {
+    CHECK_GPU(step);
+    hipLaunchKernelGGL(adam_prep_kernel, dim3(1), dim3(64), 0,
+                       at::cuda::getCurrentCUDAStream(),
+                       step.data_ptr<int64_t>(), sched.data_ptr<float>(),
+                       (float)lr, (float)decay, (float)b1, (float)b2);
+}
+
+void fused_adam_sched(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+                      torch::Tensor v, torch::Tensor sched, double b1,
+                      double b2, double eps) {
+    CHECK_GPU(p);
+    auto gc = g.contiguous();
+    int64_t total = p.numel();
+    int blocks = std::min<int64_t>(ceildiv(total, 256), 2048);
+    hipLaunchKernelGGL(fused_adam_sched_kernel, dim3(blocks), dim3(256), 0,
+                       at::cuda::getCurrentCUDAStream(), p.data_ptr<float>(),
+                       gc.data_ptr<float>(), m.data_ptr<float>(),
+                       v.data_ptr<float>(), total, sched.data_ptr<float>(),
+                       (float)b1, (float)b2, (float)eps);
 }
 
 torch::Tensor relu_bwd(torch::Tensor dy, torch::Tensor y) {

@@ -47,10 +47,18 @@ class SyntheticMedicalImages:
         idx = indices.to(self.device)
         y = self.labels[idx]
         n = idx.numel()
-        g = torch.Generator(device="cpu").manual_seed(self.seed)
         # Noise is drawn deterministically per call position rather than per
         # index (cheap); the class template carries the learnable signal.
-        noise = torch.randn(n, self.H, self.W, self.C, generator=g).to(self.device)
+        # Sampled directly on the device — the training loop never touches host.
+        if self.device.type == "cuda":
+            if not hasattr(self, "_gen"):
+                self._gen = torch.Generator(device=self.device)
+                self._gen.manual_seed(self.seed)
+            noise = torch.randn(n, self.H, self.W, self.C, generator=self._gen,
+                                device=self.device)
+        else:
+            g = torch.Generator(device="cpu").manual_seed(self.seed)
+            noise = torch.randn(n, self.H, self.W, self.C, generator=g)
         x = 0.6 * self.templates[y] + 0.4 * noise
         x = torch.sigmoid(x)  # [0, 1] like rescale=1/255 images
         return x.to(self.dtype), y

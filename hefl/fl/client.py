@@ -46,6 +46,9 @@ class LocalClient:
                              beta1=t.beta1, beta2=t.beta2, eps=t.eps)
         self.compute_dtype = (torch.bfloat16 if (self.device.type == "cuda"
                               and t.dtype == "bf16") else torch.float32)
+        self.use_graphs = (self.device.type == "cuda"
+                           and getattr(t, "hip_graphs", True))
+        self._graphs = {}
         # Data: the full federated dataset is conceptually
         # n_clients * samples_per_client samples; this client materializes its
         # contiguous shard (reference sharding semantics, FLPyfhelin.py:75-78).
@@ -58,30 +61,84 @@ class LocalClient:
                                    seed=cfg.fl.seed + client_id)
 
     def train_step(self, x: torch.Tensor, y: torch.Tensor):
+        if self.use_graphs:
+            return self._graphed_step(x, y)
+        return self._eager_step(x, y)
+
+    def _eager_step(self, x: torch.Tensor, y: torch.Tensor):
         x = x.to(self.compute_dtype)
         logits = self.model(x)
         loss = softmax_xent(logits, y)
-        self.opt.zero_grad()
+        self.opt.zero_grad_() if self.use_graphs else self.opt.zero_grad()
         loss.backward()
         self.opt.step()
         return loss, logits
 
+    # ----- hipGraph-captured training step (MI355X): one graph replay per
+    # step instead of ~40 eager kernel launches + Python autograd overhead.
+    # One graph per batch shape (the last shard batch may be partial). -----
+    def _graphed_step(self, x: torch.Tensor, y: torch.Tensor):
+        key = tuple(x.shape)
+        ent = self._graphs.get(key)
+        if ent is None:
+            ent = self._capture(x, y)
+            self._graphs[key] = ent
+        ent["x"].copy_(x)
+        ent["y"].copy_(y)
+        ent["graph"].replay()
+        return ent["loss"], ent["logits"]
+
+    def _capture(self, x: torch.Tensor, y: torch.Tensor):
+        sx, sy = x.clone(), y.clone()
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(3):  # warmup (real steps; grads stay materialized)
+                self._eager_warmup(sx, sy)
+        torch.cuda.current_stream().wait_stream(side)
+        # device-side Adam schedule buffers, created OUTSIDE the graph and
+        # seeded with the warmup step count
+        dev = sx.device
+        self.opt._step_t = torch.tensor([self.opt.step_count],
+                                        dtype=torch.int64, device=dev)
+        self.opt._sched = torch.zeros(3, dtype=torch.float32, device=dev)
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            self.opt.zero_grad_()
+            logits = self.model(sx.to(self.compute_dtype))
+            loss = softmax_xent(logits, sy)
+            loss.backward()
+            self.opt.step_graphed()
+        return {"graph": g, "x": sx, "y": sy, "loss": loss, "logits": logits}
+
+    def _eager_warmup(self, x, y):
+        logits = self.model(x.to(self.compute_dtype))
+        loss = softmax_xent(logits, y)
+        self.opt.zero_grad_()  # in place; first call grads are None -> no-op
+        loss.backward()
+        self.opt.step()
+
     def local_train(self, epochs: Optional[int] = None) -> RoundStats:
         epochs = self.cfg.train.local_epochs if epochs is None else epochs
         stats = RoundStats()
+        dev = self.device
+        loss_sum = torch.zeros((), dtype=torch.float32, device=dev)
+        acc_sum = torch.zeros((), dtype=torch.float32, device=dev)
         t0 = time.perf_counter()
         for _ in range(epochs):
             for x, y in self.loader:
                 loss, logits = self.train_step(x, y)
                 stats.steps += 1
                 stats.samples += y.numel()
-                stats.train_loss += float(loss.detach())
-                stats.train_acc += float((logits.detach().float().argmax(-1) == y)
-                                         .float().sum())
+                # device-side accumulation: no per-step host sync
+                loss_sum += loss.detach().float()
+                acc_sum += (logits.detach().float().argmax(-1) == y).float().sum()
+        if dev.type == "cuda":
+            torch.cuda.synchronize()
         stats.seconds = time.perf_counter() - t0
         if stats.steps:
-            stats.train_loss /= stats.steps
-            stats.train_acc /= max(stats.samples, 1)
+            stats.train_loss = float(loss_sum) / stats.steps
+            stats.train_acc = float(acc_sum) / max(stats.samples, 1)
         return stats
 
     def get_weights(self) -> torch.Tensor:

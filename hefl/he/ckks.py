@@ -191,6 +191,20 @@ class CKKSContext:
         keygen, so ciphertext noise is independent across clients."""
         seed = None if self.cfg.seed is None else self.cfg.seed + 1000003 * (salt + 1)
         self._cpu_rng = np.random.default_rng(seed)
+        if self.device.type == "cuda":
+            g = torch.Generator(device=self.device)
+            if seed is not None:
+                g.manual_seed(seed)
+            self._gpu_gen = g
+
+    @property
+    def gpu_gen(self):
+        if getattr(self, "_gpu_gen", None) is None:
+            g = torch.Generator(device=self.device)
+            if self.cfg.seed is not None:
+                g.manual_seed(self.cfg.seed)
+            self._gpu_gen = g
+        return self._gpu_gen
 
     # ----- helpers -----
     def _q(self, limb: int) -> int:
@@ -211,13 +225,27 @@ class CKKSContext:
         return torch.remainder(a - b, qs.view(shape))
 
     # ----- sampling (coefficient domain) -----
-    def _sample_ternary(self, shape) -> torch.Tensor:
+    # Keygen always samples on the host RNG (deterministic across CPU/GPU
+    # contexts with the same seed — all FL ranks must derive the SAME keys).
+    # Per-message encryption noise samples on the DEVICE RNG when on GPU:
+    # host sampling of u/e0/e1 for a packed weight tensor costs tens of ms
+    # of numpy + transfer, device sampling is microseconds.
+    def _sample_ternary(self, shape, host: bool = False) -> torch.Tensor:
         # {-1, 0, 1} uniform (hamming-weight variant not needed at these n)
+        if not host and self.device.type == "cuda":
+            return torch.randint(-1, 2, tuple(shape), generator=self.gpu_gen,
+                                 device=self.device, dtype=torch.int64)
         v = torch.from_numpy(self._cpu_rng.integers(-1, 2, size=shape))
         return v.to(torch.int64)
 
-    def _sample_err(self, shape, eta: int = 21) -> torch.Tensor:
+    def _sample_err(self, shape, eta: int = 21, host: bool = False) -> torch.Tensor:
         # centered binomial, sigma = sqrt(eta/2) ~= 3.24 (SEAL sigma 3.2)
+        if not host and self.device.type == "cuda":
+            b = torch.randint(0, 2, (2 * eta,) + tuple(shape),
+                              generator=self.gpu_gen, device=self.device,
+                              dtype=torch.int8)
+            return (b[:eta].sum(0, dtype=torch.int64)
+                    - b[eta:].sum(0, dtype=torch.int64))
         b = self._cpu_rng.integers(0, 2, size=(eta,) + tuple(shape)).sum(axis=0)
         b2 = self._cpu_rng.integers(0, 2, size=(eta,) + tuple(shape)).sum(axis=0)
         return torch.from_numpy(b - b2).to(torch.int64)
@@ -245,8 +273,8 @@ class CKKSContext:
     # ----- keys -----
     def keygen(self) -> KeyPair:
         n = self.n
-        s = self._sample_ternary((n,))
-        e = self._sample_err((n,))
+        s = self._sample_ternary((n,), host=True)
+        e = self._sample_err((n,), host=True)
         sk = self._to_rns_ntt(s)                       # [L, n]
         ehat = self._to_rns_ntt(e)
         a = self._sample_uniform((self.L, n)).to(self.device)   # NTT-domain uniform
@@ -458,7 +486,7 @@ class CKKSContext:
         rlk = torch.empty((L, 2, Lp, n), dtype=torch.int64, device=self.device)
         for d in range(L):
             a = self._sample_uniform((Lp, n)).to(self.device)
-            e = self._to_rns_ntt(self._sample_err((n,)), Lp)
+            e = self._to_rns_ntt(self._sample_err((n,), host=True), Lp)
             for i in range(Lp):
                 q = self._q(i)
                 b = torch.remainder(-(self.backend.modmul(a[i], sk_ext[i], i)

@@ -49,9 +49,31 @@ class FusedAdam:
                 vh = v / bc2
                 p.data.add_(-lr_t * mh / (vh.sqrt() + self.eps))
 
+    @torch.no_grad()
+    def step_graphed(self):
+        """hipGraph-capturable step: the lr/bias-correction schedule advances
+        in a device buffer (adam_prep kernel), so a captured step replays with
+        zero host work. Call only on GPU params with materialized .grad."""
+        C = hefl.load_extension()
+        if not hasattr(self, "_sched"):
+            dev = self.params[0].device
+            self._step_t = torch.zeros(1, dtype=torch.int64, device=dev)
+            self._sched = torch.zeros(3, dtype=torch.float32, device=dev)
+        C.adam_prep(self._step_t, self._sched, self.lr, self.decay,
+                    self.beta1, self.beta2)
+        for p, m, v in zip(self.params, self.m, self.v):
+            C.fused_adam_sched(p.data, p.grad, m, v, self._sched,
+                               self.beta1, self.beta2, self.eps)
+
     def zero_grad(self):
         for p in self.params:
             p.grad = None
+
+    def zero_grad_(self):
+        """In-place grad zeroing (graph-capture safe: keeps buffers alive)."""
+        for p in self.params:
+            if p.grad is not None:
+                p.grad.zero_()
 
     def state_dict(self):
         return {"step": self.step_count, "m": self.m, "v": self.v,
