@@ -31,7 +31,7 @@ torch::Tensor maxpool2x2_bwd(torch::Tensor dy, torch::Tensor idx, int64_t H,
 std::vector<torch::Tensor> softmax_xent_fwd(torch::Tensor logits,
                                             torch::Tensor labels);
 torch::Tensor softmax_xent_bwd(torch::Tensor probs, torch::Tensor labels,
-                               double scale);
+                               torch::Tensor dloss);
 void fused_adam(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                 torch::Tensor v, double lr, double b1, double b2, double eps,
                 double bc1, double bc2);

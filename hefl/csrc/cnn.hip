@@ -451,10 +451,14 @@ __global__ void softmax_xent_fwd_kernel(const unsigned short* __restrict__ logit
     }
 }
 
+// dloss is a device scalar (graph-capture safe: no host readback of the
+// upstream gradient); scale = dloss / M.
 __global__ void softmax_xent_bwd_kernel(const float* __restrict__ probs,
                                         const int64_t* __restrict__ labels,
+                                        const float* __restrict__ dloss,
                                         float* __restrict__ dlogits, int64_t M,
-                                        int C, float scale) {
+                                        int C) {
+    const float scale = dloss[0] / (float)M;
     int64_t total = M * C;
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
          i += (int64_t)gridDim.x * blockDim.x) {
@@ -730,7 +734,7 @@ std::vector<torch::Tensor> softmax_xent_fwd(torch::Tensor logits,
 }
 
 torch::Tensor softmax_xent_bwd(torch::Tensor probs, torch::Tensor labels,
-                               double scale) {
+                               torch::Tensor dloss) {
     CHECK_GPU(probs);
     const int64_t M = probs.size(0);
     const int C = (int)probs.size(1);
@@ -739,8 +743,8 @@ torch::Tensor softmax_xent_bwd(torch::Tensor probs, torch::Tensor labels,
     int blocks = std::min<int64_t>(ceildiv(total, 256), 2048);
     hipLaunchKernelGGL(softmax_xent_bwd_kernel, dim3(blocks), dim3(256), 0,
                        at::cuda::getCurrentCUDAStream(), probs.data_ptr<float>(),
-                       labels.data_ptr<int64_t>(), dlogits.data_ptr<float>(), M,
-                       C, (float)scale);
+                       labels.data_ptr<int64_t>(), dloss.data_ptr<float>(),
+                       dlogits.data_ptr<float>(), M, C);
     return dlogits;
 }
 

@@ -198,7 +198,11 @@ class _SoftmaxXentFn(torch.autograd.Function):
         probs, labels = ctx.saved_tensors
         M = probs.shape[0]
         if probs.is_cuda:
-            dlogits = _C().softmax_xent_bwd(probs, labels, float(dloss) / M)
+            # dloss stays a device scalar (no host sync: hipGraph-capturable)
+            dl = dloss if torch.is_tensor(dloss) else torch.tensor(
+                float(dloss), device=probs.device)
+            dlogits = _C().softmax_xent_bwd(probs, labels,
+                                            dl.to(probs.device, torch.float32))
             dlogits = dlogits.to(ctx.out_dtype)
         else:
             onehot = F.one_hot(labels, probs.shape[-1]).float()
