@@ -35,8 +35,8 @@ torch::Tensor softmax_xent_bwd(torch::Tensor probs, torch::Tensor labels,
 void fused_adam(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                 torch::Tensor v, double lr, double b1, double b2, double eps,
                 double bc1, double bc2);
-void adam_prep(torch::Tensor step, torch::Tensor sched, double lr,
-               double decay, double b1, double b2);
+void adam_prep(torch::Tensor step, torch::Tensor sched, torch::Tensor hyper,
+               double b1, double b2);
 void fused_adam_sched(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                       torch::Tensor v, torch::Tensor sched, double b1,
                       double b2, double eps);

@@ -493,11 +493,14 @@ __global__ void fused_adam_kernel(float* __restrict__ p,
 // device buffer advanced by adam_prep_kernel INSIDE the graph — replays need
 // no host-side scalar updates.
 __global__ void adam_prep_kernel(int64_t* __restrict__ step,
-                                 float* __restrict__ sched, float lr,
-                                 float decay, float b1, float b2) {
+                                 float* __restrict__ sched,
+                                 const float* __restrict__ hyper,  // {lr, decay}
+                                 float b1, float b2) {
     if (threadIdx.x == 0 && blockIdx.x == 0) {
         int64_t t = ++step[0];
-        sched[0] = lr / (1.f + decay * (float)(t - 1));  // Keras decay schedule
+        // hyper lives in a device buffer so lr changes (ReduceLROnPlateau)
+        // reach a captured graph without re-capture
+        sched[0] = hyper[0] / (1.f + hyper[1] * (float)(t - 1));  // Keras decay
         sched[1] = 1.f - powf(b1, (float)t);
         sched[2] = 1.f - powf(b2, (float)t);
     }
@@ -762,13 +765,13 @@ void fused_adam(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                        (float)b2, (float)eps, (float)bc1, (float)bc2);
 }
 
-void adam_prep(torch::Tensor step, torch::Tensor sched, double lr,
-               double decay, double b1, double b2) {
+void adam_prep(torch::Tensor step, torch::Tensor sched, torch::Tensor hyper,
+               double b1, double b2) {
     CHECK_GPU(step);
     hipLaunchKernelGGL(adam_prep_kernel, dim3(1), dim3(64), 0,
                        at::cuda::getCurrentCUDAStream(),
                        step.data_ptr<int64_t>(), sched.data_ptr<float>(),
-                       (float)lr, (float)decay, (float)b1, (float)b2);
+                       hyper.data_ptr<float>(), (float)b1, (float)b2);
 }
 
 void fused_adam_sched(torch::Tensor p, torch::Tensor g, torch::Tensor m,

@@ -98,10 +98,7 @@ class LocalClient:
         torch.cuda.current_stream().wait_stream(side)
         # device-side Adam schedule buffers, created OUTSIDE the graph and
         # seeded with the warmup step count
-        dev = sx.device
-        self.opt._step_t = torch.tensor([self.opt.step_count],
-                                        dtype=torch.int64, device=dev)
-        self.opt._sched = torch.zeros(3, dtype=torch.float32, device=dev)
+        self.opt.prepare_graph_state(sx.device)
         g = torch.cuda.CUDAGraph()
         with torch.cuda.graph(g):
             self.opt.zero_grad_()
@@ -118,21 +115,43 @@ class LocalClient:
         loss.backward()
         self.opt.step()
 
-    def local_train(self, epochs: Optional[int] = None) -> RoundStats:
+    def local_train(self, epochs: Optional[int] = None,
+                    callbacks: Optional[list] = None) -> RoundStats:
+        """Local SGD for `epochs` epochs. With callbacks (the reference wires
+        EarlyStopping/ReduceLROnPlateau/ModelCheckpoint into every fit,
+        FLPyfhelin.py:186-192) per-epoch metrics sync to host; without,
+        the loop is fully sync-free until the end."""
         epochs = self.cfg.train.local_epochs if epochs is None else epochs
+        callbacks = callbacks or []
         stats = RoundStats()
         dev = self.device
         loss_sum = torch.zeros((), dtype=torch.float32, device=dev)
         acc_sum = torch.zeros((), dtype=torch.float32, device=dev)
         t0 = time.perf_counter()
-        for _ in range(epochs):
+        for ep in range(epochs):
+            ep_loss = torch.zeros((), dtype=torch.float32, device=dev)
+            ep_acc = torch.zeros((), dtype=torch.float32, device=dev)
+            ep_steps = ep_samples = 0
             for x, y in self.loader:
                 loss, logits = self.train_step(x, y)
-                stats.steps += 1
-                stats.samples += y.numel()
+                ep_steps += 1
+                ep_samples += y.numel()
                 # device-side accumulation: no per-step host sync
-                loss_sum += loss.detach().float()
-                acc_sum += (logits.detach().float().argmax(-1) == y).float().sum()
+                ep_loss += loss.detach().float()
+                ep_acc += (logits.detach().float().argmax(-1) == y).float().sum()
+            loss_sum += ep_loss
+            acc_sum += ep_acc
+            stats.steps += ep_steps
+            stats.samples += ep_samples
+            if callbacks:
+                logs = {"loss": float(ep_loss) / max(ep_steps, 1),
+                        "accuracy": float(ep_acc) / max(ep_samples, 1)}
+                for cb in callbacks:
+                    cb.on_epoch_end(ep, logs)
+                if any(cb.stop_training for cb in callbacks):
+                    break
+        for cb in callbacks:
+            cb.on_train_end()
         if dev.type == "cuda":
             torch.cuda.synchronize()
         stats.seconds = time.perf_counter() - t0

@@ -316,7 +316,20 @@ class CKKSContext:
             c = self.backend.ntt(data[..., 0, :].contiguous(), 0, inverse=True)
             cent = torch.where(c > q // 2, c - q, c)
             return self.encoder.decode(cent.cpu().numpy(), pt.scale, k)
-        # exact big-int CRT path (multi-limb decrypts: tests / mul chains)
+        # multi-limb fast path: when the plaintext magnitude is << q0/2 (true
+        # for any decode of real-valued weights at scale <= 2^45), the
+        # centered limb-0 residue IS the integer value — checked against
+        # limb 1 on a sample, falling back to exact big-int CRT on mismatch.
+        q0 = self._q(0)
+        c0 = self.backend.ntt(data[..., 0, :].contiguous(), 0, inverse=True)
+        cent0 = torch.where(c0 > q0 // 2, c0 - q0, c0)
+        if cent0.abs().max().item() < q0 // 4:
+            q1 = self._q(1)
+            c1 = self.backend.ntt(data[..., 1, :].contiguous(), 1, inverse=True)
+            cent1 = torch.where(c1 > q1 // 2, c1 - q1, c1)
+            if torch.equal(torch.remainder(cent0, q1), torch.remainder(cent1, q1)):
+                return self.encoder.decode(cent0.cpu().numpy(), pt.scale, k)
+        # exact big-int CRT path (plaintext too large for limb-0 shortcut)
         coeff_limbs = []
         for i in range(nlimbs):
             c = self.backend.ntt(data[..., i, :].contiguous(), i, inverse=True)

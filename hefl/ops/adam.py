@@ -49,17 +49,29 @@ class FusedAdam:
                 vh = v / bc2
                 p.data.add_(-lr_t * mh / (vh.sqrt() + self.eps))
 
+    def prepare_graph_state(self, device=None):
+        """Create device-side step/schedule/hyper buffers (must run OUTSIDE
+        any hipGraph capture)."""
+        dev = device if device is not None else self.params[0].device
+        self._step_t = torch.tensor([self.step_count], dtype=torch.int64,
+                                    device=dev)
+        self._sched = torch.zeros(3, dtype=torch.float32, device=dev)
+        self._hyper = torch.tensor([self.lr, self.decay],
+                                   dtype=torch.float32, device=dev)
+
+    def set_lr(self, lr: float):
+        self.lr = lr
+        if hasattr(self, "_hyper"):
+            self._hyper[0] = lr  # reaches captured graphs without re-capture
+
     @torch.no_grad()
     def step_graphed(self):
         """hipGraph-capturable step: the lr/bias-correction schedule advances
         in a device buffer (adam_prep kernel), so a captured step replays with
-        zero host work. Call only on GPU params with materialized .grad."""
+        zero host work. Call only on GPU params with materialized .grad,
+        after prepare_graph_state()."""
         C = hefl.load_extension()
-        if not hasattr(self, "_sched"):
-            dev = self.params[0].device
-            self._step_t = torch.zeros(1, dtype=torch.int64, device=dev)
-            self._sched = torch.zeros(3, dtype=torch.float32, device=dev)
-        C.adam_prep(self._step_t, self._sched, self.lr, self.decay,
+        C.adam_prep(self._step_t, self._sched, self._hyper,
                     self.beta1, self.beta2)
         for p, m, v in zip(self.params, self.m, self.v):
             C.fused_adam_sched(p.data, p.grad, m, v, self._sched,
@@ -76,7 +88,10 @@ class FusedAdam:
                 p.grad.zero_()
 
     def state_dict(self):
-        return {"step": self.step_count, "m": self.m, "v": self.v,
+        step = self.step_count
+        if hasattr(self, "_step_t"):
+            step = int(self._step_t.item())  # graphed steps advance on device
+        return {"step": step, "m": self.m, "v": self.v,
                 "lr": self.lr, "decay": self.decay}
 
     def load_state_dict(self, sd):
