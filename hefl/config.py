@@ -71,6 +71,7 @@ class FLConfig:
     val_samples_per_client: int = 80
     test_samples: int = 400
     encrypted: bool = True                # False -> plaintext FedAvg (config #1)
+    denom_mode: str = "plain"             # "encrypted" -> ct x ct + relin in aggregation (config #3)
     seed: int = 1234
 
 
@@ -105,7 +106,7 @@ def preset(name: str) -> RunConfig:
         # 8-client LeNet-5 on 32x32x3, CKKS n=2^14 with homomorphic mult + rescale
         return RunConfig(
             model=ModelConfig("lenet5", (32, 32, 3), 10),
-            fl=FLConfig(n_clients=8, encrypted=True),
+            fl=FLConfig(n_clients=8, encrypted=True, denom_mode="encrypted"),
             he=HEConfig(m=16384, scale_bits=40, q_bits=(60, 40, 40)),
         )
     if name in ("config4", "cnn4-xray"):

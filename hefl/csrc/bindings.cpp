@@ -41,6 +41,21 @@ void fused_adam_sched(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                       torch::Tensor v, torch::Tensor sched, double b1,
                       double b2, double eps);
 torch::Tensor relu_bwd(torch::Tensor dy, torch::Tensor y);
+std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
+                                  torch::Tensor beta, double eps, bool relu);
+torch::Tensor bn_apply(torch::Tensor x, torch::Tensor mean,
+                       torch::Tensor invstd, torch::Tensor gamma,
+                       torch::Tensor beta, bool relu);
+std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
+                                  torch::Tensor mean, torch::Tensor invstd,
+                                  torch::Tensor gamma, bool train);
+std::vector<torch::Tensor> maxpool_fwd(torch::Tensor x, int64_t k, int64_t s,
+                                       int64_t p);
+torch::Tensor maxpool_bwd(torch::Tensor dy, torch::Tensor idx, int64_t H,
+                          int64_t W, int64_t k, int64_t s, int64_t p);
+torch::Tensor avgpool_global_fwd(torch::Tensor x);
+torch::Tensor avgpool_global_bwd(torch::Tensor dy, int64_t H, int64_t W);
+torch::Tensor add_relu(torch::Tensor a, torch::Tensor b);
 torch::Tensor bias_grad(torch::Tensor dy);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -68,5 +83,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("adam_prep", &adam_prep);
     m.def("fused_adam_sched", &fused_adam_sched);
     m.def("relu_bwd", &relu_bwd);
+    m.def("bn_fwd", &bn_fwd);
+    m.def("bn_apply", &bn_apply);
+    m.def("bn_bwd", &bn_bwd);
+    m.def("maxpool_fwd", &maxpool_fwd);
+    m.def("maxpool_bwd", &maxpool_bwd);
+    m.def("avgpool_global_fwd", &avgpool_global_fwd);
+    m.def("avgpool_global_bwd", &avgpool_global_bwd);
+    m.def("add_relu", &add_relu);
     m.def("bias_grad", &bias_grad);
 }

@@ -525,6 +525,217 @@ __global__ void fused_adam_sched_kernel(float* __restrict__ p,
 }
 
 // ---------------------------------------------------------------------------
+// BatchNorm (NHWC, per-channel over N*H*W) — ResNet-18 (config #5) support.
+// ---------------------------------------------------------------------------
+
+// One block per channel: mean + invstd in fp32.
+__global__ void bn_stats_kernel(const unsigned short* __restrict__ x,
+                                float* __restrict__ mean,
+                                float* __restrict__ invstd, int64_t M, int C,
+                                float eps) {
+    __shared__ float s_sum[256], s_sq[256];
+    const int c = blockIdx.x;
+    float acc = 0.f, acc2 = 0.f;
+    for (int64_t r = threadIdx.x; r < M; r += blockDim.x) {
+        float v = bf2f(x[r * C + c]);
+        acc += v;
+        acc2 += v * v;
+    }
+    s_sum[threadIdx.x] = acc;
+    s_sq[threadIdx.x] = acc2;
+    __syncthreads();
+    for (int off = 128; off > 0; off >>= 1) {
+        if (threadIdx.x < off) {
+            s_sum[threadIdx.x] += s_sum[threadIdx.x + off];
+            s_sq[threadIdx.x] += s_sq[threadIdx.x + off];
+        }
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) {
+        float mu = s_sum[0] / (float)M;
+        float var = fmaxf(s_sq[0] / (float)M - mu * mu, 0.f);
+        mean[c] = mu;
+        invstd[c] = rsqrtf(var + eps);
+    }
+}
+
+__global__ void bn_apply_kernel(const unsigned short* __restrict__ x,
+                                const float* __restrict__ mean,
+                                const float* __restrict__ invstd,
+                                const float* __restrict__ gamma,
+                                const float* __restrict__ beta,
+                                unsigned short* __restrict__ y, int64_t total,
+                                int C, int relu) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        int c = i % C;
+        float v = (bf2f(x[i]) - mean[c]) * invstd[c] * gamma[c] + beta[c];
+        if (relu) v = v > 0.f ? v : 0.f;
+        y[i] = f2bf(v);
+    }
+}
+
+// dgamma[c] = sum dy*xhat; dbeta[c] = sum dy. One block per channel.
+__global__ void bn_reduce_bwd_kernel(const unsigned short* __restrict__ dy,
+                                     const unsigned short* __restrict__ x,
+                                     const float* __restrict__ mean,
+                                     const float* __restrict__ invstd,
+                                     float* __restrict__ dgamma,
+                                     float* __restrict__ dbeta, int64_t M,
+                                     int C) {
+    __shared__ float s_dg[256], s_db[256];
+    const int c = blockIdx.x;
+    float dg = 0.f, db = 0.f;
+    for (int64_t r = threadIdx.x; r < M; r += blockDim.x) {
+        float g = bf2f(dy[r * C + c]);
+        float xh = (bf2f(x[r * C + c]) - mean[c]) * invstd[c];
+        dg += g * xh;
+        db += g;
+    }
+    s_dg[threadIdx.x] = dg;
+    s_db[threadIdx.x] = db;
+    __syncthreads();
+    for (int off = 128; off > 0; off >>= 1) {
+        if (threadIdx.x < off) {
+            s_dg[threadIdx.x] += s_dg[threadIdx.x + off];
+            s_db[threadIdx.x] += s_db[threadIdx.x + off];
+        }
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) {
+        dgamma[c] = s_dg[0];
+        dbeta[c] = s_db[0];
+    }
+}
+
+__global__ void bn_dx_kernel(const unsigned short* __restrict__ dy,
+                             const unsigned short* __restrict__ x,
+                             const float* __restrict__ mean,
+                             const float* __restrict__ invstd,
+                             const float* __restrict__ gamma,
+                             const float* __restrict__ dgamma,
+                             const float* __restrict__ dbeta,
+                             unsigned short* __restrict__ dx, int64_t total,
+                             int C, int64_t M, int train) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        int c = i % C;
+        float g = bf2f(dy[i]);
+        float v;
+        if (train) {
+            float xh = (bf2f(x[i]) - mean[c]) * invstd[c];
+            v = gamma[c] * invstd[c] *
+                (g - dbeta[c] / (float)M - xh * dgamma[c] / (float)M);
+        } else {
+            v = gamma[c] * invstd[c] * g;  // frozen stats
+        }
+        dx[i] = f2bf(v);
+    }
+}
+
+// ---------------------------------------------------------------------------
+// Generic MaxPool k x k, stride s, padding p (ResNet stem: 3x3 s2 p1).
+// Overlapping windows => backward accumulates into fp32 with atomics.
+// ---------------------------------------------------------------------------
+
+__global__ void maxpool_gen_fwd_kernel(const unsigned short* __restrict__ x,
+                                       unsigned short* __restrict__ y,
+                                       uint8_t* __restrict__ idx, int N, int H,
+                                       int W, int C, int OH, int OW, int k,
+                                       int s, int p) {
+    int64_t total = (int64_t)N * OH * OW * C;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        int c = i % C;
+        int64_t t = i / C;
+        int ow = t % OW;
+        t /= OW;
+        int oh = t % OH;
+        int n = t / OH;
+        float best = -3.4e38f;
+        int bi = 0;
+        for (int d = 0; d < k * k; ++d) {
+            int ih = oh * s + d / k - p, iw = ow * s + d % k - p;
+            if (ih < 0 || ih >= H || iw < 0 || iw >= W) continue;
+            float v = bf2f(x[(((int64_t)n * H + ih) * W + iw) * C + c]);
+            if (v > best) { best = v; bi = d; }
+        }
+        y[i] = f2bf(best);
+        idx[i] = (uint8_t)bi;
+    }
+}
+
+__global__ void maxpool_gen_bwd_kernel(const unsigned short* __restrict__ dy,
+                                       const uint8_t* __restrict__ idx,
+                                       float* __restrict__ dx32, int N, int H,
+                                       int W, int C, int OH, int OW, int k,
+                                       int s, int p) {
+    int64_t total = (int64_t)N * OH * OW * C;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        int c = i % C;
+        int64_t t = i / C;
+        int ow = t % OW;
+        t /= OW;
+        int oh = t % OH;
+        int n = t / OH;
+        int d = idx[i];
+        int ih = oh * s + d / k - p, iw = ow * s + d % k - p;
+        if (ih < 0 || ih >= H || iw < 0 || iw >= W) continue;
+        atomicAdd(dx32 + (((int64_t)n * H + ih) * W + iw) * C + c,
+                  bf2f(dy[i]));
+    }
+}
+
+// ---------------------------------------------------------------------------
+// Global average pool [N,H,W,C] -> [N,C] + backward broadcast.
+// ---------------------------------------------------------------------------
+
+__global__ void avgpool_global_fwd_kernel(const unsigned short* __restrict__ x,
+                                          unsigned short* __restrict__ y,
+                                          int N, int HW, int C) {
+    int64_t total = (int64_t)N * C;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        int c = i % C;
+        int n = i / C;
+        float acc = 0.f;
+        const unsigned short* base = x + (int64_t)n * HW * C + c;
+        for (int r = 0; r < HW; ++r) acc += bf2f(base[(int64_t)r * C]);
+        y[i] = f2bf(acc / (float)HW);
+    }
+}
+
+__global__ void avgpool_global_bwd_kernel(const unsigned short* __restrict__ dy,
+                                          unsigned short* __restrict__ dx,
+                                          int N, int HW, int C) {
+    int64_t total = (int64_t)N * HW * C;
+    float inv;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        int c = i % C;
+        int64_t t = i / C;
+        int n = t / HW;
+        inv = 1.f / (float)HW;
+        dx[i] = f2bf(bf2f(dy[(int64_t)n * C + c]) * inv);
+    }
+}
+
+// ---------------------------------------------------------------------------
+// Fused residual add + ReLU.
+// ---------------------------------------------------------------------------
+
+__global__ void add_relu_kernel(const unsigned short* __restrict__ a,
+                                const unsigned short* __restrict__ b,
+                                unsigned short* __restrict__ y, int64_t total) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        float v = bf2f(a[i]) + bf2f(b[i]);
+        y[i] = f2bf(v > 0.f ? v : 0.f);
+    }
+}
+
+// ---------------------------------------------------------------------------
 // ReLU backward + bias grad
 // ---------------------------------------------------------------------------
 
@@ -797,6 +1008,143 @@ torch::Tensor relu_bwd(torch::Tensor dy, torch::Tensor y) {
                        at::cuda::getCurrentCUDAStream(), bf_ptr(dy), bf_ptr(y),
                        bf_ptr_mut(dx), total);
     return dx;
+}
+
+std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
+                                  torch::Tensor beta, double eps, bool relu) {
+    CHECK_GPU(x);
+    TORCH_CHECK(x.is_contiguous());
+    const int C = (int)x.size(-1);
+    const int64_t M = x.numel() / C;
+    auto f32 = x.options().dtype(torch::kFloat32);
+    auto mean = torch::empty({C}, f32);
+    auto invstd = torch::empty({C}, f32);
+    auto y = torch::empty_like(x);
+    auto stream = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(bn_stats_kernel, dim3(C), dim3(256), 0, stream,
+                       bf_ptr(x), mean.data_ptr<float>(),
+                       invstd.data_ptr<float>(), M, C, (float)eps);
+    int64_t total = x.numel();
+    int blocks = std::min<int64_t>(ceildiv(total, 256), 2048);
+    hipLaunchKernelGGL(bn_apply_kernel, dim3(blocks), dim3(256), 0, stream,
+                       bf_ptr(x), mean.data_ptr<float>(),
+                       invstd.data_ptr<float>(), gamma.data_ptr<float>(),
+                       beta.data_ptr<float>(), bf_ptr_mut(y), total, C,
+                       relu ? 1 : 0);
+    return {y, mean, invstd};
+}
+
+torch::Tensor bn_apply(torch::Tensor x, torch::Tensor mean,
+                       torch::Tensor invstd, torch::Tensor gamma,
+                       torch::Tensor beta, bool relu) {
+    CHECK_GPU(x);
+    const int C = (int)x.size(-1);
+    auto y = torch::empty_like(x);
+    int64_t total = x.numel();
+    int blocks = std::min<int64_t>(ceildiv(total, 256), 2048);
+    hipLaunchKernelGGL(bn_apply_kernel, dim3(blocks), dim3(256), 0,
+                       at::cuda::getCurrentCUDAStream(), bf_ptr(x),
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                       bf_ptr_mut(y), total, C, relu ? 1 : 0);
+    return y;
+}
+
+std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
+                                  torch::Tensor mean, torch::Tensor invstd,
+                                  torch::Tensor gamma, bool train) {
+    CHECK_GPU(dy);
+    auto dyc = dy.contiguous();
+    const int C = (int)x.size(-1);
+    const int64_t M = x.numel() / C;
+    auto f32 = x.options().dtype(torch::kFloat32);
+    auto dgamma = torch::empty({C}, f32);
+    auto dbeta = torch::empty({C}, f32);
+    auto dx = torch::empty_like(x);
+    auto stream = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(bn_reduce_bwd_kernel, dim3(C), dim3(256), 0, stream,
+                       bf_ptr(dyc), bf_ptr(x), mean.data_ptr<float>(),
+                       invstd.data_ptr<float>(), dgamma.data_ptr<float>(),
+                       dbeta.data_ptr<float>(), M, C);
+    int64_t total = x.numel();
+    int blocks = std::min<int64_t>(ceildiv(total, 256), 2048);
+    hipLaunchKernelGGL(bn_dx_kernel, dim3(blocks), dim3(256), 0, stream,
+                       bf_ptr(dyc), bf_ptr(x), mean.data_ptr<float>(),
+                       invstd.data_ptr<float>(), gamma.data_ptr<float>(),
+                       dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
+                       bf_ptr_mut(dx), total, C, M, train ? 1 : 0);
+    return {dx, dgamma, dbeta};
+}
+
+std::vector<torch::Tensor> maxpool_fwd(torch::Tensor x, int64_t k, int64_t s,
+                                       int64_t p) {
+    CHECK_GPU(x);
+    TORCH_CHECK(x.is_contiguous());
+    const int N = (int)x.size(0), H = (int)x.size(1), W = (int)x.size(2),
+              C = (int)x.size(3);
+    const int OH = (H + 2 * (int)p - (int)k) / (int)s + 1;
+    const int OW = (W + 2 * (int)p - (int)k) / (int)s + 1;
+    auto y = torch::empty({N, OH, OW, C}, x.options());
+    auto idx = torch::empty({N, OH, OW, C}, x.options().dtype(torch::kUInt8));
+    int64_t total = (int64_t)N * OH * OW * C;
+    int blocks = std::min<int64_t>(ceildiv(total, 256), 2048);
+    hipLaunchKernelGGL(maxpool_gen_fwd_kernel, dim3(blocks), dim3(256), 0,
+                       at::cuda::getCurrentCUDAStream(), bf_ptr(x),
+                       bf_ptr_mut(y), idx.data_ptr<uint8_t>(), N, H, W, C, OH,
+                       OW, (int)k, (int)s, (int)p);
+    return {y, idx};
+}
+
+torch::Tensor maxpool_bwd(torch::Tensor dy, torch::Tensor idx, int64_t H,
+                          int64_t W, int64_t k, int64_t s, int64_t p) {
+    CHECK_GPU(dy);
+    const int N = (int)dy.size(0), OH = (int)dy.size(1), OW = (int)dy.size(2),
+              C = (int)dy.size(3);
+    auto dx32 = torch::zeros({N, H, W, C}, dy.options().dtype(torch::kFloat32));
+    int64_t total = (int64_t)N * OH * OW * C;
+    int blocks = std::min<int64_t>(ceildiv(total, 256), 2048);
+    hipLaunchKernelGGL(maxpool_gen_bwd_kernel, dim3(blocks), dim3(256), 0,
+                       at::cuda::getCurrentCUDAStream(), bf_ptr(dy),
+                       idx.data_ptr<uint8_t>(), dx32.data_ptr<float>(), N,
+                       (int)H, (int)W, C, OH, OW, (int)k, (int)s, (int)p);
+    return dx32.to(torch::kBFloat16);
+}
+
+torch::Tensor avgpool_global_fwd(torch::Tensor x) {
+    CHECK_GPU(x);
+    TORCH_CHECK(x.is_contiguous());
+    const int N = (int)x.size(0), HW = (int)(x.size(1) * x.size(2)),
+              C = (int)x.size(3);
+    auto y = torch::empty({N, C}, x.options());
+    int64_t total = (int64_t)N * C;
+    int blocks = std::min<int64_t>(ceildiv(total, 256), 2048);
+    hipLaunchKernelGGL(avgpool_global_fwd_kernel, dim3(blocks), dim3(256), 0,
+                       at::cuda::getCurrentCUDAStream(), bf_ptr(x),
+                       bf_ptr_mut(y), N, HW, C);
+    return y;
+}
+
+torch::Tensor avgpool_global_bwd(torch::Tensor dy, int64_t H, int64_t W) {
+    CHECK_GPU(dy);
+    const int N = (int)dy.size(0), C = (int)dy.size(1);
+    auto dx = torch::empty({N, H, W, C}, dy.options());
+    int64_t total = dx.numel();
+    int blocks = std::min<int64_t>(ceildiv(total, 256), 2048);
+    hipLaunchKernelGGL(avgpool_global_bwd_kernel, dim3(blocks), dim3(256), 0,
+                       at::cuda::getCurrentCUDAStream(), bf_ptr(dy),
+                       bf_ptr_mut(dx), N, (int)(H * W), C);
+    return dx;
+}
+
+torch::Tensor add_relu(torch::Tensor a, torch::Tensor b) {
+    CHECK_GPU(a);
+    auto y = torch::empty_like(a);
+    int64_t total = a.numel();
+    int blocks = std::min<int64_t>(ceildiv(total, 256), 2048);
+    hipLaunchKernelGGL(add_relu_kernel, dim3(blocks), dim3(256), 0,
+                       at::cuda::getCurrentCUDAStream(), bf_ptr(a), bf_ptr(b),
+                       bf_ptr_mut(y), total);
+    return y;
 }
 
 torch::Tensor bias_grad(torch::Tensor dy) {

@@ -41,7 +41,9 @@ class FLRunner:
         self.agg: Optional[SecureAggregator] = None
         if cfg.fl.encrypted:
             ctx = CKKSContext(cfg.he, device=device)
-            self.agg = SecureAggregator(ctx, rank=self.rank, verbose=verbose)
+            self.agg = SecureAggregator(ctx, rank=self.rank, verbose=verbose,
+                                        denom_mode=cfg.fl.denom_mode,
+                                        n_clients=cfg.fl.n_clients)
 
     def run_round(self, epochs: Optional[int] = None) -> FLRoundResult:
         t0 = time.perf_counter()

@@ -44,9 +44,36 @@ class SecureAggregator:
     ciphertexts are independent.
     """
 
-    def __init__(self, ctx: CKKSContext, rank: int = 0, verbose: bool = False):
+    def __init__(self, ctx: CKKSContext, rank: int = 0, verbose: bool = False,
+                 denom_mode: str = "plain", n_clients: Optional[int] = None):
+        """denom_mode:
+        - "plain":     divide the summed ciphertext by plaintext 1/n
+                       (ct x plain mult + rescale — what the reference
+                       actually exercises, FLPyfhelin.py:385);
+        - "encrypted": multiply by an ENCRYPTED 1/n (ct x ct + relinearize +
+                       rescale — the reference's commented-out intent,
+                       c_denom at FLPyfhelin.py:371; BASELINE.json config #3).
+                       Requires n_clients at construction: the denominator
+                       ciphertext is derived from the SHARED seed before the
+                       per-rank reseed, so every rank aggregates with the
+                       byte-identical ciphertext (no model drift).
+        """
+        assert denom_mode in ("plain", "encrypted")
         self.ctx = ctx
         self.keys: KeyPair = ctx.keygen()
+        self.denom_mode = denom_mode
+        self.enc_denom = None
+        if denom_mode == "encrypted":
+            if ctx.L < 3:
+                raise ValueError(
+                    "encrypted denominator needs a >=3-limb chain "
+                    "(ct x ct consumes one rescale level)")
+            if n_clients is None:
+                raise ValueError("denom_mode='encrypted' needs n_clients")
+            self.keys.relin = ctx.relin_keygen(self.keys.sk)
+            import numpy as np
+            pt = ctx.encode(np.full(ctx.slots, 1.0 / n_clients))
+            self.enc_denom = ctx.encrypt(pt, self.keys.pk)
         ctx.reseed(rank)
         self.verbose = verbose
         self.stats = AggStats()
@@ -74,7 +101,15 @@ class SecureAggregator:
         self.ctx.modreduce_tensor_(ct)
         out = ct
         if n > 1:
-            out = self.ctx.rescale_tensor(self.ctx.mul_scalar_tensor(ct, 1.0 / n))
+            if self.denom_mode == "encrypted":
+                from ..he.ckks import Ciphertext
+                prod = self.ctx.mul_ct(Ciphertext(ct.data, ct.scale),
+                                       self.enc_denom, self.keys.relin)
+                out = CtxtTensor(prod.data, prod.scale, ct.count)
+                out = self.ctx.rescale_tensor(out)
+            else:
+                out = self.ctx.rescale_tensor(
+                    self.ctx.mul_scalar_tensor(ct, 1.0 / n))
         self._t("aggregate", t0)
         return out
 

@@ -12,15 +12,26 @@ from typing import List
 import torch
 
 
+def _agg_tensors(model: torch.nn.Module):
+    """Parameters + floating-point buffers (BN running stats must be averaged
+    across clients too), in deterministic module order."""
+    for p in model.parameters():
+        yield p
+    for b in model.buffers():
+        if b.is_floating_point():
+            yield b
+
+
 def flat_params(model: torch.nn.Module) -> torch.Tensor:
     with torch.no_grad():
-        return torch.cat([p.detach().float().reshape(-1) for p in model.parameters()])
+        return torch.cat([p.detach().float().reshape(-1)
+                          for p in _agg_tensors(model)])
 
 
 def load_flat_params(model: torch.nn.Module, vec: torch.Tensor) -> None:
     with torch.no_grad():
         off = 0
-        for p in model.parameters():
+        for p in _agg_tensors(model):
             n = p.numel()
             p.copy_(vec[off:off + n].reshape(p.shape).to(p.dtype))
             off += n

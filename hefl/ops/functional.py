@@ -213,3 +213,167 @@ class _SoftmaxXentFn(torch.autograd.Function):
 
 def softmax_xent(logits, labels):
     return _SoftmaxXentFn.apply(logits, labels)
+
+
+# ---------------------------------------------------------------------------
+# BatchNorm2d (NHWC, per-channel over N*H*W) — ResNet-18 support (config #5).
+# ---------------------------------------------------------------------------
+
+class _BatchNorm2dFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, gamma, beta, running_mean, running_var, training: bool,
+                momentum: float, eps: float, relu: bool):
+        ctx.relu = relu
+        ctx.training = training
+        C = x.shape[-1]
+        if x.is_cuda:
+            if training:
+                y, mean, invstd = _C().bn_fwd(x.contiguous(), gamma.detach().float(),
+                                              beta.detach().float(), eps, relu)
+            else:
+                mean = running_mean
+                invstd = (running_var + eps).rsqrt()
+                y = _C().bn_apply(x.contiguous(), mean, invstd,
+                                  gamma.detach().float(), beta.detach().float(),
+                                  relu)
+        else:
+            xf = x.float().reshape(-1, C)
+            if training:
+                mean = xf.mean(0)
+                var = xf.var(0, unbiased=False)
+                invstd = (var + eps).rsqrt()
+            else:
+                mean = running_mean
+                invstd = (running_var + eps).rsqrt()
+            y = ((xf - mean) * invstd * gamma + beta).reshape(x.shape)
+            if relu:
+                y = F.relu(y)
+        if training and running_mean is not None:
+            with torch.no_grad():
+                M = x.numel() // C
+                # torch stores the UNBIASED variance in running_var
+                var_b = (1.0 / invstd.double() ** 2 - eps).float()
+                var_b = var_b * (M / max(M - 1, 1))
+                running_mean.mul_(1 - momentum).add_(mean, alpha=momentum)
+                running_var.mul_(1 - momentum).add_(var_b, alpha=momentum)
+        ctx.save_for_backward(x, gamma.detach(), mean, invstd, y)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, gamma, mean, invstd, y = ctx.saved_tensors
+        C = x.shape[-1]
+        M = x.numel() // C
+        if dy.is_cuda:
+            dy = dy.contiguous()
+            if ctx.relu:
+                dy = _C().relu_bwd(dy, y)
+            dx, dgamma, dbeta = _C().bn_bwd(dy, x.contiguous(), mean, invstd,
+                                            gamma.float(), ctx.training)
+        else:
+            dy = dy.float()
+            if ctx.relu:
+                dy = dy * (y > 0).float()
+            dyf = dy.reshape(-1, C)
+            xf = x.float().reshape(-1, C)
+            xhat = (xf - mean) * invstd
+            dgamma = (dyf * xhat).sum(0)
+            dbeta = dyf.sum(0)
+            if ctx.training:
+                dx = (gamma * invstd) * (dyf - dbeta / M - xhat * dgamma / M)
+            else:
+                dx = (gamma * invstd) * dyf
+            dx = dx.reshape(x.shape)
+        return dx, dgamma, dbeta, None, None, None, None, None, None
+
+
+def batchnorm2d(x, gamma, beta, running_mean, running_var, training=True,
+                momentum=0.1, eps=1e-5, relu=False):
+    return _BatchNorm2dFn.apply(x, gamma, beta, running_mean, running_var,
+                                training, momentum, eps, relu)
+
+
+# ---------------------------------------------------------------------------
+# Generic MaxPool k/s/p (ResNet stem 3x3 s2 p1), global avg pool, add+relu.
+# ---------------------------------------------------------------------------
+
+class _MaxPoolFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, k: int, s: int, p: int):
+        ctx.ksp = (k, s, p)
+        ctx.in_shape = x.shape
+        if x.is_cuda:
+            y, idx = _C().maxpool_fwd(x.contiguous(), k, s, p)
+            ctx.save_for_backward(idx)
+        else:
+            xf = x.float().permute(0, 3, 1, 2)
+            yn, idx = F.max_pool2d(xf, k, s, p, return_indices=True)
+            y = yn.permute(0, 2, 3, 1).contiguous()
+            ctx.save_for_backward(idx)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (idx,) = ctx.saved_tensors
+        k, s, p = ctx.ksp
+        N, H, W, C = ctx.in_shape
+        if dy.is_cuda:
+            dx = _C().maxpool_bwd(dy.contiguous(), idx, H, W, k, s, p)
+        else:
+            # scatter-ADD (max_unpool2d overwrites, wrong for overlapping
+            # windows like the ResNet stem's 3x3 s2)
+            dyn = dy.float().permute(0, 3, 1, 2)
+            dxn = torch.zeros(N, C, H * W, dtype=dyn.dtype)
+            dxn.scatter_add_(2, idx.reshape(N, C, -1), dyn.reshape(N, C, -1))
+            dx = dxn.reshape(N, C, H, W).permute(0, 2, 3, 1).contiguous()
+        return dx, None, None, None
+
+
+def maxpool(x, k: int = 2, s: int = 2, p: int = 0):
+    if k == 2 and s == 2 and p == 0:
+        return maxpool2x2(x)
+    return _MaxPoolFn.apply(x, k, s, p)
+
+
+class _GlobalAvgPoolFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        ctx.in_shape = x.shape
+        if x.is_cuda:
+            return _C().avgpool_global_fwd(x.contiguous())
+        return x.float().mean(dim=(1, 2))
+
+    @staticmethod
+    def backward(ctx, dy):
+        N, H, W, C = ctx.in_shape
+        if dy.is_cuda:
+            return _C().avgpool_global_bwd(dy.contiguous(), H, W)
+        return (dy / (H * W)).reshape(N, 1, 1, C).expand(N, H, W, C).contiguous()
+
+
+def global_avgpool(x):
+    return _GlobalAvgPoolFn.apply(x)
+
+
+class _AddReluFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, a, b):
+        if a.is_cuda:
+            y = _C().add_relu(a.contiguous(), b.contiguous())
+        else:
+            y = F.relu(a.float() + b.float())
+        ctx.save_for_backward(y)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (y,) = ctx.saved_tensors
+        if dy.is_cuda:
+            d = _C().relu_bwd(dy.contiguous(), y)
+        else:
+            d = dy * (y > 0).float()
+        return d, d
+
+
+def add_relu(a, b):
+    return _AddReluFn.apply(a, b)

@@ -62,3 +62,36 @@ class Dense(nn.Module):
 
     def forward(self, x):
         return Fx.linear(x, self.weight, self.bias, self.relu)
+
+
+class BatchNorm2d(nn.Module):
+    """NHWC batch norm over (N,H,W) per channel, optional fused ReLU.
+    fp32 affine params + running stats; bf16 activations on GPU."""
+
+    def __init__(self, c: int, eps: float = 1e-5, momentum: float = 0.1,
+                 relu: bool = False):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(c))
+        self.bias = nn.Parameter(torch.zeros(c))
+        self.register_buffer("running_mean", torch.zeros(c))
+        self.register_buffer("running_var", torch.ones(c))
+        self.eps, self.momentum, self.relu = eps, momentum, relu
+
+    def forward(self, x):
+        return Fx.batchnorm2d(x, self.weight, self.bias, self.running_mean,
+                              self.running_var, self.training, self.momentum,
+                              self.eps, self.relu)
+
+
+class MaxPool(nn.Module):
+    def __init__(self, k: int = 2, s: int = 2, p: int = 0):
+        super().__init__()
+        self.k, self.s, self.p = k, s, p
+
+    def forward(self, x):
+        return Fx.maxpool(x, self.k, self.s, self.p)
+
+
+class GlobalAvgPool(nn.Module):
+    def forward(self, x):
+        return Fx.global_avgpool(x)

@@ -1,0 +1,125 @@
+"""GPU tests: ResNet-support kernels (BN, generic maxpool, global avgpool,
+add+relu) vs CPU fp32 references; ResNet-18 training; config #3 encrypted-
+denominator aggregation on the GPU CKKS kernels."""
+import pytest
+import torch
+
+from hefl.ops import functional as Fx
+
+pytestmark = pytest.mark.gpu
+
+
+def _close(a, b, atol=None, rel=3e-2):
+    a = a.detach().float().cpu()
+    b = b.detach().float().cpu()
+    scale = b.abs().max().item() + 1e-8
+    err = (a - b).abs().max().item()
+    bound = max(atol if atol is not None else 0.0, rel * scale)
+    assert err < bound, f"max abs err {err:.4g} vs scale {scale:.4g}"
+
+
+def test_batchnorm_gpu():
+    torch.manual_seed(0)
+    N, H, W, C = 8, 14, 14, 32
+    x = torch.randn(N, H, W, C)
+    gamma = torch.randn(C) * 0.5 + 1
+    beta = torch.randn(C) * 0.1
+
+    xc = x.clone().requires_grad_(True)
+    gc = gamma.clone().requires_grad_(True)
+    bc = beta.clone().requires_grad_(True)
+    rmc, rvc = torch.zeros(C), torch.ones(C)
+    yc = Fx.batchnorm2d(xc, gc, bc, rmc, rvc, training=True)
+    g = torch.randn_like(yc)
+    yc.backward(g)
+
+    xg = x.to("cuda", torch.bfloat16).requires_grad_(True)
+    gg = gamma.cuda().requires_grad_(True)
+    bg = beta.cuda().requires_grad_(True)
+    rmg, rvg = torch.zeros(C, device="cuda"), torch.ones(C, device="cuda")
+    yg = Fx.batchnorm2d(xg, gg, bg, rmg, rvg, training=True)
+    yg.backward(g.to("cuda", torch.bfloat16))
+
+    _close(yg, yc)
+    _close(rmg, rmc, atol=1e-2)
+    _close(rvg, rvc, atol=1e-2)
+    _close(gg.grad, gc.grad, rel=5e-2)
+    _close(bg.grad, bc.grad, rel=5e-2)
+    _close(xg.grad, xc.grad, atol=2e-2, rel=5e-2)
+
+    # eval mode uses running stats
+    ye_c = Fx.batchnorm2d(xc.detach(), gc.detach(), bc.detach(), rmc, rvc,
+                          training=False)
+    ye_g = Fx.batchnorm2d(xg.detach(), gg.detach(), bg.detach(), rmg, rvg,
+                          training=False)
+    _close(ye_g, ye_c)
+
+
+def test_maxpool3x3_gpu():
+    torch.manual_seed(1)
+    x = torch.randn(4, 15, 15, 16).to(torch.bfloat16).float()
+    xc = x.clone().requires_grad_(True)
+    yc = Fx.maxpool(xc, 3, 2, 1)
+    g = torch.randn_like(yc)
+    yc.backward(g)
+    xg = x.to("cuda", torch.bfloat16).requires_grad_(True)
+    yg = Fx.maxpool(xg, 3, 2, 1)
+    yg.backward(g.to("cuda", torch.bfloat16))
+    _close(yg, yc)
+    _close(xg.grad, xc.grad, rel=5e-2)
+
+
+def test_avgpool_add_relu_gpu():
+    torch.manual_seed(2)
+    x = torch.randn(4, 8, 8, 64)
+    xc = x.clone().requires_grad_(True)
+    yc = Fx.global_avgpool(xc)
+    yc.sum().backward()
+    xg = x.to("cuda", torch.bfloat16).requires_grad_(True)
+    yg = Fx.global_avgpool(xg)
+    yg.sum().backward()
+    _close(yg, yc)
+    _close(xg.grad, xc.grad)
+
+    a = torch.randn(1000)
+    b = torch.randn(1000)
+    yc2 = Fx.add_relu(a.clone().requires_grad_(True), b.clone().requires_grad_(True))
+    yg2 = Fx.add_relu(a.to("cuda", torch.bfloat16), b.to("cuda", torch.bfloat16))
+    _close(yg2, yc2)
+
+
+def test_resnet18_trains_gpu():
+    from hefl.config import ModelConfig
+    from hefl.models import build_model
+    from hefl.ops.adam import FusedAdam
+    from hefl.ops.functional import softmax_xent
+
+    torch.manual_seed(3)
+    m = build_model(ModelConfig("resnet18", (64, 64, 3), 10), seed=1).cuda()
+    opt = FusedAdam(m.parameters(), lr=1e-3)
+    x = torch.randn(8, 64, 64, 3, device="cuda", dtype=torch.bfloat16)
+    y = torch.randint(0, 10, (8,), device="cuda")
+    losses = []
+    for _ in range(10):
+        loss = softmax_xent(m(x), y)
+        opt.zero_grad()
+        loss.backward()
+        opt.step()
+        losses.append(loss.item())
+    assert losses[-1] < losses[0], losses
+
+
+def test_encrypted_denominator_aggregation_gpu():
+    """Config #3 semantics: ct x ct multiply + relinearize + rescale in the
+    aggregation path, on the GPU CKKS kernels (n=2^14, 3-limb chain)."""
+    from hefl.config import HEConfig
+    from hefl.fl.secure import SecureAggregator
+    from hefl.he.ckks import CKKSContext
+
+    cfg = HEConfig(m=16384, scale_bits=40, q_bits=(60, 40, 40), seed=11)
+    ctx = CKKSContext(cfg, device="cuda")
+    agg = SecureAggregator(ctx, rank=0, denom_mode="encrypted", n_clients=8)
+    vec = torch.randn(30000)
+    ct = agg.encrypt(vec * 8)  # as if 8 clients summed
+    out = agg.decrypt(agg.aggregate(ct, n_clients=8))
+    assert (out - vec).abs().max().item() < 5e-3
