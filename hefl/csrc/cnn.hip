@@ -142,168 +142,312 @@ gemm_kernel(const unsigned short* __restrict__ A,
 }
 
 // ---------------------------------------------------------------------------
-// Conv2d forward, implicit GEMM: M = N*OH*OW, N-dim = Kout, K-dim = R*S*C.
-// x [N,H,W,C] bf16; w [Kout,R,S,C] bf16; y [N,OH,OW,Kout] bf16.
+// Conv2d implicit-GEMM kernels, v2.
+//
+// Shared structure (fwd / dgrad / wgrad): 64xBN output tile per 256-thread
+// workgroup (BN=64: 2x2 waves each computing 2x2 MFMA fragments; BN=16:
+// 4x1 waves, 1 fragment), BK=32 K-steps staged through LDS. Staging uses
+// 16-B vector loads whenever the innermost gathered dim is a multiple of 8
+// (channels for fwd A / wgrad B, output channels for dgrad A / wgrad A —
+// dgrad reorders its K-dim to (r,s,ko) to make the dy gather contiguous);
+// per-row im2col coordinate decode is hoisted out of the K loop.
 // ---------------------------------------------------------------------------
 
 struct ConvShape {
     int N, H, W, C, Kout, R, S, OH, OW, stride, pad;
 };
 
+template <int BN, int WM, int WN, int FM, int FN>
+struct ConvTile {
+    static constexpr int BM = 64;
+    static constexpr int BK = 32;
+    unsigned short As[BM][BK];
+    unsigned short Bs[BN][BK];
+};
+
+template <int BN, int WM, int WN, int FM, int FN>
+__device__ __forceinline__ void tile_mfma(
+    const unsigned short (&As)[64][32], const unsigned short (&Bs)[BN][32],
+    f32x4 (&acc)[FM][FN], int wave, int lane) {
+    const int wm = wave / WN, wn = wave % WN;
+    const int half = lane >> 4, sub = lane & 15;
+    bf16x8 a[FM], b[FN];
+#pragma unroll
+    for (int i = 0; i < FM; ++i)
+        a[i] = *reinterpret_cast<const bf16x8*>(
+            &As[wm * FM * 16 + i * 16 + sub][half * 8]);
+#pragma unroll
+    for (int j = 0; j < FN; ++j)
+        b[j] = *reinterpret_cast<const bf16x8*>(
+            &Bs[wn * FN * 16 + j * 16 + sub][half * 8]);
+#pragma unroll
+    for (int i = 0; i < FM; ++i)
+#pragma unroll
+        for (int j = 0; j < FN; ++j)
+            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[i], b[j],
+                                                                acc[i][j], 0, 0, 0);
+}
+
+__device__ __forceinline__ void copy16(unsigned short* dst,
+                                       const unsigned short* src) {
+    *reinterpret_cast<uint4*>(dst) = *reinterpret_cast<const uint4*>(src);
+}
+
+__device__ __forceinline__ void zero16(unsigned short* dst) {
+    *reinterpret_cast<uint4*>(dst) = uint4{0, 0, 0, 0};
+}
+
+// ---- forward: y[m=(n,oh,ow), ko] = sum_k A(m,k) * w[ko, k],
+//      k = (r, s, c) with c fastest ----
+template <int BN, int WM, int WN, int FM, int FN>
 __global__ void __launch_bounds__(TPB)
 conv_fwd_kernel(const unsigned short* __restrict__ x,
                 const unsigned short* __restrict__ w,
                 const float* __restrict__ bias, unsigned short* __restrict__ y,
                 ConvShape s, int relu) {
-    __shared__ TileSmem sm;
+    __shared__ ConvTile<BN, WM, WN, FM, FN> sm;
     const int tid = threadIdx.x;
     const int wave = tid >> 6;
     const int lane = tid & 63;
-    const int m0 = blockIdx.x * BM;
+    const int m0 = blockIdx.x * 64;
     const int n0 = blockIdx.y * BN;
     const int M = s.N * s.OH * s.OW;
     const int KK = s.R * s.S * s.C;
+    const bool fast = (s.C % 8 == 0);
 
-    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-    for (int k0 = 0; k0 < KK; k0 += BK) {
-        for (int i = tid; i < BM * BK / 8; i += TPB) {
-            int row = i / (BK / 8);
-            int kc = (i % (BK / 8)) * 8;
-            int m = m0 + row;
-            int n_ = m / (s.OH * s.OW);
-            int rem = m % (s.OH * s.OW);
-            int oh = rem / s.OW, ow = rem % s.OW;
+    // hoisted per-thread A-row coordinates (row = tid/4, k-chunk = tid%4)
+    const int arow = tid >> 2;
+    const int akc = (tid & 3) * 8;
+    const int am = m0 + arow;
+    int a_n = 0, a_oh = 0, a_ow = 0;
+    if (am < M) {
+        a_n = am / (s.OH * s.OW);
+        int rem = am % (s.OH * s.OW);
+        a_oh = rem / s.OW;
+        a_ow = rem % s.OW;
+    }
+
+    f32x4 acc[FM][FN];
 #pragma unroll
-            for (int j = 0; j < 8; ++j) {
-                int k = k0 + kc + j;
-                unsigned short v = 0;
-                if (m < M && k < KK) {
-                    int c = k % s.C;
-                    int rs = k / s.C;        // k-dim ordered (r, s, c)
+    for (int i = 0; i < FM; ++i)
+#pragma unroll
+        for (int j = 0; j < FN; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+    for (int k0 = 0; k0 < KK; k0 += 32) {
+        {  // ---- stage A (one 8-elem chunk per thread) ----
+            const int k = k0 + akc;
+            unsigned short* dst = &sm.As[arow][akc];
+            if (am < M && k < KK) {
+                if (fast) {
+                    int rs = k / s.C, c = k % s.C;
                     int r = rs / s.S, ss = rs % s.S;
-                    int ih = oh * s.stride + r - s.pad;
-                    int iw = ow * s.stride + ss - s.pad;
+                    int ih = a_oh * s.stride + r - s.pad;
+                    int iw = a_ow * s.stride + ss - s.pad;
                     if (ih >= 0 && ih < s.H && iw >= 0 && iw < s.W)
-                        v = x[(((int64_t)n_ * s.H + ih) * s.W + iw) * s.C + c];
+                        copy16(dst, x + ((((int64_t)a_n * s.H + ih) * s.W + iw)
+                                             * s.C + c));
+                    else
+                        zero16(dst);
+                } else {
+#pragma unroll
+                    for (int j = 0; j < 8; ++j) {
+                        int kk = k + j;
+                        unsigned short v = 0;
+                        if (kk < KK) {
+                            int c = kk % s.C;
+                            int rs = kk / s.C;
+                            int r = rs / s.S, ss = rs % s.S;
+                            int ih = a_oh * s.stride + r - s.pad;
+                            int iw = a_ow * s.stride + ss - s.pad;
+                            if (ih >= 0 && ih < s.H && iw >= 0 && iw < s.W)
+                                v = x[((((int64_t)a_n * s.H + ih) * s.W + iw)
+                                           * s.C + c)];
+                        }
+                        dst[j] = v;
+                    }
                 }
-                sm.As[row][kc + j] = v;
+            } else {
+                zero16(dst);
             }
         }
-        for (int i = tid; i < BN * BK / 8; i += TPB) {
-            int col = i / (BK / 8);
-            int kc = (i % (BK / 8)) * 8;
+        // ---- stage B: w[ko][k] contiguous in k ----
+        for (int i = tid; i < BN * 4; i += TPB) {
+            const int col = i >> 2;
+            const int kc = (i & 3) * 8;
+            const int ko = n0 + col, k = k0 + kc;
+            unsigned short* dst = &sm.Bs[col][kc];
+            if (ko < s.Kout && k < KK) {
+                if ((KK & 7) == 0) {
+                    copy16(dst, w + (int64_t)ko * KK + k);
+                } else {
 #pragma unroll
-            for (int j = 0; j < 8; ++j) {
-                int ko = n0 + col, k = k0 + kc + j;
-                unsigned short v = 0;
-                if (ko < s.Kout && k < KK) {
-                    int c = k % s.C;
-                    int rs = k / s.C;
-                    v = w[((int64_t)ko * s.R * s.S + rs) * s.C + c];
+                    for (int j = 0; j < 8; ++j)
+                        dst[j] = (k + j < KK) ? w[(int64_t)ko * KK + k + j] : 0;
                 }
-                sm.Bs[col][kc + j] = v;
+            } else {
+                zero16(dst);
             }
         }
         __syncthreads();
-        acc = tile_mfma_step(sm, wave, lane, acc);
+        tile_mfma<BN, WM, WN, FM, FN>(sm.As, sm.Bs, acc, wave, lane);
         __syncthreads();
     }
-    const int col = n0 + (lane & 15);
-    if (col >= s.Kout) return;
-    float bv = bias ? bias[col] : 0.f;
+    // ---- epilogue ----
+    const int wm = wave / WN, wn = wave % WN;
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-        int row = m0 + wave * 16 + (lane >> 4) * 4 + r;
-        if (row >= M) continue;
-        float v = acc[r] + bv;
-        if (relu) v = v > 0.f ? v : 0.f;
-        y[(int64_t)row * s.Kout + col] = f2bf(v);
+    for (int j = 0; j < FN; ++j) {
+        const int col = n0 + wn * FN * 16 + j * 16 + (lane & 15);
+        if (col >= s.Kout) continue;
+        const float bv = bias ? bias[col] : 0.f;
+#pragma unroll
+        for (int i = 0; i < FM; ++i)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int row = m0 + wm * FM * 16 + i * 16 + (lane >> 4) * 4 + r;
+                if (row >= M) continue;
+                float v = acc[i][j][r] + bv;
+                if (relu) v = v > 0.f ? v : 0.f;
+                y[(int64_t)row * s.Kout + col] = f2bf(v);
+            }
     }
 }
 
-// Conv2d dgrad: dx[n,ih,iw,c] = sum_{ko,r,s} dy[n,oh,ow,ko] * w[ko,r,s,c]
-// where oh = (ih + pad - r)/stride when divisible. Implicit GEMM:
-// M = N*H*W, N-dim = C, K-dim = Kout*R*S, B = w viewed [Kout*R*S, C].
+// ---- dgrad: dx[m=(n,ih,iw), c] = sum_k A(m,k) * B(k,c),
+//      k = (r, s, ko) with ko FASTEST so the dy gather is contiguous;
+//      B(k, c) = w[ko, r, s, c] (strided, small tile) ----
+template <int BN, int WM, int WN, int FM, int FN>
 __global__ void __launch_bounds__(TPB)
 conv_dgrad_kernel(const unsigned short* __restrict__ dy,
                   const unsigned short* __restrict__ w,
                   unsigned short* __restrict__ dx, ConvShape s) {
-    __shared__ TileSmem sm;
+    __shared__ ConvTile<BN, WM, WN, FM, FN> sm;
     const int tid = threadIdx.x;
     const int wave = tid >> 6;
     const int lane = tid & 63;
-    const int m0 = blockIdx.x * BM;
+    const int m0 = blockIdx.x * 64;
     const int n0 = blockIdx.y * BN;
     const int M = s.N * s.H * s.W;
     const int KK = s.Kout * s.R * s.S;
+    const bool fast = (s.Kout % 8 == 0);
 
-    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-    for (int k0 = 0; k0 < KK; k0 += BK) {
-        for (int i = tid; i < BM * BK / 8; i += TPB) {
-            int row = i / (BK / 8);
-            int kc = (i % (BK / 8)) * 8;
-            int m = m0 + row;
-            int n_ = m / (s.H * s.W);
-            int rem = m % (s.H * s.W);
-            int ih = rem / s.W, iw = rem % s.W;
+    const int arow = tid >> 2;
+    const int akc = (tid & 3) * 8;
+    const int am = m0 + arow;
+    int a_n = 0, a_ih = 0, a_iw = 0;
+    if (am < M) {
+        a_n = am / (s.H * s.W);
+        int rem = am % (s.H * s.W);
+        a_ih = rem / s.W;
+        a_iw = rem % s.W;
+    }
+
+    f32x4 acc[FM][FN];
 #pragma unroll
-            for (int j = 0; j < 8; ++j) {
-                int k = k0 + kc + j;
-                unsigned short v = 0;
-                if (m < M && k < KK) {
-                    int rs = k % (s.R * s.S);
-                    int ko = k / (s.R * s.S);
-                    int r = rs / s.S, ss = rs % s.S;
-                    int oh_num = ih + s.pad - r, ow_num = iw + s.pad - ss;
-                    if (oh_num >= 0 && ow_num >= 0 && oh_num % s.stride == 0 &&
-                        ow_num % s.stride == 0) {
-                        int oh = oh_num / s.stride, ow = ow_num / s.stride;
-                        if (oh < s.OH && ow < s.OW)
-                            v = dy[(((int64_t)n_ * s.OH + oh) * s.OW + ow) *
-                                       s.Kout + ko];
-                    }
+    for (int i = 0; i < FM; ++i)
+#pragma unroll
+        for (int j = 0; j < FN; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+    for (int k0 = 0; k0 < KK; k0 += 32) {
+        {  // ---- stage A from dy ----
+            const int k = k0 + akc;
+            unsigned short* dst = &sm.As[arow][akc];
+            bool done = false;
+            if (am < M && k < KK && fast) {
+                // chunk shares (r,s): ko = k % Kout, rs = k / Kout
+                int rs = k / s.Kout, ko = k % s.Kout;
+                int r = rs / s.S, ss = rs % s.S;
+                int oh_num = a_ih + s.pad - r, ow_num = a_iw + s.pad - ss;
+                done = true;
+                if (oh_num >= 0 && ow_num >= 0 && oh_num % s.stride == 0 &&
+                    ow_num % s.stride == 0 && oh_num / s.stride < s.OH &&
+                    ow_num / s.stride < s.OW) {
+                    copy16(dst, dy + ((((int64_t)a_n * s.OH + oh_num / s.stride)
+                                          * s.OW + ow_num / s.stride)
+                                         * s.Kout + ko));
+                } else {
+                    zero16(dst);
                 }
-                sm.As[row][kc + j] = v;
+            }
+            if (!done) {
+                if (am < M && k < KK) {
+#pragma unroll
+                    for (int j = 0; j < 8; ++j) {
+                        int kk = k + j;
+                        unsigned short v = 0;
+                        if (kk < KK) {
+                            int rs = kk / s.Kout, ko = kk % s.Kout;
+                            int r = rs / s.S, ss = rs % s.S;
+                            int oh_num = a_ih + s.pad - r;
+                            int ow_num = a_iw + s.pad - ss;
+                            if (oh_num >= 0 && ow_num >= 0 &&
+                                oh_num % s.stride == 0 &&
+                                ow_num % s.stride == 0 &&
+                                oh_num / s.stride < s.OH &&
+                                ow_num / s.stride < s.OW)
+                                v = dy[((((int64_t)a_n * s.OH +
+                                          oh_num / s.stride) * s.OW +
+                                         ow_num / s.stride) * s.Kout + ko)];
+                        }
+                        dst[j] = v;
+                    }
+                } else {
+                    zero16(dst);
+                }
             }
         }
-        for (int i = tid; i < BN * BK / 8; i += TPB) {
-            int col = i / (BK / 8);
-            int kc = (i % (BK / 8)) * 8;
+        // ---- stage B: B(k=(rs,ko), c) = w[ko*RS*C + rs*C + c] (scalar) ----
+        for (int i = tid; i < BN * 4; i += TPB) {
+            const int col = i >> 2;
+            const int kc = (i & 3) * 8;
+            const int c = n0 + col, k = k0 + kc;
+            unsigned short* dst = &sm.Bs[col][kc];
 #pragma unroll
             for (int j = 0; j < 8; ++j) {
-                int c = n0 + col, k = k0 + kc + j;
+                int kk = k + j;
                 unsigned short v = 0;
-                if (c < s.C && k < KK)
-                    v = w[(int64_t)k * s.C + c];  // w flat [(ko,r,s), c]
-                sm.Bs[col][kc + j] = v;
+                if (c < s.C && kk < KK) {
+                    int rs = kk / s.Kout, ko = kk % s.Kout;
+                    v = w[((int64_t)ko * s.R * s.S + rs) * s.C + c];
+                }
+                dst[j] = v;
             }
         }
         __syncthreads();
-        acc = tile_mfma_step(sm, wave, lane, acc);
+        tile_mfma<BN, WM, WN, FM, FN>(sm.As, sm.Bs, acc, wave, lane);
         __syncthreads();
     }
-    const int col = n0 + (lane & 15);
-    if (col >= s.C) return;
+    const int wm = wave / WN, wn = wave % WN;
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-        int row = m0 + wave * 16 + (lane >> 4) * 4 + r;
-        if (row >= M) continue;
-        dx[(int64_t)row * s.C + col] = f2bf(acc[r]);
+    for (int j = 0; j < FN; ++j) {
+        const int col = n0 + wn * FN * 16 + j * 16 + (lane & 15);
+        if (col >= s.C) continue;
+#pragma unroll
+        for (int i = 0; i < FM; ++i)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int row = m0 + wm * FM * 16 + i * 16 + (lane >> 4) * 4 + r;
+                if (row >= M) continue;
+                dx[(int64_t)row * s.C + col] = f2bf(acc[i][j][r]);
+            }
     }
 }
 
-// Conv2d wgrad: dw[ko, r, s, c] = sum_pix dy[pix, ko] * x[pix->(ih,iw), c].
-// GEMM: M = Kout, N-dim = R*S*C, K-dim = N*OH*OW; split-K over grid.z with
-// fp32 atomics into dw (zeroed by the host wrapper).
+// ---- wgrad: dw[ko, rsc] = sum_pix dy[pix, ko] * xcol[pix, rsc].
+// LDS tiles are PIXEL-major (Dys[32][64], Xs[32][BN]) so BOTH global
+// gathers are 16-B vector loads (ko contiguous in dy; c contiguous in x);
+// fragments read LDS with k-strided scalar loads instead. Split-K over
+// grid.z with fp32 atomics. ----
+template <int BN, int WM, int WN, int FM, int FN>
 __global__ void __launch_bounds__(TPB)
 conv_wgrad_kernel(const unsigned short* __restrict__ dy,
                   const unsigned short* __restrict__ x,
                   float* __restrict__ dw, ConvShape s, int k_chunks) {
-    __shared__ TileSmem sm;
+    __shared__ unsigned short Dys[32][64 + 8];  // [pix][ko] (+pad)
+    __shared__ unsigned short Xs[32][BN + 8];   // [pix][rsc]
     const int tid = threadIdx.x;
     const int wave = tid >> 6;
     const int lane = tid & 63;
-    const int m0 = blockIdx.x * BM;   // over Kout
+    const int m0 = blockIdx.x * 64;   // over Kout
     const int n0 = blockIdx.y * BN;   // over RSC
     const int M = s.Kout;
     const int NN = s.R * s.S * s.C;
@@ -311,61 +455,128 @@ conv_wgrad_kernel(const unsigned short* __restrict__ dy,
     const int chunk = (KK + k_chunks - 1) / k_chunks;
     const int kbeg = blockIdx.z * chunk;
     const int kend = min(kbeg + chunk, KK);
+    const bool fast_dy = (s.Kout % 8 == 0);
+    const bool fast_x = (s.C % 8 == 0);
 
-    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-    for (int k0 = kbeg; k0 < kend; k0 += BK) {
-        for (int i = tid; i < BM * BK / 8; i += TPB) {
-            int row = i / (BK / 8);
-            int kc = (i % (BK / 8)) * 8;
-            int ko = m0 + row;
+    f32x4 acc[FM][FN];
 #pragma unroll
-            for (int j = 0; j < 8; ++j) {
-                int k = k0 + kc + j;  // pixel index
-                unsigned short v = 0;
-                if (ko < M && k < kend)
-                    v = dy[(int64_t)k * s.Kout + ko];
-                sm.As[row][kc + j] = v;
+    for (int i = 0; i < FM; ++i)
+#pragma unroll
+        for (int j = 0; j < FN; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+    for (int k0 = kbeg; k0 < kend; k0 += 32) {
+        // ---- stage dy tile [32 pix][64 ko]: 256 chunks of 8 ko ----
+        {
+            const int pix = tid >> 3;
+            const int koc = (tid & 7) * 8;
+            const int kpix = k0 + pix;
+            const int ko = m0 + koc;
+            unsigned short* dst = &Dys[pix][koc];
+            if (kpix < kend && ko < M && fast_dy) {
+                copy16(dst, dy + (int64_t)kpix * s.Kout + ko);
+            } else if (kpix < kend && ko < M) {
+#pragma unroll
+                for (int j = 0; j < 8; ++j)
+                    dst[j] = (ko + j < M) ? dy[(int64_t)kpix * s.Kout + ko + j]
+                                          : 0;
+            } else {
+                zero16(dst);
             }
         }
-        for (int i = tid; i < BN * BK / 8; i += TPB) {
-            int col = i / (BK / 8);
-            int kc = (i % (BK / 8)) * 8;
-            int nn = n0 + col;
-            int c = nn % s.C;
-            int rs = nn / s.C;
-            int r = rs / s.S, ss = rs % s.S;
-#pragma unroll
-            for (int j = 0; j < 8; ++j) {
-                int k = k0 + kc + j;
-                unsigned short v = 0;
-                if (nn < NN && k < kend) {
-                    int n_ = k / (s.OH * s.OW);
-                    int rem = k % (s.OH * s.OW);
-                    int oh = rem / s.OW, ow = rem % s.OW;
-                    int ih = oh * s.stride + r - s.pad;
-                    int iw = ow * s.stride + ss - s.pad;
+        // ---- stage x tile [32 pix][BN rsc]: BN*4 chunks of 8 c ----
+        for (int i = tid; i < 32 * (BN / 8); i += TPB) {
+            const int pix = i / (BN / 8);
+            const int cc = (i % (BN / 8)) * 8;
+            const int kpix = k0 + pix;
+            const int nn = n0 + cc;
+            unsigned short* dst = &Xs[pix][cc];
+            bool done = false;
+            if (kpix < kend && nn < NN) {
+                int n_ = kpix / (s.OH * s.OW);
+                int rem = kpix % (s.OH * s.OW);
+                int oh = rem / s.OW, ow = rem % s.OW;
+                int c = nn % s.C;
+                int rs = nn / s.C;
+                int r = rs / s.S, ss = rs % s.S;
+                int ih = oh * s.stride + r - s.pad;
+                int iw = ow * s.stride + ss - s.pad;
+                if (fast_x && c + 8 <= s.C) {
+                    done = true;
                     if (ih >= 0 && ih < s.H && iw >= 0 && iw < s.W)
-                        v = x[(((int64_t)n_ * s.H + ih) * s.W + iw) * s.C + c];
+                        copy16(dst, x + ((((int64_t)n_ * s.H + ih) * s.W + iw)
+                                             * s.C + c));
+                    else
+                        zero16(dst);
                 }
-                sm.Bs[col][kc + j] = v;
+                if (!done) {
+#pragma unroll
+                    for (int j = 0; j < 8; ++j) {
+                        int n2 = nn + j;
+                        unsigned short v = 0;
+                        if (n2 < NN) {
+                            int c2 = n2 % s.C;
+                            int rs2 = n2 / s.C;
+                            int r2 = rs2 / s.S, ss2 = rs2 % s.S;
+                            int ih2 = oh * s.stride + r2 - s.pad;
+                            int iw2 = ow * s.stride + ss2 - s.pad;
+                            if (ih2 >= 0 && ih2 < s.H && iw2 >= 0 && iw2 < s.W)
+                                v = x[((((int64_t)n_ * s.H + ih2) * s.W + iw2)
+                                           * s.C + c2)];
+                        }
+                        dst[j] = v;
+                    }
+                    done = true;
+                }
             }
+            if (!done) zero16(dst);
         }
         __syncthreads();
-        acc = tile_mfma_step(sm, wave, lane, acc);
+        // ---- MFMA: A(m=ko, k=pix) from Dys[k][m]; B(k=pix, n) from Xs ----
+        {
+            const int wm = wave / WN, wn = wave % WN;
+            const int half = lane >> 4, sub = lane & 15;
+            bf16x8 a[FM], b[FN];
+#pragma unroll
+            for (int i = 0; i < FM; ++i) {
+                const int mrow = wm * FM * 16 + i * 16 + sub;
+#pragma unroll
+                for (int j = 0; j < 8; ++j)
+                    a[i][j] = *reinterpret_cast<const bf16_t*>(
+                        &Dys[half * 8 + j][mrow]);
+            }
+#pragma unroll
+            for (int j = 0; j < FN; ++j) {
+                const int col = wn * FN * 16 + j * 16 + sub;
+#pragma unroll
+                for (int t = 0; t < 8; ++t)
+                    b[j][t] = *reinterpret_cast<const bf16_t*>(
+                        &Xs[half * 8 + t][col]);
+            }
+#pragma unroll
+            for (int i = 0; i < FM; ++i)
+#pragma unroll
+                for (int j = 0; j < FN; ++j)
+                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        a[i], b[j], acc[i][j], 0, 0, 0);
+        }
         __syncthreads();
     }
-    const int col = n0 + (lane & 15);
-    if (col >= NN) return;
-    // dw layout [Kout, R, S, C] flat = [ko][rsc] with rsc ordered (r,s,c):
-    // col indexes (r,s,c) in that same order.
+    const int wm = wave / WN, wn = wave % WN;
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-        int ko = m0 + wave * 16 + (lane >> 4) * 4 + r;
-        if (ko >= M) continue;
-        if (k_chunks > 1)
-            atomicAdd(dw + (int64_t)ko * NN + col, acc[r]);
-        else
-            dw[(int64_t)ko * NN + col] = acc[r];
+    for (int j = 0; j < FN; ++j) {
+        const int col = n0 + wn * FN * 16 + j * 16 + (lane & 15);
+        if (col >= NN) continue;
+#pragma unroll
+        for (int i = 0; i < FM; ++i)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int ko = m0 + wm * FM * 16 + i * 16 + (lane >> 4) * 4 + r;
+                if (ko >= M) continue;
+                if (k_chunks > 1)
+                    atomicAdd(dw + (int64_t)ko * NN + col, acc[i][j][r]);
+                else
+                    dw[(int64_t)ko * NN + col] = acc[i][j][r];
+            }
     }
 }
 
@@ -805,11 +1016,19 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
     ConvShape s = make_shape(x, w, (int)stride, (int)pad);
     auto y = torch::empty({s.N, s.OH, s.OW, s.Kout}, x.options());
     const int M = s.N * s.OH * s.OW;
-    dim3 grid(ceildiv(M, BM), ceildiv(s.Kout, BN));
     const float* bias = b.numel() ? b.data_ptr<float>() : nullptr;
-    hipLaunchKernelGGL(conv_fwd_kernel, grid, dim3(TPB), 0,
-                       at::cuda::getCurrentCUDAStream(), bf_ptr(x), bf_ptr(w),
-                       bias, bf_ptr_mut(y), s, relu ? 1 : 0);
+    auto stream = at::cuda::getCurrentCUDAStream();
+    if (s.Kout > 16) {
+        dim3 grid(ceildiv(M, 64), ceildiv(s.Kout, 64));
+        hipLaunchKernelGGL((conv_fwd_kernel<64, 2, 2, 2, 2>), grid, dim3(TPB),
+                           0, stream, bf_ptr(x), bf_ptr(w), bias, bf_ptr_mut(y),
+                           s, relu ? 1 : 0);
+    } else {
+        dim3 grid(ceildiv(M, 64), ceildiv(s.Kout, 16));
+        hipLaunchKernelGGL((conv_fwd_kernel<16, 4, 1, 1, 1>), grid, dim3(TPB),
+                           0, stream, bf_ptr(x), bf_ptr(w), bias, bf_ptr_mut(y),
+                           s, relu ? 1 : 0);
+    }
     return y;
 }
 
@@ -826,10 +1045,16 @@ torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor w, int64_t stride,
     s.OH = (int)dy.size(1); s.OW = (int)dy.size(2);
     auto dx = torch::empty({N, (int64_t)H, (int64_t)W, C}, dy.options());
     const int M = N * (int)H * (int)W;
-    dim3 grid(ceildiv(M, BM), ceildiv(C, BN));
-    hipLaunchKernelGGL(conv_dgrad_kernel, grid, dim3(TPB), 0,
-                       at::cuda::getCurrentCUDAStream(), bf_ptr(dy), bf_ptr(w),
-                       bf_ptr_mut(dx), s);
+    auto stream = at::cuda::getCurrentCUDAStream();
+    if (C > 16) {
+        dim3 grid(ceildiv(M, 64), ceildiv(C, 64));
+        hipLaunchKernelGGL((conv_dgrad_kernel<64, 2, 2, 2, 2>), grid, dim3(TPB),
+                           0, stream, bf_ptr(dy), bf_ptr(w), bf_ptr_mut(dx), s);
+    } else {
+        dim3 grid(ceildiv(M, 64), ceildiv(C, 16));
+        hipLaunchKernelGGL((conv_dgrad_kernel<16, 4, 1, 1, 1>), grid, dim3(TPB),
+                           0, stream, bf_ptr(dy), bf_ptr(w), bf_ptr_mut(dx), s);
+    }
     return dx;
 }
 
@@ -845,18 +1070,28 @@ torch::Tensor conv2d_wgrad(torch::Tensor dy, torch::Tensor x, int64_t stride,
     s.OH = (int)dy.size(1); s.OW = (int)dy.size(2);
     const int NN = s.R * s.S * s.C;
     const int KK = s.N * s.OH * s.OW;
+    const bool big = NN > 16;
+    const int bn = big ? 64 : 16;
     // split-K to fill the chip: target >= 512 blocks
-    int tiles = ceildiv(s.Kout, BM) * ceildiv(NN, BN);
-    int k_chunks = std::max(1, std::min(ceildiv(KK, BK * 8), 512 / std::max(tiles, 1)));
+    int tiles = ceildiv(s.Kout, 64) * ceildiv(NN, bn);
+    int k_chunks = std::max(1, std::min(ceildiv(KK, 32 * 8),
+                                        512 / std::max(tiles, 1)));
     auto dw = k_chunks > 1
                   ? torch::zeros({s.Kout, R, S, s.C},
                                  x.options().dtype(torch::kFloat32))
                   : torch::empty({s.Kout, R, S, s.C},
                                  x.options().dtype(torch::kFloat32));
-    dim3 grid(ceildiv(s.Kout, BM), ceildiv(NN, BN), k_chunks);
-    hipLaunchKernelGGL(conv_wgrad_kernel, grid, dim3(TPB), 0,
-                       at::cuda::getCurrentCUDAStream(), bf_ptr(dy), bf_ptr(x),
-                       dw.data_ptr<float>(), s, k_chunks);
+    dim3 grid(ceildiv(s.Kout, 64), ceildiv(NN, bn), k_chunks);
+    auto stream = at::cuda::getCurrentCUDAStream();
+    if (big) {
+        hipLaunchKernelGGL((conv_wgrad_kernel<64, 2, 2, 2, 2>), grid, dim3(TPB),
+                           0, stream, bf_ptr(dy), bf_ptr(x),
+                           dw.data_ptr<float>(), s, k_chunks);
+    } else {
+        hipLaunchKernelGGL((conv_wgrad_kernel<16, 4, 1, 1, 1>), grid, dim3(TPB),
+                           0, stream, bf_ptr(dy), bf_ptr(x),
+                           dw.data_ptr<float>(), s, k_chunks);
+    }
     return dw;
 }
 
