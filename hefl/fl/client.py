@@ -100,7 +100,11 @@ class LocalClient:
         # seeded with the warmup step count
         self.opt.prepare_graph_state(sx.device)
         g = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(g):
+        # capture on the SAME stream the warmup ran on: AccumulateGrad nodes
+        # bind to the stream that first materialized each .grad, and a capture
+        # on a different stream lets the accumulation escape the graph
+        # (symptom: replays run but weights never learn)
+        with torch.cuda.graph(g, stream=side):
             self.opt.zero_grad_()
             logits = self.model(sx.to(self.compute_dtype))
             loss = softmax_xent(logits, sy)

@@ -51,7 +51,11 @@ class FusedAdam:
 
     def prepare_graph_state(self, device=None):
         """Create device-side step/schedule/hyper buffers (must run OUTSIDE
-        any hipGraph capture)."""
+        any hipGraph capture). Created ONCE: captured graphs hold raw
+        pointers to these buffers, so re-creating them would leave earlier
+        graphs reading freed memory."""
+        if hasattr(self, "_step_t"):
+            return  # keep live buffers; device counter keeps running
         dev = device if device is not None else self.params[0].device
         self._step_t = torch.tensor([self.step_count], dtype=torch.int64,
                                     device=dev)

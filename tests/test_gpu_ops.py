@@ -184,3 +184,27 @@ def test_training_reduces_loss_gpu():
         opt.step()
         losses.append(loss.item())
     assert losses[-1] < losses[0] * 0.5, losses[::5]
+
+
+def test_graphed_client_training_converges():
+    """LocalClient with hipGraphs ON and a shard whose last batch is PARTIAL
+    (two captured graphs share Adam's device-side schedule buffers — a
+    re-created buffer would leave graph #1 reading freed memory and silently
+    stop learning; regression test for that exact bug)."""
+    from hefl.config import preset
+    from hefl.fl.client import LocalClient
+
+    cfg = preset("config2")
+    cfg.fl.n_clients = 1
+    cfg.fl.samples_per_client = 80  # 2 full batches + 1 partial (16)
+    cfg.train.hip_graphs = True
+    c = LocalClient(cfg, 0, device="cuda:0")
+    assert c.use_graphs
+    first = c.local_train(epochs=1)
+    for _ in range(12):
+        last = c.local_train(epochs=1)
+    assert len(c._graphs) == 2, "expected full + partial batch graphs"
+    assert last.train_loss < first.train_loss * 0.5, (first.train_loss,
+                                                      last.train_loss)
+    w = c.get_weights()
+    assert not torch.isnan(w).any()
