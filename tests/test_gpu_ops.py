@@ -201,7 +201,7 @@ def test_graphed_client_training_converges():
     c = LocalClient(cfg, 0, device="cuda:0")
     assert c.use_graphs
     first = c.local_train(epochs=1)
-    for _ in range(12):
+    for _ in range(40):
         last = c.local_train(epochs=1)
     assert len(c._graphs) == 2, "expected full + partial batch graphs"
     assert last.train_loss < first.train_loss * 0.5, (first.train_loss,
