@@ -320,7 +320,8 @@ __global__ void __launch_bounds__(TPB)
 conv_dgrad_kernel(const unsigned short* __restrict__ dy,
                   const unsigned short* __restrict__ w,
                   unsigned short* __restrict__ dx, ConvShape s) {
-    __shared__ ConvTile<BN, WM, WN, FM, FN> sm;
+    __shared__ unsigned short As[64][32];
+    __shared__ unsigned short Bst[32][BN + 8];  // k-major: B(k, c) tile
     const int tid = threadIdx.x;
     const int wave = tid >> 6;
     const int lane = tid & 63;
@@ -350,7 +351,7 @@ conv_dgrad_kernel(const unsigned short* __restrict__ dy,
     for (int k0 = 0; k0 < KK; k0 += 32) {
         {  // ---- stage A from dy ----
             const int k = k0 + akc;
-            unsigned short* dst = &sm.As[arow][akc];
+            unsigned short* dst = &As[arow][akc];
             bool done = false;
             if (am < M && k < KK && fast) {
                 // chunk shares (r,s): ko = k % Kout, rs = k / Kout
@@ -395,25 +396,53 @@ conv_dgrad_kernel(const unsigned short* __restrict__ dy,
                 }
             }
         }
-        // ---- stage B: B(k=(rs,ko), c) = w[ko*RS*C + rs*C + c] (scalar) ----
-        for (int i = tid; i < BN * 4; i += TPB) {
-            const int col = i >> 2;
-            const int kc = (i & 3) * 8;
-            const int c = n0 + col, k = k0 + kc;
-            unsigned short* dst = &sm.Bs[col][kc];
+        // ---- stage B k-major: Bst[k][c..c+8] <- w[ko*RS*C + rs*C + c],
+        //      contiguous in c (16-B loads when C % 8 == 0) ----
+        for (int i = tid; i < 32 * (BN / 8); i += TPB) {
+            const int kk = i / (BN / 8);
+            const int cc = (i % (BN / 8)) * 8;
+            const int k = k0 + kk;
+            const int c = n0 + cc;
+            unsigned short* dst = &Bst[kk][cc];
+            if (k < KK && c < s.C) {
+                int rs = k / s.Kout, ko = k % s.Kout;
+                const unsigned short* src =
+                    w + ((int64_t)ko * s.R * s.S + rs) * s.C + c;
+                if (s.C % 8 == 0) {
+                    copy16(dst, src);
+                } else {
 #pragma unroll
-            for (int j = 0; j < 8; ++j) {
-                int kk = k + j;
-                unsigned short v = 0;
-                if (c < s.C && kk < KK) {
-                    int rs = kk / s.Kout, ko = kk % s.Kout;
-                    v = w[((int64_t)ko * s.R * s.S + rs) * s.C + c];
+                    for (int j = 0; j < 8; ++j)
+                        dst[j] = (c + j < s.C) ? src[j] : 0;
                 }
-                dst[j] = v;
+            } else {
+                zero16(dst);
             }
         }
         __syncthreads();
-        tile_mfma<BN, WM, WN, FM, FN>(sm.As, sm.Bs, acc, wave, lane);
+        {   // ---- MFMA: A from As (k-minor), B from Bst (k-major) ----
+            const int wm = wave / WN, wn = wave % WN;
+            const int half = lane >> 4, sub = lane & 15;
+            bf16x8 a[FM], b[FN];
+#pragma unroll
+            for (int i = 0; i < FM; ++i)
+                a[i] = *reinterpret_cast<const bf16x8*>(
+                    &As[wm * FM * 16 + i * 16 + sub][half * 8]);
+#pragma unroll
+            for (int j = 0; j < FN; ++j) {
+                const int col = wn * FN * 16 + j * 16 + sub;
+#pragma unroll
+                for (int t = 0; t < 8; ++t)
+                    b[j][t] = *reinterpret_cast<const bf16_t*>(
+                        &Bst[half * 8 + t][col]);
+            }
+#pragma unroll
+            for (int i = 0; i < FM; ++i)
+#pragma unroll
+                for (int j = 0; j < FN; ++j)
+                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        a[i], b[j], acc[i][j], 0, 0, 0);
+        }
         __syncthreads();
     }
     const int wm = wave / WN, wn = wave % WN;
@@ -739,32 +768,36 @@ __global__ void fused_adam_sched_kernel(float* __restrict__ p,
 // BatchNorm (NHWC, per-channel over N*H*W) — ResNet-18 (config #5) support.
 // ---------------------------------------------------------------------------
 
-// One block per channel: mean + invstd in fp32.
-__global__ void bn_stats_kernel(const unsigned short* __restrict__ x,
-                                float* __restrict__ mean,
-                                float* __restrict__ invstd, int64_t M, int C,
-                                float eps) {
-    __shared__ float s_sum[256], s_sq[256];
-    const int c = blockIdx.x;
-    float acc = 0.f, acc2 = 0.f;
-    for (int64_t r = threadIdx.x; r < M; r += blockDim.x) {
-        float v = bf2f(x[r * C + c]);
-        acc += v;
-        acc2 += v * v;
-    }
-    s_sum[threadIdx.x] = acc;
-    s_sq[threadIdx.x] = acc2;
-    __syncthreads();
-    for (int off = 128; off > 0; off >>= 1) {
-        if (threadIdx.x < off) {
-            s_sum[threadIdx.x] += s_sum[threadIdx.x + off];
-            s_sq[threadIdx.x] += s_sq[threadIdx.x + off];
+// Coalesced two-stage reduction: stage 1 blocks each cover a row-chunk x
+// all C channels (consecutive threads -> consecutive channels), partials
+// atomically added into fp32 accumulators; stage 2 finalizes mean/invstd.
+__global__ void bn_partial_kernel(const unsigned short* __restrict__ x,
+                                  float* __restrict__ gsum,
+                                  float* __restrict__ gsq, int64_t M, int C,
+                                  int rows_per_block) {
+    const int64_t r0 = (int64_t)blockIdx.x * rows_per_block;
+    const int64_t r1 = min(r0 + rows_per_block, M);
+    for (int c = threadIdx.x; c < C; c += blockDim.x) {
+        float acc = 0.f, acc2 = 0.f;
+        for (int64_t r = r0; r < r1; ++r) {
+            float v = bf2f(x[r * C + c]);
+            acc += v;
+            acc2 += v * v;
         }
-        __syncthreads();
+        atomicAdd(gsum + c, acc);
+        atomicAdd(gsq + c, acc2);
     }
-    if (threadIdx.x == 0) {
-        float mu = s_sum[0] / (float)M;
-        float var = fmaxf(s_sq[0] / (float)M - mu * mu, 0.f);
+}
+
+__global__ void bn_finalize_kernel(const float* __restrict__ gsum,
+                                   const float* __restrict__ gsq,
+                                   float* __restrict__ mean,
+                                   float* __restrict__ invstd, int64_t M,
+                                   int C, float eps) {
+    for (int c = blockIdx.x * blockDim.x + threadIdx.x; c < C;
+         c += gridDim.x * blockDim.x) {
+        float mu = gsum[c] / (float)M;
+        float var = fmaxf(gsq[c] / (float)M - mu * mu, 0.f);
         mean[c] = mu;
         invstd[c] = rsqrtf(var + eps);
     }
@@ -786,36 +819,27 @@ __global__ void bn_apply_kernel(const unsigned short* __restrict__ x,
     }
 }
 
-// dgamma[c] = sum dy*xhat; dbeta[c] = sum dy. One block per channel.
-__global__ void bn_reduce_bwd_kernel(const unsigned short* __restrict__ dy,
-                                     const unsigned short* __restrict__ x,
-                                     const float* __restrict__ mean,
-                                     const float* __restrict__ invstd,
-                                     float* __restrict__ dgamma,
-                                     float* __restrict__ dbeta, int64_t M,
-                                     int C) {
-    __shared__ float s_dg[256], s_db[256];
-    const int c = blockIdx.x;
-    float dg = 0.f, db = 0.f;
-    for (int64_t r = threadIdx.x; r < M; r += blockDim.x) {
-        float g = bf2f(dy[r * C + c]);
-        float xh = (bf2f(x[r * C + c]) - mean[c]) * invstd[c];
-        dg += g * xh;
-        db += g;
-    }
-    s_dg[threadIdx.x] = dg;
-    s_db[threadIdx.x] = db;
-    __syncthreads();
-    for (int off = 128; off > 0; off >>= 1) {
-        if (threadIdx.x < off) {
-            s_dg[threadIdx.x] += s_dg[threadIdx.x + off];
-            s_db[threadIdx.x] += s_db[threadIdx.x + off];
+// dgamma[c] = sum dy*xhat; dbeta[c] = sum dy — same coalesced two-stage
+// shape (partials straight into dgamma/dbeta, zeroed by the wrapper).
+__global__ void bn_bwd_partial_kernel(const unsigned short* __restrict__ dy,
+                                      const unsigned short* __restrict__ x,
+                                      const float* __restrict__ mean,
+                                      const float* __restrict__ invstd,
+                                      float* __restrict__ dgamma,
+                                      float* __restrict__ dbeta, int64_t M,
+                                      int C, int rows_per_block) {
+    const int64_t r0 = (int64_t)blockIdx.x * rows_per_block;
+    const int64_t r1 = min(r0 + rows_per_block, M);
+    for (int c = threadIdx.x; c < C; c += blockDim.x) {
+        float dg = 0.f, db = 0.f;
+        const float mu = mean[c], is = invstd[c];
+        for (int64_t r = r0; r < r1; ++r) {
+            float g = bf2f(dy[r * C + c]);
+            dg += g * (bf2f(x[r * C + c]) - mu) * is;
+            db += g;
         }
-        __syncthreads();
-    }
-    if (threadIdx.x == 0) {
-        dgamma[c] = s_dg[0];
-        dbeta[c] = s_db[0];
+        atomicAdd(dgamma + c, dg);
+        atomicAdd(dbeta + c, db);
     }
 }
 
@@ -1256,9 +1280,17 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
     auto invstd = torch::empty({C}, f32);
     auto y = torch::empty_like(x);
     auto stream = at::cuda::getCurrentCUDAStream();
-    hipLaunchKernelGGL(bn_stats_kernel, dim3(C), dim3(256), 0, stream,
-                       bf_ptr(x), mean.data_ptr<float>(),
-                       invstd.data_ptr<float>(), M, C, (float)eps);
+    auto gsum = torch::zeros({C}, f32);
+    auto gsq = torch::zeros({C}, f32);
+    int rpb = (int)std::max<int64_t>(64, (M + 511) / 512);
+    int nblk = (int)((M + rpb - 1) / rpb);
+    hipLaunchKernelGGL(bn_partial_kernel, dim3(nblk), dim3(256), 0, stream,
+                       bf_ptr(x), gsum.data_ptr<float>(), gsq.data_ptr<float>(),
+                       M, C, rpb);
+    hipLaunchKernelGGL(bn_finalize_kernel, dim3(ceildiv(C, 256)), dim3(256), 0,
+                       stream, gsum.data_ptr<float>(), gsq.data_ptr<float>(),
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(), M, C,
+                       (float)eps);
     int64_t total = x.numel();
     int blocks = std::min<int64_t>(ceildiv(total, 256), 2048);
     hipLaunchKernelGGL(bn_apply_kernel, dim3(blocks), dim3(256), 0, stream,
@@ -1293,14 +1325,16 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
     const int C = (int)x.size(-1);
     const int64_t M = x.numel() / C;
     auto f32 = x.options().dtype(torch::kFloat32);
-    auto dgamma = torch::empty({C}, f32);
-    auto dbeta = torch::empty({C}, f32);
+    auto dgamma = torch::zeros({C}, f32);
+    auto dbeta = torch::zeros({C}, f32);
     auto dx = torch::empty_like(x);
     auto stream = at::cuda::getCurrentCUDAStream();
-    hipLaunchKernelGGL(bn_reduce_bwd_kernel, dim3(C), dim3(256), 0, stream,
+    int rpb = (int)std::max<int64_t>(64, (M + 511) / 512);
+    int nblk = (int)((M + rpb - 1) / rpb);
+    hipLaunchKernelGGL(bn_bwd_partial_kernel, dim3(nblk), dim3(256), 0, stream,
                        bf_ptr(dyc), bf_ptr(x), mean.data_ptr<float>(),
                        invstd.data_ptr<float>(), dgamma.data_ptr<float>(),
-                       dbeta.data_ptr<float>(), M, C);
+                       dbeta.data_ptr<float>(), M, C, rpb);
     int64_t total = x.numel();
     int blocks = std::min<int64_t>(ceildiv(total, 256), 2048);
     hipLaunchKernelGGL(bn_dx_kernel, dim3(blocks), dim3(256), 0, stream,
