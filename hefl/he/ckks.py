@@ -291,6 +291,15 @@ class CKKSContext:
         """vals: real [..., k<=slots] -> NTT-form RNS plaintext."""
         scale = self.scale if scale is None else scale
         nlimbs = self.L if nlimbs is None else nlimbs
+        if self.device.type == "cuda":
+            # full-device path: special FFT + rounding on the GPU
+            vt = (vals if torch.is_tensor(vals)
+                  else torch.from_numpy(np.asarray(vals, dtype=np.float64)))
+            ct = self.encoder.encode_torch(vt.to(self.device), scale)
+            out = []
+            for i in range(nlimbs):
+                out.append(self.backend.ntt(torch.remainder(ct, self._q(i)), i))
+            return Plaintext(torch.stack(out, dim=-2), scale)
         coeffs = self.encoder.encode(np.asarray(vals, dtype=np.float64), scale)
         out = []
         if coeffs.dtype == np.int64:
@@ -315,6 +324,8 @@ class CKKSContext:
             q = self._q(0)
             c = self.backend.ntt(data[..., 0, :].contiguous(), 0, inverse=True)
             cent = torch.where(c > q // 2, c - q, c)
+            if self.device.type == "cuda":
+                return self.encoder.decode_torch(cent, pt.scale, k)
             return self.encoder.decode(cent.cpu().numpy(), pt.scale, k)
         # multi-limb fast path: when the plaintext magnitude is << q0/2 (true
         # for any decode of real-valued weights at scale <= 2^45), the
@@ -328,6 +339,8 @@ class CKKSContext:
             c1 = self.backend.ntt(data[..., 1, :].contiguous(), 1, inverse=True)
             cent1 = torch.where(c1 > q1 // 2, c1 - q1, c1)
             if torch.equal(torch.remainder(cent0, q1), torch.remainder(cent1, q1)):
+                if self.device.type == "cuda":
+                    return self.encoder.decode_torch(cent0, pt.scale, k)
                 return self.encoder.decode(cent0.cpu().numpy(), pt.scale, k)
         # exact big-int CRT path (plaintext too large for limb-0 shortcut)
         coeff_limbs = []
@@ -443,15 +456,23 @@ class CKKSContext:
         """Slot-pack a flat fp32 vector into ceil(len/slots) ciphertexts."""
         count = vec.numel()
         B = (count + self.slots - 1) // self.slots
-        buf = np.zeros((B, self.slots), dtype=np.float64)
-        buf.reshape(-1)[:count] = vec.detach().float().cpu().numpy().reshape(-1)
-        pt = self.encode(buf)                     # [B, L, n]
+        if self.device.type == "cuda":
+            buf = torch.zeros(B * self.slots, dtype=torch.float64,
+                              device=self.device)
+            buf[:count] = vec.detach().double().reshape(-1).to(self.device)
+            pt = self.encode(buf.reshape(B, self.slots))
+        else:
+            buf = np.zeros((B, self.slots), dtype=np.float64)
+            buf.reshape(-1)[:count] = vec.detach().float().cpu().numpy().reshape(-1)
+            pt = self.encode(buf)                 # [B, L, n]
         data = self._encrypt_data(pt.data, pk)    # [B, 2, L, n]
         return CtxtTensor(data, pt.scale, count)
 
     def decrypt_tensor(self, ct: CtxtTensor, sk: torch.Tensor) -> torch.Tensor:
         pt = self._decrypt_data(ct.data, sk)      # [B, L, n]
         vals = self.decode(Plaintext(pt, ct.scale), self.slots)  # [B, slots]
+        if torch.is_tensor(vals):
+            return vals.reshape(-1)[:ct.count].to(torch.float32)
         flat = torch.from_numpy(np.ascontiguousarray(vals.reshape(-1)[:ct.count]))
         return flat.to(torch.float32)
 

@@ -101,3 +101,114 @@ class Encoder:
         z = cf[..., :half] + 1j * cf[..., half:]
         v = self.fft_special(z / scale)
         return v.real[..., :k]
+
+
+# ---------------------------------------------------------------------------
+# Torch port of the special FFT pair: runs on the context device (MI355X) in
+# complex128. The numpy path above stays as the CPU oracle; this path exists
+# because host-side encode/decode of a ResNet-sized weight vector (hundreds
+# of ciphertexts x 2^14 slots) costs seconds in numpy and ~1 ms on device.
+# ---------------------------------------------------------------------------
+
+import torch as _torch
+
+
+class TorchEncoderMixin:
+    def _torch_tables(self, device):
+        cache = getattr(self, "_tcache", None)
+        if cache is None:
+            cache = self._tcache = {}
+        key = str(device)
+        if key in cache:
+            return cache[key]
+        perm = _torch.tensor(
+            [int(bit_reverse(i, self.slots.bit_length() - 1))
+             for i in range(self.slots)], dtype=_torch.long, device=device)
+        fwd, inv = [], []
+        size = self.slots
+        length = 2
+        while length <= size:
+            lenh, lenq = length // 2, length * 4
+            idx = [(self.rot[j] % lenq) * (self.M // lenq) for j in range(lenh)]
+            fwd.append(_torch.from_numpy(self.ksi[idx]).to(device))
+            length *= 2
+        length = size
+        while length >= 2:
+            lenh, lenq = length // 2, length * 4
+            idx = [(lenq - (self.rot[j] % lenq)) * (self.M // lenq)
+                   for j in range(lenh)]
+            inv.append(_torch.from_numpy(self.ksi[idx]).to(device))
+            length //= 2
+        cache[key] = (perm, fwd, inv)
+        return cache[key]
+
+    def fft_special_torch(self, v: "_torch.Tensor") -> "_torch.Tensor":
+        """Decode direction. v: complex128 [..., slots] on device."""
+        perm, fwd, _ = self._torch_tables(v.device)
+        v = v.index_select(-1, perm)
+        size = self.slots
+        length = 2
+        si = 0
+        while length <= size:
+            lenh = length // 2
+            w = fwd[si]
+            lead = v.shape[:-1]
+            v = v.reshape(*lead, size // length, 2, lenh)
+            u = v[..., 0, :]
+            t = v[..., 1, :] * w
+            v = _torch.stack([u + t, u - t], dim=-2).reshape(*lead, size)
+            length *= 2
+            si += 1
+        return v
+
+    def fft_special_inv_torch(self, v: "_torch.Tensor") -> "_torch.Tensor":
+        """Encode direction. v: complex128 [..., slots] on device."""
+        perm, _, inv = self._torch_tables(v.device)
+        size = self.slots
+        length = size
+        si = 0
+        while length >= 2:
+            lenh = length // 2
+            w = inv[si]
+            lead = v.shape[:-1]
+            v = v.reshape(*lead, size // length, 2, lenh)
+            u = v[..., 0, :] + v[..., 1, :]
+            t = (v[..., 0, :] - v[..., 1, :]) * w
+            v = _torch.stack([u, t], dim=-2).reshape(*lead, size)
+            length //= 2
+            si += 1
+        v = v.index_select(-1, perm)
+        return v / size
+
+    def encode_torch(self, vals: "_torch.Tensor", scale: float) -> "_torch.Tensor":
+        """vals: real [..., k<=slots] on device -> int64 coeffs [..., n]
+        (centered; exact for |coeff| < 2^52)."""
+        vals = vals.to(_torch.float64)
+        pad = self.slots - vals.shape[-1]
+        if pad < 0:
+            raise ValueError("too many values for slot count")
+        if pad:
+            vals = _torch.nn.functional.pad(vals, (0, pad))
+        z = self.fft_special_inv_torch(vals.to(_torch.complex128))
+        re = _torch.round(z.real * scale)
+        im = _torch.round(z.imag * scale)
+        peak = max(re.abs().max().item(), im.abs().max().item())
+        if peak >= 2.0 ** 52:
+            raise OverflowError("encode overflow: scale too large for f64 path")
+        return _torch.cat([re, im], dim=-1).to(_torch.int64)
+
+    def decode_torch(self, coeffs: "_torch.Tensor", scale: float,
+                     k: int) -> "_torch.Tensor":
+        """Centered int64 coeffs [..., n] on device -> float32 [..., k]."""
+        half = self.slots
+        cf = coeffs.to(_torch.float64)
+        z = _torch.complex(cf[..., :half], cf[..., half:]) / scale
+        v = self.fft_special_torch(z)
+        return v.real[..., :k].to(_torch.float32)
+
+
+# graft the torch methods onto Encoder (mixin-by-assignment; __bases__
+# reassignment is not allowed on classes deriving from object directly)
+for _name in ("_torch_tables", "fft_special_torch", "fft_special_inv_torch",
+              "encode_torch", "decode_torch"):
+    setattr(Encoder, _name, getattr(TorchEncoderMixin, _name))
