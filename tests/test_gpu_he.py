@@ -94,7 +94,7 @@ def test_gpu_context_encrypt_decrypt_matches_cpu():
     vec = torch.randn(10000)
     ct = gpu.encrypt_tensor(vec, kp.pk)
     assert ct.data.is_cuda
-    back = gpu.decrypt_tensor(ct, kp.sk)
+    back = gpu.decrypt_tensor(ct, kp.sk).cpu()
     assert (back - vec).abs().max().item() < 1e-3
 
     # homomorphic FedAvg semantics on GPU
@@ -102,7 +102,7 @@ def test_gpu_context_encrypt_decrypt_matches_cpu():
     lazy = CtxtTensor(ct.data + ct2.data, ct.scale, ct.count)
     gpu.modreduce_tensor_(lazy)
     avg = gpu.rescale_tensor(gpu.mul_scalar_tensor(lazy, 0.5))
-    out = gpu.decrypt_tensor(avg, kp.sk)
+    out = gpu.decrypt_tensor(avg, kp.sk).cpu()
     assert (out - 2 * vec).abs().max().item() < 1e-2
 
 

@@ -121,5 +121,5 @@ def test_encrypted_denominator_aggregation_gpu():
     agg = SecureAggregator(ctx, rank=0, denom_mode="encrypted", n_clients=8)
     vec = torch.randn(30000)
     ct = agg.encrypt(vec * 8)  # as if 8 clients summed
-    out = agg.decrypt(agg.aggregate(ct, n_clients=8))
+    out = agg.decrypt(agg.aggregate(ct, n_clients=8)).cpu()
     assert (out - vec).abs().max().item() < 5e-3
