@@ -57,6 +57,11 @@ torch::Tensor avgpool_global_fwd(torch::Tensor x);
 torch::Tensor avgpool_global_bwd(torch::Tensor dy, int64_t H, int64_t W);
 torch::Tensor add_relu(torch::Tensor a, torch::Tensor b);
 torch::Tensor bias_grad(torch::Tensor dy);
+void fused_adam_mt(torch::Tensor meta, torch::Tensor ptrs, torch::Tensor sizes,
+                   int64_t nchunks, torch::Tensor sched, double b1, double b2,
+                   double eps);
+void zero_grads_mt(torch::Tensor meta, torch::Tensor ptrs, torch::Tensor sizes,
+                   int64_t nchunks);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "hefl gfx950 (MI355X/CDNA4) HIP kernels";
@@ -92,4 +97,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("avgpool_global_bwd", &avgpool_global_bwd);
     m.def("add_relu", &add_relu);
     m.def("bias_grad", &bias_grad);
+    m.def("fused_adam_mt", &fused_adam_mt);
+    m.def("zero_grads_mt", &zero_grads_mt);
 }

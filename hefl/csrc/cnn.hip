@@ -765,6 +765,117 @@ __global__ void fused_adam_sched_kernel(float* __restrict__ p,
 }
 
 // ---------------------------------------------------------------------------
+// Skinny linear fwd (M ~ batch 32): one wave per (16x16 tile, K-chunk),
+// fragments loaded STRAIGHT from global (both operands contiguous along K,
+// L2-resident at these sizes), fp32 atomics into y32, tiny epilogue kernel.
+// Replaces the LDS-staged path that was latency-bound at 4 workgroups.
+// ---------------------------------------------------------------------------
+
+__global__ void linear_splitk_kernel(const unsigned short* __restrict__ x,
+                                     const unsigned short* __restrict__ w,
+                                     float* __restrict__ y32, int M, int N,
+                                     int K, int kc_len) {
+    const int lane = threadIdx.x & 63;
+    const int m0 = blockIdx.x * 16;
+    const int n0 = blockIdx.y * 16;
+    const int k0 = blockIdx.z * kc_len;
+    const int kend = min(k0 + kc_len, K);
+    const int sub = lane & 15, half = lane >> 4;
+    const int row = m0 + sub;         // A row
+    const int col = n0 + sub;         // B col (w row)
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    for (int k = k0; k < kend; k += 32) {
+        bf16x8 a{}, b{};
+        const int kk = k + half * 8;
+        if (row < M) {
+            if (kk + 8 <= K && (kk & 7) == 0)
+                a = *reinterpret_cast<const bf16x8*>(&x[(int64_t)row * K + kk]);
+            else
+#pragma unroll
+                for (int j = 0; j < 8; ++j)
+                    a[j] = (kk + j < K) ? *reinterpret_cast<const bf16_t*>(
+                                              &x[(int64_t)row * K + kk + j])
+                                        : (bf16_t)0.f;
+        }
+        if (col < N) {
+            if (kk + 8 <= K && (kk & 7) == 0)
+                b = *reinterpret_cast<const bf16x8*>(&w[(int64_t)col * K + kk]);
+            else
+#pragma unroll
+                for (int j = 0; j < 8; ++j)
+                    b[j] = (kk + j < K) ? *reinterpret_cast<const bf16_t*>(
+                                              &w[(int64_t)col * K + kk + j])
+                                        : (bf16_t)0.f;
+        }
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        int orow = m0 + (lane >> 4) * 4 + r;
+        int ocol = n0 + (lane & 15);
+        if (orow < M && ocol < N)
+            atomicAdd(y32 + (int64_t)orow * N + ocol, acc[r]);
+    }
+}
+
+__global__ void linear_epilogue_kernel(const float* __restrict__ y32,
+                                       const float* __restrict__ bias,
+                                       unsigned short* __restrict__ y,
+                                       int64_t total, int N, int relu) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        float v = y32[i] + (bias ? bias[i % N] : 0.f);
+        if (relu) v = v > 0.f ? v : 0.f;
+        y[i] = f2bf(v);
+    }
+}
+
+// ---------------------------------------------------------------------------
+// Multi-tensor fused Adam + multi-tensor grad zero: ONE launch over all
+// parameter tensors via a precomputed chunk table (built once per optimizer;
+// hipGraph replays touch no host state).
+// meta: int64 [nchunks][2] = {tensor index, element offset};
+// ptrs: int64 [T][4] = {p, g, m, v} device addresses; sizes: int64 [T].
+// ---------------------------------------------------------------------------
+
+constexpr int MT_CHUNK = 8192;
+
+__global__ void fused_adam_mt_kernel(const int64_t* __restrict__ meta,
+                                     const int64_t* __restrict__ ptrs,
+                                     const int64_t* __restrict__ sizes,
+                                     const float* __restrict__ sched, float b1,
+                                     float b2, float eps) {
+    const int c = blockIdx.x;
+    const int t = (int)meta[c * 2];
+    const int64_t off = meta[c * 2 + 1];
+    float* p = reinterpret_cast<float*>(ptrs[t * 4 + 0]);
+    const float* g = reinterpret_cast<const float*>(ptrs[t * 4 + 1]);
+    float* m = reinterpret_cast<float*>(ptrs[t * 4 + 2]);
+    float* v = reinterpret_cast<float*>(ptrs[t * 4 + 3]);
+    const int64_t n = min(off + (int64_t)MT_CHUNK, sizes[t]);
+    const float lr = sched[0], bc1 = sched[1], bc2 = sched[2];
+    for (int64_t i = off + threadIdx.x; i < n; i += blockDim.x) {
+        float gi = g[i];
+        float mi = b1 * m[i] + (1.f - b1) * gi;
+        float vi = b2 * v[i] + (1.f - b2) * gi * gi;
+        m[i] = mi;
+        v[i] = vi;
+        p[i] -= lr * (mi / bc1) / (sqrtf(vi / bc2) + eps);
+    }
+}
+
+__global__ void zero_mt_kernel(const int64_t* __restrict__ meta,
+                               const int64_t* __restrict__ ptrs,
+                               const int64_t* __restrict__ sizes) {
+    const int c = blockIdx.x;
+    const int t = (int)meta[c * 2];
+    const int64_t off = meta[c * 2 + 1];
+    float* g = reinterpret_cast<float*>(ptrs[t * 4 + 1]);
+    const int64_t n = min(off + (int64_t)MT_CHUNK, sizes[t]);
+    for (int64_t i = off + threadIdx.x; i < n; i += blockDim.x) g[i] = 0.f;
+}
+
+// ---------------------------------------------------------------------------
 // BatchNorm (NHWC, per-channel over N*H*W) — ResNet-18 (config #5) support.
 // ---------------------------------------------------------------------------
 
@@ -983,13 +1094,17 @@ __global__ void relu_bwd_kernel(const unsigned short* __restrict__ dy,
     }
 }
 
-// db[k] = sum over rows of dy[., k]; dy bf16 [M, K]; one block per k.
+// db[k] = sum over rows of dy[., k]: 2-D grid (k x row-chunks), coalesced
+// row sweeps, fp32 atomics into db (zeroed by the wrapper).
 __global__ void bias_grad_kernel(const unsigned short* __restrict__ dy,
-                                 float* __restrict__ db, int64_t M, int K) {
+                                 float* __restrict__ db, int64_t M, int K,
+                                 int rows_per_block) {
     __shared__ float red[256];
     const int k = blockIdx.x;
+    const int64_t r0 = (int64_t)blockIdx.y * rows_per_block;
+    const int64_t r1 = min(r0 + rows_per_block, M);
     float acc = 0.f;
-    for (int64_t r = threadIdx.x; r < M; r += blockDim.x)
+    for (int64_t r = r0 + threadIdx.x; r < r1; r += blockDim.x)
         acc += bf2f(dy[r * K + k]);
     red[threadIdx.x] = acc;
     __syncthreads();
@@ -997,7 +1112,7 @@ __global__ void bias_grad_kernel(const unsigned short* __restrict__ dy,
         if (threadIdx.x < off) red[threadIdx.x] += red[threadIdx.x + off];
         __syncthreads();
     }
-    if (threadIdx.x == 0) db[k] = red[0];
+    if (threadIdx.x == 0) atomicAdd(db + k, red[0]);
 }
 
 inline int ceildiv(int64_t a, int64_t b) { return (int)((a + b - 1) / b); }
@@ -1124,11 +1239,27 @@ torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
     CHECK_GPU(x);
     TORCH_CHECK(x.is_contiguous() && w.is_contiguous());
     const int M = (int)x.size(0), K = (int)x.size(1), N = (int)w.size(0);
-    auto y = torch::empty({M, N}, x.options());
-    dim3 grid(ceildiv(M, BM), ceildiv(N, BN));
+    auto stream = at::cuda::getCurrentCUDAStream();
     const float* bias = b.numel() ? b.data_ptr<float>() : nullptr;
+    auto y = torch::empty({M, N}, x.options());
+    if (M <= 64) {
+        auto y32 = torch::zeros({M, N}, x.options().dtype(torch::kFloat32));
+        int kc = std::max(256, K / 4);
+        kc = ((kc + 31) / 32) * 32;
+        dim3 grid(ceildiv(M, 16), ceildiv(N, 16), ceildiv(K, kc));
+        hipLaunchKernelGGL(linear_splitk_kernel, grid, dim3(64), 0, stream,
+                           bf_ptr(x), bf_ptr(w), y32.data_ptr<float>(), M, N,
+                           K, kc);
+        int64_t total = (int64_t)M * N;
+        hipLaunchKernelGGL(linear_epilogue_kernel,
+                           dim3((int)std::min<int64_t>(ceildiv(total, 256), 2048)),
+                           dim3(256), 0, stream, y32.data_ptr<float>(), bias,
+                           bf_ptr_mut(y), total, N, relu ? 1 : 0);
+        return y;
+    }
+    dim3 grid(ceildiv(M, BM), ceildiv(N, BN));
     hipLaunchKernelGGL((gemm_kernel<false, true, true, false>), grid, dim3(TPB),
-                       0, at::cuda::getCurrentCUDAStream(), bf_ptr(x), bf_ptr(w),
+                       0, stream, bf_ptr(x), bf_ptr(w),
                        bf_ptr_mut(y), bias, M, N, K, relu ? 1 : 0);
     return y;
 }
@@ -1421,9 +1552,31 @@ torch::Tensor bias_grad(torch::Tensor dy) {
     auto dyc = dy.contiguous();
     const int K = (int)dyc.size(-1);
     const int64_t M = dyc.numel() / K;
-    auto db = torch::empty({K}, dyc.options().dtype(torch::kFloat32));
-    hipLaunchKernelGGL(bias_grad_kernel, dim3(K), dim3(256), 0,
+    auto db = torch::zeros({K}, dyc.options().dtype(torch::kFloat32));
+    int rpb = (int)std::max<int64_t>(2048, (M + 63) / 64);
+    dim3 grid(K, (unsigned)((M + rpb - 1) / rpb));
+    hipLaunchKernelGGL(bias_grad_kernel, grid, dim3(256), 0,
                        at::cuda::getCurrentCUDAStream(), bf_ptr(dyc),
-                       db.data_ptr<float>(), M, K);
+                       db.data_ptr<float>(), M, K, rpb);
     return db;
+}
+
+void fused_adam_mt(torch::Tensor meta, torch::Tensor ptrs, torch::Tensor sizes,
+                   int64_t nchunks, torch::Tensor sched, double b1, double b2,
+                   double eps) {
+    CHECK_GPU(meta);
+    hipLaunchKernelGGL(fused_adam_mt_kernel, dim3((unsigned)nchunks), dim3(256),
+                       0, at::cuda::getCurrentCUDAStream(),
+                       meta.data_ptr<int64_t>(), ptrs.data_ptr<int64_t>(),
+                       sizes.data_ptr<int64_t>(), sched.data_ptr<float>(),
+                       (float)b1, (float)b2, (float)eps);
+}
+
+void zero_grads_mt(torch::Tensor meta, torch::Tensor ptrs, torch::Tensor sizes,
+                   int64_t nchunks) {
+    CHECK_GPU(meta);
+    hipLaunchKernelGGL(zero_mt_kernel, dim3((unsigned)nchunks), dim3(256), 0,
+                       at::cuda::getCurrentCUDAStream(),
+                       meta.data_ptr<int64_t>(), ptrs.data_ptr<int64_t>(),
+                       sizes.data_ptr<int64_t>());
 }

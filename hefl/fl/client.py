@@ -55,7 +55,7 @@ class LocalClient:
         n_total = cfg.fl.n_clients * cfg.fl.samples_per_client
         self.dataset = SyntheticMedicalImages(
             n_total, cfg.model.in_shape, cfg.model.n_classes,
-            seed=cfg.fl.seed, device=device)
+            seed=cfg.fl.seed, device=device, dtype=self.compute_dtype)
         idx = shard_indices(n_total, client_id, cfg.fl.n_clients)
         self.loader = ClientLoader(self.dataset, idx, t.batch_size,
                                    seed=cfg.fl.seed + client_id)
@@ -104,12 +104,20 @@ class LocalClient:
         # bind to the stream that first materialized each .grad, and a capture
         # on a different stream lets the accumulation escape the graph
         # (symptom: replays run but weights never learn)
+        if not hasattr(self, "_acc_loss"):
+            self._acc_loss = torch.zeros((), dtype=torch.float32, device=sx.device)
+            self._acc_correct = torch.zeros((), dtype=torch.float32,
+                                            device=sx.device)
         with torch.cuda.graph(g, stream=side):
             self.opt.zero_grad_()
             logits = self.model(sx.to(self.compute_dtype))
             loss = softmax_xent(logits, sy)
             loss.backward()
             self.opt.step_graphed()
+            # stats accumulate INSIDE the graph into persistent buffers
+            self._acc_loss += loss.detach().float()
+            self._acc_correct += (logits.detach().float().argmax(-1)
+                                  == sy).float().sum()
         return {"graph": g, "x": sx, "y": sy, "loss": loss, "logits": logits}
 
     def _eager_warmup(self, x, y):
@@ -129,10 +137,22 @@ class LocalClient:
         callbacks = callbacks or []
         stats = RoundStats()
         dev = self.device
+        graphed_stats = self.use_graphs and not callbacks
+        if graphed_stats and hasattr(self, "_acc_loss"):
+            self._acc_loss.zero_()
+            self._acc_correct.zero_()
         loss_sum = torch.zeros((), dtype=torch.float32, device=dev)
         acc_sum = torch.zeros((), dtype=torch.float32, device=dev)
         t0 = time.perf_counter()
         for ep in range(epochs):
+            if graphed_stats:
+                # stats accumulate inside the captured graphs: the loop body
+                # is replay + input copies only
+                for x, y in self.loader:
+                    loss, logits = self.train_step(x, y)
+                    stats.steps += 1
+                    stats.samples += y.numel()
+                continue
             ep_loss = torch.zeros((), dtype=torch.float32, device=dev)
             ep_acc = torch.zeros((), dtype=torch.float32, device=dev)
             ep_steps = ep_samples = 0
@@ -160,8 +180,12 @@ class LocalClient:
             torch.cuda.synchronize()
         stats.seconds = time.perf_counter() - t0
         if stats.steps:
-            stats.train_loss = float(loss_sum) / stats.steps
-            stats.train_acc = float(acc_sum) / max(stats.samples, 1)
+            if graphed_stats and hasattr(self, "_acc_loss"):
+                stats.train_loss = float(self._acc_loss) / stats.steps
+                stats.train_acc = float(self._acc_correct) / max(stats.samples, 1)
+            else:
+                stats.train_loss = float(loss_sum) / stats.steps
+                stats.train_acc = float(acc_sum) / max(stats.samples, 1)
         return stats
 
     def get_weights(self) -> torch.Tensor:

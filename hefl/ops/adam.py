@@ -49,9 +49,12 @@ class FusedAdam:
                 vh = v / bc2
                 p.data.add_(-lr_t * mh / (vh.sqrt() + self.eps))
 
+    _MT_CHUNK = 8192  # matches MT_CHUNK in hefl/csrc/cnn.hip
+
     def prepare_graph_state(self, device=None):
-        """Create device-side step/schedule/hyper buffers (must run OUTSIDE
-        any hipGraph capture). Created ONCE: captured graphs hold raw
+        """Create device-side step/schedule/hyper buffers plus the
+        multi-tensor chunk table (must run OUTSIDE any hipGraph capture,
+        with .grad materialized). Created ONCE: captured graphs hold raw
         pointers to these buffers, so re-creating them would leave earlier
         graphs reading freed memory."""
         if hasattr(self, "_step_t"):
@@ -62,6 +65,19 @@ class FusedAdam:
         self._sched = torch.zeros(3, dtype=torch.float32, device=dev)
         self._hyper = torch.tensor([self.lr, self.decay],
                                    dtype=torch.float32, device=dev)
+        # multi-tensor chunk table: ONE kernel updates every parameter
+        assert all(p.grad is not None for p in self.params),             "grads must be materialized before prepare_graph_state"
+        ptrs, sizes, meta = [], [], []
+        for t, (p, m, v) in enumerate(zip(self.params, self.m, self.v)):
+            ptrs.append([p.data.data_ptr(), p.grad.data_ptr(),
+                         m.data_ptr(), v.data_ptr()])
+            sizes.append(p.numel())
+            for off in range(0, p.numel(), self._MT_CHUNK):
+                meta.append([t, off])
+        self._mt_ptrs = torch.tensor(ptrs, dtype=torch.int64, device=dev)
+        self._mt_sizes = torch.tensor(sizes, dtype=torch.int64, device=dev)
+        self._mt_meta = torch.tensor(meta, dtype=torch.int64, device=dev)
+        self._mt_nchunks = len(meta)
 
     def set_lr(self, lr: float):
         self.lr = lr
@@ -71,22 +87,27 @@ class FusedAdam:
     @torch.no_grad()
     def step_graphed(self):
         """hipGraph-capturable step: the lr/bias-correction schedule advances
-        in a device buffer (adam_prep kernel), so a captured step replays with
-        zero host work. Call only on GPU params with materialized .grad,
-        after prepare_graph_state()."""
+        in a device buffer (adam_prep kernel) and ONE multi-tensor kernel
+        updates every parameter. Call only on GPU params with materialized
+        .grad, after prepare_graph_state()."""
         C = hefl.load_extension()
         C.adam_prep(self._step_t, self._sched, self._hyper,
                     self.beta1, self.beta2)
-        for p, m, v in zip(self.params, self.m, self.v):
-            C.fused_adam_sched(p.data, p.grad, m, v, self._sched,
-                               self.beta1, self.beta2, self.eps)
+        C.fused_adam_mt(self._mt_meta, self._mt_ptrs, self._mt_sizes,
+                        self._mt_nchunks, self._sched,
+                        self.beta1, self.beta2, self.eps)
 
     def zero_grad(self):
         for p in self.params:
             p.grad = None
 
     def zero_grad_(self):
-        """In-place grad zeroing (graph-capture safe: keeps buffers alive)."""
+        """In-place grad zeroing (graph-capture safe: keeps buffers alive).
+        One multi-tensor kernel once the chunk table exists."""
+        if hasattr(self, "_mt_meta"):
+            hefl.load_extension().zero_grads_mt(
+                self._mt_meta, self._mt_ptrs, self._mt_sizes, self._mt_nchunks)
+            return
         for p in self.params:
             if p.grad is not None:
                 p.grad.zero_()
