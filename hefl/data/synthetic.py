@@ -78,12 +78,14 @@ class ClientLoader:
     """Minimal epoch iterator over a shard (drop_last=False, like Keras)."""
 
     def __init__(self, ds: SyntheticMedicalImages, indices: torch.Tensor,
-                 batch_size: int, seed: int = 0, shuffle: bool = True):
+                 batch_size: int, seed: int = 0, shuffle: bool = True,
+                 augment=None):
         self.ds = ds
         self.indices = indices.clone()
         self.batch_size = int(batch_size)
         self.seed = int(seed)
         self.shuffle = shuffle
+        self.augment = augment  # callable(x) -> x, e.g. pipeline.hflip_augment
         self._epoch = 0
 
     def __len__(self) -> int:
@@ -96,4 +98,7 @@ class ClientLoader:
             order = order[torch.randperm(order.numel(), generator=g)]
         self._epoch += 1
         for i in range(0, order.numel(), self.batch_size):
-            yield self.ds.batch(order[i:i + self.batch_size])
+            x, y = self.ds.batch(order[i:i + self.batch_size])
+            if self.augment is not None:
+                x = self.augment(x)
+            yield x, y

@@ -165,6 +165,8 @@ class Pyfhel:
         FLPyfhelin.py:357-364; its version is dead code with a NameError)."""
         if self._sk is None:
             raise ValueError("keyGen must run before relinKeyGen")
+        if self._keys is None:  # keys restored from bytes, not keyGen()
+            self._keys = KeyPair(sk=self._sk, pk=self._pk)
         self._keys.relin = self.context.relin_keygen(self._sk)
 
     # ----- scalar API (reference-exact shape) -----
