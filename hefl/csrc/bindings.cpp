@@ -29,9 +29,11 @@ std::vector<torch::Tensor> maxpool2x2_fwd(torch::Tensor x);
 torch::Tensor maxpool2x2_bwd(torch::Tensor dy, torch::Tensor idx, int64_t H,
                              int64_t W);
 std::vector<torch::Tensor> softmax_xent_fwd(torch::Tensor logits,
-                                            torch::Tensor labels);
+                                            torch::Tensor labels,
+                                            torch::Tensor acc_loss,
+                                            torch::Tensor acc_correct);
 torch::Tensor softmax_xent_bwd(torch::Tensor probs, torch::Tensor labels,
-                               torch::Tensor dloss);
+                               torch::Tensor dloss, bool out_bf16);
 void fused_adam(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                 torch::Tensor v, double lr, double b1, double b2, double eps,
                 double bc1, double bc2);

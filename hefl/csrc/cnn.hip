@@ -664,19 +664,31 @@ __global__ void maxpool_bwd_kernel(const unsigned short* __restrict__ dy,
 // logits bf16 [M, C]; probs f32 out; loss = sum(-log p[label]) / M.
 // ---------------------------------------------------------------------------
 
+// Also fuses the training statistics: when acc_loss/acc_correct are given,
+// the kernel accumulates mean loss and argmax==label counts into those
+// persistent buffers — replacing a ~7-kernel torch chain per step.
 __global__ void softmax_xent_fwd_kernel(const unsigned short* __restrict__ logits,
                                         const int64_t* __restrict__ labels,
                                         float* __restrict__ probs,
-                                        float* __restrict__ loss, int M, int C) {
+                                        float* __restrict__ loss, int M, int C,
+                                        float* __restrict__ acc_loss,
+                                        float* __restrict__ acc_correct) {
     const int row = blockIdx.x * (blockDim.x / 64) + (threadIdx.x >> 6);
     const int lane = threadIdx.x & 63;
     if (row >= M) return;
     const unsigned short* lr = logits + (int64_t)row * C;
     float mx = -3.4e38f;
-    for (int c = lane; c < C; c += 64) mx = fmaxf(mx, bf2f(lr[c]));
+    int arg = 0;
+    for (int c = lane; c < C; c += 64) {
+        float v = bf2f(lr[c]);
+        if (v > mx) { mx = v; arg = c; }
+    }
 #pragma unroll
-    for (int off = 32; off > 0; off >>= 1)
-        mx = fmaxf(mx, __shfl_xor(mx, off, 64));
+    for (int off = 32; off > 0; off >>= 1) {
+        float omx = __shfl_xor(mx, off, 64);
+        int oarg = __shfl_xor(arg, off, 64);
+        if (omx > mx || (omx == mx && oarg < arg)) { mx = omx; arg = oarg; }
+    }
     float sum = 0.f;
     for (int c = lane; c < C; c += 64) sum += __expf(bf2f(lr[c]) - mx);
 #pragma unroll
@@ -687,16 +699,20 @@ __global__ void softmax_xent_fwd_kernel(const unsigned short* __restrict__ logit
     if (lane == 0) {
         int64_t lab = labels[row];
         float p = __expf(bf2f(lr[lab]) - mx) * inv;
-        atomicAdd(loss, -__logf(fmaxf(p, 1e-30f)) / M);
+        float l = -__logf(fmaxf(p, 1e-30f)) / M;
+        atomicAdd(loss, l);
+        if (acc_loss) atomicAdd(acc_loss, l);
+        if (acc_correct && arg == (int)lab) atomicAdd(acc_correct, 1.f);
     }
 }
 
 // dloss is a device scalar (graph-capture safe: no host readback of the
 // upstream gradient); scale = dloss / M.
+template <bool OUT_BF16>
 __global__ void softmax_xent_bwd_kernel(const float* __restrict__ probs,
                                         const int64_t* __restrict__ labels,
                                         const float* __restrict__ dloss,
-                                        float* __restrict__ dlogits, int64_t M,
+                                        void* __restrict__ dlogits, int64_t M,
                                         int C) {
     const float scale = dloss[0] / (float)M;
     int64_t total = M * C;
@@ -704,8 +720,11 @@ __global__ void softmax_xent_bwd_kernel(const float* __restrict__ probs,
          i += (int64_t)gridDim.x * blockDim.x) {
         int64_t row = i / C;
         int c = i % C;
-        float g = probs[i] - (labels[row] == c ? 1.f : 0.f);
-        dlogits[i] = g * scale;
+        float g = (probs[i] - (labels[row] == c ? 1.f : 0.f)) * scale;
+        if (OUT_BF16)
+            reinterpret_cast<unsigned short*>(dlogits)[i] = f2bf(g);
+        else
+            reinterpret_cast<float*>(dlogits)[i] = g;
     }
 }
 
@@ -838,7 +857,7 @@ __global__ void linear_epilogue_kernel(const float* __restrict__ y32,
 // ptrs: int64 [T][4] = {p, g, m, v} device addresses; sizes: int64 [T].
 // ---------------------------------------------------------------------------
 
-constexpr int MT_CHUNK = 8192;
+constexpr int MT_CHUNK = 2048;
 
 __global__ void fused_adam_mt_kernel(const int64_t* __restrict__ meta,
                                      const int64_t* __restrict__ ptrs,
@@ -1321,7 +1340,9 @@ torch::Tensor maxpool2x2_bwd(torch::Tensor dy, torch::Tensor idx, int64_t H,
 }
 
 std::vector<torch::Tensor> softmax_xent_fwd(torch::Tensor logits,
-                                            torch::Tensor labels) {
+                                            torch::Tensor labels,
+                                            torch::Tensor acc_loss,
+                                            torch::Tensor acc_correct) {
     CHECK_GPU(logits);
     TORCH_CHECK(logits.is_contiguous());
     const int M = (int)logits.size(0), C = (int)logits.size(1);
@@ -1333,22 +1354,34 @@ std::vector<torch::Tensor> softmax_xent_fwd(torch::Tensor logits,
                        dim3(64 * waves_per_block), 0,
                        at::cuda::getCurrentCUDAStream(), bf_ptr(logits),
                        labels.data_ptr<int64_t>(), probs.data_ptr<float>(),
-                       loss.data_ptr<float>(), M, C);
+                       loss.data_ptr<float>(), M, C,
+                       acc_loss.numel() ? acc_loss.data_ptr<float>() : nullptr,
+                       acc_correct.numel() ? acc_correct.data_ptr<float>()
+                                           : nullptr);
     return {loss, probs};
 }
 
 torch::Tensor softmax_xent_bwd(torch::Tensor probs, torch::Tensor labels,
-                               torch::Tensor dloss) {
+                               torch::Tensor dloss, bool out_bf16) {
     CHECK_GPU(probs);
     const int64_t M = probs.size(0);
     const int C = (int)probs.size(1);
-    auto dlogits = torch::empty_like(probs);
+    auto dlogits = torch::empty({M, C}, probs.options().dtype(
+                                    out_bf16 ? torch::kBFloat16 : torch::kFloat32));
     int64_t total = M * C;
     int blocks = std::min<int64_t>(ceildiv(total, 256), 2048);
-    hipLaunchKernelGGL(softmax_xent_bwd_kernel, dim3(blocks), dim3(256), 0,
-                       at::cuda::getCurrentCUDAStream(), probs.data_ptr<float>(),
-                       labels.data_ptr<int64_t>(), dloss.data_ptr<float>(),
-                       dlogits.data_ptr<float>(), M, C);
+    if (out_bf16)
+        hipLaunchKernelGGL((softmax_xent_bwd_kernel<true>), dim3(blocks),
+                           dim3(256), 0, at::cuda::getCurrentCUDAStream(),
+                           probs.data_ptr<float>(), labels.data_ptr<int64_t>(),
+                           dloss.data_ptr<float>(),
+                           (void*)bf_ptr_mut(dlogits), M, C);
+    else
+        hipLaunchKernelGGL((softmax_xent_bwd_kernel<false>), dim3(blocks),
+                           dim3(256), 0, at::cuda::getCurrentCUDAStream(),
+                           probs.data_ptr<float>(), labels.data_ptr<int64_t>(),
+                           dloss.data_ptr<float>(),
+                           (void*)dlogits.data_ptr<float>(), M, C);
     return dlogits;
 }
 

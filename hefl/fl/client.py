@@ -111,13 +111,10 @@ class LocalClient:
         with torch.cuda.graph(g, stream=side):
             self.opt.zero_grad_()
             logits = self.model(sx.to(self.compute_dtype))
-            loss = softmax_xent(logits, sy)
+            # stats accumulate INSIDE the loss kernel into persistent buffers
+            loss = softmax_xent(logits, sy, self._acc_loss, self._acc_correct)
             loss.backward()
             self.opt.step_graphed()
-            # stats accumulate INSIDE the graph into persistent buffers
-            self._acc_loss += loss.detach().float()
-            self._acc_correct += (logits.detach().float().argmax(-1)
-                                  == sy).float().sum()
         return {"graph": g, "x": sx, "y": sy, "loss": loss, "logits": logits}
 
     def _eager_warmup(self, x, y):

@@ -49,7 +49,7 @@ class FusedAdam:
                 vh = v / bc2
                 p.data.add_(-lr_t * mh / (vh.sqrt() + self.eps))
 
-    _MT_CHUNK = 8192  # matches MT_CHUNK in hefl/csrc/cnn.hip
+    _MT_CHUNK = 2048  # matches MT_CHUNK in hefl/csrc/cnn.hip
 
     def prepare_graph_state(self, device=None):
         """Create device-side step/schedule/hyper buffers plus the

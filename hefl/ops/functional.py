@@ -182,9 +182,13 @@ def linear(x, w, b=None, relu: bool = False):
 
 class _SoftmaxXentFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, logits, labels):
+    def forward(ctx, logits, labels, acc_loss=None, acc_correct=None):
         if logits.is_cuda:
-            loss, probs = _C().softmax_xent_fwd(logits, labels)
+            empty = torch.empty(0, device=logits.device)
+            loss, probs = _C().softmax_xent_fwd(
+                logits.contiguous(), labels,
+                acc_loss if acc_loss is not None else empty,
+                acc_correct if acc_correct is not None else empty)
         else:
             lf = logits.float()
             probs = F.softmax(lf, dim=-1)
@@ -201,18 +205,20 @@ class _SoftmaxXentFn(torch.autograd.Function):
             # dloss stays a device scalar (no host sync: hipGraph-capturable)
             dl = dloss if torch.is_tensor(dloss) else torch.tensor(
                 float(dloss), device=probs.device)
-            dlogits = _C().softmax_xent_bwd(probs, labels,
-                                            dl.to(probs.device, torch.float32))
-            dlogits = dlogits.to(ctx.out_dtype)
+            dlogits = _C().softmax_xent_bwd(
+                probs, labels, dl.to(probs.device, torch.float32),
+                ctx.out_dtype == torch.bfloat16)
         else:
             onehot = F.one_hot(labels, probs.shape[-1]).float()
             dlogits = (probs - onehot) * (dloss / M)
             dlogits = dlogits.to(ctx.out_dtype)
-        return dlogits, None
+        return dlogits, None, None, None
 
 
-def softmax_xent(logits, labels):
-    return _SoftmaxXentFn.apply(logits, labels)
+def softmax_xent(logits, labels, acc_loss=None, acc_correct=None):
+    """Fused softmax + CE; on GPU optionally accumulates mean-loss and
+    correct-count into persistent device buffers (stats fused in-kernel)."""
+    return _SoftmaxXentFn.apply(logits, labels, acc_loss, acc_correct)
 
 
 # ---------------------------------------------------------------------------
