@@ -86,6 +86,10 @@ class LocalClient:
         ent["x"].copy_(x)
         ent["y"].copy_(y)
         ent["graph"].replay()
+        # Adam as two direct launches against this graph's grad pointers —
+        # the captured backward STEALS fresh pooled grad tensors (stable
+        # addresses across replays), so no zero/accumulate kernels exist
+        self.opt.step_mt(ent["mt"])
         return ent["loss"], ent["logits"]
 
     def _capture(self, x: torch.Tensor, y: torch.Tensor):
@@ -108,19 +112,22 @@ class LocalClient:
             self._acc_loss = torch.zeros((), dtype=torch.float32, device=sx.device)
             self._acc_correct = torch.zeros((), dtype=torch.float32,
                                             device=sx.device)
+        # grads set to None: the captured backward steals fresh grad tensors
+        # from the graph pool; replays rewrite them in place
+        self.opt.zero_grad()
         with torch.cuda.graph(g, stream=side):
-            self.opt.zero_grad_()
             logits = self.model(sx.to(self.compute_dtype))
             # stats accumulate INSIDE the loss kernel into persistent buffers
             loss = softmax_xent(logits, sy, self._acc_loss, self._acc_correct)
             loss.backward()
-            self.opt.step_graphed()
-        return {"graph": g, "x": sx, "y": sy, "loss": loss, "logits": logits}
+        mt = self.opt.build_mt_table()  # this graph's stolen-grad pointers
+        return {"graph": g, "x": sx, "y": sy, "loss": loss, "logits": logits,
+                "mt": mt}
 
     def _eager_warmup(self, x, y):
+        self.opt.zero_grad()  # grads=None: fresh tensors each warmup step
         logits = self.model(x.to(self.compute_dtype))
         loss = softmax_xent(logits, y)
-        self.opt.zero_grad_()  # in place; first call grads are None -> no-op
         loss.backward()
         self.opt.step()
 

@@ -65,8 +65,19 @@ class FusedAdam:
         self._sched = torch.zeros(3, dtype=torch.float32, device=dev)
         self._hyper = torch.tensor([self.lr, self.decay],
                                    dtype=torch.float32, device=dev)
-        # multi-tensor chunk table: ONE kernel updates every parameter
-        assert all(p.grad is not None for p in self.params),             "grads must be materialized before prepare_graph_state"
+
+    def set_lr(self, lr: float):
+        self.lr = lr
+        if hasattr(self, "_hyper"):
+            self._hyper[0] = lr  # reaches captured graphs without re-capture
+
+    def build_mt_table(self):
+        """Multi-tensor chunk table from the CURRENT .grad pointers — built
+        per captured graph (each capture's backward steals fresh pooled grad
+        tensors, so every graph gets its own table)."""
+        assert all(p.grad is not None for p in self.params), \
+            "grads must be materialized before build_mt_table"
+        dev = self.params[0].device
         ptrs, sizes, meta = [], [], []
         for t, (p, m, v) in enumerate(zip(self.params, self.m, self.v)):
             ptrs.append([p.data.data_ptr(), p.grad.data_ptr(),
@@ -74,40 +85,29 @@ class FusedAdam:
             sizes.append(p.numel())
             for off in range(0, p.numel(), self._MT_CHUNK):
                 meta.append([t, off])
-        self._mt_ptrs = torch.tensor(ptrs, dtype=torch.int64, device=dev)
-        self._mt_sizes = torch.tensor(sizes, dtype=torch.int64, device=dev)
-        self._mt_meta = torch.tensor(meta, dtype=torch.int64, device=dev)
-        self._mt_nchunks = len(meta)
-
-    def set_lr(self, lr: float):
-        self.lr = lr
-        if hasattr(self, "_hyper"):
-            self._hyper[0] = lr  # reaches captured graphs without re-capture
+        return {"ptrs": torch.tensor(ptrs, dtype=torch.int64, device=dev),
+                "sizes": torch.tensor(sizes, dtype=torch.int64, device=dev),
+                "meta": torch.tensor(meta, dtype=torch.int64, device=dev),
+                "n": len(meta)}
 
     @torch.no_grad()
-    def step_graphed(self):
-        """hipGraph-capturable step: the lr/bias-correction schedule advances
-        in a device buffer (adam_prep kernel) and ONE multi-tensor kernel
-        updates every parameter. Call only on GPU params with materialized
-        .grad, after prepare_graph_state()."""
+    def step_mt(self, table):
+        """Two direct kernel launches (schedule advance + one multi-tensor
+        Adam over all params) against a per-graph pointer table. Runs AFTER
+        a graph replay — the replay wrote the grads the table points at."""
         C = hefl.load_extension()
         C.adam_prep(self._step_t, self._sched, self._hyper,
                     self.beta1, self.beta2)
-        C.fused_adam_mt(self._mt_meta, self._mt_ptrs, self._mt_sizes,
-                        self._mt_nchunks, self._sched,
-                        self.beta1, self.beta2, self.eps)
+        C.fused_adam_mt(table["meta"], table["ptrs"], table["sizes"],
+                        table["n"], self._sched, self.beta1, self.beta2,
+                        self.eps)
 
     def zero_grad(self):
         for p in self.params:
             p.grad = None
 
     def zero_grad_(self):
-        """In-place grad zeroing (graph-capture safe: keeps buffers alive).
-        One multi-tensor kernel once the chunk table exists."""
-        if hasattr(self, "_mt_meta"):
-            hefl.load_extension().zero_grads_mt(
-                self._mt_meta, self._mt_ptrs, self._mt_sizes, self._mt_nchunks)
-            return
+        """In-place grad zeroing (keeps buffers alive)."""
         for p in self.params:
             if p.grad is not None:
                 p.grad.zero_()
