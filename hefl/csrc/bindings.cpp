@@ -13,6 +13,16 @@ torch::Tensor modmul_scalar(torch::Tensor a, int64_t s, int64_t q);
 torch::Tensor modadd(torch::Tensor a, torch::Tensor b, int64_t q);
 torch::Tensor modsub(torch::Tensor a, torch::Tensor b, int64_t q);
 void modreduce_(torch::Tensor x, torch::Tensor qs);
+void ntt_limbs(torch::Tensor x, torch::Tensor w, torch::Tensor wsh,
+               torch::Tensor qs, int64_t L);
+void intt_limbs(torch::Tensor x, torch::Tensor winv, torch::Tensor winvsh,
+                torch::Tensor qs, torch::Tensor ninv, torch::Tensor ninvsh,
+                int64_t L);
+torch::Tensor modmul_limbs(torch::Tensor a, torch::Tensor b, torch::Tensor qs,
+                           torch::Tensor ratios, int64_t L, int64_t n);
+torch::Tensor modmul_scalar_limbs(torch::Tensor a, torch::Tensor scalars,
+                                  torch::Tensor shoups, torch::Tensor qs,
+                                  int64_t L, int64_t n);
 
 // cnn.hip
 torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
@@ -75,6 +85,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("modadd", &modadd);
     m.def("modsub", &modsub);
     m.def("modreduce_", &modreduce_, "in-place per-limb reduction after lazy sum");
+    m.def("ntt_limbs", &ntt_limbs, "fused multi-limb forward NTT [R, L, n]");
+    m.def("intt_limbs", &intt_limbs, "fused multi-limb inverse NTT");
+    m.def("modmul_limbs", &modmul_limbs);
+    m.def("modmul_scalar_limbs", &modmul_scalar_limbs);
     // CNN
     m.def("conv2d_fwd", &conv2d_fwd);
     m.def("conv2d_dgrad", &conv2d_dgrad);

@@ -315,7 +315,7 @@ conv_fwd_kernel(const unsigned short* __restrict__ x,
 // ---- dgrad: dx[m=(n,ih,iw), c] = sum_k A(m,k) * B(k,c),
 //      k = (r, s, ko) with ko FASTEST so the dy gather is contiguous;
 //      B(k, c) = w[ko, r, s, c] (strided, small tile) ----
-template <int BN, int WM, int WN, int FM, int FN>
+template <int BN, int WM, int WN, int FM, int FN, bool S1 = false>
 __global__ void __launch_bounds__(TPB)
 conv_dgrad_kernel(const unsigned short* __restrict__ dy,
                   const unsigned short* __restrict__ w,
@@ -359,9 +359,18 @@ conv_dgrad_kernel(const unsigned short* __restrict__ dy,
                 int r = rs / s.S, ss = rs % s.S;
                 int oh_num = a_ih + s.pad - r, ow_num = a_iw + s.pad - ss;
                 done = true;
-                if (oh_num >= 0 && ow_num >= 0 && oh_num % s.stride == 0 &&
-                    ow_num % s.stride == 0 && oh_num / s.stride < s.OH &&
-                    ow_num / s.stride < s.OW) {
+                if (S1) {  // stride 1: no divisibility checks or divisions
+                    if (oh_num >= 0 && ow_num >= 0 && oh_num < s.OH &&
+                        ow_num < s.OW) {
+                        copy16(dst, dy + ((((int64_t)a_n * s.OH + oh_num)
+                                              * s.OW + ow_num) * s.Kout + ko));
+                    } else {
+                        zero16(dst);
+                    }
+                } else if (oh_num >= 0 && ow_num >= 0 &&
+                           oh_num % s.stride == 0 && ow_num % s.stride == 0 &&
+                           oh_num / s.stride < s.OH &&
+                           ow_num / s.stride < s.OW) {
                     copy16(dst, dy + ((((int64_t)a_n * s.OH + oh_num / s.stride)
                                           * s.OW + ow_num / s.stride)
                                          * s.Kout + ko));
@@ -1206,12 +1215,24 @@ torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor w, int64_t stride,
     auto stream = at::cuda::getCurrentCUDAStream();
     if (C > 16) {
         dim3 grid(ceildiv(M, 64), ceildiv(C, 64));
-        hipLaunchKernelGGL((conv_dgrad_kernel<64, 2, 2, 2, 2>), grid, dim3(TPB),
-                           0, stream, bf_ptr(dy), bf_ptr(w), bf_ptr_mut(dx), s);
+        if (stride == 1)
+            hipLaunchKernelGGL((conv_dgrad_kernel<64, 2, 2, 2, 2, true>), grid,
+                               dim3(TPB), 0, stream, bf_ptr(dy), bf_ptr(w),
+                               bf_ptr_mut(dx), s);
+        else
+            hipLaunchKernelGGL((conv_dgrad_kernel<64, 2, 2, 2, 2, false>), grid,
+                               dim3(TPB), 0, stream, bf_ptr(dy), bf_ptr(w),
+                               bf_ptr_mut(dx), s);
     } else {
         dim3 grid(ceildiv(M, 64), ceildiv(C, 16));
-        hipLaunchKernelGGL((conv_dgrad_kernel<16, 4, 1, 1, 1>), grid, dim3(TPB),
-                           0, stream, bf_ptr(dy), bf_ptr(w), bf_ptr_mut(dx), s);
+        if (stride == 1)
+            hipLaunchKernelGGL((conv_dgrad_kernel<16, 4, 1, 1, 1, true>), grid,
+                               dim3(TPB), 0, stream, bf_ptr(dy), bf_ptr(w),
+                               bf_ptr_mut(dx), s);
+        else
+            hipLaunchKernelGGL((conv_dgrad_kernel<16, 4, 1, 1, 1, false>), grid,
+                               dim3(TPB), 0, stream, bf_ptr(dy), bf_ptr(w),
+                               bf_ptr_mut(dx), s);
     }
     return dx;
 }

@@ -166,6 +166,190 @@ __global__ void intt_global_stage_kernel(int64_t* __restrict__ x,
 }
 
 // ---------------------------------------------------------------------------
+// Multi-limb fused variants: one launch covers every RNS limb of a
+// [R, L, n] tensor (flat row r -> limb r % L, tables indexed per limb).
+// Cuts the per-limb Python/launch loop out of encrypt/decrypt/keyswitch.
+// ---------------------------------------------------------------------------
+
+__global__ void ntt_global_stage_ml_kernel(int64_t* __restrict__ x,
+                                           const int64_t* __restrict__ w,
+                                           const int64_t* __restrict__ wsh,
+                                           const int64_t* __restrict__ qs,
+                                           int L, int n, int m) {
+    const int64_t nhalf = n >> 1;
+    const int64_t row = blockIdx.y;
+    const int limb = (int)(row % L);
+    const uint64_t q = (uint64_t)qs[limb];
+    const int64_t* wl = w + (int64_t)limb * n;
+    const int64_t* wshl = wsh + (int64_t)limb * n;
+    int64_t base_off = row * (int64_t)n;
+    const uint32_t t = (uint32_t)(n / (2 * m));
+    for (int64_t k = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; k < nhalf;
+         k += (int64_t)gridDim.x * blockDim.x) {
+        uint32_t j = (uint32_t)(k / t);
+        uint32_t pos = (uint32_t)(k % t);
+        int64_t i0 = base_off + (int64_t)j * 2 * t + pos;
+        uint64_t U = (uint64_t)x[i0];
+        uint64_t V = mulmod_shoup((uint64_t)x[i0 + t], (uint64_t)wl[m + j],
+                                  (uint64_t)wshl[m + j], q);
+        x[i0] = (int64_t)addmod_u64(U, V, q);
+        x[i0 + t] = (int64_t)submod_u64(U, V, q);
+    }
+}
+
+__global__ void __launch_bounds__(kThreads)
+ntt_lds_ml_kernel(int64_t* __restrict__ x, const int64_t* __restrict__ w,
+                  const int64_t* __restrict__ wsh,
+                  const int64_t* __restrict__ qs, int L, int n, int nblk) {
+    extern __shared__ __attribute__((aligned(16))) int64_t smem[];
+    const int tid = threadIdx.x;
+    const int blk = blockIdx.x;
+    const int64_t row = blockIdx.y;
+    const int limb = (int)(row % L);
+    const uint64_t q = (uint64_t)qs[limb];
+    const int64_t* wl = w + (int64_t)limb * n;
+    const int64_t* wshl = wsh + (int64_t)limb * n;
+    int64_t* xr = x + row * n + (int64_t)blk * nblk;
+    for (int i = tid; i < nblk; i += kThreads) smem[i] = xr[i];
+    __syncthreads();
+    const int nb2 = nblk >> 1;
+    for (int m = n / nblk; m < n; m <<= 1) {
+        const uint32_t t = (uint32_t)(n / (2 * m));
+        for (int lb = tid; lb < nb2; lb += kThreads) {
+            uint32_t jloc = (uint32_t)lb / t;
+            uint32_t pos = (uint32_t)lb % t;
+            uint32_t base = jloc * 2 * t + pos;
+            uint32_t jglob = (uint32_t)blk * ((uint32_t)nb2 / t) + jloc;
+            uint64_t U = (uint64_t)smem[base];
+            uint64_t V = mulmod_shoup((uint64_t)smem[base + t],
+                                      (uint64_t)wl[m + jglob],
+                                      (uint64_t)wshl[m + jglob], q);
+            smem[base] = (int64_t)addmod_u64(U, V, q);
+            smem[base + t] = (int64_t)submod_u64(U, V, q);
+        }
+        __syncthreads();
+    }
+    for (int i = tid; i < nblk; i += kThreads) xr[i] = smem[i];
+}
+
+__global__ void __launch_bounds__(kThreads)
+intt_lds_ml_kernel(int64_t* __restrict__ x, const int64_t* __restrict__ winv,
+                   const int64_t* __restrict__ winvsh,
+                   const int64_t* __restrict__ qs,
+                   const int64_t* __restrict__ ninv,
+                   const int64_t* __restrict__ ninvsh, int L, int n, int nblk,
+                   int scale_here) {
+    extern __shared__ __attribute__((aligned(16))) int64_t smem[];
+    const int tid = threadIdx.x;
+    const int blk = blockIdx.x;
+    const int64_t row = blockIdx.y;
+    const int limb = (int)(row % L);
+    const uint64_t q = (uint64_t)qs[limb];
+    const int64_t* wl = winv + (int64_t)limb * n;
+    const int64_t* wshl = winvsh + (int64_t)limb * n;
+    int64_t* xr = x + row * n + (int64_t)blk * nblk;
+    for (int i = tid; i < nblk; i += kThreads) smem[i] = xr[i];
+    __syncthreads();
+    const int nb2 = nblk >> 1;
+    for (int m = n; m >= 2 * (n / nblk); m >>= 1) {
+        const int h = m >> 1;
+        const uint32_t t = (uint32_t)(n / m);
+        for (int lb = tid; lb < nb2; lb += kThreads) {
+            uint32_t jloc = (uint32_t)lb / t;
+            uint32_t pos = (uint32_t)lb % t;
+            uint32_t base = jloc * 2 * t + pos;
+            uint32_t jglob = (uint32_t)blk * ((uint32_t)nb2 / t) + jloc;
+            uint64_t U = (uint64_t)smem[base];
+            uint64_t V = (uint64_t)smem[base + t];
+            smem[base] = (int64_t)addmod_u64(U, V, q);
+            smem[base + t] = (int64_t)mulmod_shoup(
+                submod_u64(U, V, q), (uint64_t)wl[h + jglob],
+                (uint64_t)wshl[h + jglob], q);
+        }
+        __syncthreads();
+    }
+    if (scale_here) {
+        for (int i = tid; i < nblk; i += kThreads)
+            smem[i] = (int64_t)mulmod_shoup((uint64_t)smem[i],
+                                            (uint64_t)ninv[limb],
+                                            (uint64_t)ninvsh[limb], q);
+        __syncthreads();
+    }
+    for (int i = tid; i < nblk; i += kThreads) xr[i] = smem[i];
+}
+
+__global__ void intt_global_stage_ml_kernel(int64_t* __restrict__ x,
+                                            const int64_t* __restrict__ winv,
+                                            const int64_t* __restrict__ winvsh,
+                                            const int64_t* __restrict__ qs,
+                                            const int64_t* __restrict__ ninv,
+                                            const int64_t* __restrict__ ninvsh,
+                                            int L, int n, int m,
+                                            int scale_here) {
+    const int64_t nhalf = n >> 1;
+    const int64_t row = blockIdx.y;
+    const int limb = (int)(row % L);
+    const uint64_t q = (uint64_t)qs[limb];
+    const int64_t* wl = winv + (int64_t)limb * n;
+    const int64_t* wshl = winvsh + (int64_t)limb * n;
+    int64_t base_off = row * (int64_t)n;
+    const int h = m >> 1;
+    const uint32_t t = (uint32_t)(n / m);
+    for (int64_t k = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; k < nhalf;
+         k += (int64_t)gridDim.x * blockDim.x) {
+        uint32_t j = (uint32_t)(k / t);
+        uint32_t pos = (uint32_t)(k % t);
+        int64_t i0 = base_off + (int64_t)j * 2 * t + pos;
+        uint64_t U = (uint64_t)x[i0];
+        uint64_t V = (uint64_t)x[i0 + t];
+        uint64_t a = addmod_u64(U, V, q);
+        uint64_t b = mulmod_shoup(submod_u64(U, V, q), (uint64_t)wl[h + j],
+                                  (uint64_t)wshl[h + j], q);
+        if (scale_here) {
+            a = mulmod_shoup(a, (uint64_t)ninv[limb], (uint64_t)ninvsh[limb], q);
+            b = mulmod_shoup(b, (uint64_t)ninv[limb], (uint64_t)ninvsh[limb], q);
+        }
+        x[i0] = (int64_t)a;
+        x[i0 + t] = (int64_t)b;
+    }
+}
+
+// a [.., L, n] x b (numel divides a's, same limb layout) with per-limb
+// primes; ratios [L][2] = floor(2^128/q) words.
+__global__ void modmul_limbs_kernel(const int64_t* __restrict__ a,
+                                    const int64_t* __restrict__ b,
+                                    int64_t* __restrict__ out, int64_t total,
+                                    int64_t b_numel,
+                                    const int64_t* __restrict__ qs,
+                                    const int64_t* __restrict__ ratios, int L,
+                                    int64_t n) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        const int limb = (int)((i / n) % L);
+        out[i] = (int64_t)mulmod_barrett(
+            (uint64_t)a[i], (uint64_t)b[i % b_numel], (uint64_t)qs[limb],
+            (uint64_t)ratios[limb * 2], (uint64_t)ratios[limb * 2 + 1]);
+    }
+}
+
+__global__ void modmul_scalar_limbs_kernel(const int64_t* __restrict__ a,
+                                           int64_t* __restrict__ out,
+                                           int64_t total,
+                                           const int64_t* __restrict__ scalars,
+                                           const int64_t* __restrict__ shoups,
+                                           const int64_t* __restrict__ qs,
+                                           int L, int64_t n) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        const int limb = (int)((i / n) % L);
+        out[i] = (int64_t)mulmod_shoup((uint64_t)a[i],
+                                       (uint64_t)scalars[limb],
+                                       (uint64_t)shoups[limb],
+                                       (uint64_t)qs[limb]);
+    }
+}
+
+// ---------------------------------------------------------------------------
 // Pointwise modular ops
 // ---------------------------------------------------------------------------
 
@@ -355,4 +539,86 @@ void modreduce_(torch::Tensor x, torch::Tensor qs) {
     hipLaunchKernelGGL(modreduce_kernel, dim3(blocks), dim3(kThreads), 0,
                        at::cuda::getCurrentCUDAStream(), x.data_ptr<int64_t>(),
                        qs.data_ptr<int64_t>(), total, n, L);
+}
+
+// In-place fused forward NTT over x [R, L, n] (limb = row % L).
+void ntt_limbs(torch::Tensor x, torch::Tensor w, torch::Tensor wsh,
+               torch::Tensor qs, int64_t L) {
+    CHECK_CUDA_OK(x);
+    TORCH_CHECK(x.is_contiguous());
+    const int n = (int)x.size(-1);
+    const int64_t rows = x.numel() / n;
+    const int nblk = n < kNblkMax ? n : kNblkMax;
+    auto stream = at::cuda::getCurrentCUDAStream();
+    for (int m = 1; m < n / nblk; m <<= 1) {
+        int blocks = (int)std::min<int64_t>((n / 2 + kThreads - 1) / kThreads, 1024);
+        hipLaunchKernelGGL(ntt_global_stage_ml_kernel, rows_grid(blocks, rows),
+                           dim3(kThreads), 0, stream, x.data_ptr<int64_t>(),
+                           w.data_ptr<int64_t>(), wsh.data_ptr<int64_t>(),
+                           qs.data_ptr<int64_t>(), (int)L, n, m);
+    }
+    hipLaunchKernelGGL(ntt_lds_ml_kernel, rows_grid(n / nblk, rows),
+                       dim3(kThreads), nblk * sizeof(int64_t), stream,
+                       x.data_ptr<int64_t>(), w.data_ptr<int64_t>(),
+                       wsh.data_ptr<int64_t>(), qs.data_ptr<int64_t>(), (int)L,
+                       n, nblk);
+}
+
+void intt_limbs(torch::Tensor x, torch::Tensor winv, torch::Tensor winvsh,
+                torch::Tensor qs, torch::Tensor ninv, torch::Tensor ninvsh,
+                int64_t L) {
+    CHECK_CUDA_OK(x);
+    TORCH_CHECK(x.is_contiguous());
+    const int n = (int)x.size(-1);
+    const int64_t rows = x.numel() / n;
+    const int nblk = n < kNblkMax ? n : kNblkMax;
+    auto stream = at::cuda::getCurrentCUDAStream();
+    const bool has_global = nblk < n;
+    hipLaunchKernelGGL(intt_lds_ml_kernel, rows_grid(n / nblk, rows),
+                       dim3(kThreads), nblk * sizeof(int64_t), stream,
+                       x.data_ptr<int64_t>(), winv.data_ptr<int64_t>(),
+                       winvsh.data_ptr<int64_t>(), qs.data_ptr<int64_t>(),
+                       ninv.data_ptr<int64_t>(), ninvsh.data_ptr<int64_t>(),
+                       (int)L, n, nblk, has_global ? 0 : 1);
+    for (int m = n / nblk; m >= 2; m >>= 1) {
+        int blocks = (int)std::min<int64_t>((n / 2 + kThreads - 1) / kThreads, 1024);
+        hipLaunchKernelGGL(intt_global_stage_ml_kernel, rows_grid(blocks, rows),
+                           dim3(kThreads), 0, stream, x.data_ptr<int64_t>(),
+                           winv.data_ptr<int64_t>(), winvsh.data_ptr<int64_t>(),
+                           qs.data_ptr<int64_t>(), ninv.data_ptr<int64_t>(),
+                           ninvsh.data_ptr<int64_t>(), (int)L, n, m,
+                           m == 2 ? 1 : 0);
+    }
+}
+
+torch::Tensor modmul_limbs(torch::Tensor a, torch::Tensor b, torch::Tensor qs,
+                           torch::Tensor ratios, int64_t L, int64_t n) {
+    CHECK_CUDA_OK(a);
+    TORCH_CHECK(a.is_contiguous() && b.is_contiguous());
+    TORCH_CHECK(a.numel() % b.numel() == 0);
+    auto out = torch::empty_like(a);
+    int64_t total = a.numel();
+    int blocks = (int)std::min<int64_t>((total + kThreads - 1) / kThreads, 2048);
+    hipLaunchKernelGGL(modmul_limbs_kernel, dim3(blocks), dim3(kThreads), 0,
+                       at::cuda::getCurrentCUDAStream(), a.data_ptr<int64_t>(),
+                       b.data_ptr<int64_t>(), out.data_ptr<int64_t>(), total,
+                       b.numel(), qs.data_ptr<int64_t>(),
+                       ratios.data_ptr<int64_t>(), (int)L, n);
+    return out;
+}
+
+torch::Tensor modmul_scalar_limbs(torch::Tensor a, torch::Tensor scalars,
+                                  torch::Tensor shoups, torch::Tensor qs,
+                                  int64_t L, int64_t n) {
+    CHECK_CUDA_OK(a);
+    TORCH_CHECK(a.is_contiguous());
+    auto out = torch::empty_like(a);
+    int64_t total = a.numel();
+    int blocks = (int)std::min<int64_t>((total + kThreads - 1) / kThreads, 2048);
+    hipLaunchKernelGGL(modmul_scalar_limbs_kernel, dim3(blocks), dim3(kThreads),
+                       0, at::cuda::getCurrentCUDAStream(),
+                       a.data_ptr<int64_t>(), out.data_ptr<int64_t>(), total,
+                       scalars.data_ptr<int64_t>(), shoups.data_ptr<int64_t>(),
+                       qs.data_ptr<int64_t>(), (int)L, n);
+    return out;
 }

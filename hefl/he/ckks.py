@@ -96,6 +96,14 @@ class GpuBackend:
         self.ninv = ninv.to(device)
         self.ninv_shoup = torch.stack(
             [shoup(ninv[i:i + 1], q) for i, q in enumerate(primes)]).reshape(-1).to(device)
+        # Barrett ratio words floor(2^128/q) per limb, for fused modmul
+        words = []
+        for q in primes:
+            r = ((1 << 128) - 1) // q
+            words.append([np.int64(np.uint64(r & ((1 << 64) - 1))),
+                          np.int64(np.uint64(r >> 64))])
+        self.ratio_words = torch.from_numpy(
+            np.array(words, dtype=np.int64)).to(device)
 
     def ntt(self, x: torch.Tensor, limb: int, inverse: bool = False) -> torch.Tensor:
         out = x.contiguous().clone()
@@ -115,6 +123,40 @@ class GpuBackend:
     def modmul_scalar(self, a: torch.Tensor, s: int, limb: int) -> torch.Tensor:
         return self._C.modmul_scalar(a.contiguous(), int(s % self.primes[limb]),
                                      int(self.primes[limb]))
+
+    # ----- fused multi-limb ops: one launch across the first L limbs -----
+    def ntt_all(self, x: torch.Tensor, inverse: bool = False) -> torch.Tensor:
+        """x [..., L, n] -> NTT per limb in ONE fused launch set."""
+        L = x.shape[-2]
+        out = x.contiguous().clone()
+        flat = out.reshape(-1, L, self.n)
+        if inverse:
+            self._C.intt_limbs(flat, self.winv, self.winv_shoup, self.qs,
+                               self.ninv, self.ninv_shoup, L)
+        else:
+            self._C.ntt_limbs(flat, self.w, self.w_shoup, self.qs, L)
+        return out
+
+    def modmul_limbs(self, a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+        """a [..., L, n] x b ([..., L, n] or [L, n] tiling a), per-limb primes."""
+        L = a.shape[-2]
+        return self._C.modmul_limbs(a.contiguous(), b.contiguous(), self.qs,
+                                    self.ratio_words, L, self.n)
+
+    def modmul_scalar_limbs(self, a: torch.Tensor, scalars) -> torch.Tensor:
+        """a [..., L, n] x per-limb plain scalars (python ints)."""
+        L = a.shape[-2]
+        sc, sh = [], []
+        for i in range(L):
+            q = self.primes[i]
+            v = int(scalars[i]) % q
+            sc.append(v)
+            sh.append(np.int64(np.uint64((v << 64) // q)))
+        dev = a.device
+        sct = torch.tensor(sc, dtype=torch.int64, device=dev)
+        sht = torch.tensor(np.array(sh, dtype=np.int64), device=dev)
+        return self._C.modmul_scalar_limbs(a.contiguous(), sct, sht,
+                                           self.qs, L, self.n)
 
 
 # ---------------------------------------------------------------------------
@@ -263,6 +305,13 @@ class CKKSContext:
     def _to_rns_ntt(self, coeffs: torch.Tensor, nlimbs: Optional[int] = None) -> torch.Tensor:
         """Small centered int64 coeffs [..., n] -> NTT-form RNS [..., L, n]."""
         nlimbs = self.L if nlimbs is None else nlimbs
+        if self.device.type == "cuda":
+            qs = self.backend.qs[:nlimbs]
+            shape = [1] * coeffs.dim() + [1]
+            shape[-2] = nlimbs
+            rem = torch.remainder(coeffs.to(self.device).unsqueeze(-2),
+                                  qs.view(shape))
+            return self.backend.ntt_all(rem)
         out = []
         for i in range(nlimbs):
             q = self._q(i)
@@ -292,14 +341,12 @@ class CKKSContext:
         scale = self.scale if scale is None else scale
         nlimbs = self.L if nlimbs is None else nlimbs
         if self.device.type == "cuda":
-            # full-device path: special FFT + rounding on the GPU
+            # full-device path: special FFT + rounding on the GPU, then one
+            # fused multi-limb remainder + NTT
             vt = (vals if torch.is_tensor(vals)
                   else torch.from_numpy(np.asarray(vals, dtype=np.float64)))
             ct = self.encoder.encode_torch(vt.to(self.device), scale)
-            out = []
-            for i in range(nlimbs):
-                out.append(self.backend.ntt(torch.remainder(ct, self._q(i)), i))
-            return Plaintext(torch.stack(out, dim=-2), scale)
+            return Plaintext(self._to_rns_ntt(ct, nlimbs), scale)
         coeffs = self.encoder.encode(np.asarray(vals, dtype=np.float64), scale)
         out = []
         if coeffs.dtype == np.int64:
@@ -370,6 +417,13 @@ class CKKSContext:
         u = self._to_rns_ntt(self._sample_ternary(lead + (n,)), nlimbs)
         e0 = self._to_rns_ntt(self._sample_err(lead + (n,)), nlimbs)
         e1 = self._to_rns_ntt(self._sample_err(lead + (n,)), nlimbs)
+        if self.device.type == "cuda":
+            qs = self.backend.qs[:nlimbs].view((1,) * len(lead) + (nlimbs, 1))
+            bu = self.backend.modmul_limbs(u, pk[0, :nlimbs])
+            au = self.backend.modmul_limbs(u, pk[1, :nlimbs])
+            c0 = torch.remainder(bu + e0 + ptdata, qs)
+            c1 = torch.remainder(au + e1, qs)
+            return torch.stack([c0, c1], dim=-3)
         c0 = torch.empty_like(ptdata)
         c1 = torch.empty_like(ptdata)
         for i in range(nlimbs):
@@ -389,6 +443,11 @@ class CKKSContext:
         nlimbs = ctdata.shape[-2]
         c0 = ctdata[..., 0, :, :]
         c1 = ctdata[..., 1, :, :]
+        if self.device.type == "cuda":
+            cs = self.backend.modmul_limbs(c1.contiguous(), sk[:nlimbs])
+            qs = self.backend.qs[:nlimbs].view((1,) * (c0.dim() - 2)
+                                               + (nlimbs, 1))
+            return torch.remainder(c0 + cs, qs)
         out = torch.empty_like(c0)
         for i in range(nlimbs):
             cs = self.backend.modmul(c1[..., i, :], sk[i].expand_as(c1[..., i, :]).contiguous(), i)

@@ -128,3 +128,38 @@ def test_gpu_vs_cpu_identical_ntt_path():
     out_g = gpu.decode(pt_g, 16)
     og = out_g.cpu().numpy() if torch.is_tensor(out_g) else out_g
     assert np.abs(out_c - og).max() < 1e-6
+
+
+def test_fused_multilimb_ops_match_per_limb():
+    """ntt_limbs / modmul_limbs / modmul_scalar_limbs must equal the
+    per-limb kernels on identical inputs."""
+    cfg = HEConfig(m=1024, scale_bits=30, q_bits=(50, 40, 30), seed=9)
+    ctx = CKKSContext(cfg, device="cuda")
+    be = ctx.backend
+    L, n = 3, 1024
+    rng = np.random.default_rng(4)
+    x = torch.from_numpy(np.stack(
+        [rng.integers(0, ctx.primes[i], size=(5, n), dtype=np.int64)
+         for i in range(L)], axis=1)).cuda()  # [5, L, n]
+
+    fused = be.ntt_all(x)
+    per = torch.stack([be.ntt(x[:, i, :], i) for i in range(L)], dim=1)
+    assert torch.equal(fused, per)
+
+    back = be.ntt_all(fused, inverse=True)
+    assert torch.equal(back, x)
+
+    b = torch.from_numpy(np.stack(
+        [rng.integers(0, ctx.primes[i], size=(n,), dtype=np.int64)
+         for i in range(L)], axis=0)).cuda()  # [L, n]
+    mm = be.modmul_limbs(x, b)
+    per = torch.stack(
+        [be.modmul(x[:, i, :].contiguous(),
+                   b[i].expand(5, n).contiguous(), i) for i in range(L)], dim=1)
+    assert torch.equal(mm, per)
+
+    scalars = [123456789, 987654321, 55555]
+    ms = be.modmul_scalar_limbs(x, scalars)
+    per = torch.stack([be.modmul_scalar(x[:, i, :].contiguous(),
+                                        scalars[i], i) for i in range(L)], dim=1)
+    assert torch.equal(ms, per)
