@@ -157,17 +157,16 @@ struct ConvShape {
     int N, H, W, C, Kout, R, S, OH, OW, stride, pad;
 };
 
-template <int BN, int WM, int WN, int FM, int FN>
+template <int BM, int BN, int WM, int WN, int FM, int FN>
 struct ConvTile {
-    static constexpr int BM = 64;
     static constexpr int BK = 32;
     unsigned short As[BM][BK];
     unsigned short Bs[BN][BK];
 };
 
-template <int BN, int WM, int WN, int FM, int FN>
+template <int BM, int BN, int WM, int WN, int FM, int FN>
 __device__ __forceinline__ void tile_mfma(
-    const unsigned short (&As)[64][32], const unsigned short (&Bs)[BN][32],
+    const unsigned short (&As)[BM][32], const unsigned short (&Bs)[BN][32],
     f32x4 (&acc)[FM][FN], int wave, int lane) {
     const int wm = wave / WN, wn = wave % WN;
     const int half = lane >> 4, sub = lane & 15;
@@ -199,32 +198,38 @@ __device__ __forceinline__ void zero16(unsigned short* dst) {
 
 // ---- forward: y[m=(n,oh,ow), ko] = sum_k A(m,k) * w[ko, k],
 //      k = (r, s, c) with c fastest ----
-template <int BN, int WM, int WN, int FM, int FN>
+template <int BM, int BN, int WM, int WN, int FM, int FN>
 __global__ void __launch_bounds__(TPB)
 conv_fwd_kernel(const unsigned short* __restrict__ x,
                 const unsigned short* __restrict__ w,
                 const float* __restrict__ bias, unsigned short* __restrict__ y,
                 ConvShape s, int relu) {
-    __shared__ ConvTile<BN, WM, WN, FM, FN> sm;
+    __shared__ ConvTile<BM, BN, WM, WN, FM, FN> sm;
+    constexpr int RPT = BM * 4 / TPB;  // A-chunks per thread
     const int tid = threadIdx.x;
     const int wave = tid >> 6;
     const int lane = tid & 63;
-    const int m0 = blockIdx.x * 64;
+    const int m0 = blockIdx.x * BM;
     const int n0 = blockIdx.y * BN;
     const int M = s.N * s.OH * s.OW;
     const int KK = s.R * s.S * s.C;
     const bool fast = (s.C % 8 == 0);
 
-    // hoisted per-thread A-row coordinates (row = tid/4, k-chunk = tid%4)
-    const int arow = tid >> 2;
+    // hoisted per-thread A-row coordinates (RPT rows of 4 chunks each)
+    int arows[RPT], am_[RPT], a_n[RPT], a_oh[RPT], a_ow[RPT];
     const int akc = (tid & 3) * 8;
-    const int am = m0 + arow;
-    int a_n = 0, a_oh = 0, a_ow = 0;
-    if (am < M) {
-        a_n = am / (s.OH * s.OW);
-        int rem = am % (s.OH * s.OW);
-        a_oh = rem / s.OW;
-        a_ow = rem % s.OW;
+#pragma unroll
+    for (int t = 0; t < RPT; ++t) {
+        arows[t] = (tid + t * TPB) >> 2;
+        const int am = m0 + arows[t];
+        am_[t] = am;
+        a_n[t] = a_oh[t] = a_ow[t] = 0;
+        if (am < M) {
+            a_n[t] = am / (s.OH * s.OW);
+            int rem = am % (s.OH * s.OW);
+            a_oh[t] = rem / s.OW;
+            a_ow[t] = rem % s.OW;
+        }
     }
 
     f32x4 acc[FM][FN];
@@ -234,17 +239,18 @@ conv_fwd_kernel(const unsigned short* __restrict__ x,
         for (int j = 0; j < FN; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
     for (int k0 = 0; k0 < KK; k0 += 32) {
-        {  // ---- stage A (one 8-elem chunk per thread) ----
+#pragma unroll
+        for (int t = 0; t < RPT; ++t) {  // ---- stage A ----
             const int k = k0 + akc;
-            unsigned short* dst = &sm.As[arow][akc];
-            if (am < M && k < KK) {
+            unsigned short* dst = &sm.As[arows[t]][akc];
+            if (am_[t] < M && k < KK) {
                 if (fast) {
                     int rs = k / s.C, c = k % s.C;
                     int r = rs / s.S, ss = rs % s.S;
-                    int ih = a_oh * s.stride + r - s.pad;
-                    int iw = a_ow * s.stride + ss - s.pad;
+                    int ih = a_oh[t] * s.stride + r - s.pad;
+                    int iw = a_ow[t] * s.stride + ss - s.pad;
                     if (ih >= 0 && ih < s.H && iw >= 0 && iw < s.W)
-                        copy16(dst, x + ((((int64_t)a_n * s.H + ih) * s.W + iw)
+                        copy16(dst, x + ((((int64_t)a_n[t] * s.H + ih) * s.W + iw)
                                              * s.C + c));
                     else
                         zero16(dst);
@@ -257,10 +263,10 @@ conv_fwd_kernel(const unsigned short* __restrict__ x,
                             int c = kk % s.C;
                             int rs = kk / s.C;
                             int r = rs / s.S, ss = rs % s.S;
-                            int ih = a_oh * s.stride + r - s.pad;
-                            int iw = a_ow * s.stride + ss - s.pad;
+                            int ih = a_oh[t] * s.stride + r - s.pad;
+                            int iw = a_ow[t] * s.stride + ss - s.pad;
                             if (ih >= 0 && ih < s.H && iw >= 0 && iw < s.W)
-                                v = x[((((int64_t)a_n * s.H + ih) * s.W + iw)
+                                v = x[((((int64_t)a_n[t] * s.H + ih) * s.W + iw)
                                            * s.C + c)];
                         }
                         dst[j] = v;
@@ -289,7 +295,7 @@ conv_fwd_kernel(const unsigned short* __restrict__ x,
             }
         }
         __syncthreads();
-        tile_mfma<BN, WM, WN, FM, FN>(sm.As, sm.Bs, acc, wave, lane);
+        tile_mfma<BM, BN, WM, WN, FM, FN>(sm.As, sm.Bs, acc, wave, lane);
         __syncthreads();
     }
     // ---- epilogue ----
@@ -315,31 +321,37 @@ conv_fwd_kernel(const unsigned short* __restrict__ x,
 // ---- dgrad: dx[m=(n,ih,iw), c] = sum_k A(m,k) * B(k,c),
 //      k = (r, s, ko) with ko FASTEST so the dy gather is contiguous;
 //      B(k, c) = w[ko, r, s, c] (strided, small tile) ----
-template <int BN, int WM, int WN, int FM, int FN, bool S1 = false>
+template <int BM, int BN, int WM, int WN, int FM, int FN, bool S1 = false>
 __global__ void __launch_bounds__(TPB)
 conv_dgrad_kernel(const unsigned short* __restrict__ dy,
                   const unsigned short* __restrict__ w,
                   unsigned short* __restrict__ dx, ConvShape s) {
-    __shared__ unsigned short As[64][32];
+    __shared__ unsigned short As[BM][32];
     __shared__ unsigned short Bst[32][BN + 8];  // k-major: B(k, c) tile
+    constexpr int RPT = BM * 4 / TPB;
     const int tid = threadIdx.x;
     const int wave = tid >> 6;
     const int lane = tid & 63;
-    const int m0 = blockIdx.x * 64;
+    const int m0 = blockIdx.x * BM;
     const int n0 = blockIdx.y * BN;
     const int M = s.N * s.H * s.W;
     const int KK = s.Kout * s.R * s.S;
     const bool fast = (s.Kout % 8 == 0);
 
-    const int arow = tid >> 2;
+    int arows[RPT], am_[RPT], a_n[RPT], a_ih[RPT], a_iw[RPT];
     const int akc = (tid & 3) * 8;
-    const int am = m0 + arow;
-    int a_n = 0, a_ih = 0, a_iw = 0;
-    if (am < M) {
-        a_n = am / (s.H * s.W);
-        int rem = am % (s.H * s.W);
-        a_ih = rem / s.W;
-        a_iw = rem % s.W;
+#pragma unroll
+    for (int t = 0; t < RPT; ++t) {
+        arows[t] = (tid + t * TPB) >> 2;
+        const int am = m0 + arows[t];
+        am_[t] = am;
+        a_n[t] = a_ih[t] = a_iw[t] = 0;
+        if (am < M) {
+            a_n[t] = am / (s.H * s.W);
+            int rem = am % (s.H * s.W);
+            a_ih[t] = rem / s.W;
+            a_iw[t] = rem % s.W;
+        }
     }
 
     f32x4 acc[FM][FN];
@@ -349,20 +361,21 @@ conv_dgrad_kernel(const unsigned short* __restrict__ dy,
         for (int j = 0; j < FN; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
     for (int k0 = 0; k0 < KK; k0 += 32) {
-        {  // ---- stage A from dy ----
+#pragma unroll
+        for (int t = 0; t < RPT; ++t) {  // ---- stage A from dy ----
             const int k = k0 + akc;
-            unsigned short* dst = &As[arow][akc];
+            unsigned short* dst = &As[arows[t]][akc];
             bool done = false;
-            if (am < M && k < KK && fast) {
+            if (am_[t] < M && k < KK && fast) {
                 // chunk shares (r,s): ko = k % Kout, rs = k / Kout
                 int rs = k / s.Kout, ko = k % s.Kout;
                 int r = rs / s.S, ss = rs % s.S;
-                int oh_num = a_ih + s.pad - r, ow_num = a_iw + s.pad - ss;
+                int oh_num = a_ih[t] + s.pad - r, ow_num = a_iw[t] + s.pad - ss;
                 done = true;
                 if (S1) {  // stride 1: no divisibility checks or divisions
                     if (oh_num >= 0 && ow_num >= 0 && oh_num < s.OH &&
                         ow_num < s.OW) {
-                        copy16(dst, dy + ((((int64_t)a_n * s.OH + oh_num)
+                        copy16(dst, dy + ((((int64_t)a_n[t] * s.OH + oh_num)
                                               * s.OW + ow_num) * s.Kout + ko));
                     } else {
                         zero16(dst);
@@ -371,7 +384,7 @@ conv_dgrad_kernel(const unsigned short* __restrict__ dy,
                            oh_num % s.stride == 0 && ow_num % s.stride == 0 &&
                            oh_num / s.stride < s.OH &&
                            ow_num / s.stride < s.OW) {
-                    copy16(dst, dy + ((((int64_t)a_n * s.OH + oh_num / s.stride)
+                    copy16(dst, dy + ((((int64_t)a_n[t] * s.OH + oh_num / s.stride)
                                           * s.OW + ow_num / s.stride)
                                          * s.Kout + ko));
                 } else {
@@ -379,7 +392,7 @@ conv_dgrad_kernel(const unsigned short* __restrict__ dy,
                 }
             }
             if (!done) {
-                if (am < M && k < KK) {
+                if (am_[t] < M && k < KK) {
 #pragma unroll
                     for (int j = 0; j < 8; ++j) {
                         int kk = k + j;
@@ -387,14 +400,14 @@ conv_dgrad_kernel(const unsigned short* __restrict__ dy,
                         if (kk < KK) {
                             int rs = kk / s.Kout, ko = kk % s.Kout;
                             int r = rs / s.S, ss = rs % s.S;
-                            int oh_num = a_ih + s.pad - r;
-                            int ow_num = a_iw + s.pad - ss;
+                            int oh_num = a_ih[t] + s.pad - r;
+                            int ow_num = a_iw[t] + s.pad - ss;
                             if (oh_num >= 0 && ow_num >= 0 &&
                                 oh_num % s.stride == 0 &&
                                 ow_num % s.stride == 0 &&
                                 oh_num / s.stride < s.OH &&
                                 ow_num / s.stride < s.OW)
-                                v = dy[((((int64_t)a_n * s.OH +
+                                v = dy[((((int64_t)a_n[t] * s.OH +
                                           oh_num / s.stride) * s.OW +
                                          ow_num / s.stride) * s.Kout + ko)];
                         }
@@ -1185,16 +1198,21 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
     const int M = s.N * s.OH * s.OW;
     const float* bias = b.numel() ? b.data_ptr<float>() : nullptr;
     auto stream = at::cuda::getCurrentCUDAStream();
-    if (s.Kout > 16) {
+    if (s.Kout > 16 && M >= 4096) {
+        dim3 grid(ceildiv(M, 128), ceildiv(s.Kout, 64));
+        hipLaunchKernelGGL((conv_fwd_kernel<128, 64, 2, 2, 4, 2>), grid,
+                           dim3(TPB), 0, stream, bf_ptr(x), bf_ptr(w), bias,
+                           bf_ptr_mut(y), s, relu ? 1 : 0);
+    } else if (s.Kout > 16) {
         dim3 grid(ceildiv(M, 64), ceildiv(s.Kout, 64));
-        hipLaunchKernelGGL((conv_fwd_kernel<64, 2, 2, 2, 2>), grid, dim3(TPB),
-                           0, stream, bf_ptr(x), bf_ptr(w), bias, bf_ptr_mut(y),
-                           s, relu ? 1 : 0);
+        hipLaunchKernelGGL((conv_fwd_kernel<64, 64, 2, 2, 2, 2>), grid,
+                           dim3(TPB), 0, stream, bf_ptr(x), bf_ptr(w), bias,
+                           bf_ptr_mut(y), s, relu ? 1 : 0);
     } else {
         dim3 grid(ceildiv(M, 64), ceildiv(s.Kout, 16));
-        hipLaunchKernelGGL((conv_fwd_kernel<16, 4, 1, 1, 1>), grid, dim3(TPB),
-                           0, stream, bf_ptr(x), bf_ptr(w), bias, bf_ptr_mut(y),
-                           s, relu ? 1 : 0);
+        hipLaunchKernelGGL((conv_fwd_kernel<64, 16, 4, 1, 1, 1>), grid,
+                           dim3(TPB), 0, stream, bf_ptr(x), bf_ptr(w), bias,
+                           bf_ptr_mut(y), s, relu ? 1 : 0);
     }
     return y;
 }
@@ -1213,26 +1231,36 @@ torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor w, int64_t stride,
     auto dx = torch::empty({N, (int64_t)H, (int64_t)W, C}, dy.options());
     const int M = N * (int)H * (int)W;
     auto stream = at::cuda::getCurrentCUDAStream();
-    if (C > 16) {
+    if (C > 16 && M >= 4096) {
+        dim3 grid(ceildiv(M, 128), ceildiv(C, 64));
+        if (stride == 1)
+            hipLaunchKernelGGL((conv_dgrad_kernel<128, 64, 2, 2, 4, 2, true>),
+                               grid, dim3(TPB), 0, stream, bf_ptr(dy),
+                               bf_ptr(w), bf_ptr_mut(dx), s);
+        else
+            hipLaunchKernelGGL((conv_dgrad_kernel<128, 64, 2, 2, 4, 2, false>),
+                               grid, dim3(TPB), 0, stream, bf_ptr(dy),
+                               bf_ptr(w), bf_ptr_mut(dx), s);
+    } else if (C > 16) {
         dim3 grid(ceildiv(M, 64), ceildiv(C, 64));
         if (stride == 1)
-            hipLaunchKernelGGL((conv_dgrad_kernel<64, 2, 2, 2, 2, true>), grid,
-                               dim3(TPB), 0, stream, bf_ptr(dy), bf_ptr(w),
-                               bf_ptr_mut(dx), s);
+            hipLaunchKernelGGL((conv_dgrad_kernel<64, 64, 2, 2, 2, 2, true>),
+                               grid, dim3(TPB), 0, stream, bf_ptr(dy),
+                               bf_ptr(w), bf_ptr_mut(dx), s);
         else
-            hipLaunchKernelGGL((conv_dgrad_kernel<64, 2, 2, 2, 2, false>), grid,
-                               dim3(TPB), 0, stream, bf_ptr(dy), bf_ptr(w),
-                               bf_ptr_mut(dx), s);
+            hipLaunchKernelGGL((conv_dgrad_kernel<64, 64, 2, 2, 2, 2, false>),
+                               grid, dim3(TPB), 0, stream, bf_ptr(dy),
+                               bf_ptr(w), bf_ptr_mut(dx), s);
     } else {
         dim3 grid(ceildiv(M, 64), ceildiv(C, 16));
         if (stride == 1)
-            hipLaunchKernelGGL((conv_dgrad_kernel<16, 4, 1, 1, 1, true>), grid,
-                               dim3(TPB), 0, stream, bf_ptr(dy), bf_ptr(w),
-                               bf_ptr_mut(dx), s);
+            hipLaunchKernelGGL((conv_dgrad_kernel<64, 16, 4, 1, 1, 1, true>),
+                               grid, dim3(TPB), 0, stream, bf_ptr(dy),
+                               bf_ptr(w), bf_ptr_mut(dx), s);
         else
-            hipLaunchKernelGGL((conv_dgrad_kernel<16, 4, 1, 1, 1, false>), grid,
-                               dim3(TPB), 0, stream, bf_ptr(dy), bf_ptr(w),
-                               bf_ptr_mut(dx), s);
+            hipLaunchKernelGGL((conv_dgrad_kernel<64, 16, 4, 1, 1, 1, false>),
+                               grid, dim3(TPB), 0, stream, bf_ptr(dy),
+                               bf_ptr(w), bf_ptr_mut(dx), s);
     }
     return dx;
 }
