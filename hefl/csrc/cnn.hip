@@ -1198,7 +1198,7 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
     const int M = s.N * s.OH * s.OW;
     const float* bias = b.numel() ? b.data_ptr<float>() : nullptr;
     auto stream = at::cuda::getCurrentCUDAStream();
-    if (s.Kout > 16 && M >= 4096) {
+    if (false) {  // 128-tile: measured neutral-to-worse (barrier-bound)
         dim3 grid(ceildiv(M, 128), ceildiv(s.Kout, 64));
         hipLaunchKernelGGL((conv_fwd_kernel<128, 64, 2, 2, 4, 2>), grid,
                            dim3(TPB), 0, stream, bf_ptr(x), bf_ptr(w), bias,
@@ -1231,7 +1231,7 @@ torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor w, int64_t stride,
     auto dx = torch::empty({N, (int64_t)H, (int64_t)W, C}, dy.options());
     const int M = N * (int)H * (int)W;
     auto stream = at::cuda::getCurrentCUDAStream();
-    if (C > 16 && M >= 4096) {
+    if (false) {  // 128-tile: measured neutral-to-worse (barrier-bound)
         dim3 grid(ceildiv(M, 128), ceildiv(C, 64));
         if (stride == 1)
             hipLaunchKernelGGL((conv_dgrad_kernel<128, 64, 2, 2, 4, 2, true>),
