@@ -318,6 +318,134 @@ conv_fwd_kernel(const unsigned short* __restrict__ x,
     }
 }
 
+// ---- glds-pipelined forward: 2 LDS buffers, global_load_lds staging
+// (dest is lane-linear: thread t's chunk lands at byte t*16 of its wave's
+// 1 KiB slice), stage of tile k+1 issued BEFORE the MFMAs of tile k so the
+// DMA flight hides under compute; __syncthreads drains it at the barrier.
+// Out-of-range lanes (M tail, padding border, K tail) read a 16-B zero
+// scratch buffer instead of branching around the DMA. Requires C % 8 == 0
+// and KK % 8 == 0 (wrapper falls back to the gather kernel otherwise). ----
+
+typedef uint32_t __attribute__((address_space(3))) lds_u32_t;
+typedef const uint32_t __attribute__((address_space(1))) glb_u32_t;
+
+__device__ __forceinline__ void glds16(const unsigned short* src, void* lds_base) {
+    __builtin_amdgcn_global_load_lds((glb_u32_t*)src, (lds_u32_t*)lds_base, 16,
+                                     0, 0);
+}
+
+template <int BM, int BN, int WM, int WN, int FM, int FN>
+__global__ void __launch_bounds__(TPB)
+conv_fwd_glds_kernel(const unsigned short* __restrict__ x,
+                     const unsigned short* __restrict__ w,
+                     const float* __restrict__ bias,
+                     unsigned short* __restrict__ y,
+                     const unsigned short* __restrict__ zbuf, ConvShape s,
+                     int relu) {
+    // ONE shared object (two makes hipcc emit a vmcnt(0) drain before each
+    // k-step's first ds_read, defeating the glds pipeline — guide trap 4a)
+    __shared__ unsigned short smem[2 * (BM + BN) * 32];
+    auto As = [&](int buf) -> unsigned short (*)[32] {
+        return reinterpret_cast<unsigned short(*)[32]>(smem + buf * (BM + BN) * 32);
+    };
+    auto Bs = [&](int buf) -> unsigned short (*)[32] {
+        return reinterpret_cast<unsigned short(*)[32]>(smem + buf * (BM + BN) * 32
+                                                       + BM * 32);
+    };
+    constexpr int RPT = BM * 4 / TPB;
+    constexpr int BPT = BN * 4 / TPB;  // B chunks per thread (>=1 when BN=64)
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+    const int m0 = blockIdx.x * BM;
+    const int n0 = blockIdx.y * BN;
+    const int M = s.N * s.OH * s.OW;
+    const int KK = s.R * s.S * s.C;
+
+    int a_n[RPT], a_oh[RPT], a_ow[RPT];
+    bool a_ok[RPT];
+    const int akc = (tid & 3) * 8;
+#pragma unroll
+    for (int t = 0; t < RPT; ++t) {
+        const int am = m0 + ((tid + t * TPB) >> 2);
+        a_ok[t] = am < M;
+        a_n[t] = a_oh[t] = a_ow[t] = 0;
+        if (a_ok[t]) {
+            a_n[t] = am / (s.OH * s.OW);
+            int rem = am % (s.OH * s.OW);
+            a_oh[t] = rem / s.OW;
+            a_ow[t] = rem % s.OW;
+        }
+    }
+    const int bcol0 = tid >> 2;  // first B column this thread stages
+
+    auto stage = [&](int buf, int k0) {
+        // A: RPT chunks, each one 16-B DMA; lane-linear within the wave
+#pragma unroll
+        for (int t = 0; t < RPT; ++t) {
+            const int k = k0 + akc;
+            const unsigned short* src = zbuf;
+            if (a_ok[t] && k < KK) {
+                int rs = k / s.C, c = k % s.C;
+                int r = rs / s.S, ss = rs % s.S;
+                int ih = a_oh[t] * s.stride + r - s.pad;
+                int iw = a_ow[t] * s.stride + ss - s.pad;
+                if (ih >= 0 && ih < s.H && iw >= 0 && iw < s.W)
+                    src = x + ((((int64_t)a_n[t] * s.H + ih) * s.W + iw) * s.C
+                               + c);
+            }
+            char* base = (char*)&As(buf)[0][0] + (wave + t * 4) * 1024;
+            glds16(src, base);
+        }
+#pragma unroll
+        for (int t = 0; t < BPT; ++t) {
+            const int i = tid + t * TPB;
+            const int ko = n0 + (i >> 2);
+            const int k = k0 + (i & 3) * 8;
+            const unsigned short* src = zbuf;
+            if (ko < s.Kout && k < KK) src = w + (int64_t)ko * KK + k;
+            char* base = (char*)&Bs(buf)[0][0] + (wave + t * 4) * 1024;
+            glds16(src, base);
+        }
+    };
+
+    f32x4 acc[FM][FN];
+#pragma unroll
+    for (int i = 0; i < FM; ++i)
+#pragma unroll
+        for (int j = 0; j < FN; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+    stage(0, 0);
+    __syncthreads();  // drains the DMA (vmcnt 0) + barrier
+    int buf = 0;
+    for (int k0 = 0; k0 < KK; k0 += 32) {
+        if (k0 + 32 < KK) stage(buf ^ 1, k0 + 32);  // issue BEFORE compute
+        tile_mfma<BM, BN, WM, WN, FM, FN>(
+            *reinterpret_cast<const unsigned short(*)[BM][32]>(As(buf)),
+            *reinterpret_cast<const unsigned short(*)[BN][32]>(Bs(buf)), acc,
+            wave, lane);
+        __syncthreads();
+        buf ^= 1;
+    }
+    const int wm = wave / WN, wn = wave % WN;
+#pragma unroll
+    for (int j = 0; j < FN; ++j) {
+        const int col = n0 + wn * FN * 16 + j * 16 + (lane & 15);
+        if (col >= s.Kout) continue;
+        const float bv = bias ? bias[col] : 0.f;
+#pragma unroll
+        for (int i = 0; i < FM; ++i)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int row = m0 + wm * FM * 16 + i * 16 + (lane >> 4) * 4 + r;
+                if (row >= M) continue;
+                float v = acc[i][j][r] + bv;
+                if (relu) v = v > 0.f ? v : 0.f;
+                y[(int64_t)row * s.Kout + col] = f2bf(v);
+            }
+    }
+}
+
 // ---- dgrad: dx[m=(n,ih,iw), c] = sum_k A(m,k) * B(k,c),
 //      k = (r, s, ko) with ko FASTEST so the dy gather is contiguous;
 //      B(k, c) = w[ko, r, s, c] (strided, small tile) ----
@@ -1198,11 +1326,16 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
     const int M = s.N * s.OH * s.OW;
     const float* bias = b.numel() ? b.data_ptr<float>() : nullptr;
     auto stream = at::cuda::getCurrentCUDAStream();
-    if (false) {  // 128-tile: measured neutral-to-worse (barrier-bound)
-        dim3 grid(ceildiv(M, 128), ceildiv(s.Kout, 64));
-        hipLaunchKernelGGL((conv_fwd_kernel<128, 64, 2, 2, 4, 2>), grid,
+    const int KKf = s.R * s.S * s.C;
+    if (s.Kout > 16 && s.C % 8 == 0 && KKf % 8 == 0) {
+        // glds double-buffered pipeline (DMA flight hides under MFMA)
+        static torch::Tensor zbuf;
+        if (!zbuf.defined() || zbuf.device() != x.device())
+            zbuf = torch::zeros({8}, x.options());
+        dim3 grid(ceildiv(M, 64), ceildiv(s.Kout, 64));
+        hipLaunchKernelGGL((conv_fwd_glds_kernel<64, 64, 2, 2, 2, 2>), grid,
                            dim3(TPB), 0, stream, bf_ptr(x), bf_ptr(w), bias,
-                           bf_ptr_mut(y), s, relu ? 1 : 0);
+                           bf_ptr_mut(y), bf_ptr(zbuf), s, relu ? 1 : 0);
     } else if (s.Kout > 16) {
         dim3 grid(ceildiv(M, 64), ceildiv(s.Kout, 64));
         hipLaunchKernelGGL((conv_fwd_kernel<64, 64, 2, 2, 2, 2>), grid,
