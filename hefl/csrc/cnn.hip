@@ -611,6 +611,141 @@ conv_dgrad_kernel(const unsigned short* __restrict__ dy,
     }
 }
 
+// ---- glds-pipelined dgrad (same 2-buffer DMA structure as fwd; B tile is
+// k-major so its staging is also lane-linear). Requires stride==1 paths NOT
+// needed: S1 handled in the source-address computation. ----
+template <int BM, int BN, int WM, int WN, int FM, int FN, bool S1>
+__global__ void __launch_bounds__(TPB)
+conv_dgrad_glds_kernel(const unsigned short* __restrict__ dy,
+                       const unsigned short* __restrict__ w,
+                       unsigned short* __restrict__ dx,
+                       const unsigned short* __restrict__ zbuf, ConvShape s) {
+    __shared__ unsigned short smem[2 * (BM + BN) * 32];
+    auto As = [&](int buf) -> unsigned short (*)[32] {
+        return reinterpret_cast<unsigned short(*)[32]>(smem + buf * (BM + BN) * 32);
+    };
+    auto Bst = [&](int buf) -> unsigned short (*)[BN] {
+        return reinterpret_cast<unsigned short(*)[BN]>(smem + buf * (BM + BN) * 32
+                                                        + BM * 32);
+    };
+    constexpr int RPT = BM * 4 / TPB;
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+    const int m0 = blockIdx.x * BM;
+    const int n0 = blockIdx.y * BN;
+    const int M = s.N * s.H * s.W;
+    const int KK = s.Kout * s.R * s.S;
+
+    int a_n[RPT], a_ih[RPT], a_iw[RPT];
+    bool a_ok[RPT];
+    const int akc = (tid & 3) * 8;
+#pragma unroll
+    for (int t = 0; t < RPT; ++t) {
+        const int am = m0 + ((tid + t * TPB) >> 2);
+        a_ok[t] = am < M;
+        a_n[t] = a_ih[t] = a_iw[t] = 0;
+        if (a_ok[t]) {
+            a_n[t] = am / (s.H * s.W);
+            int rem = am % (s.H * s.W);
+            a_ih[t] = rem / s.W;
+            a_iw[t] = rem % s.W;
+        }
+    }
+
+    auto stage = [&](int buf, int k0) {
+#pragma unroll
+        for (int t = 0; t < RPT; ++t) {
+            const int k = k0 + akc;
+            const unsigned short* src = zbuf;
+            if (a_ok[t] && k < KK) {
+                int rs = k / s.Kout, ko = k % s.Kout;
+                int r = rs / s.S, ss = rs % s.S;
+                int oh_num = a_ih[t] + s.pad - r, ow_num = a_iw[t] + s.pad - ss;
+                if (S1) {
+                    if (oh_num >= 0 && ow_num >= 0 && oh_num < s.OH &&
+                        ow_num < s.OW)
+                        src = dy + ((((int64_t)a_n[t] * s.OH + oh_num) * s.OW
+                                     + ow_num) * s.Kout + ko);
+                } else if (oh_num >= 0 && ow_num >= 0 &&
+                           oh_num % s.stride == 0 && ow_num % s.stride == 0 &&
+                           oh_num / s.stride < s.OH &&
+                           ow_num / s.stride < s.OW) {
+                    src = dy + ((((int64_t)a_n[t] * s.OH + oh_num / s.stride)
+                                 * s.OW + ow_num / s.stride) * s.Kout + ko);
+                }
+            }
+            char* base = (char*)&As(buf)[0][0] + (wave + t * 4) * 1024;
+            glds16(src, base);
+        }
+        {   // B: Bst[k][c..c+8], chunk i = tid -> byte tid*16 (BN == 64)
+            const int kk = tid >> 3;
+            const int cc = (tid & 7) * 8;
+            const int k = k0 + kk;
+            const int c = n0 + cc;
+            const unsigned short* src = zbuf;
+            if (k < KK && c < s.C) {
+                int rs = k / s.Kout, ko = k % s.Kout;
+                src = w + ((int64_t)ko * s.R * s.S + rs) * s.C + c;
+            }
+            char* base = (char*)&Bst(buf)[0][0] + wave * 1024;
+            glds16(src, base);
+        }
+    };
+
+    f32x4 acc[FM][FN];
+#pragma unroll
+    for (int i = 0; i < FM; ++i)
+#pragma unroll
+        for (int j = 0; j < FN; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+    stage(0, 0);
+    __syncthreads();
+    int buf = 0;
+    for (int k0 = 0; k0 < KK; k0 += 32) {
+        if (k0 + 32 < KK) stage(buf ^ 1, k0 + 32);
+        {
+            const int wm = wave / WN, wn = wave % WN;
+            const int half = lane >> 4, sub = lane & 15;
+            bf16x8 a[FM], b[FN];
+#pragma unroll
+            for (int i = 0; i < FM; ++i)
+                a[i] = *reinterpret_cast<const bf16x8*>(
+                    &As(buf)[wm * FM * 16 + i * 16 + sub][half * 8]);
+#pragma unroll
+            for (int j = 0; j < FN; ++j) {
+                const int col = wn * FN * 16 + j * 16 + sub;
+#pragma unroll
+                for (int t = 0; t < 8; ++t)
+                    b[j][t] = *reinterpret_cast<const bf16_t*>(
+                        &Bst(buf)[half * 8 + t][col]);
+            }
+#pragma unroll
+            for (int i = 0; i < FM; ++i)
+#pragma unroll
+                for (int j = 0; j < FN; ++j)
+                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        a[i], b[j], acc[i][j], 0, 0, 0);
+        }
+        __syncthreads();
+        buf ^= 1;
+    }
+    const int wm = wave / WN, wn = wave % WN;
+#pragma unroll
+    for (int j = 0; j < FN; ++j) {
+        const int col = n0 + wn * FN * 16 + j * 16 + (lane & 15);
+        if (col >= s.C) continue;
+#pragma unroll
+        for (int i = 0; i < FM; ++i)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int row = m0 + wm * FM * 16 + i * 16 + (lane >> 4) * 4 + r;
+                if (row >= M) continue;
+                dx[(int64_t)row * s.C + col] = f2bf(acc[i][j][r]);
+            }
+    }
+}
+
 // ---- wgrad: dw[ko, rsc] = sum_pix dy[pix, ko] * xcol[pix, rsc].
 // LDS tiles are PIXEL-major (Dys[32][64], Xs[32][BN]) so BOTH global
 // gathers are 16-B vector loads (ko contiguous in dy; c contiguous in x);
@@ -1364,7 +1499,20 @@ torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor w, int64_t stride,
     auto dx = torch::empty({N, (int64_t)H, (int64_t)W, C}, dy.options());
     const int M = N * (int)H * (int)W;
     auto stream = at::cuda::getCurrentCUDAStream();
-    if (false) {  // 128-tile: measured neutral-to-worse (barrier-bound)
+    if (C > 16 && C % 8 == 0 && s.Kout % 8 == 0) {
+        static torch::Tensor zbuf;
+        if (!zbuf.defined() || zbuf.device() != dy.device())
+            zbuf = torch::zeros({8}, dy.options());
+        dim3 grid(ceildiv(M, 64), ceildiv(C, 64));
+        if (stride == 1)
+            hipLaunchKernelGGL((conv_dgrad_glds_kernel<64, 64, 2, 2, 2, 2, true>),
+                               grid, dim3(TPB), 0, stream, bf_ptr(dy),
+                               bf_ptr(w), bf_ptr_mut(dx), bf_ptr(zbuf), s);
+        else
+            hipLaunchKernelGGL((conv_dgrad_glds_kernel<64, 64, 2, 2, 2, 2, false>),
+                               grid, dim3(TPB), 0, stream, bf_ptr(dy),
+                               bf_ptr(w), bf_ptr_mut(dx), bf_ptr(zbuf), s);
+    } else if (false) {  // 128-tile: measured neutral-to-worse (barrier-bound)
         dim3 grid(ceildiv(M, 128), ceildiv(C, 64));
         if (stride == 1)
             hipLaunchKernelGGL((conv_dgrad_kernel<128, 64, 2, 2, 4, 2, true>),
