@@ -1192,16 +1192,32 @@ __global__ void bn_partial_kernel(const unsigned short* __restrict__ x,
                                   int rows_per_block) {
     const int64_t r0 = (int64_t)blockIdx.x * rows_per_block;
     const int64_t r1 = min(r0 + rows_per_block, M);
-    for (int c = threadIdx.x; c < C; c += blockDim.x) {
-        float acc = 0.f, acc2 = 0.f;
-        for (int64_t r = r0; r < r1; ++r) {
-            float v = bf2f(x[r * C + c]);
-            acc += v;
-            acc2 += v * v;
+    if (C >= (int)blockDim.x) {  // one channel per thread, strided
+        for (int c = threadIdx.x; c < C; c += blockDim.x) {
+            float acc = 0.f, acc2 = 0.f;
+            for (int64_t r = r0; r < r1; ++r) {
+                float v = bf2f(x[r * C + c]);
+                acc += v;
+                acc2 += v * v;
+            }
+            atomicAdd(gsum + c, acc);
+            atomicAdd(gsq + c, acc2);
         }
-        atomicAdd(gsum + c, acc);
-        atomicAdd(gsq + c, acc2);
+        return;
     }
+    // thread = (channel, row-lane): all 256 lanes stay busy when C < 256
+    const int lanes = (int)blockDim.x / C;           // row-parallel lanes
+    const int c = threadIdx.x % C;
+    const int rl = threadIdx.x / C;
+    if (rl >= lanes) return;
+    float acc = 0.f, acc2 = 0.f;
+    for (int64_t r = r0 + rl; r < r1; r += lanes) {
+        float v = bf2f(x[r * C + c]);
+        acc += v;
+        acc2 += v * v;
+    }
+    atomicAdd(gsum + c, acc);
+    atomicAdd(gsq + c, acc2);
 }
 
 __global__ void bn_finalize_kernel(const float* __restrict__ gsum,
@@ -1245,17 +1261,33 @@ __global__ void bn_bwd_partial_kernel(const unsigned short* __restrict__ dy,
                                       int C, int rows_per_block) {
     const int64_t r0 = (int64_t)blockIdx.x * rows_per_block;
     const int64_t r1 = min(r0 + rows_per_block, M);
-    for (int c = threadIdx.x; c < C; c += blockDim.x) {
-        float dg = 0.f, db = 0.f;
-        const float mu = mean[c], is = invstd[c];
-        for (int64_t r = r0; r < r1; ++r) {
-            float g = bf2f(dy[r * C + c]);
-            dg += g * (bf2f(x[r * C + c]) - mu) * is;
-            db += g;
+    if (C >= (int)blockDim.x) {
+        for (int c = threadIdx.x; c < C; c += blockDim.x) {
+            float dg = 0.f, db = 0.f;
+            const float mu = mean[c], is = invstd[c];
+            for (int64_t r = r0; r < r1; ++r) {
+                float g = bf2f(dy[r * C + c]);
+                dg += g * (bf2f(x[r * C + c]) - mu) * is;
+                db += g;
+            }
+            atomicAdd(dgamma + c, dg);
+            atomicAdd(dbeta + c, db);
         }
-        atomicAdd(dgamma + c, dg);
-        atomicAdd(dbeta + c, db);
+        return;
     }
+    const int lanes = (int)blockDim.x / C;
+    const int c = threadIdx.x % C;
+    const int rl = threadIdx.x / C;
+    if (rl >= lanes) return;
+    float dg = 0.f, db = 0.f;
+    const float mu = mean[c], is = invstd[c];
+    for (int64_t r = r0 + rl; r < r1; r += lanes) {
+        float g = bf2f(dy[r * C + c]);
+        dg += g * (bf2f(x[r * C + c]) - mu) * is;
+        db += g;
+    }
+    atomicAdd(dgamma + c, dg);
+    atomicAdd(dbeta + c, db);
 }
 
 __global__ void bn_dx_kernel(const unsigned short* __restrict__ dy,
