@@ -786,7 +786,7 @@ conv_wgrad_kernel(const unsigned short* __restrict__ dy,
             const int kpix = k0 + pix;
             const int ko = m0 + koc;
             unsigned short* dst = &Dys[pix][koc];
-            if (kpix < kend && ko < M && fast_dy) {
+            if (kpix < kend && ko + 8 <= M && fast_dy) {
                 copy16(dst, dy + (int64_t)kpix * s.Kout + ko);
             } else if (kpix < kend && ko < M) {
 #pragma unroll
