@@ -314,6 +314,112 @@ __global__ void intt_global_stage_ml_kernel(int64_t* __restrict__ x,
     }
 }
 
+// Radix-4 fused cross-block stages for n = 4*nblk (n = 2^15): ONE global
+// pass instead of two. Forward: CT stages m=1 (w[1]) then m=2 (w[2], w[3]).
+__global__ void ntt_global_radix4_ml_kernel(int64_t* __restrict__ x,
+                                            const int64_t* __restrict__ w,
+                                            const int64_t* __restrict__ wsh,
+                                            const int64_t* __restrict__ qs,
+                                            int L, int n) {
+    const int64_t quarter = n >> 2;
+    const int64_t row = blockIdx.y;
+    const int limb = (int)(row % L);
+    const uint64_t q = (uint64_t)qs[limb];
+    const int64_t* wl = w + (int64_t)limb * n;
+    const int64_t* wshl = wsh + (int64_t)limb * n;
+    int64_t* xr = x + row * (int64_t)n;
+    for (int64_t k = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         k < quarter; k += (int64_t)gridDim.x * blockDim.x) {
+        uint64_t x0 = (uint64_t)xr[k];
+        uint64_t x1 = (uint64_t)xr[k + quarter];
+        uint64_t x2 = (uint64_t)xr[k + 2 * quarter];
+        uint64_t x3 = (uint64_t)xr[k + 3 * quarter];
+        // stage m=1 (t = n/2, twiddle w[1]): pairs (x0,x2), (x1,x3)
+        {
+            uint64_t W = (uint64_t)wl[1], Wsh = (uint64_t)wshl[1];
+            uint64_t v = mulmod_shoup(x2, W, Wsh, q);
+            uint64_t u = x0;
+            x0 = addmod_u64(u, v, q);
+            x2 = submod_u64(u, v, q);
+            v = mulmod_shoup(x3, W, Wsh, q);
+            u = x1;
+            x1 = addmod_u64(u, v, q);
+            x3 = submod_u64(u, v, q);
+        }
+        // stage m=2 (t = n/4): pair (x0,x1) w[2]; pair (x2,x3) w[3]
+        {
+            uint64_t W = (uint64_t)wl[2], Wsh = (uint64_t)wshl[2];
+            uint64_t v = mulmod_shoup(x1, W, Wsh, q);
+            uint64_t u = x0;
+            x0 = addmod_u64(u, v, q);
+            x1 = submod_u64(u, v, q);
+            W = (uint64_t)wl[3];
+            Wsh = (uint64_t)wshl[3];
+            v = mulmod_shoup(x3, W, Wsh, q);
+            u = x2;
+            x2 = addmod_u64(u, v, q);
+            x3 = submod_u64(u, v, q);
+        }
+        xr[k] = (int64_t)x0;
+        xr[k + quarter] = (int64_t)x1;
+        xr[k + 2 * quarter] = (int64_t)x2;
+        xr[k + 3 * quarter] = (int64_t)x3;
+    }
+}
+
+// Inverse: GS stages m=4 (winv[2], winv[3]) then m=2 (winv[1]) + 1/n scale.
+__global__ void intt_global_radix4_ml_kernel(int64_t* __restrict__ x,
+                                             const int64_t* __restrict__ winv,
+                                             const int64_t* __restrict__ winvsh,
+                                             const int64_t* __restrict__ qs,
+                                             const int64_t* __restrict__ ninv,
+                                             const int64_t* __restrict__ ninvsh,
+                                             int L, int n) {
+    const int64_t quarter = n >> 2;
+    const int64_t row = blockIdx.y;
+    const int limb = (int)(row % L);
+    const uint64_t q = (uint64_t)qs[limb];
+    const int64_t* wl = winv + (int64_t)limb * n;
+    const int64_t* wshl = winvsh + (int64_t)limb * n;
+    const uint64_t nv = (uint64_t)ninv[limb], nvs = (uint64_t)ninvsh[limb];
+    int64_t* xr = x + row * (int64_t)n;
+    for (int64_t k = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         k < quarter; k += (int64_t)gridDim.x * blockDim.x) {
+        uint64_t x0 = (uint64_t)xr[k];
+        uint64_t x1 = (uint64_t)xr[k + quarter];
+        uint64_t x2 = (uint64_t)xr[k + 2 * quarter];
+        uint64_t x3 = (uint64_t)xr[k + 3 * quarter];
+        // stage m=4 (t = n/4): pairs (x0,x1) winv[2]; (x2,x3) winv[3]
+        {
+            uint64_t W = (uint64_t)wl[2], Wsh = (uint64_t)wshl[2];
+            uint64_t u = x0, v = x1;
+            x0 = addmod_u64(u, v, q);
+            x1 = mulmod_shoup(submod_u64(u, v, q), W, Wsh, q);
+            W = (uint64_t)wl[3];
+            Wsh = (uint64_t)wshl[3];
+            u = x2;
+            v = x3;
+            x2 = addmod_u64(u, v, q);
+            x3 = mulmod_shoup(submod_u64(u, v, q), W, Wsh, q);
+        }
+        // stage m=2 (t = n/2): pairs (x0,x2) winv[1]; (x1,x3) winv[1]
+        {
+            uint64_t W = (uint64_t)wl[1], Wsh = (uint64_t)wshl[1];
+            uint64_t u = x0, v = x2;
+            x0 = addmod_u64(u, v, q);
+            x2 = mulmod_shoup(submod_u64(u, v, q), W, Wsh, q);
+            u = x1;
+            v = x3;
+            x1 = addmod_u64(u, v, q);
+            x3 = mulmod_shoup(submod_u64(u, v, q), W, Wsh, q);
+        }
+        xr[k] = (int64_t)mulmod_shoup(x0, nv, nvs, q);
+        xr[k + quarter] = (int64_t)mulmod_shoup(x1, nv, nvs, q);
+        xr[k + 2 * quarter] = (int64_t)mulmod_shoup(x2, nv, nvs, q);
+        xr[k + 3 * quarter] = (int64_t)mulmod_shoup(x3, nv, nvs, q);
+    }
+}
+
 // a [.., L, n] x b (numel divides a's, same limb layout) with per-limb
 // primes; ratios [L][2] = floor(2^128/q) words.
 __global__ void modmul_limbs_kernel(const int64_t* __restrict__ a,
@@ -402,10 +508,14 @@ __global__ void modreduce_kernel(int64_t* __restrict__ x,
                                  int64_t limb_stride, int L) {
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
          i += (int64_t)gridDim.x * blockDim.x) {
-        uint64_t q = (uint64_t)qs[(i / limb_stride) % L];
+        const uint64_t q = (uint64_t)qs[(i / limb_stride) % L];
         uint64_t v = (uint64_t)x[i];
-        // v < 8q < 2^63: at most 3 conditional halving steps then compare
-        v %= q;
+        // v < 8q < 2^63 (lazy sum of <= 8 residues): three conditional
+        // subtracts replace the software u64 division
+        const uint64_t q4 = q << 2, q2 = q << 1;
+        if (v >= q4) v -= q4;
+        if (v >= q2) v -= q2;
+        if (v >= q) v -= q;
         x[i] = (int64_t)v;
     }
 }
@@ -550,12 +660,20 @@ void ntt_limbs(torch::Tensor x, torch::Tensor w, torch::Tensor wsh,
     const int64_t rows = x.numel() / n;
     const int nblk = n < kNblkMax ? n : kNblkMax;
     auto stream = at::cuda::getCurrentCUDAStream();
-    for (int m = 1; m < n / nblk; m <<= 1) {
-        int blocks = (int)std::min<int64_t>((n / 2 + kThreads - 1) / kThreads, 1024);
-        hipLaunchKernelGGL(ntt_global_stage_ml_kernel, rows_grid(blocks, rows),
+    if (n / nblk == 4) {
+        int blocks = (int)std::min<int64_t>((n / 4 + kThreads - 1) / kThreads, 1024);
+        hipLaunchKernelGGL(ntt_global_radix4_ml_kernel, rows_grid(blocks, rows),
                            dim3(kThreads), 0, stream, x.data_ptr<int64_t>(),
                            w.data_ptr<int64_t>(), wsh.data_ptr<int64_t>(),
-                           qs.data_ptr<int64_t>(), (int)L, n, m);
+                           qs.data_ptr<int64_t>(), (int)L, n);
+    } else {
+        for (int m = 1; m < n / nblk; m <<= 1) {
+            int blocks = (int)std::min<int64_t>((n / 2 + kThreads - 1) / kThreads, 1024);
+            hipLaunchKernelGGL(ntt_global_stage_ml_kernel, rows_grid(blocks, rows),
+                               dim3(kThreads), 0, stream, x.data_ptr<int64_t>(),
+                               w.data_ptr<int64_t>(), wsh.data_ptr<int64_t>(),
+                               qs.data_ptr<int64_t>(), (int)L, n, m);
+        }
     }
     hipLaunchKernelGGL(ntt_lds_ml_kernel, rows_grid(n / nblk, rows),
                        dim3(kThreads), nblk * sizeof(int64_t), stream,
@@ -580,14 +698,23 @@ void intt_limbs(torch::Tensor x, torch::Tensor winv, torch::Tensor winvsh,
                        winvsh.data_ptr<int64_t>(), qs.data_ptr<int64_t>(),
                        ninv.data_ptr<int64_t>(), ninvsh.data_ptr<int64_t>(),
                        (int)L, n, nblk, has_global ? 0 : 1);
-    for (int m = n / nblk; m >= 2; m >>= 1) {
-        int blocks = (int)std::min<int64_t>((n / 2 + kThreads - 1) / kThreads, 1024);
-        hipLaunchKernelGGL(intt_global_stage_ml_kernel, rows_grid(blocks, rows),
+    if (n / nblk == 4) {
+        int blocks = (int)std::min<int64_t>((n / 4 + kThreads - 1) / kThreads, 1024);
+        hipLaunchKernelGGL(intt_global_radix4_ml_kernel, rows_grid(blocks, rows),
                            dim3(kThreads), 0, stream, x.data_ptr<int64_t>(),
                            winv.data_ptr<int64_t>(), winvsh.data_ptr<int64_t>(),
                            qs.data_ptr<int64_t>(), ninv.data_ptr<int64_t>(),
-                           ninvsh.data_ptr<int64_t>(), (int)L, n, m,
-                           m == 2 ? 1 : 0);
+                           ninvsh.data_ptr<int64_t>(), (int)L, n);
+    } else {
+        for (int m = n / nblk; m >= 2; m >>= 1) {
+            int blocks = (int)std::min<int64_t>((n / 2 + kThreads - 1) / kThreads, 1024);
+            hipLaunchKernelGGL(intt_global_stage_ml_kernel, rows_grid(blocks, rows),
+                               dim3(kThreads), 0, stream, x.data_ptr<int64_t>(),
+                               winv.data_ptr<int64_t>(), winvsh.data_ptr<int64_t>(),
+                               qs.data_ptr<int64_t>(), ninv.data_ptr<int64_t>(),
+                               ninvsh.data_ptr<int64_t>(), (int)L, n, m,
+                               m == 2 ? 1 : 0);
+        }
     }
 }
 

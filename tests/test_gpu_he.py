@@ -130,13 +130,15 @@ def test_gpu_vs_cpu_identical_ntt_path():
     assert np.abs(out_c - og).max() < 1e-6
 
 
-def test_fused_multilimb_ops_match_per_limb():
+@pytest.mark.parametrize("n", [1024, 32768])
+def test_fused_multilimb_ops_match_per_limb(n):
     """ntt_limbs / modmul_limbs / modmul_scalar_limbs must equal the
-    per-limb kernels on identical inputs."""
-    cfg = HEConfig(m=1024, scale_bits=30, q_bits=(50, 40, 30), seed=9)
+    per-limb kernels on identical inputs (n=2^15 exercises the fused
+    radix-4 cross-block stage)."""
+    cfg = HEConfig(m=n, scale_bits=30, q_bits=(50, 40, 30), seed=9)
     ctx = CKKSContext(cfg, device="cuda")
     be = ctx.backend
-    L, n = 3, 1024
+    L = 3
     rng = np.random.default_rng(4)
     x = torch.from_numpy(np.stack(
         [rng.integers(0, ctx.primes[i], size=(5, n), dtype=np.int64)
