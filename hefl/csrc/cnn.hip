@@ -1592,9 +1592,9 @@ torch::Tensor conv2d_wgrad(torch::Tensor dy, torch::Tensor x, int64_t stride,
     const int KK = s.N * s.OH * s.OW;
     const bool big = NN > 16;
     const int bn = big ? 64 : 16;
-    // split-K to fill the chip: target >= 512 blocks
+    // split-K to fill the chip: target >= 512 blocks (2 per CU)
     int tiles = ceildiv(s.Kout, 64) * ceildiv(NN, bn);
-    int k_chunks = std::max(1, std::min(ceildiv(KK, 32 * 8),
+    int k_chunks = std::max(1, std::min(ceildiv(KK, 32 * 2),
                                         512 / std::max(tiles, 1)));
     auto dw = k_chunks > 1
                   ? torch::zeros({s.Kout, R, S, s.C},
