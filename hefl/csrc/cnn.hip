@@ -340,8 +340,9 @@ conv_fwd_glds_kernel(const unsigned short* __restrict__ x,
                      const unsigned short* __restrict__ w,
                      const float* __restrict__ bias,
                      unsigned short* __restrict__ y,
+                     float* __restrict__ y32,  // split-K accumulator (or null)
                      const unsigned short* __restrict__ zbuf, ConvShape s,
-                     int relu) {
+                     int relu, int k_chunks) {
     // ONE shared object (two makes hipcc emit a vmcnt(0) drain before each
     // k-step's first ds_read, defeating the glds pipeline — guide trap 4a)
     __shared__ unsigned short smem[2 * (BM + BN) * 32];
@@ -415,11 +416,15 @@ conv_fwd_glds_kernel(const unsigned short* __restrict__ x,
 #pragma unroll
         for (int j = 0; j < FN; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-    stage(0, 0);
+    const int ksteps = (KK + 31) / 32;
+    const int kc_len = (ksteps + k_chunks - 1) / k_chunks;
+    const int kbeg = blockIdx.z * kc_len * 32;
+    const int kend = min(kbeg + kc_len * 32, KK);
+    stage(0, kbeg);
     __syncthreads();  // drains the DMA (vmcnt 0) + barrier
     int buf = 0;
-    for (int k0 = 0; k0 < KK; k0 += 32) {
-        if (k0 + 32 < KK) stage(buf ^ 1, k0 + 32);  // issue BEFORE compute
+    for (int k0 = kbeg; k0 < kend; k0 += 32) {
+        if (k0 + 32 < kend) stage(buf ^ 1, k0 + 32);  // issue BEFORE compute
         tile_mfma<BM, BN, WM, WN, FM, FN>(
             *reinterpret_cast<const unsigned short(*)[BM][32]>(As(buf)),
             *reinterpret_cast<const unsigned short(*)[BN][32]>(Bs(buf)), acc,
@@ -439,9 +444,13 @@ conv_fwd_glds_kernel(const unsigned short* __restrict__ x,
             for (int r = 0; r < 4; ++r) {
                 int row = m0 + wm * FM * 16 + i * 16 + (lane >> 4) * 4 + r;
                 if (row >= M) continue;
-                float v = acc[i][j][r] + bv;
-                if (relu) v = v > 0.f ? v : 0.f;
-                y[(int64_t)row * s.Kout + col] = f2bf(v);
+                if (k_chunks > 1) {
+                    atomicAdd(y32 + (int64_t)row * s.Kout + col, acc[i][j][r]);
+                } else {
+                    float v = acc[i][j][r] + bv;
+                    if (relu) v = v > 0.f ? v : 0.f;
+                    y[(int64_t)row * s.Kout + col] = f2bf(v);
+                }
             }
     }
 }
@@ -619,7 +628,9 @@ __global__ void __launch_bounds__(TPB)
 conv_dgrad_glds_kernel(const unsigned short* __restrict__ dy,
                        const unsigned short* __restrict__ w,
                        unsigned short* __restrict__ dx,
-                       const unsigned short* __restrict__ zbuf, ConvShape s) {
+                       float* __restrict__ dx32,
+                       const unsigned short* __restrict__ zbuf, ConvShape s,
+                       int k_chunks) {
     __shared__ unsigned short smem[2 * (BM + BN) * 32];
     auto As = [&](int buf) -> unsigned short (*)[32] {
         return reinterpret_cast<unsigned short(*)[32]>(smem + buf * (BM + BN) * 32);
@@ -699,11 +710,15 @@ conv_dgrad_glds_kernel(const unsigned short* __restrict__ dy,
 #pragma unroll
         for (int j = 0; j < FN; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-    stage(0, 0);
+    const int ksteps = (KK + 31) / 32;
+    const int kc_len = (ksteps + k_chunks - 1) / k_chunks;
+    const int kbeg = blockIdx.z * kc_len * 32;
+    const int kend = min(kbeg + kc_len * 32, KK);
+    stage(0, kbeg);
     __syncthreads();
     int buf = 0;
-    for (int k0 = 0; k0 < KK; k0 += 32) {
-        if (k0 + 32 < KK) stage(buf ^ 1, k0 + 32);
+    for (int k0 = kbeg; k0 < kend; k0 += 32) {
+        if (k0 + 32 < kend) stage(buf ^ 1, k0 + 32);
         {
             const int wm = wave / WN, wn = wave % WN;
             const int half = lane >> 4, sub = lane & 15;
@@ -741,9 +756,21 @@ conv_dgrad_glds_kernel(const unsigned short* __restrict__ dy,
             for (int r = 0; r < 4; ++r) {
                 int row = m0 + wm * FM * 16 + i * 16 + (lane >> 4) * 4 + r;
                 if (row >= M) continue;
-                dx[(int64_t)row * s.C + col] = f2bf(acc[i][j][r]);
+                if (k_chunks > 1)
+                    atomicAdd(dx32 + (int64_t)row * s.C + col, acc[i][j][r]);
+                else
+                    dx[(int64_t)row * s.C + col] = f2bf(acc[i][j][r]);
             }
     }
+}
+
+// f32 accumulator -> bf16 (no bias/relu): split-K epilogue for dgrad
+__global__ void cast_f32_bf16_kernel(const float* __restrict__ src,
+                                     unsigned short* __restrict__ dst,
+                                     int64_t total) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
+         i += (int64_t)gridDim.x * blockDim.x)
+        dst[i] = f2bf(src[i]);
 }
 
 // ---- wgrad: dw[ko, rsc] = sum_pix dy[pix, ko] * xcol[pix, rsc].
@@ -1495,14 +1522,35 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
     auto stream = at::cuda::getCurrentCUDAStream();
     const int KKf = s.R * s.S * s.C;
     if (s.Kout > 16 && s.C % 8 == 0 && KKf % 8 == 0) {
-        // glds double-buffered pipeline (DMA flight hides under MFMA)
+        // glds double-buffered pipeline (DMA flight hides under MFMA);
+        // small-M/N late layers split the K loop over grid.z to fill the chip
         static torch::Tensor zbuf;
         if (!zbuf.defined() || zbuf.device() != x.device())
             zbuf = torch::zeros({8}, x.options());
-        dim3 grid(ceildiv(M, 64), ceildiv(s.Kout, 64));
-        hipLaunchKernelGGL((conv_fwd_glds_kernel<64, 64, 2, 2, 2, 2>), grid,
-                           dim3(TPB), 0, stream, bf_ptr(x), bf_ptr(w), bias,
-                           bf_ptr_mut(y), bf_ptr(zbuf), s, relu ? 1 : 0);
+        int tiles = ceildiv(M, 64) * ceildiv(s.Kout, 64);
+        int ksteps = ceildiv(KKf, 32);
+        int k_chunks = 1;
+        if (tiles < 256)
+            k_chunks = std::max(1, std::min(ksteps / 4, 512 / tiles));
+        dim3 grid(ceildiv(M, 64), ceildiv(s.Kout, 64), k_chunks);
+        if (k_chunks > 1) {
+            auto y32 = torch::zeros({M, s.Kout},
+                                    x.options().dtype(torch::kFloat32));
+            hipLaunchKernelGGL((conv_fwd_glds_kernel<64, 64, 2, 2, 2, 2>), grid,
+                               dim3(TPB), 0, stream, bf_ptr(x), bf_ptr(w), bias,
+                               bf_ptr_mut(y), y32.data_ptr<float>(),
+                               bf_ptr(zbuf), s, relu ? 1 : 0, k_chunks);
+            int64_t total = (int64_t)M * s.Kout;
+            hipLaunchKernelGGL(linear_epilogue_kernel,
+                               dim3((int)std::min<int64_t>(ceildiv(total, 256), 2048)),
+                               dim3(256), 0, stream, y32.data_ptr<float>(), bias,
+                               bf_ptr_mut(y), total, s.Kout, relu ? 1 : 0);
+        } else {
+            hipLaunchKernelGGL((conv_fwd_glds_kernel<64, 64, 2, 2, 2, 2>), grid,
+                               dim3(TPB), 0, stream, bf_ptr(x), bf_ptr(w), bias,
+                               bf_ptr_mut(y), (float*)nullptr, bf_ptr(zbuf), s,
+                               relu ? 1 : 0, 1);
+        }
     } else if (s.Kout > 16) {
         dim3 grid(ceildiv(M, 64), ceildiv(s.Kout, 64));
         hipLaunchKernelGGL((conv_fwd_kernel<64, 64, 2, 2, 2, 2>), grid,
@@ -1535,15 +1583,36 @@ torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor w, int64_t stride,
         static torch::Tensor zbuf;
         if (!zbuf.defined() || zbuf.device() != dy.device())
             zbuf = torch::zeros({8}, dy.options());
-        dim3 grid(ceildiv(M, 64), ceildiv(C, 64));
+        const int KKd = s.Kout * s.R * s.S;
+        int tiles = ceildiv(M, 64) * ceildiv(C, 64);
+        int ksteps = ceildiv(KKd, 32);
+        int k_chunks = 1;
+        if (tiles < 256)
+            k_chunks = std::max(1, std::min(ksteps / 4, 512 / tiles));
+        dim3 grid(ceildiv(M, 64), ceildiv(C, 64), k_chunks);
+        torch::Tensor dx32;
+        float* dx32p = nullptr;
+        if (k_chunks > 1) {
+            dx32 = torch::zeros({M, C}, dy.options().dtype(torch::kFloat32));
+            dx32p = dx32.data_ptr<float>();
+        }
         if (stride == 1)
             hipLaunchKernelGGL((conv_dgrad_glds_kernel<64, 64, 2, 2, 2, 2, true>),
                                grid, dim3(TPB), 0, stream, bf_ptr(dy),
-                               bf_ptr(w), bf_ptr_mut(dx), bf_ptr(zbuf), s);
+                               bf_ptr(w), bf_ptr_mut(dx), dx32p, bf_ptr(zbuf),
+                               s, k_chunks);
         else
             hipLaunchKernelGGL((conv_dgrad_glds_kernel<64, 64, 2, 2, 2, 2, false>),
                                grid, dim3(TPB), 0, stream, bf_ptr(dy),
-                               bf_ptr(w), bf_ptr_mut(dx), bf_ptr(zbuf), s);
+                               bf_ptr(w), bf_ptr_mut(dx), dx32p, bf_ptr(zbuf),
+                               s, k_chunks);
+        if (k_chunks > 1) {
+            int64_t total = (int64_t)M * C;
+            hipLaunchKernelGGL(cast_f32_bf16_kernel,
+                               dim3((int)std::min<int64_t>(ceildiv(total, 256), 2048)),
+                               dim3(256), 0, stream, dx32p, bf_ptr_mut(dx),
+                               total);
+        }
     } else if (false) {  // 128-tile: measured neutral-to-worse (barrier-bound)
         dim3 grid(ceildiv(M, 128), ceildiv(C, 64));
         if (stride == 1)
