@@ -921,6 +921,134 @@ conv_wgrad_kernel(const unsigned short* __restrict__ dy,
     }
 }
 
+// ---- glds-pipelined wgrad: both tiles (Dys pixel-major [32][64] and
+// Xs pixel-major [32][BN]) stage as lane-linear 16-B DMAs; 2-buffer
+// pipeline like fwd/dgrad. Requires Kout % 8 == 0 and C % 8 == 0. ----
+template <int BN, int WM, int WN, int FM, int FN>
+__global__ void __launch_bounds__(TPB)
+conv_wgrad_glds_kernel(const unsigned short* __restrict__ dy,
+                       const unsigned short* __restrict__ x,
+                       float* __restrict__ dw,
+                       const unsigned short* __restrict__ zbuf, ConvShape s,
+                       int k_chunks) {
+    __shared__ unsigned short smem[2 * (64 + BN) * 32];
+    auto Dys = [&](int buf) -> unsigned short (*)[64] {
+        return reinterpret_cast<unsigned short(*)[64]>(smem + buf * (64 + BN) * 32);
+    };
+    auto Xs = [&](int buf) -> unsigned short (*)[BN] {
+        return reinterpret_cast<unsigned short(*)[BN]>(smem + buf * (64 + BN) * 32
+                                                        + 64 * 32);
+    };
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+    const int m0 = blockIdx.x * 64;   // over Kout
+    const int n0 = blockIdx.y * BN;   // over RSC
+    const int M = s.Kout;
+    const int NN = s.R * s.S * s.C;
+    const int KK = s.N * s.OH * s.OW;
+    const int chunk = (KK + k_chunks - 1) / k_chunks;
+    const int kbeg = blockIdx.z * chunk;
+    const int kend = min(kbeg + chunk, KK);
+
+    auto stage = [&](int buf, int k0) {
+        {   // Dys[pix][ko]: chunk tid -> byte tid*16 of its wave slice
+            const int pix = tid >> 3;
+            const int koc = (tid & 7) * 8;
+            const int kpix = k0 + pix;
+            const int ko = m0 + koc;
+            const unsigned short* src = zbuf;
+            if (kpix < kend && ko + 8 <= M)
+                src = dy + (int64_t)kpix * s.Kout + ko;
+            glds16(src, (char*)&Dys(buf)[0][0] + wave * 1024);
+        }
+#pragma unroll
+        for (int t = 0; t < BN * 4 / TPB; ++t) {  // Xs chunks
+            const int i = tid + t * TPB;
+            const int pix = i / (BN / 8);
+            const int cc = (i % (BN / 8)) * 8;
+            const int kpix = k0 + pix;
+            const int nn = n0 + cc;
+            const unsigned short* src = zbuf;
+            if (kpix < kend && nn < NN) {
+                int n_ = kpix / (s.OH * s.OW);
+                int rem = kpix % (s.OH * s.OW);
+                int oh = rem / s.OW, ow = rem % s.OW;
+                int c = nn % s.C;
+                int rs = nn / s.C;
+                int r = rs / s.S, ss = rs % s.S;
+                if (c + 8 <= s.C) {
+                    int ih = oh * s.stride + r - s.pad;
+                    int iw = ow * s.stride + ss - s.pad;
+                    if (ih >= 0 && ih < s.H && iw >= 0 && iw < s.W)
+                        src = x + ((((int64_t)n_ * s.H + ih) * s.W + iw) * s.C
+                                   + c);
+                }
+            }
+            glds16(src, (char*)&Xs(buf)[0][0] + (wave + t * 4) * 1024);
+        }
+    };
+
+    f32x4 acc[FM][FN];
+#pragma unroll
+    for (int i = 0; i < FM; ++i)
+#pragma unroll
+        for (int j = 0; j < FN; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+    stage(0, kbeg);
+    __syncthreads();
+    int buf = 0;
+    for (int k0 = kbeg; k0 < kend; k0 += 32) {
+        if (k0 + 32 < kend) stage(buf ^ 1, k0 + 32);
+        {
+            const int wm = wave / WN, wn = wave % WN;
+            const int half = lane >> 4, sub = lane & 15;
+            bf16x8 a[FM], b[FN];
+#pragma unroll
+            for (int i = 0; i < FM; ++i) {
+                const int mrow = wm * FM * 16 + i * 16 + sub;
+#pragma unroll
+                for (int j = 0; j < 8; ++j)
+                    a[i][j] = *reinterpret_cast<const bf16_t*>(
+                        &Dys(buf)[half * 8 + j][mrow]);
+            }
+#pragma unroll
+            for (int j = 0; j < FN; ++j) {
+                const int col = wn * FN * 16 + j * 16 + sub;
+#pragma unroll
+                for (int t = 0; t < 8; ++t)
+                    b[j][t] = *reinterpret_cast<const bf16_t*>(
+                        &Xs(buf)[half * 8 + t][col]);
+            }
+#pragma unroll
+            for (int i = 0; i < FM; ++i)
+#pragma unroll
+                for (int j = 0; j < FN; ++j)
+                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        a[i], b[j], acc[i][j], 0, 0, 0);
+        }
+        __syncthreads();
+        buf ^= 1;
+    }
+    const int wm = wave / WN, wn = wave % WN;
+#pragma unroll
+    for (int j = 0; j < FN; ++j) {
+        const int col = n0 + wn * FN * 16 + j * 16 + (lane & 15);
+        if (col >= NN) continue;
+#pragma unroll
+        for (int i = 0; i < FM; ++i)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int ko = m0 + wm * FM * 16 + i * 16 + (lane >> 4) * 4 + r;
+                if (ko >= M) continue;
+                if (k_chunks > 1)
+                    atomicAdd(dw + (int64_t)ko * NN + col, acc[i][j][r]);
+                else
+                    dw[(int64_t)ko * NN + col] = acc[i][j][r];
+            }
+    }
+}
+
 // ---------------------------------------------------------------------------
 // MaxPool 2x2 stride 2 (NHWC bf16), argmax corner saved for backward.
 // ---------------------------------------------------------------------------
@@ -1660,6 +1788,7 @@ torch::Tensor conv2d_wgrad(torch::Tensor dy, torch::Tensor x, int64_t stride,
     const int NN = s.R * s.S * s.C;
     const int KK = s.N * s.OH * s.OW;
     const bool big = NN > 16;
+    const bool glds_ok = big && (s.C % 8 == 0) && (s.Kout % 8 == 0);
     const int bn = big ? 64 : 16;
     // split-K to fill the chip: target >= 512 blocks (2 per CU)
     int tiles = ceildiv(s.Kout, 64) * ceildiv(NN, bn);
@@ -1672,7 +1801,14 @@ torch::Tensor conv2d_wgrad(torch::Tensor dy, torch::Tensor x, int64_t stride,
                                  x.options().dtype(torch::kFloat32));
     dim3 grid(ceildiv(s.Kout, 64), ceildiv(NN, bn), k_chunks);
     auto stream = at::cuda::getCurrentCUDAStream();
-    if (big) {
+    if (glds_ok) {
+        static torch::Tensor zbuf;
+        if (!zbuf.defined() || zbuf.device() != dy.device())
+            zbuf = torch::zeros({8}, dy.options());
+        hipLaunchKernelGGL((conv_wgrad_glds_kernel<64, 2, 2, 2, 2>), grid,
+                           dim3(TPB), 0, stream, bf_ptr(dy), bf_ptr(x),
+                           dw.data_ptr<float>(), bf_ptr(zbuf), s, k_chunks);
+    } else if (big) {
         hipLaunchKernelGGL((conv_wgrad_kernel<64, 2, 2, 2, 2>), grid, dim3(TPB),
                            0, stream, bf_ptr(dy), bf_ptr(x),
                            dw.data_ptr<float>(), s, k_chunks);
