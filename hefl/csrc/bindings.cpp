@@ -69,6 +69,7 @@ torch::Tensor avgpool_global_fwd(torch::Tensor x);
 torch::Tensor avgpool_global_bwd(torch::Tensor dy, int64_t H, int64_t W);
 torch::Tensor add_relu(torch::Tensor a, torch::Tensor b);
 torch::Tensor bias_grad(torch::Tensor dy);
+std::vector<torch::Tensor> relu_bias_bwd(torch::Tensor dy, torch::Tensor y);
 void fused_adam_mt(torch::Tensor meta, torch::Tensor ptrs, torch::Tensor sizes,
                    int64_t nchunks, torch::Tensor sched, double b1, double b2,
                    double eps);
@@ -113,6 +114,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("avgpool_global_bwd", &avgpool_global_bwd);
     m.def("add_relu", &add_relu);
     m.def("bias_grad", &bias_grad);
+    m.def("relu_bias_bwd", &relu_bias_bwd);
     m.def("fused_adam_mt", &fused_adam_mt);
     m.def("zero_grads_mt", &zero_grads_mt);
 }

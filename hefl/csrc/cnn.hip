@@ -1080,6 +1080,61 @@ __global__ void maxpool_fwd_kernel(const unsigned short* __restrict__ x,
     }
 }
 
+// Gather form: one thread per INPUT cell, no zero-fill pass needed
+// (2x2 stride-2 windows never overlap; odd tail rows/cols get 0).
+__global__ void maxpool2x2_bwd_gather_kernel(const unsigned short* __restrict__ dy,
+                                             const uint8_t* __restrict__ idx,
+                                             unsigned short* __restrict__ dx,
+                                             int N, int H, int W, int C,
+                                             int OH, int OW) {
+    int64_t total = (int64_t)N * H * W * C;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        int c = i % C;
+        int64_t t = i / C;
+        int iw = t % W;
+        t /= W;
+        int ih = t % H;
+        int n = t / H;
+        int oh = ih >> 1, ow = iw >> 1;
+        unsigned short v = 0;
+        if (oh < OH && ow < OW) {
+            int64_t oidx = (((int64_t)n * OH + oh) * OW + ow) * C + c;
+            int d = idx[oidx];
+            if ((d >> 1) == (ih & 1) && (d & 1) == (iw & 1)) v = dy[oidx];
+        }
+        dx[i] = v;
+    }
+}
+
+// Fused ReLU-mask + bias grad: dy' = dy * (y > 0), db[k] = sum dy'[., k]
+// — one pass over dy instead of {relu_bwd, zero-fill, bias_grad}.
+__global__ void relu_bias_bwd_kernel(const unsigned short* __restrict__ dy,
+                                     const unsigned short* __restrict__ y,
+                                     unsigned short* __restrict__ dym,
+                                     float* __restrict__ db, int64_t M, int K,
+                                     int rows_per_block) {
+    __shared__ float red[256];
+    const int k = blockIdx.x;
+    const int64_t r0 = (int64_t)blockIdx.y * rows_per_block;
+    const int64_t r1 = min(r0 + rows_per_block, M);
+    float acc = 0.f;
+    for (int64_t r = r0 + threadIdx.x; r < r1; r += blockDim.x) {
+        int64_t i = r * K + k;
+        unsigned short yv = y[i];
+        unsigned short g = ((yv & 0x7fffu) != 0 && !(yv & 0x8000u)) ? dy[i] : 0;
+        dym[i] = g;
+        acc += bf2f(g);
+    }
+    red[threadIdx.x] = acc;
+    __syncthreads();
+    for (int off = 128; off > 0; off >>= 1) {
+        if (threadIdx.x < off) red[threadIdx.x] += red[threadIdx.x + off];
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) atomicAdd(db + k, red[0]);
+}
+
 __global__ void maxpool_bwd_kernel(const unsigned short* __restrict__ dy,
                                    const uint8_t* __restrict__ idx,
                                    unsigned short* __restrict__ dx, int N,
@@ -1896,14 +1951,29 @@ torch::Tensor maxpool2x2_bwd(torch::Tensor dy, torch::Tensor idx, int64_t H,
     CHECK_GPU(dy);
     const int N = (int)dy.size(0), OH = (int)dy.size(1), OW = (int)dy.size(2),
               C = (int)dy.size(3);
-    auto dx = torch::zeros({N, H, W, C}, dy.options());
-    int64_t total = (int64_t)N * OH * OW * C;
+    auto dx = torch::empty({N, H, W, C}, dy.options());
+    int64_t total = (int64_t)N * H * W * C;
     int blocks = std::min<int64_t>(ceildiv(total, 256), 2048);
-    hipLaunchKernelGGL(maxpool_bwd_kernel, dim3(blocks), dim3(256), 0,
+    hipLaunchKernelGGL(maxpool2x2_bwd_gather_kernel, dim3(blocks), dim3(256), 0,
                        at::cuda::getCurrentCUDAStream(), bf_ptr(dy),
                        idx.data_ptr<uint8_t>(), bf_ptr_mut(dx), N, (int)H,
                        (int)W, C, OH, OW);
     return dx;
+}
+
+std::vector<torch::Tensor> relu_bias_bwd(torch::Tensor dy, torch::Tensor y) {
+    CHECK_GPU(dy);
+    auto dyc = dy.contiguous();
+    const int K = (int)dyc.size(-1);
+    const int64_t M = dyc.numel() / K;
+    auto dym = torch::empty_like(dyc);
+    auto db = torch::zeros({K}, dyc.options().dtype(torch::kFloat32));
+    int rpb = (int)std::max<int64_t>(2048, (M + 63) / 64);
+    dim3 grid(K, (unsigned)((M + rpb - 1) / rpb));
+    hipLaunchKernelGGL(relu_bias_bwd_kernel, grid, dim3(256), 0,
+                       at::cuda::getCurrentCUDAStream(), bf_ptr(dyc), bf_ptr(y),
+                       bf_ptr_mut(dym), db.data_ptr<float>(), M, K, rpb);
+    return {dym, db};
 }
 
 std::vector<torch::Tensor> softmax_xent_fwd(torch::Tensor logits,

@@ -63,13 +63,17 @@ class _Conv2dFn(torch.autograd.Function):
         stride, relu, pad = ctx.stride, ctx.relu, ctx.pad
         if dy.is_cuda:
             dy = dy.contiguous()
-            if relu:
+            db = None
+            if relu and ctx.has_bias:
+                dy, db = _C().relu_bias_bwd(dy, y)  # one fused pass
+            elif relu:
                 dy = _C().relu_bwd(dy, y)
+            elif ctx.has_bias:
+                db = _C().bias_grad(dy)
             dx = _C().conv2d_dgrad(dy, w, stride, x.shape[1], x.shape[2], pad) \
                 if ctx.needs_input_grad[0] else None
             dw = _C().conv2d_wgrad(dy, x.contiguous(), stride, w.shape[1],
                                    w.shape[2], pad)
-            db = _C().bias_grad(dy) if ctx.has_bias else None
         else:
             dy = dy.float()
             if relu:
@@ -156,11 +160,15 @@ class _LinearFn(torch.autograd.Function):
         x, w, y = ctx.saved_tensors
         if dy.is_cuda:
             dy = dy.contiguous()
-            if ctx.relu:
+            db = None
+            if ctx.relu and ctx.has_bias:
+                dy, db = _C().relu_bias_bwd(dy, y)
+            elif ctx.relu:
                 dy = _C().relu_bwd(dy, y)
+            elif ctx.has_bias:
+                db = _C().bias_grad(dy.view(-1, dy.shape[-1]))
             dx = _C().linear_dgrad(dy, w) if ctx.needs_input_grad[0] else None
             dw = _C().linear_wgrad(dy, x).to(torch.float32)
-            db = _C().bias_grad(dy.view(-1, dy.shape[-1])) if ctx.has_bias else None
         else:
             dy = dy.float()
             if ctx.relu:
