@@ -1049,6 +1049,127 @@ conv_wgrad_glds_kernel(const unsigned short* __restrict__ dy,
     }
 }
 
+// ---- small-shape wgrad (first convs: Kout <= 64, R*S*C <= 16, e.g. the
+// 28x28x1 model's conv1): the generic 64xBN tile wastes 3/4 of its waves on
+// out-of-range rows. Here the 4 waves SPLIT over pixels (each owns 32 of a
+// 128-pixel iteration) and reduce their partial accumulators through LDS at
+// the end; grid.z chunks pixels further. ----
+__global__ void __launch_bounds__(TPB)
+conv_wgrad_small_kernel(const unsigned short* __restrict__ dy,
+                        const unsigned short* __restrict__ x,
+                        float* __restrict__ dw, ConvShape s, int k_chunks) {
+    __shared__ unsigned short Dys[128][64];
+    __shared__ unsigned short Xp[128][16];
+    __shared__ float red[4][16 * 16];
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+    const int M = s.Kout;            // <= 64
+    const int NN = s.R * s.S * s.C;  // <= 16
+    const int KK = s.N * s.OH * s.OW;
+    const int FM = (M + 15) / 16;
+    const int chunk = (KK + k_chunks - 1) / k_chunks;
+    const int kbeg = blockIdx.z * chunk;
+    const int kend = min(kbeg + chunk, KK);
+    const bool fast_dy = (s.Kout % 8 == 0);
+
+    f32x4 acc[4];  // up to FM=4 fragments
+#pragma unroll
+    for (int i = 0; i < 4; ++i) acc[i] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+    for (int p0 = kbeg; p0 < kend; p0 += 128) {
+        // stage dy [128][Kout]
+        for (int i = tid; i < 128 * (M / 8 > 0 ? M / 8 : 1); i += TPB) {
+            if (fast_dy) {
+                const int pix = i / (M / 8);
+                const int koc = (i % (M / 8)) * 8;
+                unsigned short* dst = &Dys[pix][koc];
+                if (p0 + pix < kend)
+                    copy16(dst, dy + (int64_t)(p0 + pix) * s.Kout + koc);
+                else
+                    zero16(dst);
+            } else {
+                const int pix = i;  // M < 8: scalar per pixel
+                if (pix < 128) {
+                    for (int ko = 0; ko < M; ++ko)
+                        Dys[pix][ko] = (p0 + pix < kend)
+                            ? dy[(int64_t)(p0 + pix) * s.Kout + ko] : 0;
+                }
+            }
+        }
+        // stage x patches [128][NN]
+        for (int i = tid; i < 128 * NN; i += TPB) {
+            const int pix = i / NN;
+            const int nn = i % NN;
+            unsigned short v = 0;
+            const int kpix = p0 + pix;
+            if (kpix < kend) {
+                int n_ = kpix / (s.OH * s.OW);
+                int rem = kpix % (s.OH * s.OW);
+                int oh = rem / s.OW, ow = rem % s.OW;
+                int c = nn % s.C;
+                int rs = nn / s.C;
+                int r = rs / s.S, ss = rs % s.S;
+                int ih = oh * s.stride + r - s.pad;
+                int iw = ow * s.stride + ss - s.pad;
+                if (ih >= 0 && ih < s.H && iw >= 0 && iw < s.W)
+                    v = x[((((int64_t)n_ * s.H + ih) * s.W + iw) * s.C + c)];
+            }
+            Xp[pix][nn] = v;
+        }
+        __syncthreads();
+        // each wave: its own 32 pixels as the MFMA K dim
+        {
+            const int half = lane >> 4, sub = lane & 15;
+            const int pbase = wave * 32 + half * 8;
+            bf16x8 b{};
+#pragma unroll
+            for (int j = 0; j < 8; ++j)
+                if (sub < 16)
+                    b[j] = *reinterpret_cast<const bf16_t*>(&Xp[pbase + j][sub]);
+#pragma unroll
+            for (int fm = 0; fm < 4; ++fm) {
+                if (fm >= FM) break;
+                bf16x8 a;
+#pragma unroll
+                for (int j = 0; j < 8; ++j)
+                    a[j] = *reinterpret_cast<const bf16_t*>(
+                        &Dys[pbase + j][fm * 16 + sub]);
+                acc[fm] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[fm],
+                                                                  0, 0, 0);
+            }
+        }
+        __syncthreads();
+    }
+    // cross-wave reduction per fragment, then atomics into dw
+    for (int fm = 0; fm < FM; ++fm) {
+        const int sub = lane & 15;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            int row = (lane >> 4) * 4 + r;  // 0..15 within fragment
+            red[wave][row * 16 + sub] = acc[fm][r];
+        }
+        __syncthreads();
+        // wave 0 sums and writes
+        if (wave == 0) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int row = (lane >> 4) * 4 + r;
+                float v = red[0][row * 16 + sub] + red[1][row * 16 + sub] +
+                          red[2][row * 16 + sub] + red[3][row * 16 + sub];
+                int ko = fm * 16 + row;
+                if (ko < M && sub < NN) {
+                    if (k_chunks > 1)
+                        atomicAdd(dw + (int64_t)ko * NN + sub, v);
+                    else
+                        dw[(int64_t)ko * NN + sub] = v;
+                }
+            }
+        }
+        __syncthreads();
+    }
+}
+
 // ---------------------------------------------------------------------------
 // MaxPool 2x2 stride 2 (NHWC bf16), argmax corner saved for backward.
 // ---------------------------------------------------------------------------
@@ -1843,6 +1964,7 @@ torch::Tensor conv2d_wgrad(torch::Tensor dy, torch::Tensor x, int64_t stride,
     const int NN = s.R * s.S * s.C;
     const int KK = s.N * s.OH * s.OW;
     const bool big = NN > 16;
+    const bool small_fast = (!big) && s.Kout <= 64;
     const bool glds_ok = big && (s.C % 8 == 0) && (s.Kout % 8 == 0);
     const int bn = big ? 64 : 16;
     // split-K to fill the chip: target >= 512 blocks (2 per CU)
@@ -1854,8 +1976,18 @@ torch::Tensor conv2d_wgrad(torch::Tensor dy, torch::Tensor x, int64_t stride,
                                  x.options().dtype(torch::kFloat32))
                   : torch::empty({s.Kout, R, S, s.C},
                                  x.options().dtype(torch::kFloat32));
-    dim3 grid(ceildiv(s.Kout, 64), ceildiv(NN, bn), k_chunks);
     auto stream = at::cuda::getCurrentCUDAStream();
+    if (small_fast) {
+        int kc = std::max(1, std::min(ceildiv(KK, 256), 512));
+        auto dw2 = kc > 1 ? torch::zeros({s.Kout, R, S, s.C},
+                                         x.options().dtype(torch::kFloat32))
+                          : dw;
+        hipLaunchKernelGGL(conv_wgrad_small_kernel, dim3(1, 1, kc), dim3(TPB),
+                           0, stream, bf_ptr(dy), bf_ptr(x),
+                           dw2.data_ptr<float>(), s, kc);
+        return dw2;
+    }
+    dim3 grid(ceildiv(s.Kout, 64), ceildiv(NN, bn), k_chunks);
     if (glds_ok) {
         static torch::Tensor zbuf;
         if (!zbuf.defined() || zbuf.device() != dy.device())

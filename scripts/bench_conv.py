@@ -17,6 +17,7 @@ def bench(fn, iters=200):
     return e0.elapsed_time(e1) * 1000 / iters  # us
 
 shapes = [
+    ("cnn2-conv1-28x28-c1", 32, 28, 28, 1, 16, 3, 1, 0),
     ("resnet3x3-64c-32x32", 32, 32, 32, 64, 64, 3, 1, 1),
     ("resnet3x3-128c-16x16", 32, 16, 16, 128, 128, 3, 1, 1),
     ("resnet3x3-512c-4x4", 32, 4, 4, 512, 512, 3, 1, 1),
