@@ -36,3 +36,36 @@ def test_gpu_phase_timer():
         y = (x * 2).sum()
     s = t.summary()
     assert s["matmul-ish"]["device_s"] >= 0
+
+
+def test_gpu_export_import_checkpoint_roundtrip(tmp_path):
+    """Encrypted weight file export/import + round-state checkpoint with the
+    GPU CKKS context (device tensors through the serialization path)."""
+    from hefl.fl.checkpoint import load_round_state, save_round_state
+    from hefl.fl.export import (decrypt_into_model, encrypt_export_weights,
+                                import_encrypted_weights)
+    from hefl.he import Pyfhel
+    from hefl.models import CNN2
+    from hefl.ops.adam import FusedAdam
+
+    he = Pyfhel()
+    he.contextGen(m=8192, q_bits=(60, 40), scale_bits=40, seed=3,
+                  device="cuda")
+    he.keyGen()
+    m = CNN2((28, 28, 1), 10, seed=4).cuda()
+    path = encrypt_export_weights(he, m, client_id=0, directory=str(tmp_path))
+    he2, val = import_encrypted_weights(path, device="cuda")
+    he2._sk = he._sk
+    m2 = CNN2((28, 28, 1), 10, seed=99).cuda()
+    decrypt_into_model(he2, val, m2)
+    for a, b in zip(m.parameters(), m2.parameters()):
+        assert (a - b).abs().max().item() < 1e-3
+
+    opt = FusedAdam(m.parameters(), lr=1e-3)
+    save_round_state(str(tmp_path / "r.pt"), m, opt, round_idx=2, he=he)
+    m3 = CNN2((28, 28, 1), 10, seed=123).cuda()
+    opt3 = FusedAdam(m3.parameters(), lr=1e-3)
+    rnd, _ = load_round_state(str(tmp_path / "r.pt"), m3, opt3)
+    assert rnd == 2
+    for a, b in zip(m.parameters(), m3.parameters()):
+        assert torch.equal(a, b)
