@@ -334,7 +334,7 @@ __device__ __forceinline__ void glds16(const unsigned short* src, void* lds_base
                                      0, 0);
 }
 
-template <int BM, int BN, int WM, int WN, int FM, int FN>
+template <int BM, int BN, int WM, int WN, int FM, int FN, int BK = 32>
 __global__ void __launch_bounds__(TPB)
 conv_fwd_glds_kernel(const unsigned short* __restrict__ x,
                      const unsigned short* __restrict__ w,
@@ -345,16 +345,16 @@ conv_fwd_glds_kernel(const unsigned short* __restrict__ x,
                      int relu, int k_chunks) {
     // ONE shared object (two makes hipcc emit a vmcnt(0) drain before each
     // k-step's first ds_read, defeating the glds pipeline — guide trap 4a)
-    __shared__ unsigned short smem[2 * (BM + BN) * 32];
-    auto As = [&](int buf) -> unsigned short (*)[32] {
-        return reinterpret_cast<unsigned short(*)[32]>(smem + buf * (BM + BN) * 32);
+    __shared__ unsigned short smem[2 * (BM + BN) * BK];
+    auto As = [&](int buf) -> unsigned short (*)[BK] {
+        return reinterpret_cast<unsigned short(*)[BK]>(smem + buf * (BM + BN) * BK);
     };
-    auto Bs = [&](int buf) -> unsigned short (*)[32] {
-        return reinterpret_cast<unsigned short(*)[32]>(smem + buf * (BM + BN) * 32
-                                                       + BM * 32);
+    auto Bs = [&](int buf) -> unsigned short (*)[BK] {
+        return reinterpret_cast<unsigned short(*)[BK]>(smem + buf * (BM + BN) * BK
+                                                       + BM * BK);
     };
-    constexpr int RPT = BM * 4 / TPB;
-    constexpr int BPT = BN * 4 / TPB;  // B chunks per thread (>=1 when BN=64)
+    constexpr int RPT = BM * (BK / 8) / TPB;
+    constexpr int BPT = BN * (BK / 8) / TPB;  // B chunks per thread
     const int tid = threadIdx.x;
     const int wave = tid >> 6;
     const int lane = tid & 63;
@@ -363,12 +363,13 @@ conv_fwd_glds_kernel(const unsigned short* __restrict__ x,
     const int M = s.N * s.OH * s.OW;
     const int KK = s.R * s.S * s.C;
 
+    constexpr int CPR = BK / 8;  // chunks per row
     int a_n[RPT], a_oh[RPT], a_ow[RPT];
     bool a_ok[RPT];
-    const int akc = (tid & 3) * 8;
+    const int akc = (tid % CPR) * 8;
 #pragma unroll
     for (int t = 0; t < RPT; ++t) {
-        const int am = m0 + ((tid + t * TPB) >> 2);
+        const int am = m0 + ((tid + t * TPB) / CPR);
         a_ok[t] = am < M;
         a_n[t] = a_oh[t] = a_ow[t] = 0;
         if (a_ok[t]) {
@@ -378,7 +379,6 @@ conv_fwd_glds_kernel(const unsigned short* __restrict__ x,
             a_ow[t] = rem % s.OW;
         }
     }
-    const int bcol0 = tid >> 2;  // first B column this thread stages
 
     auto stage = [&](int buf, int k0) {
         // A: RPT chunks, each one 16-B DMA; lane-linear within the wave
@@ -401,8 +401,8 @@ conv_fwd_glds_kernel(const unsigned short* __restrict__ x,
 #pragma unroll
         for (int t = 0; t < BPT; ++t) {
             const int i = tid + t * TPB;
-            const int ko = n0 + (i >> 2);
-            const int k = k0 + (i & 3) * 8;
+            const int ko = n0 + i / CPR;
+            const int k = k0 + (i % CPR) * 8;
             const unsigned short* src = zbuf;
             if (ko < s.Kout && k < KK) src = w + (int64_t)ko * KK + k;
             char* base = (char*)&Bs(buf)[0][0] + (wave + t * 4) * 1024;
@@ -416,19 +416,37 @@ conv_fwd_glds_kernel(const unsigned short* __restrict__ x,
 #pragma unroll
         for (int j = 0; j < FN; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-    const int ksteps = (KK + 31) / 32;
+    const int ksteps = (KK + BK - 1) / BK;
     const int kc_len = (ksteps + k_chunks - 1) / k_chunks;
-    const int kbeg = blockIdx.z * kc_len * 32;
-    const int kend = min(kbeg + kc_len * 32, KK);
+    const int kbeg = blockIdx.z * kc_len * BK;
+    const int kend = min(kbeg + kc_len * BK, KK);
     stage(0, kbeg);
     __syncthreads();  // drains the DMA (vmcnt 0) + barrier
     int buf = 0;
-    for (int k0 = kbeg; k0 < kend; k0 += 32) {
-        if (k0 + 32 < kend) stage(buf ^ 1, k0 + 32);  // issue BEFORE compute
-        tile_mfma<BM, BN, WM, WN, FM, FN>(
-            *reinterpret_cast<const unsigned short(*)[BM][32]>(As(buf)),
-            *reinterpret_cast<const unsigned short(*)[BN][32]>(Bs(buf)), acc,
-            wave, lane);
+    for (int k0 = kbeg; k0 < kend; k0 += BK) {
+        if (k0 + BK < kend) stage(buf ^ 1, k0 + BK);  // issue BEFORE compute
+        {   // MFMA over the BK-deep tile, 32 k per instruction
+            const int wm = wave / WN, wn = wave % WN;
+            const int half = lane >> 4, sub = lane & 15;
+#pragma unroll
+            for (int kk = 0; kk < BK; kk += 32) {
+                bf16x8 a[FM], b[FN];
+#pragma unroll
+                for (int i = 0; i < FM; ++i)
+                    a[i] = *reinterpret_cast<const bf16x8*>(
+                        &As(buf)[wm * FM * 16 + i * 16 + sub][kk + half * 8]);
+#pragma unroll
+                for (int j = 0; j < FN; ++j)
+                    b[j] = *reinterpret_cast<const bf16x8*>(
+                        &Bs(buf)[wn * FN * 16 + j * 16 + sub][kk + half * 8]);
+#pragma unroll
+                for (int i = 0; i < FM; ++i)
+#pragma unroll
+                    for (int j = 0; j < FN; ++j)
+                        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            a[i], b[j], acc[i][j], 0, 0, 0);
+            }
+        }
         __syncthreads();
         buf ^= 1;
     }
@@ -1840,9 +1858,9 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
         if (k_chunks > 1) {
             auto y32 = torch::zeros({M, s.Kout},
                                     x.options().dtype(torch::kFloat32));
-            hipLaunchKernelGGL((conv_fwd_glds_kernel<64, 64, 2, 2, 2, 2>), grid,
-                               dim3(TPB), 0, stream, bf_ptr(x), bf_ptr(w), bias,
-                               bf_ptr_mut(y), y32.data_ptr<float>(),
+            hipLaunchKernelGGL((conv_fwd_glds_kernel<64, 64, 2, 2, 2, 2, 32>),
+                               grid, dim3(TPB), 0, stream, bf_ptr(x), bf_ptr(w),
+                               bias, bf_ptr_mut(y), y32.data_ptr<float>(),
                                bf_ptr(zbuf), s, relu ? 1 : 0, k_chunks);
             int64_t total = (int64_t)M * s.Kout;
             hipLaunchKernelGGL(linear_epilogue_kernel,
@@ -1850,10 +1868,10 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
                                dim3(256), 0, stream, y32.data_ptr<float>(), bias,
                                bf_ptr_mut(y), total, s.Kout, relu ? 1 : 0);
         } else {
-            hipLaunchKernelGGL((conv_fwd_glds_kernel<64, 64, 2, 2, 2, 2>), grid,
-                               dim3(TPB), 0, stream, bf_ptr(x), bf_ptr(w), bias,
-                               bf_ptr_mut(y), (float*)nullptr, bf_ptr(zbuf), s,
-                               relu ? 1 : 0, 1);
+            hipLaunchKernelGGL((conv_fwd_glds_kernel<64, 64, 2, 2, 2, 2, 32>),
+                               grid, dim3(TPB), 0, stream, bf_ptr(x), bf_ptr(w),
+                               bias, bf_ptr_mut(y), (float*)nullptr,
+                               bf_ptr(zbuf), s, relu ? 1 : 0, 1);
         }
     } else if (s.Kout > 16) {
         dim3 grid(ceildiv(M, 64), ceildiv(s.Kout, 64));
