@@ -9,7 +9,7 @@ LocalClient.local_train / train_server.
 from __future__ import annotations
 
 import copy
-from typing import Dict, List, Optional
+from typing import Dict
 
 import torch
 

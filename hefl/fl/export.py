@@ -96,3 +96,18 @@ def decrypt_into_model(he: Pyfhel, val: Dict[str, CtxtTensor],
             p.copy_(vec.reshape(p.shape).to(p.dtype))
     print("Time to decrypt:", time.time() - start)
     return model
+
+
+def export_plain_weights(model: torch.nn.Module, file_name: str = "plainweights.pickle") -> str:
+    """Plaintext weight export for size/time comparison — API parity with
+    the reference's notebook cell 6 (load_weights + export_weights of
+    unencrypted arrays to plainweights.pickle)."""
+    start = time.time()
+    os.makedirs(os.path.dirname(file_name) or ".", exist_ok=True)
+    val = {f"c_{i}_0": p.detach().float().cpu().numpy()
+           for i, p in enumerate(model.parameters())}
+    with open(file_name, "wb") as f:
+        pickle.dump({"key": None, "val": val}, f,
+                    protocol=pickle.HIGHEST_PROTOCOL)
+    print("Time to export weights to pickle:", time.time() - start)
+    return file_name

@@ -20,11 +20,9 @@ from typing import Dict, List, Optional
 import torch
 
 from ..config import RunConfig
-from ..data.shard import shard_indices
 from ..data.synthetic import SyntheticMedicalImages
 from ..he.ckks import CKKSContext, CtxtTensor
 from ..models import build_model
-from ..ops.adam import FusedAdam
 from .callbacks import EarlyStopping, ModelCheckpoint, ReduceLROnPlateau
 from .client import LocalClient
 from .metrics import classification_metrics

@@ -20,7 +20,7 @@ from __future__ import annotations
 import io
 import pickle
 import struct
-from typing import Optional, Union
+from typing import Optional
 
 import numpy as np
 import torch

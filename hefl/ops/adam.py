@@ -8,7 +8,6 @@ correction, decayed lr, parameter write) — replacing TF's Adam kernels
 """
 from __future__ import annotations
 
-import math
 from typing import Iterable
 
 import torch

@@ -141,3 +141,16 @@ def test_export_import_encrypted_weights(tmp_path, capsys):
     assert "Time to export weights to pickle:" in out  # :239
     assert "Time to import:" in out               # :327
     assert "Time to decrypt:" in out              # :267
+
+
+def test_export_plain_weights(tmp_path, capsys):
+    from hefl.fl.export import export_plain_weights
+    from hefl.models import CNN2
+    import pickle
+
+    m = CNN2((28, 28, 1), 2, seed=1)
+    p = export_plain_weights(m, str(tmp_path / "plainweights.pickle"))
+    with open(p, "rb") as f:
+        d = pickle.load(f)
+    assert len(d["val"]) == len(list(m.parameters()))
+    assert "Time to export weights to pickle:" in capsys.readouterr().out
