@@ -365,11 +365,19 @@ conv_fwd_glds_kernel(const unsigned short* __restrict__ x,
 
     constexpr int CPR = BK / 8;  // chunks per row
     int a_n[RPT], a_oh[RPT], a_ow[RPT];
+    int akcs[RPT], arow3[RPT];
     bool a_ok[RPT];
-    const int akc = (tid % CPR) * 8;
 #pragma unroll
     for (int t = 0; t < RPT; ++t) {
-        const int am = m0 + ((tid + t * TPB) / CPR);
+        const int arow = (tid + t * TPB) / CPR;
+        // XOR-swizzled chunk assignment (rule 21: permute the SOURCE, read
+        // with the same XOR): kills the 4-way ds_read_b128 bank conflict of
+        // rows-at-same-column fragment reads. Only defined for CPR == 4.
+        // key = (row>>2)&3: rows 4 apart share a 256-B LDS bank row quarter,
+        // so the slot choice must differ among them (row&3 would alias)
+        arow3[t] = (arow >> 2) & 3;
+        akcs[t] = ((tid % CPR) ^ (CPR == 4 ? arow3[t] : 0)) * 8;
+        const int am = m0 + arow;
         a_ok[t] = am < M;
         a_n[t] = a_oh[t] = a_ow[t] = 0;
         if (a_ok[t]) {
@@ -384,7 +392,7 @@ conv_fwd_glds_kernel(const unsigned short* __restrict__ x,
         // A: RPT chunks, each one 16-B DMA; lane-linear within the wave
 #pragma unroll
         for (int t = 0; t < RPT; ++t) {
-            const int k = k0 + akc;
+            const int k = k0 + akcs[t];
             const unsigned short* src = zbuf;
             if (a_ok[t] && k < KK) {
                 int rs = k / s.C, c = k % s.C;
@@ -401,8 +409,10 @@ conv_fwd_glds_kernel(const unsigned short* __restrict__ x,
 #pragma unroll
         for (int t = 0; t < BPT; ++t) {
             const int i = tid + t * TPB;
-            const int ko = n0 + i / CPR;
-            const int k = k0 + (i % CPR) * 8;
+            const int korow = i / CPR;
+            const int ko = n0 + korow;
+            const int kc = ((i % CPR) ^ (CPR == 4 ? ((korow >> 2) & 3) : 0)) * 8;
+            const int k = k0 + kc;
             const unsigned short* src = zbuf;
             if (ko < s.Kout && k < KK) src = w + (int64_t)ko * KK + k;
             char* base = (char*)&Bs(buf)[0][0] + (wave + t * 4) * 1024;
@@ -428,17 +438,20 @@ conv_fwd_glds_kernel(const unsigned short* __restrict__ x,
         {   // MFMA over the BK-deep tile, 32 k per instruction
             const int wm = wave / WN, wn = wave % WN;
             const int half = lane >> 4, sub = lane & 15;
+            const int swz = (BK == 32) ? ((sub >> 2) & 3) : 0;  // stage XOR key
 #pragma unroll
             for (int kk = 0; kk < BK; kk += 32) {
                 bf16x8 a[FM], b[FN];
 #pragma unroll
                 for (int i = 0; i < FM; ++i)
                     a[i] = *reinterpret_cast<const bf16x8*>(
-                        &As(buf)[wm * FM * 16 + i * 16 + sub][kk + half * 8]);
+                        &As(buf)[wm * FM * 16 + i * 16 + sub]
+                                [kk + ((half ^ swz) * 8)]);
 #pragma unroll
                 for (int j = 0; j < FN; ++j)
                     b[j] = *reinterpret_cast<const bf16x8*>(
-                        &Bs(buf)[wn * FN * 16 + j * 16 + sub][kk + half * 8]);
+                        &Bs(buf)[wn * FN * 16 + j * 16 + sub]
+                                [kk + ((half ^ swz) * 8)]);
 #pragma unroll
                 for (int i = 0; i < FM; ++i)
 #pragma unroll
