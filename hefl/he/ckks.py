@@ -493,12 +493,23 @@ class CKKSContext:
         qL = self._q(last)
         qL_half = qL // 2
         # coefficient-domain last limb, centered for round-to-nearest
-        cl = self.backend.ntt(data[..., last, :], last, inverse=True)
+        cl = self.backend.ntt(data[..., last, :].contiguous(), last,
+                              inverse=True)
+        cl_c = torch.where(cl > qL_half, cl - qL, cl)
+        if self.device.type == "cuda":
+            # fused across target limbs: one remainder broadcast, one fused
+            # NTT launch set, one scalar-limbs multiply
+            qs = self.backend.qs[:last]
+            shape = [1] * cl_c.dim() + [1]
+            shape[-2] = last
+            r = torch.remainder(cl_c.unsqueeze(-2), qs.view(shape))
+            r_ntt = self.backend.ntt_all(r)
+            diff = torch.remainder(data[..., :last, :] - r_ntt, qs.view(shape))
+            inv = [pow(qL % self._q(i), -1, self._q(i)) for i in range(last)]
+            return self.backend.modmul_scalar_limbs(diff, inv)
         out = data[..., :last, :].clone()
         for i in range(last):
             q = self._q(i)
-            # centered residue of cl mod q_i
-            cl_c = torch.where(cl > qL_half, cl - qL, cl)
             r = torch.remainder(cl_c, q)
             r_ntt = self.backend.ntt(r, i)
             diff = torch.remainder(out[..., i, :] - r_ntt, q)
