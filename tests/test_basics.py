@@ -115,3 +115,26 @@ def test_key_file_workflow(tmp_path):
     assert abs(sk.decryptFrac(s) - 2.5) < 1e-3
     gen_rekey(sk)        # reference's dead gen_rekey, functional here
     assert sk._keys is not None
+
+
+def test_metrics_match_sklearn():
+    # the reference computes precision/recall/f1 with sklearn
+    # average='weighted' + accuracy_score (notebook cell 3)
+    import numpy as np
+    from sklearn.metrics import (accuracy_score, f1_score, precision_score,
+                                 recall_score)
+
+    from hefl.fl.metrics import classification_metrics
+
+    rng = np.random.default_rng(0)
+    y_true = torch.from_numpy(rng.integers(0, 4, 300))
+    y_pred = torch.from_numpy(rng.integers(0, 4, 300))
+    m = classification_metrics(y_true, y_pred, 4)
+    assert abs(m["accuracy"] - accuracy_score(y_true, y_pred)) < 1e-9
+    assert abs(m["precision"] - precision_score(y_true, y_pred,
+                                                average="weighted",
+                                                zero_division=0)) < 1e-9
+    assert abs(m["recall"] - recall_score(y_true, y_pred, average="weighted",
+                                          zero_division=0)) < 1e-9
+    assert abs(m["f1"] - f1_score(y_true, y_pred, average="weighted",
+                                  zero_division=0)) < 1e-9
