@@ -1314,15 +1314,22 @@ __global__ void maxpool_bwd_kernel(const unsigned short* __restrict__ dy,
 // Also fuses the training statistics: when acc_loss/acc_correct are given,
 // the kernel accumulates mean loss and argmax==label counts into those
 // persistent buffers — replacing a ~7-kernel torch chain per step.
+// SINGLE_BLOCK: grid(1) with an intra-block reduction — loss is written
+// directly (no zero-fill + atomics); rows strided over the block's waves.
+template <bool SINGLE_BLOCK>
 __global__ void softmax_xent_fwd_kernel(const unsigned short* __restrict__ logits,
                                         const int64_t* __restrict__ labels,
                                         float* __restrict__ probs,
                                         float* __restrict__ loss, int M, int C,
                                         float* __restrict__ acc_loss,
                                         float* __restrict__ acc_correct) {
-    const int row = blockIdx.x * (blockDim.x / 64) + (threadIdx.x >> 6);
+    __shared__ float lred[2][8];
     const int lane = threadIdx.x & 63;
-    if (row >= M) return;
+    const int nw = blockDim.x / 64;
+    float my_loss = 0.f, my_corr = 0.f;
+    for (int row = SINGLE_BLOCK ? (threadIdx.x >> 6)
+                                : blockIdx.x * nw + (threadIdx.x >> 6);
+         row < M; row += SINGLE_BLOCK ? nw : gridDim.x * nw) {
     const unsigned short* lr = logits + (int64_t)row * C;
     float mx = -3.4e38f;
     int arg = 0;
@@ -1347,9 +1354,33 @@ __global__ void softmax_xent_fwd_kernel(const unsigned short* __restrict__ logit
         int64_t lab = labels[row];
         float p = __expf(bf2f(lr[lab]) - mx) * inv;
         float l = -__logf(fmaxf(p, 1e-30f)) / M;
-        atomicAdd(loss, l);
-        if (acc_loss) atomicAdd(acc_loss, l);
-        if (acc_correct && arg == (int)lab) atomicAdd(acc_correct, 1.f);
+        if (SINGLE_BLOCK) {
+            my_loss += l;
+            if (arg == (int)lab) my_corr += 1.f;
+        } else {
+            atomicAdd(loss, l);
+            if (acc_loss) atomicAdd(acc_loss, l);
+            if (acc_correct && arg == (int)lab) atomicAdd(acc_correct, 1.f);
+        }
+    }
+    }  // row loop
+    if (SINGLE_BLOCK) {
+        const int w = threadIdx.x >> 6;
+        if (lane == 0) {
+            lred[0][w] = my_loss;
+            lred[1][w] = my_corr;
+        }
+        __syncthreads();
+        if (threadIdx.x == 0) {
+            float L = 0.f, Cc = 0.f;
+            for (int i = 0; i < nw; ++i) {
+                L += lred[0][i];
+                Cc += lred[1][i];
+            }
+            loss[0] = L;                      // direct write: no zero-fill
+            if (acc_loss) atomicAdd(acc_loss, L);
+            if (acc_correct) atomicAdd(acc_correct, Cc);
+        }
     }
 }
 
@@ -1475,22 +1506,26 @@ __global__ void linear_splitk_kernel(const unsigned short* __restrict__ x,
         }
         acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
     }
+    // slab write [kc][M][N]: no zero-init, no atomics; epilogue sums slabs
+    float* slab = y32 + (int64_t)blockIdx.z * M * N;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
         int orow = m0 + (lane >> 4) * 4 + r;
         int ocol = n0 + (lane & 15);
         if (orow < M && ocol < N)
-            atomicAdd(y32 + (int64_t)orow * N + ocol, acc[r]);
+            slab[(int64_t)orow * N + ocol] = acc[r];
     }
 }
 
 __global__ void linear_epilogue_kernel(const float* __restrict__ y32,
                                        const float* __restrict__ bias,
                                        unsigned short* __restrict__ y,
-                                       int64_t total, int N, int relu) {
+                                       int64_t total, int N, int relu,
+                                       int slabs) {
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
          i += (int64_t)gridDim.x * blockDim.x) {
-        float v = y32[i] + (bias ? bias[i % N] : 0.f);
+        float v = bias ? bias[i % N] : 0.f;
+        for (int sl = 0; sl < slabs; ++sl) v += y32[sl * total + i];
         if (relu) v = v > 0.f ? v : 0.f;
         y[i] = f2bf(v);
     }
@@ -1514,10 +1549,13 @@ __global__ void fused_adam_mt_kernel(const int64_t* __restrict__ meta,
     const int c = blockIdx.x;
     const int t = (int)meta[c * 2];
     const int64_t off = meta[c * 2 + 1];
-    float* p = reinterpret_cast<float*>(ptrs[t * 4 + 0]);
-    const float* g = reinterpret_cast<const float*>(ptrs[t * 4 + 1]);
-    float* m = reinterpret_cast<float*>(ptrs[t * 4 + 2]);
-    float* v = reinterpret_cast<float*>(ptrs[t * 4 + 3]);
+    float* p = reinterpret_cast<float*>(ptrs[t * 5 + 0]);
+    const float* g = reinterpret_cast<const float*>(ptrs[t * 5 + 1]);
+    float* m = reinterpret_cast<float*>(ptrs[t * 5 + 2]);
+    float* v = reinterpret_cast<float*>(ptrs[t * 5 + 3]);
+    // bf16 shadow of the fp32 master (what conv/linear forward reads):
+    // refreshed here so no per-step cast kernels exist
+    unsigned short* sh = reinterpret_cast<unsigned short*>(ptrs[t * 5 + 4]);
     const int64_t n = min(off + (int64_t)MT_CHUNK, sizes[t]);
     const float lr = sched[0], bc1 = sched[1], bc2 = sched[2];
     for (int64_t i = off + threadIdx.x; i < n; i += blockDim.x) {
@@ -1526,7 +1564,9 @@ __global__ void fused_adam_mt_kernel(const int64_t* __restrict__ meta,
         float vi = b2 * v[i] + (1.f - b2) * gi * gi;
         m[i] = mi;
         v[i] = vi;
-        p[i] -= lr * (mi / bc1) / (sqrtf(vi / bc2) + eps);
+        float pv = p[i] - lr * (mi / bc1) / (sqrtf(vi / bc2) + eps);
+        p[i] = pv;
+        if (sh) sh[i] = f2bf(pv);
     }
 }
 
@@ -1865,7 +1905,7 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
         int tiles = ceildiv(M, 64) * ceildiv(s.Kout, 64);
         int ksteps = ceildiv(KKf, 32);
         int k_chunks = 1;
-        if (tiles < 256)
+        if (tiles < 256 && ksteps >= 16)
             k_chunks = std::max(1, std::min(ksteps / 4, 512 / tiles));
         dim3 grid(ceildiv(M, 64), ceildiv(s.Kout, 64), k_chunks);
         if (k_chunks > 1) {
@@ -1879,7 +1919,7 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
             hipLaunchKernelGGL(linear_epilogue_kernel,
                                dim3((int)std::min<int64_t>(ceildiv(total, 256), 2048)),
                                dim3(256), 0, stream, y32.data_ptr<float>(), bias,
-                               bf_ptr_mut(y), total, s.Kout, relu ? 1 : 0);
+                               bf_ptr_mut(y), total, s.Kout, relu ? 1 : 0, 1);
         } else {
             hipLaunchKernelGGL((conv_fwd_glds_kernel<64, 64, 2, 2, 2, 2, 32>),
                                grid, dim3(TPB), 0, stream, bf_ptr(x), bf_ptr(w),
@@ -1922,7 +1962,7 @@ torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor w, int64_t stride,
         int tiles = ceildiv(M, 64) * ceildiv(C, 64);
         int ksteps = ceildiv(KKd, 32);
         int k_chunks = 1;
-        if (tiles < 256)
+        if (tiles < 256 && ksteps >= 16)
             k_chunks = std::max(1, std::min(ksteps / 4, 512 / tiles));
         dim3 grid(ceildiv(M, 64), ceildiv(C, 64), k_chunks);
         torch::Tensor dx32;
@@ -2047,10 +2087,12 @@ torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
     const float* bias = b.numel() ? b.data_ptr<float>() : nullptr;
     auto y = torch::empty({M, N}, x.options());
     if (M <= 64) {
-        auto y32 = torch::zeros({M, N}, x.options().dtype(torch::kFloat32));
         int kc = std::max(256, K / 4);
         kc = ((kc + 31) / 32) * 32;
-        dim3 grid(ceildiv(M, 16), ceildiv(N, 16), ceildiv(K, kc));
+        const int slabs = ceildiv(K, kc);
+        auto y32 = torch::empty({slabs, M, N},
+                                x.options().dtype(torch::kFloat32));
+        dim3 grid(ceildiv(M, 16), ceildiv(N, 16), slabs);
         hipLaunchKernelGGL(linear_splitk_kernel, grid, dim3(64), 0, stream,
                            bf_ptr(x), bf_ptr(w), y32.data_ptr<float>(), M, N,
                            K, kc);
@@ -2058,7 +2100,7 @@ torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
         hipLaunchKernelGGL(linear_epilogue_kernel,
                            dim3((int)std::min<int64_t>(ceildiv(total, 256), 2048)),
                            dim3(256), 0, stream, y32.data_ptr<float>(), bias,
-                           bf_ptr_mut(y), total, N, relu ? 1 : 0);
+                           bf_ptr_mut(y), total, N, relu ? 1 : 0, slabs);
         return y;
     }
     dim3 grid(ceildiv(M, BM), ceildiv(N, BN));
@@ -2147,12 +2189,25 @@ std::vector<torch::Tensor> softmax_xent_fwd(torch::Tensor logits,
     TORCH_CHECK(logits.is_contiguous());
     const int M = (int)logits.size(0), C = (int)logits.size(1);
     auto probs = torch::empty({M, C}, logits.options().dtype(torch::kFloat32));
+    const int waves_per_block = 8;
+    auto stream = at::cuda::getCurrentCUDAStream();
+    if (M <= 1024) {
+        // single block: loss written directly, no zero-fill kernel
+        auto loss = torch::empty({}, logits.options().dtype(torch::kFloat32));
+        hipLaunchKernelGGL((softmax_xent_fwd_kernel<true>), dim3(1),
+                           dim3(64 * waves_per_block), 0, stream,
+                           bf_ptr(logits), labels.data_ptr<int64_t>(),
+                           probs.data_ptr<float>(), loss.data_ptr<float>(), M,
+                           C,
+                           acc_loss.numel() ? acc_loss.data_ptr<float>() : nullptr,
+                           acc_correct.numel() ? acc_correct.data_ptr<float>()
+                                               : nullptr);
+        return {loss, probs};
+    }
     auto loss = torch::zeros({}, logits.options().dtype(torch::kFloat32));
-    const int waves_per_block = 4;
     int blocks = ceildiv(M, waves_per_block);
-    hipLaunchKernelGGL(softmax_xent_fwd_kernel, dim3(blocks),
-                       dim3(64 * waves_per_block), 0,
-                       at::cuda::getCurrentCUDAStream(), bf_ptr(logits),
+    hipLaunchKernelGGL((softmax_xent_fwd_kernel<false>), dim3(blocks),
+                       dim3(64 * waves_per_block), 0, stream, bf_ptr(logits),
                        labels.data_ptr<int64_t>(), probs.data_ptr<float>(),
                        loss.data_ptr<float>(), M, C,
                        acc_loss.numel() ? acc_loss.data_ptr<float>() : nullptr,

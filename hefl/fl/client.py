@@ -112,6 +112,9 @@ class LocalClient:
             self._acc_loss = torch.zeros((), dtype=torch.float32, device=sx.device)
             self._acc_correct = torch.zeros((), dtype=torch.float32,
                                             device=sx.device)
+        # bf16 shadow weights must exist BEFORE capture so the captured
+        # forward reads them (instead of recording per-step cast kernels)
+        self.opt.ensure_shadows()
         # grads set to None: the captured backward steals fresh grad tensors
         # from the graph pool; replays rewrite them in place
         self.opt.zero_grad()
@@ -197,6 +200,8 @@ class LocalClient:
 
     def set_weights(self, vec: torch.Tensor) -> None:
         load_flat_params(self.model, vec.to(self.device))
+        if hasattr(self.opt, "refresh_shadows"):
+            self.opt.refresh_shadows()  # bf16 shadows must follow FedAvg loads
 
     @torch.no_grad()
     def evaluate(self, dataset: SyntheticMedicalImages, indices: torch.Tensor,

@@ -40,6 +40,9 @@ class FusedAdam:
             if p.is_cuda:
                 hefl.load_extension().fused_adam(
                     p.data, g, m, v, lr_t, self.beta1, self.beta2, self.eps, bc1, bc2)
+                sh = getattr(p, "_bf16", None)
+                if sh is not None:
+                    sh.copy_(p.data.to(torch.bfloat16))
             else:
                 g = g.float()
                 m.mul_(self.beta1).add_(g, alpha=1 - self.beta1)
@@ -77,10 +80,13 @@ class FusedAdam:
         assert all(p.grad is not None for p in self.params), \
             "grads must be materialized before build_mt_table"
         dev = self.params[0].device
+        self.ensure_shadows()
         ptrs, sizes, meta = [], [], []
         for t, (p, m, v) in enumerate(zip(self.params, self.m, self.v)):
+            sh = getattr(p, "_bf16", None)
             ptrs.append([p.data.data_ptr(), p.grad.data_ptr(),
-                         m.data_ptr(), v.data_ptr()])
+                         m.data_ptr(), v.data_ptr(),
+                         sh.data_ptr() if sh is not None else 0])
             sizes.append(p.numel())
             for off in range(0, p.numel(), self._MT_CHUNK):
                 meta.append([t, off])
@@ -88,6 +94,21 @@ class FusedAdam:
                 "sizes": torch.tensor(sizes, dtype=torch.int64, device=dev),
                 "meta": torch.tensor(meta, dtype=torch.int64, device=dev),
                 "n": len(meta)}
+
+    def ensure_shadows(self):
+        """bf16 shadow weights: the forward kernels read these instead of
+        casting the fp32 masters every step; the multi-tensor Adam kernel
+        refreshes them in the same pass that updates the masters."""
+        for p in self.params:
+            if p.is_cuda and getattr(p, "_bf16", None) is None:
+                p._bf16 = p.detach().to(torch.bfloat16).contiguous()
+
+    def refresh_shadows(self):
+        """Re-sync shadows after an external weight write (FedAvg load)."""
+        for p in self.params:
+            sh = getattr(p, "_bf16", None)
+            if sh is not None:
+                sh.copy_(p.detach().to(torch.bfloat16))
 
     @torch.no_grad()
     def step_mt(self, table):

@@ -24,6 +24,10 @@ def _C():
 
 
 def _gpu_dtype(t: torch.Tensor) -> torch.Tensor:
+    # prefer the optimizer-maintained bf16 shadow (no per-step cast kernel)
+    sh = getattr(t, "_bf16", None)
+    if sh is not None:
+        return sh
     return t if t.dtype == torch.bfloat16 else t.to(torch.bfloat16)
 
 
@@ -41,7 +45,7 @@ class _Conv2dFn(torch.autograd.Function):
         ctx.pad = pad
         ctx.has_bias = b is not None
         if x.is_cuda:
-            wb = _gpu_dtype(w.detach())
+            wb = _gpu_dtype(w)
             bb = b.detach().float() if b is not None else torch.empty(0, device=x.device)
             y = _C().conv2d_fwd(x.contiguous(), wb.contiguous(), bb, stride, relu, pad)
             ctx.save_for_backward(x, wb, y)
@@ -144,7 +148,7 @@ class _LinearFn(torch.autograd.Function):
         ctx.relu = relu
         ctx.has_bias = b is not None
         if x.is_cuda:
-            wb = _gpu_dtype(w.detach())
+            wb = _gpu_dtype(w)
             bb = b.detach().float() if b is not None else torch.empty(0, device=x.device)
             y = _C().linear_fwd(x, wb, bb, relu)
             ctx.save_for_backward(x, wb, y)
