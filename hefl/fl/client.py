@@ -112,6 +112,7 @@ class LocalClient:
             self._acc_loss = torch.zeros((), dtype=torch.float32, device=sx.device)
             self._acc_correct = torch.zeros((), dtype=torch.float32,
                                             device=sx.device)
+            self._one = torch.ones((), dtype=torch.float32, device=sx.device)
         # bf16 shadow weights must exist BEFORE capture so the captured
         # forward reads them (instead of recording per-step cast kernels)
         self.opt.ensure_shadows()
@@ -122,7 +123,8 @@ class LocalClient:
             logits = self.model(sx.to(self.compute_dtype))
             # stats accumulate INSIDE the loss kernel into persistent buffers
             loss = softmax_xent(logits, sy, self._acc_loss, self._acc_correct)
-            loss.backward()
+            # persistent gradient seed: no per-step ones() fill kernel
+            loss.backward(gradient=self._one)
         mt = self.opt.build_mt_table()  # this graph's stolen-grad pointers
         return {"graph": g, "x": sx, "y": sy, "loss": loss, "logits": logits,
                 "mt": mt}
