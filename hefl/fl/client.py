@@ -126,8 +126,12 @@ class LocalClient:
             # persistent gradient seed: no per-step ones() fill kernel
             loss.backward(gradient=self._one)
         mt = self.opt.build_mt_table()  # this graph's stolen-grad pointers
+        # hold the stolen grad tensors: a later capture re-steals p.grad and
+        # would otherwise drop the refs, returning these pool blocks to the
+        # allocator while this graph's kernels still write them
+        grads = [p.grad for p in self.opt.params]
         return {"graph": g, "x": sx, "y": sy, "loss": loss, "logits": logits,
-                "mt": mt}
+                "mt": mt, "grads": grads}
 
     def _eager_warmup(self, x, y):
         self.opt.zero_grad()  # grads=None: fresh tensors each warmup step
