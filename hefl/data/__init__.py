@@ -1,4 +1,7 @@
-from .synthetic import SyntheticMedicalImages, make_client_loader
-from .shard import shard_indices, prep_df
+from .pipeline import get_test_data, get_train_data, hflip_augment
+from .shard import prep_df, shard_indices
+from .synthetic import ClientLoader, SyntheticMedicalImages, make_client_loader
 
-__all__ = ["SyntheticMedicalImages", "make_client_loader", "shard_indices", "prep_df"]
+__all__ = ["SyntheticMedicalImages", "ClientLoader", "make_client_loader",
+           "shard_indices", "prep_df", "get_train_data", "get_test_data",
+           "hflip_augment"]

@@ -1,6 +1,10 @@
-from .functional import conv2d, maxpool2x2, linear, softmax_xent
-from .modules import Conv2dValid, MaxPool2x2, Dense, Flatten
+from .functional import (add_relu, batchnorm2d, conv2d, global_avgpool,
+                         linear, maxpool, maxpool2x2, softmax_xent)
+from .modules import (BatchNorm2d, Conv2dValid, Dense, Flatten,
+                      GlobalAvgPool, MaxPool, MaxPool2x2)
 from .adam import FusedAdam
 
-__all__ = ["conv2d", "maxpool2x2", "linear", "softmax_xent",
-           "Conv2dValid", "MaxPool2x2", "Dense", "Flatten", "FusedAdam"]
+__all__ = ["conv2d", "maxpool2x2", "maxpool", "linear", "softmax_xent",
+           "batchnorm2d", "global_avgpool", "add_relu",
+           "Conv2dValid", "MaxPool2x2", "MaxPool", "Dense", "Flatten",
+           "BatchNorm2d", "GlobalAvgPool", "FusedAdam"]
