@@ -13,6 +13,7 @@ torch::Tensor modmul_scalar(torch::Tensor a, int64_t s, int64_t q);
 torch::Tensor modadd(torch::Tensor a, torch::Tensor b, int64_t q);
 torch::Tensor modsub(torch::Tensor a, torch::Tensor b, int64_t q);
 void modreduce_(torch::Tensor x, torch::Tensor qs);
+torch::Tensor cbd21(torch::Tensor bits);
 void ntt_limbs(torch::Tensor x, torch::Tensor w, torch::Tensor wsh,
                torch::Tensor qs, int64_t L);
 void intt_limbs(torch::Tensor x, torch::Tensor winv, torch::Tensor winvsh,
@@ -86,6 +87,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("modadd", &modadd);
     m.def("modsub", &modsub);
     m.def("modreduce_", &modreduce_, "in-place per-limb reduction after lazy sum");
+    m.def("cbd21", &cbd21, "centered binomial eta=21 from 64-bit draws");
     m.def("ntt_limbs", &ntt_limbs, "fused multi-limb forward NTT [R, L, n]");
     m.def("intt_limbs", &intt_limbs, "fused multi-limb inverse NTT");
     m.def("modmul_limbs", &modmul_limbs);

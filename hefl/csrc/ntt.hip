@@ -520,6 +520,19 @@ __global__ void modreduce_kernel(int64_t* __restrict__ x,
     }
 }
 
+// Centered binomial eta=21 from ONE uniform 64-bit draw per coefficient:
+// e = popcount(bits[0:21]) - popcount(bits[21:42]) — replaces 42 separate
+// int8 Bernoulli draws (the sampling half of encrypt's cost at ResNet scale).
+__global__ void cbd21_kernel(const int64_t* __restrict__ bits,
+                             int64_t* __restrict__ out, int64_t total) {
+    constexpr uint64_t M21 = (1ull << 21) - 1;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        uint64_t x = (uint64_t)bits[i];
+        out[i] = (int64_t)(__popcll(x & M21) - __popcll((x >> 21) & M21));
+    }
+}
+
 int64_t shoup_of(uint64_t w, uint64_t q) {
     unsigned __int128 t = ((unsigned __int128)w) << 64;
     return (int64_t)(uint64_t)(t / q);
@@ -747,5 +760,16 @@ torch::Tensor modmul_scalar_limbs(torch::Tensor a, torch::Tensor scalars,
                        a.data_ptr<int64_t>(), out.data_ptr<int64_t>(), total,
                        scalars.data_ptr<int64_t>(), shoups.data_ptr<int64_t>(),
                        qs.data_ptr<int64_t>(), (int)L, n);
+    return out;
+}
+
+torch::Tensor cbd21(torch::Tensor bits) {
+    CHECK_CUDA_OK(bits);
+    auto out = torch::empty_like(bits);
+    int64_t total = bits.numel();
+    int blocks = (int)std::min<int64_t>((total + kThreads - 1) / kThreads, 2048);
+    hipLaunchKernelGGL(cbd21_kernel, dim3(blocks), dim3(kThreads), 0,
+                       at::cuda::getCurrentCUDAStream(),
+                       bits.data_ptr<int64_t>(), out.data_ptr<int64_t>(), total);
     return out;
 }

@@ -283,11 +283,12 @@ class CKKSContext:
     def _sample_err(self, shape, eta: int = 21, host: bool = False) -> torch.Tensor:
         # centered binomial, sigma = sqrt(eta/2) ~= 3.24 (SEAL sigma 3.2)
         if not host and self.device.type == "cuda":
-            b = torch.randint(0, 2, (2 * eta,) + tuple(shape),
-                              generator=self.gpu_gen, device=self.device,
-                              dtype=torch.int8)
-            return (b[:eta].sum(0, dtype=torch.int64)
-                    - b[eta:].sum(0, dtype=torch.int64))
+            # one uniform 64-bit draw per coefficient; the HIP kernel takes
+            # popcount(bits[0:21]) - popcount(bits[21:42])
+            bits = torch.randint(-(2 ** 63), 2 ** 63 - 1, tuple(shape),
+                                 generator=self.gpu_gen, device=self.device,
+                                 dtype=torch.int64)
+            return self.backend._C.cbd21(bits)
         b = self._cpu_rng.integers(0, 2, size=(eta,) + tuple(shape)).sum(axis=0)
         b2 = self._cpu_rng.integers(0, 2, size=(eta,) + tuple(shape)).sum(axis=0)
         return torch.from_numpy(b - b2).to(torch.int64)
