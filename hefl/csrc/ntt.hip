@@ -529,7 +529,9 @@ __global__ void cbd21_kernel(const int64_t* __restrict__ bits,
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
          i += (int64_t)gridDim.x * blockDim.x) {
         uint64_t x = (uint64_t)bits[i];
-        out[i] = (int64_t)(__popcll(x & M21) - __popcll((x >> 21) & M21));
+        // __popcll returns unsigned: cast BEFORE subtracting or negative
+        // differences wrap to ~2^32
+        out[i] = (int64_t)((int)__popcll(x & M21) - (int)__popcll((x >> 21) & M21));
     }
 }
 
