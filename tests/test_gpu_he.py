@@ -165,3 +165,19 @@ def test_fused_multilimb_ops_match_per_limb(n):
     per = torch.stack([be.modmul_scalar(x[:, i, :].contiguous(),
                                         scalars[i], i) for i in range(L)], dim=1)
     assert torch.equal(ms, per)
+
+
+def test_device_noise_sampler_distribution():
+    """cbd21 must be a centered binomial with sigma = sqrt(21/2) ~ 3.24
+    (regression: an unsigned popcount subtraction once wrapped negatives to
+    ~2^32, silently destroying every ciphertext)."""
+    import hefl
+    C = hefl.load_extension()
+    g = torch.Generator(device="cuda")
+    g.manual_seed(3)
+    bits = torch.randint(-(2 ** 63), 2 ** 63 - 1, (1_000_000,), generator=g,
+                         device="cuda", dtype=torch.int64)
+    e = C.cbd21(bits).float()
+    assert abs(e.mean().item()) < 0.02
+    assert abs(e.std().item() - (21 / 2) ** 0.5) < 0.05
+    assert e.abs().max().item() <= 21
