@@ -1592,6 +1592,7 @@ __global__ void bn_partial_kernel(const unsigned short* __restrict__ x,
                                   float* __restrict__ gsum,
                                   float* __restrict__ gsq, int64_t M, int C,
                                   int rows_per_block) {
+    __shared__ float red[2][256];
     const int64_t r0 = (int64_t)blockIdx.x * rows_per_block;
     const int64_t r1 = min(r0 + rows_per_block, M);
     if (C >= (int)blockDim.x) {  // one channel per thread, strided
@@ -1607,19 +1608,33 @@ __global__ void bn_partial_kernel(const unsigned short* __restrict__ x,
         }
         return;
     }
-    // thread = (channel, row-lane): all 256 lanes stay busy when C < 256
+    // thread = (channel, row-lane); LDS-reduce the row-lanes so each block
+    // issues ONE atomic per channel (hot-word atomics dominated before)
     const int lanes = (int)blockDim.x / C;           // row-parallel lanes
     const int c = threadIdx.x % C;
     const int rl = threadIdx.x / C;
-    if (rl >= lanes) return;
     float acc = 0.f, acc2 = 0.f;
-    for (int64_t r = r0 + rl; r < r1; r += lanes) {
-        float v = bf2f(x[r * C + c]);
-        acc += v;
-        acc2 += v * v;
+    if (rl < lanes) {
+        for (int64_t r = r0 + rl; r < r1; r += lanes) {
+            float v = bf2f(x[r * C + c]);
+            acc += v;
+            acc2 += v * v;
+        }
     }
-    atomicAdd(gsum + c, acc);
-    atomicAdd(gsq + c, acc2);
+    red[0][threadIdx.x] = acc;
+    red[1][threadIdx.x] = acc2;
+    __syncthreads();
+    for (int off = lanes >> 1; off >= 1; off >>= 1) {
+        if (rl < off) {
+            red[0][threadIdx.x] += red[0][threadIdx.x + off * C];
+            red[1][threadIdx.x] += red[1][threadIdx.x + off * C];
+        }
+        __syncthreads();
+    }
+    if (rl == 0) {
+        atomicAdd(gsum + c, red[0][threadIdx.x]);
+        atomicAdd(gsq + c, red[1][threadIdx.x]);
+    }
 }
 
 __global__ void bn_finalize_kernel(const float* __restrict__ gsum,
@@ -1661,6 +1676,7 @@ __global__ void bn_bwd_partial_kernel(const unsigned short* __restrict__ dy,
                                       float* __restrict__ dgamma,
                                       float* __restrict__ dbeta, int64_t M,
                                       int C, int rows_per_block) {
+    __shared__ float red[2][256];
     const int64_t r0 = (int64_t)blockIdx.x * rows_per_block;
     const int64_t r1 = min(r0 + rows_per_block, M);
     if (C >= (int)blockDim.x) {
@@ -1680,16 +1696,29 @@ __global__ void bn_bwd_partial_kernel(const unsigned short* __restrict__ dy,
     const int lanes = (int)blockDim.x / C;
     const int c = threadIdx.x % C;
     const int rl = threadIdx.x / C;
-    if (rl >= lanes) return;
     float dg = 0.f, db = 0.f;
-    const float mu = mean[c], is = invstd[c];
-    for (int64_t r = r0 + rl; r < r1; r += lanes) {
-        float g = bf2f(dy[r * C + c]);
-        dg += g * (bf2f(x[r * C + c]) - mu) * is;
-        db += g;
+    if (rl < lanes) {
+        const float mu = mean[c], is = invstd[c];
+        for (int64_t r = r0 + rl; r < r1; r += lanes) {
+            float g = bf2f(dy[r * C + c]);
+            dg += g * (bf2f(x[r * C + c]) - mu) * is;
+            db += g;
+        }
     }
-    atomicAdd(dgamma + c, dg);
-    atomicAdd(dbeta + c, db);
+    red[0][threadIdx.x] = dg;
+    red[1][threadIdx.x] = db;
+    __syncthreads();
+    for (int off = lanes >> 1; off >= 1; off >>= 1) {
+        if (rl < off) {
+            red[0][threadIdx.x] += red[0][threadIdx.x + off * C];
+            red[1][threadIdx.x] += red[1][threadIdx.x + off * C];
+        }
+        __syncthreads();
+    }
+    if (rl == 0) {
+        atomicAdd(dgamma + c, red[0][threadIdx.x]);
+        atomicAdd(dbeta + c, red[1][threadIdx.x]);
+    }
 }
 
 __global__ void bn_dx_kernel(const unsigned short* __restrict__ dy,
@@ -2301,7 +2330,7 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
     auto stream = at::cuda::getCurrentCUDAStream();
     auto gsum = torch::zeros({C}, f32);
     auto gsq = torch::zeros({C}, f32);
-    int rpb = (int)std::max<int64_t>(64, (M + 511) / 512);
+    int rpb = (int)std::max<int64_t>(64, (M + 127) / 128);
     int nblk = (int)((M + rpb - 1) / rpb);
     hipLaunchKernelGGL(bn_partial_kernel, dim3(nblk), dim3(256), 0, stream,
                        bf_ptr(x), gsum.data_ptr<float>(), gsq.data_ptr<float>(),
@@ -2348,7 +2377,7 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
     auto dbeta = torch::zeros({C}, f32);
     auto dx = torch::empty_like(x);
     auto stream = at::cuda::getCurrentCUDAStream();
-    int rpb = (int)std::max<int64_t>(64, (M + 511) / 512);
+    int rpb = (int)std::max<int64_t>(64, (M + 127) / 128);
     int nblk = (int)((M + rpb - 1) / rpb);
     hipLaunchKernelGGL(bn_bwd_partial_kernel, dim3(nblk), dim3(256), 0, stream,
                        bf_ptr(dyc), bf_ptr(x), mean.data_ptr<float>(),
