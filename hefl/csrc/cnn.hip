@@ -1609,8 +1609,10 @@ __global__ void bn_partial_kernel(const unsigned short* __restrict__ x,
         return;
     }
     // thread = (channel, row-lane); LDS-reduce the row-lanes so each block
-    // issues ONE atomic per channel (hot-word atomics dominated before)
-    const int lanes = (int)blockDim.x / C;           // row-parallel lanes
+    // issues ONE atomic per channel (hot-word atomics dominated before).
+    // lanes rounded down to a power of two so the tree reduction is exact.
+    int lanes = (int)blockDim.x / C;                 // row-parallel lanes
+    lanes = 1 << (31 - __clz(lanes));
     const int c = threadIdx.x % C;
     const int rl = threadIdx.x / C;
     float acc = 0.f, acc2 = 0.f;
@@ -1693,7 +1695,8 @@ __global__ void bn_bwd_partial_kernel(const unsigned short* __restrict__ dy,
         }
         return;
     }
-    const int lanes = (int)blockDim.x / C;
+    int lanes = (int)blockDim.x / C;
+    lanes = 1 << (31 - __clz(lanes));
     const int c = threadIdx.x % C;
     const int rl = threadIdx.x / C;
     float dg = 0.f, db = 0.f;
