@@ -46,10 +46,11 @@ __global__ void ntt_global_stage_kernel(int64_t* __restrict__ x,
     const int64_t row = blockIdx.y;
     int64_t base_off = row * (int64_t)n;
     const uint32_t t = (uint32_t)(n / (2 * m));
+    const int tlog = 31 - __clz(t);  // t is a power of two
     for (int64_t k = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; k < nhalf;
          k += (int64_t)gridDim.x * blockDim.x) {
-        uint32_t j = (uint32_t)(k / t);
-        uint32_t pos = (uint32_t)(k % t);
+        uint32_t j = (uint32_t)(k >> tlog);
+        uint32_t pos = (uint32_t)k & (t - 1);
         int64_t i0 = base_off + (int64_t)j * 2 * t + pos;
         uint64_t W = (uint64_t)w[m + j];
         uint64_t Wsh = (uint64_t)wsh[m + j];
@@ -75,11 +76,12 @@ ntt_lds_kernel(int64_t* __restrict__ x, const int64_t* __restrict__ w,
     const int nb2 = nblk >> 1;
     for (int m = n / nblk; m < n; m <<= 1) {
         const uint32_t t = (uint32_t)(n / (2 * m));  // <= nblk/2 here
+        const int tlog = 31 - __clz(t);              // t is a power of two
         for (int lb = tid; lb < nb2; lb += kThreads) {
-            uint32_t jloc = (uint32_t)lb / t;
-            uint32_t pos = (uint32_t)lb % t;
+            uint32_t jloc = (uint32_t)lb >> tlog;
+            uint32_t pos = (uint32_t)lb & (t - 1);
             uint32_t base = jloc * 2 * t + pos;
-            uint32_t jglob = (uint32_t)blk * ((uint32_t)nb2 / t) + jloc;
+            uint32_t jglob = (uint32_t)blk * ((uint32_t)nb2 >> tlog) + jloc;
             uint64_t W = (uint64_t)w[m + jglob];
             uint64_t Wsh = (uint64_t)wsh[m + jglob];
             uint64_t U = (uint64_t)smem[base];
@@ -112,11 +114,12 @@ intt_lds_kernel(int64_t* __restrict__ x, const int64_t* __restrict__ winv,
     for (int m = n; m >= 2 * (n / nblk); m >>= 1) {
         const int h = m >> 1;
         const uint32_t t = (uint32_t)(n / m);
+        const int tlog = 31 - __clz(t);              // t is a power of two
         for (int lb = tid; lb < nb2; lb += kThreads) {
-            uint32_t jloc = (uint32_t)lb / t;
-            uint32_t pos = (uint32_t)lb % t;
+            uint32_t jloc = (uint32_t)lb >> tlog;
+            uint32_t pos = (uint32_t)lb & (t - 1);
             uint32_t base = jloc * 2 * t + pos;
-            uint32_t jglob = (uint32_t)blk * ((uint32_t)nb2 / t) + jloc;
+            uint32_t jglob = (uint32_t)blk * ((uint32_t)nb2 >> tlog) + jloc;
             uint64_t S = (uint64_t)winv[h + jglob];
             uint64_t Ssh = (uint64_t)winvsh[h + jglob];
             uint64_t U = (uint64_t)smem[base];
@@ -145,10 +148,11 @@ __global__ void intt_global_stage_kernel(int64_t* __restrict__ x,
     int64_t base_off = row * (int64_t)n;
     const int h = m >> 1;
     const uint32_t t = (uint32_t)(n / m);
+    const int tlog = 31 - __clz(t);  // t is a power of two
     for (int64_t k = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; k < nhalf;
          k += (int64_t)gridDim.x * blockDim.x) {
-        uint32_t j = (uint32_t)(k / t);
-        uint32_t pos = (uint32_t)(k % t);
+        uint32_t j = (uint32_t)(k >> tlog);
+        uint32_t pos = (uint32_t)k & (t - 1);
         int64_t i0 = base_off + (int64_t)j * 2 * t + pos;
         uint64_t S = (uint64_t)winv[h + j];
         uint64_t Ssh = (uint64_t)winvsh[h + j];
@@ -184,10 +188,11 @@ __global__ void ntt_global_stage_ml_kernel(int64_t* __restrict__ x,
     const int64_t* wshl = wsh + (int64_t)limb * n;
     int64_t base_off = row * (int64_t)n;
     const uint32_t t = (uint32_t)(n / (2 * m));
+    const int tlog = 31 - __clz(t);  // t is a power of two
     for (int64_t k = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; k < nhalf;
          k += (int64_t)gridDim.x * blockDim.x) {
-        uint32_t j = (uint32_t)(k / t);
-        uint32_t pos = (uint32_t)(k % t);
+        uint32_t j = (uint32_t)(k >> tlog);
+        uint32_t pos = (uint32_t)k & (t - 1);
         int64_t i0 = base_off + (int64_t)j * 2 * t + pos;
         uint64_t U = (uint64_t)x[i0];
         uint64_t V = mulmod_shoup((uint64_t)x[i0 + t], (uint64_t)wl[m + j],
@@ -219,7 +224,7 @@ ntt_lds_ml_kernel(int64_t* __restrict__ x, const int64_t* __restrict__ w,
             uint32_t jloc = (uint32_t)lb / t;
             uint32_t pos = (uint32_t)lb % t;
             uint32_t base = jloc * 2 * t + pos;
-            uint32_t jglob = (uint32_t)blk * ((uint32_t)nb2 / t) + jloc;
+            uint32_t jglob = (uint32_t)blk * ((uint32_t)nb2 >> tlog) + jloc;
             uint64_t U = (uint64_t)smem[base];
             uint64_t V = mulmod_shoup((uint64_t)smem[base + t],
                                       (uint64_t)wl[m + jglob],
@@ -254,11 +259,12 @@ intt_lds_ml_kernel(int64_t* __restrict__ x, const int64_t* __restrict__ winv,
     for (int m = n; m >= 2 * (n / nblk); m >>= 1) {
         const int h = m >> 1;
         const uint32_t t = (uint32_t)(n / m);
+        const int tlog = 31 - __clz(t);              // t is a power of two
         for (int lb = tid; lb < nb2; lb += kThreads) {
-            uint32_t jloc = (uint32_t)lb / t;
-            uint32_t pos = (uint32_t)lb % t;
+            uint32_t jloc = (uint32_t)lb >> tlog;
+            uint32_t pos = (uint32_t)lb & (t - 1);
             uint32_t base = jloc * 2 * t + pos;
-            uint32_t jglob = (uint32_t)blk * ((uint32_t)nb2 / t) + jloc;
+            uint32_t jglob = (uint32_t)blk * ((uint32_t)nb2 >> tlog) + jloc;
             uint64_t U = (uint64_t)smem[base];
             uint64_t V = (uint64_t)smem[base + t];
             smem[base] = (int64_t)addmod_u64(U, V, q);
@@ -295,10 +301,11 @@ __global__ void intt_global_stage_ml_kernel(int64_t* __restrict__ x,
     int64_t base_off = row * (int64_t)n;
     const int h = m >> 1;
     const uint32_t t = (uint32_t)(n / m);
+    const int tlog = 31 - __clz(t);  // t is a power of two
     for (int64_t k = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; k < nhalf;
          k += (int64_t)gridDim.x * blockDim.x) {
-        uint32_t j = (uint32_t)(k / t);
-        uint32_t pos = (uint32_t)(k % t);
+        uint32_t j = (uint32_t)(k >> tlog);
+        uint32_t pos = (uint32_t)k & (t - 1);
         int64_t i0 = base_off + (int64_t)j * 2 * t + pos;
         uint64_t U = (uint64_t)x[i0];
         uint64_t V = (uint64_t)x[i0 + t];
