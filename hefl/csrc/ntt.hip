@@ -220,9 +220,10 @@ ntt_lds_ml_kernel(int64_t* __restrict__ x, const int64_t* __restrict__ w,
     const int nb2 = nblk >> 1;
     for (int m = n / nblk; m < n; m <<= 1) {
         const uint32_t t = (uint32_t)(n / (2 * m));
+        const int tlog = 31 - __clz(t);              // t is a power of two
         for (int lb = tid; lb < nb2; lb += kThreads) {
-            uint32_t jloc = (uint32_t)lb / t;
-            uint32_t pos = (uint32_t)lb % t;
+            uint32_t jloc = (uint32_t)lb >> tlog;
+            uint32_t pos = (uint32_t)lb & (t - 1);
             uint32_t base = jloc * 2 * t + pos;
             uint32_t jglob = (uint32_t)blk * ((uint32_t)nb2 >> tlog) + jloc;
             uint64_t U = (uint64_t)smem[base];
