@@ -227,6 +227,19 @@ class CKKSContext:
         else:
             self.backend = CpuBackend(self.all_primes, self.n)
         self._cpu_rng = np.random.default_rng(cfg.seed)
+        # security check against the HE standard (the reference passes
+        # sec=128 to Pyfhel, FLPyfhelin.py:332; here it is verified):
+        # total modulus = chain + special prime
+        from .primes import max_logqp_128
+        total_bits = sum(q.bit_length() for q in self.all_primes)
+        limit = max_logqp_128(self.n)
+        self.secure_128 = total_bits <= limit
+        if cfg.sec >= 128 and not self.secure_128 and self.n >= 1024:
+            import warnings
+            warnings.warn(
+                f"CKKS parameters (n={self.n}, logQP={total_bits}) exceed the "
+                f"HE-standard 128-bit bound ({limit}); reduce q_bits or raise m",
+                stacklevel=2)
 
     def reseed(self, salt: int) -> None:
         """Re-seed encryption randomness (e.g. per FL rank) after a shared-seed

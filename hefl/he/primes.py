@@ -74,3 +74,17 @@ def bit_reverse(x: int, bits: int) -> int:
         r = (r << 1) | (x & 1)
         x >>= 1
     return r
+
+
+# HE-standard (homomorphicencryption.org) maximum log2(Q*P) for classical
+# 128-bit security with ternary secrets, per ring dimension.
+HE_STD_128_CLASSICAL = {
+    1024: 27, 2048: 54, 4096: 109, 8192: 218, 16384: 438, 32768: 881,
+}
+
+
+def max_logqp_128(n: int) -> int:
+    """Largest total modulus (chain + special prime) bit count meeting
+    classical 128-bit security at ring dimension n. Dimensions below 1024
+    (test-only) return 0 — never secure."""
+    return HE_STD_128_CLASSICAL.get(n, 0)

@@ -124,3 +124,24 @@ def test_rescale_matches_exact_division(ctx, keys):
     r = ctx.rescale(ctx.mul_scalar(ct, 1.0))
     out = ctx.decode(ctx.decrypt(r, keys.sk), ctx.slots)
     assert np.abs(out - v).max() < 1e-4
+
+
+def test_security_standard_check():
+    """BASELINE configs meet the HE-standard 128-bit classical bound
+    (chain + special prime)."""
+    from hefl.config import preset
+    for name in ("config2", "config3", "config4", "config5"):
+        cfg = preset(name).he
+        ctx = CKKSContext(HEConfig(m=cfg.m, scale_bits=cfg.scale_bits,
+                                   q_bits=cfg.q_bits, seed=0))
+        assert ctx.secure_128, (name, sum(q.bit_length()
+                                          for q in ctx.all_primes))
+
+
+def test_insecure_params_warn():
+    import warnings
+    with warnings.catch_warnings(record=True) as w:
+        warnings.simplefilter("always")
+        CKKSContext(HEConfig(m=1024, scale_bits=40, q_bits=(60, 40, 60),
+                             seed=0))  # logQP ~ 220 >> 27
+        assert any("128-bit" in str(x.message) for x in w)
