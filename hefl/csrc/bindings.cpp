@@ -55,7 +55,9 @@ void fused_adam_sched(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                       double b2, double eps);
 torch::Tensor relu_bwd(torch::Tensor dy, torch::Tensor y);
 std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
-                                  torch::Tensor beta, double eps, bool relu);
+                                  torch::Tensor beta, torch::Tensor rmean,
+                                  torch::Tensor rvar, double eps,
+                                  double momentum, bool relu);
 torch::Tensor bn_apply(torch::Tensor x, torch::Tensor mean,
                        torch::Tensor invstd, torch::Tensor gamma,
                        torch::Tensor beta, bool relu);

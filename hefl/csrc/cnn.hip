@@ -1642,14 +1642,21 @@ __global__ void bn_partial_kernel(const unsigned short* __restrict__ x,
 __global__ void bn_finalize_kernel(const float* __restrict__ gsum,
                                    const float* __restrict__ gsq,
                                    float* __restrict__ mean,
-                                   float* __restrict__ invstd, int64_t M,
-                                   int C, float eps) {
+                                   float* __restrict__ invstd,
+                                   float* __restrict__ running_mean,
+                                   float* __restrict__ running_var, int64_t M,
+                                   int C, float eps, float momentum) {
     for (int c = blockIdx.x * blockDim.x + threadIdx.x; c < C;
          c += gridDim.x * blockDim.x) {
         float mu = gsum[c] / (float)M;
         float var = fmaxf(gsq[c] / (float)M - mu * mu, 0.f);
         mean[c] = mu;
         invstd[c] = rsqrtf(var + eps);
+        if (running_mean) {  // torch semantics: UNBIASED var in running_var
+            float unb = var * ((float)M / (float)max(M - 1, (int64_t)1));
+            running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mu;
+            running_var[c] = (1.f - momentum) * running_var[c] + momentum * unb;
+        }
     }
 }
 
@@ -2321,7 +2328,9 @@ torch::Tensor relu_bwd(torch::Tensor dy, torch::Tensor y) {
 }
 
 std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
-                                  torch::Tensor beta, double eps, bool relu) {
+                                  torch::Tensor beta, torch::Tensor rmean,
+                                  torch::Tensor rvar, double eps,
+                                  double momentum, bool relu) {
     CHECK_GPU(x);
     TORCH_CHECK(x.is_contiguous());
     const int C = (int)x.size(-1);
@@ -2340,8 +2349,10 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
                        M, C, rpb);
     hipLaunchKernelGGL(bn_finalize_kernel, dim3(ceildiv(C, 256)), dim3(256), 0,
                        stream, gsum.data_ptr<float>(), gsq.data_ptr<float>(),
-                       mean.data_ptr<float>(), invstd.data_ptr<float>(), M, C,
-                       (float)eps);
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       rmean.numel() ? rmean.data_ptr<float>() : nullptr,
+                       rvar.numel() ? rvar.data_ptr<float>() : nullptr, M, C,
+                       (float)eps, (float)momentum);
     int64_t total = x.numel();
     int blocks = std::min<int64_t>(ceildiv(total, 256), 2048);
     hipLaunchKernelGGL(bn_apply_kernel, dim3(blocks), dim3(256), 0, stream,

@@ -246,8 +246,15 @@ class _BatchNorm2dFn(torch.autograd.Function):
         C = x.shape[-1]
         if x.is_cuda:
             if training:
-                y, mean, invstd = _C().bn_fwd(x.contiguous(), gamma.detach().float(),
-                                              beta.detach().float(), eps, relu)
+                # running stats update fused into the finalize kernel (the
+                # torch mul_/add_ chain was 4 kernels per BN layer per step)
+                empty = torch.empty(0, device=x.device)
+                y, mean, invstd = _C().bn_fwd(
+                    x.contiguous(), gamma.detach().float(),
+                    beta.detach().float(),
+                    running_mean if running_mean is not None else empty,
+                    running_var if running_var is not None else empty,
+                    eps, momentum, relu)
             else:
                 mean = running_mean
                 invstd = (running_var + eps).rsqrt()
@@ -266,7 +273,7 @@ class _BatchNorm2dFn(torch.autograd.Function):
             y = ((xf - mean) * invstd * gamma + beta).reshape(x.shape)
             if relu:
                 y = F.relu(y)
-        if training and running_mean is not None:
+        if training and running_mean is not None and not x.is_cuda:
             with torch.no_grad():
                 M = x.numel() // C
                 # torch stores the UNBIASED variance in running_var
