@@ -1259,32 +1259,55 @@ __global__ void maxpool2x2_bwd_gather_kernel(const unsigned short* __restrict__ 
     }
 }
 
-// Fused ReLU-mask + bias grad: dy' = dy * (y > 0), db[k] = sum dy'[., k]
-// — one pass over dy instead of {relu_bwd, zero-fill, bias_grad}.
+// Fused ReLU-mask + bias grad: dy' = dy * (y > 0), db[k] = sum dy'[., k].
+// Row-major traversal (thread = (channel, row-lane) like the BN partials:
+// consecutive threads touch consecutive channels => coalesced reads AND
+// writes), per-block LDS lane reduction, one atomic per channel per block.
 __global__ void relu_bias_bwd_kernel(const unsigned short* __restrict__ dy,
                                      const unsigned short* __restrict__ y,
                                      unsigned short* __restrict__ dym,
                                      float* __restrict__ db, int64_t M, int K,
                                      int rows_per_block) {
     __shared__ float red[256];
-    const int k = blockIdx.x;
-    const int64_t r0 = (int64_t)blockIdx.y * rows_per_block;
+    const int64_t r0 = (int64_t)blockIdx.x * rows_per_block;
     const int64_t r1 = min(r0 + rows_per_block, M);
+    if (K >= (int)blockDim.x) {
+        for (int c = threadIdx.x; c < K; c += blockDim.x) {
+            float acc = 0.f;
+            for (int64_t r = r0; r < r1; ++r) {
+                int64_t i = r * K + c;
+                unsigned short yv = y[i];
+                unsigned short g =
+                    ((yv & 0x7fffu) != 0 && !(yv & 0x8000u)) ? dy[i] : 0;
+                dym[i] = g;
+                acc += bf2f(g);
+            }
+            atomicAdd(db + c, acc);
+        }
+        return;
+    }
+    int lanes = (int)blockDim.x / K;
+    lanes = 1 << (31 - __clz(lanes));
+    const int c = threadIdx.x % K;
+    const int rl = threadIdx.x / K;
     float acc = 0.f;
-    for (int64_t r = r0 + threadIdx.x; r < r1; r += blockDim.x) {
-        int64_t i = r * K + k;
-        unsigned short yv = y[i];
-        unsigned short g = ((yv & 0x7fffu) != 0 && !(yv & 0x8000u)) ? dy[i] : 0;
-        dym[i] = g;
-        acc += bf2f(g);
+    if (rl < lanes) {
+        for (int64_t r = r0 + rl; r < r1; r += lanes) {
+            int64_t i = r * K + c;
+            unsigned short yv = y[i];
+            unsigned short g =
+                ((yv & 0x7fffu) != 0 && !(yv & 0x8000u)) ? dy[i] : 0;
+            dym[i] = g;
+            acc += bf2f(g);
+        }
     }
     red[threadIdx.x] = acc;
     __syncthreads();
-    for (int off = 128; off > 0; off >>= 1) {
-        if (threadIdx.x < off) red[threadIdx.x] += red[threadIdx.x + off];
+    for (int off = lanes >> 1; off >= 1; off >>= 1) {
+        if (rl < off) red[threadIdx.x] += red[threadIdx.x + off * K];
         __syncthreads();
     }
-    if (threadIdx.x == 0) atomicAdd(db + k, red[0]);
+    if (rl == 0) atomicAdd(db + c, red[threadIdx.x]);
 }
 
 __global__ void maxpool_bwd_kernel(const unsigned short* __restrict__ dy,
@@ -2212,9 +2235,9 @@ std::vector<torch::Tensor> relu_bias_bwd(torch::Tensor dy, torch::Tensor y) {
     const int64_t M = dyc.numel() / K;
     auto dym = torch::empty_like(dyc);
     auto db = torch::zeros({K}, dyc.options().dtype(torch::kFloat32));
-    int rpb = (int)std::max<int64_t>(2048, (M + 63) / 64);
-    dim3 grid(K, (unsigned)((M + rpb - 1) / rpb));
-    hipLaunchKernelGGL(relu_bias_bwd_kernel, grid, dim3(256), 0,
+    int rpb = (int)std::max<int64_t>(64, (M + 511) / 512);
+    int nblk = (int)((M + rpb - 1) / rpb);
+    hipLaunchKernelGGL(relu_bias_bwd_kernel, dim3(nblk), dim3(256), 0,
                        at::cuda::getCurrentCUDAStream(), bf_ptr(dyc), bf_ptr(y),
                        bf_ptr_mut(dym), db.data_ptr<float>(), M, K, rpb);
     return {dym, db};
