@@ -1611,10 +1611,13 @@ __global__ void zero_mt_kernel(const int64_t* __restrict__ meta,
 // Coalesced two-stage reduction: stage 1 blocks each cover a row-chunk x
 // all C channels (consecutive threads -> consecutive channels), partials
 // atomically added into fp32 accumulators; stage 2 finalizes mean/invstd.
+// writes per-block partial sums into slab[b][2][C] (torch::empty — every
+// cell is stored exactly once, so no zero-fill or atomics are needed)
 __global__ void bn_partial_kernel(const unsigned short* __restrict__ x,
-                                  float* __restrict__ gsum,
-                                  float* __restrict__ gsq, int64_t M, int C,
+                                  float* __restrict__ slab, int64_t M, int C,
                                   int rows_per_block) {
+    float* gsum = slab + (int64_t)blockIdx.x * 2 * C;
+    float* gsq = gsum + C;
     __shared__ float red[2][256];
     const int64_t r0 = (int64_t)blockIdx.x * rows_per_block;
     const int64_t r1 = min(r0 + rows_per_block, M);
@@ -1626,8 +1629,8 @@ __global__ void bn_partial_kernel(const unsigned short* __restrict__ x,
                 acc += v;
                 acc2 += v * v;
             }
-            atomicAdd(gsum + c, acc);
-            atomicAdd(gsq + c, acc2);
+            gsum[c] = acc;
+            gsq[c] = acc2;
         }
         return;
     }
@@ -1657,13 +1660,12 @@ __global__ void bn_partial_kernel(const unsigned short* __restrict__ x,
         __syncthreads();
     }
     if (rl == 0) {
-        atomicAdd(gsum + c, red[0][threadIdx.x]);
-        atomicAdd(gsq + c, red[1][threadIdx.x]);
+        gsum[c] = red[0][threadIdx.x];
+        gsq[c] = red[1][threadIdx.x];
     }
 }
 
-__global__ void bn_finalize_kernel(const float* __restrict__ gsum,
-                                   const float* __restrict__ gsq,
+__global__ void bn_finalize_kernel(const float* __restrict__ slab, int nblk,
                                    float* __restrict__ mean,
                                    float* __restrict__ invstd,
                                    float* __restrict__ running_mean,
@@ -1671,8 +1673,13 @@ __global__ void bn_finalize_kernel(const float* __restrict__ gsum,
                                    int C, float eps, float momentum) {
     for (int c = blockIdx.x * blockDim.x + threadIdx.x; c < C;
          c += gridDim.x * blockDim.x) {
-        float mu = gsum[c] / (float)M;
-        float var = fmaxf(gsq[c] / (float)M - mu * mu, 0.f);
+        float s1 = 0.f, s2 = 0.f;
+        for (int b = 0; b < nblk; ++b) {
+            s1 += slab[(int64_t)b * 2 * C + c];
+            s2 += slab[(int64_t)b * 2 * C + C + c];
+        }
+        float mu = s1 / (float)M;
+        float var = fmaxf(s2 / (float)M - mu * mu, 0.f);
         mean[c] = mu;
         invstd[c] = rsqrtf(var + eps);
         if (running_mean) {  // torch semantics: UNBIASED var in running_var
@@ -1705,9 +1712,10 @@ __global__ void bn_bwd_partial_kernel(const unsigned short* __restrict__ dy,
                                       const unsigned short* __restrict__ x,
                                       const float* __restrict__ mean,
                                       const float* __restrict__ invstd,
-                                      float* __restrict__ dgamma,
-                                      float* __restrict__ dbeta, int64_t M,
+                                      float* __restrict__ slab, int64_t M,
                                       int C, int rows_per_block) {
+    float* dgamma = slab + (int64_t)blockIdx.x * 2 * C;
+    float* dbeta = dgamma + C;
     __shared__ float red[2][256];
     const int64_t r0 = (int64_t)blockIdx.x * rows_per_block;
     const int64_t r1 = min(r0 + rows_per_block, M);
@@ -1720,8 +1728,8 @@ __global__ void bn_bwd_partial_kernel(const unsigned short* __restrict__ dy,
                 dg += g * (bf2f(x[r * C + c]) - mu) * is;
                 db += g;
             }
-            atomicAdd(dgamma + c, dg);
-            atomicAdd(dbeta + c, db);
+            dgamma[c] = dg;
+            dbeta[c] = db;
         }
         return;
     }
@@ -1749,8 +1757,23 @@ __global__ void bn_bwd_partial_kernel(const unsigned short* __restrict__ dy,
         __syncthreads();
     }
     if (rl == 0) {
-        atomicAdd(dgamma + c, red[0][threadIdx.x]);
-        atomicAdd(dbeta + c, red[1][threadIdx.x]);
+        dgamma[c] = red[0][threadIdx.x];
+        dbeta[c] = red[1][threadIdx.x];
+    }
+}
+
+__global__ void bn_bwd_finalize_kernel(const float* __restrict__ slab,
+                                       int nblk, float* __restrict__ dgamma,
+                                       float* __restrict__ dbeta, int C) {
+    for (int c = blockIdx.x * blockDim.x + threadIdx.x; c < C;
+         c += gridDim.x * blockDim.x) {
+        float dg = 0.f, db = 0.f;
+        for (int b = 0; b < nblk; ++b) {
+            dg += slab[(int64_t)b * 2 * C + c];
+            db += slab[(int64_t)b * 2 * C + C + c];
+        }
+        dgamma[c] = dg;
+        dbeta[c] = db;
     }
 }
 
@@ -2363,15 +2386,13 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
     auto invstd = torch::empty({C}, f32);
     auto y = torch::empty_like(x);
     auto stream = at::cuda::getCurrentCUDAStream();
-    auto gsum = torch::zeros({C}, f32);
-    auto gsq = torch::zeros({C}, f32);
     int rpb = (int)std::max<int64_t>(64, (M + 127) / 128);
     int nblk = (int)((M + rpb - 1) / rpb);
+    auto slab = torch::empty({nblk, 2, C}, f32);
     hipLaunchKernelGGL(bn_partial_kernel, dim3(nblk), dim3(256), 0, stream,
-                       bf_ptr(x), gsum.data_ptr<float>(), gsq.data_ptr<float>(),
-                       M, C, rpb);
+                       bf_ptr(x), slab.data_ptr<float>(), M, C, rpb);
     hipLaunchKernelGGL(bn_finalize_kernel, dim3(ceildiv(C, 256)), dim3(256), 0,
-                       stream, gsum.data_ptr<float>(), gsq.data_ptr<float>(),
+                       stream, slab.data_ptr<float>(), nblk,
                        mean.data_ptr<float>(), invstd.data_ptr<float>(),
                        rmean.numel() ? rmean.data_ptr<float>() : nullptr,
                        rvar.numel() ? rvar.data_ptr<float>() : nullptr, M, C,
@@ -2410,16 +2431,20 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
     const int C = (int)x.size(-1);
     const int64_t M = x.numel() / C;
     auto f32 = x.options().dtype(torch::kFloat32);
-    auto dgamma = torch::zeros({C}, f32);
-    auto dbeta = torch::zeros({C}, f32);
+    auto dgamma = torch::empty({C}, f32);
+    auto dbeta = torch::empty({C}, f32);
     auto dx = torch::empty_like(x);
     auto stream = at::cuda::getCurrentCUDAStream();
     int rpb = (int)std::max<int64_t>(64, (M + 127) / 128);
     int nblk = (int)((M + rpb - 1) / rpb);
+    auto slab = torch::empty({nblk, 2, C}, f32);
     hipLaunchKernelGGL(bn_bwd_partial_kernel, dim3(nblk), dim3(256), 0, stream,
                        bf_ptr(dyc), bf_ptr(x), mean.data_ptr<float>(),
-                       invstd.data_ptr<float>(), dgamma.data_ptr<float>(),
-                       dbeta.data_ptr<float>(), M, C, rpb);
+                       invstd.data_ptr<float>(), slab.data_ptr<float>(), M, C,
+                       rpb);
+    hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3(ceildiv(C, 256)), dim3(256),
+                       0, stream, slab.data_ptr<float>(), nblk,
+                       dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), C);
     int64_t total = x.numel();
     int blocks = std::min<int64_t>(ceildiv(total, 256), 2048);
     hipLaunchKernelGGL(bn_dx_kernel, dim3(blocks), dim3(256), 0, stream,
