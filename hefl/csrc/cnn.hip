@@ -1665,21 +1665,33 @@ __global__ void bn_partial_kernel(const unsigned short* __restrict__ x,
     }
 }
 
+// one block per channel: 256 threads tree-reduce the nblk slab rows
 __global__ void bn_finalize_kernel(const float* __restrict__ slab, int nblk,
                                    float* __restrict__ mean,
                                    float* __restrict__ invstd,
                                    float* __restrict__ running_mean,
                                    float* __restrict__ running_var, int64_t M,
                                    int C, float eps, float momentum) {
-    for (int c = blockIdx.x * blockDim.x + threadIdx.x; c < C;
-         c += gridDim.x * blockDim.x) {
-        float s1 = 0.f, s2 = 0.f;
-        for (int b = 0; b < nblk; ++b) {
-            s1 += slab[(int64_t)b * 2 * C + c];
-            s2 += slab[(int64_t)b * 2 * C + C + c];
+    __shared__ float red[2][256];
+    const int c = blockIdx.x;
+    float s1 = 0.f, s2 = 0.f;
+    for (int b = threadIdx.x; b < nblk; b += blockDim.x) {
+        s1 += slab[(int64_t)b * 2 * C + c];
+        s2 += slab[(int64_t)b * 2 * C + C + c];
+    }
+    red[0][threadIdx.x] = s1;
+    red[1][threadIdx.x] = s2;
+    __syncthreads();
+    for (int off = 128; off > 0; off >>= 1) {
+        if ((int)threadIdx.x < off) {
+            red[0][threadIdx.x] += red[0][threadIdx.x + off];
+            red[1][threadIdx.x] += red[1][threadIdx.x + off];
         }
-        float mu = s1 / (float)M;
-        float var = fmaxf(s2 / (float)M - mu * mu, 0.f);
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) {
+        float mu = red[0][0] / (float)M;
+        float var = fmaxf(red[1][0] / (float)M - mu * mu, 0.f);
         mean[c] = mu;
         invstd[c] = rsqrtf(var + eps);
         if (running_mean) {  // torch semantics: UNBIASED var in running_var
@@ -1765,15 +1777,26 @@ __global__ void bn_bwd_partial_kernel(const unsigned short* __restrict__ dy,
 __global__ void bn_bwd_finalize_kernel(const float* __restrict__ slab,
                                        int nblk, float* __restrict__ dgamma,
                                        float* __restrict__ dbeta, int C) {
-    for (int c = blockIdx.x * blockDim.x + threadIdx.x; c < C;
-         c += gridDim.x * blockDim.x) {
-        float dg = 0.f, db = 0.f;
-        for (int b = 0; b < nblk; ++b) {
-            dg += slab[(int64_t)b * 2 * C + c];
-            db += slab[(int64_t)b * 2 * C + C + c];
+    __shared__ float red[2][256];
+    const int c = blockIdx.x;
+    float dg = 0.f, db = 0.f;
+    for (int b = threadIdx.x; b < nblk; b += blockDim.x) {
+        dg += slab[(int64_t)b * 2 * C + c];
+        db += slab[(int64_t)b * 2 * C + C + c];
+    }
+    red[0][threadIdx.x] = dg;
+    red[1][threadIdx.x] = db;
+    __syncthreads();
+    for (int off = 128; off > 0; off >>= 1) {
+        if ((int)threadIdx.x < off) {
+            red[0][threadIdx.x] += red[0][threadIdx.x + off];
+            red[1][threadIdx.x] += red[1][threadIdx.x + off];
         }
-        dgamma[c] = dg;
-        dbeta[c] = db;
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) {
+        dgamma[c] = red[0][0];
+        dbeta[c] = red[1][0];
     }
 }
 
@@ -2391,7 +2414,7 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
     auto slab = torch::empty({nblk, 2, C}, f32);
     hipLaunchKernelGGL(bn_partial_kernel, dim3(nblk), dim3(256), 0, stream,
                        bf_ptr(x), slab.data_ptr<float>(), M, C, rpb);
-    hipLaunchKernelGGL(bn_finalize_kernel, dim3(ceildiv(C, 256)), dim3(256), 0,
+    hipLaunchKernelGGL(bn_finalize_kernel, dim3(C), dim3(256), 0,
                        stream, slab.data_ptr<float>(), nblk,
                        mean.data_ptr<float>(), invstd.data_ptr<float>(),
                        rmean.numel() ? rmean.data_ptr<float>() : nullptr,
@@ -2442,8 +2465,8 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
                        bf_ptr(dyc), bf_ptr(x), mean.data_ptr<float>(),
                        invstd.data_ptr<float>(), slab.data_ptr<float>(), M, C,
                        rpb);
-    hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3(ceildiv(C, 256)), dim3(256),
-                       0, stream, slab.data_ptr<float>(), nblk,
+    hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3(C), dim3(256), 0,
+                       stream, slab.data_ptr<float>(), nblk,
                        dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), C);
     int64_t total = x.numel();
     int blocks = std::min<int64_t>(ceildiv(total, 256), 2048);
