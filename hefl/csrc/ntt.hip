@@ -218,9 +218,57 @@ ntt_lds_ml_kernel(int64_t* __restrict__ x, const int64_t* __restrict__ w,
     for (int i = tid; i < nblk; i += kThreads) smem[i] = xr[i];
     __syncthreads();
     const int nb2 = nblk >> 1;
-    for (int m = n / nblk; m < n; m <<= 1) {
+    const int nb4 = nblk >> 2;
+    int m = n / nblk;
+    // radix-4 passes: two CT stages (m, 2m) per barrier — halves the LDS
+    // round trips and __syncthreads count of the radix-2 ladder
+    for (; 2 * m < n; m <<= 2) {
         const uint32_t t = (uint32_t)(n / (2 * m));
-        const int tlog = 31 - __clz(t);              // t is a power of two
+        const uint32_t t2 = t >> 1;
+        const int t2log = 31 - __clz(t2);
+        for (int lb = tid; lb < nb4; lb += kThreads) {
+            uint32_t g = (uint32_t)lb >> t2log;
+            uint32_t p = (uint32_t)lb & (t2 - 1);
+            uint32_t base = g * 2 * t + p;
+            uint32_t jm = (uint32_t)blk * ((uint32_t)nb2 >> (t2log + 1)) + g;
+            uint64_t x0 = (uint64_t)smem[base];
+            uint64_t x1 = (uint64_t)smem[base + t2];
+            uint64_t x2 = (uint64_t)smem[base + t];
+            uint64_t x3 = (uint64_t)smem[base + t + t2];
+            // stage m (twiddle w[m + jm] for both pairs)
+            {
+                uint64_t W = (uint64_t)wl[m + jm];
+                uint64_t Wsh = (uint64_t)wshl[m + jm];
+                uint64_t v = mulmod_shoup(x2, W, Wsh, q);
+                x2 = submod_u64(x0, v, q);
+                x0 = addmod_u64(x0, v, q);
+                v = mulmod_shoup(x3, W, Wsh, q);
+                x3 = submod_u64(x1, v, q);
+                x1 = addmod_u64(x1, v, q);
+            }
+            // stage 2m: groups 2jm (x0,x1) and 2jm+1 (x2,x3)
+            {
+                uint64_t W = (uint64_t)wl[2 * m + 2 * jm];
+                uint64_t Wsh = (uint64_t)wshl[2 * m + 2 * jm];
+                uint64_t v = mulmod_shoup(x1, W, Wsh, q);
+                x1 = submod_u64(x0, v, q);
+                x0 = addmod_u64(x0, v, q);
+                W = (uint64_t)wl[2 * m + 2 * jm + 1];
+                Wsh = (uint64_t)wshl[2 * m + 2 * jm + 1];
+                v = mulmod_shoup(x3, W, Wsh, q);
+                x3 = submod_u64(x2, v, q);
+                x2 = addmod_u64(x2, v, q);
+            }
+            smem[base] = (int64_t)x0;
+            smem[base + t2] = (int64_t)x1;
+            smem[base + t] = (int64_t)x2;
+            smem[base + t + t2] = (int64_t)x3;
+        }
+        __syncthreads();
+    }
+    for (; m < n; m <<= 1) {  // at most one radix-2 tail stage
+        const uint32_t t = (uint32_t)(n / (2 * m));
+        const int tlog = 31 - __clz(t);
         for (int lb = tid; lb < nb2; lb += kThreads) {
             uint32_t jloc = (uint32_t)lb >> tlog;
             uint32_t pos = (uint32_t)lb & (t - 1);
@@ -257,10 +305,59 @@ intt_lds_ml_kernel(int64_t* __restrict__ x, const int64_t* __restrict__ winv,
     for (int i = tid; i < nblk; i += kThreads) smem[i] = xr[i];
     __syncthreads();
     const int nb2 = nblk >> 1;
-    for (int m = n; m >= 2 * (n / nblk); m >>= 1) {
+    const int nb4 = nblk >> 2;
+    int m = n;
+    // radix-4 passes: two GS stages (m, m/2) per barrier — halves LDS
+    // round trips and __syncthreads vs the radix-2 ladder
+    for (; m >= 4 * (n / nblk); m >>= 2) {
         const int h = m >> 1;
+        const int h2 = m >> 2;
         const uint32_t t = (uint32_t)(n / m);
         const int tlog = 31 - __clz(t);              // t is a power of two
+        for (int lb = tid; lb < nb4; lb += kThreads) {
+            uint32_t j2 = (uint32_t)lb >> tlog;
+            uint32_t p = (uint32_t)lb & (t - 1);
+            uint32_t base = j2 * 4 * t + p;
+            uint32_t j2g = (uint32_t)blk * ((uint32_t)nb4 >> tlog) + j2;
+            uint64_t x0 = (uint64_t)smem[base];
+            uint64_t x1 = (uint64_t)smem[base + t];
+            uint64_t x2 = (uint64_t)smem[base + 2 * t];
+            uint64_t x3 = (uint64_t)smem[base + 3 * t];
+            // stage m: (x0,x1) group 2*j2g, (x2,x3) group 2*j2g+1
+            {
+                uint64_t W = (uint64_t)wl[h + 2 * j2g];
+                uint64_t Wsh = (uint64_t)wshl[h + 2 * j2g];
+                uint64_t u = addmod_u64(x0, x1, q);
+                x1 = mulmod_shoup(submod_u64(x0, x1, q), W, Wsh, q);
+                x0 = u;
+                W = (uint64_t)wl[h + 2 * j2g + 1];
+                Wsh = (uint64_t)wshl[h + 2 * j2g + 1];
+                u = addmod_u64(x2, x3, q);
+                x3 = mulmod_shoup(submod_u64(x2, x3, q), W, Wsh, q);
+                x2 = u;
+            }
+            // stage m/2: (x0,x2) and (x1,x3), both group j2g
+            {
+                uint64_t W = (uint64_t)wl[h2 + j2g];
+                uint64_t Wsh = (uint64_t)wshl[h2 + j2g];
+                uint64_t u = addmod_u64(x0, x2, q);
+                x2 = mulmod_shoup(submod_u64(x0, x2, q), W, Wsh, q);
+                x0 = u;
+                u = addmod_u64(x1, x3, q);
+                x3 = mulmod_shoup(submod_u64(x1, x3, q), W, Wsh, q);
+                x1 = u;
+            }
+            smem[base] = (int64_t)x0;
+            smem[base + t] = (int64_t)x1;
+            smem[base + 2 * t] = (int64_t)x2;
+            smem[base + 3 * t] = (int64_t)x3;
+        }
+        __syncthreads();
+    }
+    for (; m >= 2 * (n / nblk); m >>= 1) {  // at most one radix-2 tail stage
+        const int h = m >> 1;
+        const uint32_t t = (uint32_t)(n / m);
+        const int tlog = 31 - __clz(t);
         for (int lb = tid; lb < nb2; lb += kThreads) {
             uint32_t jloc = (uint32_t)lb >> tlog;
             uint32_t pos = (uint32_t)lb & (t - 1);
