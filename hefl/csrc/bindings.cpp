@@ -75,7 +75,7 @@ torch::Tensor bias_grad(torch::Tensor dy);
 std::vector<torch::Tensor> relu_bias_bwd(torch::Tensor dy, torch::Tensor y);
 void fused_adam_mt(torch::Tensor meta, torch::Tensor ptrs, torch::Tensor sizes,
                    int64_t nchunks, torch::Tensor sched, double b1, double b2,
-                   double eps);
+                   double eps, int64_t zero_g);
 void zero_grads_mt(torch::Tensor meta, torch::Tensor ptrs, torch::Tensor sizes,
                    int64_t nchunks);
 

@@ -1568,12 +1568,12 @@ __global__ void fused_adam_mt_kernel(const int64_t* __restrict__ meta,
                                      const int64_t* __restrict__ ptrs,
                                      const int64_t* __restrict__ sizes,
                                      const float* __restrict__ sched, float b1,
-                                     float b2, float eps) {
+                                     float b2, float eps, int zero_g) {
     const int c = blockIdx.x;
     const int t = (int)meta[c * 2];
     const int64_t off = meta[c * 2 + 1];
     float* p = reinterpret_cast<float*>(ptrs[t * 5 + 0]);
-    const float* g = reinterpret_cast<const float*>(ptrs[t * 5 + 1]);
+    float* g = reinterpret_cast<float*>(ptrs[t * 5 + 1]);
     float* m = reinterpret_cast<float*>(ptrs[t * 5 + 2]);
     float* v = reinterpret_cast<float*>(ptrs[t * 5 + 3]);
     // bf16 shadow of the fp32 master (what conv/linear forward reads):
@@ -1590,6 +1590,10 @@ __global__ void fused_adam_mt_kernel(const int64_t* __restrict__ meta,
         float pv = p[i] - lr * (mi / bc1) / (sqrtf(vi / bc2) + eps);
         p[i] = pv;
         if (sh) sh[i] = f2bf(pv);
+        // zero_g: consume-and-clear so the next captured step's backward
+        // accumulates into zeroed buffers (epoch-graph mode: one pointer
+        // table serves every step in the graph)
+        if (zero_g) g[i] = 0.f;
     }
 }
 
@@ -2565,13 +2569,13 @@ torch::Tensor bias_grad(torch::Tensor dy) {
 
 void fused_adam_mt(torch::Tensor meta, torch::Tensor ptrs, torch::Tensor sizes,
                    int64_t nchunks, torch::Tensor sched, double b1, double b2,
-                   double eps) {
+                   double eps, int64_t zero_g) {
     CHECK_GPU(meta);
     hipLaunchKernelGGL(fused_adam_mt_kernel, dim3((unsigned)nchunks), dim3(256),
                        0, at::cuda::getCurrentCUDAStream(),
                        meta.data_ptr<int64_t>(), ptrs.data_ptr<int64_t>(),
                        sizes.data_ptr<int64_t>(), sched.data_ptr<float>(),
-                       (float)b1, (float)b2, (float)eps);
+                       (float)b1, (float)b2, (float)eps, (int)zero_g);
 }
 
 void zero_grads_mt(torch::Tensor meta, torch::Tensor ptrs, torch::Tensor sizes,

@@ -91,12 +91,18 @@ class ClientLoader:
     def __len__(self) -> int:
         return (self.indices.numel() + self.batch_size - 1) // self.batch_size
 
-    def __iter__(self) -> Iterator[Tuple[torch.Tensor, torch.Tensor]]:
+    def epoch_order(self) -> torch.Tensor:
+        """Consume one epoch's (shuffled) sample order — used by the
+        epoch-graph path, which stages the whole epoch at once."""
         order = self.indices
         if self.shuffle:
             g = torch.Generator().manual_seed(self.seed + self._epoch)
             order = order[torch.randperm(order.numel(), generator=g)]
         self._epoch += 1
+        return order
+
+    def __iter__(self) -> Iterator[Tuple[torch.Tensor, torch.Tensor]]:
+        order = self.epoch_order()
         for i in range(0, order.numel(), self.batch_size):
             x, y = self.ds.batch(order[i:i + self.batch_size])
             if self.augment is not None:

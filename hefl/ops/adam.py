@@ -111,16 +111,19 @@ class FusedAdam:
                 sh.copy_(p.detach().to(torch.bfloat16))
 
     @torch.no_grad()
-    def step_mt(self, table):
+    def step_mt(self, table, zero_grad: bool = False):
         """Two direct kernel launches (schedule advance + one multi-tensor
         Adam over all params) against a per-graph pointer table. Runs AFTER
-        a graph replay — the replay wrote the grads the table points at."""
+        a graph replay — the replay wrote the grads the table points at.
+        With zero_grad the Adam kernel clears each grad as it consumes it
+        (epoch-graph mode: the next captured step's backward accumulates
+        into zeroed, address-stable buffers)."""
         C = hefl.load_extension()
         C.adam_prep(self._step_t, self._sched, self._hyper,
                     self.beta1, self.beta2)
         C.fused_adam_mt(table["meta"], table["ptrs"], table["sizes"],
                         table["n"], self._sched, self.beta1, self.beta2,
-                        self.eps)
+                        self.eps, 1 if zero_grad else 0)
 
     def zero_grad(self):
         for p in self.params:

@@ -204,8 +204,35 @@ def test_graphed_client_training_converges():
     first = c.local_train(epochs=1)
     for _ in range(40):
         last = c.local_train(epochs=1)
-    assert len(c._graphs) == 2, "expected full + partial batch graphs"
+    # whole-epoch graph: all 3 steps (2 full + 1 partial batch) in ONE graph
+    ent = c._ep_ent
+    assert ent["steps"] == 3 and ent["n"] == 80
     assert last.train_loss < first.train_loss * 0.5, (first.train_loss,
                                                       last.train_loss)
     w = c.get_weights()
     assert not torch.isnan(w).any()
+
+
+@pytest.mark.gpu
+def test_graphed_per_step_path_with_callbacks():
+    """The callbacks path uses per-step graphs (one per batch shape) with
+    Adam applied as direct launches post-replay; both captured graphs share
+    Adam's device-side schedule buffers (regression: a re-created buffer
+    left graph #1 reading freed memory and silently stopped learning)."""
+    from hefl.config import preset
+    from hefl.fl.callbacks import EarlyStopping
+    from hefl.fl.client import LocalClient
+
+    cfg = preset("config2")
+    cfg.fl.n_clients = 1
+    cfg.fl.samples_per_client = 80  # 2 full batches + 1 partial (16)
+    cfg.train.hip_graphs = True
+    c = LocalClient(cfg, 0, device="cuda:0")
+    cb = EarlyStopping(c.model, patience=10_000)
+    first = c.local_train(epochs=1, callbacks=[cb])
+    for _ in range(40):
+        last = c.local_train(epochs=1, callbacks=[cb])
+    assert len(c._graphs) == 2, "expected full + partial batch graphs"
+    assert last.train_loss < first.train_loss * 0.5, (first.train_loss,
+                                                      last.train_loss)
+    assert not torch.isnan(c.get_weights()).any()
