@@ -1282,7 +1282,8 @@ __global__ void relu_bias_bwd_kernel(const unsigned short* __restrict__ dy,
                 dym[i] = g;
                 acc += bf2f(g);
             }
-            atomicAdd(db + c, acc);
+            if (gridDim.x == 1) db[c] = acc;  // single block: direct store
+            else atomicAdd(db + c, acc);
         }
         return;
     }
@@ -1307,7 +1308,10 @@ __global__ void relu_bias_bwd_kernel(const unsigned short* __restrict__ dy,
         if (rl < off) red[threadIdx.x] += red[threadIdx.x + off * K];
         __syncthreads();
     }
-    if (rl == 0) atomicAdd(db + c, red[threadIdx.x]);
+    if (rl == 0) {
+        if (gridDim.x == 1) db[c] = red[threadIdx.x];
+        else atomicAdd(db + c, red[threadIdx.x]);
+    }
 }
 
 __global__ void maxpool_bwd_kernel(const unsigned short* __restrict__ dy,
@@ -1466,6 +1470,27 @@ __global__ void adam_prep_kernel(int64_t* __restrict__ step,
     }
 }
 
+__global__ void adam_prep_epoch_kernel(int64_t* __restrict__ step,
+                                       float* __restrict__ sched,
+                                       const float* __restrict__ hyper,
+                                       float b1, float b2, int S) {
+    if (threadIdx.x == 0 && blockIdx.x == 0) {
+        int64_t base = step[0];
+        float b1p = powf(b1, (float)base);
+        float b2p = powf(b2, (float)base);
+        for (int s = 0; s < S; ++s) {
+            int64_t t = base + 1 + s;
+            b1p *= b1;
+            b2p *= b2;
+            sched[s * 3 + 0] =
+                hyper[0] / (1.f + hyper[1] * (float)(t - 1));  // Keras decay
+            sched[s * 3 + 1] = 1.f - b1p;
+            sched[s * 3 + 2] = 1.f - b2p;
+        }
+        step[0] = base + S;
+    }
+}
+
 __global__ void fused_adam_sched_kernel(float* __restrict__ p,
                                         const float* __restrict__ g,
                                         float* __restrict__ m,
@@ -1568,7 +1593,9 @@ __global__ void fused_adam_mt_kernel(const int64_t* __restrict__ meta,
                                      const int64_t* __restrict__ ptrs,
                                      const int64_t* __restrict__ sizes,
                                      const float* __restrict__ sched, float b1,
-                                     float b2, float eps, int zero_g) {
+                                     float b2, float eps, int zero_g,
+                                     int sched_off) {
+    sched += 3 * sched_off;
     const int c = blockIdx.x;
     const int t = (int)meta[c * 2];
     const int64_t off = meta[c * 2 + 1];
@@ -1962,7 +1989,10 @@ __global__ void bias_grad_kernel(const unsigned short* __restrict__ dy,
         if (threadIdx.x < off) red[threadIdx.x] += red[threadIdx.x + off];
         __syncthreads();
     }
-    if (threadIdx.x == 0) atomicAdd(db + k, red[0]);
+    if (threadIdx.x == 0) {
+        if (gridDim.y == 1) db[k] = red[0];  // single chunk: direct store
+        else atomicAdd(db + k, red[0]);
+    }
 }
 
 inline int ceildiv(int64_t a, int64_t b) { return (int)((a + b - 1) / b); }
@@ -2284,9 +2314,11 @@ std::vector<torch::Tensor> relu_bias_bwd(torch::Tensor dy, torch::Tensor y) {
     const int K = (int)dyc.size(-1);
     const int64_t M = dyc.numel() / K;
     auto dym = torch::empty_like(dyc);
-    auto db = torch::zeros({K}, dyc.options().dtype(torch::kFloat32));
     int rpb = (int)std::max<int64_t>(64, (M + 511) / 512);
     int nblk = (int)((M + rpb - 1) / rpb);
+    auto db = nblk == 1
+                  ? torch::empty({K}, dyc.options().dtype(torch::kFloat32))
+                  : torch::zeros({K}, dyc.options().dtype(torch::kFloat32));
     hipLaunchKernelGGL(relu_bias_bwd_kernel, dim3(nblk), dim3(256), 0,
                        at::cuda::getCurrentCUDAStream(), bf_ptr(dyc), bf_ptr(y),
                        bf_ptr_mut(dym), db.data_ptr<float>(), M, K, rpb);
@@ -2558,9 +2590,11 @@ torch::Tensor bias_grad(torch::Tensor dy) {
     auto dyc = dy.contiguous();
     const int K = (int)dyc.size(-1);
     const int64_t M = dyc.numel() / K;
-    auto db = torch::zeros({K}, dyc.options().dtype(torch::kFloat32));
     int rpb = (int)std::max<int64_t>(2048, (M + 63) / 64);
     dim3 grid(K, (unsigned)((M + rpb - 1) / rpb));
+    auto db = grid.y == 1
+                  ? torch::empty({K}, dyc.options().dtype(torch::kFloat32))
+                  : torch::zeros({K}, dyc.options().dtype(torch::kFloat32));
     hipLaunchKernelGGL(bias_grad_kernel, grid, dim3(256), 0,
                        at::cuda::getCurrentCUDAStream(), bf_ptr(dyc),
                        db.data_ptr<float>(), M, K, rpb);
@@ -2569,13 +2603,23 @@ torch::Tensor bias_grad(torch::Tensor dy) {
 
 void fused_adam_mt(torch::Tensor meta, torch::Tensor ptrs, torch::Tensor sizes,
                    int64_t nchunks, torch::Tensor sched, double b1, double b2,
-                   double eps, int64_t zero_g) {
+                   double eps, int64_t zero_g, int64_t sched_off) {
     CHECK_GPU(meta);
     hipLaunchKernelGGL(fused_adam_mt_kernel, dim3((unsigned)nchunks), dim3(256),
                        0, at::cuda::getCurrentCUDAStream(),
                        meta.data_ptr<int64_t>(), ptrs.data_ptr<int64_t>(),
                        sizes.data_ptr<int64_t>(), sched.data_ptr<float>(),
-                       (float)b1, (float)b2, (float)eps, (int)zero_g);
+                       (float)b1, (float)b2, (float)eps, (int)zero_g,
+                       (int)sched_off);
+}
+
+void adam_prep_epoch(torch::Tensor step, torch::Tensor sched,
+                     torch::Tensor hyper, double b1, double b2, int64_t S) {
+    CHECK_GPU(step);
+    hipLaunchKernelGGL(adam_prep_epoch_kernel, dim3(1), dim3(64), 0,
+                       at::cuda::getCurrentCUDAStream(),
+                       step.data_ptr<int64_t>(), sched.data_ptr<float>(),
+                       hyper.data_ptr<float>(), (float)b1, (float)b2, (int)S);
 }
 
 void zero_grads_mt(torch::Tensor meta, torch::Tensor ptrs, torch::Tensor sizes,

@@ -176,22 +176,31 @@ class LocalClient:
                                             device=X.device)
             self._one = torch.ones((), dtype=torch.float32, device=X.device)
         self.opt.ensure_shadows()
-        # grads stay MATERIALIZED (warmup's) and are zeroed in place: every
-        # captured backward accumulates into these fixed buffers and the
-        # captured Adam clears them, so the same table works for all steps
-        self.opt.zero_grad_()
-        mt = self.opt.build_mt_table()
+        # Per-step pointer tables: each captured backward STEALS fresh
+        # pooled grad tensors (no zero/accumulate kernels at all); the
+        # tables the captured Adam kernels read are filled AFTER capture
+        # with the addresses recorded during it (capture-pool addresses
+        # are replay-stable). One schedule-prep kernel serves the epoch.
+        steps = (n + B - 1) // B
+        shells = [self.opt.alloc_mt_shell() for _ in range(steps)]
+        rows_per_step = []
+        grads_hold = []
         g = torch.cuda.CUDAGraph()
         with torch.cuda.graph(g, stream=side):
-            for i in range(0, n, B):
+            self.opt.prep_epoch(steps)
+            for s, i in enumerate(range(0, n, B)):
+                self.opt.zero_grad()  # grads=None -> backward steals
                 logits = self.model(X[i:i + B])
                 loss = softmax_xent(logits, Y[i:i + B], self._acc_loss,
                                     self._acc_correct)
                 loss.backward(gradient=self._one)
-                self.opt.step_mt(mt, zero_grad=True)
-        ent = {"graph": g, "X": X, "Y": Y, "mt": mt,
-               "grads": [p.grad for p in self.opt.params],
-               "steps": (n + B - 1) // B, "n": n}
+                rows_per_step.append(self.opt.current_ptr_rows())
+                self.opt.step_mt_at(shells[s], s)
+                grads_hold.append([p.grad for p in self.opt.params])
+        for shell, rows in zip(shells, rows_per_step):
+            self.opt.fill_mt_shell(shell, rows)
+        ent = {"graph": g, "X": X, "Y": Y, "mt": shells,
+               "grads": grads_hold, "steps": steps, "n": n}
         self._ep_ent = ent
         return ent
 

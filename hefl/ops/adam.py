@@ -115,15 +115,71 @@ class FusedAdam:
         """Two direct kernel launches (schedule advance + one multi-tensor
         Adam over all params) against a per-graph pointer table. Runs AFTER
         a graph replay — the replay wrote the grads the table points at.
-        With zero_grad the Adam kernel clears each grad as it consumes it
-        (epoch-graph mode: the next captured step's backward accumulates
-        into zeroed, address-stable buffers)."""
+        With zero_grad the Adam kernel clears each grad as it consumes it."""
         C = hefl.load_extension()
         C.adam_prep(self._step_t, self._sched, self._hyper,
                     self.beta1, self.beta2)
         C.fused_adam_mt(table["meta"], table["ptrs"], table["sizes"],
                         table["n"], self._sched, self.beta1, self.beta2,
-                        self.eps, 1 if zero_grad else 0)
+                        self.eps, 1 if zero_grad else 0, 0)
+
+    # ----- epoch-graph mode: S captured steps share ONE schedule-prep
+    # kernel (S rows of (lr, bc1, bc2)) and get per-step pointer tables
+    # whose contents are filled AFTER capture (the captured backward steals
+    # fresh pooled grad tensors per step; addresses are replay-stable). -----
+    def prep_epoch(self, steps: int):
+        """One kernel: schedule rows for `steps` captured steps; advances
+        the device step counter by `steps` per replay."""
+        if not hasattr(self, "_sched_ep") or self._sched_ep.numel() < 3 * steps:
+            self._sched_ep = torch.zeros(3 * steps, dtype=torch.float32,
+                                         device=self.params[0].device)
+        hefl.load_extension().adam_prep_epoch(
+            self._step_t, self._sched_ep, self._hyper, self.beta1, self.beta2,
+            steps)
+
+    def _mt_shared(self):
+        if not hasattr(self, "_mt_meta"):
+            dev = self.params[0].device
+            sizes, meta = [], []
+            for t, p in enumerate(self.params):
+                sizes.append(p.numel())
+                for off in range(0, p.numel(), self._MT_CHUNK):
+                    meta.append([t, off])
+            self._mt_meta = torch.tensor(meta, dtype=torch.int64, device=dev)
+            self._mt_sizes = torch.tensor(sizes, dtype=torch.int64, device=dev)
+        return self._mt_meta, self._mt_sizes
+
+    def alloc_mt_shell(self):
+        """Empty per-step pointer table; captured kernels reference its
+        (stable) storage, contents are written post-capture."""
+        meta, sizes = self._mt_shared()
+        ptrs = torch.empty(len(self.params), 5, dtype=torch.int64,
+                           device=self.params[0].device)
+        return {"ptrs": ptrs, "meta": meta, "sizes": sizes,
+                "n": meta.shape[0]}
+
+    def current_ptr_rows(self):
+        """Snapshot {p, grad, m, v, shadow} device addresses (called during
+        capture, right after a step's backward stole its grad tensors)."""
+        self.ensure_shadows()
+        rows = []
+        for p, m, v in zip(self.params, self.m, self.v):
+            sh = getattr(p, "_bf16", None)
+            rows.append([p.data.data_ptr(), p.grad.data_ptr(), m.data_ptr(),
+                         v.data_ptr(), sh.data_ptr() if sh is not None else 0])
+        return rows
+
+    def fill_mt_shell(self, shell, rows):
+        shell["ptrs"].copy_(torch.tensor(rows, dtype=torch.int64))
+
+    @torch.no_grad()
+    def step_mt_at(self, table, sched_off: int):
+        """Adam-only launch (no prep) against the epoch schedule row
+        `sched_off` — captured once per step inside the epoch graph."""
+        C = hefl.load_extension()
+        C.fused_adam_mt(table["meta"], table["ptrs"], table["sizes"],
+                        table["n"], self._sched_ep, self.beta1, self.beta2,
+                        self.eps, 0, sched_off)
 
     def zero_grad(self):
         for p in self.params:
