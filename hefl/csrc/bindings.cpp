@@ -73,6 +73,9 @@ torch::Tensor avgpool_global_bwd(torch::Tensor dy, int64_t H, int64_t W);
 torch::Tensor add_relu(torch::Tensor a, torch::Tensor b);
 torch::Tensor bias_grad(torch::Tensor dy);
 std::vector<torch::Tensor> relu_bias_bwd(torch::Tensor dy, torch::Tensor y);
+std::vector<torch::Tensor> pool_relu_bias_bwd(torch::Tensor dy,
+                                              torch::Tensor idx,
+                                              torch::Tensor y);
 void adam_prep_epoch(torch::Tensor step, torch::Tensor sched,
                      torch::Tensor hyper, double b1, double b2, int64_t S);
 void fused_adam_mt(torch::Tensor meta, torch::Tensor ptrs, torch::Tensor sizes,
@@ -121,6 +124,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("add_relu", &add_relu);
     m.def("bias_grad", &bias_grad);
     m.def("relu_bias_bwd", &relu_bias_bwd);
+    m.def("pool_relu_bias_bwd", &pool_relu_bias_bwd);
     m.def("fused_adam_mt", &fused_adam_mt);
     m.def("adam_prep_epoch", &adam_prep_epoch);
     m.def("zero_grads_mt", &zero_grads_mt);

@@ -1314,6 +1314,76 @@ __global__ void relu_bias_bwd_kernel(const unsigned short* __restrict__ dy,
     }
 }
 
+// Fused maxpool2x2-backward + ReLU-mask + bias grad for the conv->relu->pool
+// trunk pattern (every _SeqCNN trunk layer): one pass produces
+// dym[i] = (pool-argmax hit ? dy_pool[o] : 0) gated by (y_conv[i] > 0) and
+// db[c] = sum_i dym[i,c] — replacing maxpool2x2_bwd_gather + relu_bias_bwd
+// (two full passes over the conv-activation gradient).
+__device__ inline unsigned short pool_relu_gate(
+    const unsigned short* __restrict__ dy, const uint8_t* __restrict__ idx,
+    const unsigned short* __restrict__ y, int64_t r, int c, int K, int H,
+    int W, int OH, int OW) {
+    int64_t t = r;
+    const int iw = (int)(t % W);
+    t /= W;
+    const int ih = (int)(t % H);
+    const int n = (int)(t / H);
+    const int oh = ih >> 1, ow = iw >> 1;
+    if (oh >= OH || ow >= OW) return 0;
+    const int64_t o = (((int64_t)n * OH + oh) * OW + ow) * K + c;
+    const int d = idx[o];
+    if ((d >> 1) != (ih & 1) || (d & 1) != (iw & 1)) return 0;
+    const unsigned short yv = y[r * K + c];
+    return ((yv & 0x7fffu) != 0 && !(yv & 0x8000u)) ? dy[o] : (unsigned short)0;
+}
+
+__global__ void pool_relu_bias_bwd_kernel(
+    const unsigned short* __restrict__ dy, const uint8_t* __restrict__ idx,
+    const unsigned short* __restrict__ y, unsigned short* __restrict__ dym,
+    float* __restrict__ db, int64_t M, int K, int rows_per_block, int H,
+    int W, int OH, int OW) {
+    __shared__ float red[256];
+    const int64_t r0 = (int64_t)blockIdx.x * rows_per_block;
+    const int64_t r1 = min(r0 + rows_per_block, M);
+    if (K >= (int)blockDim.x) {  // one channel per thread, strided
+        for (int c = threadIdx.x; c < K; c += blockDim.x) {
+            float acc = 0.f;
+            for (int64_t r = r0; r < r1; ++r) {
+                unsigned short g =
+                    pool_relu_gate(dy, idx, y, r, c, K, H, W, OH, OW);
+                dym[r * K + c] = g;
+                acc += bf2f(g);
+            }
+            if (gridDim.x == 1) db[c] = acc;
+            else atomicAdd(db + c, acc);
+        }
+        return;
+    }
+    int lanes = (int)blockDim.x / K;
+    lanes = 1 << (31 - __clz(lanes));
+    const int c = threadIdx.x % K;
+    const int rl = threadIdx.x / K;
+    float acc = 0.f;
+    if (rl < lanes) {
+        for (int64_t r = r0 + rl; r < r1; r += lanes) {
+            unsigned short g =
+                pool_relu_gate(dy, idx, y, r, c, K, H, W, OH, OW);
+            dym[r * K + c] = g;
+            acc += bf2f(g);
+        }
+    }
+    red[threadIdx.x] = acc;
+    __syncthreads();
+    for (int off = lanes >> 1; off >= 1; off >>= 1) {
+        if (rl < off) red[threadIdx.x] += red[threadIdx.x + off * K];
+        __syncthreads();
+    }
+    if (rl == 0) {
+        if (gridDim.x == 1) db[c] = red[threadIdx.x];
+        else atomicAdd(db + c, red[threadIdx.x]);
+    }
+}
+
 __global__ void maxpool_bwd_kernel(const unsigned short* __restrict__ dy,
                                    const uint8_t* __restrict__ idx,
                                    unsigned short* __restrict__ dx, int N,
@@ -1516,10 +1586,14 @@ __global__ void fused_adam_sched_kernel(float* __restrict__ p,
 // Replaces the LDS-staged path that was latency-bound at 4 workgroups.
 // ---------------------------------------------------------------------------
 
+template <int DIRECT>  // DIRECT: one slab -> write bf16 y with bias/relu
 __global__ void linear_splitk_kernel(const unsigned short* __restrict__ x,
                                      const unsigned short* __restrict__ w,
                                      float* __restrict__ y32, int M, int N,
-                                     int K, int kc_len) {
+                                     int K, int kc_len,
+                                     const float* __restrict__ bias,
+                                     unsigned short* __restrict__ yout,
+                                     int relu) {
     const int lane = threadIdx.x & 63;
     const int m0 = blockIdx.x * 16;
     const int n0 = blockIdx.y * 16;
@@ -1553,6 +1627,19 @@ __global__ void linear_splitk_kernel(const unsigned short* __restrict__ x,
                                         : (bf16_t)0.f;
         }
         acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+    }
+    if (DIRECT) {  // single slab: fused bias + relu + bf16 store, no epilogue
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            int orow = m0 + (lane >> 4) * 4 + r;
+            int ocol = n0 + (lane & 15);
+            if (orow < M && ocol < N) {
+                float v = acc[r] + (bias ? bias[ocol] : 0.f);
+                if (relu) v = v > 0.f ? v : 0.f;
+                yout[(int64_t)orow * N + ocol] = f2bf(v);
+            }
+        }
+        return;
     }
     // slab write [kc][M][N]: no zero-init, no atomics; epilogue sums slabs
     float* slab = y32 + (int64_t)blockIdx.z * M * N;
@@ -2232,12 +2319,19 @@ torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
         int kc = std::max(256, K / 4);
         kc = ((kc + 31) / 32) * 32;
         const int slabs = ceildiv(K, kc);
+        if (slabs == 1) {  // whole K in one pass: direct bf16 write
+            dim3 grid(ceildiv(M, 16), ceildiv(N, 16), 1);
+            hipLaunchKernelGGL((linear_splitk_kernel<1>), grid, dim3(64, 1, 1),
+                               0, stream, bf_ptr(x), bf_ptr(w), nullptr, M, N,
+                               K, kc, bias, bf_ptr_mut(y), relu ? 1 : 0);
+            return y;
+        }
         auto y32 = torch::empty({slabs, M, N},
                                 x.options().dtype(torch::kFloat32));
         dim3 grid(ceildiv(M, 16), ceildiv(N, 16), slabs);
-        hipLaunchKernelGGL(linear_splitk_kernel, grid, dim3(64), 0, stream,
+        hipLaunchKernelGGL((linear_splitk_kernel<0>), grid, dim3(64), 0, stream,
                            bf_ptr(x), bf_ptr(w), y32.data_ptr<float>(), M, N,
-                           K, kc);
+                           K, kc, nullptr, nullptr, 0);
         int64_t total = (int64_t)M * N;
         hipLaunchKernelGGL(linear_epilogue_kernel,
                            dim3((int)std::min<int64_t>(ceildiv(total, 256), 2048)),
@@ -2322,6 +2416,28 @@ std::vector<torch::Tensor> relu_bias_bwd(torch::Tensor dy, torch::Tensor y) {
     hipLaunchKernelGGL(relu_bias_bwd_kernel, dim3(nblk), dim3(256), 0,
                        at::cuda::getCurrentCUDAStream(), bf_ptr(dyc), bf_ptr(y),
                        bf_ptr_mut(dym), db.data_ptr<float>(), M, K, rpb);
+    return {dym, db};
+}
+
+std::vector<torch::Tensor> pool_relu_bias_bwd(torch::Tensor dy,
+                                              torch::Tensor idx,
+                                              torch::Tensor y) {
+    CHECK_GPU(dy);
+    auto dyc = dy.contiguous();
+    const int N = (int)y.size(0), H = (int)y.size(1), W = (int)y.size(2),
+              K = (int)y.size(3);
+    const int OH = (int)dyc.size(1), OW = (int)dyc.size(2);
+    const int64_t M = (int64_t)N * H * W;
+    auto dym = torch::empty_like(y);
+    int rpb = (int)std::max<int64_t>(64, (M + 511) / 512);
+    int nblk = (int)((M + rpb - 1) / rpb);
+    auto db = nblk == 1
+                  ? torch::empty({K}, dyc.options().dtype(torch::kFloat32))
+                  : torch::zeros({K}, dyc.options().dtype(torch::kFloat32));
+    hipLaunchKernelGGL(pool_relu_bias_bwd_kernel, dim3(nblk), dim3(256), 0,
+                       at::cuda::getCurrentCUDAStream(), bf_ptr(dyc),
+                       idx.data_ptr<uint8_t>(), bf_ptr(y), bf_ptr_mut(dym),
+                       db.data_ptr<float>(), M, K, rpb, H, W, OH, OW);
     return {dym, db};
 }
 

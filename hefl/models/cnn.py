@@ -17,7 +17,7 @@ import torch
 import torch.nn as nn
 
 from ..config import ModelConfig
-from ..ops.modules import Conv2dValid, Dense, Flatten, MaxPool2x2
+from ..ops.modules import ConvPool, Dense, Flatten
 
 
 class _SeqCNN(nn.Module):
@@ -31,10 +31,9 @@ class _SeqCNN(nn.Module):
         layers = []
         cin = C
         for f in conv_filters:
-            conv = Conv2dValid(cin, f, k=conv_k, relu=True, gen=gen)
-            layers += [conv, MaxPool2x2()]
-            H, W = conv.out_hw(H, W)
-            H, W = H // 2, W // 2
+            block = ConvPool(cin, f, k=conv_k, gen=gen)
+            layers.append(block)  # fused conv+relu+pool trunk block
+            H, W = block.out_hw(H, W)
             cin = f
         self.trunk = nn.ModuleList(layers)
         self.flatten = Flatten()

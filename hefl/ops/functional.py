@@ -138,6 +138,43 @@ def maxpool2x2(x):
     return _MaxPool2x2Fn.apply(x)
 
 
+class _ConvReluPoolFn(torch.autograd.Function):
+    """Fused conv(+bias+ReLU) -> maxpool2x2 trunk block (GPU only).
+
+    Forward launches the same two kernels as the composed ops; the win is in
+    backward: ONE pool_relu_bias_bwd kernel produces the gated conv-activation
+    gradient AND the bias gradient (replacing maxpool2x2_bwd + relu_bias_bwd,
+    i.e. two full passes over the activation-gradient tensor)."""
+
+    @staticmethod
+    def forward(ctx, x, w, b, pad: int):
+        ctx.pad = pad
+        wb = _gpu_dtype(w)
+        bb = b.detach().float()
+        y = _C().conv2d_fwd(x.contiguous(), wb.contiguous(), bb, 1, True, pad)
+        p, idx = _C().maxpool2x2_fwd(y)
+        ctx.save_for_backward(x, wb, y, idx)
+        return p
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w, y, idx = ctx.saved_tensors
+        dym, db = _C().pool_relu_bias_bwd(dy.contiguous(), idx, y)
+        dx = _C().conv2d_dgrad(dym, w, 1, x.shape[1], x.shape[2], ctx.pad) \
+            if ctx.needs_input_grad[0] else None
+        dw = _C().conv2d_wgrad(dym, x.contiguous(), 1, w.shape[1], w.shape[2],
+                               ctx.pad)
+        return dx, dw, db, None
+
+
+def conv_relu_pool(x, w, b, pad: int = 0):
+    """conv3x3(+bias,+relu) then maxpool2x2. GPU: fused-backward Function;
+    CPU: composed reference ops (same autograd semantics)."""
+    if x.is_cuda:
+        return _ConvReluPoolFn.apply(x, w, b, pad)
+    return maxpool2x2(conv2d(x, w, b, 1, True, pad))
+
+
 # ---------------------------------------------------------------------------
 # Linear (Dense): y = x @ w^T + b, optional fused ReLU.
 # ---------------------------------------------------------------------------

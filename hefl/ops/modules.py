@@ -47,6 +47,23 @@ class MaxPool2x2(nn.Module):
         return Fx.maxpool2x2(x)
 
 
+class ConvPool(Conv2dValid):
+    """Conv2dValid(relu=True) + MaxPool2x2 as one trunk block with a fused
+    backward on GPU (one kernel for pool-bwd + ReLU mask + bias grad).
+    Same parameters/init stream as Conv2dValid."""
+
+    def __init__(self, cin: int, cout: int, k: int = 3, gen=None):
+        super().__init__(cin, cout, k=k, stride=1, relu=True, bias=True,
+                         pad=0, gen=gen)
+
+    def out_hw(self, h: int, w: int):
+        oh, ow = super().out_hw(h, w)
+        return oh // 2, ow // 2
+
+    def forward(self, x):
+        return Fx.conv_relu_pool(x, self.weight, self.bias, self.pad)
+
+
 class Flatten(nn.Module):
     def forward(self, x):
         return x.reshape(x.shape[0], -1)

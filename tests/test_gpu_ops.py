@@ -236,3 +236,41 @@ def test_graphed_per_step_path_with_callbacks():
     assert last.train_loss < first.train_loss * 0.5, (first.train_loss,
                                                       last.train_loss)
     assert not torch.isnan(c.get_weights()).any()
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("shape", [
+    (4, 28, 28, 1, 16, 3),   # cnn2 block1
+    (4, 13, 13, 16, 32, 3),  # cnn2 block2 (odd H/W: floor pooling)
+    (2, 32, 32, 3, 6, 5),    # lenet5 block1
+])
+def test_conv_relu_pool_fused_matches_composed(shape):
+    """Fused trunk block (one pool+relu+bias backward kernel) against the
+    COMPOSED GPU ops on identical bf16 inputs: same forward kernels run in
+    both, so relu-gate and pool-argmax decisions are identical and the only
+    difference is the fused backward — grads must match near-exactly.
+    (CPU-vs-GPU parity is covered by the per-op tests + training tests;
+    comparing this block against fp32 CPU would re-test bf16 tie flips.)"""
+    N, H, W, C, K, ks = shape
+    torch.manual_seed(0)
+    x = torch.randn(N, H, W, C, device="cuda", dtype=torch.bfloat16)
+    w = (torch.randn(K, ks, ks, C, device="cuda") * 0.1)
+    b = torch.randn(K, device="cuda") * 0.05
+
+    def run(fused):
+        xd = x.clone().requires_grad_(True)
+        wd = torch.nn.Parameter(w.clone())
+        bd = torch.nn.Parameter(b.clone())
+        if fused:
+            y = Fx.conv_relu_pool(xd, wd, bd)
+        else:
+            y = Fx.maxpool2x2(Fx.conv2d(xd, wd, bd, 1, True, 0))
+        y.float().pow(2).sum().backward()
+        return y, xd.grad, wd.grad, bd.grad
+
+    yf, dxf, dwf, dbf = run(True)
+    yc, dxc, dwc, dbc = run(False)
+    _close(yf, yc, rel=1e-5)
+    _close(dxf, dxc, rel=1e-3)
+    _close(dwf, dwc, rel=1e-3)
+    _close(dbf, dbc, rel=1e-3)
