@@ -35,6 +35,9 @@ torch::Tensor conv2d_wgrad(torch::Tensor dy, torch::Tensor x, int64_t stride,
 torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
                          bool relu);
 torch::Tensor linear_dgrad(torch::Tensor dy, torch::Tensor w);
+std::vector<torch::Tensor> dense_bwd_small(torch::Tensor dy, torch::Tensor y,
+                                           torch::Tensor x, torch::Tensor w,
+                                           bool need_dx);
 torch::Tensor linear_wgrad(torch::Tensor dy, torch::Tensor x);
 std::vector<torch::Tensor> maxpool2x2_fwd(torch::Tensor x);
 torch::Tensor maxpool2x2_bwd(torch::Tensor dy, torch::Tensor idx, int64_t H,
@@ -105,6 +108,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("conv2d_wgrad", &conv2d_wgrad);
     m.def("linear_fwd", &linear_fwd);
     m.def("linear_dgrad", &linear_dgrad);
+    m.def("dense_bwd_small", &dense_bwd_small);
     m.def("linear_wgrad", &linear_wgrad);
     m.def("maxpool2x2_fwd", &maxpool2x2_fwd);
     m.def("maxpool2x2_bwd", &maxpool2x2_bwd);

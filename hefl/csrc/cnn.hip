@@ -2318,6 +2318,7 @@ torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
     if (M <= 64) {
         int kc = std::max(256, K / 4);
         kc = ((kc + 31) / 32) * 32;
+        if (K <= 1024) kc = ((K + 31) / 32) * 32;  // one slab: direct write
         const int slabs = ceildiv(K, kc);
         if (slabs == 1) {  // whole K in one pass: direct bf16 write
             dim3 grid(ceildiv(M, 16), ceildiv(N, 16), 1);
@@ -2344,6 +2345,70 @@ torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
                        0, stream, bf_ptr(x), bf_ptr(w),
                        bf_ptr_mut(y), bias, M, N, K, relu ? 1 : 0);
     return y;
+}
+
+// Whole backward of a small Dense layer in ONE kernel: applies the ReLU
+// mask (optional), and produces dx[M,K] = g @ w, dw[N,K] = g^T @ x and
+// db[N] = colsum(g) — replacing relu_bias_bwd/bias_grad + two GEMM launches.
+// Dot lengths are M or N (<= 64/256): plain VALU FMAs beat MFMA setup at
+// these sizes, and the batch-32 FL configs are launch-latency bound.
+__global__ void dense_bwd_small_kernel(
+    const unsigned short* __restrict__ dy, const unsigned short* __restrict__ y,
+    const unsigned short* __restrict__ x, const unsigned short* __restrict__ w,
+    unsigned short* __restrict__ dx, float* __restrict__ dw,
+    float* __restrict__ db, int M, int N, int K, int64_t ndx) {
+    const int64_t total = ndx + (int64_t)N * K;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        if (i < ndx) {  // dx[m,k] = sum_n g[m,n] * w[n,k]
+            const int m = (int)(i / K), k = (int)(i % K);
+            float acc = 0.f;
+            for (int n = 0; n < N; ++n) {
+                float g = bf2f(dy[m * N + n]);
+                if (y) {
+                    unsigned short yv = y[m * N + n];
+                    if ((yv & 0x7fffu) == 0 || (yv & 0x8000u)) g = 0.f;
+                }
+                acc += g * bf2f(w[(int64_t)n * K + k]);
+            }
+            dx[i] = f2bf(acc);
+        } else {  // dw[n,k] = sum_m g[m,n] * x[m,k]; db from the k==0 thread
+            const int64_t j = i - ndx;
+            const int n = (int)(j / K), k = (int)(j % K);
+            float acc = 0.f, accb = 0.f;
+            for (int m = 0; m < M; ++m) {
+                float g = bf2f(dy[m * N + n]);
+                if (y) {
+                    unsigned short yv = y[m * N + n];
+                    if ((yv & 0x7fffu) == 0 || (yv & 0x8000u)) g = 0.f;
+                }
+                acc += g * bf2f(x[(int64_t)m * K + k]);
+                accb += g;
+            }
+            dw[j] = acc;
+            if (k == 0) db[n] = accb;
+        }
+    }
+}
+
+std::vector<torch::Tensor> dense_bwd_small(torch::Tensor dy, torch::Tensor y,
+                                           torch::Tensor x, torch::Tensor w,
+                                           bool need_dx) {
+    CHECK_GPU(dy);
+    auto dyc = dy.contiguous();
+    const int M = (int)dyc.size(0), N = (int)dyc.size(1), K = (int)w.size(1);
+    auto dx = torch::empty({need_dx ? M : 0, K}, dyc.options());
+    auto dw = torch::empty({N, K}, dyc.options().dtype(torch::kFloat32));
+    auto db = torch::empty({N}, dyc.options().dtype(torch::kFloat32));
+    const int64_t ndx = need_dx ? (int64_t)M * K : 0;
+    const int64_t total = ndx + (int64_t)N * K;
+    int blocks = (int)std::min<int64_t>((total + 255) / 256, 2048);
+    hipLaunchKernelGGL(dense_bwd_small_kernel, dim3(blocks), dim3(256), 0,
+                       at::cuda::getCurrentCUDAStream(), bf_ptr(dyc),
+                       y.numel() ? bf_ptr(y) : nullptr, bf_ptr(x), bf_ptr(w),
+                       bf_ptr_mut(dx), dw.data_ptr<float>(),
+                       db.data_ptr<float>(), M, N, K, ndx);
+    return {dx, dw, db};
 }
 
 torch::Tensor linear_dgrad(torch::Tensor dy, torch::Tensor w) {
