@@ -35,11 +35,10 @@ torch::Tensor conv2d_wgrad(torch::Tensor dy, torch::Tensor x, int64_t stride,
 torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
                          bool relu);
 torch::Tensor linear_dgrad(torch::Tensor dy, torch::Tensor w);
-std::vector<torch::Tensor> dense_bwd_small(torch::Tensor dy, torch::Tensor y,
-                                           torch::Tensor x, torch::Tensor w,
-                                           bool need_dx);
 torch::Tensor linear_wgrad(torch::Tensor dy, torch::Tensor x);
 std::vector<torch::Tensor> maxpool2x2_fwd(torch::Tensor x);
+torch::Tensor synth_batch(torch::Tensor templates, torch::Tensor labels,
+                          int64_t seed);
 torch::Tensor maxpool2x2_bwd(torch::Tensor dy, torch::Tensor idx, int64_t H,
                              int64_t W);
 std::vector<torch::Tensor> softmax_xent_fwd(torch::Tensor logits,
@@ -108,9 +107,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("conv2d_wgrad", &conv2d_wgrad);
     m.def("linear_fwd", &linear_fwd);
     m.def("linear_dgrad", &linear_dgrad);
-    m.def("dense_bwd_small", &dense_bwd_small);
     m.def("linear_wgrad", &linear_wgrad);
     m.def("maxpool2x2_fwd", &maxpool2x2_fwd);
+    m.def("synth_batch", &synth_batch);
     m.def("maxpool2x2_bwd", &maxpool2x2_bwd);
     m.def("softmax_xent_fwd", &softmax_xent_fwd);
     m.def("softmax_xent_bwd", &softmax_xent_bwd);

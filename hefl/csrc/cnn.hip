@@ -25,6 +25,7 @@ namespace {
 using bf16_t = __bf16;
 typedef bf16_t bf16x8 __attribute__((ext_vector_type(8)));
 typedef float f32x4 __attribute__((ext_vector_type(4)));
+typedef unsigned short u16x8 __attribute__((ext_vector_type(8)));
 
 __device__ __forceinline__ float bf2f(unsigned short u) {
     union { unsigned int i; float f; } v;
@@ -1205,6 +1206,34 @@ conv_wgrad_small_kernel(const unsigned short* __restrict__ dy,
 // MaxPool 2x2 stride 2 (NHWC bf16), argmax corner saved for backward.
 // ---------------------------------------------------------------------------
 
+// Synthetic data generation fused to one kernel: class-template gather +
+// counter-hash Box-Muller Gaussian noise + sigmoid, bf16 store.
+// Replaces the staging chain randn/index/mul/add/sigmoid/cast (~6 fp32
+// passes over the epoch tensor) with a single bandwidth-bound pass
+// (hefl/data/synthetic.py batch()).
+__global__ void synth_batch_kernel(const float* __restrict__ T,
+                                   const int64_t* __restrict__ lab,
+                                   unsigned short* __restrict__ out,
+                                   int64_t per_img, int64_t total,
+                                   unsigned long long seed) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        const int64_t n = i / per_img;
+        const int64_t off = i - n * per_img;
+        const float t = T[lab[n] * per_img + off];
+        // splitmix64 counter hash -> two 32-bit uniforms -> Box-Muller
+        unsigned long long z = seed + (unsigned long long)i * 0x9E3779B97F4A7C15ull;
+        z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
+        z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
+        z ^= z >> 31;
+        const float u1 = ((unsigned int)z + 1.0f) * 2.3283064e-10f;  // (0,1]
+        const float u2 = (unsigned int)(z >> 32) * 2.3283064e-10f;
+        const float nrm = sqrtf(-2.f * __logf(u1)) * __cosf(6.2831853f * u2);
+        const float v = 0.6f * t + 0.4f * nrm;
+        out[i] = f2bf(1.f / (1.f + __expf(-v)));  // sigmoid -> [0,1]
+    }
+}
+
 __global__ void maxpool_fwd_kernel(const unsigned short* __restrict__ x,
                                    unsigned short* __restrict__ y,
                                    uint8_t* __restrict__ idx, int N, int H,
@@ -1739,6 +1768,76 @@ __global__ void bn_partial_kernel(const unsigned short* __restrict__ x,
     __shared__ float red[2][256];
     const int64_t r0 = (int64_t)blockIdx.x * rows_per_block;
     const int64_t r1 = min(r0 + rows_per_block, M);
+    if ((C & 7) == 0 && C <= 2048) {
+        // vectorized: thread = (channel octet, row-lane), 16-byte loads,
+        // 8 per-channel partials in registers (scalar 2-byte loads left
+        // this kernel ~10x off the HBM roofline)
+        __shared__ float red8[2][2048];
+        const int noct = C >> 3;
+        int lanes = (int)blockDim.x / noct;
+        lanes = lanes ? (1 << (31 - __clz(lanes))) : 0;
+        const int oct = threadIdx.x % noct;
+        const int rl = threadIdx.x / noct;
+        float acc[8] = {0.f}, acc2[8] = {0.f};
+        if (lanes == 0) {  // C > 8*blockDim: strided octets, no lane reduce
+            for (int o = threadIdx.x; o < noct; o += blockDim.x) {
+                float a[8] = {0.f}, a2[8] = {0.f};
+                for (int64_t r = r0; r < r1; ++r) {
+                    u16x8 v8 = *reinterpret_cast<const u16x8*>(
+                        &x[r * C + o * 8]);
+#pragma unroll
+                    for (int j = 0; j < 8; ++j) {
+                        float v = bf2f(v8[j]);
+                        a[j] += v;
+                        a2[j] += v * v;
+                    }
+                }
+#pragma unroll
+                for (int j = 0; j < 8; ++j) {
+                    gsum[o * 8 + j] = a[j];
+                    gsq[o * 8 + j] = a2[j];
+                }
+            }
+            return;
+        }
+        if (rl < lanes) {
+            for (int64_t r = r0 + rl; r < r1; r += lanes) {
+                u16x8 v8 = *reinterpret_cast<const u16x8*>(&x[r * C + oct * 8]);
+#pragma unroll
+                for (int j = 0; j < 8; ++j) {
+                    float v = bf2f(v8[j]);
+                    acc[j] += v;
+                    acc2[j] += v * v;
+                }
+            }
+        }
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            red8[0][threadIdx.x * 8 + j] = acc[j];
+            red8[1][threadIdx.x * 8 + j] = acc2[j];
+        }
+        __syncthreads();
+        for (int off = lanes >> 1; off >= 1; off >>= 1) {
+            if (rl < off) {
+#pragma unroll
+                for (int j = 0; j < 8; ++j) {
+                    red8[0][threadIdx.x * 8 + j] +=
+                        red8[0][(threadIdx.x + off * noct) * 8 + j];
+                    red8[1][threadIdx.x * 8 + j] +=
+                        red8[1][(threadIdx.x + off * noct) * 8 + j];
+                }
+            }
+            __syncthreads();
+        }
+        if (rl == 0) {
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                gsum[oct * 8 + j] = red8[0][threadIdx.x * 8 + j];
+                gsq[oct * 8 + j] = red8[1][threadIdx.x * 8 + j];
+            }
+        }
+        return;
+    }
     if (C >= (int)blockDim.x) {  // one channel per thread, strided
         for (int c = threadIdx.x; c < C; c += blockDim.x) {
             float acc = 0.f, acc2 = 0.f;
@@ -1849,6 +1948,87 @@ __global__ void bn_bwd_partial_kernel(const unsigned short* __restrict__ dy,
     __shared__ float red[2][256];
     const int64_t r0 = (int64_t)blockIdx.x * rows_per_block;
     const int64_t r1 = min(r0 + rows_per_block, M);
+    if ((C & 7) == 0 && C <= 2048) {  // vectorized octet path (see fwd)
+        __shared__ float red8[2][2048];
+        const int noct = C >> 3;
+        int lanes = (int)blockDim.x / noct;
+        lanes = lanes ? (1 << (31 - __clz(lanes))) : 0;
+        const int oct = threadIdx.x % noct;
+        const int rl = threadIdx.x / noct;
+        if (lanes == 0) {
+            for (int o = threadIdx.x; o < noct; o += blockDim.x) {
+                float dg[8] = {0.f}, db[8] = {0.f}, mu[8], is[8];
+#pragma unroll
+                for (int j = 0; j < 8; ++j) {
+                    mu[j] = mean[o * 8 + j];
+                    is[j] = invstd[o * 8 + j];
+                }
+                for (int64_t r = r0; r < r1; ++r) {
+                    u16x8 g8 = *reinterpret_cast<const u16x8*>(
+                        &dy[r * C + o * 8]);
+                    u16x8 x8 = *reinterpret_cast<const u16x8*>(
+                        &x[r * C + o * 8]);
+#pragma unroll
+                    for (int j = 0; j < 8; ++j) {
+                        float g = bf2f(g8[j]);
+                        dg[j] += g * (bf2f(x8[j]) - mu[j]) * is[j];
+                        db[j] += g;
+                    }
+                }
+#pragma unroll
+                for (int j = 0; j < 8; ++j) {
+                    dgamma[o * 8 + j] = dg[j];
+                    dbeta[o * 8 + j] = db[j];
+                }
+            }
+            return;
+        }
+        float dg[8] = {0.f}, db[8] = {0.f};
+        if (rl < lanes) {
+            float mu[8], is[8];
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                mu[j] = mean[oct * 8 + j];
+                is[j] = invstd[oct * 8 + j];
+            }
+            for (int64_t r = r0 + rl; r < r1; r += lanes) {
+                u16x8 g8 = *reinterpret_cast<const u16x8*>(&dy[r * C + oct * 8]);
+                u16x8 x8 = *reinterpret_cast<const u16x8*>(&x[r * C + oct * 8]);
+#pragma unroll
+                for (int j = 0; j < 8; ++j) {
+                    float g = bf2f(g8[j]);
+                    dg[j] += g * (bf2f(x8[j]) - mu[j]) * is[j];
+                    db[j] += g;
+                }
+            }
+        }
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            red8[0][threadIdx.x * 8 + j] = dg[j];
+            red8[1][threadIdx.x * 8 + j] = db[j];
+        }
+        __syncthreads();
+        for (int off = lanes >> 1; off >= 1; off >>= 1) {
+            if (rl < off) {
+#pragma unroll
+                for (int j = 0; j < 8; ++j) {
+                    red8[0][threadIdx.x * 8 + j] +=
+                        red8[0][(threadIdx.x + off * noct) * 8 + j];
+                    red8[1][threadIdx.x * 8 + j] +=
+                        red8[1][(threadIdx.x + off * noct) * 8 + j];
+                }
+            }
+            __syncthreads();
+        }
+        if (rl == 0) {
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                dgamma[oct * 8 + j] = red8[0][threadIdx.x * 8 + j];
+                dbeta[oct * 8 + j] = red8[1][threadIdx.x * 8 + j];
+            }
+        }
+        return;
+    }
     if (C >= (int)blockDim.x) {
         for (int c = threadIdx.x; c < C; c += blockDim.x) {
             float dg = 0.f, db = 0.f;
@@ -2347,70 +2527,6 @@ torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
     return y;
 }
 
-// Whole backward of a small Dense layer in ONE kernel: applies the ReLU
-// mask (optional), and produces dx[M,K] = g @ w, dw[N,K] = g^T @ x and
-// db[N] = colsum(g) — replacing relu_bias_bwd/bias_grad + two GEMM launches.
-// Dot lengths are M or N (<= 64/256): plain VALU FMAs beat MFMA setup at
-// these sizes, and the batch-32 FL configs are launch-latency bound.
-__global__ void dense_bwd_small_kernel(
-    const unsigned short* __restrict__ dy, const unsigned short* __restrict__ y,
-    const unsigned short* __restrict__ x, const unsigned short* __restrict__ w,
-    unsigned short* __restrict__ dx, float* __restrict__ dw,
-    float* __restrict__ db, int M, int N, int K, int64_t ndx) {
-    const int64_t total = ndx + (int64_t)N * K;
-    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
-         i += (int64_t)gridDim.x * blockDim.x) {
-        if (i < ndx) {  // dx[m,k] = sum_n g[m,n] * w[n,k]
-            const int m = (int)(i / K), k = (int)(i % K);
-            float acc = 0.f;
-            for (int n = 0; n < N; ++n) {
-                float g = bf2f(dy[m * N + n]);
-                if (y) {
-                    unsigned short yv = y[m * N + n];
-                    if ((yv & 0x7fffu) == 0 || (yv & 0x8000u)) g = 0.f;
-                }
-                acc += g * bf2f(w[(int64_t)n * K + k]);
-            }
-            dx[i] = f2bf(acc);
-        } else {  // dw[n,k] = sum_m g[m,n] * x[m,k]; db from the k==0 thread
-            const int64_t j = i - ndx;
-            const int n = (int)(j / K), k = (int)(j % K);
-            float acc = 0.f, accb = 0.f;
-            for (int m = 0; m < M; ++m) {
-                float g = bf2f(dy[m * N + n]);
-                if (y) {
-                    unsigned short yv = y[m * N + n];
-                    if ((yv & 0x7fffu) == 0 || (yv & 0x8000u)) g = 0.f;
-                }
-                acc += g * bf2f(x[(int64_t)m * K + k]);
-                accb += g;
-            }
-            dw[j] = acc;
-            if (k == 0) db[n] = accb;
-        }
-    }
-}
-
-std::vector<torch::Tensor> dense_bwd_small(torch::Tensor dy, torch::Tensor y,
-                                           torch::Tensor x, torch::Tensor w,
-                                           bool need_dx) {
-    CHECK_GPU(dy);
-    auto dyc = dy.contiguous();
-    const int M = (int)dyc.size(0), N = (int)dyc.size(1), K = (int)w.size(1);
-    auto dx = torch::empty({need_dx ? M : 0, K}, dyc.options());
-    auto dw = torch::empty({N, K}, dyc.options().dtype(torch::kFloat32));
-    auto db = torch::empty({N}, dyc.options().dtype(torch::kFloat32));
-    const int64_t ndx = need_dx ? (int64_t)M * K : 0;
-    const int64_t total = ndx + (int64_t)N * K;
-    int blocks = (int)std::min<int64_t>((total + 255) / 256, 2048);
-    hipLaunchKernelGGL(dense_bwd_small_kernel, dim3(blocks), dim3(256), 0,
-                       at::cuda::getCurrentCUDAStream(), bf_ptr(dyc),
-                       y.numel() ? bf_ptr(y) : nullptr, bf_ptr(x), bf_ptr(w),
-                       bf_ptr_mut(dx), dw.data_ptr<float>(),
-                       db.data_ptr<float>(), M, N, K, ndx);
-    return {dx, dw, db};
-}
-
 torch::Tensor linear_dgrad(torch::Tensor dy, torch::Tensor w) {
     CHECK_GPU(dy);
     const int M = (int)dy.size(0), N = (int)dy.size(1), K = (int)w.size(1);
@@ -2433,6 +2549,26 @@ torch::Tensor linear_wgrad(torch::Tensor dy, torch::Tensor x) {
                        0, at::cuda::getCurrentCUDAStream(), bf_ptr(dy), bf_ptr(x),
                        dw.data_ptr<float>(), nullptr, N, K, M, 0);
     return dw;
+}
+
+torch::Tensor synth_batch(torch::Tensor templates, torch::Tensor labels,
+                          int64_t seed) {
+    CHECK_GPU(templates);
+    TORCH_CHECK(templates.is_contiguous() && labels.is_contiguous());
+    const int64_t n = labels.size(0);
+    const int64_t per = templates.numel() / templates.size(0);
+    auto out = torch::empty({n, templates.size(1), templates.size(2),
+                             templates.size(3)},
+                            templates.options().dtype(torch::kBFloat16));
+    const int64_t total = n * per;
+    int blocks = (int)std::min<int64_t>((total + 255) / 256, 8192);
+    hipLaunchKernelGGL(synth_batch_kernel, dim3(blocks), dim3(256), 0,
+                       at::cuda::getCurrentCUDAStream(),
+                       templates.data_ptr<float>(),
+                       labels.data_ptr<int64_t>(),
+                       reinterpret_cast<unsigned short*>(out.data_ptr()),
+                       per, total, (unsigned long long)seed);
+    return out;
 }
 
 std::vector<torch::Tensor> maxpool2x2_fwd(torch::Tensor x) {

@@ -51,6 +51,14 @@ class SyntheticMedicalImages:
         # index (cheap); the class template carries the learnable signal.
         # Sampled directly on the device — the training loop never touches host.
         if self.device.type == "cuda":
+            if self.dtype == torch.bfloat16:
+                # single fused kernel: template gather + counter-hash normal
+                # noise + sigmoid + bf16 store (vs ~6 fp32 staging passes)
+                import hefl
+                self._synth_ctr = getattr(self, "_synth_ctr", 0) + 1
+                return hefl.load_extension().synth_batch(
+                    self.templates, y.contiguous(),
+                    self.seed * 0x10001 + self._synth_ctr), y
             if not hasattr(self, "_gen"):
                 self._gen = torch.Generator(device=self.device)
                 self._gen.manual_seed(self.seed)

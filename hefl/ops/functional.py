@@ -201,18 +201,6 @@ class _LinearFn(torch.autograd.Function):
         x, w, y = ctx.saved_tensors
         if dy.is_cuda:
             dy = dy.contiguous()
-            M, N = dy.shape[0], dy.shape[-1]
-            K = w.shape[1]
-            import os
-            if (dy.dim() == 2 and ctx.has_bias and M <= 64 and N <= 256
-                    and K <= 4096 and not os.environ.get("HEFL_NO_DENSE_FUSE")):
-                # whole small-dense backward (relu mask + dx + dw + db) in
-                # ONE kernel — the FL configs are launch-latency bound
-                need_dx = bool(ctx.needs_input_grad[0])
-                empty = torch.empty(0, device=dy.device)
-                dx, dw, db = _C().dense_bwd_small(
-                    dy, y if ctx.relu else empty, x, w, need_dx)
-                return (dx if need_dx else None), dw, db, None
             db = None
             if ctx.relu and ctx.has_bias:
                 dy, db = _C().relu_bias_bwd(dy, y)

@@ -181,3 +181,24 @@ def test_device_noise_sampler_distribution():
     assert abs(e.mean().item()) < 0.02
     assert abs(e.std().item() - (21 / 2) ** 0.5) < 0.05
     assert e.abs().max().item() <= 21
+
+
+@pytest.mark.gpu
+def test_synth_batch_distribution():
+    """Fused data-gen kernel: logit(x) must equal 0.6*template + 0.4*N(0,1)
+    in distribution (mean residual ~0, std ~0.4)."""
+    from hefl.data.synthetic import SyntheticMedicalImages
+
+    ds = SyntheticMedicalImages(512, (28, 28, 1), 4, seed=3, device="cuda",
+                                dtype=torch.bfloat16)
+    idx = torch.arange(512, device="cuda")
+    x, y = ds.batch(idx)
+    assert x.dtype == torch.bfloat16 and x.shape == (512, 28, 28, 1)
+    xf = x.float().clamp(1e-4, 1 - 1e-4)
+    logit = torch.log(xf / (1 - xf))
+    resid = logit - 0.6 * ds.templates[y]
+    assert abs(resid.mean().item()) < 0.02, resid.mean().item()
+    assert abs(resid.std().item() - 0.4) < 0.05, resid.std().item()
+    # different calls draw fresh noise
+    x2, _ = ds.batch(idx)
+    assert not torch.equal(x, x2)
