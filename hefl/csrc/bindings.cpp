@@ -65,7 +65,8 @@ torch::Tensor bn_apply(torch::Tensor x, torch::Tensor mean,
                        torch::Tensor beta, bool relu);
 std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
                                   torch::Tensor mean, torch::Tensor invstd,
-                                  torch::Tensor gamma, bool train);
+                                  torch::Tensor gamma, bool train,
+                                  torch::Tensor relu_y);
 std::vector<torch::Tensor> maxpool_fwd(torch::Tensor x, int64_t k, int64_t s,
                                        int64_t p);
 torch::Tensor maxpool_bwd(torch::Tensor dy, torch::Tensor idx, int64_t H,

@@ -1937,8 +1937,21 @@ __global__ void bn_apply_kernel(const unsigned short* __restrict__ x,
 
 // dgamma[c] = sum dy*xhat; dbeta[c] = sum dy — same coalesced two-stage
 // shape (partials straight into dgamma/dbeta, zeroed by the wrapper).
+// relu_y: optional post-ReLU BN output — gates dy in place of a separate
+// relu_bwd pre-pass (one fewer full read+write of the activation grad)
+__device__ __forceinline__ float bf_gated(const unsigned short* dy,
+                                          const unsigned short* relu_y,
+                                          int64_t i) {
+    if (relu_y) {
+        unsigned short yv = relu_y[i];
+        if ((yv & 0x7fffu) == 0 || (yv & 0x8000u)) return 0.f;
+    }
+    return bf2f(dy[i]);
+}
+
 __global__ void bn_bwd_partial_kernel(const unsigned short* __restrict__ dy,
                                       const unsigned short* __restrict__ x,
+                                      const unsigned short* __restrict__ relu_y,
                                       const float* __restrict__ mean,
                                       const float* __restrict__ invstd,
                                       float* __restrict__ slab, int64_t M,
@@ -1968,9 +1981,16 @@ __global__ void bn_bwd_partial_kernel(const unsigned short* __restrict__ dy,
                         &dy[r * C + o * 8]);
                     u16x8 x8 = *reinterpret_cast<const u16x8*>(
                         &x[r * C + o * 8]);
+                    u16x8 y8{};
+                    if (relu_y)
+                        y8 = *reinterpret_cast<const u16x8*>(
+                            &relu_y[r * C + o * 8]);
 #pragma unroll
                     for (int j = 0; j < 8; ++j) {
                         float g = bf2f(g8[j]);
+                        if (relu_y &&
+                            ((y8[j] & 0x7fffu) == 0 || (y8[j] & 0x8000u)))
+                            g = 0.f;
                         dg[j] += g * (bf2f(x8[j]) - mu[j]) * is[j];
                         db[j] += g;
                     }
@@ -1994,9 +2014,15 @@ __global__ void bn_bwd_partial_kernel(const unsigned short* __restrict__ dy,
             for (int64_t r = r0 + rl; r < r1; r += lanes) {
                 u16x8 g8 = *reinterpret_cast<const u16x8*>(&dy[r * C + oct * 8]);
                 u16x8 x8 = *reinterpret_cast<const u16x8*>(&x[r * C + oct * 8]);
+                u16x8 y8{};
+                if (relu_y)
+                    y8 = *reinterpret_cast<const u16x8*>(
+                        &relu_y[r * C + oct * 8]);
 #pragma unroll
                 for (int j = 0; j < 8; ++j) {
                     float g = bf2f(g8[j]);
+                    if (relu_y && ((y8[j] & 0x7fffu) == 0 || (y8[j] & 0x8000u)))
+                        g = 0.f;
                     dg[j] += g * (bf2f(x8[j]) - mu[j]) * is[j];
                     db[j] += g;
                 }
@@ -2034,7 +2060,7 @@ __global__ void bn_bwd_partial_kernel(const unsigned short* __restrict__ dy,
             float dg = 0.f, db = 0.f;
             const float mu = mean[c], is = invstd[c];
             for (int64_t r = r0; r < r1; ++r) {
-                float g = bf2f(dy[r * C + c]);
+                float g = bf_gated(dy, relu_y, r * C + c);
                 dg += g * (bf2f(x[r * C + c]) - mu) * is;
                 db += g;
             }
@@ -2051,7 +2077,7 @@ __global__ void bn_bwd_partial_kernel(const unsigned short* __restrict__ dy,
     if (rl < lanes) {
         const float mu = mean[c], is = invstd[c];
         for (int64_t r = r0 + rl; r < r1; r += lanes) {
-            float g = bf2f(dy[r * C + c]);
+            float g = bf_gated(dy, relu_y, r * C + c);
             dg += g * (bf2f(x[r * C + c]) - mu) * is;
             db += g;
         }
@@ -2100,6 +2126,7 @@ __global__ void bn_bwd_finalize_kernel(const float* __restrict__ slab,
 
 __global__ void bn_dx_kernel(const unsigned short* __restrict__ dy,
                              const unsigned short* __restrict__ x,
+                             const unsigned short* __restrict__ relu_y,
                              const float* __restrict__ mean,
                              const float* __restrict__ invstd,
                              const float* __restrict__ gamma,
@@ -2110,7 +2137,7 @@ __global__ void bn_dx_kernel(const unsigned short* __restrict__ dy,
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
          i += (int64_t)gridDim.x * blockDim.x) {
         int c = i % C;
-        float g = bf2f(dy[i]);
+        float g = bf_gated(dy, relu_y, i);
         float v;
         if (train) {
             float xh = (bf2f(x[i]) - mean[c]) * invstd[c];
@@ -2801,9 +2828,12 @@ torch::Tensor bn_apply(torch::Tensor x, torch::Tensor mean,
 
 std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
                                   torch::Tensor mean, torch::Tensor invstd,
-                                  torch::Tensor gamma, bool train) {
+                                  torch::Tensor gamma, bool train,
+                                  torch::Tensor relu_y) {
     CHECK_GPU(dy);
     auto dyc = dy.contiguous();
+    const unsigned short* ry =
+        relu_y.numel() ? bf_ptr(relu_y) : nullptr;
     const int C = (int)x.size(-1);
     const int64_t M = x.numel() / C;
     auto f32 = x.options().dtype(torch::kFloat32);
@@ -2815,7 +2845,7 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
     int nblk = (int)((M + rpb - 1) / rpb);
     auto slab = torch::empty({nblk, 2, C}, f32);
     hipLaunchKernelGGL(bn_bwd_partial_kernel, dim3(nblk), dim3(256), 0, stream,
-                       bf_ptr(dyc), bf_ptr(x), mean.data_ptr<float>(),
+                       bf_ptr(dyc), bf_ptr(x), ry, mean.data_ptr<float>(),
                        invstd.data_ptr<float>(), slab.data_ptr<float>(), M, C,
                        rpb);
     hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3(C), dim3(256), 0,
@@ -2824,7 +2854,7 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
     int64_t total = x.numel();
     int blocks = std::min<int64_t>(ceildiv(total, 256), 2048);
     hipLaunchKernelGGL(bn_dx_kernel, dim3(blocks), dim3(256), 0, stream,
-                       bf_ptr(dyc), bf_ptr(x), mean.data_ptr<float>(),
+                       bf_ptr(dyc), bf_ptr(x), ry, mean.data_ptr<float>(),
                        invstd.data_ptr<float>(), gamma.data_ptr<float>(),
                        dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
                        bf_ptr_mut(dx), total, C, M, train ? 1 : 0);

@@ -328,10 +328,10 @@ class _BatchNorm2dFn(torch.autograd.Function):
         M = x.numel() // C
         if dy.is_cuda:
             dy = dy.contiguous()
-            if ctx.relu:
-                dy = _C().relu_bwd(dy, y)
+            # relu gate fused into the BN backward kernels (no relu_bwd pass)
+            ry = y if ctx.relu else torch.empty(0, device=dy.device)
             dx, dgamma, dbeta = _C().bn_bwd(dy, x.contiguous(), mean, invstd,
-                                            gamma.float(), ctx.training)
+                                            gamma.float(), ctx.training, ry)
         else:
             dy = dy.float()
             if ctx.relu:
