@@ -1191,10 +1191,12 @@ conv_wgrad_small_kernel(const unsigned short* __restrict__ dy,
                           red[2][row * 16 + sub] + red[3][row * 16 + sub];
                 int ko = fm * 16 + row;
                 if (ko < M && sub < NN) {
-                    if (k_chunks > 1)
-                        atomicAdd(dw + (int64_t)ko * NN + sub, v);
-                    else
-                        dw[(int64_t)ko * NN + sub] = v;
+                    // split-K: per-chunk slab slice (every [M*NN] cell is
+                    // stored exactly once -> no atomics, no zero-fill; a
+                    // tiny epilogue sums the slices)
+                    float* dst = dw + (k_chunks > 1
+                                           ? (int64_t)blockIdx.z * M * NN : 0);
+                    dst[(int64_t)ko * NN + sub] = v;
                 }
             }
         }
@@ -2289,6 +2291,17 @@ __global__ void bias_grad_kernel(const unsigned short* __restrict__ dy,
     }
 }
 
+__global__ void sum_slabs_f32_kernel(const float* __restrict__ slab,
+                                     float* __restrict__ out, int64_t total,
+                                     int nslab) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        float a = 0.f;
+        for (int sl = 0; sl < nslab; ++sl) a += slab[sl * total + i];
+        out[i] = a;
+    }
+}
+
 inline int ceildiv(int64_t a, int64_t b) { return (int)((a + b - 1) / b); }
 
 inline ConvShape make_shape(const torch::Tensor& x, const torch::Tensor& w,
@@ -2478,22 +2491,35 @@ torch::Tensor conv2d_wgrad(torch::Tensor dy, torch::Tensor x, int64_t stride,
     int tiles = ceildiv(s.Kout, 64) * ceildiv(NN, bn);
     int k_chunks = std::max(1, std::min(ceildiv(KK, 32 * 2),
                                         512 / std::max(tiles, 1)));
+    auto stream = at::cuda::getCurrentCUDAStream();
+    if (small_fast) {
+        auto dw = torch::empty({s.Kout, R, S, s.C},
+                               x.options().dtype(torch::kFloat32));
+        int kc = std::max(1, std::min(ceildiv(KK, 256), 512));
+        if (kc == 1) {
+            hipLaunchKernelGGL(conv_wgrad_small_kernel, dim3(1, 1, 1),
+                               dim3(TPB), 0, stream, bf_ptr(dy), bf_ptr(x),
+                               dw.data_ptr<float>(), s, 1);
+            return dw;
+        }
+        const int64_t total = (int64_t)s.Kout * R * S * s.C;
+        auto slab = torch::empty({kc, total},
+                                 x.options().dtype(torch::kFloat32));
+        hipLaunchKernelGGL(conv_wgrad_small_kernel, dim3(1, 1, kc), dim3(TPB),
+                           0, stream, bf_ptr(dy), bf_ptr(x),
+                           slab.data_ptr<float>(), s, kc);
+        hipLaunchKernelGGL(sum_slabs_f32_kernel,
+                           dim3((int)std::min<int64_t>((total + 255) / 256,
+                                                       1024)),
+                           dim3(256), 0, stream, slab.data_ptr<float>(),
+                           dw.data_ptr<float>(), total, kc);
+        return dw;
+    }
     auto dw = k_chunks > 1
                   ? torch::zeros({s.Kout, R, S, s.C},
                                  x.options().dtype(torch::kFloat32))
                   : torch::empty({s.Kout, R, S, s.C},
                                  x.options().dtype(torch::kFloat32));
-    auto stream = at::cuda::getCurrentCUDAStream();
-    if (small_fast) {
-        int kc = std::max(1, std::min(ceildiv(KK, 256), 512));
-        auto dw2 = kc > 1 ? torch::zeros({s.Kout, R, S, s.C},
-                                         x.options().dtype(torch::kFloat32))
-                          : dw;
-        hipLaunchKernelGGL(conv_wgrad_small_kernel, dim3(1, 1, kc), dim3(TPB),
-                           0, stream, bf_ptr(dy), bf_ptr(x),
-                           dw2.data_ptr<float>(), s, kc);
-        return dw2;
-    }
     dim3 grid(ceildiv(s.Kout, 64), ceildiv(NN, bn), k_chunks);
     if (glds_ok) {
         static torch::Tensor zbuf;
