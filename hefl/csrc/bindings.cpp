@@ -81,6 +81,11 @@ std::vector<torch::Tensor> pool_relu_bias_bwd(torch::Tensor dy,
                                               torch::Tensor y);
 void adam_prep_epoch(torch::Tensor step, torch::Tensor sched,
                      torch::Tensor hyper, double b1, double b2, int64_t S);
+void pack_mt(torch::Tensor meta, torch::Tensor ptrs, torch::Tensor sizes,
+             torch::Tensor offs, int64_t nchunks, torch::Tensor flat);
+void unpack_mt(torch::Tensor flat, torch::Tensor meta, torch::Tensor ptrs,
+               torch::Tensor shptrs, torch::Tensor sizes, torch::Tensor offs,
+               int64_t nchunks);
 void fused_adam_mt(torch::Tensor meta, torch::Tensor ptrs, torch::Tensor sizes,
                    int64_t nchunks, torch::Tensor sched, double b1, double b2,
                    double eps, int64_t zero_g, int64_t sched_off);
@@ -130,6 +135,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("relu_bias_bwd", &relu_bias_bwd);
     m.def("pool_relu_bias_bwd", &pool_relu_bias_bwd);
     m.def("fused_adam_mt", &fused_adam_mt);
+    m.def("pack_mt", &pack_mt);
+    m.def("unpack_mt", &unpack_mt);
     m.def("adam_prep_epoch", &adam_prep_epoch);
     m.def("zero_grads_mt", &zero_grads_mt);
 }

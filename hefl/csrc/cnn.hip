@@ -1742,6 +1742,45 @@ __global__ void fused_adam_mt_kernel(const int64_t* __restrict__ meta,
     }
 }
 
+// Multi-tensor weight pack/unpack: the per-round FedAvg weight transport
+// (flat fp32 vector <-> parameter tensors + bf16 shadows) as ONE kernel
+// instead of ~3 per parameter (cat / per-param copy_ / shadow cast).
+// meta: [nchunk][2] = {tensor idx, elem offset}; offs[t] = flat offset.
+__global__ void pack_mt_kernel(const int64_t* __restrict__ meta,
+                               const int64_t* __restrict__ ptrs,
+                               const int64_t* __restrict__ sizes,
+                               const int64_t* __restrict__ offs,
+                               float* __restrict__ flat) {
+    const int c = blockIdx.x;
+    const int t = (int)meta[c * 2];
+    const int64_t off = meta[c * 2 + 1];
+    const float* p = reinterpret_cast<const float*>(ptrs[t]);
+    const int64_t n = min(off + (int64_t)MT_CHUNK, sizes[t]);
+    const int64_t base = offs[t];
+    for (int64_t i = off + threadIdx.x; i < n; i += blockDim.x)
+        flat[base + i] = p[i];
+}
+
+__global__ void unpack_mt_kernel(const float* __restrict__ flat,
+                                 const int64_t* __restrict__ meta,
+                                 const int64_t* __restrict__ ptrs,
+                                 const int64_t* __restrict__ shptrs,
+                                 const int64_t* __restrict__ sizes,
+                                 const int64_t* __restrict__ offs) {
+    const int c = blockIdx.x;
+    const int t = (int)meta[c * 2];
+    const int64_t off = meta[c * 2 + 1];
+    float* p = reinterpret_cast<float*>(ptrs[t]);
+    unsigned short* sh = reinterpret_cast<unsigned short*>(shptrs[t]);
+    const int64_t n = min(off + (int64_t)MT_CHUNK, sizes[t]);
+    const int64_t base = offs[t];
+    for (int64_t i = off + threadIdx.x; i < n; i += blockDim.x) {
+        float v = flat[base + i];
+        p[i] = v;
+        if (sh) sh[i] = f2bf(v);  // bf16 shadow refreshed in the same pass
+    }
+}
+
 __global__ void zero_mt_kernel(const int64_t* __restrict__ meta,
                                const int64_t* __restrict__ ptrs,
                                const int64_t* __restrict__ sizes) {
@@ -2993,6 +3032,27 @@ void adam_prep_epoch(torch::Tensor step, torch::Tensor sched,
                        at::cuda::getCurrentCUDAStream(),
                        step.data_ptr<int64_t>(), sched.data_ptr<float>(),
                        hyper.data_ptr<float>(), (float)b1, (float)b2, (int)S);
+}
+
+void pack_mt(torch::Tensor meta, torch::Tensor ptrs, torch::Tensor sizes,
+             torch::Tensor offs, int64_t nchunks, torch::Tensor flat) {
+    CHECK_GPU(meta);
+    hipLaunchKernelGGL(pack_mt_kernel, dim3((unsigned)nchunks), dim3(256), 0,
+                       at::cuda::getCurrentCUDAStream(),
+                       meta.data_ptr<int64_t>(), ptrs.data_ptr<int64_t>(),
+                       sizes.data_ptr<int64_t>(), offs.data_ptr<int64_t>(),
+                       flat.data_ptr<float>());
+}
+
+void unpack_mt(torch::Tensor flat, torch::Tensor meta, torch::Tensor ptrs,
+               torch::Tensor shptrs, torch::Tensor sizes, torch::Tensor offs,
+               int64_t nchunks) {
+    CHECK_GPU(meta);
+    hipLaunchKernelGGL(unpack_mt_kernel, dim3((unsigned)nchunks), dim3(256), 0,
+                       at::cuda::getCurrentCUDAStream(),
+                       flat.data_ptr<float>(), meta.data_ptr<int64_t>(),
+                       ptrs.data_ptr<int64_t>(), shptrs.data_ptr<int64_t>(),
+                       sizes.data_ptr<int64_t>(), offs.data_ptr<int64_t>());
 }
 
 void zero_grads_mt(torch::Tensor meta, torch::Tensor ptrs, torch::Tensor sizes,

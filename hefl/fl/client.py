@@ -275,10 +275,63 @@ class LocalClient:
                 stats.train_acc = float(acc_sum) / max(stats.samples, 1)
         return stats
 
+    # ----- FedAvg weight transport: on GPU, ONE multi-tensor kernel packs
+    # (or unpacks + refreshes bf16 shadows) the whole parameter vector,
+    # replacing ~3 kernels per parameter tensor. -----
+    def _weight_tables(self):
+        wt = getattr(self, "_wt", None)
+        if wt is not None:
+            return wt
+        import hefl  # extension must be present on GPU (fails loudly if not)
+        hefl.load_extension()
+        from .weights import _agg_tensors
+        self.opt.ensure_shadows()
+        dev = self.device
+        tensors = list(_agg_tensors(self.model))
+        ptrs, shp, sizes, offs, meta = [], [], [], [], []
+        total = 0
+        for t, p in enumerate(tensors):
+            assert p.dtype == torch.float32 and p.is_contiguous()
+            ptrs.append(p.data_ptr())
+            sh = getattr(p, "_bf16", None)
+            shp.append(sh.data_ptr() if sh is not None else 0)
+            sizes.append(p.numel())
+            offs.append(total)
+            total += p.numel()
+            for off in range(0, p.numel(), FusedAdam._MT_CHUNK):
+                meta.append([t, off])
+        self._wt = {
+            "meta": torch.tensor(meta, dtype=torch.int64, device=dev),
+            "ptrs": torch.tensor(ptrs, dtype=torch.int64, device=dev),
+            "shptrs": torch.tensor(shp, dtype=torch.int64, device=dev),
+            "sizes": torch.tensor(sizes, dtype=torch.int64, device=dev),
+            "offs": torch.tensor(offs, dtype=torch.int64, device=dev),
+            "n": len(meta), "total": total,
+        }
+        return self._wt
+
     def get_weights(self) -> torch.Tensor:
+        if self.device.type == "cuda":
+            import hefl
+            wt = self._weight_tables()
+            flat = torch.empty(wt["total"], dtype=torch.float32,
+                               device=self.device)
+            hefl.load_extension().pack_mt(wt["meta"], wt["ptrs"], wt["sizes"],
+                                          wt["offs"], wt["n"], flat)
+            return flat
         return flat_params(self.model)
 
     def set_weights(self, vec: torch.Tensor) -> None:
+        if self.device.type == "cuda":
+            import hefl
+            wt = self._weight_tables()
+            v = vec.to(device=self.device, dtype=torch.float32).contiguous()
+            if v.numel() != wt["total"]:
+                raise ValueError(f"vector length {v.numel()} != {wt['total']}")
+            hefl.load_extension().unpack_mt(v, wt["meta"], wt["ptrs"],
+                                            wt["shptrs"], wt["sizes"],
+                                            wt["offs"], wt["n"])
+            return
         load_flat_params(self.model, vec.to(self.device))
         if hasattr(self.opt, "refresh_shadows"):
             self.opt.refresh_shadows()  # bf16 shadows must follow FedAvg loads

@@ -112,3 +112,28 @@ def test_gpu_resnet_encrypted_round():
     for b in fl.global_model.buffers():
         if b.is_floating_point():
             assert t.isfinite(b).all()
+
+
+@pytest.mark.gpu
+def test_mt_weight_pack_unpack():
+    """Multi-tensor pack/unpack kernels vs the torch reference path, incl.
+    bf16 shadow refresh on load."""
+    from hefl.config import preset
+    from hefl.fl.client import LocalClient
+    from hefl.fl.weights import flat_params
+
+    cfg = preset("config2")
+    cfg.fl.n_clients = 1
+    cfg.fl.samples_per_client = 64
+    c = LocalClient(cfg, 0, device="cuda:0")
+    ref = flat_params(c.model)
+    got = c.get_weights()
+    assert torch.equal(got.cpu(), ref.cpu())
+    vec = torch.randn_like(ref) * 0.05
+    c.set_weights(vec)
+    back = flat_params(c.model)
+    assert torch.equal(back.cpu(), vec.cpu())
+    for p in c.opt.params:  # shadows refreshed in the same kernel
+        sh = getattr(p, "_bf16", None)
+        assert sh is not None
+        assert torch.equal(sh, p.detach().to(torch.bfloat16))
