@@ -154,8 +154,22 @@ gemm_kernel(const unsigned short* __restrict__ A,
 // per-row im2col coordinate decode is hoisted out of the K loop.
 // ---------------------------------------------------------------------------
 
+// Branch-free unsigned division by a runtime constant (magic multiply):
+// exact for all n < 2^31, d < 2^16 (Hacker's Delight round-up magic with a
+// 64-bit multiply). The im2col index decode in the conv stage loops was a
+// chain of ~6 u32 divisions per k-step — more VALU cycles than the MFMAs.
+struct FastDiv {
+    unsigned long long m;
+    int p;
+};
+
+__device__ __forceinline__ unsigned fdiv(unsigned n, FastDiv f) {
+    return (unsigned)(((unsigned long long)n * f.m) >> f.p);
+}
+
 struct ConvShape {
     int N, H, W, C, Kout, R, S, OH, OW, stride, pad;
+    FastDiv fC, fS, fOW, fOHOW, fKout;
 };
 
 template <int BM, int BN, int WM, int WN, int FM, int FN>
@@ -346,7 +360,7 @@ conv_fwd_glds_kernel(const unsigned short* __restrict__ x,
                      int relu, int k_chunks) {
     // ONE shared object (two makes hipcc emit a vmcnt(0) drain before each
     // k-step's first ds_read, defeating the glds pipeline — guide trap 4a)
-    __shared__ unsigned short smem[2 * (BM + BN) * BK];
+    __shared__ unsigned short smem[3 * (BM + BN) * BK];
     auto As = [&](int buf) -> unsigned short (*)[BK] {
         return reinterpret_cast<unsigned short(*)[BK]>(smem + buf * (BM + BN) * BK);
     };
@@ -396,8 +410,8 @@ conv_fwd_glds_kernel(const unsigned short* __restrict__ x,
             const int k = k0 + akcs[t];
             const unsigned short* src = zbuf;
             if (a_ok[t] && k < KK) {
-                int rs = k / s.C, c = k % s.C;
-                int r = rs / s.S, ss = rs % s.S;
+                int rs = (int)fdiv((unsigned)k, s.fC), c = k - rs * s.C;
+                int r = (int)fdiv((unsigned)rs, s.fS), ss = rs - r * s.S;
                 int ih = a_oh[t] * s.stride + r - s.pad;
                 int iw = a_ow[t] * s.stride + ss - s.pad;
                 if (ih >= 0 && ih < s.H && iw >= 0 && iw < s.W)
@@ -431,11 +445,20 @@ conv_fwd_glds_kernel(const unsigned short* __restrict__ x,
     const int kc_len = (ksteps + k_chunks - 1) / k_chunks;
     const int kbeg = blockIdx.z * kc_len * BK;
     const int kend = min(kbeg + kc_len * BK, KK);
+    // 3-buffer pipeline, one stage always in flight: counted s_waitcnt
+    // vmcnt(2) (= 2 glds instructions per stage per wave) + raw s_barrier.
+    // __syncthreads here would emit vmcnt(0) and drain the in-flight stage
+    // (guide: 3-buf span +83% vs sync, 2-buf +40%).
     stage(0, kbeg);
-    __syncthreads();  // drains the DMA (vmcnt 0) + barrier
+    stage(1, kbeg + BK);
     int buf = 0;
     for (int k0 = kbeg; k0 < kend; k0 += BK) {
-        if (k0 + BK < kend) stage(buf ^ 1, k0 + BK);  // issue BEFORE compute
+        if (k0 + BK < kend)
+            asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+        else
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        asm volatile("s_barrier" ::: "memory");
+        if (k0 + 2 * BK < kend) stage(buf == 0 ? 2 : buf - 1, k0 + 2 * BK);
         {   // MFMA over the BK-deep tile, 32 k per instruction
             const int wm = wave / WN, wn = wave % WN;
             const int half = lane >> 4, sub = lane & 15;
@@ -461,9 +484,9 @@ conv_fwd_glds_kernel(const unsigned short* __restrict__ x,
                             a[i], b[j], acc[i][j], 0, 0, 0);
             }
         }
-        __syncthreads();
-        buf ^= 1;
+        buf = buf == 2 ? 0 : buf + 1;
     }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // degenerate-range drain
     const int wm = wave / WN, wn = wave % WN;
 #pragma unroll
     for (int j = 0; j < FN; ++j) {
@@ -663,7 +686,7 @@ conv_dgrad_glds_kernel(const unsigned short* __restrict__ dy,
                        float* __restrict__ dx32,
                        const unsigned short* __restrict__ zbuf, ConvShape s,
                        int k_chunks) {
-    __shared__ unsigned short smem[2 * (BM + BN) * 32];
+    __shared__ unsigned short smem[3 * (BM + BN) * 32];
     auto As = [&](int buf) -> unsigned short (*)[32] {
         return reinterpret_cast<unsigned short(*)[32]>(smem + buf * (BM + BN) * 32);
     };
@@ -702,8 +725,8 @@ conv_dgrad_glds_kernel(const unsigned short* __restrict__ dy,
             const int k = k0 + akc;
             const unsigned short* src = zbuf;
             if (a_ok[t] && k < KK) {
-                int rs = k / s.Kout, ko = k % s.Kout;
-                int r = rs / s.S, ss = rs % s.S;
+                int rs = (int)fdiv((unsigned)k, s.fKout), ko = k - rs * s.Kout;
+                int r = (int)fdiv((unsigned)rs, s.fS), ss = rs - r * s.S;
                 int oh_num = a_ih[t] + s.pad - r, ow_num = a_iw[t] + s.pad - ss;
                 if (S1) {
                     if (oh_num >= 0 && ow_num >= 0 && oh_num < s.OH &&
@@ -728,7 +751,7 @@ conv_dgrad_glds_kernel(const unsigned short* __restrict__ dy,
             const int c = n0 + cc;
             const unsigned short* src = zbuf;
             if (k < KK && c < s.C) {
-                int rs = k / s.Kout, ko = k % s.Kout;
+                int rs = (int)fdiv((unsigned)k, s.fKout), ko = k - rs * s.Kout;
                 src = w + ((int64_t)ko * s.R * s.S + rs) * s.C + c;
             }
             char* base = (char*)&Bst(buf)[0][0] + wave * 1024;
@@ -746,11 +769,17 @@ conv_dgrad_glds_kernel(const unsigned short* __restrict__ dy,
     const int kc_len = (ksteps + k_chunks - 1) / k_chunks;
     const int kbeg = blockIdx.z * kc_len * 32;
     const int kend = min(kbeg + kc_len * 32, KK);
+    // 3-buffer counted-vmcnt pipeline (see conv_fwd_glds_kernel)
     stage(0, kbeg);
-    __syncthreads();
+    stage(1, kbeg + 32);
     int buf = 0;
     for (int k0 = kbeg; k0 < kend; k0 += 32) {
-        if (k0 + 32 < kend) stage(buf ^ 1, k0 + 32);
+        if (k0 + 32 < kend)
+            asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+        else
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        asm volatile("s_barrier" ::: "memory");
+        if (k0 + 64 < kend) stage(buf == 0 ? 2 : buf - 1, k0 + 64);
         {
             const int wm = wave / WN, wn = wave % WN;
             const int half = lane >> 4, sub = lane & 15;
@@ -774,9 +803,9 @@ conv_dgrad_glds_kernel(const unsigned short* __restrict__ dy,
                     acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         a[i], b[j], acc[i][j], 0, 0, 0);
         }
-        __syncthreads();
-        buf ^= 1;
+        buf = buf == 2 ? 0 : buf + 1;
     }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     const int wm = wave / WN, wn = wave % WN;
 #pragma unroll
     for (int j = 0; j < FN; ++j) {
@@ -865,12 +894,12 @@ conv_wgrad_kernel(const unsigned short* __restrict__ dy,
             unsigned short* dst = &Xs[pix][cc];
             bool done = false;
             if (kpix < kend && nn < NN) {
-                int n_ = kpix / (s.OH * s.OW);
-                int rem = kpix % (s.OH * s.OW);
-                int oh = rem / s.OW, ow = rem % s.OW;
-                int c = nn % s.C;
-                int rs = nn / s.C;
-                int r = rs / s.S, ss = rs % s.S;
+                int n_ = (int)fdiv((unsigned)kpix, s.fOHOW);
+                int rem = kpix - n_ * (s.OH * s.OW);
+                int oh = (int)fdiv((unsigned)rem, s.fOW), ow = rem - oh * s.OW;
+                int rs = (int)fdiv((unsigned)nn, s.fC);
+                int c = nn - rs * s.C;
+                int r = (int)fdiv((unsigned)rs, s.fS), ss = rs - r * s.S;
                 int ih = oh * s.stride + r - s.pad;
                 int iw = ow * s.stride + ss - s.pad;
                 if (fast_x && c + 8 <= s.C) {
@@ -963,7 +992,7 @@ conv_wgrad_glds_kernel(const unsigned short* __restrict__ dy,
                        float* __restrict__ dw,
                        const unsigned short* __restrict__ zbuf, ConvShape s,
                        int k_chunks) {
-    __shared__ unsigned short smem[2 * (64 + BN) * 32];
+    __shared__ unsigned short smem[3 * (64 + BN) * 32];
     auto Dys = [&](int buf) -> unsigned short (*)[64] {
         return reinterpret_cast<unsigned short(*)[64]>(smem + buf * (64 + BN) * 32);
     };
@@ -1003,12 +1032,12 @@ conv_wgrad_glds_kernel(const unsigned short* __restrict__ dy,
             const int nn = n0 + cc;
             const unsigned short* src = zbuf;
             if (kpix < kend && nn < NN) {
-                int n_ = kpix / (s.OH * s.OW);
-                int rem = kpix % (s.OH * s.OW);
-                int oh = rem / s.OW, ow = rem % s.OW;
-                int c = nn % s.C;
-                int rs = nn / s.C;
-                int r = rs / s.S, ss = rs % s.S;
+                int n_ = (int)fdiv((unsigned)kpix, s.fOHOW);
+                int rem = kpix - n_ * (s.OH * s.OW);
+                int oh = (int)fdiv((unsigned)rem, s.fOW), ow = rem - oh * s.OW;
+                int rs = (int)fdiv((unsigned)nn, s.fC);
+                int c = nn - rs * s.C;
+                int r = (int)fdiv((unsigned)rs, s.fS), ss = rs - r * s.S;
                 if (c + 8 <= s.C) {
                     int ih = oh * s.stride + r - s.pad;
                     int iw = ow * s.stride + ss - s.pad;
@@ -1027,11 +1056,17 @@ conv_wgrad_glds_kernel(const unsigned short* __restrict__ dy,
 #pragma unroll
         for (int j = 0; j < FN; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
+    // 3-buffer counted-vmcnt pipeline (see conv_fwd_glds_kernel)
     stage(0, kbeg);
-    __syncthreads();
+    stage(1, kbeg + 32);
     int buf = 0;
     for (int k0 = kbeg; k0 < kend; k0 += 32) {
-        if (k0 + 32 < kend) stage(buf ^ 1, k0 + 32);
+        if (k0 + 32 < kend)
+            asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+        else
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        asm volatile("s_barrier" ::: "memory");
+        if (k0 + 64 < kend) stage(buf == 0 ? 2 : buf - 1, k0 + 64);
         {
             const int wm = wave / WN, wn = wave % WN;
             const int half = lane >> 4, sub = lane & 15;
@@ -1059,9 +1094,9 @@ conv_wgrad_glds_kernel(const unsigned short* __restrict__ dy,
                     acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         a[i], b[j], acc[i][j], 0, 0, 0);
         }
-        __syncthreads();
-        buf ^= 1;
+        buf = buf == 2 ? 0 : buf + 1;
     }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     const int wm = wave / WN, wn = wave % WN;
 #pragma unroll
     for (int j = 0; j < FN; ++j) {
@@ -2343,6 +2378,24 @@ __global__ void sum_slabs_f32_kernel(const float* __restrict__ slab,
 
 inline int ceildiv(int64_t a, int64_t b) { return (int)((a + b - 1) / b); }
 
+inline FastDiv fdiv_make(unsigned d) {
+    int sh = 0;
+    while ((1ull << sh) < d) ++sh;  // ceil log2(d)
+    FastDiv f;
+    f.p = 32 + sh;
+    f.m = (unsigned long long)(((static_cast<unsigned __int128>(1) << f.p) +
+                                d - 1) / d);
+    return f;
+}
+
+inline void fill_magic(ConvShape& s) {
+    s.fC = fdiv_make((unsigned)s.C);
+    s.fS = fdiv_make((unsigned)s.S);
+    s.fOW = fdiv_make((unsigned)s.OW);
+    s.fOHOW = fdiv_make((unsigned)(s.OH * s.OW));
+    s.fKout = fdiv_make((unsigned)s.Kout);
+}
+
 inline ConvShape make_shape(const torch::Tensor& x, const torch::Tensor& w,
                             int stride, int pad) {
     ConvShape s;
@@ -2357,6 +2410,7 @@ inline ConvShape make_shape(const torch::Tensor& x, const torch::Tensor& w,
     s.pad = pad;
     s.OH = (s.H + 2 * pad - s.R) / stride + 1;
     s.OW = (s.W + 2 * pad - s.S) / stride + 1;
+    fill_magic(s);
     return s;
 }
 
@@ -2439,6 +2493,7 @@ torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor w, int64_t stride,
     s.Kout = (int)w.size(0); s.R = (int)w.size(1); s.S = (int)w.size(2);
     s.stride = (int)stride; s.pad = (int)pad;
     s.OH = (int)dy.size(1); s.OW = (int)dy.size(2);
+    fill_magic(s);
     auto dx = torch::empty({N, (int64_t)H, (int64_t)W, C}, dy.options());
     const int M = N * (int)H * (int)W;
     auto stream = at::cuda::getCurrentCUDAStream();
@@ -2520,6 +2575,7 @@ torch::Tensor conv2d_wgrad(torch::Tensor dy, torch::Tensor x, int64_t stride,
     s.Kout = (int)dy.size(3); s.R = (int)R; s.S = (int)S;
     s.stride = (int)stride; s.pad = (int)pad;
     s.OH = (int)dy.size(1); s.OW = (int)dy.size(2);
+    fill_magic(s);
     const int NN = s.R * s.S * s.C;
     const int KK = s.N * s.OH * s.OW;
     const bool big = NN > 16;
