@@ -260,8 +260,8 @@ conv_fwd_kernel(const unsigned short* __restrict__ x,
             unsigned short* dst = &sm.As[arows[t]][akc];
             if (am_[t] < M && k < KK) {
                 if (fast) {
-                    int rs = k / s.C, c = k % s.C;
-                    int r = rs / s.S, ss = rs % s.S;
+                    int rs = (int)fdiv((unsigned)k, s.fC), c = k - rs * s.C;
+                    int r = (int)fdiv((unsigned)rs, s.fS), ss = rs - r * s.S;
                     int ih = a_oh[t] * s.stride + r - s.pad;
                     int iw = a_ow[t] * s.stride + ss - s.pad;
                     if (ih >= 0 && ih < s.H && iw >= 0 && iw < s.W)
@@ -275,9 +275,9 @@ conv_fwd_kernel(const unsigned short* __restrict__ x,
                         int kk = k + j;
                         unsigned short v = 0;
                         if (kk < KK) {
-                            int c = kk % s.C;
-                            int rs = kk / s.C;
-                            int r = rs / s.S, ss = rs % s.S;
+                            int rs = (int)fdiv((unsigned)kk, s.fC);
+                            int c = kk - rs * s.C;
+                            int r = (int)fdiv((unsigned)rs, s.fS), ss = rs - r * s.S;
                             int ih = a_oh[t] * s.stride + r - s.pad;
                             int iw = a_ow[t] * s.stride + ss - s.pad;
                             if (ih >= 0 && ih < s.H && iw >= 0 && iw < s.W)
@@ -560,8 +560,8 @@ conv_dgrad_kernel(const unsigned short* __restrict__ dy,
             bool done = false;
             if (am_[t] < M && k < KK && fast) {
                 // chunk shares (r,s): ko = k % Kout, rs = k / Kout
-                int rs = k / s.Kout, ko = k % s.Kout;
-                int r = rs / s.S, ss = rs % s.S;
+                int rs = (int)fdiv((unsigned)k, s.fKout), ko = k - rs * s.Kout;
+                int r = (int)fdiv((unsigned)rs, s.fS), ss = rs - r * s.S;
                 int oh_num = a_ih[t] + s.pad - r, ow_num = a_iw[t] + s.pad - ss;
                 done = true;
                 if (S1) {  // stride 1: no divisibility checks or divisions
@@ -590,8 +590,8 @@ conv_dgrad_kernel(const unsigned short* __restrict__ dy,
                         int kk = k + j;
                         unsigned short v = 0;
                         if (kk < KK) {
-                            int rs = kk / s.Kout, ko = kk % s.Kout;
-                            int r = rs / s.S, ss = rs % s.S;
+                            int rs = (int)fdiv((unsigned)kk, s.fKout), ko = kk - rs * s.Kout;
+                            int r = (int)fdiv((unsigned)rs, s.fS), ss = rs - r * s.S;
                             int oh_num = a_ih[t] + s.pad - r;
                             int ow_num = a_iw[t] + s.pad - ss;
                             if (oh_num >= 0 && ow_num >= 0 &&
@@ -619,7 +619,7 @@ conv_dgrad_kernel(const unsigned short* __restrict__ dy,
             const int c = n0 + cc;
             unsigned short* dst = &Bst[kk][cc];
             if (k < KK && c < s.C) {
-                int rs = k / s.Kout, ko = k % s.Kout;
+                int rs = (int)fdiv((unsigned)k, s.fKout), ko = k - rs * s.Kout;
                 const unsigned short* src =
                     w + ((int64_t)ko * s.R * s.S + rs) * s.C + c;
                 if (s.C % 8 == 0) {
@@ -916,9 +916,9 @@ conv_wgrad_kernel(const unsigned short* __restrict__ dy,
                         int n2 = nn + j;
                         unsigned short v = 0;
                         if (n2 < NN) {
-                            int c2 = n2 % s.C;
-                            int rs2 = n2 / s.C;
-                            int r2 = rs2 / s.S, ss2 = rs2 % s.S;
+                            int rs2 = (int)fdiv((unsigned)n2, s.fC);
+                            int c2 = n2 - rs2 * s.C;
+                            int r2 = (int)fdiv((unsigned)rs2, s.fS), ss2 = rs2 - r2 * s.S;
                             int ih2 = oh * s.stride + r2 - s.pad;
                             int iw2 = ow * s.stride + ss2 - s.pad;
                             if (ih2 >= 0 && ih2 < s.H && iw2 >= 0 && iw2 < s.W)
@@ -1171,12 +1171,12 @@ conv_wgrad_small_kernel(const unsigned short* __restrict__ dy,
             unsigned short v = 0;
             const int kpix = p0 + pix;
             if (kpix < kend) {
-                int n_ = kpix / (s.OH * s.OW);
-                int rem = kpix % (s.OH * s.OW);
-                int oh = rem / s.OW, ow = rem % s.OW;
-                int c = nn % s.C;
-                int rs = nn / s.C;
-                int r = rs / s.S, ss = rs % s.S;
+                int n_ = (int)fdiv((unsigned)kpix, s.fOHOW);
+                int rem = kpix - n_ * (s.OH * s.OW);
+                int oh = (int)fdiv((unsigned)rem, s.fOW), ow = rem - oh * s.OW;
+                int rs = (int)fdiv((unsigned)nn, s.fC);
+                int c = nn - rs * s.C;
+                int r = (int)fdiv((unsigned)rs, s.fS), ss = rs - r * s.S;
                 int ih = oh * s.stride + r - s.pad;
                 int iw = ow * s.stride + ss - s.pad;
                 if (ih >= 0 && ih < s.H && iw >= 0 && iw < s.W)
