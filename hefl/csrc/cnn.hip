@@ -1252,10 +1252,10 @@ __global__ void synth_batch_kernel(const float* __restrict__ T,
                                    const int64_t* __restrict__ lab,
                                    unsigned short* __restrict__ out,
                                    int64_t per_img, int64_t total,
-                                   unsigned long long seed) {
+                                   unsigned long long seed, FastDiv fPer) {
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
          i += (int64_t)gridDim.x * blockDim.x) {
-        const int64_t n = i / per_img;
+        const int64_t n = (int64_t)fdiv((unsigned)i, fPer);
         const int64_t off = i - n * per_img;
         const float t = T[lab[n] * per_img + off];
         // splitmix64 counter hash -> two 32-bit uniforms -> Box-Muller
@@ -1274,16 +1274,18 @@ __global__ void synth_batch_kernel(const float* __restrict__ T,
 __global__ void maxpool_fwd_kernel(const unsigned short* __restrict__ x,
                                    unsigned short* __restrict__ y,
                                    uint8_t* __restrict__ idx, int N, int H,
-                                   int W, int C, int OH, int OW) {
+                                   int W, int C, int OH, int OW, FastDiv fC,
+                                   FastDiv fOW, FastDiv fOH) {
     int64_t total = (int64_t)N * OH * OW * C;
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
          i += (int64_t)gridDim.x * blockDim.x) {
-        int c = i % C;
-        int64_t t = i / C;
-        int ow = t % OW;
-        t /= OW;
-        int oh = t % OH;
-        int n = t / OH;
+        unsigned t = fdiv((unsigned)i, fC);
+        int c = (int)((unsigned)i - t * C);
+        unsigned t2 = fdiv(t, fOW);
+        int ow = (int)(t - t2 * OW);
+        unsigned t3 = fdiv(t2, fOH);
+        int oh = (int)(t2 - t3 * OH);
+        int n = (int)t3;
         int ih = oh * 2, iw = ow * 2;
         float best = -3.4e38f;
         int bi = 0;
@@ -1304,16 +1306,18 @@ __global__ void maxpool2x2_bwd_gather_kernel(const unsigned short* __restrict__ 
                                              const uint8_t* __restrict__ idx,
                                              unsigned short* __restrict__ dx,
                                              int N, int H, int W, int C,
-                                             int OH, int OW) {
+                                             int OH, int OW, FastDiv fC,
+                                             FastDiv fW, FastDiv fH) {
     int64_t total = (int64_t)N * H * W * C;
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
          i += (int64_t)gridDim.x * blockDim.x) {
-        int c = i % C;
-        int64_t t = i / C;
-        int iw = t % W;
-        t /= W;
-        int ih = t % H;
-        int n = t / H;
+        unsigned t = fdiv((unsigned)i, fC);
+        int c = (int)((unsigned)i - t * C);
+        unsigned t2 = fdiv(t, fW);
+        int iw = (int)(t - t2 * W);
+        unsigned t3 = fdiv(t2, fH);
+        int ih = (int)(t2 - t3 * H);
+        int n = (int)t3;
         int oh = ih >> 1, ow = iw >> 1;
         unsigned short v = 0;
         if (oh < OH && ow < OW) {
@@ -1388,12 +1392,12 @@ __global__ void relu_bias_bwd_kernel(const unsigned short* __restrict__ dy,
 __device__ inline unsigned short pool_relu_gate(
     const unsigned short* __restrict__ dy, const uint8_t* __restrict__ idx,
     const unsigned short* __restrict__ y, int64_t r, int c, int K, int H,
-    int W, int OH, int OW) {
-    int64_t t = r;
-    const int iw = (int)(t % W);
-    t /= W;
-    const int ih = (int)(t % H);
-    const int n = (int)(t / H);
+    int W, int OH, int OW, FastDiv fW, FastDiv fH) {
+    const unsigned t1 = fdiv((unsigned)r, fW);
+    const int iw = (int)((unsigned)r - t1 * W);
+    const unsigned t2 = fdiv(t1, fH);
+    const int ih = (int)(t1 - t2 * H);
+    const int n = (int)t2;
     const int oh = ih >> 1, ow = iw >> 1;
     if (oh >= OH || ow >= OW) return 0;
     const int64_t o = (((int64_t)n * OH + oh) * OW + ow) * K + c;
@@ -1407,7 +1411,7 @@ __global__ void pool_relu_bias_bwd_kernel(
     const unsigned short* __restrict__ dy, const uint8_t* __restrict__ idx,
     const unsigned short* __restrict__ y, unsigned short* __restrict__ dym,
     float* __restrict__ db, int64_t M, int K, int rows_per_block, int H,
-    int W, int OH, int OW) {
+    int W, int OH, int OW, FastDiv fW, FastDiv fH) {
     __shared__ float red[256];
     const int64_t r0 = (int64_t)blockIdx.x * rows_per_block;
     const int64_t r1 = min(r0 + rows_per_block, M);
@@ -1416,7 +1420,7 @@ __global__ void pool_relu_bias_bwd_kernel(
             float acc = 0.f;
             for (int64_t r = r0; r < r1; ++r) {
                 unsigned short g =
-                    pool_relu_gate(dy, idx, y, r, c, K, H, W, OH, OW);
+                    pool_relu_gate(dy, idx, y, r, c, K, H, W, OH, OW, fW, fH);
                 dym[r * K + c] = g;
                 acc += bf2f(g);
             }
@@ -1433,7 +1437,7 @@ __global__ void pool_relu_bias_bwd_kernel(
     if (rl < lanes) {
         for (int64_t r = r0 + rl; r < r1; r += lanes) {
             unsigned short g =
-                pool_relu_gate(dy, idx, y, r, c, K, H, W, OH, OW);
+                pool_relu_gate(dy, idx, y, r, c, K, H, W, OH, OW, fW, fH);
             dym[r * K + c] = g;
             acc += bf2f(g);
         }
@@ -2235,20 +2239,24 @@ __global__ void maxpool_gen_fwd_kernel(const unsigned short* __restrict__ x,
                                        unsigned short* __restrict__ y,
                                        uint8_t* __restrict__ idx, int N, int H,
                                        int W, int C, int OH, int OW, int k,
-                                       int s, int p) {
+                                       int s, int p, FastDiv fC, FastDiv fOW,
+                                       FastDiv fOH) {
     int64_t total = (int64_t)N * OH * OW * C;
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
          i += (int64_t)gridDim.x * blockDim.x) {
-        int c = i % C;
-        int64_t t = i / C;
-        int ow = t % OW;
-        t /= OW;
-        int oh = t % OH;
-        int n = t / OH;
+        unsigned t1 = fdiv((unsigned)i, fC);
+        int c = (int)((unsigned)i - t1 * C);
+        unsigned t2 = fdiv(t1, fOW);
+        int ow = (int)(t1 - t2 * OW);
+        unsigned t3 = fdiv(t2, fOH);
+        int oh = (int)(t2 - t3 * OH);
+        int n = (int)t3;
         float best = -3.4e38f;
         int bi = 0;
+        int dh = -1, dwc = k;  // incremental d/k, d%k
         for (int d = 0; d < k * k; ++d) {
-            int ih = oh * s + d / k - p, iw = ow * s + d % k - p;
+            if (++dwc >= k) { dwc = 0; ++dh; }
+            int ih = oh * s + dh - p, iw = ow * s + dwc - p;
             if (ih < 0 || ih >= H || iw < 0 || iw >= W) continue;
             float v = bf2f(x[(((int64_t)n * H + ih) * W + iw) * C + c]);
             if (v > best) { best = v; bi = d; }
@@ -2262,18 +2270,21 @@ __global__ void maxpool_gen_bwd_kernel(const unsigned short* __restrict__ dy,
                                        const uint8_t* __restrict__ idx,
                                        float* __restrict__ dx32, int N, int H,
                                        int W, int C, int OH, int OW, int k,
-                                       int s, int p) {
+                                       int s, int p, FastDiv fC, FastDiv fOW,
+                                       FastDiv fOH, FastDiv fk) {
     int64_t total = (int64_t)N * OH * OW * C;
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
          i += (int64_t)gridDim.x * blockDim.x) {
-        int c = i % C;
-        int64_t t = i / C;
-        int ow = t % OW;
-        t /= OW;
-        int oh = t % OH;
-        int n = t / OH;
+        unsigned t1 = fdiv((unsigned)i, fC);
+        int c = (int)((unsigned)i - t1 * C);
+        unsigned t2 = fdiv(t1, fOW);
+        int ow = (int)(t1 - t2 * OW);
+        unsigned t3 = fdiv(t2, fOH);
+        int oh = (int)(t2 - t3 * OH);
+        int n = (int)t3;
         int d = idx[i];
-        int ih = oh * s + d / k - p, iw = ow * s + d % k - p;
+        int dk = (int)fdiv((unsigned)d, fk);
+        int ih = oh * s + dk - p, iw = ow * s + (d - dk * k) - p;
         if (ih < 0 || ih >= H || iw < 0 || iw >= W) continue;
         atomicAdd(dx32 + (((int64_t)n * H + ih) * W + iw) * C + c,
                   bf2f(dy[i]));
@@ -2715,7 +2726,8 @@ torch::Tensor synth_batch(torch::Tensor templates, torch::Tensor labels,
                        templates.data_ptr<float>(),
                        labels.data_ptr<int64_t>(),
                        reinterpret_cast<unsigned short*>(out.data_ptr()),
-                       per, total, (unsigned long long)seed);
+                       per, total, (unsigned long long)seed,
+                       fdiv_make((unsigned)per));
     return out;
 }
 
@@ -2732,7 +2744,8 @@ std::vector<torch::Tensor> maxpool2x2_fwd(torch::Tensor x) {
     hipLaunchKernelGGL(maxpool_fwd_kernel, dim3(blocks), dim3(256), 0,
                        at::cuda::getCurrentCUDAStream(), bf_ptr(x),
                        bf_ptr_mut(y), idx.data_ptr<uint8_t>(), N, H, W, C, OH,
-                       OW);
+                       OW, fdiv_make((unsigned)C), fdiv_make((unsigned)OW),
+                       fdiv_make((unsigned)OH));
     return {y, idx};
 }
 
@@ -2747,7 +2760,8 @@ torch::Tensor maxpool2x2_bwd(torch::Tensor dy, torch::Tensor idx, int64_t H,
     hipLaunchKernelGGL(maxpool2x2_bwd_gather_kernel, dim3(blocks), dim3(256), 0,
                        at::cuda::getCurrentCUDAStream(), bf_ptr(dy),
                        idx.data_ptr<uint8_t>(), bf_ptr_mut(dx), N, (int)H,
-                       (int)W, C, OH, OW);
+                       (int)W, C, OH, OW, fdiv_make((unsigned)C),
+                       fdiv_make((unsigned)W), fdiv_make((unsigned)H));
     return dx;
 }
 
@@ -2786,7 +2800,8 @@ std::vector<torch::Tensor> pool_relu_bias_bwd(torch::Tensor dy,
     hipLaunchKernelGGL(pool_relu_bias_bwd_kernel, dim3(nblk), dim3(256), 0,
                        at::cuda::getCurrentCUDAStream(), bf_ptr(dyc),
                        idx.data_ptr<uint8_t>(), bf_ptr(y), bf_ptr_mut(dym),
-                       db.data_ptr<float>(), M, K, rpb, H, W, OH, OW);
+                       db.data_ptr<float>(), M, K, rpb, H, W, OH, OW, fdiv_make((unsigned)W),
+                       fdiv_make((unsigned)H));
     return {dym, db};
 }
 
@@ -2997,7 +3012,9 @@ std::vector<torch::Tensor> maxpool_fwd(torch::Tensor x, int64_t k, int64_t s,
     hipLaunchKernelGGL(maxpool_gen_fwd_kernel, dim3(blocks), dim3(256), 0,
                        at::cuda::getCurrentCUDAStream(), bf_ptr(x),
                        bf_ptr_mut(y), idx.data_ptr<uint8_t>(), N, H, W, C, OH,
-                       OW, (int)k, (int)s, (int)p);
+                       OW, (int)k, (int)s, (int)p,
+                       fdiv_make((unsigned)C), fdiv_make((unsigned)OW),
+                       fdiv_make((unsigned)OH));
     return {y, idx};
 }
 
@@ -3012,7 +3029,9 @@ torch::Tensor maxpool_bwd(torch::Tensor dy, torch::Tensor idx, int64_t H,
     hipLaunchKernelGGL(maxpool_gen_bwd_kernel, dim3(blocks), dim3(256), 0,
                        at::cuda::getCurrentCUDAStream(), bf_ptr(dy),
                        idx.data_ptr<uint8_t>(), dx32.data_ptr<float>(), N,
-                       (int)H, (int)W, C, OH, OW, (int)k, (int)s, (int)p);
+                       (int)H, (int)W, C, OH, OW, (int)k, (int)s, (int)p,
+                       fdiv_make((unsigned)C), fdiv_make((unsigned)OW),
+                       fdiv_make((unsigned)OH), fdiv_make((unsigned)k));
     return dx32.to(torch::kBFloat16);
 }
 
