@@ -2005,10 +2005,10 @@ __global__ void bn_apply_kernel(const unsigned short* __restrict__ x,
                                 const float* __restrict__ gamma,
                                 const float* __restrict__ beta,
                                 unsigned short* __restrict__ y, int64_t total,
-                                int C, int relu) {
+                                int C, int relu, FastDiv fC) {
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
          i += (int64_t)gridDim.x * blockDim.x) {
-        int c = i % C;
+        int c = (int)((unsigned)i - fdiv((unsigned)i, fC) * C);
         float v = (bf2f(x[i]) - mean[c]) * invstd[c] * gamma[c] + beta[c];
         if (relu) v = v > 0.f ? v : 0.f;
         y[i] = f2bf(v);
@@ -2213,10 +2213,11 @@ __global__ void bn_dx_kernel(const unsigned short* __restrict__ dy,
                              const float* __restrict__ dgamma,
                              const float* __restrict__ dbeta,
                              unsigned short* __restrict__ dx, int64_t total,
-                             int C, int64_t M, int train) {
+                             int C, int64_t M, int train,
+                             FastDiv fC) {
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
          i += (int64_t)gridDim.x * blockDim.x) {
-        int c = i % C;
+        int c = (int)((unsigned)i - fdiv((unsigned)i, fC) * C);
         float g = bf_gated(dy, relu_y, i);
         float v;
         if (train) {
@@ -2297,12 +2298,13 @@ __global__ void maxpool_gen_bwd_kernel(const unsigned short* __restrict__ dy,
 
 __global__ void avgpool_global_fwd_kernel(const unsigned short* __restrict__ x,
                                           unsigned short* __restrict__ y,
-                                          int N, int HW, int C) {
+                                          int N, int HW, int C, FastDiv fC) {
     int64_t total = (int64_t)N * C;
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
          i += (int64_t)gridDim.x * blockDim.x) {
-        int c = i % C;
-        int n = i / C;
+        unsigned t = fdiv((unsigned)i, fC);
+        int c = (int)((unsigned)i - t * C);
+        int n = (int)t;
         float acc = 0.f;
         const unsigned short* base = x + (int64_t)n * HW * C + c;
         for (int r = 0; r < HW; ++r) acc += bf2f(base[(int64_t)r * C]);
@@ -2312,14 +2314,14 @@ __global__ void avgpool_global_fwd_kernel(const unsigned short* __restrict__ x,
 
 __global__ void avgpool_global_bwd_kernel(const unsigned short* __restrict__ dy,
                                           unsigned short* __restrict__ dx,
-                                          int N, int HW, int C) {
+                                          int N, int HW, int C, FastDiv fC, FastDiv fHW) {
     int64_t total = (int64_t)N * HW * C;
     float inv;
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
          i += (int64_t)gridDim.x * blockDim.x) {
-        int c = i % C;
-        int64_t t = i / C;
-        int n = t / HW;
+        unsigned t0 = fdiv((unsigned)i, fC);
+        int c = (int)((unsigned)i - t0 * C);
+        int n = (int)fdiv(t0, fHW);
         inv = 1.f / (float)HW;
         dx[i] = f2bf(bf2f(dy[(int64_t)n * C + c]) * inv);
     }
@@ -2942,7 +2944,7 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
                        bf_ptr(x), mean.data_ptr<float>(),
                        invstd.data_ptr<float>(), gamma.data_ptr<float>(),
                        beta.data_ptr<float>(), bf_ptr_mut(y), total, C,
-                       relu ? 1 : 0);
+                       relu ? 1 : 0, fdiv_make((unsigned)C));
     return {y, mean, invstd};
 }
 
@@ -2958,7 +2960,7 @@ torch::Tensor bn_apply(torch::Tensor x, torch::Tensor mean,
                        at::cuda::getCurrentCUDAStream(), bf_ptr(x),
                        mean.data_ptr<float>(), invstd.data_ptr<float>(),
                        gamma.data_ptr<float>(), beta.data_ptr<float>(),
-                       bf_ptr_mut(y), total, C, relu ? 1 : 0);
+                       bf_ptr_mut(y), total, C, relu ? 1 : 0, fdiv_make((unsigned)C));
     return y;
 }
 
@@ -2993,7 +2995,7 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
                        bf_ptr(dyc), bf_ptr(x), ry, mean.data_ptr<float>(),
                        invstd.data_ptr<float>(), gamma.data_ptr<float>(),
                        dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
-                       bf_ptr_mut(dx), total, C, M, train ? 1 : 0);
+                       bf_ptr_mut(dx), total, C, M, train ? 1 : 0, fdiv_make((unsigned)C));
     return {dx, dgamma, dbeta};
 }
 
@@ -3045,7 +3047,7 @@ torch::Tensor avgpool_global_fwd(torch::Tensor x) {
     int blocks = std::min<int64_t>(ceildiv(total, 256), 2048);
     hipLaunchKernelGGL(avgpool_global_fwd_kernel, dim3(blocks), dim3(256), 0,
                        at::cuda::getCurrentCUDAStream(), bf_ptr(x),
-                       bf_ptr_mut(y), N, HW, C);
+                       bf_ptr_mut(y), N, HW, C, fdiv_make((unsigned)C));
     return y;
 }
 
@@ -3057,7 +3059,9 @@ torch::Tensor avgpool_global_bwd(torch::Tensor dy, int64_t H, int64_t W) {
     int blocks = std::min<int64_t>(ceildiv(total, 256), 2048);
     hipLaunchKernelGGL(avgpool_global_bwd_kernel, dim3(blocks), dim3(256), 0,
                        at::cuda::getCurrentCUDAStream(), bf_ptr(dy),
-                       bf_ptr_mut(dx), N, (int)(H * W), C);
+                       bf_ptr_mut(dx), N, (int)(H * W), C,
+                       fdiv_make((unsigned)C),
+                       fdiv_make((unsigned)(H * W)));
     return dx;
 }
 
