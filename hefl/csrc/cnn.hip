@@ -2381,12 +2381,24 @@ __global__ void bias_grad_kernel(const unsigned short* __restrict__ dy,
 __global__ void sum_slabs_f32_kernel(const float* __restrict__ slab,
                                      float* __restrict__ out, int64_t total,
                                      int nslab) {
-    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
-         i += (int64_t)gridDim.x * blockDim.x) {
-        float a = 0.f;
-        for (int sl = 0; sl < nslab; ++sl) a += slab[sl * total + i];
-        out[i] = a;
+    // thread = (slab lane, elem): 16 lanes sum strided slabs in parallel,
+    // LDS tree-reduce (a serial slab loop left ~90% of the block idle for
+    // the small dw tensors this sums)
+    __shared__ float red[256];
+    const int lane = threadIdx.x & 15;
+    const int eloc = threadIdx.x >> 4;
+    const int64_t e = (int64_t)blockIdx.x * 16 + eloc;
+    float a = 0.f;
+    if (e < total)
+        for (int sl = lane; sl < nslab; sl += 16)
+            a += slab[(int64_t)sl * total + e];
+    red[threadIdx.x] = a;
+    __syncthreads();
+    for (int off = 8; off >= 1; off >>= 1) {
+        if (lane < off) red[threadIdx.x] += red[threadIdx.x + off];
+        __syncthreads();
     }
+    if (lane == 0 && e < total) out[e] = red[threadIdx.x];
 }
 
 inline int ceildiv(int64_t a, int64_t b) { return (int)((a + b - 1) / b); }
@@ -2617,9 +2629,8 @@ torch::Tensor conv2d_wgrad(torch::Tensor dy, torch::Tensor x, int64_t stride,
                            0, stream, bf_ptr(dy), bf_ptr(x),
                            slab.data_ptr<float>(), s, kc);
         hipLaunchKernelGGL(sum_slabs_f32_kernel,
-                           dim3((int)std::min<int64_t>((total + 255) / 256,
-                                                       1024)),
-                           dim3(256), 0, stream, slab.data_ptr<float>(),
+                           dim3((int)((total + 15) / 16)), dim3(256), 0,
+                           stream, slab.data_ptr<float>(),
                            dw.data_ptr<float>(), total, kc);
         return dw;
     }
