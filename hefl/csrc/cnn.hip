@@ -1673,6 +1673,22 @@ __global__ void linear_splitk_kernel(const unsigned short* __restrict__ x,
     const int row = m0 + sub;         // A row
     const int col = n0 + sub;         // B col (w row)
     f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    if ((K & 31) == 0 && m0 + 16 <= M && n0 + 16 <= N) {
+        // aligned full-tile fast path: hoisted bounds, unrolled so the
+        // compiler keeps several 16-B loads in flight ahead of the MFMAs
+        // (the checked loop serialized ~300 ns global latency per k-step).
+        // The guard is WAVE-UNIFORM (m0/n0, not row/col): MFMA is a
+        // wave-level op, so mixed fast/slow lanes would execute separate
+        // MFMAs with garbage operands in the inactive lanes.
+        const unsigned short* xr = x + (int64_t)row * K + half * 8;
+        const unsigned short* wr = w + (int64_t)col * K + half * 8;
+#pragma unroll 4
+        for (int k = k0; k < kend; k += 32) {
+            bf16x8 a = *reinterpret_cast<const bf16x8*>(xr + k);
+            bf16x8 b = *reinterpret_cast<const bf16x8*>(wr + k);
+            acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+        }
+    } else
     for (int k = k0; k < kend; k += 32) {
         bf16x8 a{}, b{};
         const int kk = k + half * 8;
