@@ -39,6 +39,8 @@ torch::Tensor linear_wgrad(torch::Tensor dy, torch::Tensor x);
 std::vector<torch::Tensor> maxpool2x2_fwd(torch::Tensor x);
 torch::Tensor synth_batch(torch::Tensor templates, torch::Tensor labels,
                           int64_t seed);
+torch::Tensor synth_batch_g(torch::Tensor templates, torch::Tensor labels,
+                            torch::Tensor seed_buf, int64_t salt);
 torch::Tensor maxpool2x2_bwd(torch::Tensor dy, torch::Tensor idx, int64_t H,
                              int64_t W);
 std::vector<torch::Tensor> softmax_xent_fwd(torch::Tensor logits,
@@ -116,6 +118,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("linear_wgrad", &linear_wgrad);
     m.def("maxpool2x2_fwd", &maxpool2x2_fwd);
     m.def("synth_batch", &synth_batch);
+    m.def("synth_batch_g", &synth_batch_g);
     m.def("maxpool2x2_bwd", &maxpool2x2_bwd);
     m.def("softmax_xent_fwd", &softmax_xent_fwd);
     m.def("softmax_xent_bwd", &softmax_xent_bwd);

@@ -1252,7 +1252,9 @@ __global__ void synth_batch_kernel(const float* __restrict__ T,
                                    const int64_t* __restrict__ lab,
                                    unsigned short* __restrict__ out,
                                    int64_t per_img, int64_t total,
-                                   unsigned long long seed, FastDiv fPer) {
+                                   unsigned long long seed, FastDiv fPer,
+                                   const long long* __restrict__ seed_buf) {
+    if (seed_buf) seed += (unsigned long long)seed_buf[0];  // graph-replay seed
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
          i += (int64_t)gridDim.x * blockDim.x) {
         const int64_t n = (int64_t)fdiv((unsigned)i, fPer);
@@ -2756,7 +2758,31 @@ torch::Tensor synth_batch(torch::Tensor templates, torch::Tensor labels,
                        labels.data_ptr<int64_t>(),
                        reinterpret_cast<unsigned short*>(out.data_ptr()),
                        per, total, (unsigned long long)seed,
-                       fdiv_make((unsigned)per));
+                       fdiv_make((unsigned)per), (const long long*)nullptr);
+    return out;
+}
+
+torch::Tensor synth_batch_g(torch::Tensor templates, torch::Tensor labels,
+                            torch::Tensor seed_buf, int64_t salt) {
+    // graph-capturable variant: base seed read from a device buffer the
+    // host rewrites before each replay; salt distinguishes captured calls
+    CHECK_GPU(templates);
+    TORCH_CHECK(templates.is_contiguous() && labels.is_contiguous());
+    const int64_t n = labels.size(0);
+    const int64_t per = templates.numel() / templates.size(0);
+    auto out = torch::empty({n, templates.size(1), templates.size(2),
+                             templates.size(3)},
+                            templates.options().dtype(torch::kBFloat16));
+    const int64_t total = n * per;
+    int blocks = (int)std::min<int64_t>((total + 255) / 256, 8192);
+    hipLaunchKernelGGL(synth_batch_kernel, dim3(blocks), dim3(256), 0,
+                       at::cuda::getCurrentCUDAStream(),
+                       templates.data_ptr<float>(),
+                       labels.data_ptr<int64_t>(),
+                       reinterpret_cast<unsigned short*>(out.data_ptr()),
+                       per, total, (unsigned long long)salt,
+                       fdiv_make((unsigned)per),
+                       (const long long*)seed_buf.data_ptr<int64_t>());
     return out;
 }
 

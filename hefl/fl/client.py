@@ -156,6 +156,14 @@ class LocalClient:
             return ent
         B = self.loader.batch_size
         n = self.loader.indices.numel()
+        # in-graph data staging: the captured graph gathers labels and
+        # generates the epoch's samples itself (synth kernel reads its seed
+        # from a device buffer the host rewrites before each replay), so
+        # per epoch the host does ONE small H2D copy + ONE replay. Falls
+        # back to host staging when an augment hook or non-bf16 dtype is in
+        # play (those paths go through the torch ops).
+        in_graph_data = (self.loader.augment is None
+                         and self.compute_dtype == torch.bfloat16)
         x0, y0 = self.dataset.batch(self.loader.indices)
         if self.loader.augment is not None:
             x0 = self.loader.augment(x0)
@@ -185,8 +193,16 @@ class LocalClient:
         shells = [self.opt.alloc_mt_shell() for _ in range(steps)]
         rows_per_step = []
         grads_hold = []
+        order_buf = torch.empty(n, dtype=torch.int64, device=self.device)
+        seed_buf = torch.zeros(1, dtype=torch.int64, device=self.device)
+        order_buf.copy_(self.loader.indices.to(self.device))
         g = torch.cuda.CUDAGraph()
+        import hefl
+        C = hefl.load_extension()
         with torch.cuda.graph(g, stream=side):
+            if in_graph_data:
+                Y = self.dataset.labels.index_select(0, order_buf)
+                X = C.synth_batch_g(self.dataset.templates, Y, seed_buf, 0)
             self.opt.prep_epoch(steps)
             for s, i in enumerate(range(0, n, B)):
                 self.opt.zero_grad()  # grads=None -> backward steals
@@ -200,13 +216,23 @@ class LocalClient:
         for shell, rows in zip(shells, rows_per_step):
             self.opt.fill_mt_shell(shell, rows)
         ent = {"graph": g, "X": X, "Y": Y, "mt": shells,
-               "grads": grads_hold, "steps": steps, "n": n}
+               "grads": grads_hold, "steps": steps, "n": n,
+               "order_buf": order_buf, "seed_buf": seed_buf,
+               "in_graph_data": in_graph_data, "seed_ctr": 0}
         self._ep_ent = ent
         return ent
 
     def _epoch_replay(self, ent) -> None:
         """Stage one (shuffled) epoch of data, then replay the epoch graph."""
         order = self.loader.epoch_order()
+        if ent["in_graph_data"]:
+            # data generation is captured: host work is two small copies
+            ent["order_buf"].copy_(order.to(self.device))
+            ent["seed_ctr"] += 1
+            ent["seed_buf"].fill_(self.dataset.seed * 0x20003
+                                  + ent["seed_ctr"])
+            ent["graph"].replay()
+            return
         x, y = self.dataset.batch(order)
         if self.loader.augment is not None:
             x = self.loader.augment(x)
