@@ -2121,6 +2121,27 @@ __global__ void bn_apply_kernel(const unsigned short* __restrict__ x,
                                 const float* __restrict__ beta,
                                 unsigned short* __restrict__ y, int64_t total,
                                 int C, int relu, FastDiv fC) {
+    if ((C & 7) == 0) {  // octet path: one 16-B load/store per thread.
+        // NOTE: the host passes fC built over C/8 (octet count) here.
+        const int64_t t8 = total >> 3;
+        const int noct = C >> 3;
+        for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+             i < t8; i += (int64_t)gridDim.x * blockDim.x) {
+            const int c0 = 8 * (int)((unsigned)i - fdiv((unsigned)i, fC) * noct);
+            u16x8 x8 = *reinterpret_cast<const u16x8*>(&x[i * 8]);
+            u16x8 y8;
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                const int c = c0 + j;
+                float v = (bf2f(x8[j]) - mean[c]) * invstd[c] * gamma[c] +
+                          beta[c];
+                if (relu) v = v > 0.f ? v : 0.f;
+                y8[j] = f2bf(v);
+            }
+            *reinterpret_cast<u16x8*>(&y[i * 8]) = y8;
+        }
+        return;
+    }
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
          i += (int64_t)gridDim.x * blockDim.x) {
         int c = (int)((unsigned)i - fdiv((unsigned)i, fC) * C);
@@ -2330,6 +2351,38 @@ __global__ void bn_dx_kernel(const unsigned short* __restrict__ dy,
                              unsigned short* __restrict__ dx, int64_t total,
                              int C, int64_t M, int train,
                              FastDiv fC) {
+    const float invM = 1.f / (float)M;
+    if ((C & 7) == 0) {  // octet path (fC built over C/8 by the host)
+        const int64_t t8 = total >> 3;
+        const int noct = C >> 3;
+        for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+             i < t8; i += (int64_t)gridDim.x * blockDim.x) {
+            const int c0 = 8 * (int)((unsigned)i - fdiv((unsigned)i, fC) * noct);
+            u16x8 g8 = *reinterpret_cast<const u16x8*>(&dy[i * 8]);
+            u16x8 x8 = *reinterpret_cast<const u16x8*>(&x[i * 8]);
+            u16x8 y8{};
+            if (relu_y) y8 = *reinterpret_cast<const u16x8*>(&relu_y[i * 8]);
+            u16x8 o8;
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                const int c = c0 + j;
+                float g = bf2f(g8[j]);
+                if (relu_y && ((y8[j] & 0x7fffu) == 0 || (y8[j] & 0x8000u)))
+                    g = 0.f;
+                float v;
+                if (train) {
+                    float xh = (bf2f(x8[j]) - mean[c]) * invstd[c];
+                    v = gamma[c] * invstd[c] *
+                        (g - dbeta[c] * invM - xh * dgamma[c] * invM);
+                } else {
+                    v = gamma[c] * invstd[c] * g;
+                }
+                o8[j] = f2bf(v);
+            }
+            *reinterpret_cast<u16x8*>(&dx[i * 8]) = o8;
+        }
+        return;
+    }
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
          i += (int64_t)gridDim.x * blockDim.x) {
         int c = (int)((unsigned)i - fdiv((unsigned)i, fC) * C);
@@ -2338,7 +2391,7 @@ __global__ void bn_dx_kernel(const unsigned short* __restrict__ dy,
         if (train) {
             float xh = (bf2f(x[i]) - mean[c]) * invstd[c];
             v = gamma[c] * invstd[c] *
-                (g - dbeta[c] / (float)M - xh * dgamma[c] / (float)M);
+                (g - dbeta[c] * invM - xh * dgamma[c] * invM);
         } else {
             v = gamma[c] * invstd[c] * g;  // frozen stats
         }
@@ -2449,8 +2502,21 @@ __global__ void avgpool_global_bwd_kernel(const unsigned short* __restrict__ dy,
 __global__ void add_relu_kernel(const unsigned short* __restrict__ a,
                                 const unsigned short* __restrict__ b,
                                 unsigned short* __restrict__ y, int64_t total) {
-    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
+    const int64_t t8 = total >> 3;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < t8;
          i += (int64_t)gridDim.x * blockDim.x) {
+        u16x8 a8 = *reinterpret_cast<const u16x8*>(&a[i * 8]);
+        u16x8 b8 = *reinterpret_cast<const u16x8*>(&b[i * 8]);
+        u16x8 y8;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            float v = bf2f(a8[j]) + bf2f(b8[j]);
+            y8[j] = f2bf(v > 0.f ? v : 0.f);
+        }
+        *reinterpret_cast<u16x8*>(&y[i * 8]) = y8;
+    }
+    for (int64_t i = (t8 << 3) + blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         i < total; i += (int64_t)gridDim.x * blockDim.x) {
         float v = bf2f(a[i]) + bf2f(b[i]);
         y[i] = f2bf(v > 0.f ? v : 0.f);
     }
@@ -2463,10 +2529,20 @@ __global__ void add_relu_kernel(const unsigned short* __restrict__ a,
 __global__ void relu_bwd_kernel(const unsigned short* __restrict__ dy,
                                 const unsigned short* __restrict__ y,
                                 unsigned short* __restrict__ dx, int64_t total) {
-    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
+    const int64_t t8 = total >> 3;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < t8;
          i += (int64_t)gridDim.x * blockDim.x) {
-        dx[i] = (y[i] & 0x7fffu) != 0 && !(y[i] & 0x8000u) ? dy[i] : 0;
+        u16x8 d8 = *reinterpret_cast<const u16x8*>(&dy[i * 8]);
+        u16x8 y8 = *reinterpret_cast<const u16x8*>(&y[i * 8]);
+        u16x8 o8;
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+            o8[j] = (y8[j] & 0x7fffu) != 0 && !(y8[j] & 0x8000u) ? d8[j] : 0;
+        *reinterpret_cast<u16x8*>(&dx[i * 8]) = o8;
     }
+    for (int64_t i = (t8 << 3) + blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         i < total; i += (int64_t)gridDim.x * blockDim.x)
+        dx[i] = (y[i] & 0x7fffu) != 0 && !(y[i] & 0x8000u) ? dy[i] : 0;
 }
 
 // db[k] = sum over rows of dy[., k]: 2-D grid (k x row-chunks), coalesced
@@ -3101,7 +3177,7 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
                        bf_ptr(x), mean.data_ptr<float>(),
                        invstd.data_ptr<float>(), gamma.data_ptr<float>(),
                        beta.data_ptr<float>(), bf_ptr_mut(y), total, C,
-                       relu ? 1 : 0, fdiv_make((unsigned)C));
+                       relu ? 1 : 0, fdiv_make((unsigned)((C & 7) == 0 ? C / 8 : C)));
     return {y, mean, invstd};
 }
 
@@ -3117,7 +3193,7 @@ torch::Tensor bn_apply(torch::Tensor x, torch::Tensor mean,
                        at::cuda::getCurrentCUDAStream(), bf_ptr(x),
                        mean.data_ptr<float>(), invstd.data_ptr<float>(),
                        gamma.data_ptr<float>(), beta.data_ptr<float>(),
-                       bf_ptr_mut(y), total, C, relu ? 1 : 0, fdiv_make((unsigned)C));
+                       bf_ptr_mut(y), total, C, relu ? 1 : 0, fdiv_make((unsigned)((C & 7) == 0 ? C / 8 : C)));
     return y;
 }
 
@@ -3152,7 +3228,7 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
                        bf_ptr(dyc), bf_ptr(x), ry, mean.data_ptr<float>(),
                        invstd.data_ptr<float>(), gamma.data_ptr<float>(),
                        dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
-                       bf_ptr_mut(dx), total, C, M, train ? 1 : 0, fdiv_make((unsigned)C));
+                       bf_ptr_mut(dx), total, C, M, train ? 1 : 0, fdiv_make((unsigned)((C & 7) == 0 ? C / 8 : C)));
     return {dx, dgamma, dbeta};
 }
 
