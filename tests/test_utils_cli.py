@@ -30,3 +30,32 @@ def test_cli_runs_tiny_experiment():
     d = json.loads(last)
     assert d["clients"] == 2 and d["encrypted"] is False
     assert 0.0 <= d["metrics"]["accuracy"] <= 1.0
+
+
+def test_bench_json_contract(tmp_path):
+    """bench.py is the driver's measurement contract: run it tiny on CPU and
+    validate the single JSON line it prints (keys, metric name, weak
+    scaling, vs_baseline arithmetic)."""
+    import json
+    import os
+    import subprocess
+    import sys
+
+    repo = os.path.join(os.path.dirname(__file__), "..")
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--steps", "1", "--warmup", "0",
+         "--local-epochs", "1"],
+        cwd=repo, capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = out.stdout.strip().splitlines()[-1]
+    d = json.loads(line)
+    assert d["metric"] == "fl_rounds_per_sec"
+    assert d["unit"] == "rounds/s" and d["higher_is_better"] is True
+    assert d["scaling"] == "weak" and d["n_gpus"] == 1
+    assert d["steps"] == 1 and d["warmup"] == 0
+    assert d["data"] == "synthetic"
+    assert abs(d["vs_baseline"] - d["value"] / 1.52e-4) < 1e-6
+    assert abs(d["ms_per_step"] - 1000.0 / d["value"]) < 1e-6
+    cfg = d["config"]
+    assert cfg["model"] == "cnn2" and cfg["samples_per_client"] == 720
+    assert cfg["he"]["scheme"] == "CKKS" and cfg["he"]["encrypted"] is True
