@@ -54,9 +54,6 @@ void fused_adam(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                 double bc1, double bc2);
 void adam_prep(torch::Tensor step, torch::Tensor sched, torch::Tensor hyper,
                double b1, double b2);
-void fused_adam_sched(torch::Tensor p, torch::Tensor g, torch::Tensor m,
-                      torch::Tensor v, torch::Tensor sched, double b1,
-                      double b2, double eps);
 torch::Tensor relu_bwd(torch::Tensor dy, torch::Tensor y);
 std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
                                   torch::Tensor beta, torch::Tensor rmean,
@@ -91,8 +88,6 @@ void unpack_mt(torch::Tensor flat, torch::Tensor meta, torch::Tensor ptrs,
 void fused_adam_mt(torch::Tensor meta, torch::Tensor ptrs, torch::Tensor sizes,
                    int64_t nchunks, torch::Tensor sched, double b1, double b2,
                    double eps, int64_t zero_g, int64_t sched_off);
-void zero_grads_mt(torch::Tensor meta, torch::Tensor ptrs, torch::Tensor sizes,
-                   int64_t nchunks);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "hefl gfx950 (MI355X/CDNA4) HIP kernels";
@@ -124,7 +119,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("softmax_xent_bwd", &softmax_xent_bwd);
     m.def("fused_adam", &fused_adam);
     m.def("adam_prep", &adam_prep);
-    m.def("fused_adam_sched", &fused_adam_sched);
     m.def("relu_bwd", &relu_bwd);
     m.def("bn_fwd", &bn_fwd);
     m.def("bn_apply", &bn_apply);
@@ -141,5 +135,4 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("pack_mt", &pack_mt);
     m.def("unpack_mt", &unpack_mt);
     m.def("adam_prep_epoch", &adam_prep_epoch);
-    m.def("zero_grads_mt", &zero_grads_mt);
 }

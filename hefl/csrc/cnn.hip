@@ -1553,24 +1553,6 @@ __global__ void pool_relu_bias_bwd_kernel(
     }
 }
 
-__global__ void maxpool_bwd_kernel(const unsigned short* __restrict__ dy,
-                                   const uint8_t* __restrict__ idx,
-                                   unsigned short* __restrict__ dx, int N,
-                                   int H, int W, int C, int OH, int OW) {
-    int64_t total = (int64_t)N * OH * OW * C;
-    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
-         i += (int64_t)gridDim.x * blockDim.x) {
-        int c = i % C;
-        int64_t t = i / C;
-        int ow = t % OW;
-        t /= OW;
-        int oh = t % OH;
-        int n = t / OH;
-        int d = idx[i];
-        int ih = oh * 2 + (d >> 1), iw = ow * 2 + (d & 1);
-        dx[(((int64_t)n * H + ih) * W + iw) * C + c] = dy[i];
-    }
-}
 
 // ---------------------------------------------------------------------------
 // Fused softmax + categorical cross-entropy (mean), one wave per row.
@@ -1730,23 +1712,6 @@ __global__ void adam_prep_epoch_kernel(int64_t* __restrict__ step,
     }
 }
 
-__global__ void fused_adam_sched_kernel(float* __restrict__ p,
-                                        const float* __restrict__ g,
-                                        float* __restrict__ m,
-                                        float* __restrict__ v, int64_t total,
-                                        const float* __restrict__ sched,
-                                        float b1, float b2, float eps) {
-    const float lr = sched[0], bc1 = sched[1], bc2 = sched[2];
-    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
-         i += (int64_t)gridDim.x * blockDim.x) {
-        float gi = g[i];
-        float mi = b1 * m[i] + (1.f - b1) * gi;
-        float vi = b2 * v[i] + (1.f - b2) * gi * gi;
-        m[i] = mi;
-        v[i] = vi;
-        p[i] -= lr * (mi / bc1) / (sqrtf(vi / bc2) + eps);
-    }
-}
 
 // ---------------------------------------------------------------------------
 // Skinny linear fwd (M ~ batch 32): one wave per (16x16 tile, K-chunk),
@@ -1935,16 +1900,6 @@ __global__ void unpack_mt_kernel(const float* __restrict__ flat,
     }
 }
 
-__global__ void zero_mt_kernel(const int64_t* __restrict__ meta,
-                               const int64_t* __restrict__ ptrs,
-                               const int64_t* __restrict__ sizes) {
-    const int c = blockIdx.x;
-    const int t = (int)meta[c * 2];
-    const int64_t off = meta[c * 2 + 1];
-    float* g = reinterpret_cast<float*>(ptrs[t * 4 + 1]);
-    const int64_t n = min(off + (int64_t)MT_CHUNK, sizes[t]);
-    for (int64_t i = off + threadIdx.x; i < n; i += blockDim.x) g[i] = 0.f;
-}
 
 // ---------------------------------------------------------------------------
 // BatchNorm (NHWC, per-channel over N*H*W) — ResNet-18 (config #5) support.
@@ -3122,19 +3077,6 @@ void adam_prep(torch::Tensor step, torch::Tensor sched, torch::Tensor hyper,
                        hyper.data_ptr<float>(), (float)b1, (float)b2);
 }
 
-void fused_adam_sched(torch::Tensor p, torch::Tensor g, torch::Tensor m,
-                      torch::Tensor v, torch::Tensor sched, double b1,
-                      double b2, double eps) {
-    CHECK_GPU(p);
-    auto gc = g.contiguous();
-    int64_t total = p.numel();
-    int blocks = std::min<int64_t>(ceildiv(total, 256), 2048);
-    hipLaunchKernelGGL(fused_adam_sched_kernel, dim3(blocks), dim3(256), 0,
-                       at::cuda::getCurrentCUDAStream(), p.data_ptr<float>(),
-                       gc.data_ptr<float>(), m.data_ptr<float>(),
-                       v.data_ptr<float>(), total, sched.data_ptr<float>(),
-                       (float)b1, (float)b2, (float)eps);
-}
 
 torch::Tensor relu_bwd(torch::Tensor dy, torch::Tensor y) {
     CHECK_GPU(dy);
@@ -3367,11 +3309,3 @@ void unpack_mt(torch::Tensor flat, torch::Tensor meta, torch::Tensor ptrs,
                        sizes.data_ptr<int64_t>(), offs.data_ptr<int64_t>());
 }
 
-void zero_grads_mt(torch::Tensor meta, torch::Tensor ptrs, torch::Tensor sizes,
-                   int64_t nchunks) {
-    CHECK_GPU(meta);
-    hipLaunchKernelGGL(zero_mt_kernel, dim3((unsigned)nchunks), dim3(256), 0,
-                       at::cuda::getCurrentCUDAStream(),
-                       meta.data_ptr<int64_t>(), ptrs.data_ptr<int64_t>(),
-                       sizes.data_ptr<int64_t>());
-}
