@@ -145,3 +145,60 @@ def test_insecure_params_warn():
         CKKSContext(HEConfig(m=1024, scale_bits=40, q_bits=(60, 40, 60),
                              seed=0))  # logQP ~ 220 >> 27
         assert any("128-bit" in str(x.message) for x in w)
+
+
+def test_rescale_level_exhaustion_raises(ctx, keys):
+    """A ciphertext at the last level cannot rescale again — the engine must
+    say so instead of producing garbage (the reference's Pyfhel would abort
+    deep in SEAL)."""
+    v = np.random.default_rng(7).normal(size=ctx.slots)
+    ct = ctx.encrypt(ctx.encode(v), keys.pk)
+    r = ctx.rescale(ctx.mul_scalar(ct, 0.5))  # L=2 -> L=1
+    assert r.level == 1
+    with pytest.raises(Exception):
+        ctx.rescale(ctx.mul_scalar(r, 0.5))   # no level left
+
+
+def test_additive_noise_growth_bounded(ctx, keys):
+    """Summing 8 independently-encrypted copies (the lazy all-reduce bound)
+    keeps decryption error << one slot-value ulp."""
+    rng = np.random.default_rng(8)
+    v = rng.normal(size=ctx.slots)
+    acc = None
+    for _ in range(8):
+        ct = ctx.encrypt(ctx.encode(v), keys.pk)
+        acc = ct if acc is None else ctx.add(acc, ct)
+    out = ctx.decode(ctx.decrypt(acc, keys.sk), ctx.slots)
+    assert np.abs(out - 8 * v).max() < 1e-3
+
+
+def test_ciphertexts_differ_across_ranks_same_plaintext():
+    """After reseed(rank), two ranks encrypting the same vector under the
+    same shared-seed keys must produce DIFFERENT ciphertexts (client
+    privacy), while both still decrypt correctly."""
+    cfgs = [HEConfig(m=64, scale_bits=30, q_bits=(50, 30), seed=11)
+            for _ in range(2)]
+    ctxs = [CKKSContext(c) for c in cfgs]
+    kps = [c.keygen() for c in ctxs]
+    assert torch.equal(kps[0].pk, kps[1].pk)  # shared-seed keygen
+    for rank, c in enumerate(ctxs):
+        c.reseed(rank)
+    v = np.random.default_rng(9).normal(size=ctxs[0].slots)
+    c0 = ctxs[0].encrypt(ctxs[0].encode(v), kps[0].pk)
+    c1 = ctxs[1].encrypt(ctxs[1].encode(v), kps[1].pk)
+    assert not torch.equal(c0.data, c1.data)
+    for c, ct, kp in ((ctxs[0], c0, kps[0]), (ctxs[1], c1, kps[1])):
+        out = c.decode(c.decrypt(ct, kp.sk), c.slots)
+        assert np.abs(out - v).max() < 1e-4
+
+
+def test_encrypt_tensor_empty_and_exact_slot_fit(ctx, keys):
+    """Boundary shapes: exactly slots-many values (no padding ct) and a
+    1-element vector."""
+    vec = torch.randn(ctx.slots)
+    ct = ctx.encrypt_tensor(vec, keys.pk)
+    assert ct.data.shape[0] == 1
+    assert (ctx.decrypt_tensor(ct, keys.sk) - vec).abs().max() < 1e-4
+    one = torch.tensor([2.5])
+    ct1 = ctx.encrypt_tensor(one, keys.pk)
+    assert abs(float(ctx.decrypt_tensor(ct1, keys.sk)[0]) - 2.5) < 1e-4
