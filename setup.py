@@ -6,7 +6,7 @@ travels with the repo snapshot to the GPU box).
 """
 import os
 
-from setuptools import setup
+from setuptools import find_packages, setup
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
@@ -15,7 +15,7 @@ from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E40
 setup(
     name="hefl",
     version="0.1.0",
-    packages=["hefl"],
+    packages=find_packages(include=["hefl", "hefl.*"]),
     ext_modules=[
         CUDAExtension(
             name="hefl._C",
