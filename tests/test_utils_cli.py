@@ -59,3 +59,28 @@ def test_bench_json_contract(tmp_path):
     cfg = d["config"]
     assert cfg["model"] == "cnn2" and cfg["samples_per_client"] == 720
     assert cfg["he"]["scheme"] == "CKKS" and cfg["he"]["encrypted"] is True
+
+
+def test_bench_two_rank_driver_invocation():
+    """The driver launches bench.py via torch.distributed.run for the
+    scaling bench; exercise that exact invocation at world=2 on CPU/gloo
+    and check the rank-0 JSON (weak scaling: n_gpus=2, global batch 2x)."""
+    import json
+    import os
+    import subprocess
+    import sys
+
+    repo = os.path.join(os.path.dirname(__file__), "..")
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29781", "bench.py", "--gpus", "2",
+         "--steps", "1", "--warmup", "0", "--local-epochs", "1"],
+        cwd=repo, capture_output=True, text=True, timeout=600,
+        env=dict(os.environ, MASTER_ADDR="127.0.0.1"))
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith("{")][-1]
+    d = json.loads(line)
+    assert d["n_gpus"] == 2 and d["scaling"] == "weak"
+    assert d["config"]["global_batch"] == 64
+    assert d["config"]["parallelism"].startswith("fl-dp2")
