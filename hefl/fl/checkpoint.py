@@ -77,9 +77,18 @@ def load_round_state(path: str, model: torch.nn.Module, optimizer, he=None):
     state = torch.load(path, weights_only=False)
     model.load_state_dict(state["model"])
     opt = state["optimizer"]
-    optimizer.step_count = opt["step"]
-    optimizer.lr = opt["lr"]
-    optimizer.decay = opt["decay"]
+    # go through the optimizer API: a graphed client's captured kernels read
+    # DEVICE-side step/lr buffers, which raw attribute writes would miss
+    if hasattr(optimizer, "set_step"):
+        optimizer.set_step(opt["step"])
+        optimizer.set_lr(opt["lr"])
+        optimizer.decay = opt["decay"]
+        if hasattr(optimizer, "_hyper"):
+            optimizer._hyper[1] = opt["decay"]
+    else:
+        optimizer.step_count = opt["step"]
+        optimizer.lr = opt["lr"]
+        optimizer.decay = opt["decay"]
     for dst, src in zip(optimizer.m, opt["m"]):
         dst.copy_(src.to(dst.device))
     for dst, src in zip(optimizer.v, opt["v"]):

@@ -73,6 +73,13 @@ class FusedAdam:
         if hasattr(self, "_hyper"):
             self._hyper[0] = lr  # reaches captured graphs without re-capture
 
+    def set_step(self, step: int):
+        """Set the step counter, keeping the device-side counter captured
+        graphs read in sync (checkpoint resume on a graphed client)."""
+        self.step_count = int(step)
+        if hasattr(self, "_step_t"):
+            self._step_t.fill_(int(step))
+
     def build_mt_table(self):
         """Multi-tensor chunk table from the CURRENT .grad pointers — built
         per captured graph (each capture's backward steals fresh pooled grad
@@ -199,7 +206,13 @@ class FusedAdam:
                 "lr": self.lr, "decay": self.decay}
 
     def load_state_dict(self, sd):
-        self.step_count = sd["step"]
+        self.set_step(sd["step"])
+        if "lr" in sd:
+            self.set_lr(sd["lr"])
+        if "decay" in sd:
+            self.decay = sd["decay"]
+            if hasattr(self, "_hyper"):
+                self._hyper[1] = sd["decay"]
         for dst, src in zip(self.m, sd["m"]):
             dst.copy_(src)
         for dst, src in zip(self.v, sd["v"]):

@@ -92,3 +92,32 @@ def test_presets_match_baseline_json():
     assert cfg.train.local_epochs == 10 and cfg.train.batch_size == 32
     assert cfg.fl.samples_per_client == 720
     assert "metric" in base or "benchmarks" in base or len(base) > 0
+
+
+def test_optimizer_resume_syncs_device_schedule_buffers():
+    """Checkpoint resume on a graphed client must reach the DEVICE-side
+    step/lr buffers the captured Adam kernels read (regression: raw
+    attribute writes left them stale)."""
+    m = CNN2((28, 28, 1), 2, seed=0)
+    opt = FusedAdam(m.parameters(), lr=1e-3, decay=1e-4)
+    opt.prepare_graph_state(torch.device("cpu"))  # creates _step_t/_hyper
+    opt.step_count = 7
+    sd = {"step": 41, "lr": 5e-4, "decay": 2e-4,
+          "m": [t.clone() for t in opt.m], "v": [t.clone() for t in opt.v]}
+    opt.load_state_dict(sd)
+    assert opt.step_count == 41 and int(opt._step_t.item()) == 41
+    assert abs(float(opt._hyper[0]) - 5e-4) < 1e-9
+    assert abs(float(opt._hyper[1]) - 2e-4) < 1e-9
+    # checkpoint round-trip path hits the same buffers
+    from hefl.fl.checkpoint import load_round_state, save_round_state
+    import tempfile, os
+    opt.set_step(99)
+    opt.set_lr(3e-4)
+    with tempfile.TemporaryDirectory() as d:
+        p = os.path.join(d, "r.pt")
+        save_round_state(p, m, opt, round_idx=1)
+        opt2 = FusedAdam(CNN2((28, 28, 1), 2, seed=1).parameters(), lr=9e-9)
+        opt2.prepare_graph_state(torch.device("cpu"))
+        load_round_state(p, CNN2((28, 28, 1), 2, seed=1), opt2)
+        assert int(opt2._step_t.item()) == 99
+        assert abs(float(opt2._hyper[0]) - 3e-4) < 1e-9
