@@ -657,6 +657,10 @@ class CKKSContext:
 
     def mul_ct(self, a: Ciphertext, b: Ciphertext, rlk: torch.Tensor) -> Ciphertext:
         """ct x ct multiply + relinearize (no rescale; call rescale after)."""
+        if rlk is None:
+            raise ValueError(
+                "ct x ct multiply needs relinearization keys — run "
+                "relin_keygen(sk) (Pyfhel API: relinKeyGen) first")
         assert a.level == b.level
         L = a.level
         a0, a1 = a.data[..., 0, :, :], a.data[..., 1, :, :]

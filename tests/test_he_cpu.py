@@ -202,3 +202,29 @@ def test_encrypt_tensor_empty_and_exact_slot_fit(ctx, keys):
     one = torch.tensor([2.5])
     ct1 = ctx.encrypt_tensor(one, keys.pk)
     assert abs(float(ctx.decrypt_tensor(ct1, keys.sk)[0]) - 2.5) < 1e-4
+
+
+def test_error_paths_raise_clearly():
+    """Misuse raises descriptive errors instead of crashing deep in the
+    engine (the reference's Pyfhel aborts inside SEAL on most of these)."""
+    from hefl.he.pyfhel_compat import Pyfhel
+
+    he = Pyfhel()
+    with pytest.raises(ValueError, match="public key"):
+        he.encryptFrac(1.0)
+    he.contextGen(m=64, q_bits=(50, 30), scale_bits=30, seed=0)
+    with pytest.raises(ValueError, match="public key"):
+        he.encryptFrac(1.0)
+    with pytest.raises(ValueError, match="keyGen"):
+        he.relinKeyGen()
+
+    cfg = HEConfig(m=64, scale_bits=26, q_bits=(55, 26, 26), seed=0)
+    c = CKKSContext(cfg)
+    kp = c.keygen()
+    ct = c.encrypt(c.encode(np.ones(32)), kp.pk)
+    with pytest.raises(ValueError, match="relinearization"):
+        c.mul_ct(ct, ct, None)
+
+    from hefl.config import preset
+    with pytest.raises(KeyError, match="unknown preset"):
+        preset("nope")
