@@ -214,6 +214,9 @@ class CKKSContext:
         # chain primes + ONE special prime for hybrid key-switching (the
         # relinearization path, configs #3/#5); special prime is never part
         # of a ciphertext level, only of relin keys.
+        if not cfg.q_bits:
+            raise ValueError("q_bits must name at least one chain prime "
+                             "(L >= 1); an empty chain cannot encrypt")
         allp = gen_prime_chain(self.n, tuple(cfg.q_bits) + (60,))
         self.primes: List[int] = allp[:-1]
         self.special: int = allp[-1]

@@ -228,3 +228,7 @@ def test_error_paths_raise_clearly():
     from hefl.config import preset
     with pytest.raises(KeyError, match="unknown preset"):
         preset("nope")
+    with pytest.raises(ValueError, match="power of two"):
+        CKKSContext(HEConfig(m=100, scale_bits=30, q_bits=(50, 30), seed=0))
+    with pytest.raises(ValueError, match="at least one"):
+        CKKSContext(HEConfig(m=64, scale_bits=30, q_bits=(), seed=0))
