@@ -100,3 +100,16 @@ def test_batched_tensor_api():
     ct = he.encrypt_tensor(vec)
     out = he.decrypt_tensor(ct)
     assert (out - vec).abs().max().item() < 1e-4
+
+
+def test_readme_migration_example():
+    """The README migration snippet, executed verbatim."""
+    from hefl.he import Pyfhel
+
+    HE = Pyfhel()
+    HE.contextGen(p=65537, m=2048, sec=128)
+    HE.keyGen()
+    c = HE.encryptFrac(0.5)
+    c2 = c + c
+    c3 = c2 * 0.5
+    assert abs(HE.decryptFrac(c3) - 0.5) < 1e-3
