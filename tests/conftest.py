@@ -19,3 +19,12 @@ def pytest_collection_modifyitems(config, items):
     for item in items:
         if "gpu" in item.keywords:
             item.add_marker(skip_gpu)
+
+
+def free_port() -> int:
+    """OS-assigned free TCP port for torch.distributed rendezvous — fixed
+    port numbers collide when multiprocess tests run back-to-back."""
+    import socket
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]

@@ -83,12 +83,14 @@ def _run(target, world, port):
 
 @pytest.mark.timeout(300)
 def test_two_process_encrypted_fedavg_direct():
-    results = _run(_worker_direct, 2, 29611)
+    from conftest import free_port
+    results = _run(_worker_direct, 2, free_port())
     for rank, err in results:
         assert err < 1e-3, (rank, err)
 
 
 @pytest.mark.timeout(300)
 def test_two_process_encrypted_fl_round():
-    results = _run(_worker, 2, 29621)
+    from conftest import free_port
+    results = _run(_worker, 2, free_port())
     assert all(same for _, same, _ in results), results

@@ -48,7 +48,8 @@ def _worker(rank, world, port, q):
 def test_two_process_plaintext_fedavg():
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    port = 29601
+    from conftest import free_port
+    port = free_port()
     procs = [ctx.Process(target=_worker, args=(r, 2, port, q)) for r in range(2)]
     for p in procs:
         p.start()
@@ -98,7 +99,9 @@ def test_distributed_matches_sequential():
 
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    procs = [ctx.Process(target=_equiv_worker, args=(r, 2, 29611, q))
+    from conftest import free_port
+    port = free_port()
+    procs = [ctx.Process(target=_equiv_worker, args=(r, 2, port, q))
              for r in range(2)]
     for p in procs:
         p.start()

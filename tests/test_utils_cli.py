@@ -74,7 +74,7 @@ def test_bench_two_rank_driver_invocation():
     out = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-         "--master-port", "29781", "bench.py", "--gpus", "2",
+         "--master-port", str(__import__("conftest").free_port()), "bench.py", "--gpus", "2",
          "--steps", "1", "--warmup", "0", "--local-epochs", "1"],
         cwd=repo, capture_output=True, text=True, timeout=600,
         env=dict(os.environ, MASTER_ADDR="127.0.0.1"))
