@@ -10,6 +10,43 @@
 // fp32 master copy in Python), fp32 accumulation via
 // v_mfma_f32_16x16x32_bf16 (one 16x16 output fragment per wave,
 // 4 waves per workgroup covering 64 output rows).
+//
+// Kernel index (in file order):
+//   FastDiv / fdiv                 magic-multiply division (all index decode)
+//   conv_fwd_kernel / tile_mfma    gather-form implicit-GEMM conv fwd
+//   glds16                         global_load_lds 16-B DMA helper
+//   conv_fwd_glds_kernel           fwd with LDS-DMA 3-buffer counted-vmcnt
+//                                  pipeline (raw s_barrier, 1 stage in flight)
+//   conv_dgrad_kernel/_glds_       dgrad, K reordered to (r,s,ko) for
+//                                  contiguous dy gathers
+//   conv_wgrad_kernel/_glds_       wgrad, pixel-major tiles, split-K slabs
+//   conv_wgrad_small_kernel        Kout<=64 & RSC<=16 first-layer wgrad:
+//                                  waves split over pixels, LDS reduce
+//   sum_slabs_f32_kernel           lane-parallel split-K slab reduction
+//   maxpool_fwd_kernel (2x2)       pool fwd + argmax byte
+//   maxpool2x2_bwd_gather_kernel   gather-form pool bwd (no scatter/fill)
+//   pool_relu_gate / pool_relu_bias_bwd_kernel
+//                                  FUSED conv->relu->pool backward: pool
+//                                  gather + ReLU mask + bias grad, bf16x8
+//   relu_bias_bwd_kernel           fused ReLU-mask + bias grad (fc layers)
+//   linear_splitk_kernel           skinny-M wave GEMM, direct-from-global,
+//                                  unrolled wave-uniform fast path
+//   linear_epilogue_kernel         slab sum + bias + relu + bf16 store
+//   gemm_kernel<AT,BT,...>         general MFMA GEMM (fc d/wgrad)
+//   softmax_xent_fwd/bwd_kernel    fused softmax-CE with in-kernel
+//                                  loss/accuracy accumulators
+//   adam_prep(_epoch)_kernel       device-side Keras-decay schedule
+//   fused_adam(_mt)_kernel         fused Adam (+bf16 shadow refresh);
+//                                  _mt = one launch over all params
+//   pack_mt/unpack_mt_kernel       FedAvg weight vector pack/unpack
+//   bn_partial/bn_finalize/bn_apply/bn_bwd_partial/bn_bwd_finalize/bn_dx
+//                                  two-stage BatchNorm, bf16x8 octet paths,
+//                                  ReLU gate fused into backward
+//   synth_batch_kernel             one-pass synthetic data generation
+//   maxpool_gen_fwd/bwd_kernel     generic k/s/p pooling (ResNet stem)
+//   avgpool_global_fwd/bwd_kernel  global average pool
+//   add_relu_kernel                residual add + ReLU (bf16x8)
+//   relu_bwd_kernel, bias_grad_kernel, cast_f32_bf16_kernel   elementwise
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>

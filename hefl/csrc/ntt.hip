@@ -7,6 +7,19 @@
 // with merged psi powers in bit-reversed table order (tables built in
 // Python, uploaded once per context).
 //
+// Kernel index (in file order):
+//   ntt_lds_kernel / intt_lds_kernel         single-q LDS-resident stages
+//   ntt_global_stage_kernel / intt_...       strided stages for n=2^15
+//   *_ml variants                            fused multi-limb (limb = row%L),
+//                                            radix-4 in-LDS (2 butterfly
+//                                            levels per barrier)
+//   ntt/intt_global_radix4_ml_kernel         fused radix-4 global stage pair
+//   modmul/modadd/modsub/modmul_scalar       Barrett/Shoup pointwise ops
+//   modreduce_ kernel                        lazy-sum -> [0,q) (post
+//                                            all-reduce), branchless
+//   cbd21_kernel                             centered-binomial noise from
+//                                            one 64-bit draw (popcounts)
+//
 // Structure per row of length n (n = 2^6 .. 2^15):
 //  * stages whose butterfly span exceeds NBLK run in a strided global-memory
 //    kernel (only n=2^15 needs any on MI355X with NBLK=8192);
