@@ -23,7 +23,12 @@ class SyntheticMedicalImages:
 
     def __init__(self, n_samples: int, in_shape: Tuple[int, int, int],
                  n_classes: int, seed: int = 0, device: str = "cpu",
-                 dtype: torch.dtype = torch.float32):
+                 dtype: torch.dtype = torch.float32,
+                 template_seed: int = None):
+        """`template_seed` pins the per-class templates independently of
+        `seed`: a held-out TEST set must share the train templates (same
+        classification task) while drawing disjoint labels and noise —
+        pass template_seed=<train seed>, seed=<train seed + offset>."""
         self.n_samples = int(n_samples)
         self.H, self.W, self.C = in_shape
         self.n_classes = int(n_classes)
@@ -33,12 +38,16 @@ class SyntheticMedicalImages:
         # Per-class signal template: fixed low-frequency pattern per class so a
         # CNN can actually learn (accuracy-parity validation mirrors the
         # reference's end-to-end statistical check).
-        g = torch.Generator(device="cpu").manual_seed(self.seed ^ 0x5EED)
+        g = torch.Generator(device="cpu").manual_seed(
+            (self.seed if template_seed is None else int(template_seed))
+            ^ 0x5EED)
         base = torch.randn(self.n_classes, 8, 8, self.C, generator=g)
         self.templates = torch.nn.functional.interpolate(
             base.permute(0, 3, 1, 2), size=(self.H, self.W), mode="bilinear",
             align_corners=False).permute(0, 2, 3, 1).contiguous()
         self.templates = self.templates.to(self.device)
+        if template_seed is not None:
+            g = torch.Generator(device="cpu").manual_seed(self.seed ^ 0x1AB315)
         self.labels = torch.randint(0, self.n_classes, (self.n_samples,),
                                     generator=g).to(self.device)
 

@@ -49,8 +49,16 @@ def load_model_weights(model: torch.nn.Module, path: str):
 
 
 def save_round_state(path: str, model: torch.nn.Module, optimizer,
-                     round_idx: int, he=None, extra: Optional[dict] = None):
-    """One durable artifact per round: enough to resume the federation."""
+                     round_idx: int, he=None, extra: Optional[dict] = None,
+                     private_path: Optional[str] = None):
+    """One durable artifact per round: enough to resume the federation.
+
+    Key separation mirrors the reference's publickey/privatekey split
+    (notebook cell 1, FLPyfhelin.py:251-261): the round file carries only
+    PUBLIC HE material (context + pk); the secret key goes to a separate
+    private artifact (default `<path>.private`) that only the decrypting
+    party needs — so a round checkpoint can be shared with the aggregation
+    server without leaking sk."""
     os.makedirs(os.path.dirname(path) or ".", exist_ok=True)
     state = {
         "round": round_idx,
@@ -68,12 +76,16 @@ def save_round_state(path: str, model: torch.nn.Module, optimizer,
         state["he"] = {
             "context": he.to_bytes_context(),
             "public_key": he.to_bytes_publicKey(),
-            "secret_key": he.to_bytes_secretKey(),
         }
+        if he._sk is not None:
+            torch.save({"context": he.to_bytes_context(),
+                        "secret_key": he.to_bytes_secretKey()},
+                       private_path or path + ".private")
     torch.save(state, path)
 
 
-def load_round_state(path: str, model: torch.nn.Module, optimizer, he=None):
+def load_round_state(path: str, model: torch.nn.Module, optimizer, he=None,
+                     private_path: Optional[str] = None):
     state = torch.load(path, weights_only=False)
     model.load_state_dict(state["model"])
     opt = state["optimizer"]
@@ -96,5 +108,11 @@ def load_round_state(path: str, model: torch.nn.Module, optimizer, he=None):
     if he is not None and "he" in state:
         he.from_bytes_context(state["he"]["context"])
         he.from_bytes_publicKey(state["he"]["public_key"])
-        he.from_bytes_secretKey(state["he"]["secret_key"])
+        if "secret_key" in state["he"]:  # legacy bundled-sk round files
+            he.from_bytes_secretKey(state["he"]["secret_key"])
+        else:
+            sk_file = private_path or path + ".private"
+            if os.path.exists(sk_file):
+                priv = torch.load(sk_file, weights_only=False)
+                he.from_bytes_secretKey(priv["secret_key"])
     return state["round"], state.get("extra", {})

@@ -50,15 +50,26 @@ class LocalClient:
                            and getattr(t, "hip_graphs", True))
         self._graphs = {}
         # Data: the full federated dataset is conceptually
-        # n_clients * samples_per_client samples; this client materializes its
-        # contiguous shard (reference sharding semantics, FLPyfhelin.py:75-78).
-        n_total = cfg.fl.n_clients * cfg.fl.samples_per_client
+        # n_clients * (train + val) samples; this client materializes its
+        # contiguous shard (reference sharding semantics, FLPyfhelin.py:75-78)
+        # and splits off the TRAILING val fraction (reference:
+        # validation_split=0.1 of an 800-sample shard -> 720 train / 80 val,
+        # FLPyfhelin.py:85-99). Training runs on the train subset only;
+        # val metrics feed the callbacks (fit(validation_data=...), :193).
+        per_client = cfg.fl.samples_per_client + cfg.fl.val_samples_per_client
+        n_total = cfg.fl.n_clients * per_client
         self.dataset = SyntheticMedicalImages(
             n_total, cfg.model.in_shape, cfg.model.n_classes,
             seed=cfg.fl.seed, device=device, dtype=self.compute_dtype)
         idx = shard_indices(n_total, client_id, cfg.fl.n_clients)
-        self.loader = ClientLoader(self.dataset, idx, t.batch_size,
+        n_val = cfg.fl.val_samples_per_client
+        train_idx = idx[: idx.numel() - n_val]
+        self.loader = ClientLoader(self.dataset, train_idx, t.batch_size,
                                    seed=cfg.fl.seed + client_id)
+        self.val_loader = (ClientLoader(self.dataset,
+                                        idx[idx.numel() - n_val:],
+                                        t.batch_size, shuffle=False)
+                           if n_val else None)
 
     def train_step(self, x: torch.Tensor, y: torch.Tensor):
         if self.use_graphs:
@@ -283,6 +294,9 @@ class LocalClient:
             if callbacks:
                 logs = {"loss": float(ep_loss) / max(ep_steps, 1),
                         "accuracy": float(ep_acc) / max(ep_samples, 1)}
+                if self.val_loader is not None:
+                    vl, va = self.evaluate_loss(self.val_loader)
+                    logs["val_loss"], logs["val_accuracy"] = vl, va
                 for cb in callbacks:
                     cb.on_epoch_end(ep, logs)
                 if any(cb.stop_training for cb in callbacks):
@@ -361,6 +375,23 @@ class LocalClient:
         load_flat_params(self.model, vec.to(self.device))
         if hasattr(self.opt, "refresh_shadows"):
             self.opt.refresh_shadows()  # bf16 shadows must follow FedAvg loads
+
+    @torch.no_grad()
+    def evaluate_loss(self, loader: ClientLoader):
+        """Mean loss + accuracy over a loader (the per-epoch validation pass
+        the reference gets from fit(validation_data=...), FLPyfhelin.py:193)."""
+        dev = self.device
+        loss_sum = torch.zeros((), dtype=torch.float32, device=dev)
+        acc_sum = torch.zeros((), dtype=torch.float32, device=dev)
+        steps = samples = 0
+        for x, y in loader:
+            logits = self.model(x.to(self.compute_dtype))
+            loss_sum += softmax_xent(logits, y).detach().float()
+            acc_sum += (logits.detach().float().argmax(-1) == y).float().sum()
+            steps += 1
+            samples += y.numel()
+        return (float(loss_sum) / max(steps, 1),
+                float(acc_sum) / max(samples, 1))
 
     @torch.no_grad()
     def evaluate(self, dataset: SyntheticMedicalImages, indices: torch.Tensor,

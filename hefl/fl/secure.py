@@ -5,15 +5,24 @@ This collapses the reference's whole aggregation stack — pickle export
 add/mult loop (:366-390, timed at 231 s for 2 clients) — into:
 
   1. encrypt the client's flat fp32 weight vector into slot-packed CKKS
-     ciphertexts resident in HBM (one [B, 2, L, n] int64 tensor),
-  2. ONE all-reduce(SUM) on the raw RNS coefficient tensor. Limb primes are
-     < 2**60 (hefl/he/primes.py), so an int64 SUM over <= 8 clients is
-     overflow-free lazy reduction — no per-hop modular arithmetic needed,
-     and RCCL's stock int64 sum rings over the 7 xGMI links do the work,
+     ciphertexts resident in HBM (one [B, 2, L, n] int64 tensor), in SLABS,
+  2. per slab, ONE async all-reduce(SUM) on the raw RNS coefficients,
+     issued as soon as that slab is encrypted — the collective for slab i
+     rides the 7 xGMI links while slab i+1 is still encrypting on the
+     compute stream. Limb primes are < 2**60 (hefl/he/primes.py), so an
+     int64 SUM over <= 8 clients is overflow-free lazy reduction — no
+     per-hop modular arithmetic, RCCL's stock int64 sum rings do the work,
   3. one modreduce kernel back to [0, q_i), one ct x plain(1/n) multiply,
-     one rescale,
+     one rescale — run for EVERY n including n=1, so a 1-GPU round is
+     structurally identical to the 8-GPU round minus the collective,
   4. decrypt (key-holder only; the aggregation itself used no key at all,
      preserving the reference's pk-only-server model, FLPyfhelin.py:370).
+
+Key agreement: the reference shares ONE keypair via pickle files (cell 1 +
+FLPyfhelin.py:346-355). Here rank 0's keys are broadcast over the process
+group at init — correctness no longer depends on every rank seeding keygen
+identically (with he.seed=None each rank's keygen draws OS entropy; summing
+ciphertexts under different keys would decrypt to noise with no error).
 
 The timing labels mirror the reference's prints (FLPyfhelin.py:224,267,389)
 for log familiarity.
@@ -29,6 +38,12 @@ import torch.distributed as dist
 
 from ..he.ckks import CKKSContext, CtxtTensor, KeyPair
 
+# Per-slab payload for the bucketed all-reduce. xGMI is 7 point-to-point
+# links x ~153 GB/s per GPU; ~64 MiB slabs are big enough to amortize ring
+# setup yet small enough that several are in flight while later slabs
+# encrypt (collectives row).
+DEFAULT_BUCKET_BYTES = 64 << 20
+
 
 @dataclass
 class AggStats:
@@ -38,14 +53,15 @@ class AggStats:
 class SecureAggregator:
     """Per-rank helper owning the CKKS context + keys for encrypted FedAvg.
 
-    All ranks derive identical keys from the shared config seed (the
-    reference shares one keypair via pickle files, cell 1 + FLPyfhelin.py
-    :346-355); encryption randomness is then re-seeded per rank so client
-    ciphertexts are independent.
+    Rank 0 generates the keypair (and relin keys / encrypted denominator if
+    configured) and broadcasts it; every other rank adopts it. Encryption
+    randomness is then re-seeded per rank so client ciphertexts are
+    independent.
     """
 
     def __init__(self, ctx: CKKSContext, rank: int = 0, verbose: bool = False,
-                 denom_mode: str = "plain", n_clients: Optional[int] = None):
+                 denom_mode: str = "plain", n_clients: Optional[int] = None,
+                 bucket_bytes: int = DEFAULT_BUCKET_BYTES):
         """denom_mode:
         - "plain":     divide the summed ciphertext by plaintext 1/n
                        (ct x plain mult + rescale — what the reference
@@ -53,15 +69,13 @@ class SecureAggregator:
         - "encrypted": multiply by an ENCRYPTED 1/n (ct x ct + relinearize +
                        rescale — the reference's commented-out intent,
                        c_denom at FLPyfhelin.py:371; BASELINE.json config #3).
-                       Requires n_clients at construction: the denominator
-                       ciphertext is derived from the SHARED seed before the
-                       per-rank reseed, so every rank aggregates with the
-                       byte-identical ciphertext (no model drift).
+                       Requires n_clients at construction.
         """
         assert denom_mode in ("plain", "encrypted")
         self.ctx = ctx
         self.keys: KeyPair = ctx.keygen()
         self.denom_mode = denom_mode
+        self.bucket_bytes = bucket_bytes
         self.enc_denom = None
         if denom_mode == "encrypted":
             if ctx.L < 3:
@@ -74,9 +88,22 @@ class SecureAggregator:
             import numpy as np
             pt = ctx.encode(np.full(ctx.slots, 1.0 / n_clients))
             self.enc_denom = ctx.encrypt(pt, self.keys.pk)
+        self._sync_keys()
         ctx.reseed(rank)
         self.verbose = verbose
         self.stats = AggStats()
+
+    def _sync_keys(self) -> None:
+        """Broadcast rank 0's key material so all ranks share ONE keypair
+        (the reference's single publickey.pickle/privatekey.pickle model)."""
+        if not (dist.is_initialized() and dist.get_world_size() > 1):
+            return
+        dist.broadcast(self.keys.sk, src=0)
+        dist.broadcast(self.keys.pk, src=0)
+        if self.keys.relin is not None:
+            dist.broadcast(self.keys.relin, src=0)
+        if self.enc_denom is not None:
+            dist.broadcast(self.enc_denom.data, src=0)
 
     def _t(self, label: str, t0: float):
         dt = time.perf_counter() - t0
@@ -84,33 +111,85 @@ class SecureAggregator:
         if self.verbose:
             print(f"Time to {label}:", dt)
 
+    def _slab_cts(self) -> int:
+        """Ciphertexts per all-reduce bucket (>=1)."""
+        per_ct = 2 * self.ctx.L * self.ctx.n * 8  # int64 bytes
+        return max(1, self.bucket_bytes // per_ct)
+
     def encrypt(self, vec: torch.Tensor) -> CtxtTensor:
         t0 = time.perf_counter()
         ct = self.ctx.encrypt_tensor(vec, self.keys.pk)
         self._t("encrypt weights", t0)
         return ct
 
+    def _divide(self, ct: CtxtTensor, n: int) -> CtxtTensor:
+        """ct x (1/n) + rescale — ALWAYS executed (n=1 included) so the
+        per-round op sequence does not depend on world size."""
+        if self.denom_mode == "encrypted":
+            from ..he.ckks import Ciphertext
+            prod = self.ctx.mul_ct(Ciphertext(ct.data, ct.scale),
+                                   self.enc_denom, self.keys.relin)
+            out = CtxtTensor(prod.data, prod.scale, ct.count)
+            return self.ctx.rescale_tensor(out)
+        return self.ctx.rescale_tensor(
+            self.ctx.mul_scalar_tensor(ct, 1.0 / n))
+
     def aggregate(self, ct: CtxtTensor, n_clients: Optional[int] = None) -> CtxtTensor:
-        """Sum ciphertexts across ranks (lazy int64 all-reduce) and divide by n."""
+        """Sum ciphertexts across ranks (bucketed lazy int64 all-reduce) and
+        divide by n."""
         t0 = time.perf_counter()
         world = dist.get_world_size() if dist.is_initialized() else 1
         n = n_clients if n_clients is not None else world
         assert n <= 8, "lazy int64 reduction is proven for <= 8 summands"
         if world > 1:
-            dist.all_reduce(ct.data, op=dist.ReduceOp.SUM)
+            slab = self._slab_cts()
+            works = [dist.all_reduce(ct.data[i:i + slab], op=dist.ReduceOp.SUM,
+                                     async_op=True)
+                     for i in range(0, ct.data.shape[0], slab)]
+            for w in works:
+                w.wait()
         self.ctx.modreduce_tensor_(ct)
-        out = ct
-        if n > 1:
-            if self.denom_mode == "encrypted":
-                from ..he.ckks import Ciphertext
-                prod = self.ctx.mul_ct(Ciphertext(ct.data, ct.scale),
-                                       self.enc_denom, self.keys.relin)
-                out = CtxtTensor(prod.data, prod.scale, ct.count)
-                out = self.ctx.rescale_tensor(out)
-            else:
-                out = self.ctx.rescale_tensor(
-                    self.ctx.mul_scalar_tensor(ct, 1.0 / n))
+        out = self._divide(ct, n)
         self._t("aggregate", t0)
+        return out
+
+    def fedavg_ct(self, vec: torch.Tensor,
+                  n_clients: Optional[int] = None) -> CtxtTensor:
+        """Encrypt + all-reduce with slab pipelining: the all-reduce for
+        slab i overlaps the encryption of slab i+1 (encrypt runs on the
+        compute stream, the collective on RCCL's own streams)."""
+        t0 = time.perf_counter()
+        world = dist.get_world_size() if dist.is_initialized() else 1
+        n = n_clients if n_clients is not None else world
+        assert n <= 8, "lazy int64 reduction is proven for <= 8 summands"
+        ctx = self.ctx
+        slots = ctx.slots
+        count = vec.numel()
+        B = (count + slots - 1) // slots
+        slab = self._slab_cts()
+        if world <= 1 or B <= slab:
+            ct = self.encrypt(vec)
+            return self.aggregate(ct, n_clients=n)
+        data = torch.empty((B, 2, ctx.L, ctx.n), dtype=torch.int64,
+                           device=ctx.device)
+        works = []
+        scale = None
+        for b0 in range(0, B, slab):
+            b1 = min(B, b0 + slab)
+            sub = vec[b0 * slots: min(count, b1 * slots)]
+            ct_slab = ctx.encrypt_tensor(sub, self.keys.pk)
+            data[b0:b1].copy_(ct_slab.data)
+            scale = ct_slab.scale
+            works.append(dist.all_reduce(data[b0:b1], op=dist.ReduceOp.SUM,
+                                         async_op=True))
+        self._t("encrypt weights", t0)
+        t1 = time.perf_counter()
+        for w in works:
+            w.wait()
+        ct = CtxtTensor(data, scale, count)
+        self.ctx.modreduce_tensor_(ct)
+        out = self._divide(ct, n)
+        self._t("aggregate", t1)
         return out
 
     def decrypt(self, ct: CtxtTensor) -> torch.Tensor:
@@ -121,4 +200,4 @@ class SecureAggregator:
 
     def fedavg(self, vec: torch.Tensor, n_clients: Optional[int] = None) -> torch.Tensor:
         """Full encrypted FedAvg of a flat fp32 weight vector."""
-        return self.decrypt(self.aggregate(self.encrypt(vec), n_clients)).to(vec.device)
+        return self.decrypt(self.fedavg_ct(vec, n_clients)).to(vec.device)

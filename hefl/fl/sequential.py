@@ -28,6 +28,9 @@ from .client import LocalClient
 from .metrics import classification_metrics
 from .weights import flat_params, load_flat_params
 
+# Seed offset separating held-out test data from every training stream.
+TEST_SEED_OFFSET = 77777
+
 
 @dataclass
 class SeqRoundReport:
@@ -51,20 +54,28 @@ class SequentialFL:
         if cfg.fl.encrypted:
             self.ctx = CKKSContext(cfg.he, device=device)
             self.keys = self.ctx.keygen()
-        # held-out test set, disjoint seed from train (reference: 400 test
-        # images, notebook cell 0/3)
+        # held-out test set (reference: 400 test images, notebook cell 0/3):
+        # SAME class templates as training (template_seed) so it is the same
+        # task, but a disjoint seed for labels/noise — with the raw fl.seed
+        # the test labels replicated the first training labels and test
+        # noise could coincide with training noise, inflating the metrics.
         self.test_ds = SyntheticMedicalImages(
             cfg.fl.test_samples, cfg.model.in_shape, cfg.model.n_classes,
-            seed=cfg.fl.seed, device=device)
+            seed=cfg.fl.seed + TEST_SEED_OFFSET, device=device,
+            template_seed=cfg.fl.seed)
         n_train = cfg.fl.n_clients * cfg.fl.samples_per_client
         self.test_idx = torch.arange(n_train, n_train + cfg.fl.test_samples) \
             % cfg.fl.test_samples
 
     def _default_callbacks(self, client: LocalClient):
+        # monitor validation loss when the client holds a val split (the
+        # reference passes validation_data to every fit, FLPyfhelin.py:193);
+        # fall back to train loss when no val samples are configured
+        mon = "val_loss" if client.val_loader is not None else "loss"
         return [
-            EarlyStopping(client.model, monitor="loss", patience=5,
+            EarlyStopping(client.model, monitor=mon, patience=5,
                           restore_best=True),
-            ReduceLROnPlateau(client.opt, monitor="loss", factor=0.3,
+            ReduceLROnPlateau(client.opt, monitor=mon, factor=0.3,
                               patience=2),
         ]
 
@@ -84,7 +95,10 @@ class SequentialFL:
         t1 = time.perf_counter()
         n = len(self.clients)
         if self.ctx is not None:
-            # encrypted aggregation, key-separated: sum + 1/n under pk only
+            # encrypted aggregation, key-separated: sum + 1/n under pk only.
+            # Same lazy-int64 bound as the distributed path (secure.py):
+            # limb primes < 2**60, so > 8 summands could overflow int64.
+            assert n <= 8, "lazy int64 ciphertext sum is proven for <= 8 clients"
             agg: Optional[CtxtTensor] = None
             for i, v in enumerate(vecs):
                 self.ctx.reseed(i)

@@ -472,8 +472,19 @@ class CKKSContext:
         return out
 
     # ----- homomorphic ops -----
+    @staticmethod
+    def _check_add_scales(sa: float, sb: float) -> None:
+        # adding ciphertexts at different scales silently decodes to garbage
+        # (the slots are Delta-scaled integers; sums only make sense at one
+        # Delta) — fail loudly instead
+        if abs(sa - sb) > 1e-9 * max(sa, sb):
+            raise ValueError(
+                f"ciphertext scales differ ({sa} vs {sb}); rescale or "
+                f"re-encode one operand before adding")
+
     def add(self, a: Ciphertext, b: Ciphertext) -> Ciphertext:
         assert a.level == b.level
+        self._check_add_scales(a.scale, b.scale)
         limbs = list(range(a.level))
         return Ciphertext(self._modadd(a.data, b.data, limbs), max(a.scale, b.scale))
 
@@ -565,6 +576,7 @@ class CKKSContext:
 
     def add_tensor(self, a: CtxtTensor, b: CtxtTensor) -> CtxtTensor:
         assert a.count == b.count and a.level == b.level
+        self._check_add_scales(a.scale, b.scale)
         limbs = list(range(a.level))
         return CtxtTensor(self._modadd(a.data, b.data, limbs),
                           max(a.scale, b.scale), a.count)

@@ -73,7 +73,11 @@ class PyCtxt:
 
     def __add__(self, other):
         if isinstance(other, (int, float)) and other == 0:
-            return self
+            # clone, not self: the reference seeds accumulators with int 0
+            # (FLPyfhelin.py:380-381) — returning self would alias the
+            # accumulator to the client's ciphertext, so a later in-place
+            # mutation of the sum would corrupt the client's upload
+            return PyCtxt(self._ct.clone(), self._pyfhel)
         if isinstance(other, PyCtxt):
             he = self._he()
             return PyCtxt(he._ctx.add(self._ct, other._ct), he)
@@ -233,15 +237,18 @@ class Pyfhel:
         self._sk = self._key_from_bytes(b"SK\x00\x00", buf)
 
     # Pyfhel pickles whole objects in the reference's export dict
-    # (FLPyfhelin.py:233); support that directly.
+    # (FLPyfhelin.py:233); support that directly. The SECRET key is never
+    # pickled: the reference's C-backed pickled Pyfhel carries no usable key
+    # material, and a pickled object lands in 'public' artifacts
+    # (publickey.pickle, per-client uploads) that the pk-only aggregation
+    # server reads — sk must travel only via the explicit
+    # to_bytes_secretKey/privatekey.pickle path (hefl/fl/keys.py).
     def __getstate__(self):
-        state = {"cfg": None, "pk": None, "sk": None}
+        state = {"cfg": None, "pk": None}
         if self._ctx is not None:
             state["cfg"] = self._ctx.cfg
         if self._pk is not None:
             state["pk"] = self._pk.cpu().numpy()
-        if self._sk is not None:
-            state["sk"] = self._sk.cpu().numpy()
         return state
 
     def __setstate__(self, state):
@@ -250,5 +257,5 @@ class Pyfhel:
             self.contextGen_cfg(state["cfg"])
         if state["pk"] is not None:
             self._pk = torch.from_numpy(state["pk"])
-        if state["sk"] is not None:
+        if state.get("sk") is not None:  # legacy pickles only
             self._sk = torch.from_numpy(state["sk"])

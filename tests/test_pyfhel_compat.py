@@ -76,13 +76,23 @@ def test_key_serialization_roundtrip():
 
 def test_ctxt_pickle_roundtrip_standalone():
     # the reference pickles ndarray-of-PyCtxt and re-attaches contexts on
-    # load (FLPyfhelin.py:236,309,321); ours round-trips standalone
+    # load (FLPyfhelin.py:236,309,321); ours round-trips standalone.
+    # A pickled Pyfhel deliberately does NOT carry the secret key (the
+    # export dict lands in 'public' artifacts the aggregation server
+    # reads); sk travels only via the explicit to_bytes_secretKey path.
     he = make_he()
     ct = he.encryptFrac(4.5)
+    sk = he.to_bytes_secretKey()
     blob = pickle.dumps({"key": he, "val": {"c_0_0": ct}},
                         protocol=pickle.HIGHEST_PROTOCOL)
     loaded = pickle.loads(blob)
     he2, ct2 = loaded["key"], loaded["val"]["c_0_0"]
+    with pytest.raises(ValueError):
+        he2.decryptFrac(ct2)  # no sk in the pickle: must fail loudly
+    # pk survives the pickle (the server can keep aggregating under it)
+    he2.encryptFrac(1.0)
+    # the key-holder restores sk explicitly and decrypts
+    he2.from_bytes_secretKey(sk)
     assert abs(he2.decryptFrac(ct2) - 4.5) < 1e-4
 
 
