@@ -24,6 +24,20 @@ torch::Tensor modmul_limbs(torch::Tensor a, torch::Tensor b, torch::Tensor qs,
 torch::Tensor modmul_scalar_limbs(torch::Tensor a, torch::Tensor scalars,
                                   torch::Tensor shoups, torch::Tensor qs,
                                   int64_t L, int64_t n);
+torch::Tensor modadd_limbs(torch::Tensor a, torch::Tensor b, torch::Tensor qs,
+                           int64_t L, int64_t n);
+torch::Tensor modadd3_limbs(torch::Tensor a, torch::Tensor b, torch::Tensor c,
+                            torch::Tensor qs, int64_t L, int64_t n);
+torch::Tensor modsub_limbs(torch::Tensor a, torch::Tensor b, torch::Tensor qs,
+                           int64_t L, int64_t aL, int64_t n);
+torch::Tensor bcast_center_mod(torch::Tensor x, int64_t qc, torch::Tensor qs,
+                               torch::Tensor ratios, int64_t L);
+std::vector<torch::Tensor> ct_mul(torch::Tensor a, torch::Tensor b,
+                                  torch::Tensor qs, torch::Tensor ratios,
+                                  int64_t L, int64_t n);
+std::vector<torch::Tensor> ks_inner(torch::Tensor dig, torch::Tensor rlk,
+                                    torch::Tensor qs, torch::Tensor ratios,
+                                    int64_t D, int64_t Lp, int64_t n);
 
 // cnn.hip
 torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
@@ -104,6 +118,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("intt_limbs", &intt_limbs, "fused multi-limb inverse NTT");
     m.def("modmul_limbs", &modmul_limbs);
     m.def("modmul_scalar_limbs", &modmul_scalar_limbs);
+    m.def("modadd_limbs", &modadd_limbs, "per-limb modular add over [.., L, n]");
+    m.def("modadd3_limbs", &modadd3_limbs, "fused 3-way per-limb modular add");
+    m.def("modsub_limbs", &modsub_limbs, "per-limb modular sub over [.., L, n]");
+    m.def("bcast_center_mod", &bcast_center_mod,
+          "[.., n] -> [.., L, n] center + per-limb Barrett reduce");
+    m.def("ct_mul", &ct_mul, "fused ct x ct tensor product (d0, d1, d2)");
+    m.def("ks_inner", &ks_inner, "fused key-switch digit inner product");
     // CNN
     m.def("conv2d_fwd", &conv2d_fwd);
     m.def("conv2d_dgrad", &conv2d_dgrad);

@@ -638,6 +638,161 @@ __global__ void modreduce_kernel(int64_t* __restrict__ x,
     }
 }
 
+// ---------------------------------------------------------------------------
+// Fused per-limb pointwise ops over [..., L, n] tensors (limb = (i/n) % L).
+// These replace the torch.remainder / int64 add glue that at::native was
+// otherwise running between the NTT launches (round-1 profiles showed ~4%
+// of the config #5 round in generic elementwise remainder kernels).
+// ---------------------------------------------------------------------------
+
+__global__ void modadd_limbs_kernel(const int64_t* __restrict__ a,
+                                    const int64_t* __restrict__ b,
+                                    int64_t* __restrict__ out, int64_t total,
+                                    int64_t b_numel,
+                                    const int64_t* __restrict__ qs, int L,
+                                    int64_t n) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        const uint64_t q = (uint64_t)qs[(i / n) % L];
+        out[i] = (int64_t)addmod_u64((uint64_t)a[i], (uint64_t)b[i % b_numel], q);
+    }
+}
+
+// c0 of encrypt: (b*u + e + pt) with all three residues already in [0, q).
+__global__ void modadd3_limbs_kernel(const int64_t* __restrict__ a,
+                                     const int64_t* __restrict__ b,
+                                     const int64_t* __restrict__ c,
+                                     int64_t* __restrict__ out, int64_t total,
+                                     const int64_t* __restrict__ qs, int L,
+                                     int64_t n) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        const uint64_t q = (uint64_t)qs[(i / n) % L];
+        uint64_t v = (uint64_t)a[i] + (uint64_t)b[i];  // < 2q < 2^61
+        if (v >= q) v -= q;
+        out[i] = (int64_t)addmod_u64(v, (uint64_t)c[i], q);
+    }
+}
+
+// a may carry MORE limbs than the output (aL >= L): rescale subtracts over
+// the first L of a's aL limbs without materializing the slice first.
+__global__ void modsub_limbs_kernel(const int64_t* __restrict__ a,
+                                    const int64_t* __restrict__ b,
+                                    int64_t* __restrict__ out, int64_t total,
+                                    int64_t b_numel,
+                                    const int64_t* __restrict__ qs, int L,
+                                    int aL, int64_t n) {
+    const int64_t Ln = (int64_t)L * n;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        const uint64_t q = (uint64_t)qs[(i / n) % L];
+        const int64_t ai = (aL == L) ? i
+                                     : (i / Ln) * ((int64_t)aL * n) + (i % Ln);
+        out[i] = (int64_t)submod_u64((uint64_t)a[ai], (uint64_t)b[i % b_numel], q);
+    }
+}
+
+// Broadcast one coefficient row [.., n] into L limb residues [.., L, n]:
+//   v = x (centered signed if qc == 0, else x in [0, qc) centered mod qc),
+//   out[.., l, j] = v mod q_l  (Barrett; |v| can exceed q_l).
+// One launch replaces torch.where + the [.., 1, n] -> [.., L, n] broadcast
+// remainder in _to_rns_ntt / rescale / key-switch mod-down.
+__global__ void bcast_center_mod_kernel(const int64_t* __restrict__ x,
+                                        int64_t* __restrict__ out,
+                                        int64_t total_out, int64_t n, int L,
+                                        uint64_t qc,
+                                        const int64_t* __restrict__ qs,
+                                        const int64_t* __restrict__ ratios) {
+    const int64_t Ln = (int64_t)L * n;
+    for (int64_t o = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         o < total_out; o += (int64_t)gridDim.x * blockDim.x) {
+        const int64_t j = o % n;
+        const int l = (int)((o / n) % L);
+        const int64_t row = o / Ln;
+        int64_t v = x[row * n + j];
+        if (qc) {
+            if ((uint64_t)v > (qc >> 1)) v -= (int64_t)qc;
+        }
+        const uint64_t q = (uint64_t)qs[l];
+        const uint64_t r0 = (uint64_t)ratios[2 * l];
+        const uint64_t r1 = (uint64_t)ratios[2 * l + 1];
+        const uint64_t m = v < 0 ? (uint64_t)(-v) : (uint64_t)v;
+        const uint64_t red = barrett_red128(m, 0, q, r0, r1);
+        out[o] = (int64_t)((v < 0 && red) ? q - red : red);
+    }
+}
+
+// ---------------------------------------------------------------------------
+// Fused ct x ct + hybrid key-switch kernels (the Python
+// per-digit/per-limb loop was O(L*(L+1)) small launches).
+// ---------------------------------------------------------------------------
+
+// Tensor product of two ciphertexts: d0 = a0*b0, d1 = a0*b1 + a1*b0,
+// d2 = a1*b1, per limb. a is [R, 2, L, n]; b is [Rb, 2, L, n] with Rb == R
+// or Rb == 1 (a batched ciphertext times one shared ct, e.g. the encrypted
+// 1/n denominator). One launch replaces 3L modmuls + L remainders.
+__global__ void ct_mul_kernel(const int64_t* __restrict__ a,
+                              const int64_t* __restrict__ b,
+                              int64_t* __restrict__ d0,
+                              int64_t* __restrict__ d1,
+                              int64_t* __restrict__ d2, int64_t total,
+                              int64_t b_total, int L, int64_t n,
+                              const int64_t* __restrict__ qs,
+                              const int64_t* __restrict__ ratios) {
+    const int64_t Ln = (int64_t)L * n;
+    for (int64_t o = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         o < total; o += (int64_t)gridDim.x * blockDim.x) {
+        const int l = (int)((o / n) % L);
+        const int64_t row = o / Ln;
+        const int64_t inner = o % Ln;
+        const uint64_t q = (uint64_t)qs[l];
+        const uint64_t r0 = (uint64_t)ratios[2 * l];
+        const uint64_t r1 = (uint64_t)ratios[2 * l + 1];
+        const int64_t ia = row * 2 * Ln + inner;
+        const int64_t ib = (row * 2 * Ln + inner) % b_total;
+        const uint64_t a0 = (uint64_t)a[ia], a1 = (uint64_t)a[ia + Ln];
+        const uint64_t b0 = (uint64_t)b[ib], b1 = (uint64_t)b[ib + Ln];
+        d0[o] = (int64_t)mulmod_barrett(a0, b0, q, r0, r1);
+        d1[o] = (int64_t)addmod_u64(mulmod_barrett(a0, b1, q, r0, r1),
+                                    mulmod_barrett(a1, b0, q, r0, r1), q);
+        d2[o] = (int64_t)mulmod_barrett(a1, b1, q, r0, r1);
+    }
+}
+
+// Key-switch inner product: acc_c[row, i, j] = sum_d dig[row, d, i, j] *
+// rlk[d, c, i, j] mod q_i, for c in {0, 1}. dig is the NTT-form digit
+// decomposition [R, D, Lp, n]; rlk is [D, 2, Lp, n]. Reads dig once for
+// both accumulators; one launch replaces 2*D*Lp modmul+add rounds.
+__global__ void ks_inner_kernel(const int64_t* __restrict__ dig,
+                                const int64_t* __restrict__ rlk,
+                                int64_t* __restrict__ acc0,
+                                int64_t* __restrict__ acc1, int64_t total,
+                                int D, int Lp, int64_t n,
+                                const int64_t* __restrict__ qs,
+                                const int64_t* __restrict__ ratios) {
+    const int64_t Lpn = (int64_t)Lp * n;
+    for (int64_t o = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         o < total; o += (int64_t)gridDim.x * blockDim.x) {
+        const int64_t j = o % n;
+        const int i = (int)((o / n) % Lp);
+        const int64_t row = o / Lpn;
+        const uint64_t q = (uint64_t)qs[i];
+        const uint64_t r0 = (uint64_t)ratios[2 * i];
+        const uint64_t r1 = (uint64_t)ratios[2 * i + 1];
+        uint64_t a0 = 0, a1 = 0;
+        const int64_t dig_base = row * D * Lpn + (int64_t)i * n + j;
+        const int64_t rlk_base = (int64_t)i * n + j;
+        for (int d = 0; d < D; ++d) {
+            const uint64_t x = (uint64_t)dig[dig_base + (int64_t)d * Lpn];
+            const int64_t rb = rlk_base + (int64_t)d * 2 * Lpn;
+            a0 = addmod_u64(a0, mulmod_barrett(x, (uint64_t)rlk[rb], q, r0, r1), q);
+            a1 = addmod_u64(a1, mulmod_barrett(x, (uint64_t)rlk[rb + Lpn], q, r0, r1), q);
+        }
+        acc0[o] = (int64_t)a0;
+        acc1[o] = (int64_t)a1;
+    }
+}
+
 // Centered binomial eta=21 from ONE uniform 64-bit draw per coefficient:
 // e = popcount(bits[0:21]) - popcount(bits[21:42]) — replaces 42 separate
 // int8 Bernoulli draws (the sampling half of encrypt's cost at ResNet scale).
@@ -891,5 +1046,115 @@ torch::Tensor cbd21(torch::Tensor bits) {
     hipLaunchKernelGGL(cbd21_kernel, dim3(blocks), dim3(kThreads), 0,
                        at::cuda::getCurrentCUDAStream(),
                        bits.data_ptr<int64_t>(), out.data_ptr<int64_t>(), total);
+    return out;
+}
+
+torch::Tensor modadd_limbs(torch::Tensor a, torch::Tensor b, torch::Tensor qs,
+                           int64_t L, int64_t n) {
+    CHECK_CUDA_OK(a);
+    TORCH_CHECK(a.is_contiguous() && b.is_contiguous());
+    TORCH_CHECK(a.numel() % b.numel() == 0);
+    auto out = torch::empty_like(a);
+    int64_t total = a.numel();
+    int blocks = (int)std::min<int64_t>((total + kThreads - 1) / kThreads, 2048);
+    hipLaunchKernelGGL(modadd_limbs_kernel, dim3(blocks), dim3(kThreads), 0,
+                       at::cuda::getCurrentCUDAStream(), a.data_ptr<int64_t>(),
+                       b.data_ptr<int64_t>(), out.data_ptr<int64_t>(), total,
+                       b.numel(), qs.data_ptr<int64_t>(), (int)L, n);
+    return out;
+}
+
+torch::Tensor modadd3_limbs(torch::Tensor a, torch::Tensor b, torch::Tensor c,
+                            torch::Tensor qs, int64_t L, int64_t n) {
+    CHECK_CUDA_OK(a);
+    TORCH_CHECK(a.is_contiguous() && b.is_contiguous() && c.is_contiguous());
+    TORCH_CHECK(a.numel() == b.numel() && a.numel() == c.numel());
+    auto out = torch::empty_like(a);
+    int64_t total = a.numel();
+    int blocks = (int)std::min<int64_t>((total + kThreads - 1) / kThreads, 2048);
+    hipLaunchKernelGGL(modadd3_limbs_kernel, dim3(blocks), dim3(kThreads), 0,
+                       at::cuda::getCurrentCUDAStream(), a.data_ptr<int64_t>(),
+                       b.data_ptr<int64_t>(), c.data_ptr<int64_t>(),
+                       out.data_ptr<int64_t>(), total, qs.data_ptr<int64_t>(),
+                       (int)L, n);
+    return out;
+}
+
+torch::Tensor modsub_limbs(torch::Tensor a, torch::Tensor b, torch::Tensor qs,
+                           int64_t L, int64_t aL, int64_t n) {
+    CHECK_CUDA_OK(a);
+    TORCH_CHECK(a.is_contiguous() && b.is_contiguous());
+    TORCH_CHECK(a.size(-2) == aL && a.size(-1) == n);
+    auto sizes = a.sizes().vec();
+    sizes[sizes.size() - 2] = L;  // output carries the first L of a's limbs
+    auto out = torch::empty(sizes, a.options());
+    int64_t total = out.numel();
+    int blocks = (int)std::min<int64_t>((total + kThreads - 1) / kThreads, 2048);
+    hipLaunchKernelGGL(modsub_limbs_kernel, dim3(blocks), dim3(kThreads), 0,
+                       at::cuda::getCurrentCUDAStream(), a.data_ptr<int64_t>(),
+                       b.data_ptr<int64_t>(), out.data_ptr<int64_t>(), total,
+                       b.numel(), qs.data_ptr<int64_t>(), (int)L, (int)aL, n);
+    return out;
+}
+
+// a [R, 2, L, n] x b [Rb, 2, L, n] (Rb == R or 1) -> (d0, d1, d2) each [R, L, n].
+std::vector<torch::Tensor> ct_mul(torch::Tensor a, torch::Tensor b,
+                                  torch::Tensor qs, torch::Tensor ratios,
+                                  int64_t L, int64_t n) {
+    CHECK_CUDA_OK(a);
+    TORCH_CHECK(a.is_contiguous() && b.is_contiguous());
+    auto sizes = a.sizes().vec();       // [.., 2, L, n]
+    sizes.erase(sizes.end() - 3);       // drop the (c0, c1) dim
+    auto d0 = torch::empty(sizes, a.options());
+    auto d1 = torch::empty(sizes, a.options());
+    auto d2 = torch::empty(sizes, a.options());
+    int64_t total = d0.numel();
+    int blocks = (int)std::min<int64_t>((total + kThreads - 1) / kThreads, 4096);
+    hipLaunchKernelGGL(ct_mul_kernel, dim3(blocks), dim3(kThreads), 0,
+                       at::cuda::getCurrentCUDAStream(), a.data_ptr<int64_t>(),
+                       b.data_ptr<int64_t>(), d0.data_ptr<int64_t>(),
+                       d1.data_ptr<int64_t>(), d2.data_ptr<int64_t>(), total,
+                       b.numel(), (int)L, n, qs.data_ptr<int64_t>(),
+                       ratios.data_ptr<int64_t>());
+    return {d0, d1, d2};
+}
+
+// dig [R, D, Lp, n] x rlk [D, 2, Lp, n] -> (acc0, acc1) each [R, Lp, n].
+std::vector<torch::Tensor> ks_inner(torch::Tensor dig, torch::Tensor rlk,
+                                    torch::Tensor qs, torch::Tensor ratios,
+                                    int64_t D, int64_t Lp, int64_t n) {
+    CHECK_CUDA_OK(dig);
+    TORCH_CHECK(dig.is_contiguous() && rlk.is_contiguous());
+    auto sizes = dig.sizes().vec();     // [.., D, Lp, n]
+    sizes.erase(sizes.end() - 3);       // drop the digit dim
+    auto acc0 = torch::empty(sizes, dig.options());
+    auto acc1 = torch::empty(sizes, dig.options());
+    int64_t total = acc0.numel();
+    int blocks = (int)std::min<int64_t>((total + kThreads - 1) / kThreads, 4096);
+    hipLaunchKernelGGL(ks_inner_kernel, dim3(blocks), dim3(kThreads), 0,
+                       at::cuda::getCurrentCUDAStream(),
+                       dig.data_ptr<int64_t>(), rlk.data_ptr<int64_t>(),
+                       acc0.data_ptr<int64_t>(), acc1.data_ptr<int64_t>(),
+                       total, (int)D, (int)Lp, n, qs.data_ptr<int64_t>(),
+                       ratios.data_ptr<int64_t>());
+    return {acc0, acc1};
+}
+
+// x [..., n] -> out [..., L, n]: per-limb Barrett reduction of the (optionally
+// qc-centered) coefficient row. qc = 0 means x is already centered signed.
+torch::Tensor bcast_center_mod(torch::Tensor x, int64_t qc, torch::Tensor qs,
+                               torch::Tensor ratios, int64_t L) {
+    CHECK_CUDA_OK(x);
+    TORCH_CHECK(x.is_contiguous());
+    const int64_t n = x.size(-1);
+    auto sizes = x.sizes().vec();
+    sizes.insert(sizes.end() - 1, L);
+    auto out = torch::empty(sizes, x.options());
+    int64_t total = out.numel();
+    int blocks = (int)std::min<int64_t>((total + kThreads - 1) / kThreads, 4096);
+    hipLaunchKernelGGL(bcast_center_mod_kernel, dim3(blocks), dim3(kThreads), 0,
+                       at::cuda::getCurrentCUDAStream(), x.data_ptr<int64_t>(),
+                       out.data_ptr<int64_t>(), total, n, (int)L, (uint64_t)qc,
+                       qs.data_ptr<int64_t>(), ratios.data_ptr<int64_t>());
     return out;
 }

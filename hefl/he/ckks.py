@@ -158,6 +158,32 @@ class GpuBackend:
         return self._C.modmul_scalar_limbs(a.contiguous(), sct, sht,
                                            self.qs, L, self.n)
 
+    # ----- fused per-limb pointwise glue (replaces torch.remainder paths) ---
+    def modadd_limbs(self, a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+        return self._C.modadd_limbs(a.contiguous(), b.contiguous(), self.qs,
+                                    a.shape[-2], self.n)
+
+    def modadd3_limbs(self, a, b, c) -> torch.Tensor:
+        return self._C.modadd3_limbs(a.contiguous(), b.contiguous(),
+                                     c.contiguous(), self.qs, a.shape[-2],
+                                     self.n)
+
+    def modsub_limbs(self, a: torch.Tensor, b: torch.Tensor,
+                     L: Optional[int] = None) -> torch.Tensor:
+        """out[.., l, :] = a[.., l, :] - b over the first L limbs (a may
+        carry more limbs than L — the rescale slice without a copy)."""
+        aL = a.shape[-2]
+        L = aL if L is None else L
+        return self._C.modsub_limbs(a.contiguous(), b.contiguous(), self.qs,
+                                    L, aL, self.n)
+
+    def bcast_center_mod(self, x: torch.Tensor, qc: int, L: int) -> torch.Tensor:
+        """x [..., n] -> [..., L, n]: center mod qc (qc=0: already centered
+        signed), then per-limb Barrett reduce. One launch replacing
+        torch.where + broadcast remainder."""
+        return self._C.bcast_center_mod(x.contiguous(), qc, self.qs,
+                                        self.ratio_words, L)
+
 
 # ---------------------------------------------------------------------------
 # Data containers
@@ -269,6 +295,9 @@ class CKKSContext:
         return self.all_primes[limb]  # index L = the special prime
 
     def _modadd(self, a: torch.Tensor, b: torch.Tensor, limbs: Sequence[int]) -> torch.Tensor:
+        if self.device.type == "cuda" and list(limbs) == list(range(len(limbs))):
+            # fused per-limb HIP kernel (inputs are valid residues < q)
+            return self.backend.modadd_limbs(a, b)
         qs = torch.tensor([self._q(i) for i in limbs], dtype=torch.int64,
                           device=a.device)
         shape = [1] * a.dim()
@@ -276,6 +305,8 @@ class CKKSContext:
         return torch.remainder(a + b, qs.view(shape))
 
     def _modsub(self, a: torch.Tensor, b: torch.Tensor, limbs: Sequence[int]) -> torch.Tensor:
+        if self.device.type == "cuda" and list(limbs) == list(range(len(limbs))):
+            return self.backend.modsub_limbs(a, b)
         qs = torch.tensor([self._q(i) for i in limbs], dtype=torch.int64,
                           device=a.device)
         shape = [1] * a.dim()
@@ -323,11 +354,8 @@ class CKKSContext:
         """Small centered int64 coeffs [..., n] -> NTT-form RNS [..., L, n]."""
         nlimbs = self.L if nlimbs is None else nlimbs
         if self.device.type == "cuda":
-            qs = self.backend.qs[:nlimbs]
-            shape = [1] * coeffs.dim() + [1]
-            shape[-2] = nlimbs
-            rem = torch.remainder(coeffs.to(self.device).unsqueeze(-2),
-                                  qs.view(shape))
+            rem = self.backend.bcast_center_mod(coeffs.to(self.device), 0,
+                                                nlimbs)
             return self.backend.ntt_all(rem)
         out = []
         for i in range(nlimbs):
@@ -435,11 +463,10 @@ class CKKSContext:
         e0 = self._to_rns_ntt(self._sample_err(lead + (n,)), nlimbs)
         e1 = self._to_rns_ntt(self._sample_err(lead + (n,)), nlimbs)
         if self.device.type == "cuda":
-            qs = self.backend.qs[:nlimbs].view((1,) * len(lead) + (nlimbs, 1))
             bu = self.backend.modmul_limbs(u, pk[0, :nlimbs])
             au = self.backend.modmul_limbs(u, pk[1, :nlimbs])
-            c0 = torch.remainder(bu + e0 + ptdata, qs)
-            c1 = torch.remainder(au + e1, qs)
+            c0 = self.backend.modadd3_limbs(bu, e0, ptdata.contiguous())
+            c1 = self.backend.modadd_limbs(au, e1)
             return torch.stack([c0, c1], dim=-3)
         c0 = torch.empty_like(ptdata)
         c1 = torch.empty_like(ptdata)
@@ -462,9 +489,7 @@ class CKKSContext:
         c1 = ctdata[..., 1, :, :]
         if self.device.type == "cuda":
             cs = self.backend.modmul_limbs(c1.contiguous(), sk[:nlimbs])
-            qs = self.backend.qs[:nlimbs].view((1,) * (c0.dim() - 2)
-                                               + (nlimbs, 1))
-            return torch.remainder(c0 + cs, qs)
+            return self.backend.modadd_limbs(c0.contiguous(), cs)
         out = torch.empty_like(c0)
         for i in range(nlimbs):
             cs = self.backend.modmul(c1[..., i, :], sk[i].expand_as(c1[..., i, :]).contiguous(), i)
@@ -494,6 +519,9 @@ class CKKSContext:
         delta = self.scale if scale is None else scale
         sc = int(round(x * delta))
         nlimbs = data.shape[-2]
+        if self.device.type == "cuda":
+            return self.backend.modmul_scalar_limbs(
+                data.contiguous(), [sc] * nlimbs)
         out = torch.empty_like(data)
         for i in range(nlimbs):
             q = self._q(i)
@@ -520,21 +548,22 @@ class CKKSContext:
         last = nlimbs - 1
         qL = self._q(last)
         qL_half = qL // 2
+        if self.device.type == "cuda":
+            # fully fused: INTT of the last limb, one center+broadcast-reduce
+            # launch, one fused NTT set, one strided-slice modsub (no
+            # materialized slice), one scalar-limbs multiply — zero
+            # at::native kernels on this path
+            cl = self.backend.ntt(data[..., last, :].contiguous(), last,
+                                  inverse=True)
+            r = self.backend.bcast_center_mod(cl, qL, last)
+            r_ntt = self.backend.ntt_all(r)
+            diff = self.backend.modsub_limbs(data.contiguous(), r_ntt, last)
+            inv = [pow(qL % self._q(i), -1, self._q(i)) for i in range(last)]
+            return self.backend.modmul_scalar_limbs(diff, inv)
         # coefficient-domain last limb, centered for round-to-nearest
         cl = self.backend.ntt(data[..., last, :].contiguous(), last,
                               inverse=True)
         cl_c = torch.where(cl > qL_half, cl - qL, cl)
-        if self.device.type == "cuda":
-            # fused across target limbs: one remainder broadcast, one fused
-            # NTT launch set, one scalar-limbs multiply
-            qs = self.backend.qs[:last]
-            shape = [1] * cl_c.dim() + [1]
-            shape[-2] = last
-            r = torch.remainder(cl_c.unsqueeze(-2), qs.view(shape))
-            r_ntt = self.backend.ntt_all(r)
-            diff = torch.remainder(data[..., :last, :] - r_ntt, qs.view(shape))
-            inv = [pow(qL % self._q(i), -1, self._q(i)) for i in range(last)]
-            return self.backend.modmul_scalar_limbs(diff, inv)
         out = data[..., :last, :].clone()
         for i in range(last):
             q = self._q(i)
@@ -633,10 +662,31 @@ class CKKSContext:
 
     def _keyswitch(self, d2: torch.Tensor, rlk: torch.Tensor):
         """Key-switch NTT-form d2 [..., L, n] through rlk; returns the pair
-        (ks0, ks1) each [..., L, n] to add to (c0, c1)."""
+        (ks0, ks1) each [..., L, n] to add to (c0, c1).
+
+        GPU path is fully fused: one INTT set over all
+        digits, one lift launch, one NTT set, ONE inner-product kernel over
+        all digits x limbs, then a fused mod-down — ~12 launches total
+        instead of the O(L*(L+1)) per-digit/per-limb Python loop."""
         L = d2.shape[-2]
         Lp = L + 1
         n = self.n
+        if self.device.type == "cuda":
+            be = self.backend
+            c = be.ntt_all(d2, inverse=True)           # digits -> coeff domain
+            dig = be.bcast_center_mod(c, 0, Lp)        # [..., L, Lp, n]
+            dig = be.ntt_all(dig)                      # NTT over the Lp limbs
+            acc0, acc1 = be._C.ks_inner(dig.contiguous(), rlk.contiguous(),
+                                        be.qs, be.ratio_words, L, Lp, n)
+            P = self.special
+            inv = [pow(P % self._q(i), -1, self._q(i)) for i in range(L)]
+            outs = []
+            for acc in (acc0, acc1):
+                cl = be.ntt(acc[..., L, :].contiguous(), L, inverse=True)
+                r_ntt = be.ntt_all(be.bcast_center_mod(cl, P, L))
+                diff = be.modsub_limbs(acc, r_ntt, L)
+                outs.append(be.modmul_scalar_limbs(diff, inv))
+            return outs[0], outs[1]
         lead = d2.shape[:-2]
         acc0 = torch.zeros(lead + (Lp, n), dtype=torch.int64, device=d2.device)
         acc1 = torch.zeros_like(acc0)
@@ -678,18 +728,24 @@ class CKKSContext:
                 "relin_keygen(sk) (Pyfhel API: relinKeyGen) first")
         assert a.level == b.level
         L = a.level
-        a0, a1 = a.data[..., 0, :, :], a.data[..., 1, :, :]
-        b0, b1 = b.data[..., 0, :, :], b.data[..., 1, :, :]
-        d0 = torch.empty_like(a0)
-        d1 = torch.empty_like(a0)
-        d2 = torch.empty_like(a0)
-        for i in range(L):
-            q = self._q(i)
-            d0[..., i, :] = self.backend.modmul(a0[..., i, :], b0[..., i, :], i)
-            cross = (self.backend.modmul(a0[..., i, :], b1[..., i, :], i)
-                     + self.backend.modmul(a1[..., i, :], b0[..., i, :], i))
-            d1[..., i, :] = torch.remainder(cross, q)
-            d2[..., i, :] = self.backend.modmul(a1[..., i, :], b1[..., i, :], i)
+        if self.device.type == "cuda":
+            # one fused tensor-product launch (d0, d1, d2 together)
+            d0, d1, d2 = self.backend._C.ct_mul(
+                a.data.contiguous(), b.data.contiguous(), self.backend.qs,
+                self.backend.ratio_words, L, self.n)
+        else:
+            a0, a1 = a.data[..., 0, :, :], a.data[..., 1, :, :]
+            b0, b1 = b.data[..., 0, :, :], b.data[..., 1, :, :]
+            d0 = torch.empty_like(a0)
+            d1 = torch.empty_like(a0)
+            d2 = torch.empty_like(a0)
+            for i in range(L):
+                q = self._q(i)
+                d0[..., i, :] = self.backend.modmul(a0[..., i, :], b0[..., i, :], i)
+                cross = (self.backend.modmul(a0[..., i, :], b1[..., i, :], i)
+                         + self.backend.modmul(a1[..., i, :], b0[..., i, :], i))
+                d1[..., i, :] = torch.remainder(cross, q)
+                d2[..., i, :] = self.backend.modmul(a1[..., i, :], b1[..., i, :], i)
         ks0, ks1 = self._keyswitch(d2, rlk)
         limbs = list(range(L))
         c0 = self._modadd(d0, ks0, limbs)
@@ -699,6 +755,11 @@ class CKKSContext:
     def modreduce_tensor_(self, ct: CtxtTensor) -> CtxtTensor:
         """Reduce lazily-summed (int64) limb values back to [0, q_i) in place
         — the step after the RCCL all-reduce of raw coefficient tensors."""
+        if self.device.type == "cuda":
+            # branchless conditional-subtract kernel (sum of <= 8 residues)
+            self.backend._C.modreduce_(
+                ct.data, self.backend.qs[:ct.level].contiguous())
+            return ct
         qs = torch.tensor(self.primes[:ct.level], dtype=torch.int64,
                           device=ct.data.device)
         ct.data.remainder_(qs.view(1, 1, -1, 1))
