@@ -94,6 +94,7 @@ std::vector<torch::Tensor> pool_relu_bias_bwd(torch::Tensor dy,
                                               torch::Tensor y);
 void adam_prep_epoch(torch::Tensor step, torch::Tensor sched,
                      torch::Tensor hyper, double b1, double b2, int64_t S);
+torch::Tensor pad_channels(torch::Tensor x, int64_t C8);
 void pack_mt(torch::Tensor meta, torch::Tensor ptrs, torch::Tensor sizes,
              torch::Tensor offs, int64_t nchunks, torch::Tensor flat);
 void unpack_mt(torch::Tensor flat, torch::Tensor meta, torch::Tensor ptrs,
@@ -153,6 +154,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("relu_bias_bwd", &relu_bias_bwd);
     m.def("pool_relu_bias_bwd", &pool_relu_bias_bwd);
     m.def("fused_adam_mt", &fused_adam_mt);
+    m.def("pad_channels", &pad_channels,
+          "zero-pad NHWC channel dim to C8 (stem -> glds MFMA path)");
     m.def("pack_mt", &pack_mt);
     m.def("unpack_mt", &unpack_mt);
     m.def("adam_prep_epoch", &adam_prep_epoch);

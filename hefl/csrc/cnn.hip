@@ -871,6 +871,21 @@ __global__ void cast_f32_bf16_kernel(const float* __restrict__ src,
         dst[i] = f2bf(src[i]);
 }
 
+// Zero-pad the channel (last) dim C -> C8: routes C % 8 != 0 stem convs
+// (RefCNN6 conv1 C=3, ResNet stem, cnn4 conv1 C=1) onto the glds MFMA
+// pipeline, which needs 16-B-aligned channel vectors. The padded FLOPs are
+// multiply-by-zero; the pipeline is ~5-10x faster per FLOP than the
+// generic-gather fallback at these shapes (profiles/r02_refcnn6_*).
+__global__ void pad_channels_kernel(const unsigned short* __restrict__ x,
+                                    unsigned short* __restrict__ out,
+                                    int64_t total_out, int C, int C8) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         i < total_out; i += (int64_t)gridDim.x * blockDim.x) {
+        const int c = (int)(i % C8);
+        out[i] = c < C ? x[(i / C8) * C + c] : 0;
+    }
+}
+
 // ---- wgrad: dw[ko, rsc] = sum_pix dy[pix, ko] * xcol[pix, rsc].
 // LDS tiles are PIXEL-major (Dys[32][64], Xs[32][BN]) so BOTH global
 // gathers are 16-B vector loads (ko contiguous in dy; c contiguous in x);
@@ -3275,6 +3290,22 @@ torch::Tensor avgpool_global_bwd(torch::Tensor dy, int64_t H, int64_t W) {
                        fdiv_make((unsigned)C),
                        fdiv_make((unsigned)(H * W)));
     return dx;
+}
+
+torch::Tensor pad_channels(torch::Tensor x, int64_t C8) {
+    CHECK_GPU(x);
+    TORCH_CHECK(x.is_contiguous() && x.dtype() == torch::kBFloat16);
+    const int C = (int)x.size(-1);
+    TORCH_CHECK(C <= C8);
+    auto sizes = x.sizes().vec();
+    sizes.back() = C8;
+    auto out = torch::empty(sizes, x.options());
+    int64_t total = out.numel();
+    auto stream = at::cuda::getCurrentCUDAStream();
+    int blocks = (int)std::min<int64_t>(ceildiv(total, 256), 4096);
+    hipLaunchKernelGGL(pad_channels_kernel, dim3(blocks), dim3(256), 0, stream,
+                       bf_ptr(x), bf_ptr_mut(out), total, C, (int)C8);
+    return out;
 }
 
 torch::Tensor add_relu(torch::Tensor a, torch::Tensor b) {

@@ -31,6 +31,20 @@ def _gpu_dtype(t: torch.Tensor) -> torch.Tensor:
     return t if t.dtype == torch.bfloat16 else t.to(torch.bfloat16)
 
 
+def _want_cpad(C: int, Kout: int) -> bool:
+    """Channel-pad a C % 8 != 0 conv (stem shapes: RefCNN6/ResNet C=3,
+    cnn4 C=1) onto the glds MFMA pipeline. The padded lanes multiply zeros,
+    but the pipeline is ~5-10x faster per FLOP than the generic-gather
+    fallback at these shapes (profiles/r02_refcnn6_kernel_stats.csv:
+    stem fwd 25 TF/s, stem wgrad 10 TF/s on the fallback)."""
+    return C % 8 != 0 and Kout > 16 and Kout % 8 == 0
+
+
+def _pad8(t: torch.Tensor) -> torch.Tensor:
+    C = t.shape[-1]
+    return _C().pad_channels(t.contiguous(), ((C + 7) // 8) * 8)
+
+
 # ---------------------------------------------------------------------------
 # Conv2d (valid padding or explicit pre-pad, square stride), NHWC.
 # ---------------------------------------------------------------------------
@@ -44,8 +58,14 @@ class _Conv2dFn(torch.autograd.Function):
         ctx.relu = relu
         ctx.pad = pad
         ctx.has_bias = b is not None
+        ctx.cpad = False
         if x.is_cuda:
             wb = _gpu_dtype(w)
+            if _want_cpad(x.shape[-1], w.shape[0]):
+                ctx.cpad = True
+                ctx.in_C = x.shape[-1]
+                x = _pad8(x)
+                wb = _pad8(wb)
             bb = b.detach().float() if b is not None else torch.empty(0, device=x.device)
             y = _C().conv2d_fwd(x.contiguous(), wb.contiguous(), bb, stride, relu, pad)
             ctx.save_for_backward(x, wb, y)
@@ -78,6 +98,10 @@ class _Conv2dFn(torch.autograd.Function):
                 if ctx.needs_input_grad[0] else None
             dw = _C().conv2d_wgrad(dy, x.contiguous(), stride, w.shape[1],
                                    w.shape[2], pad)
+            if ctx.cpad:  # drop the zero-padded channel lanes
+                dw = dw[..., :ctx.in_C].contiguous()
+                if dx is not None:
+                    dx = dx[..., :ctx.in_C].contiguous()
         else:
             dy = dy.float()
             if relu:
@@ -150,6 +174,11 @@ class _ConvReluPoolFn(torch.autograd.Function):
     def forward(ctx, x, w, b, pad: int):
         ctx.pad = pad
         wb = _gpu_dtype(w)
+        ctx.cpad = _want_cpad(x.shape[-1], w.shape[0])
+        if ctx.cpad:  # stem shapes ride the glds pipeline on padded channels
+            ctx.in_C = x.shape[-1]
+            x = _pad8(x)
+            wb = _pad8(wb)
         bb = b.detach().float()
         y = _C().conv2d_fwd(x.contiguous(), wb.contiguous(), bb, 1, True, pad)
         p, idx = _C().maxpool2x2_fwd(y)
@@ -164,6 +193,10 @@ class _ConvReluPoolFn(torch.autograd.Function):
             if ctx.needs_input_grad[0] else None
         dw = _C().conv2d_wgrad(dym, x.contiguous(), 1, w.shape[1], w.shape[2],
                                ctx.pad)
+        if ctx.cpad:
+            dw = dw[..., :ctx.in_C].contiguous()
+            if dx is not None:
+                dx = dx[..., :ctx.in_C].contiguous()
         return dx, dw, db, None
 
 

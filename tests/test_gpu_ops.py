@@ -30,6 +30,9 @@ def _close(a, b, atol=None, rel=3e-2):
     (4, 13, 13, 16, 32, 3, 1, 0),    # cnn2 conv2
     (2, 32, 32, 3, 6, 5, 1, 0),      # lenet5 conv1
     (2, 16, 16, 32, 64, 3, 2, 1),    # resnet-style strided+padded
+    (2, 32, 32, 3, 32, 3, 1, 0),     # RefCNN6 conv1 shape class (C=3 -> pad8)
+    (2, 34, 34, 3, 64, 7, 2, 3),     # ResNet stem 7x7 s2 (C=3 -> pad8)
+    (2, 28, 28, 1, 32, 3, 1, 0),     # cnn4 conv1 (C=1 -> pad8)
 ])
 def test_conv2d_fwd_bwd(shape):
     N, H, W, C, K, ks, stride, pad = shape
