@@ -30,6 +30,17 @@ for name, cfg, nw in cases:
     ms_rs, rs = t(lambda: ctx.rescale_tensor(sc))
     ms_dec, _ = t(lambda: ctx.decrypt_tensor(rs, kp.sk))
     mb = ct.data.numel() * 8 / 1e6
-    print(f"{name}: B={B} cts ({mb:.0f} MB) | encrypt {ms_enc:7.2f}ms | "
-          f"modreduce {ms_red:6.2f}ms | ct*plain {ms_mul:6.2f}ms | "
-          f"rescale {ms_rs:6.2f}ms | decrypt {ms_dec:7.2f}ms")
+    line = (f"{name}: B={B} cts ({mb:.0f} MB) | encrypt {ms_enc:7.2f}ms | "
+            f"modreduce {ms_red:6.2f}ms | ct*plain {ms_mul:6.2f}ms | "
+            f"rescale {ms_rs:6.2f}ms | decrypt {ms_dec:7.2f}ms")
+    if ctx.L >= 3:
+        # ct x ct + relinearize (the config #3/#5 aggregation inner op):
+        # batched ciphertext tensor times one shared encrypted denominator
+        import numpy as np
+        from hefl.he.ckks import Ciphertext
+        rlk = ctx.relin_keygen(kp.sk)
+        denom = ctx.encrypt(ctx.encode(np.full(ctx.slots, 0.125)), kp.pk)
+        ctc = Ciphertext(ct.data, ct.scale)
+        ms_mc, _ = t(lambda: ctx.mul_ct(ctc, denom, rlk), iters=5)
+        line += f" | ct*ct+relin {ms_mc:7.2f}ms"
+    print(line)
