@@ -39,6 +39,12 @@ std::vector<torch::Tensor> ks_inner(torch::Tensor dig, torch::Tensor rlk,
                                     torch::Tensor qs, torch::Tensor ratios,
                                     int64_t D, int64_t Lp, int64_t n);
 
+// fft.hip
+torch::Tensor fft_encode(torch::Tensor vals, torch::Tensor tw_enc,
+                         double scale);
+torch::Tensor fft_decode(torch::Tensor coeffs, torch::Tensor tw_dec,
+                         double scale, int64_t k);
+
 // cnn.hip
 torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
                          int64_t stride, bool relu, int64_t pad);
@@ -126,6 +132,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "[.., n] -> [.., L, n] center + per-limb Barrett reduce");
     m.def("ct_mul", &ct_mul, "fused ct x ct tensor product (d0, d1, d2)");
     m.def("ks_inner", &ks_inner, "fused key-switch digit inner product");
+    m.def("fft_encode", &fft_encode,
+          "CKKS special-FFT encode: f64 slots -> int64 coeffs");
+    m.def("fft_decode", &fft_decode,
+          "CKKS special-FFT decode: int64 coeffs -> f32 slots");
     // CNN
     m.def("conv2d_fwd", &conv2d_fwd);
     m.def("conv2d_dgrad", &conv2d_dgrad);

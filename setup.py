@@ -23,6 +23,7 @@ setup(
                 "hefl/csrc/bindings.cpp",
                 "hefl/csrc/ntt.hip",
                 "hefl/csrc/cnn.hip",
+                "hefl/csrc/fft.hip",
             ],
             extra_compile_args={
                 "cxx": ["-O3", "-std=c++17"],

@@ -202,3 +202,30 @@ def test_synth_batch_distribution():
     # different calls draw fresh noise
     x2, _ = ds.batch(idx)
     assert not torch.equal(x, x2)
+
+
+@pytest.mark.parametrize("m", [256, 8192, 16384, 32768])
+def test_hip_fft_encode_decode_matches_oracle(m):
+    """Hand-written special-FFT kernels (hefl/csrc/fft.hip) vs the CPU
+    oracle encoder — the VERDICT r1 item 5 replacement for the torch
+    complex128 at::native path. m=32768 (slots 2^14) exercises the
+    global-stage split (slots > the 8192-slot LDS block)."""
+    import hefl
+    from hefl.he.encoder import Encoder
+    C = hefl.load_extension()
+    enc = Encoder(m)
+    slots = m // 2
+    rng = np.random.default_rng(5)
+    vals = rng.standard_normal((3, slots))
+    scale = 2.0 ** 40
+    ref = enc.encode(vals, scale)              # CPU oracle
+    tw_enc, tw_dec = enc._hip_tables("cuda")
+    got = C.fft_encode(torch.from_numpy(vals).cuda(), tw_enc, scale).cpu()
+    diff = (got - torch.from_numpy(np.asarray(ref, dtype=np.int64))).abs()
+    # identical butterfly order; FMA contraction may flip an int rounding
+    assert (diff <= 1).all(), diff.max()
+    assert (diff == 0).float().mean() > 0.99
+    back = C.fft_decode(got.cuda(), tw_dec, scale, slots).cpu().numpy()
+    ref_back = enc.decode(np.asarray(ref), scale, slots)
+    assert np.abs(back - ref_back).max() < 1e-5
+    assert np.abs(back - vals).max() < 1e-4
