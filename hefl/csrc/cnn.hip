@@ -547,6 +547,179 @@ conv_fwd_glds_kernel(const unsigned short* __restrict__ x,
     }
 }
 
+// ---- BK=64 two-buffer glds forward (guide GEMM-ladder G15 structure:
+// stage -> vmcnt(0) + __syncthreads -> issue next stage -> MFMA, so the
+// next tile's DMA flies under the current tile's 16 MFMAs/wave; st_16x32
+// LDS swizzle kills the 8-way ds_read_b128 conflict of linear 128-B rows).
+// Tiles are sized to the conv shapes: BN=32 for 32-filter layers (the
+// 64-wide tile wasted half its MFMA work there), BN=64 with FM=4 depth
+// otherwise. Requires C % 8 == 0 and KK % 8 == 0 like the BK=32 kernel. ----
+
+__device__ __forceinline__ int swz128(int byte_off) {
+    // st_16x32 swizzle for 128-B LDS rows: XOR byte bit 5 with bit 9
+    // (involutive: the XOR never changes bit 9)
+    return byte_off ^ (((byte_off >> 9) & 1) << 5);
+}
+
+template <int BM, int BN, int WM, int WN, int FM, int FN, bool PAD0>
+__global__ void __launch_bounds__(TPB)
+conv_fwd_glds64_kernel(const unsigned short* __restrict__ x,
+                       const unsigned short* __restrict__ w,
+                       const float* __restrict__ bias,
+                       unsigned short* __restrict__ y,
+                       float* __restrict__ y32,  // split-K accumulator or null
+                       const unsigned short* __restrict__ zbuf, ConvShape s,
+                       int relu, int k_chunks) {
+    constexpr int BK = 64;
+    constexpr int NW = TPB / 64;
+    static_assert(WM * WN == NW, "wave grid must cover the block");
+    static_assert(BM == WM * FM * 16 && BN == WN * FN * 16, "tile mismatch");
+    constexpr int ABYTES = BM * BK * 2;
+    constexpr int RPT = (BM * BK / 8) / TPB;  // A 16-B chunks per thread
+    constexpr int BPT = (BN * BK / 8) / TPB;
+    static_assert(RPT * TPB * 8 == BM * BK && BPT * TPB * 8 == BN * BK);
+    __shared__ unsigned short smem[2 * (BM + BN) * BK];
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+    const int m0 = blockIdx.x * BM;
+    const int n0 = blockIdx.y * BN;
+    const int M = s.N * s.OH * s.OW;
+    const int KK = s.R * s.S * s.C;
+
+    // Per-thread A-chunk decode: the DMA dest is lane-linear (physical
+    // byte p), so the SOURCE chunk is the one whose swizzled logical
+    // offset is p (rule 21: permute the source, read with the same XOR).
+    int a_k8[RPT], a_oh[RPT], a_ow[RPT], a_n[RPT];
+    int64_t a_base[RPT];
+    bool a_ok[RPT];
+#pragma unroll
+    for (int t = 0; t < RPT; ++t) {
+        const int p = (wave + t * NW) * 1024 + lane * 16;
+        const int l = swz128(p);
+        const int row = l >> 7;
+        a_k8[t] = (l & 127) >> 1;  // k offset within the tile (multiple of 8)
+        const int am = m0 + row;
+        a_ok[t] = am < M;
+        a_base[t] = 0;
+        a_n[t] = a_oh[t] = a_ow[t] = 0;
+        if (a_ok[t]) {
+            a_n[t] = am / (s.OH * s.OW);
+            const int rem = am - a_n[t] * (s.OH * s.OW);
+            a_oh[t] = rem / s.OW;
+            a_ow[t] = rem - a_oh[t] * s.OW;
+            // pad == 0 (valid conv): ih = oh*stride + r <= H-1 always, so
+            // the whole gather is branchless off one precomputed base
+            a_base[t] = (((int64_t)a_n[t] * s.H + a_oh[t] * s.stride) * s.W
+                         + a_ow[t] * s.stride) * s.C;
+        }
+    }
+
+    auto stage = [&](int bufsel, int k0) {
+        char* abase = (char*)smem + bufsel * (ABYTES + BN * BK * 2);
+#pragma unroll
+        for (int t = 0; t < RPT; ++t) {
+            const int k = k0 + a_k8[t];
+            const unsigned short* src = zbuf;
+            if (a_ok[t] && k < KK) {
+                const int rs = (int)fdiv((unsigned)k, s.fC);
+                const int c = k - rs * s.C;
+                const int r = (int)fdiv((unsigned)rs, s.fS);
+                const int ss = rs - r * s.S;
+                if (PAD0) {
+                    src = x + a_base[t] + ((int64_t)r * s.W + ss) * s.C + c;
+                } else {
+                    const int ih = a_oh[t] * s.stride + r - s.pad;
+                    const int iw = a_ow[t] * s.stride + ss - s.pad;
+                    if (ih >= 0 && ih < s.H && iw >= 0 && iw < s.W)
+                        src = x + ((((int64_t)a_n[t] * s.H + ih) * s.W + iw)
+                                   * s.C + c);
+                }
+            }
+            glds16(src, abase + (wave + t * NW) * 1024 + lane * 16);
+        }
+        char* bbase = abase + ABYTES;
+#pragma unroll
+        for (int t = 0; t < BPT; ++t) {
+            const int p = (wave + t * NW) * 1024 + lane * 16;
+            const int l = swz128(p);
+            const int korow = l >> 7;
+            const int kc = (l & 127) >> 1;
+            const int ko = n0 + korow;
+            const int k = k0 + kc;
+            const unsigned short* src = zbuf;
+            if (ko < s.Kout && k < KK) src = w + (int64_t)ko * KK + k;
+            glds16(src, bbase + p);
+        }
+    };
+
+    f32x4 acc[FM][FN];
+#pragma unroll
+    for (int i = 0; i < FM; ++i)
+#pragma unroll
+        for (int j = 0; j < FN; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+    const int ksteps = (KK + BK - 1) / BK;
+    const int kc_len = (ksteps + k_chunks - 1) / k_chunks;
+    const int kbeg = blockIdx.z * kc_len * BK;
+    const int kend = min(kbeg + kc_len * BK, KK);
+    const int wm = wave / WN, wn = wave % WN;
+    const int half = lane >> 4, sub = lane & 15;
+    stage(0, kbeg);
+    int buf = 0;
+    for (int k0 = kbeg; k0 < kend; k0 += BK) {
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __syncthreads();  // tile k0 visible everywhere; last tile consumed
+        if (k0 + BK < kend) stage(buf ^ 1, k0 + BK);  // next DMA in flight
+        const char* Ab = (const char*)smem + buf * (ABYTES + BN * BK * 2);
+        const char* Bb = Ab + ABYTES;
+#pragma unroll
+        for (int kk = 0; kk < BK; kk += 32) {
+            bf16x8 a[FM], b[FN];
+#pragma unroll
+            for (int i = 0; i < FM; ++i) {
+                const int row = wm * FM * 16 + i * 16 + sub;
+                a[i] = *reinterpret_cast<const bf16x8*>(
+                    Ab + swz128(row * 128 + (kk + half * 8) * 2));
+            }
+#pragma unroll
+            for (int j = 0; j < FN; ++j) {
+                const int row = wn * FN * 16 + j * 16 + sub;
+                b[j] = *reinterpret_cast<const bf16x8*>(
+                    Bb + swz128(row * 128 + (kk + half * 8) * 2));
+            }
+#pragma unroll
+            for (int i = 0; i < FM; ++i)
+#pragma unroll
+                for (int j = 0; j < FN; ++j)
+                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        a[i], b[j], acc[i][j], 0, 0, 0);
+        }
+        buf ^= 1;
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+#pragma unroll
+    for (int j = 0; j < FN; ++j) {
+        const int col = n0 + wn * FN * 16 + j * 16 + sub;
+        if (col >= s.Kout) continue;
+        const float bv = bias ? bias[col] : 0.f;
+#pragma unroll
+        for (int i = 0; i < FM; ++i)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int row = m0 + wm * FM * 16 + i * 16 + half * 4 + r;
+                if (row >= M) continue;
+                if (k_chunks > 1) {
+                    atomicAdd(y32 + (int64_t)row * s.Kout + col, acc[i][j][r]);
+                } else {
+                    float v = acc[i][j][r] + bv;
+                    if (relu) v = v > 0.f ? v : 0.f;
+                    y[(int64_t)row * s.Kout + col] = f2bf(v);
+                }
+            }
+    }
+}
+
 // ---- dgrad: dx[m=(n,ih,iw), c] = sum_k A(m,k) * B(k,c),
 //      k = (r, s, ko) with ko FASTEST so the dy gather is contiguous;
 //      B(k, c) = w[ko, r, s, c] (strided, small tile) ----
@@ -2661,6 +2834,51 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
     const float* bias = b.numel() ? b.data_ptr<float>() : nullptr;
     auto stream = at::cuda::getCurrentCUDAStream();
     const int KKf = s.R * s.S * s.C;
+    static const bool use64 = [] {
+        const char* e = getenv("HEFL_GLDS64");
+        return !e || e[0] != '0';
+    }();
+    if (use64 && s.Kout > 16 && s.C % 8 == 0 && KKf % 8 == 0) {
+        static torch::Tensor zbuf64;
+        if (!zbuf64.defined() || zbuf64.device() != x.device())
+            zbuf64 = torch::zeros({8}, x.options());
+        const bool narrow = s.Kout <= 32;  // BN=32 tile: no wasted columns
+        const int BM = narrow ? 256 : 128, BN = narrow ? 32 : 64;
+        int tiles = ceildiv(M, BM) * ceildiv(s.Kout, BN);
+        int ksteps = ceildiv(KKf, 64);
+        int k_chunks = 1;
+        if (tiles < 256 && ksteps >= 8)
+            k_chunks = std::max(1, std::min(ksteps / 2, 512 / std::max(tiles, 1)));
+        dim3 grid(ceildiv(M, BM), ceildiv(s.Kout, BN), k_chunks);
+        torch::Tensor y32;
+        float* y32p = nullptr;
+        if (k_chunks > 1) {
+            y32 = torch::zeros({M, s.Kout}, x.options().dtype(torch::kFloat32));
+            y32p = y32.data_ptr<float>();
+        }
+        #define LAUNCH_F64(BM_, BN_, WM_, WN_, FM_, FN_, P0)                  \
+            hipLaunchKernelGGL((conv_fwd_glds64_kernel<BM_, BN_, WM_, WN_,    \
+                                                       FM_, FN_, P0>),       \
+                               grid, dim3(TPB), 0, stream, bf_ptr(x),         \
+                               bf_ptr(w), bias, bf_ptr_mut(y), y32p,          \
+                               bf_ptr(zbuf64), s, relu ? 1 : 0, k_chunks)
+        if (narrow) {
+            if (s.pad == 0) LAUNCH_F64(256, 32, 4, 1, 4, 2, true);
+            else            LAUNCH_F64(256, 32, 4, 1, 4, 2, false);
+        } else {
+            if (s.pad == 0) LAUNCH_F64(128, 64, 2, 2, 4, 2, true);
+            else            LAUNCH_F64(128, 64, 2, 2, 4, 2, false);
+        }
+        #undef LAUNCH_F64
+        if (k_chunks > 1) {
+            int64_t total = (int64_t)M * s.Kout;
+            hipLaunchKernelGGL(linear_epilogue_kernel,
+                               dim3((int)std::min<int64_t>(ceildiv(total, 256), 2048)),
+                               dim3(256), 0, stream, y32p, bias,
+                               bf_ptr_mut(y), total, s.Kout, relu ? 1 : 0, 1);
+        }
+        return y;
+    }
     if (s.Kout > 16 && s.C % 8 == 0 && KKf % 8 == 0) {
         // glds double-buffered pipeline (DMA flight hides under MFMA);
         // small-M/N late layers split the K loop over grid.z to fill the chip

@@ -23,6 +23,9 @@ shapes = [
     ("resnet3x3-512c-4x4", 32, 4, 4, 512, 512, 3, 1, 1),
     ("stem7x7-s2-128x128", 32, 128, 128, 3, 64, 7, 2, 3),
     ("cnn4-c2-111x111", 32, 111, 111, 32, 64, 3, 1, 0),
+    ("refcnn6-c2-125x125-k32", 32, 125, 125, 32, 32, 3, 1, 0),
+    ("refcnn6-c4-28x28-k64", 32, 28, 28, 32, 64, 3, 1, 0),
+    ("refcnn6-c1pad-256x256", 32, 256, 256, 8, 32, 3, 1, 0),
 ]
 for name, N, H, W, Cin, Cout, k, st, pad in shapes:
     x = torch.randn(N, H, W, Cin, device="cuda", dtype=torch.bfloat16)
