@@ -97,7 +97,8 @@ torch::Tensor bias_grad(torch::Tensor dy);
 std::vector<torch::Tensor> relu_bias_bwd(torch::Tensor dy, torch::Tensor y);
 std::vector<torch::Tensor> pool_relu_bias_bwd(torch::Tensor dy,
                                               torch::Tensor idx,
-                                              torch::Tensor y);
+                                              torch::Tensor p, int64_t H,
+                                              int64_t W);
 void adam_prep_epoch(torch::Tensor step, torch::Tensor sched,
                      torch::Tensor hyper, double b1, double b2, int64_t S);
 torch::Tensor pad_channels(torch::Tensor x, int64_t C8);

@@ -1611,170 +1611,139 @@ __global__ void relu_bias_bwd_kernel(const unsigned short* __restrict__ dy,
     }
 }
 
-// Fused maxpool2x2-backward + ReLU-mask + bias grad for the conv->relu->pool
-// trunk pattern (every _SeqCNN trunk layer): one pass produces
-// dym[i] = (pool-argmax hit ? dy_pool[o] : 0) gated by (y_conv[i] > 0) and
-// db[c] = sum_i dym[i,c] — replacing maxpool2x2_bwd_gather + relu_bias_bwd
-// (two full passes over the conv-activation gradient).
-__device__ inline unsigned short pool_relu_gate(
-    const unsigned short* __restrict__ dy, const uint8_t* __restrict__ idx,
-    const unsigned short* __restrict__ y, int64_t r, int c, int K, int H,
-    int W, int OH, int OW, FastDiv fW, FastDiv fH) {
-    const unsigned t1 = fdiv((unsigned)r, fW);
-    const int iw = (int)((unsigned)r - t1 * W);
-    const unsigned t2 = fdiv(t1, fH);
-    const int ih = (int)(t1 - t2 * H);
-    const int n = (int)t2;
-    const int oh = ih >> 1, ow = iw >> 1;
-    if (oh >= OH || ow >= OW) return 0;
-    const int64_t o = (((int64_t)n * OH + oh) * OW + ow) * K + c;
-    const int d = idx[o];
-    if ((d >> 1) != (ih & 1) || (d & 1) != (iw & 1)) return 0;
-    const unsigned short yv = y[r * K + c];
-    return ((yv & 0x7fffu) != 0 && !(yv & 0x8000u)) ? dy[o] : (unsigned short)0;
+// Octet maxpool2x2 forward: one thread per (pixel, 8-channel group), four
+// 16-B x loads + one 16-B y store + one 8-B packed idx store. The scalar
+// form was load-instruction bound ~4x off the HBM roofline.
+__global__ void maxpool_fwd_oct_kernel(const unsigned short* __restrict__ x,
+                                       unsigned short* __restrict__ y,
+                                       uint8_t* __restrict__ idx,
+                                       int64_t total8, int H, int W, int C,
+                                       int OH, int OW, FastDiv fOct,
+                                       FastDiv fOW, FastDiv fOH) {
+    const int noct = C >> 3;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         i < total8; i += (int64_t)gridDim.x * blockDim.x) {
+        unsigned t = fdiv((unsigned)i, fOct);
+        const int oct = (int)((unsigned)i - t * noct);
+        unsigned t2 = fdiv(t, fOW);
+        const int ow = (int)(t - t2 * OW);
+        unsigned t3 = fdiv(t2, fOH);
+        const int oh = (int)(t2 - t3 * OH);
+        const int n = (int)t3;
+        const int64_t r0 = ((((int64_t)n * H + 2 * oh) * W) + 2 * ow) * C
+                           + oct * 8;
+        const u16x8 a = *reinterpret_cast<const u16x8*>(&x[r0]);
+        const u16x8 b = *reinterpret_cast<const u16x8*>(&x[r0 + C]);
+        const u16x8 c8 = *reinterpret_cast<const u16x8*>(&x[r0 + (int64_t)W * C]);
+        const u16x8 d = *reinterpret_cast<const u16x8*>(&x[r0 + (int64_t)W * C + C]);
+        u16x8 best;
+        uint64_t packed = 0;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            float bv = bf2f(a[j]);
+            int bi = 0;
+            const float v1 = bf2f(b[j]), v2 = bf2f(c8[j]), v3 = bf2f(d[j]);
+            if (v1 > bv) { bv = v1; bi = 1; }
+            if (v2 > bv) { bv = v2; bi = 2; }
+            if (v3 > bv) { bv = v3; bi = 3; }
+            best[j] = f2bf(bv);
+            packed |= (uint64_t)bi << (8 * j);
+        }
+        const int64_t oo = (int64_t)t * C + oct * 8;
+        *reinterpret_cast<u16x8*>(&y[oo]) = best;
+        *reinterpret_cast<uint64_t*>(idx + oo) = packed;
+    }
 }
 
-__global__ void pool_relu_bias_bwd_kernel(
+// 2x2-block trunk pool backward: one thread per (input 2x2 block, 8-channel
+// group). The ReLU gate uses the POOLED output p (= y[argmax]; relu ran
+// before pool, so p > 0 iff the argmax cell was active) — the full pre-pool
+// activation y is neither read nor saved for backward anymore. dy/idx/p are
+// read ONCE per output cell (the row-per-thread form re-read them 4x).
+__global__ void pool_relu_bias_bwd2_kernel(
     const unsigned short* __restrict__ dy, const uint8_t* __restrict__ idx,
-    const unsigned short* __restrict__ y, unsigned short* __restrict__ dym,
-    float* __restrict__ db, int64_t M, int K, int rows_per_block, int H,
-    int W, int OH, int OW, FastDiv fW, FastDiv fH) {
-    __shared__ float red[256];
-    const int64_t r0 = (int64_t)blockIdx.x * rows_per_block;
-    const int64_t r1 = min(r0 + rows_per_block, M);
-    if ((K & 7) == 0 && K <= 2048) {
-        // vectorized octet path: 16-B y/dy and 8-B idx loads per thread,
-        // 8 per-channel bias partials in registers (the scalar form was
-        // ~25x off the HBM roofline — pure load-instruction bound)
-        __shared__ float red8[2048];
-        const int noct = K >> 3;
-        int lanes = (int)blockDim.x / noct;
-        lanes = lanes ? (1 << (31 - __clz(lanes))) : 0;
-        const int oct = (lanes ? threadIdx.x % noct : 0);
-        const int rl = (lanes ? threadIdx.x / noct : 0);
-        float acc[8] = {0.f};
-        if (lanes == 0) {  // K > 8*blockDim: strided octets, no lane reduce
-            for (int o = threadIdx.x; o < noct; o += blockDim.x) {
-                float a[8] = {0.f};
-                for (int64_t r = r0; r < r1; ++r) {
-                    const unsigned t1 = fdiv((unsigned)r, fW);
-                    const int iw = (int)((unsigned)r - t1 * W);
-                    const unsigned t2 = fdiv(t1, fH);
-                    const int ih = (int)(t1 - t2 * H);
-                    const int oh = ih >> 1, ow = iw >> 1;
-                    u16x8 g8 = {0, 0, 0, 0, 0, 0, 0, 0};
-                    if (oh < OH && ow < OW) {
-                        const int64_t orow =
-                            (((int64_t)t2 * OH + oh) * OW + ow) * K + o * 8;
-                        const int pos = ((ih & 1) << 1) | (iw & 1);
-                        u16x8 d8 = *reinterpret_cast<const u16x8*>(&dy[orow]);
-                        u16x8 y8 = *reinterpret_cast<const u16x8*>(
-                            &y[r * K + o * 8]);
-                        const uint8_t* i8 = idx + orow;
-#pragma unroll
-                        for (int j = 0; j < 8; ++j)
-                            if (i8[j] == pos && (y8[j] & 0x7fffu) != 0 &&
-                                !(y8[j] & 0x8000u))
-                                g8[j] = d8[j];
-                    }
-                    *reinterpret_cast<u16x8*>(&dym[r * K + o * 8]) = g8;
-#pragma unroll
-                    for (int j = 0; j < 8; ++j) a[j] += bf2f(g8[j]);
-                }
-#pragma unroll
-                for (int j = 0; j < 8; ++j) {
-                    if (gridDim.x == 1) db[o * 8 + j] = a[j];
-                    else atomicAdd(db + o * 8 + j, a[j]);
-                }
-            }
-            return;
-        }
-        if (rl < lanes) {
-            for (int64_t r = r0 + rl; r < r1; r += lanes) {
-                const unsigned t1 = fdiv((unsigned)r, fW);
-                const int iw = (int)((unsigned)r - t1 * W);
-                const unsigned t2 = fdiv(t1, fH);
-                const int ih = (int)(t1 - t2 * H);
-                const int oh = ih >> 1, ow = iw >> 1;
-                u16x8 g8 = {0, 0, 0, 0, 0, 0, 0, 0};
-                if (oh < OH && ow < OW) {
-                    const int64_t orow =
-                        (((int64_t)t2 * OH + oh) * OW + ow) * K + oct * 8;
-                    const int pos = ((ih & 1) << 1) | (iw & 1);
-                    u16x8 d8 = *reinterpret_cast<const u16x8*>(&dy[orow]);
-                    u16x8 y8 = *reinterpret_cast<const u16x8*>(
-                        &y[r * K + oct * 8]);
-                    const uint8_t* i8 = idx + orow;
-#pragma unroll
-                    for (int j = 0; j < 8; ++j)
-                        if (i8[j] == pos && (y8[j] & 0x7fffu) != 0 &&
-                            !(y8[j] & 0x8000u))
-                            g8[j] = d8[j];
-                }
-                *reinterpret_cast<u16x8*>(&dym[r * K + oct * 8]) = g8;
-#pragma unroll
-                for (int j = 0; j < 8; ++j) acc[j] += bf2f(g8[j]);
-            }
-        }
-#pragma unroll
-        for (int j = 0; j < 8; ++j) red8[threadIdx.x * 8 + j] = acc[j];
-        __syncthreads();
-        for (int off = lanes >> 1; off >= 1; off >>= 1) {
-            if (rl < off) {
-#pragma unroll
-                for (int j = 0; j < 8; ++j)
-                    red8[threadIdx.x * 8 + j] +=
-                        red8[(threadIdx.x + off * noct) * 8 + j];
-            }
-            __syncthreads();
-        }
-        if (rl == 0) {
+    const unsigned short* __restrict__ p, unsigned short* __restrict__ dym,
+    float* __restrict__ db, int64_t total8, int K, int H, int W, int OH,
+    int OW, int HB, int WB, FastDiv fOct, FastDiv fWB, FastDiv fHB) {
+    extern __shared__ float dbs[];  // [K]
+    const int noct = K >> 3;
+    for (int c = threadIdx.x; c < K; c += blockDim.x) dbs[c] = 0.f;
+    __syncthreads();
+    float acc[8] = {0.f};
+    int my_oct = -1;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         i < total8; i += (int64_t)gridDim.x * blockDim.x) {
+        unsigned t = fdiv((unsigned)i, fOct);
+        const int oct = (int)((unsigned)i - t * noct);
+        my_oct = oct;  // constant per thread: stride % noct == 0 (noct pow2)
+        unsigned t2 = fdiv(t, fWB);
+        const int wb = (int)(t - t2 * WB);
+        unsigned t3 = fdiv(t2, fHB);
+        const int hb = (int)(t2 - t3 * HB);
+        const int n = (int)t3;
+        const int ih = 2 * hb, iw = 2 * wb;
+        u16x8 g[4] = {};
+        if (hb < OH && wb < OW) {
+            const int64_t o = (((int64_t)n * OH + hb) * OW + wb) * K + oct * 8;
+            const u16x8 d8 = *reinterpret_cast<const u16x8*>(&dy[o]);
+            const u16x8 p8 = *reinterpret_cast<const u16x8*>(&p[o]);
+            const uint64_t i8 = *reinterpret_cast<const uint64_t*>(idx + o);
 #pragma unroll
             for (int j = 0; j < 8; ++j) {
-                if (gridDim.x == 1)
-                    db[oct * 8 + j] = red8[threadIdx.x * 8 + j];
-                else
-                    atomicAdd(db + oct * 8 + j, red8[threadIdx.x * 8 + j]);
+                if ((p8[j] & 0x7fffu) != 0 && !(p8[j] & 0x8000u)) {
+                    g[(i8 >> (8 * j)) & 3][j] = d8[j];
+                    acc[j] += bf2f(d8[j]);
+                }
             }
         }
-        return;
-    }
-    if (K >= (int)blockDim.x) {  // one channel per thread, strided
-        for (int c = threadIdx.x; c < K; c += blockDim.x) {
-            float acc = 0.f;
-            for (int64_t r = r0; r < r1; ++r) {
-                unsigned short g =
-                    pool_relu_gate(dy, idx, y, r, c, K, H, W, OH, OW, fW, fH);
-                dym[r * K + c] = g;
-                acc += bf2f(g);
-            }
-            if (gridDim.x == 1) db[c] = acc;
-            else atomicAdd(db + c, acc);
-        }
-        return;
-    }
-    int lanes = (int)blockDim.x / K;
-    lanes = 1 << (31 - __clz(lanes));
-    const int c = threadIdx.x % K;
-    const int rl = threadIdx.x / K;
-    float acc = 0.f;
-    if (rl < lanes) {
-        for (int64_t r = r0 + rl; r < r1; r += lanes) {
-            unsigned short g =
-                pool_relu_gate(dy, idx, y, r, c, K, H, W, OH, OW, fW, fH);
-            dym[r * K + c] = g;
-            acc += bf2f(g);
+        const int64_t r0 = (((int64_t)n * H + ih) * W + iw) * K + oct * 8;
+        *reinterpret_cast<u16x8*>(&dym[r0]) = g[0];
+        if (iw + 1 < W)
+            *reinterpret_cast<u16x8*>(&dym[r0 + K]) = g[1];
+        if (ih + 1 < H) {
+            *reinterpret_cast<u16x8*>(&dym[r0 + (int64_t)W * K]) = g[2];
+            if (iw + 1 < W)
+                *reinterpret_cast<u16x8*>(&dym[r0 + (int64_t)W * K + K]) = g[3];
         }
     }
-    red[threadIdx.x] = acc;
+    if (my_oct >= 0) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+            if (acc[j] != 0.f) atomicAdd(dbs + my_oct * 8 + j, acc[j]);
+    }
     __syncthreads();
-    for (int off = lanes >> 1; off >= 1; off >>= 1) {
-        if (rl < off) red[threadIdx.x] += red[threadIdx.x + off * K];
-        __syncthreads();
-    }
-    if (rl == 0) {
-        if (gridDim.x == 1) db[c] = red[threadIdx.x];
-        else atomicAdd(db + c, red[threadIdx.x]);
+    for (int c = threadIdx.x; c < K; c += blockDim.x)
+        if (dbs[c] != 0.f) atomicAdd(db + c, dbs[c]);
+}
+
+// Scalar fallback for K % 8 != 0 trunk layers (LeNet-5's 6/16 filters):
+// one thread per input cell, ReLU gate on the POOLED output p (= the
+// argmax cell's activation; relu ran before pool).
+__global__ void pool_relu_bias_bwd_scalar_kernel(
+    const unsigned short* __restrict__ dy, const uint8_t* __restrict__ idx,
+    const unsigned short* __restrict__ p, unsigned short* __restrict__ dym,
+    float* __restrict__ db, int64_t total, int K, int H, int W, int OH,
+    int OW, FastDiv fK, FastDiv fW, FastDiv fH) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         i < total; i += (int64_t)gridDim.x * blockDim.x) {
+        const unsigned t0 = fdiv((unsigned)i, fK);
+        const int c = (int)((unsigned)i - t0 * K);
+        const unsigned t1 = fdiv(t0, fW);
+        const int iw = (int)(t0 - t1 * W);
+        const unsigned t2 = fdiv(t1, fH);
+        const int ih = (int)(t1 - t2 * H);
+        const int oh = ih >> 1, ow = iw >> 1;
+        unsigned short g = 0;
+        if (oh < OH && ow < OW) {
+            const int64_t o = (((int64_t)t2 * OH + oh) * OW + ow) * K + c;
+            const int d = idx[o];
+            const unsigned short pv = p[o];
+            if ((d >> 1) == (ih & 1) && (d & 1) == (iw & 1) &&
+                (pv & 0x7fffu) != 0 && !(pv & 0x8000u))
+                g = dy[o];
+        }
+        dym[i] = g;
+        if (g) atomicAdd(db + c, bf2f(g));
     }
 }
 
@@ -2834,9 +2803,15 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
     const float* bias = b.numel() ? b.data_ptr<float>() : nullptr;
     auto stream = at::cuda::getCurrentCUDAStream();
     const int KKf = s.R * s.S * s.C;
+    // Measured on MI355X (gpurun_out/bench_conv_glds64 vs _glds32): the
+    // BK=64 two-buffer kernel LOSES 8-30% to the BK=32 three-buffer one at
+    // every conv shape here — K loops are 2-18 steps, so the prologue DMA
+    // latency (3-buf pre-stages two tiles, 2-buf one) outweighs the halved
+    // barrier count; the shapes are gather-bound, not MFMA-issue-bound.
+    // Kept behind HEFL_GLDS64=1 as the documented A/B.
     static const bool use64 = [] {
         const char* e = getenv("HEFL_GLDS64");
-        return !e || e[0] != '0';
+        return e && e[0] == '1';
     }();
     if (use64 && s.Kout > 16 && s.C % 8 == 0 && KKf % 8 == 0) {
         static torch::Tensor zbuf64;
@@ -3193,6 +3168,16 @@ std::vector<torch::Tensor> maxpool2x2_fwd(torch::Tensor x) {
     auto y = torch::empty({N, OH, OW, C}, x.options());
     auto idx = torch::empty({N, OH, OW, C}, x.options().dtype(torch::kUInt8));
     int64_t total = (int64_t)N * OH * OW * C;
+    if ((C & 7) == 0) {
+        int64_t total8 = total >> 3;
+        int blocks = (int)std::min<int64_t>(ceildiv(total8, 256), 4096);
+        hipLaunchKernelGGL(maxpool_fwd_oct_kernel, dim3(blocks), dim3(256), 0,
+                           at::cuda::getCurrentCUDAStream(), bf_ptr(x),
+                           bf_ptr_mut(y), idx.data_ptr<uint8_t>(), total8, H,
+                           W, C, OH, OW, fdiv_make((unsigned)(C >> 3)),
+                           fdiv_make((unsigned)OW), fdiv_make((unsigned)OH));
+        return {y, idx};
+    }
     int blocks = std::min<int64_t>(ceildiv(total, 256), 2048);
     hipLaunchKernelGGL(maxpool_fwd_kernel, dim3(blocks), dim3(256), 0,
                        at::cuda::getCurrentCUDAStream(), bf_ptr(x),
@@ -3237,30 +3222,37 @@ std::vector<torch::Tensor> relu_bias_bwd(torch::Tensor dy, torch::Tensor y) {
 
 std::vector<torch::Tensor> pool_relu_bias_bwd(torch::Tensor dy,
                                               torch::Tensor idx,
-                                              torch::Tensor y) {
+                                              torch::Tensor p, int64_t H,
+                                              int64_t W) {
     CHECK_GPU(dy);
     auto dyc = dy.contiguous();
-    const int N = (int)y.size(0), H = (int)y.size(1), W = (int)y.size(2),
-              K = (int)y.size(3);
-    const int OH = (int)dyc.size(1), OW = (int)dyc.size(2);
-    const int64_t M = (int64_t)N * H * W;
-    auto dym = torch::empty_like(y);
-    int rpb = (int)std::max<int64_t>(64, (M + 511) / 512);
-    if ((K & 7) == 0 && K <= 2048) {
-        // octet path: every row-lane must get >= 2 rows or the LDS reduce
-        // dominates (small K => many lanes)
-        int lanes = 256 / (K / 8);
-        lanes = lanes ? (1 << (31 - __builtin_clz(lanes))) : 1;
-        rpb = std::max(rpb, 2 * lanes);
+    const int N = (int)dyc.size(0), OH = (int)dyc.size(1),
+              OW = (int)dyc.size(2), K = (int)dyc.size(3);
+    auto dym = torch::empty({N, H, W, (int64_t)K}, dyc.options());
+    auto db = torch::zeros({K}, dyc.options().dtype(torch::kFloat32));
+    auto stream = at::cuda::getCurrentCUDAStream();
+    const int noct = K >> 3;
+    if ((K & 7) == 0 && (noct & (noct - 1)) == 0 && noct <= 256) {
+        // 2x2-block octet kernel: dy/idx/p read once per output cell
+        const int HB = (int)((H + 1) / 2), WB = (int)((W + 1) / 2);
+        const int64_t total8 = (int64_t)N * HB * WB * noct;
+        int blocks = (int)std::min<int64_t>(ceildiv(total8, 256), 4096);
+        hipLaunchKernelGGL(pool_relu_bias_bwd2_kernel, dim3(blocks),
+                           dim3(256), K * sizeof(float), stream, bf_ptr(dyc),
+                           idx.data_ptr<uint8_t>(), bf_ptr(p),
+                           bf_ptr_mut(dym), db.data_ptr<float>(), total8, K,
+                           (int)H, (int)W, OH, OW, HB, WB,
+                           fdiv_make((unsigned)noct), fdiv_make((unsigned)WB),
+                           fdiv_make((unsigned)HB));
+        return {dym, db};
     }
-    int nblk = (int)((M + rpb - 1) / rpb);
-    auto db = nblk == 1
-                  ? torch::empty({K}, dyc.options().dtype(torch::kFloat32))
-                  : torch::zeros({K}, dyc.options().dtype(torch::kFloat32));
-    hipLaunchKernelGGL(pool_relu_bias_bwd_kernel, dim3(nblk), dim3(256), 0,
-                       at::cuda::getCurrentCUDAStream(), bf_ptr(dyc),
-                       idx.data_ptr<uint8_t>(), bf_ptr(y), bf_ptr_mut(dym),
-                       db.data_ptr<float>(), M, K, rpb, H, W, OH, OW, fdiv_make((unsigned)W),
+    const int64_t total = (int64_t)N * H * W * K;
+    int blocks = (int)std::min<int64_t>(ceildiv(total, 256), 4096);
+    hipLaunchKernelGGL(pool_relu_bias_bwd_scalar_kernel, dim3(blocks),
+                       dim3(256), 0, stream, bf_ptr(dyc),
+                       idx.data_ptr<uint8_t>(), bf_ptr(p), bf_ptr_mut(dym),
+                       db.data_ptr<float>(), total, K, (int)H, (int)W, OH, OW,
+                       fdiv_make((unsigned)K), fdiv_make((unsigned)W),
                        fdiv_make((unsigned)H));
     return {dym, db};
 }

@@ -181,14 +181,18 @@ class _ConvReluPoolFn(torch.autograd.Function):
             wb = _pad8(wb)
         bb = b.detach().float()
         y = _C().conv2d_fwd(x.contiguous(), wb.contiguous(), bb, 1, True, pad)
+        ctx.conv_hw = (y.shape[1], y.shape[2])
         p, idx = _C().maxpool2x2_fwd(y)
-        ctx.save_for_backward(x, wb, y, idx)
+        # backward gates on the POOLED output p (= y at the argmax; relu ran
+        # before pool) — the full pre-pool activation y is not retained
+        ctx.save_for_backward(x, wb, p, idx)
         return p
 
     @staticmethod
     def backward(ctx, dy):
-        x, w, y, idx = ctx.saved_tensors
-        dym, db = _C().pool_relu_bias_bwd(dy.contiguous(), idx, y)
+        x, w, p, idx = ctx.saved_tensors
+        ch, cw = ctx.conv_hw
+        dym, db = _C().pool_relu_bias_bwd(dy.contiguous(), idx, p, ch, cw)
         dx = _C().conv2d_dgrad(dym, w, 1, x.shape[1], x.shape[2], ctx.pad) \
             if ctx.needs_input_grad[0] else None
         dw = _C().conv2d_wgrad(dym, x.contiguous(), 1, w.shape[1], w.shape[2],

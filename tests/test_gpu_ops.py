@@ -245,7 +245,8 @@ def test_graphed_per_step_path_with_callbacks():
 @pytest.mark.parametrize("shape", [
     (4, 28, 28, 1, 16, 3),   # cnn2 block1
     (4, 13, 13, 16, 32, 3),  # cnn2 block2 (odd H/W: floor pooling)
-    (2, 32, 32, 3, 6, 5),    # lenet5 block1
+    (2, 32, 32, 3, 6, 5),    # lenet5 block1 (K=6: scalar fallback path)
+    (2, 30, 30, 8, 64, 3),   # K=64 octet 2x2-block path
 ])
 def test_conv_relu_pool_fused_matches_composed(shape):
     """Fused trunk block (one pool+relu+bias backward kernel) against the
