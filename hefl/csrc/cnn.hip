@@ -1200,7 +1200,11 @@ conv_wgrad_kernel(const unsigned short* __restrict__ dy,
                 int ko = m0 + wm * FM * 16 + i * 16 + (lane >> 4) * 4 + r;
                 if (ko >= M) continue;
                 if (k_chunks > 1)
-                    atomicAdd(dw + (int64_t)ko * NN + col, acc[i][j][r]);
+                    // per-z-chunk slab row (plain store): blocks of one z
+                    // cover dw disjointly, so no atomics and no zero-fill;
+                    // sum_slabs_f32_kernel reduces the rows afterwards
+                    dw[(int64_t)blockIdx.z * M * NN + (int64_t)ko * NN + col]
+                        = acc[i][j][r];
                 else
                     dw[(int64_t)ko * NN + col] = acc[i][j][r];
             }
@@ -1334,7 +1338,11 @@ conv_wgrad_glds_kernel(const unsigned short* __restrict__ dy,
                 int ko = m0 + wm * FM * 16 + i * 16 + (lane >> 4) * 4 + r;
                 if (ko >= M) continue;
                 if (k_chunks > 1)
-                    atomicAdd(dw + (int64_t)ko * NN + col, acc[i][j][r]);
+                    // per-z-chunk slab row (plain store): blocks of one z
+                    // cover dw disjointly, so no atomics and no zero-fill;
+                    // sum_slabs_f32_kernel reduces the rows afterwards
+                    dw[(int64_t)blockIdx.z * M * NN + (int64_t)ko * NN + col]
+                        = acc[i][j][r];
                 else
                     dw[(int64_t)ko * NN + col] = acc[i][j][r];
             }
@@ -3025,11 +3033,18 @@ torch::Tensor conv2d_wgrad(torch::Tensor dy, torch::Tensor x, int64_t stride,
                            dw.data_ptr<float>(), total, kc);
         return dw;
     }
-    auto dw = k_chunks > 1
-                  ? torch::zeros({s.Kout, R, S, s.C},
-                                 x.options().dtype(torch::kFloat32))
-                  : torch::empty({s.Kout, R, S, s.C},
-                                 x.options().dtype(torch::kFloat32));
+    auto dw = torch::empty({s.Kout, R, S, s.C},
+                           x.options().dtype(torch::kFloat32));
+    const int64_t total = (int64_t)s.Kout * NN;
+    torch::Tensor slab;
+    float* outp = dw.data_ptr<float>();
+    if (k_chunks > 1) {
+        // per-chunk slab rows (plain stores, no zero-fill, no atomic
+        // contention across the up-to-256 z-chunks), reduced below
+        slab = torch::empty({k_chunks, total},
+                            x.options().dtype(torch::kFloat32));
+        outp = slab.data_ptr<float>();
+    }
     dim3 grid(ceildiv(s.Kout, 64), ceildiv(NN, bn), k_chunks);
     if (glds_ok) {
         static torch::Tensor zbuf;
@@ -3037,16 +3052,21 @@ torch::Tensor conv2d_wgrad(torch::Tensor dy, torch::Tensor x, int64_t stride,
             zbuf = torch::zeros({8}, dy.options());
         hipLaunchKernelGGL((conv_wgrad_glds_kernel<64, 2, 2, 2, 2>), grid,
                            dim3(TPB), 0, stream, bf_ptr(dy), bf_ptr(x),
-                           dw.data_ptr<float>(), bf_ptr(zbuf), s, k_chunks);
+                           outp, bf_ptr(zbuf), s, k_chunks);
     } else if (big) {
         hipLaunchKernelGGL((conv_wgrad_kernel<64, 2, 2, 2, 2>), grid, dim3(TPB),
                            0, stream, bf_ptr(dy), bf_ptr(x),
-                           dw.data_ptr<float>(), s, k_chunks);
+                           outp, s, k_chunks);
     } else {
         hipLaunchKernelGGL((conv_wgrad_kernel<16, 4, 1, 1, 1>), grid, dim3(TPB),
                            0, stream, bf_ptr(dy), bf_ptr(x),
-                           dw.data_ptr<float>(), s, k_chunks);
+                           outp, s, k_chunks);
     }
+    if (k_chunks > 1)
+        hipLaunchKernelGGL(sum_slabs_f32_kernel,
+                           dim3((int)((total + 15) / 16)), dim3(256), 0,
+                           stream, outp, dw.data_ptr<float>(), total,
+                           k_chunks);
     return dw;
 }
 
