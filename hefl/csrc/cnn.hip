@@ -381,6 +381,18 @@ conv_fwd_kernel(const unsigned short* __restrict__ x,
 typedef uint32_t __attribute__((address_space(3))) lds_u32_t;
 typedef const uint32_t __attribute__((address_space(1))) glb_u32_t;
 
+// XCD-aware tile remap: hardware dispatch round-robins workgroups over the
+// 8 XCDs (each with its own L2); adjacent conv M-tiles share R-1 input
+// halo rows, so map CONSECUTIVE tile ranges onto one XCD to keep the halo
+// reuse in a single L2. Bijective for any nwg (guide formula).
+__device__ __forceinline__ int xcd_remap(int wg, int nwg) {
+    if (nwg < 16) return wg;
+    const int xcd = wg & 7;
+    const int q = nwg >> 3, r = nwg & 7;
+    const int base = xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q;
+    return base + (wg >> 3);
+}
+
 __device__ __forceinline__ void glds16(const unsigned short* src, void* lds_base) {
     __builtin_amdgcn_global_load_lds((glb_u32_t*)src, (lds_u32_t*)lds_base, 16,
                                      0, 0);
@@ -410,7 +422,7 @@ conv_fwd_glds_kernel(const unsigned short* __restrict__ x,
     const int tid = threadIdx.x;
     const int wave = tid >> 6;
     const int lane = tid & 63;
-    const int m0 = blockIdx.x * BM;
+    const int m0 = xcd_remap(blockIdx.x, gridDim.x) * BM;
     const int n0 = blockIdx.y * BN;
     const int M = s.N * s.OH * s.OW;
     const int KK = s.R * s.S * s.C;
@@ -908,7 +920,7 @@ conv_dgrad_glds_kernel(const unsigned short* __restrict__ dy,
     const int tid = threadIdx.x;
     const int wave = tid >> 6;
     const int lane = tid & 63;
-    const int m0 = blockIdx.x * BM;
+    const int m0 = xcd_remap(blockIdx.x, gridDim.x) * BM;
     const int n0 = blockIdx.y * BN;
     const int M = s.N * s.H * s.W;
     const int KK = s.Kout * s.R * s.S;
@@ -1200,11 +1212,7 @@ conv_wgrad_kernel(const unsigned short* __restrict__ dy,
                 int ko = m0 + wm * FM * 16 + i * 16 + (lane >> 4) * 4 + r;
                 if (ko >= M) continue;
                 if (k_chunks > 1)
-                    // per-z-chunk slab row (plain store): blocks of one z
-                    // cover dw disjointly, so no atomics and no zero-fill;
-                    // sum_slabs_f32_kernel reduces the rows afterwards
-                    dw[(int64_t)blockIdx.z * M * NN + (int64_t)ko * NN + col]
-                        = acc[i][j][r];
+                    atomicAdd(dw + (int64_t)ko * NN + col, acc[i][j][r]);
                 else
                     dw[(int64_t)ko * NN + col] = acc[i][j][r];
             }
@@ -1338,11 +1346,7 @@ conv_wgrad_glds_kernel(const unsigned short* __restrict__ dy,
                 int ko = m0 + wm * FM * 16 + i * 16 + (lane >> 4) * 4 + r;
                 if (ko >= M) continue;
                 if (k_chunks > 1)
-                    // per-z-chunk slab row (plain store): blocks of one z
-                    // cover dw disjointly, so no atomics and no zero-fill;
-                    // sum_slabs_f32_kernel reduces the rows afterwards
-                    dw[(int64_t)blockIdx.z * M * NN + (int64_t)ko * NN + col]
-                        = acc[i][j][r];
+                    atomicAdd(dw + (int64_t)ko * NN + col, acc[i][j][r]);
                 else
                     dw[(int64_t)ko * NN + col] = acc[i][j][r];
             }
@@ -1690,7 +1694,10 @@ __global__ void pool_relu_bias_bwd2_kernel(
         const int hb = (int)(t2 - t3 * HB);
         const int n = (int)t3;
         const int ih = 2 * hb, iw = 2 * wb;
-        u16x8 g[4] = {};
+        // four separate vectors with predicated selects: a g[pos][j] array
+        // with a RUNTIME pos index would force the whole array to scratch
+        // memory (dynamic indexing defeats register allocation)
+        u16x8 g0 = {}, g1 = {}, g2 = {}, g3 = {};
         if (hb < OH && wb < OW) {
             const int64_t o = (((int64_t)n * OH + hb) * OW + wb) * K + oct * 8;
             const u16x8 d8 = *reinterpret_cast<const u16x8*>(&dy[o]);
@@ -1698,20 +1705,24 @@ __global__ void pool_relu_bias_bwd2_kernel(
             const uint64_t i8 = *reinterpret_cast<const uint64_t*>(idx + o);
 #pragma unroll
             for (int j = 0; j < 8; ++j) {
-                if ((p8[j] & 0x7fffu) != 0 && !(p8[j] & 0x8000u)) {
-                    g[(i8 >> (8 * j)) & 3][j] = d8[j];
-                    acc[j] += bf2f(d8[j]);
-                }
+                const bool act = (p8[j] & 0x7fffu) != 0 && !(p8[j] & 0x8000u);
+                const int pos = (int)((i8 >> (8 * j)) & 3);
+                const unsigned short dv = act ? d8[j] : (unsigned short)0;
+                if (act) acc[j] += bf2f(d8[j]);
+                g0[j] = pos == 0 ? dv : (unsigned short)0;
+                g1[j] = pos == 1 ? dv : (unsigned short)0;
+                g2[j] = pos == 2 ? dv : (unsigned short)0;
+                g3[j] = pos == 3 ? dv : (unsigned short)0;
             }
         }
         const int64_t r0 = (((int64_t)n * H + ih) * W + iw) * K + oct * 8;
-        *reinterpret_cast<u16x8*>(&dym[r0]) = g[0];
+        *reinterpret_cast<u16x8*>(&dym[r0]) = g0;
         if (iw + 1 < W)
-            *reinterpret_cast<u16x8*>(&dym[r0 + K]) = g[1];
+            *reinterpret_cast<u16x8*>(&dym[r0 + K]) = g1;
         if (ih + 1 < H) {
-            *reinterpret_cast<u16x8*>(&dym[r0 + (int64_t)W * K]) = g[2];
+            *reinterpret_cast<u16x8*>(&dym[r0 + (int64_t)W * K]) = g2;
             if (iw + 1 < W)
-                *reinterpret_cast<u16x8*>(&dym[r0 + (int64_t)W * K + K]) = g[3];
+                *reinterpret_cast<u16x8*>(&dym[r0 + (int64_t)W * K + K]) = g3;
         }
     }
     if (my_oct >= 0) {
@@ -3033,18 +3044,14 @@ torch::Tensor conv2d_wgrad(torch::Tensor dy, torch::Tensor x, int64_t stride,
                            dw.data_ptr<float>(), total, kc);
         return dw;
     }
-    auto dw = torch::empty({s.Kout, R, S, s.C},
-                           x.options().dtype(torch::kFloat32));
-    const int64_t total = (int64_t)s.Kout * NN;
-    torch::Tensor slab;
-    float* outp = dw.data_ptr<float>();
-    if (k_chunks > 1) {
-        // per-chunk slab rows (plain stores, no zero-fill, no atomic
-        // contention across the up-to-256 z-chunks), reduced below
-        slab = torch::empty({k_chunks, total},
-                            x.options().dtype(torch::kFloat32));
-        outp = slab.data_ptr<float>();
-    }
+    // measured: slab-rows + reduce LOSES to fp32 atomics here (CDNA4 L2
+    // atomics absorb the z-chunk contention; the slab variant paid extra
+    // write+read traffic) — gpurun_out/bench_conv_p5 vs bench_conv_glds32
+    auto dw = k_chunks > 1
+                  ? torch::zeros({s.Kout, R, S, s.C},
+                                 x.options().dtype(torch::kFloat32))
+                  : torch::empty({s.Kout, R, S, s.C},
+                                 x.options().dtype(torch::kFloat32));
     dim3 grid(ceildiv(s.Kout, 64), ceildiv(NN, bn), k_chunks);
     if (glds_ok) {
         static torch::Tensor zbuf;
@@ -3052,21 +3059,16 @@ torch::Tensor conv2d_wgrad(torch::Tensor dy, torch::Tensor x, int64_t stride,
             zbuf = torch::zeros({8}, dy.options());
         hipLaunchKernelGGL((conv_wgrad_glds_kernel<64, 2, 2, 2, 2>), grid,
                            dim3(TPB), 0, stream, bf_ptr(dy), bf_ptr(x),
-                           outp, bf_ptr(zbuf), s, k_chunks);
+                           dw.data_ptr<float>(), bf_ptr(zbuf), s, k_chunks);
     } else if (big) {
         hipLaunchKernelGGL((conv_wgrad_kernel<64, 2, 2, 2, 2>), grid, dim3(TPB),
                            0, stream, bf_ptr(dy), bf_ptr(x),
-                           outp, s, k_chunks);
+                           dw.data_ptr<float>(), s, k_chunks);
     } else {
         hipLaunchKernelGGL((conv_wgrad_kernel<16, 4, 1, 1, 1>), grid, dim3(TPB),
                            0, stream, bf_ptr(dy), bf_ptr(x),
-                           outp, s, k_chunks);
+                           dw.data_ptr<float>(), s, k_chunks);
     }
-    if (k_chunks > 1)
-        hipLaunchKernelGGL(sum_slabs_f32_kernel,
-                           dim3((int)((total + 15) / 16)), dim3(256), 0,
-                           stream, outp, dw.data_ptr<float>(), total,
-                           k_chunks);
     return dw;
 }
 
