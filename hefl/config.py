@@ -72,6 +72,10 @@ class FLConfig:
     test_samples: int = 400
     encrypted: bool = True                # False -> plaintext FedAvg (config #1)
     denom_mode: str = "plain"             # "encrypted" -> ct x ct + relin in aggregation (config #3)
+    # training-data augmentation: "none" | "hflip" | "full" ("full" = the
+    # reference's ImageDataGenerator set: shear 0.2, zoom 0.2, h-flip —
+    # FLPyfhelin.py:80-86 — applied in the data generator)
+    augment: str = "none"
     seed: int = 1234
 
 
@@ -124,10 +128,11 @@ def preset(name: str) -> RunConfig:
             he=HEConfig(m=32768, scale_bits=40, q_bits=(60, 40, 40, 40)),
         )
     if name in ("reference", "refcnn6"):
-        # The reference's own config: 2 clients, 6-conv CNN, 256x256x3, 2 classes
+        # The reference's own config: 2 clients, 6-conv CNN, 256x256x3,
+        # 2 classes, full augmentation (FLPyfhelin.py:80-86,118-146)
         return RunConfig(
             model=ModelConfig("refcnn6", (256, 256, 3), 2),
-            fl=FLConfig(n_clients=2, encrypted=True),
+            fl=FLConfig(n_clients=2, encrypted=True, augment="full"),
             he=HEConfig(m=8192),
         )
     raise KeyError(f"unknown preset: {name}")

@@ -58,9 +58,11 @@ torch::Tensor linear_dgrad(torch::Tensor dy, torch::Tensor w);
 torch::Tensor linear_wgrad(torch::Tensor dy, torch::Tensor x);
 std::vector<torch::Tensor> maxpool2x2_fwd(torch::Tensor x);
 torch::Tensor synth_batch(torch::Tensor templates, torch::Tensor labels,
-                          int64_t seed);
+                          int64_t seed, double zoom, double shear,
+                          int64_t flip);
 torch::Tensor synth_batch_g(torch::Tensor templates, torch::Tensor labels,
-                            torch::Tensor seed_buf, int64_t salt);
+                            torch::Tensor seed_buf, int64_t salt, double zoom,
+                            double shear, int64_t flip);
 torch::Tensor maxpool2x2_bwd(torch::Tensor dy, torch::Tensor idx, int64_t H,
                              int64_t W);
 std::vector<torch::Tensor> softmax_xent_fwd(torch::Tensor logits,
@@ -145,8 +147,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("linear_dgrad", &linear_dgrad);
     m.def("linear_wgrad", &linear_wgrad);
     m.def("maxpool2x2_fwd", &maxpool2x2_fwd);
-    m.def("synth_batch", &synth_batch);
-    m.def("synth_batch_g", &synth_batch_g);
+    m.def("synth_batch", &synth_batch, py::arg("templates"), py::arg("labels"),
+          py::arg("seed"), py::arg("zoom") = 0.0, py::arg("shear") = 0.0,
+          py::arg("flip") = 0);
+    m.def("synth_batch_g", &synth_batch_g, py::arg("templates"),
+          py::arg("labels"), py::arg("seed_buf"), py::arg("salt"),
+          py::arg("zoom") = 0.0, py::arg("shear") = 0.0, py::arg("flip") = 0);
     m.def("maxpool2x2_bwd", &maxpool2x2_bwd);
     m.def("softmax_xent_fwd", &softmax_xent_fwd);
     m.def("softmax_xent_bwd", &softmax_xent_bwd);

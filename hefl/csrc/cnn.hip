@@ -1485,23 +1485,71 @@ conv_wgrad_small_kernel(const unsigned short* __restrict__ dy,
 // Replaces the staging chain randn/index/mul/add/sigmoid/cast (~6 fp32
 // passes over the epoch tensor) with a single bandwidth-bound pass
 // (hefl/data/synthetic.py batch()).
+__device__ __forceinline__ unsigned long long splitmix64(unsigned long long z) {
+    z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
+    z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
+    return z ^ (z >> 31);
+}
+
+// AUG: per-sample random zoom/shear/h-flip applied by sampling the class
+// template at inverse-affine coordinates (bilinear, border-clamped) — the
+// MI355X-native analog of the reference's ImageDataGenerator augmentation
+// (shear_range 0.2, zoom_range 0.2, horizontal_flip; FLPyfhelin.py:80-86),
+// fused into the same one-pass generator so the epoch graph keeps staging
+// data on-device. zr/sr = 0 and flip = 0 reproduces the plain path.
+template <bool AUG>
 __global__ void synth_batch_kernel(const float* __restrict__ T,
                                    const int64_t* __restrict__ lab,
                                    unsigned short* __restrict__ out,
                                    int64_t per_img, int64_t total,
                                    unsigned long long seed, FastDiv fPer,
-                                   const long long* __restrict__ seed_buf) {
+                                   const long long* __restrict__ seed_buf,
+                                   int H, int W, int C, FastDiv fC, FastDiv fW,
+                                   float zr, float sr, int flip) {
     if (seed_buf) seed += (unsigned long long)seed_buf[0];  // graph-replay seed
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
          i += (int64_t)gridDim.x * blockDim.x) {
         const int64_t n = (int64_t)fdiv((unsigned)i, fPer);
         const int64_t off = i - n * per_img;
-        const float t = T[lab[n] * per_img + off];
+        float t;
+        if (!AUG) {
+            t = T[lab[n] * per_img + off];
+        } else {
+            // decode (h, w, c); per-SAMPLE transform params from their own
+            // hash chain (seed ^ sample index)
+            const unsigned hw = fdiv((unsigned)off, fC);
+            const int c = (int)((unsigned)off - hw * C);
+            const unsigned hh = fdiv(hw, fW);
+            const int w = (int)(hw - hh * W);
+            const int h = (int)hh;
+            const unsigned long long pz =
+                splitmix64(seed ^ (0xA5A5A5A5ull + (unsigned long long)n));
+            const float uz = (unsigned int)pz * 2.3283064e-10f;
+            const float us = (unsigned int)(pz >> 32) * 2.3283064e-10f;
+            const float uf = (unsigned int)splitmix64(pz) * 2.3283064e-10f;
+            const float zoom = 1.f + zr * (2.f * uz - 1.f);
+            const float shear = sr * (2.f * us - 1.f);
+            const float sf = (flip && uf < 0.5f) ? -1.f : 1.f;
+            const float cy = 0.5f * (H - 1), cx = 0.5f * (W - 1);
+            const float sy = cy + (h - cy) / zoom;
+            const float sx = cx + (w - cx) * sf / zoom + shear * (h - cy);
+            // border-clamped bilinear sample of the class template
+            const float syc = fminf(fmaxf(sy, 0.f), (float)(H - 1));
+            const float sxc = fminf(fmaxf(sx, 0.f), (float)(W - 1));
+            const int y0 = (int)syc, x0 = (int)sxc;
+            const int y1 = min(y0 + 1, H - 1), x1 = min(x0 + 1, W - 1);
+            const float fy = syc - y0, fx = sxc - x0;
+            const float* Tn = T + lab[n] * per_img;
+            const float t00 = Tn[((int64_t)y0 * W + x0) * C + c];
+            const float t01 = Tn[((int64_t)y0 * W + x1) * C + c];
+            const float t10 = Tn[((int64_t)y1 * W + x0) * C + c];
+            const float t11 = Tn[((int64_t)y1 * W + x1) * C + c];
+            t = (t00 * (1 - fx) + t01 * fx) * (1 - fy)
+                + (t10 * (1 - fx) + t11 * fx) * fy;
+        }
         // splitmix64 counter hash -> two 32-bit uniforms -> Box-Muller
-        unsigned long long z = seed + (unsigned long long)i * 0x9E3779B97F4A7C15ull;
-        z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
-        z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
-        z ^= z >> 31;
+        unsigned long long z =
+            splitmix64(seed + (unsigned long long)i * 0x9E3779B97F4A7C15ull);
         const float u1 = ((unsigned int)z + 1.0f) * 2.3283064e-10f;  // (0,1]
         const float u2 = (unsigned int)(z >> 32) * 2.3283064e-10f;
         const float nrm = sqrtf(-2.f * __logf(u1)) * __cosf(6.2831853f * u2);
@@ -3136,49 +3184,58 @@ torch::Tensor linear_wgrad(torch::Tensor dy, torch::Tensor x) {
     return dw;
 }
 
-torch::Tensor synth_batch(torch::Tensor templates, torch::Tensor labels,
-                          int64_t seed) {
+static torch::Tensor synth_launch(torch::Tensor templates,
+                                  torch::Tensor labels, int64_t seed,
+                                  const long long* seed_buf, double zoom,
+                                  double shear, int64_t flip) {
     CHECK_GPU(templates);
     TORCH_CHECK(templates.is_contiguous() && labels.is_contiguous());
     const int64_t n = labels.size(0);
+    const int H = (int)templates.size(1), W = (int)templates.size(2),
+              C = (int)templates.size(3);
     const int64_t per = templates.numel() / templates.size(0);
     auto out = torch::empty({n, templates.size(1), templates.size(2),
                              templates.size(3)},
                             templates.options().dtype(torch::kBFloat16));
     const int64_t total = n * per;
     int blocks = (int)std::min<int64_t>((total + 255) / 256, 8192);
-    hipLaunchKernelGGL(synth_batch_kernel, dim3(blocks), dim3(256), 0,
-                       at::cuda::getCurrentCUDAStream(),
-                       templates.data_ptr<float>(),
-                       labels.data_ptr<int64_t>(),
-                       reinterpret_cast<unsigned short*>(out.data_ptr()),
-                       per, total, (unsigned long long)seed,
-                       fdiv_make((unsigned)per), (const long long*)nullptr);
+    const bool aug = zoom != 0.0 || shear != 0.0 || flip != 0;
+    auto stream = at::cuda::getCurrentCUDAStream();
+    if (aug)
+        hipLaunchKernelGGL(synth_batch_kernel<true>, dim3(blocks), dim3(256),
+                           0, stream, templates.data_ptr<float>(),
+                           labels.data_ptr<int64_t>(),
+                           reinterpret_cast<unsigned short*>(out.data_ptr()),
+                           per, total, (unsigned long long)seed,
+                           fdiv_make((unsigned)per), seed_buf, H, W, C,
+                           fdiv_make((unsigned)C), fdiv_make((unsigned)W),
+                           (float)zoom, (float)shear, (int)flip);
+    else
+        hipLaunchKernelGGL(synth_batch_kernel<false>, dim3(blocks), dim3(256),
+                           0, stream, templates.data_ptr<float>(),
+                           labels.data_ptr<int64_t>(),
+                           reinterpret_cast<unsigned short*>(out.data_ptr()),
+                           per, total, (unsigned long long)seed,
+                           fdiv_make((unsigned)per), seed_buf, H, W, C,
+                           fdiv_make((unsigned)C), fdiv_make((unsigned)W),
+                           0.f, 0.f, 0);
     return out;
 }
 
+torch::Tensor synth_batch(torch::Tensor templates, torch::Tensor labels,
+                          int64_t seed, double zoom, double shear,
+                          int64_t flip) {
+    return synth_launch(templates, labels, seed, nullptr, zoom, shear, flip);
+}
+
 torch::Tensor synth_batch_g(torch::Tensor templates, torch::Tensor labels,
-                            torch::Tensor seed_buf, int64_t salt) {
+                            torch::Tensor seed_buf, int64_t salt, double zoom,
+                            double shear, int64_t flip) {
     // graph-capturable variant: base seed read from a device buffer the
     // host rewrites before each replay; salt distinguishes captured calls
-    CHECK_GPU(templates);
-    TORCH_CHECK(templates.is_contiguous() && labels.is_contiguous());
-    const int64_t n = labels.size(0);
-    const int64_t per = templates.numel() / templates.size(0);
-    auto out = torch::empty({n, templates.size(1), templates.size(2),
-                             templates.size(3)},
-                            templates.options().dtype(torch::kBFloat16));
-    const int64_t total = n * per;
-    int blocks = (int)std::min<int64_t>((total + 255) / 256, 8192);
-    hipLaunchKernelGGL(synth_batch_kernel, dim3(blocks), dim3(256), 0,
-                       at::cuda::getCurrentCUDAStream(),
-                       templates.data_ptr<float>(),
-                       labels.data_ptr<int64_t>(),
-                       reinterpret_cast<unsigned short*>(out.data_ptr()),
-                       per, total, (unsigned long long)salt,
-                       fdiv_make((unsigned)per),
-                       (const long long*)seed_buf.data_ptr<int64_t>());
-    return out;
+    return synth_launch(templates, labels, salt,
+                        (const long long*)seed_buf.data_ptr<int64_t>(), zoom,
+                        shear, flip);
 }
 
 std::vector<torch::Tensor> maxpool2x2_fwd(torch::Tensor x) {

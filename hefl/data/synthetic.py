@@ -51,23 +51,31 @@ class SyntheticMedicalImages:
         self.labels = torch.randint(0, self.n_classes, (self.n_samples,),
                                     generator=g).to(self.device)
 
-    def batch(self, indices: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
-        """Return (x[NHWC], y[N]) for the given sample indices."""
+    def batch(self, indices: torch.Tensor,
+              affine=None) -> Tuple[torch.Tensor, torch.Tensor]:
+        """Return (x[NHWC], y[N]) for the given sample indices.
+
+        `affine` = (zoom_range, shear_range, hflip) applies a per-sample
+        random zoom/shear/horizontal-flip when sampling the class template
+        — the analog of the reference's ImageDataGenerator augmentation
+        (shear 0.2, zoom 0.2, horizontal_flip; FLPyfhelin.py:80-86)."""
         idx = indices.to(self.device)
         y = self.labels[idx]
         n = idx.numel()
+        zr, sr, fl = affine if affine else (0.0, 0.0, False)
         # Noise is drawn deterministically per call position rather than per
         # index (cheap); the class template carries the learnable signal.
         # Sampled directly on the device — the training loop never touches host.
         if self.device.type == "cuda":
             if self.dtype == torch.bfloat16:
-                # single fused kernel: template gather + counter-hash normal
-                # noise + sigmoid + bf16 store (vs ~6 fp32 staging passes)
+                # single fused kernel: template gather (affine-sampled when
+                # augmenting) + counter-hash normal noise + sigmoid + bf16
                 import hefl
                 self._synth_ctr = getattr(self, "_synth_ctr", 0) + 1
                 return hefl.load_extension().synth_batch(
                     self.templates, y.contiguous(),
-                    self.seed * 0x10001 + self._synth_ctr), y
+                    self.seed * 0x10001 + self._synth_ctr,
+                    zoom=zr, shear=sr, flip=int(bool(fl))), y
             if not hasattr(self, "_gen"):
                 self._gen = torch.Generator(device=self.device)
                 self._gen.manual_seed(self.seed)
@@ -76,9 +84,43 @@ class SyntheticMedicalImages:
         else:
             g = torch.Generator(device="cpu").manual_seed(self.seed)
             noise = torch.randn(n, self.H, self.W, self.C, generator=g)
-        x = 0.6 * self.templates[y] + 0.4 * noise
+        tmpl = self.templates[y]
+        if affine and (zr or sr or fl):
+            tmpl = self._affine_sample(tmpl, zr, sr, fl)
+        x = 0.6 * tmpl + 0.4 * noise
         x = torch.sigmoid(x)  # [0, 1] like rescale=1/255 images
         return x.to(self.dtype), y
+
+    def _affine_sample(self, tmpl: torch.Tensor, zr: float, sr: float,
+                       fl: bool) -> torch.Tensor:
+        """Per-sample zoom/shear/h-flip via border-clamped bilinear sampling
+        (torch reference path; the GPU bf16 path fuses the same transform
+        into the synth kernel)."""
+        import torch.nn.functional as F
+        n = tmpl.shape[0]
+        self._aug_ctr = getattr(self, "_aug_ctr", 0) + 1
+        g = torch.Generator(device="cpu").manual_seed(
+            self.seed * 31 + self._aug_ctr)
+        zoom = 1 + zr * (2 * torch.rand(n, generator=g) - 1)
+        shear = sr * (2 * torch.rand(n, generator=g) - 1)
+        sf = torch.where(torch.rand(n, generator=g) < 0.5, -1.0, 1.0) \
+            if fl else torch.ones(n)
+        # inverse map (output -> source), about the image center; grid_sample
+        # normalized coords with align_corners=True match the (dim-1) scaling
+        H, W = self.H, self.W
+        hh = torch.arange(H, dtype=torch.float32)
+        ww = torch.arange(W, dtype=torch.float32)
+        cy, cx = 0.5 * (H - 1), 0.5 * (W - 1)
+        dy = (hh - cy).view(1, H, 1)
+        dx = (ww - cx).view(1, 1, W)
+        sy = cy + dy / zoom.view(-1, 1, 1) + torch.zeros(1, 1, W)
+        sx = (cx + dx * (sf / zoom).view(-1, 1, 1)
+              + shear.view(-1, 1, 1) * dy)
+        grid = torch.stack([2 * sx / (W - 1) - 1, 2 * sy / (H - 1) - 1],
+                           dim=-1).to(tmpl.device, tmpl.dtype)
+        out = F.grid_sample(tmpl.permute(0, 3, 1, 2), grid, mode="bilinear",
+                            padding_mode="border", align_corners=True)
+        return out.permute(0, 2, 3, 1)
 
 
 def make_client_loader(ds: SyntheticMedicalImages, client: int, n_clients: int,
@@ -96,13 +138,14 @@ class ClientLoader:
 
     def __init__(self, ds: SyntheticMedicalImages, indices: torch.Tensor,
                  batch_size: int, seed: int = 0, shuffle: bool = True,
-                 augment=None):
+                 augment=None, affine=None):
         self.ds = ds
         self.indices = indices.clone()
         self.batch_size = int(batch_size)
         self.seed = int(seed)
         self.shuffle = shuffle
         self.augment = augment  # callable(x) -> x, e.g. pipeline.hflip_augment
+        self.affine = affine    # (zoom, shear, hflip) in-generator transform
         self._epoch = 0
 
     def __len__(self) -> int:
@@ -121,7 +164,8 @@ class ClientLoader:
     def __iter__(self) -> Iterator[Tuple[torch.Tensor, torch.Tensor]]:
         order = self.epoch_order()
         for i in range(0, order.numel(), self.batch_size):
-            x, y = self.ds.batch(order[i:i + self.batch_size])
+            x, y = self.ds.batch(order[i:i + self.batch_size],
+                                 affine=self.affine)
             if self.augment is not None:
                 x = self.augment(x)
             yield x, y

@@ -64,8 +64,14 @@ class LocalClient:
         idx = shard_indices(n_total, client_id, cfg.fl.n_clients)
         n_val = cfg.fl.val_samples_per_client
         train_idx = idx[: idx.numel() - n_val]
+        # augmentation applies to TRAINING batches only (the reference's
+        # ImageDataGenerator transforms train, not val/test)
+        affine = {"none": None, "hflip": (0.0, 0.0, True),
+                  "full": (0.2, 0.2, True)}[getattr(cfg.fl, "augment", "none")]
+        self.affine = affine
         self.loader = ClientLoader(self.dataset, train_idx, t.batch_size,
-                                   seed=cfg.fl.seed + client_id)
+                                   seed=cfg.fl.seed + client_id,
+                                   affine=affine)
         self.val_loader = (ClientLoader(self.dataset,
                                         idx[idx.numel() - n_val:],
                                         t.batch_size, shuffle=False)
@@ -170,9 +176,10 @@ class LocalClient:
         # in-graph data staging: the captured graph gathers labels and
         # generates the epoch's samples itself (synth kernel reads its seed
         # from a device buffer the host rewrites before each replay), so
-        # per epoch the host does ONE small H2D copy + ONE replay. Falls
-        # back to host staging when an augment hook or non-bf16 dtype is in
-        # play (those paths go through the torch ops).
+        # per epoch the host does ONE small H2D copy + ONE replay. The
+        # affine augmentation is fused INTO the synth kernel, so it stays
+        # in-graph; only a custom augment hook or non-bf16 dtype falls back
+        # to host staging (those paths go through the torch ops).
         in_graph_data = (self.loader.augment is None
                          and self.compute_dtype == torch.bfloat16)
         x0, y0 = self.dataset.batch(self.loader.indices)
@@ -213,7 +220,9 @@ class LocalClient:
         with torch.cuda.graph(g, stream=side):
             if in_graph_data:
                 Y = self.dataset.labels.index_select(0, order_buf)
-                X = C.synth_batch_g(self.dataset.templates, Y, seed_buf, 0)
+                zr, sr, fl = self.affine or (0.0, 0.0, False)
+                X = C.synth_batch_g(self.dataset.templates, Y, seed_buf, 0,
+                                    zoom=zr, shear=sr, flip=int(bool(fl)))
             self.opt.prep_epoch(steps)
             for s, i in enumerate(range(0, n, B)):
                 self.opt.zero_grad()  # grads=None -> backward steals
@@ -244,7 +253,7 @@ class LocalClient:
                                   + ent["seed_ctr"])
             ent["graph"].replay()
             return
-        x, y = self.dataset.batch(order)
+        x, y = self.dataset.batch(order, affine=self.loader.affine)
         if self.loader.augment is not None:
             x = self.loader.augment(x)
         ent["X"].copy_(x.to(self.compute_dtype))
