@@ -1779,8 +1779,12 @@ __global__ void pool_relu_bias_bwd2_kernel(
             if (acc[j] != 0.f) atomicAdd(dbs + my_oct * 8 + j, acc[j]);
     }
     __syncthreads();
-    for (int c = threadIdx.x; c < K; c += blockDim.x)
-        if (dbs[c] != 0.f) atomicAdd(db + c, dbs[c]);
+    if (gridDim.x == 1) {  // small layers: direct store, db needs no zero-fill
+        for (int c = threadIdx.x; c < K; c += blockDim.x) db[c] = dbs[c];
+    } else {
+        for (int c = threadIdx.x; c < K; c += blockDim.x)
+            if (dbs[c] != 0.f) atomicAdd(db + c, dbs[c]);
+    }
 }
 
 // Scalar fallback for K % 8 != 0 trunk layers (LeNet-5's 6/16 filters):
@@ -3308,14 +3312,19 @@ std::vector<torch::Tensor> pool_relu_bias_bwd(torch::Tensor dy,
     const int N = (int)dyc.size(0), OH = (int)dyc.size(1),
               OW = (int)dyc.size(2), K = (int)dyc.size(3);
     auto dym = torch::empty({N, H, W, (int64_t)K}, dyc.options());
-    auto db = torch::zeros({K}, dyc.options().dtype(torch::kFloat32));
     auto stream = at::cuda::getCurrentCUDAStream();
     const int noct = K >> 3;
     if ((K & 7) == 0 && (noct & (noct - 1)) == 0 && noct <= 256) {
         // 2x2-block octet kernel: dy/idx/p read once per output cell
         const int HB = (int)((H + 1) / 2), WB = (int)((W + 1) / 2);
         const int64_t total8 = (int64_t)N * HB * WB * noct;
-        int blocks = (int)std::min<int64_t>(ceildiv(total8, 256), 4096);
+        // tiny layers run one block: direct db store, no zero-fill kernel
+        int blocks = total8 <= 16384
+                         ? 1
+                         : (int)std::min<int64_t>(ceildiv(total8, 256), 4096);
+        auto db = blocks == 1
+                      ? torch::empty({K}, dyc.options().dtype(torch::kFloat32))
+                      : torch::zeros({K}, dyc.options().dtype(torch::kFloat32));
         hipLaunchKernelGGL(pool_relu_bias_bwd2_kernel, dim3(blocks),
                            dim3(256), K * sizeof(float), stream, bf_ptr(dyc),
                            idx.data_ptr<uint8_t>(), bf_ptr(p),
@@ -3326,6 +3335,7 @@ std::vector<torch::Tensor> pool_relu_bias_bwd(torch::Tensor dy,
         return {dym, db};
     }
     const int64_t total = (int64_t)N * H * W * K;
+    auto db = torch::zeros({K}, dyc.options().dtype(torch::kFloat32));
     int blocks = (int)std::min<int64_t>(ceildiv(total, 256), 4096);
     hipLaunchKernelGGL(pool_relu_bias_bwd_scalar_kernel, dim3(blocks),
                        dim3(256), 0, stream, bf_ptr(dyc),
