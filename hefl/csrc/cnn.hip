@@ -1788,33 +1788,52 @@ __global__ void pool_relu_bias_bwd2_kernel(
 }
 
 // Scalar fallback for K % 8 != 0 trunk layers (LeNet-5's 6/16 filters):
-// one thread per input cell, ReLU gate on the POOLED output p (= the
+// thread = (channel, row-lane) so each thread's bias partial stays in a
+// register (naive per-element global atomics serialized on K=6 addresses
+// and cost config #3 ~40%); ReLU gate on the POOLED output p (= the
 // argmax cell's activation; relu ran before pool).
 __global__ void pool_relu_bias_bwd_scalar_kernel(
     const unsigned short* __restrict__ dy, const uint8_t* __restrict__ idx,
     const unsigned short* __restrict__ p, unsigned short* __restrict__ dym,
-    float* __restrict__ db, int64_t total, int K, int H, int W, int OH,
-    int OW, FastDiv fK, FastDiv fW, FastDiv fH) {
-    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
-         i < total; i += (int64_t)gridDim.x * blockDim.x) {
-        const unsigned t0 = fdiv((unsigned)i, fK);
-        const int c = (int)((unsigned)i - t0 * K);
-        const unsigned t1 = fdiv(t0, fW);
-        const int iw = (int)(t0 - t1 * W);
-        const unsigned t2 = fdiv(t1, fH);
-        const int ih = (int)(t1 - t2 * H);
-        const int oh = ih >> 1, ow = iw >> 1;
-        unsigned short g = 0;
-        if (oh < OH && ow < OW) {
-            const int64_t o = (((int64_t)t2 * OH + oh) * OW + ow) * K + c;
-            const int d = idx[o];
-            const unsigned short pv = p[o];
-            if ((d >> 1) == (ih & 1) && (d & 1) == (iw & 1) &&
-                (pv & 0x7fffu) != 0 && !(pv & 0x8000u))
-                g = dy[o];
+    float* __restrict__ db, int64_t M, int K, int rows_per_block, int H,
+    int W, int OH, int OW, FastDiv fW, FastDiv fH) {
+    __shared__ float red[256];
+    const int64_t r0 = (int64_t)blockIdx.x * rows_per_block;
+    const int64_t r1 = min(r0 + rows_per_block, M);
+    int lanes = (int)blockDim.x / K;
+    lanes = 1 << (31 - __clz(lanes));
+    const int c = threadIdx.x % K;
+    const int rl = threadIdx.x / K;
+    float acc = 0.f;
+    if (rl < lanes) {
+        for (int64_t r = r0 + rl; r < r1; r += lanes) {
+            const unsigned t1 = fdiv((unsigned)r, fW);
+            const int iw = (int)((unsigned)r - t1 * W);
+            const unsigned t2 = fdiv(t1, fH);
+            const int ih = (int)(t1 - t2 * H);
+            const int oh = ih >> 1, ow = iw >> 1;
+            unsigned short g = 0;
+            if (oh < OH && ow < OW) {
+                const int64_t o = (((int64_t)t2 * OH + oh) * OW + ow) * K + c;
+                const int d = idx[o];
+                const unsigned short pv = p[o];
+                if ((d >> 1) == (ih & 1) && (d & 1) == (iw & 1) &&
+                    (pv & 0x7fffu) != 0 && !(pv & 0x8000u))
+                    g = dy[o];
+            }
+            dym[r * K + c] = g;
+            acc += bf2f(g);
         }
-        dym[i] = g;
-        if (g) atomicAdd(db + c, bf2f(g));
+    }
+    red[threadIdx.x] = acc;
+    __syncthreads();
+    for (int off = lanes >> 1; off >= 1; off >>= 1) {
+        if (rl < off) red[threadIdx.x] += red[threadIdx.x + off * K];
+        __syncthreads();
+    }
+    if (rl == 0) {
+        if (gridDim.x == 1) db[c] = red[threadIdx.x];
+        else atomicAdd(db + c, red[threadIdx.x]);
     }
 }
 
@@ -3334,15 +3353,21 @@ std::vector<torch::Tensor> pool_relu_bias_bwd(torch::Tensor dy,
                            fdiv_make((unsigned)HB));
         return {dym, db};
     }
-    const int64_t total = (int64_t)N * H * W * K;
-    auto db = torch::zeros({K}, dyc.options().dtype(torch::kFloat32));
-    int blocks = (int)std::min<int64_t>(ceildiv(total, 256), 4096);
-    hipLaunchKernelGGL(pool_relu_bias_bwd_scalar_kernel, dim3(blocks),
+    TORCH_CHECK(K <= 256, "scalar pool-backward fallback expects K <= 256");
+    const int64_t M = (int64_t)N * H * W;
+    int rpb = (int)std::max<int64_t>(64, (M + 511) / 512);
+    int lanes = 256 / K;
+    lanes = 1 << (31 - __builtin_clz(lanes));
+    rpb = std::max(rpb, 2 * lanes);
+    int nblk = (int)((M + rpb - 1) / rpb);
+    auto db = nblk == 1
+                  ? torch::empty({K}, dyc.options().dtype(torch::kFloat32))
+                  : torch::zeros({K}, dyc.options().dtype(torch::kFloat32));
+    hipLaunchKernelGGL(pool_relu_bias_bwd_scalar_kernel, dim3(nblk),
                        dim3(256), 0, stream, bf_ptr(dyc),
                        idx.data_ptr<uint8_t>(), bf_ptr(p), bf_ptr_mut(dym),
-                       db.data_ptr<float>(), total, K, (int)H, (int)W, OH, OW,
-                       fdiv_make((unsigned)K), fdiv_make((unsigned)W),
-                       fdiv_make((unsigned)H));
+                       db.data_ptr<float>(), M, K, rpb, (int)H, (int)W, OH,
+                       OW, fdiv_make((unsigned)W), fdiv_make((unsigned)H));
     return {dym, db};
 }
 
