@@ -1353,6 +1353,147 @@ conv_wgrad_glds_kernel(const unsigned short* __restrict__ dy,
     }
 }
 
+// ---- Kout <= 32 wgrad: the 64-ko tile wasted HALF its MFMA work on
+// 32-filter layers (RefCNN6 conv1-3 — the round's #1 kernel). This variant
+// tiles [32 ko] x [BN cc] with a 64-PIXEL K-step: same 4 MFMAs per wave
+// per barrier, no ko waste, half the barriers per pixel. Pipeline and
+// gather structure mirror conv_wgrad_glds_kernel (3-buffer counted
+// vmcnt(3): three 16-B glds per thread per stage). ----
+template <int BN, int WM, int WN, int FM, int FN>
+__global__ void __launch_bounds__(TPB)
+conv_wgrad_glds_k32_kernel(const unsigned short* __restrict__ dy,
+                           const unsigned short* __restrict__ x,
+                           float* __restrict__ dw,
+                           const unsigned short* __restrict__ zbuf,
+                           ConvShape s, int k_chunks) {
+    constexpr int BKP = 64;  // pixels per K-step
+    __shared__ unsigned short smem[3 * (32 + BN) * BKP];
+    auto Dys = [&](int buf) -> unsigned short (*)[32] {
+        return reinterpret_cast<unsigned short(*)[32]>(
+            smem + buf * (32 + BN) * BKP);
+    };
+    auto Xs = [&](int buf) -> unsigned short (*)[BN] {
+        return reinterpret_cast<unsigned short(*)[BN]>(
+            smem + buf * (32 + BN) * BKP + 32 * BKP);
+    };
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+    const int m0 = blockIdx.x * 32;   // over Kout
+    const int n0 = blockIdx.y * BN;   // over RSC
+    const int M = s.Kout;
+    const int NN = s.R * s.S * s.C;
+    const int KK = s.N * s.OH * s.OW;
+    const int chunk = (KK + k_chunks - 1) / k_chunks;
+    const int kbeg = blockIdx.z * chunk;
+    const int kend = min(kbeg + chunk, KK);
+
+    auto stage = [&](int buf, int k0) {
+        {   // Dys[pix][ko]: 64 pix x 4 ko-chunks = 256 = one DMA per thread
+            const int pix = tid >> 2;
+            const int koc = (tid & 3) * 8;
+            const int kpix = k0 + pix;
+            const int ko = m0 + koc;
+            const unsigned short* src = zbuf;
+            if (kpix < kend && ko + 8 <= M)
+                src = dy + (int64_t)kpix * s.Kout + ko;
+            glds16(src, (char*)&Dys(buf)[0][0] + wave * 1024);
+        }
+#pragma unroll
+        for (int t = 0; t < BN * 8 / TPB; ++t) {  // Xs: 64 pix x BN/8 chunks
+            const int i = tid + t * TPB;
+            const int pix = i / (BN / 8);
+            const int cc = (i % (BN / 8)) * 8;
+            const int kpix = k0 + pix;
+            const int nn = n0 + cc;
+            const unsigned short* src = zbuf;
+            if (kpix < kend && nn < NN) {
+                int n_ = (int)fdiv((unsigned)kpix, s.fOHOW);
+                int rem = kpix - n_ * (s.OH * s.OW);
+                int oh = (int)fdiv((unsigned)rem, s.fOW), ow = rem - oh * s.OW;
+                int rs = (int)fdiv((unsigned)nn, s.fC);
+                int c = nn - rs * s.C;
+                int r = (int)fdiv((unsigned)rs, s.fS), ss = rs - r * s.S;
+                if (c + 8 <= s.C) {
+                    int ih = oh * s.stride + r - s.pad;
+                    int iw = ow * s.stride + ss - s.pad;
+                    if (ih >= 0 && ih < s.H && iw >= 0 && iw < s.W)
+                        src = x + ((((int64_t)n_ * s.H + ih) * s.W + iw) * s.C
+                                   + c);
+                }
+            }
+            glds16(src, (char*)&Xs(buf)[0][0] + (wave + t * 4) * 1024);
+        }
+    };
+
+    f32x4 acc[FM][FN];
+#pragma unroll
+    for (int i = 0; i < FM; ++i)
+#pragma unroll
+        for (int j = 0; j < FN; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+    stage(0, kbeg);
+    stage(1, kbeg + BKP);
+    int buf = 0;
+    for (int k0 = kbeg; k0 < kend; k0 += BKP) {
+        if (k0 + BKP < kend)
+            asm volatile("s_waitcnt vmcnt(3)" ::: "memory");
+        else
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        asm volatile("s_barrier" ::: "memory");
+        if (k0 + 2 * BKP < kend) stage(buf == 0 ? 2 : buf - 1, k0 + 2 * BKP);
+        {
+            const int wm = wave / WN, wn = wave % WN;
+            const int half = lane >> 4, sub = lane & 15;
+#pragma unroll
+            for (int kk = 0; kk < BKP; kk += 32) {
+                bf16x8 a[FM], b[FN];
+#pragma unroll
+                for (int i = 0; i < FM; ++i) {
+                    const int mrow = wm * FM * 16 + i * 16 + sub;
+#pragma unroll
+                    for (int j = 0; j < 8; ++j)
+                        a[i][j] = *reinterpret_cast<const bf16_t*>(
+                            &Dys(buf)[kk + half * 8 + j][mrow]);
+                }
+#pragma unroll
+                for (int j = 0; j < FN; ++j) {
+                    const int col = wn * FN * 16 + j * 16 + sub;
+#pragma unroll
+                    for (int t = 0; t < 8; ++t)
+                        b[j][t] = *reinterpret_cast<const bf16_t*>(
+                            &Xs(buf)[kk + half * 8 + t][col]);
+                }
+#pragma unroll
+                for (int i = 0; i < FM; ++i)
+#pragma unroll
+                    for (int j = 0; j < FN; ++j)
+                        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            a[i], b[j], acc[i][j], 0, 0, 0);
+            }
+        }
+        buf = buf == 2 ? 0 : buf + 1;
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    const int wm = wave / WN, wn = wave % WN;
+#pragma unroll
+    for (int j = 0; j < FN; ++j) {
+        const int col = n0 + wn * FN * 16 + j * 16 + (lane & 15);
+        if (col >= NN) continue;
+#pragma unroll
+        for (int i = 0; i < FM; ++i)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int ko = m0 + wm * FM * 16 + i * 16 + (lane >> 4) * 4 + r;
+                if (ko >= M) continue;
+                if (k_chunks > 1)
+                    atomicAdd(dw + (int64_t)ko * NN + col, acc[i][j][r]);
+                else
+                    dw[(int64_t)ko * NN + col] = acc[i][j][r];
+            }
+    }
+}
+
 // ---- small-shape wgrad (first convs: Kout <= 64, R*S*C <= 16, e.g. the
 // 28x28x1 model's conv1): the generic 64xBN tile wastes 3/4 of its waves on
 // out-of-range rows. Here the 4 waves SPLIT over pixels (each owns 32 of a
@@ -3124,6 +3265,23 @@ torch::Tensor conv2d_wgrad(torch::Tensor dy, torch::Tensor x, int64_t stride,
                   : torch::empty({s.Kout, R, S, s.C},
                                  x.options().dtype(torch::kFloat32));
     dim3 grid(ceildiv(s.Kout, 64), ceildiv(NN, bn), k_chunks);
+    if (glds_ok && s.Kout <= 32) {
+        // 32-ko tile, 64-pixel K-step: no ko waste on 32-filter layers
+        static torch::Tensor zbuf;
+        if (!zbuf.defined() || zbuf.device() != dy.device())
+            zbuf = torch::zeros({8}, dy.options());
+        int tiles32 = ceildiv(s.Kout, 32) * ceildiv(NN, 64);
+        int kc = std::max(1, std::min(ceildiv(KK, 128),
+                                      512 / std::max(tiles32, 1)));
+        auto dw32 = kc > 1 ? torch::zeros({s.Kout, R, S, s.C},
+                                          x.options().dtype(torch::kFloat32))
+                           : dw;
+        dim3 g32(ceildiv(s.Kout, 32), ceildiv(NN, 64), kc);
+        hipLaunchKernelGGL((conv_wgrad_glds_k32_kernel<64, 2, 2, 1, 2>), g32,
+                           dim3(TPB), 0, stream, bf_ptr(dy), bf_ptr(x),
+                           dw32.data_ptr<float>(), bf_ptr(zbuf), s, kc);
+        return dw32;
+    }
     if (glds_ok) {
         static torch::Tensor zbuf;
         if (!zbuf.defined() || zbuf.device() != dy.device())
@@ -3337,10 +3495,10 @@ std::vector<torch::Tensor> pool_relu_bias_bwd(torch::Tensor dy,
         // 2x2-block octet kernel: dy/idx/p read once per output cell
         const int HB = (int)((H + 1) / 2), WB = (int)((W + 1) / 2);
         const int64_t total8 = (int64_t)N * HB * WB * noct;
-        // tiny layers run one block: direct db store, no zero-fill kernel
-        int blocks = total8 <= 16384
-                         ? 1
-                         : (int)std::min<int64_t>(ceildiv(total8, 256), 4096);
+        // measured: forcing ONE block on small layers to skip the db
+        // zero-fill serialized the whole layer onto one CU (-30% on the
+        // headline config) — keep the parallel grid and pay the 4.7 us fill
+        int blocks = (int)std::min<int64_t>(ceildiv(total8, 256), 4096);
         auto db = blocks == 1
                       ? torch::empty({K}, dyc.options().dtype(torch::kFloat32))
                       : torch::zeros({K}, dyc.options().dtype(torch::kFloat32));
