@@ -732,6 +732,124 @@ conv_fwd_glds64_kernel(const unsigned short* __restrict__ x,
     }
 }
 
+// ---- Direct tiled 3x3 stride-1 conv forward: one block owns a TH x TW
+// output patch of one image and a BN-wide Kout tile; the (TH+2) x (TW+2) x
+// CS input tile is staged in LDS ONCE per channel slab and all NINE kernel
+// taps read from it. The implicit-GEMM kernel re-gathers the same input
+// 9x through the im2col view and runs only 4 MFMAs between barriers; here
+// one staging barrier covers 9 * FM * FN MFMAs per wave (~144), so the
+// kernel is MFMA-issue rather than stage/barrier bound. Requires stride 1,
+// R = S = 3, C % 32 == 0 (channel slabs of 32). pad 0 or 1.
+template <int TH, int TW, int BN, int WM, int WN, int FM, int FN>
+__global__ void __launch_bounds__(TPB)
+conv_fwd_tile3_kernel(const unsigned short* __restrict__ x,
+                      const unsigned short* __restrict__ w,
+                      const float* __restrict__ bias,
+                      unsigned short* __restrict__ y, ConvShape s, int relu,
+                      int tiles_h, int tiles_w) {
+    constexpr int CS = 32;               // channel slab (one MFMA K-step)
+    constexpr int XH = TH + 2, XW = TW + 2;
+    // +8 channel pad on the x-tile pixel stride: A-fragment lanes read 16
+    // consecutive pixels at 64-B stride; padding to 80 B breaks the 4-way
+    // bank alias (rule 21)
+    constexpr int XP = CS + 8;
+    __shared__ unsigned short xs[XH * XW * XP];
+    __shared__ unsigned short ws[9 * BN * CS];
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+    // tile decode: blockIdx.x = (n, th, tw)
+    const int tw_ = blockIdx.x % tiles_w;
+    const int th_ = (blockIdx.x / tiles_w) % tiles_h;
+    const int n = blockIdx.x / (tiles_w * tiles_h);
+    const int oh0 = th_ * TH, ow0 = tw_ * TW;
+    const int n0 = blockIdx.y * BN;
+    const int wm = wave / WN, wn = wave % WN;
+    const int half = lane >> 4, sub = lane & 15;
+
+    f32x4 acc[FM][FN];
+#pragma unroll
+    for (int i = 0; i < FM; ++i)
+#pragma unroll
+        for (int j = 0; j < FN; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+    const int KK = s.R * s.S * s.C;  // w rows are [ko][r][s][c], c fastest
+    for (int c0 = 0; c0 < s.C; c0 += CS) {
+        __syncthreads();  // previous slab's MFMAs done before overwrite
+        // x-tile: XH*XW pixels x CS channels, 16-B chunks; border/halo
+        // out-of-range rows load zeros
+        for (int i = tid; i < XH * XW * (CS / 8); i += TPB) {
+            const int pix = i / (CS / 8);
+            const int cc = (i % (CS / 8)) * 8;
+            const int gy = pix / XW, gx = pix - gy * XW;
+            const int ih = oh0 + gy - s.pad;
+            const int iw = ow0 + gx - s.pad;
+            u16x8 v = {};
+            if (ih >= 0 && ih < s.H && iw >= 0 && iw < s.W)
+                v = *reinterpret_cast<const u16x8*>(
+                    &x[((((int64_t)n * s.H + ih) * s.W + iw) * s.C) + c0 + cc]);
+            *reinterpret_cast<u16x8*>(&xs[pix * XP + cc]) = v;
+        }
+        // w-slab: [rs][ko][CS], ko-major rows so B-fragment reads are
+        // 16-B at stride CS*2
+        for (int i = tid; i < 9 * BN * (CS / 8); i += TPB) {
+            const int rs = i / (BN * (CS / 8));
+            const int rem = i - rs * BN * (CS / 8);
+            const int ko = rem / (CS / 8);
+            const int cc = (rem % (CS / 8)) * 8;
+            u16x8 v = {};
+            if (n0 + ko < s.Kout)
+                v = *reinterpret_cast<const u16x8*>(
+                    &w[(int64_t)(n0 + ko) * KK + rs * s.C + c0 + cc]);
+            *reinterpret_cast<u16x8*>(&ws[(rs * BN + ko) * CS + cc]) = v;
+        }
+        __syncthreads();
+#pragma unroll
+        for (int rs = 0; rs < 9; ++rs) {
+            const int r = rs / 3, ss = rs - r * 3;
+            bf16x8 a[FM], b[FN];
+#pragma unroll
+            for (int i = 0; i < FM; ++i) {
+                const int p = wm * FM * 16 + i * 16 + sub;  // local pixel
+                const int py = p / TW, px = p - py * TW;
+                a[i] = *reinterpret_cast<const bf16x8*>(
+                    &xs[((py + r) * XW + (px + ss)) * XP + half * 8]);
+            }
+#pragma unroll
+            for (int j = 0; j < FN; ++j) {
+                const int ko = wn * FN * 16 + j * 16 + sub;
+                b[j] = *reinterpret_cast<const bf16x8*>(
+                    &ws[(rs * BN + ko) * CS + half * 8]);
+            }
+#pragma unroll
+            for (int i = 0; i < FM; ++i)
+#pragma unroll
+                for (int j = 0; j < FN; ++j)
+                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        a[i], b[j], acc[i][j], 0, 0, 0);
+        }
+    }
+#pragma unroll
+    for (int j = 0; j < FN; ++j) {
+        const int col = n0 + wn * FN * 16 + j * 16 + sub;
+        if (col >= s.Kout) continue;
+        const float bv = bias ? bias[col] : 0.f;
+#pragma unroll
+        for (int i = 0; i < FM; ++i)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int p = wm * FM * 16 + i * 16 + half * 4 + r;
+                const int py = p / TW, px = p - py * TW;
+                const int oh = oh0 + py, ow = ow0 + px;
+                if (oh >= s.OH || ow >= s.OW) continue;
+                float v = acc[i][j][r] + bv;
+                if (relu) v = v > 0.f ? v : 0.f;
+                y[((((int64_t)n * s.OH + oh) * s.OW + ow) * s.Kout) + col]
+                    = f2bf(v);
+            }
+    }
+}
+
 // ---- dgrad: dx[m=(n,ih,iw), c] = sum_k A(m,k) * B(k,c),
 //      k = (r, s, ko) with ko FASTEST so the dy gather is contiguous;
 //      B(k, c) = w[ko, r, s, c] (strided, small tile) ----
@@ -1861,6 +1979,7 @@ __global__ void maxpool_fwd_oct_kernel(const unsigned short* __restrict__ x,
 // before pool, so p > 0 iff the argmax cell was active) — the full pre-pool
 // activation y is neither read nor saved for backward anymore. dy/idx/p are
 // read ONCE per output cell (the row-per-thread form re-read them 4x).
+template <int TAIL>  // 0 = per-thread LDS atomics, 1 = tree reduce, 2 = none
 __global__ void pool_relu_bias_bwd2_kernel(
     const unsigned short* __restrict__ dy, const uint8_t* __restrict__ idx,
     const unsigned short* __restrict__ p, unsigned short* __restrict__ dym,
@@ -1913,9 +2032,22 @@ __global__ void pool_relu_bias_bwd2_kernel(
                 *reinterpret_cast<u16x8*>(&dym[r0 + (int64_t)W * K + K]) = g3;
         }
     }
-    // oct is constant per thread (grid stride % noct == 0, noct pow2), so
-    // threads tid % noct == oct share a channel group: tree-reduce over the
-    // lane dimension, one global atomic per channel per block
+    if (TAIL == 2) return;
+    if (TAIL == 0) {
+        const int my_oct = (int)(((int64_t)blockIdx.x * blockDim.x
+                                  + threadIdx.x) % noct);
+        float* dbs = red8;  // reuse as [K] accumulator
+        for (int c = threadIdx.x; c < K; c += blockDim.x) dbs[c] = 0.f;
+        __syncthreads();
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+            if (acc[j] != 0.f) atomicAdd(dbs + my_oct * 8 + j, acc[j]);
+        __syncthreads();
+        for (int c = threadIdx.x; c < K; c += blockDim.x)
+            if (dbs[c] != 0.f) atomicAdd(db + c, dbs[c]);
+        return;
+    }
+    // TAIL == 1: tree reduce over the lane dimension
 #pragma unroll
     for (int j = 0; j < 8; ++j) red8[threadIdx.x * 8 + j] = acc[j];
     __syncthreads();
@@ -3047,6 +3179,40 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
     const float* bias = b.numel() ? b.data_ptr<float>() : nullptr;
     auto stream = at::cuda::getCurrentCUDAStream();
     const int KKf = s.R * s.S * s.C;
+    // Direct tiled kernel for 3x3 s1 convs with C % 32 == 0: the input
+    // tile is staged once per channel slab and all nine taps compute from
+    // it (vs 9x re-gather + 4 MFMAs/barrier in the implicit-GEMM path).
+    static const bool use_tile3 = [] {
+        const char* e = getenv("HEFL_TILE3");
+        return !e || e[0] != '0';
+    }();
+    if (use_tile3 && s.R == 3 && s.S == 3 && s.stride == 1 && s.pad <= 1 &&
+        s.C % 32 == 0 && s.Kout > 16) {
+        const int BN3 = s.Kout <= 32 ? 32 : 64;
+        const int kt = ceildiv(s.Kout, BN3);
+        auto tiles = [&](int th, int tw) {
+            return (int64_t)s.N * ceildiv(s.OH, th) * ceildiv(s.OW, tw) * kt;
+        };
+        int TH = 0;
+        if (s.OH >= 12 && tiles(16, 16) >= 256) TH = 16;
+        else if (s.OH >= 6 && tiles(8, 16) >= 256) TH = 8;
+        if (TH) {
+            const int th_ = ceildiv(s.OH, TH), tw_ = ceildiv(s.OW, 16);
+            dim3 grid((unsigned)(s.N * th_ * tw_), (unsigned)kt);
+            #define LAUNCH_T3(TH_, TW_, BN_, WM_, WN_, FM_, FN_)              \
+                hipLaunchKernelGGL(                                           \
+                    (conv_fwd_tile3_kernel<TH_, TW_, BN_, WM_, WN_, FM_,      \
+                                           FN_>),                             \
+                    grid, dim3(TPB), 0, stream, bf_ptr(x), bf_ptr(w), bias,   \
+                    bf_ptr_mut(y), s, relu ? 1 : 0, th_, tw_)
+            if (TH == 16 && BN3 == 64)      LAUNCH_T3(16, 16, 64, 4, 1, 4, 4);
+            else if (TH == 16)              LAUNCH_T3(16, 16, 32, 4, 1, 4, 2);
+            else if (BN3 == 64)             LAUNCH_T3(8, 16, 64, 2, 2, 4, 2);
+            else                            LAUNCH_T3(8, 16, 32, 2, 2, 4, 1);
+            #undef LAUNCH_T3
+            return y;
+        }
+    }
     // Measured on MI355X (gpurun_out/bench_conv_glds64 vs _glds32): the
     // BK=64 two-buffer kernel LOSES 8-30% to the BK=32 three-buffer one at
     // every conv shape here — K loops are 2-18 steps, so the prologue DMA
@@ -3515,13 +3681,21 @@ std::vector<torch::Tensor> pool_relu_bias_bwd(torch::Tensor dy,
         auto db = blocks == 1
                       ? torch::empty({K}, dyc.options().dtype(torch::kFloat32))
                       : torch::zeros({K}, dyc.options().dtype(torch::kFloat32));
-        hipLaunchKernelGGL(pool_relu_bias_bwd2_kernel, dim3(blocks),
-                           dim3(256), 0, stream, bf_ptr(dyc),
-                           idx.data_ptr<uint8_t>(), bf_ptr(p),
-                           bf_ptr_mut(dym), db.data_ptr<float>(), total8, K,
-                           (int)H, (int)W, OH, OW, HB, WB,
-                           fdiv_make((unsigned)noct), fdiv_make((unsigned)WB),
-                           fdiv_make((unsigned)HB));
+        static const int tail = [] {
+            const char* e = getenv("HEFL_POOL_TAIL");
+            return e ? e[0] - '0' : 0;
+        }();
+        #define LP(T) hipLaunchKernelGGL((pool_relu_bias_bwd2_kernel<T>),    \
+                           dim3(blocks), dim3(256), 0, stream, bf_ptr(dyc),  \
+                           idx.data_ptr<uint8_t>(), bf_ptr(p),               \
+                           bf_ptr_mut(dym), db.data_ptr<float>(), total8, K, \
+                           (int)H, (int)W, OH, OW, HB, WB,                   \
+                           fdiv_make((unsigned)noct),                        \
+                           fdiv_make((unsigned)WB), fdiv_make((unsigned)HB))
+        if (tail == 1) LP(1);
+        else if (tail == 2) LP(2);
+        else LP(0);
+        #undef LP
         return {dym, db};
     }
     TORCH_CHECK(K <= 256, "scalar pool-backward fallback expects K <= 256");
