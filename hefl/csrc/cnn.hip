@@ -1979,18 +1979,18 @@ __global__ void maxpool_fwd_oct_kernel(const unsigned short* __restrict__ x,
 // before pool, so p > 0 iff the argmax cell was active) — the full pre-pool
 // activation y is neither read nor saved for backward anymore. dy/idx/p are
 // read ONCE per output cell (the row-per-thread form re-read them 4x).
-template <int TAIL>  // 0 = per-thread LDS atomics, 1 = tree reduce, 2 = none
 __global__ void pool_relu_bias_bwd2_kernel(
     const unsigned short* __restrict__ dy, const uint8_t* __restrict__ idx,
     const unsigned short* __restrict__ p, unsigned short* __restrict__ dym,
-    float* __restrict__ db, int64_t total8, int K, int H, int W, int OH,
+    int64_t total8, int K, int H, int W, int OH,
     int OW, int HB, int WB, FastDiv fOct, FastDiv fWB, FastDiv fHB) {
-    // bias partials tree-reduce through LDS: per-thread ds_add atomics
-    // (2048/block onto K addresses) serialized into a ~55 us tail on the
-    // 224-class layers — measured 941 GB/s vs 6.5 TB/s on the forward
-    __shared__ float red8[256 * 8];
+    // NO bias reduction in this kernel: the pure scatter runs at the HBM
+    // roofline (6.2-6.5 TB/s measured); ANY reduction tail (LDS atomics or
+    // tree + barriers) serialized the block exit and cost 3-16x
+    // (gpurun_out bisect, HEFL_POOL_TAIL experiment). The bias grad is a
+    // separate gated reduction over the 4x-smaller POOLED tensors — each
+    // output cell contributes exactly once (at its argmax) iff p > 0.
     const int noct = K >> 3;
-    float acc[8] = {0.f};
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
          i < total8; i += (int64_t)gridDim.x * blockDim.x) {
         unsigned t = fdiv((unsigned)i, fOct);
@@ -2015,7 +2015,6 @@ __global__ void pool_relu_bias_bwd2_kernel(
                 const bool act = (p8[j] & 0x7fffu) != 0 && !(p8[j] & 0x8000u);
                 const int pos = (int)((i8 >> (8 * j)) & 3);
                 const unsigned short dv = act ? d8[j] : (unsigned short)0;
-                if (act) acc[j] += bf2f(d8[j]);
                 g0[j] = pos == 0 ? dv : (unsigned short)0;
                 g1[j] = pos == 1 ? dv : (unsigned short)0;
                 g2[j] = pos == 2 ? dv : (unsigned short)0;
@@ -2032,48 +2031,37 @@ __global__ void pool_relu_bias_bwd2_kernel(
                 *reinterpret_cast<u16x8*>(&dym[r0 + (int64_t)W * K + K]) = g3;
         }
     }
-    if (TAIL == 2) return;
-    if (TAIL == 0) {
-        const int my_oct = (int)(((int64_t)blockIdx.x * blockDim.x
-                                  + threadIdx.x) % noct);
-        float* dbs = red8;  // reuse as [K] accumulator
-        for (int c = threadIdx.x; c < K; c += blockDim.x) dbs[c] = 0.f;
-        __syncthreads();
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-            if (acc[j] != 0.f) atomicAdd(dbs + my_oct * 8 + j, acc[j]);
-        __syncthreads();
-        for (int c = threadIdx.x; c < K; c += blockDim.x)
-            if (dbs[c] != 0.f) atomicAdd(db + c, dbs[c]);
-        return;
+}
+
+// db[k] = sum over pooled cells of (p > 0 ? dy : 0) — the bias grad of the
+// fused trunk block, computed from the POOLED tensors (4x smaller than the
+// conv activation the old in-kernel tail walked).
+__global__ void bias_grad_gated_kernel(const unsigned short* __restrict__ dy,
+                                       const unsigned short* __restrict__ p,
+                                       float* __restrict__ db, int64_t M,
+                                       int K, int rows_per_block) {
+    __shared__ float red[256];
+    const int k = blockIdx.x;
+    const int64_t r0 = (int64_t)blockIdx.y * rows_per_block;
+    const int64_t r1 = min(r0 + rows_per_block, M);
+    float acc = 0.f;
+    for (int64_t r = r0 + threadIdx.x; r < r1; r += blockDim.x) {
+        const unsigned short pv = p[r * K + k];
+        if ((pv & 0x7fffu) != 0 && !(pv & 0x8000u)) acc += bf2f(dy[r * K + k]);
     }
-    // TAIL == 1: tree reduce over the lane dimension
-#pragma unroll
-    for (int j = 0; j < 8; ++j) red8[threadIdx.x * 8 + j] = acc[j];
+    red[threadIdx.x] = acc;
     __syncthreads();
-    const int oct0 = threadIdx.x % noct;
-    const int rl = threadIdx.x / noct;
-    const int lanes = 256 / noct;  // pow2
-    for (int off = lanes >> 1; off >= 1; off >>= 1) {
-        if (rl < off) {
-#pragma unroll
-            for (int j = 0; j < 8; ++j)
-                red8[threadIdx.x * 8 + j] +=
-                    red8[(threadIdx.x + off * noct) * 8 + j];
-        }
+    for (int off = 128; off > 0; off >>= 1) {
+        if (threadIdx.x < off) red[threadIdx.x] += red[threadIdx.x + off];
         __syncthreads();
     }
-    if (rl == 0) {
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-            const float v = red8[threadIdx.x * 8 + j];
-            if (gridDim.x == 1) db[oct0 * 8 + j] = v;
-            else if (v != 0.f) atomicAdd(db + oct0 * 8 + j, v);
-        }
+    if (threadIdx.x == 0) {
+        if (gridDim.y == 1) db[k] = red[0];
+        else atomicAdd(db + k, red[0]);
     }
 }
 
-// Scalar fallback for K % 8 != 0 trunk layers (LeNet-5's 6/16 filters):
+// Scalar fallback for K % 8 != 0 trunk layers// Scalar fallback for K % 8 != 0 trunk layers (LeNet-5's 6/16 filters):
 // thread = (channel, row-lane) so each thread's bias partial stays in a
 // register (naive per-element global atomics serialized on K=6 addresses
 // and cost config #3 ~40%); ReLU gate on the POOLED output p (= the
@@ -3674,28 +3662,23 @@ std::vector<torch::Tensor> pool_relu_bias_bwd(torch::Tensor dy,
         // 2x2-block octet kernel: dy/idx/p read once per output cell
         const int HB = (int)((H + 1) / 2), WB = (int)((W + 1) / 2);
         const int64_t total8 = (int64_t)N * HB * WB * noct;
-        // measured: forcing ONE block on small layers to skip the db
-        // zero-fill serialized the whole layer onto one CU (-30% on the
-        // headline config) — keep the parallel grid and pay the 4.7 us fill
         int blocks = (int)std::min<int64_t>(ceildiv(total8, 256), 4096);
-        auto db = blocks == 1
+        hipLaunchKernelGGL(pool_relu_bias_bwd2_kernel, dim3(blocks),
+                           dim3(256), 0, stream, bf_ptr(dyc),
+                           idx.data_ptr<uint8_t>(), bf_ptr(p),
+                           bf_ptr_mut(dym), total8, K, (int)H, (int)W, OH,
+                           OW, HB, WB, fdiv_make((unsigned)noct),
+                           fdiv_make((unsigned)WB), fdiv_make((unsigned)HB));
+        // bias grad from the POOLED tensors (4x smaller): p-gated column sum
+        const int64_t Mp = (int64_t)N * OH * OW;
+        int rpb = (int)std::max<int64_t>(256, (Mp + 511) / 512);
+        int chunks = (int)((Mp + rpb - 1) / rpb);
+        auto db = chunks == 1
                       ? torch::empty({K}, dyc.options().dtype(torch::kFloat32))
                       : torch::zeros({K}, dyc.options().dtype(torch::kFloat32));
-        static const int tail = [] {
-            const char* e = getenv("HEFL_POOL_TAIL");
-            return e ? e[0] - '0' : 0;
-        }();
-        #define LP(T) hipLaunchKernelGGL((pool_relu_bias_bwd2_kernel<T>),    \
-                           dim3(blocks), dim3(256), 0, stream, bf_ptr(dyc),  \
-                           idx.data_ptr<uint8_t>(), bf_ptr(p),               \
-                           bf_ptr_mut(dym), db.data_ptr<float>(), total8, K, \
-                           (int)H, (int)W, OH, OW, HB, WB,                   \
-                           fdiv_make((unsigned)noct),                        \
-                           fdiv_make((unsigned)WB), fdiv_make((unsigned)HB))
-        if (tail == 1) LP(1);
-        else if (tail == 2) LP(2);
-        else LP(0);
-        #undef LP
+        hipLaunchKernelGGL(bias_grad_gated_kernel, dim3(K, chunks), dim3(256),
+                           0, stream, bf_ptr(dyc), bf_ptr(p),
+                           db.data_ptr<float>(), Mp, K, rpb);
         return {dym, db};
     }
     TORCH_CHECK(K <= 256, "scalar pool-backward fallback expects K <= 256");
