@@ -278,3 +278,32 @@ def test_conv_relu_pool_fused_matches_composed(shape):
     _close(dxf, dxc, rel=1e-3)
     _close(dwf, dwc, rel=1e-3)
     _close(dbf, dbc, rel=1e-3)
+
+
+@pytest.mark.parametrize("shape", [
+    # (N, H, W, C, K, pad): 3x3 s1, C%32==0 -> direct tiled kernel
+    (4, 20, 20, 32, 64, 0),
+    (4, 21, 21, 32, 32, 1),   # odd output + pad-1 halo
+    (2, 10, 18, 64, 64, 1),   # 8x16 tile variant + C slab loop
+    (2, 33, 18, 64, 128, 0),  # tail tiles both dims, two Kout tiles
+])
+def test_conv_tile3_matches_glds(shape):
+    """Direct tiled 3x3 kernel vs the implicit-GEMM path on identical bf16
+    inputs (HEFL_TILE3=2 forces tiled dispatch on these small grids)."""
+    import os
+    import hefl
+    C_ = hefl.load_extension()
+    N, H, W, C, K, pad = shape
+    torch.manual_seed(0)
+    x = torch.randn(N, H, W, C, device="cuda", dtype=torch.bfloat16)
+    w = (torch.randn(K, 3, 3, C, device="cuda") * 0.1).to(torch.bfloat16)
+    b = torch.randn(K, device="cuda") * 0.1
+    os.environ["HEFL_TILE3"] = "2"
+    try:
+        y_tile = C_.conv2d_fwd(x, w, b, 1, True, pad)
+    finally:
+        os.environ["HEFL_TILE3"] = "0"
+    y_ref = C_.conv2d_fwd(x, w, b, 1, True, pad)
+    os.environ.pop("HEFL_TILE3", None)
+    assert torch.equal(y_tile, y_ref) or \
+        (y_tile.float() - y_ref.float()).abs().max().item() < 1e-2
