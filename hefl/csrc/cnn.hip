@@ -1982,19 +1982,19 @@ __global__ void maxpool_fwd_oct_kernel(const unsigned short* __restrict__ x,
 __global__ void pool_relu_bias_bwd2_kernel(
     const unsigned short* __restrict__ dy, const uint8_t* __restrict__ idx,
     const unsigned short* __restrict__ p, unsigned short* __restrict__ dym,
-    int64_t total8, int K, int H, int W, int OH,
+    float* __restrict__ db, int64_t total8, int K, int H, int W, int OH,
     int OW, int HB, int WB, FastDiv fOct, FastDiv fWB, FastDiv fHB) {
-    // NO bias reduction in this kernel: the pure scatter runs at the HBM
-    // roofline (6.2-6.5 TB/s measured); ANY reduction tail (LDS atomics or
-    // tree + barriers) serialized the block exit and cost 3-16x
-    // (gpurun_out bisect, HEFL_POOL_TAIL experiment). The bias grad is a
-    // separate gated reduction over the 4x-smaller POOLED tensors — each
-    // output cell contributes exactly once (at its argmax) iff p > 0.
+    extern __shared__ float dbs[];  // [K]
     const int noct = K >> 3;
+    for (int c = threadIdx.x; c < K; c += blockDim.x) dbs[c] = 0.f;
+    __syncthreads();
+    float acc[8] = {0.f};
+    int my_oct = -1;
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
          i < total8; i += (int64_t)gridDim.x * blockDim.x) {
         unsigned t = fdiv((unsigned)i, fOct);
         const int oct = (int)((unsigned)i - t * noct);
+        my_oct = oct;  // constant per thread: stride % noct == 0 (noct pow2)
         unsigned t2 = fdiv(t, fWB);
         const int wb = (int)(t - t2 * WB);
         unsigned t3 = fdiv(t2, fHB);
@@ -2015,6 +2015,7 @@ __global__ void pool_relu_bias_bwd2_kernel(
                 const bool act = (p8[j] & 0x7fffu) != 0 && !(p8[j] & 0x8000u);
                 const int pos = (int)((i8 >> (8 * j)) & 3);
                 const unsigned short dv = act ? d8[j] : (unsigned short)0;
+                if (act) acc[j] += bf2f(d8[j]);
                 g0[j] = pos == 0 ? dv : (unsigned short)0;
                 g1[j] = pos == 1 ? dv : (unsigned short)0;
                 g2[j] = pos == 2 ? dv : (unsigned short)0;
@@ -2031,54 +2032,17 @@ __global__ void pool_relu_bias_bwd2_kernel(
                 *reinterpret_cast<u16x8*>(&dym[r0 + (int64_t)W * K + K]) = g3;
         }
     }
-}
-
-// db[k] = sum over pooled cells of (p > 0 ? dy : 0) — the bias grad of the
-// fused trunk block, computed from the POOLED tensors (4x smaller than the
-// conv activation the old in-kernel tail walked).
-__global__ void bias_grad_gated_kernel(const unsigned short* __restrict__ dy,
-                                       const unsigned short* __restrict__ p,
-                                       float* __restrict__ db, int64_t M,
-                                       int K, int rows_per_block) {
-    // octet rows: thread = (8-channel group, row-lane), two 16-B loads per
-    // row (scalar 2-B loads were load-instruction bound at 0.9 TB/s).
-    // Reduction-only kernel, >100 rows per thread: the LDS tree tail
-    // amortizes (unlike inside the roofline scatter kernel).
-    __shared__ float red8[256 * 8];
-    const int noct = K >> 3;  // pow2 (host guards)
-    const int64_t r0 = (int64_t)blockIdx.x * rows_per_block;
-    const int64_t r1 = min(r0 + rows_per_block, M);
-    const int oct = threadIdx.x % noct;
-    const int rl = threadIdx.x / noct;
-    const int lanes = 256 / noct;
-    float acc[8] = {0.f};
-    for (int64_t r = r0 + rl; r < r1; r += lanes) {
-        const u16x8 d8 = *reinterpret_cast<const u16x8*>(&dy[r * K + oct * 8]);
-        const u16x8 p8 = *reinterpret_cast<const u16x8*>(&p[r * K + oct * 8]);
+    if (my_oct >= 0) {
 #pragma unroll
         for (int j = 0; j < 8; ++j)
-            if ((p8[j] & 0x7fffu) != 0 && !(p8[j] & 0x8000u))
-                acc[j] += bf2f(d8[j]);
+            if (acc[j] != 0.f) atomicAdd(dbs + my_oct * 8 + j, acc[j]);
     }
-#pragma unroll
-    for (int j = 0; j < 8; ++j) red8[threadIdx.x * 8 + j] = acc[j];
     __syncthreads();
-    for (int off = lanes >> 1; off >= 1; off >>= 1) {
-        if (rl < off) {
-#pragma unroll
-            for (int j = 0; j < 8; ++j)
-                red8[threadIdx.x * 8 + j] +=
-                    red8[(threadIdx.x + off * noct) * 8 + j];
-        }
-        __syncthreads();
-    }
-    if (rl == 0) {
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-            const float v = red8[threadIdx.x * 8 + j];
-            if (gridDim.x == 1) db[oct * 8 + j] = v;
-            else if (v != 0.f) atomicAdd(db + oct * 8 + j, v);
-        }
+    if (gridDim.x == 1) {  // small layers: direct store, db needs no zero-fill
+        for (int c = threadIdx.x; c < K; c += blockDim.x) db[c] = dbs[c];
+    } else {
+        for (int c = threadIdx.x; c < K; c += blockDim.x)
+            if (dbs[c] != 0.f) atomicAdd(db + c, dbs[c]);
     }
 }
 
@@ -3686,28 +3650,24 @@ std::vector<torch::Tensor> pool_relu_bias_bwd(torch::Tensor dy,
     auto dym = torch::empty({N, H, W, (int64_t)K}, dyc.options());
     auto stream = at::cuda::getCurrentCUDAStream();
     const int noct = K >> 3;
-    if ((K & 7) == 0 && (noct & (noct - 1)) == 0 && K <= 256) {
+    if ((K & 7) == 0 && (noct & (noct - 1)) == 0 && noct <= 256) {
         // 2x2-block octet kernel: dy/idx/p read once per output cell
         const int HB = (int)((H + 1) / 2), WB = (int)((W + 1) / 2);
         const int64_t total8 = (int64_t)N * HB * WB * noct;
+        // measured: forcing ONE block on small layers to skip the db
+        // zero-fill serialized the whole layer onto one CU (-30% on the
+        // headline config) — keep the parallel grid and pay the 4.7 us fill
         int blocks = (int)std::min<int64_t>(ceildiv(total8, 256), 4096);
-        hipLaunchKernelGGL(pool_relu_bias_bwd2_kernel, dim3(blocks),
-                           dim3(256), 0, stream, bf_ptr(dyc),
-                           idx.data_ptr<uint8_t>(), bf_ptr(p),
-                           bf_ptr_mut(dym), total8, K, (int)H, (int)W, OH,
-                           OW, HB, WB, fdiv_make((unsigned)noct),
-                           fdiv_make((unsigned)WB), fdiv_make((unsigned)HB));
-        // bias grad from the POOLED tensors (4x smaller): p-gated row sum
-        const int64_t Mp = (int64_t)N * OH * OW;
-        int lanes = 256 / noct;
-        int rpb = (int)std::max<int64_t>(2 * lanes, (Mp + 511) / 512);
-        int chunks = (int)((Mp + rpb - 1) / rpb);
-        auto db = chunks == 1
+        auto db = blocks == 1
                       ? torch::empty({K}, dyc.options().dtype(torch::kFloat32))
                       : torch::zeros({K}, dyc.options().dtype(torch::kFloat32));
-        hipLaunchKernelGGL(bias_grad_gated_kernel, dim3(chunks), dim3(256),
-                           0, stream, bf_ptr(dyc), bf_ptr(p),
-                           db.data_ptr<float>(), Mp, K, rpb);
+        hipLaunchKernelGGL(pool_relu_bias_bwd2_kernel, dim3(blocks),
+                           dim3(256), K * sizeof(float), stream, bf_ptr(dyc),
+                           idx.data_ptr<uint8_t>(), bf_ptr(p),
+                           bf_ptr_mut(dym), db.data_ptr<float>(), total8, K,
+                           (int)H, (int)W, OH, OW, HB, WB,
+                           fdiv_make((unsigned)noct), fdiv_make((unsigned)WB),
+                           fdiv_make((unsigned)HB));
         return {dym, db};
     }
     TORCH_CHECK(K <= 256, "scalar pool-backward fallback expects K <= 256");
