@@ -754,7 +754,8 @@ conv_fwd_tile3_kernel(const unsigned short* __restrict__ x,
     // bank alias (rule 21)
     constexpr int XP = CS + 8;
     __shared__ unsigned short xs[XH * XW * XP];
-    __shared__ unsigned short ws[9 * BN * CS];
+    constexpr int WP = CS + 8;  // same bank-alias pad as the x-tile
+    __shared__ unsigned short ws[9 * BN * WP];
     const int tid = threadIdx.x;
     const int wave = tid >> 6;
     const int lane = tid & 63;
@@ -801,7 +802,7 @@ conv_fwd_tile3_kernel(const unsigned short* __restrict__ x,
             if (n0 + ko < s.Kout)
                 v = *reinterpret_cast<const u16x8*>(
                     &w[(int64_t)(n0 + ko) * KK + rs * s.C + c0 + cc]);
-            *reinterpret_cast<u16x8*>(&ws[(rs * BN + ko) * CS + cc]) = v;
+            *reinterpret_cast<u16x8*>(&ws[(rs * BN + ko) * WP + cc]) = v;
         }
         __syncthreads();
 #pragma unroll
@@ -819,7 +820,7 @@ conv_fwd_tile3_kernel(const unsigned short* __restrict__ x,
             for (int j = 0; j < FN; ++j) {
                 const int ko = wn * FN * 16 + j * 16 + sub;
                 b[j] = *reinterpret_cast<const bf16x8*>(
-                    &ws[(rs * BN + ko) * CS + half * 8]);
+                    &ws[(rs * BN + ko) * WP + half * 8]);
             }
 #pragma unroll
             for (int i = 0; i < FM; ++i)
