@@ -851,6 +851,22 @@ conv_fwd_tile3_kernel(const unsigned short* __restrict__ x,
     }
 }
 
+// wT[c][r][s][ko] = w[ko][2-r][2-s][c]: the 180-rotated transpose that
+// turns dgrad into a plain forward conv of dy (tiny tensor; one launch).
+__global__ void rot180_transpose_w_kernel(const unsigned short* __restrict__ w,
+                                          unsigned short* __restrict__ wT,
+                                          int Kout, int C, int64_t total) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         i < total; i += (int64_t)gridDim.x * blockDim.x) {
+        const int ko = (int)(i % Kout);
+        const int64_t t = i / Kout;
+        const int rs = (int)(t % 9);
+        const int c = (int)(t / 9);
+        const int r = rs / 3, ss = rs - r * 3;
+        wT[i] = w[(((int64_t)ko * 3 + (2 - r)) * 3 + (2 - ss)) * C + c];
+    }
+}
+
 // ---- dgrad: dx[m=(n,ih,iw), c] = sum_k A(m,k) * B(k,c),
 //      k = (r, s, ko) with ko FASTEST so the dy gather is contiguous;
 //      B(k, c) = w[ko, r, s, c] (strided, small tile) ----
@@ -3304,6 +3320,45 @@ torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor w, int64_t stride,
     auto dx = torch::empty({N, (int64_t)H, (int64_t)W, C}, dy.options());
     const int M = N * (int)H * (int)W;
     auto stream = at::cuda::getCurrentCUDAStream();
+    // dgrad as a direct tiled conv of dy with the 180-rotated transposed
+    // weights: dx = conv3x3(dy, wT, pad' = 2 - pad). Same winner shape
+    // class as forward (output channels <= 32 -> the BN=32 16x16 tile).
+    const char* t3env_d = getenv("HEFL_TILE3");
+    const int t3mode_d = t3env_d ? t3env_d[0] - '0' : 1;
+    if (t3mode_d != 0 && s.R == 3 && s.S == 3 && stride == 1 && pad <= 1 &&
+        s.Kout % 32 == 0 && C > 16 && (t3mode_d == 2 || C <= 32)) {
+        ConvShape s3;
+        s3.N = N; s3.H = s.OH; s3.W = s.OW; s3.C = s.Kout;
+        s3.Kout = C; s3.R = 3; s3.S = 3; s3.stride = 1; s3.pad = 2 - (int)pad;
+        s3.OH = (int)H; s3.OW = (int)W;
+        fill_magic(s3);
+        const int BN3 = C <= 32 ? 32 : 64;
+        const int kt = ceildiv(C, BN3);
+        const int64_t min_tiles = t3mode_d == 2 ? 1 : 256;
+        const int64_t t16 = (int64_t)N * ceildiv(s3.OH, 16)
+                            * ceildiv(s3.OW, 16) * kt;
+        if (s3.OH >= 12 && t16 >= min_tiles) {
+            auto wT = torch::empty({C, 3, 3, s.Kout}, w.options());
+            const int64_t tot = (int64_t)C * 9 * s.Kout;
+            hipLaunchKernelGGL(rot180_transpose_w_kernel,
+                               dim3((int)std::min<int64_t>(ceildiv(tot, 256), 1024)),
+                               dim3(256), 0, stream, bf_ptr(w), bf_ptr_mut(wT),
+                               s.Kout, C, tot);
+            const int th_ = ceildiv(s3.OH, 16), tw_ = ceildiv(s3.OW, 16);
+            dim3 grid((unsigned)(N * th_ * tw_), (unsigned)kt);
+            if (BN3 == 32)
+                hipLaunchKernelGGL((conv_fwd_tile3_kernel<16, 16, 32, 4, 1, 4, 2>),
+                                   grid, dim3(TPB), 0, stream, bf_ptr(dy),
+                                   bf_ptr(wT), (const float*)nullptr,
+                                   bf_ptr_mut(dx), s3, 0, th_, tw_);
+            else
+                hipLaunchKernelGGL((conv_fwd_tile3_kernel<16, 16, 64, 4, 1, 4, 4>),
+                                   grid, dim3(TPB), 0, stream, bf_ptr(dy),
+                                   bf_ptr(wT), (const float*)nullptr,
+                                   bf_ptr_mut(dx), s3, 0, th_, tw_);
+            return dx;
+        }
+    }
     if (C > 16 && C % 8 == 0 && s.Kout % 8 == 0) {
         static torch::Tensor zbuf;
         if (!zbuf.defined() || zbuf.device() != dy.device())
