@@ -3202,7 +3202,7 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
                                            FN_>),                             \
                     grid, dim3(TPB), 0, stream, bf_ptr(x), bf_ptr(w), bias,   \
                     bf_ptr_mut(y), s, relu ? 1 : 0, th_, tw_)
-            if (TH == 16 && BN3 == 64)      LAUNCH_T3(16, 16, 64, 4, 1, 4, 4);
+            if (TH == 16 && BN3 == 64)      LAUNCH_T3(16, 16, 64, 2, 2, 8, 2);
             else if (TH == 16)              LAUNCH_T3(16, 16, 32, 4, 1, 4, 2);
             else if (BN3 == 64)             LAUNCH_T3(8, 16, 64, 2, 2, 4, 2);
             else                            LAUNCH_T3(8, 16, 32, 2, 2, 4, 1);
