@@ -3176,7 +3176,7 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
     // read per call so tests can toggle it.
     const char* t3env = getenv("HEFL_TILE3");
     const int t3mode = t3env ? t3env[0] - '0' : 1;
-    // A/B (gpurun_out/bench_conv_tile3 vs _notile): the 16x16/BN=32 variant
+    // A/B (profiles/r02_bench_conv_tile3 vs _notile): the 16x16/BN=32 variant
     // wins +34% on Kout<=32 layers (their BN=64 implicit-GEMM tile wasted
     // half its columns AND re-gathered 9x); the BN=64 and 8x16 variants
     // LOSE 10-45% to the glds pipeline (fewer resident blocks, conflictier
@@ -3210,7 +3210,7 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
             return y;
         }
     }
-    // Measured on MI355X (gpurun_out/bench_conv_glds64 vs _glds32): the
+    // Measured on MI355X (profiles/r02_bench_conv_glds64 vs _glds32): the
     // BK=64 two-buffer kernel LOSES 8-30% to the BK=32 three-buffer one at
     // every conv shape here — K loops are 2-18 steps, so the prologue DMA
     // latency (3-buf pre-stages two tiles, 2-buf one) outweighs the halved
@@ -3473,7 +3473,7 @@ torch::Tensor conv2d_wgrad(torch::Tensor dy, torch::Tensor x, int64_t stride,
     }
     // measured: slab-rows + reduce LOSES to fp32 atomics here (CDNA4 L2
     // atomics absorb the z-chunk contention; the slab variant paid extra
-    // write+read traffic) — gpurun_out/bench_conv_p5 vs bench_conv_glds32
+    // write+read traffic) — profiles/r02_bench_conv logs
     auto dw = k_chunks > 1
                   ? torch::zeros({s.Kout, R, S, s.C},
                                  x.options().dtype(torch::kFloat32))
