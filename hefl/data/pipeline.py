@@ -35,20 +35,24 @@ def get_test_data(ds: SyntheticMedicalImages, batch_size: int = 32) -> ClientLoa
 
 def get_train_data(ds: SyntheticMedicalImages, client: int, n_clients: int,
                    batch_size: int = 32, val_frac: float = 0.1,
-                   seed: int = 0, augment: bool = True
-                   ) -> Tuple[ClientLoader, Optional[ClientLoader]]:
+                   seed: int = 0, augment: bool = True,
+                   affine=None) -> Tuple[ClientLoader, Optional[ClientLoader]]:
     """Client shard -> (train loader, val loader).
 
     Shard semantics are the reference's contiguous equal slices
     (FLPyfhelin.py:75-78); validation is the trailing val_frac of the shard
-    (reference: validation_split=0.1, :88-99).
+    (reference: validation_split=0.1, :88-99). `affine` = (zoom, shear,
+    hflip) runs the full in-generator transform (the reference's shear 0.2
+    / zoom 0.2 / h-flip set, :80-86); `augment` alone keeps the post-batch
+    h-flip hook.
     """
     idx = shard_indices(ds.n_samples, client, n_clients)
     n_val = int(idx.numel() * val_frac)
     train_idx = idx[: idx.numel() - n_val]
     val_idx = idx[idx.numel() - n_val:]
-    aug = hflip_augment if augment else None
-    train = ClientLoader(ds, train_idx, batch_size, seed=seed, augment=aug)
+    aug = hflip_augment if (augment and affine is None) else None
+    train = ClientLoader(ds, train_idx, batch_size, seed=seed, augment=aug,
+                         affine=affine)
     val = (ClientLoader(ds, val_idx, batch_size, shuffle=False)
            if n_val else None)
     return train, val

@@ -67,7 +67,7 @@ class LocalClient:
         # augmentation applies to TRAINING batches only (the reference's
         # ImageDataGenerator transforms train, not val/test)
         affine = {"none": None, "hflip": (0.0, 0.0, True),
-                  "full": (0.2, 0.2, True)}[getattr(cfg.fl, "augment", "none")]
+                  "full": (0.2, 0.2, True)}[cfg.fl.augment]
         self.affine = affine
         self.loader = ClientLoader(self.dataset, train_idx, t.batch_size,
                                    seed=cfg.fl.seed + client_id,
