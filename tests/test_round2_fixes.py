@@ -257,3 +257,33 @@ def test_callbacks_react_to_val_metrics():
                         for lg in seen)
     # ES tracked the val metric (its best is a real val_loss it saw)
     assert es.best == min(lg["val_loss"] for lg in seen)
+
+
+def _worker_four(rank, world, port, q):
+    os.environ.update({
+        "RANK": str(rank), "LOCAL_RANK": str(rank), "WORLD_SIZE": str(world),
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+    })
+    import torch.distributed as dist
+    from hefl.fl.secure import SecureAggregator
+    from hefl.parallel.dist import init_distributed
+
+    init_distributed(backend="gloo")
+    # unseeded keygen + multi-slab pipeline at world 4 (broadcast + bucket
+    # paths together, above the 2-rank smoke level)
+    cfg = HEConfig(m=128, scale_bits=30, q_bits=(50, 30), seed=None)
+    agg = SecureAggregator(CKKSContext(cfg), rank=rank, bucket_bytes=4096)
+    vec = torch.randn(500, generator=torch.Generator().manual_seed(100 + rank))
+    out = agg.fedavg(vec, n_clients=world)
+    expect = torch.stack(
+        [torch.randn(500, generator=torch.Generator().manual_seed(100 + r))
+         for r in range(world)]).mean(0)
+    q.put((rank, (out - expect).abs().max().item()))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_four_rank_unseeded_bucketed_fedavg():
+    from conftest import free_port
+    for rank, err in _run(_worker_four, 4, free_port()):
+        assert err < 1e-3, (rank, err)
