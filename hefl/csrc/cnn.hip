@@ -3471,6 +3471,24 @@ torch::Tensor conv2d_wgrad(torch::Tensor dy, torch::Tensor x, int64_t stride,
                            dw.data_ptr<float>(), total, kc);
         return dw;
     }
+    if (glds_ok && s.Kout <= 32) {
+        // 32-ko tile, 64-pixel K-step: no ko waste on 32-filter layers
+        static torch::Tensor zbuf32;
+        if (!zbuf32.defined() || zbuf32.device() != dy.device())
+            zbuf32 = torch::zeros({8}, dy.options());
+        int tiles32 = ceildiv(s.Kout, 32) * ceildiv(NN, 64);
+        int kc = std::max(1, std::min(ceildiv(KK, 128),
+                                      512 / std::max(tiles32, 1)));
+        auto dw32 = kc > 1 ? torch::zeros({s.Kout, R, S, s.C},
+                                          x.options().dtype(torch::kFloat32))
+                           : torch::empty({s.Kout, R, S, s.C},
+                                          x.options().dtype(torch::kFloat32));
+        dim3 g32(ceildiv(s.Kout, 32), ceildiv(NN, 64), kc);
+        hipLaunchKernelGGL((conv_wgrad_glds_k32_kernel<64, 2, 2, 1, 2>), g32,
+                           dim3(TPB), 0, stream, bf_ptr(dy), bf_ptr(x),
+                           dw32.data_ptr<float>(), bf_ptr(zbuf32), s, kc);
+        return dw32;
+    }
     // measured: slab-rows + reduce LOSES to fp32 atomics here (CDNA4 L2
     // atomics absorb the z-chunk contention; the slab variant paid extra
     // write+read traffic) — profiles/r02_bench_conv logs
@@ -3480,23 +3498,6 @@ torch::Tensor conv2d_wgrad(torch::Tensor dy, torch::Tensor x, int64_t stride,
                   : torch::empty({s.Kout, R, S, s.C},
                                  x.options().dtype(torch::kFloat32));
     dim3 grid(ceildiv(s.Kout, 64), ceildiv(NN, bn), k_chunks);
-    if (glds_ok && s.Kout <= 32) {
-        // 32-ko tile, 64-pixel K-step: no ko waste on 32-filter layers
-        static torch::Tensor zbuf;
-        if (!zbuf.defined() || zbuf.device() != dy.device())
-            zbuf = torch::zeros({8}, dy.options());
-        int tiles32 = ceildiv(s.Kout, 32) * ceildiv(NN, 64);
-        int kc = std::max(1, std::min(ceildiv(KK, 128),
-                                      512 / std::max(tiles32, 1)));
-        auto dw32 = kc > 1 ? torch::zeros({s.Kout, R, S, s.C},
-                                          x.options().dtype(torch::kFloat32))
-                           : dw;
-        dim3 g32(ceildiv(s.Kout, 32), ceildiv(NN, 64), kc);
-        hipLaunchKernelGGL((conv_wgrad_glds_k32_kernel<64, 2, 2, 1, 2>), g32,
-                           dim3(TPB), 0, stream, bf_ptr(dy), bf_ptr(x),
-                           dw32.data_ptr<float>(), bf_ptr(zbuf), s, kc);
-        return dw32;
-    }
     if (glds_ok) {
         static torch::Tensor zbuf;
         if (!zbuf.defined() || zbuf.device() != dy.device())
