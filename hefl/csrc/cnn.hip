@@ -1494,14 +1494,15 @@ conv_wgrad_glds_kernel(const unsigned short* __restrict__ dy,
 // per barrier, no ko waste, half the barriers per pixel. Pipeline and
 // gather structure mirror conv_wgrad_glds_kernel (3-buffer counted
 // vmcnt(3): three 16-B glds per thread per stage). ----
-template <int BN, int WM, int WN, int FM, int FN>
+template <int BN, int WM, int WN, int FM, int FN, int BKP = 64>
 __global__ void __launch_bounds__(TPB)
 conv_wgrad_glds_k32_kernel(const unsigned short* __restrict__ dy,
                            const unsigned short* __restrict__ x,
                            float* __restrict__ dw,
                            const unsigned short* __restrict__ zbuf,
                            ConvShape s, int k_chunks) {
-    constexpr int BKP = 64;  // pixels per K-step
+    // BKP = pixels per K-step (64 or 128; 128 halves the barrier count
+    // again at 3 LDS buffers x 24 KB)
     __shared__ unsigned short smem[3 * (32 + BN) * BKP];
     auto Dys = [&](int buf) -> unsigned short (*)[32] {
         return reinterpret_cast<unsigned short(*)[32]>(
@@ -1524,15 +1525,17 @@ conv_wgrad_glds_k32_kernel(const unsigned short* __restrict__ dy,
     const int kend = min(kbeg + chunk, KK);
 
     auto stage = [&](int buf, int k0) {
-        {   // Dys[pix][ko]: 64 pix x 4 ko-chunks = 256 = one DMA per thread
-            const int pix = tid >> 2;
-            const int koc = (tid & 3) * 8;
+#pragma unroll
+        for (int t = 0; t < BKP * 4 / TPB; ++t) {  // Dys[pix][ko] chunks
+            const int i = tid + t * TPB;
+            const int pix = i >> 2;
+            const int koc = (i & 3) * 8;
             const int kpix = k0 + pix;
             const int ko = m0 + koc;
             const unsigned short* src = zbuf;
             if (kpix < kend && ko + 8 <= M)
                 src = dy + (int64_t)kpix * s.Kout + ko;
-            glds16(src, (char*)&Dys(buf)[0][0] + wave * 1024);
+            glds16(src, (char*)&Dys(buf)[0][0] + (wave + t * 4) * 1024);
         }
 #pragma unroll
         for (int t = 0; t < BN * 8 / TPB; ++t) {  // Xs: 64 pix x BN/8 chunks
@@ -1571,9 +1574,13 @@ conv_wgrad_glds_k32_kernel(const unsigned short* __restrict__ dy,
     stage(1, kbeg + BKP);
     int buf = 0;
     for (int k0 = kbeg; k0 < kend; k0 += BKP) {
-        if (k0 + BKP < kend)
-            asm volatile("s_waitcnt vmcnt(3)" ::: "memory");
-        else
+        if (k0 + BKP < kend) {
+            // one stage in flight = (BKP*4 + BKP*8)/TPB glds per thread
+            if constexpr (BKP == 64)
+                asm volatile("s_waitcnt vmcnt(3)" ::: "memory");
+            else
+                asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+        } else
             asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         asm volatile("s_barrier" ::: "memory");
         if (k0 + 2 * BKP < kend) stage(buf == 0 ? 2 : buf - 1, k0 + 2 * BKP);
@@ -3476,6 +3483,10 @@ torch::Tensor conv2d_wgrad(torch::Tensor dy, torch::Tensor x, int64_t stride,
         static torch::Tensor zbuf32;
         if (!zbuf32.defined() || zbuf32.device() != dy.device())
             zbuf32 = torch::zeros({8}, dy.options());
+        static const bool bkp128 = [] {
+            const char* e = getenv("HEFL_K32B");
+            return e && e[0] == '1';  // probe: 128-pixel K-steps
+        }();
         int tiles32 = ceildiv(s.Kout, 32) * ceildiv(NN, 64);
         int kc = std::max(1, std::min(ceildiv(KK, 128),
                                       512 / std::max(tiles32, 1)));
@@ -3484,9 +3495,16 @@ torch::Tensor conv2d_wgrad(torch::Tensor dy, torch::Tensor x, int64_t stride,
                            : torch::empty({s.Kout, R, S, s.C},
                                           x.options().dtype(torch::kFloat32));
         dim3 g32(ceildiv(s.Kout, 32), ceildiv(NN, 64), kc);
-        hipLaunchKernelGGL((conv_wgrad_glds_k32_kernel<64, 2, 2, 1, 2>), g32,
-                           dim3(TPB), 0, stream, bf_ptr(dy), bf_ptr(x),
-                           dw32.data_ptr<float>(), bf_ptr(zbuf32), s, kc);
+        if (bkp128)
+            hipLaunchKernelGGL((conv_wgrad_glds_k32_kernel<64, 2, 2, 1, 2, 128>),
+                               g32, dim3(TPB), 0, stream, bf_ptr(dy),
+                               bf_ptr(x), dw32.data_ptr<float>(),
+                               bf_ptr(zbuf32), s, kc);
+        else
+            hipLaunchKernelGGL((conv_wgrad_glds_k32_kernel<64, 2, 2, 1, 2>),
+                               g32, dim3(TPB), 0, stream, bf_ptr(dy),
+                               bf_ptr(x), dw32.data_ptr<float>(),
+                               bf_ptr(zbuf32), s, kc);
         return dw32;
     }
     // measured: slab-rows + reduce LOSES to fp32 atomics here (CDNA4 L2
