@@ -1357,20 +1357,20 @@ conv_wgrad_kernel(const unsigned short* __restrict__ dy,
 // ---- glds-pipelined wgrad: both tiles (Dys pixel-major [32][64] and
 // Xs pixel-major [32][BN]) stage as lane-linear 16-B DMAs; 2-buffer
 // pipeline like fwd/dgrad. Requires Kout % 8 == 0 and C % 8 == 0. ----
-template <int BN, int WM, int WN, int FM, int FN>
+template <int BN, int WM, int WN, int FM, int FN, int BKP = 32>
 __global__ void __launch_bounds__(TPB)
 conv_wgrad_glds_kernel(const unsigned short* __restrict__ dy,
                        const unsigned short* __restrict__ x,
                        float* __restrict__ dw,
                        const unsigned short* __restrict__ zbuf, ConvShape s,
                        int k_chunks) {
-    __shared__ unsigned short smem[3 * (64 + BN) * 32];
+    __shared__ unsigned short smem[3 * (64 + BN) * BKP];
     auto Dys = [&](int buf) -> unsigned short (*)[64] {
-        return reinterpret_cast<unsigned short(*)[64]>(smem + buf * (64 + BN) * 32);
+        return reinterpret_cast<unsigned short(*)[64]>(smem + buf * (64 + BN) * BKP);
     };
     auto Xs = [&](int buf) -> unsigned short (*)[BN] {
-        return reinterpret_cast<unsigned short(*)[BN]>(smem + buf * (64 + BN) * 32
-                                                        + 64 * 32);
+        return reinterpret_cast<unsigned short(*)[BN]>(smem + buf * (64 + BN) * BKP
+                                                        + 64 * BKP);
     };
     const int tid = threadIdx.x;
     const int wave = tid >> 6;
@@ -1385,18 +1385,20 @@ conv_wgrad_glds_kernel(const unsigned short* __restrict__ dy,
     const int kend = min(kbeg + chunk, KK);
 
     auto stage = [&](int buf, int k0) {
-        {   // Dys[pix][ko]: chunk tid -> byte tid*16 of its wave slice
-            const int pix = tid >> 3;
-            const int koc = (tid & 7) * 8;
+#pragma unroll
+        for (int t = 0; t < BKP * 8 / TPB; ++t) {  // Dys[pix][ko] chunks
+            const int i = tid + t * TPB;
+            const int pix = i >> 3;
+            const int koc = (i & 7) * 8;
             const int kpix = k0 + pix;
             const int ko = m0 + koc;
             const unsigned short* src = zbuf;
             if (kpix < kend && ko + 8 <= M)
                 src = dy + (int64_t)kpix * s.Kout + ko;
-            glds16(src, (char*)&Dys(buf)[0][0] + wave * 1024);
+            glds16(src, (char*)&Dys(buf)[0][0] + (wave + t * 4) * 1024);
         }
 #pragma unroll
-        for (int t = 0; t < BN * 4 / TPB; ++t) {  // Xs chunks
+        for (int t = 0; t < BKP * (BN / 8) / TPB; ++t) {  // Xs chunks
             const int i = tid + t * TPB;
             const int pix = i / (BN / 8);
             const int cc = (i % (BN / 8)) * 8;
@@ -1430,41 +1432,48 @@ conv_wgrad_glds_kernel(const unsigned short* __restrict__ dy,
 
     // 3-buffer counted-vmcnt pipeline (see conv_fwd_glds_kernel)
     stage(0, kbeg);
-    stage(1, kbeg + 32);
+    stage(1, kbeg + BKP);
     int buf = 0;
-    for (int k0 = kbeg; k0 < kend; k0 += 32) {
-        if (k0 + 32 < kend)
-            asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
-        else
+    for (int k0 = kbeg; k0 < kend; k0 += BKP) {
+        if (k0 + BKP < kend) {
+            // one stage in flight = BKP*(8 + BN/8)/TPB glds per thread
+            if constexpr (BKP == 32)
+                asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+            else
+                asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+        } else
             asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         asm volatile("s_barrier" ::: "memory");
-        if (k0 + 64 < kend) stage(buf == 0 ? 2 : buf - 1, k0 + 64);
+        if (k0 + 2 * BKP < kend) stage(buf == 0 ? 2 : buf - 1, k0 + 2 * BKP);
         {
             const int wm = wave / WN, wn = wave % WN;
             const int half = lane >> 4, sub = lane & 15;
-            bf16x8 a[FM], b[FN];
 #pragma unroll
-            for (int i = 0; i < FM; ++i) {
-                const int mrow = wm * FM * 16 + i * 16 + sub;
+            for (int kk = 0; kk < BKP; kk += 32) {
+                bf16x8 a[FM], b[FN];
 #pragma unroll
-                for (int j = 0; j < 8; ++j)
-                    a[i][j] = *reinterpret_cast<const bf16_t*>(
-                        &Dys(buf)[half * 8 + j][mrow]);
+                for (int i = 0; i < FM; ++i) {
+                    const int mrow = wm * FM * 16 + i * 16 + sub;
+#pragma unroll
+                    for (int j = 0; j < 8; ++j)
+                        a[i][j] = *reinterpret_cast<const bf16_t*>(
+                            &Dys(buf)[kk + half * 8 + j][mrow]);
+                }
+#pragma unroll
+                for (int j = 0; j < FN; ++j) {
+                    const int col = wn * FN * 16 + j * 16 + sub;
+#pragma unroll
+                    for (int t = 0; t < 8; ++t)
+                        b[j][t] = *reinterpret_cast<const bf16_t*>(
+                            &Xs(buf)[kk + half * 8 + t][col]);
+                }
+#pragma unroll
+                for (int i = 0; i < FM; ++i)
+#pragma unroll
+                    for (int j = 0; j < FN; ++j)
+                        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            a[i], b[j], acc[i][j], 0, 0, 0);
             }
-#pragma unroll
-            for (int j = 0; j < FN; ++j) {
-                const int col = wn * FN * 16 + j * 16 + sub;
-#pragma unroll
-                for (int t = 0; t < 8; ++t)
-                    b[j][t] = *reinterpret_cast<const bf16_t*>(
-                        &Xs(buf)[half * 8 + t][col]);
-            }
-#pragma unroll
-            for (int i = 0; i < FM; ++i)
-#pragma unroll
-                for (int j = 0; j < FN; ++j)
-                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        a[i], b[j], acc[i][j], 0, 0, 0);
         }
         buf = buf == 2 ? 0 : buf + 1;
     }
@@ -3483,9 +3492,11 @@ torch::Tensor conv2d_wgrad(torch::Tensor dy, torch::Tensor x, int64_t stride,
         static torch::Tensor zbuf32;
         if (!zbuf32.defined() || zbuf32.device() != dy.device())
             zbuf32 = torch::zeros({8}, dy.options());
+        // 128-pixel K-steps measured +15-16% over 64 (wgrad was
+        // barrier-amortization bound); HEFL_K32B=0 reverts for A/B
         static const bool bkp128 = [] {
             const char* e = getenv("HEFL_K32B");
-            return e && e[0] == '1';  // probe: 128-pixel K-steps
+            return !e || e[0] != '0';
         }();
         int tiles32 = ceildiv(s.Kout, 32) * ceildiv(NN, 64);
         int kc = std::max(1, std::min(ceildiv(KK, 128),
@@ -3520,9 +3531,20 @@ torch::Tensor conv2d_wgrad(torch::Tensor dy, torch::Tensor x, int64_t stride,
         static torch::Tensor zbuf;
         if (!zbuf.defined() || zbuf.device() != dy.device())
             zbuf = torch::zeros({8}, dy.options());
-        hipLaunchKernelGGL((conv_wgrad_glds_kernel<64, 2, 2, 2, 2>), grid,
-                           dim3(TPB), 0, stream, bf_ptr(dy), bf_ptr(x),
-                           dw.data_ptr<float>(), bf_ptr(zbuf), s, k_chunks);
+        static const bool wgb64 = [] {
+            const char* e = getenv("HEFL_WGB");
+            return e && e[0] == '1';  // probe: 64-pixel K-steps
+        }();
+        if (wgb64)
+            hipLaunchKernelGGL((conv_wgrad_glds_kernel<64, 2, 2, 2, 2, 64>),
+                               grid, dim3(TPB), 0, stream, bf_ptr(dy),
+                               bf_ptr(x), dw.data_ptr<float>(), bf_ptr(zbuf),
+                               s, k_chunks);
+        else
+            hipLaunchKernelGGL((conv_wgrad_glds_kernel<64, 2, 2, 2, 2>), grid,
+                               dim3(TPB), 0, stream, bf_ptr(dy), bf_ptr(x),
+                               dw.data_ptr<float>(), bf_ptr(zbuf), s,
+                               k_chunks);
     } else if (big) {
         hipLaunchKernelGGL((conv_wgrad_kernel<64, 2, 2, 2, 2>), grid, dim3(TPB),
                            0, stream, bf_ptr(dy), bf_ptr(x),
