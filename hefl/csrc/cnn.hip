@@ -1547,7 +1547,7 @@ conv_wgrad_glds_k32_kernel(const unsigned short* __restrict__ dy,
             glds16(src, (char*)&Dys(buf)[0][0] + (wave + t * 4) * 1024);
         }
 #pragma unroll
-        for (int t = 0; t < BN * 8 / TPB; ++t) {  // Xs: 64 pix x BN/8 chunks
+        for (int t = 0; t < BKP * (BN / 8) / TPB; ++t) {  // Xs[pix][cc] chunks
             const int i = tid + t * TPB;
             const int pix = i / (BN / 8);
             const int cc = (i % (BN / 8)) * 8;
