@@ -3492,8 +3492,9 @@ torch::Tensor conv2d_wgrad(torch::Tensor dy, torch::Tensor x, int64_t stride,
         static torch::Tensor zbuf32;
         if (!zbuf32.defined() || zbuf32.device() != dy.device())
             zbuf32 = torch::zeros({8}, dy.options());
-        // 128-pixel K-steps measured +15-16% over 64 (wgrad was
-        // barrier-amortization bound); HEFL_K32B=0 reverts for A/B
+        // 128-pixel K-steps measured +3-5% over 64 after the staging fix
+        // (the first +15% probe was staging only half the Xs tile — a
+        // timing-only A/B hides correctness bugs); HEFL_K32B=0 for A/B
         static const bool bkp128 = [] {
             const char* e = getenv("HEFL_K32B");
             return !e || e[0] != '0';
@@ -3531,9 +3532,10 @@ torch::Tensor conv2d_wgrad(torch::Tensor dy, torch::Tensor x, int64_t stride,
         static torch::Tensor zbuf;
         if (!zbuf.defined() || zbuf.device() != dy.device())
             zbuf = torch::zeros({8}, dy.options());
+        // 64-pixel K-steps: +3-5% at every measured shape; HEFL_WGB=0 A/B
         static const bool wgb64 = [] {
             const char* e = getenv("HEFL_WGB");
-            return e && e[0] == '1';  // probe: 64-pixel K-steps
+            return !e || e[0] != '0';
         }();
         if (wgb64)
             hipLaunchKernelGGL((conv_wgrad_glds_kernel<64, 2, 2, 2, 2, 64>),
