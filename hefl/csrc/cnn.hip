@@ -3196,11 +3196,12 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
     // wins +34% on Kout<=32 layers (their BN=64 implicit-GEMM tile wasted
     // half its columns AND re-gathered 9x); the BN=64 and 8x16 variants
     // LOSE 10-45% to the glds pipeline (fewer resident blocks, conflictier
-    // B reads) — auto mode dispatches only the measured winner.
+    // B reads) — auto (1) dispatches only the measured winner; 2 forces any
+    // grid (tests); 3 probes BN=32 column-tiles for EVERY Kout.
     if (t3mode != 0 && s.R == 3 && s.S == 3 && s.stride == 1 && s.pad <= 1 &&
         s.C % 32 == 0 && s.Kout > 16 &&
-        (t3mode == 2 || s.Kout <= 32)) {
-        const int BN3 = s.Kout <= 32 ? 32 : 64;
+        (t3mode >= 2 || s.Kout <= 32)) {
+        const int BN3 = (t3mode == 3 || s.Kout <= 32) ? 32 : 64;
         const int kt = ceildiv(s.Kout, BN3);
         const int64_t min_tiles = t3mode == 2 ? 1 : 256;
         auto tiles = [&](int th, int tw) {
@@ -3208,7 +3209,7 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
         };
         int TH = 0;
         if (s.OH >= 12 && tiles(16, 16) >= min_tiles) TH = 16;
-        else if (t3mode == 2 && s.OH >= 6 && tiles(8, 16) >= min_tiles) TH = 8;
+        else if (t3mode >= 2 && s.OH >= 6 && tiles(8, 16) >= min_tiles) TH = 8;
         if (TH) {
             const int th_ = ceildiv(s.OH, TH), tw_ = ceildiv(s.OW, 16);
             dim3 grid((unsigned)(s.N * th_ * tw_), (unsigned)kt);
