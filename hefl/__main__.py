@@ -29,6 +29,10 @@ def main():
     ap.add_argument("--device", default=None)
     ap.add_argument("--plaintext", action="store_true",
                     help="disable HE (plaintext FedAvg)")
+    ap.add_argument("--augment", choices=["none", "hflip", "full"],
+                    default=None,
+                    help="training augmentation ('full' = the reference's "
+                         "shear 0.2 / zoom 0.2 / h-flip set)")
     ap.add_argument("--callbacks", action="store_true",
                     help="enable EarlyStopping/ReduceLROnPlateau per client")
     ap.add_argument("--json", action="store_true", help="JSON line output")
@@ -42,6 +46,8 @@ def main():
         cfg.fl.n_clients = args.clients
     if args.plaintext:
         cfg.fl.encrypted = False
+    if args.augment is not None:
+        cfg.fl.augment = args.augment
     device = args.device or ("cuda" if torch.cuda.is_available() else "cpu")
 
     init_distributed()
