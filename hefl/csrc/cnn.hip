@@ -2609,6 +2609,94 @@ __global__ void bn_partial_kernel(const unsigned short* __restrict__ x,
 }
 
 // one block per channel: 256 threads tree-reduce the nblk slab rows
+// bn_partial octet path + counter-gated LAST-BLOCK finalize: saves the
+// separate bn_finalize launch (~4.7 us each, 20 BN layers x fwd/bwd per
+// ResNet step). The finalize tail runs in ONE block after a fence; every
+// other block pays only fence + one atomic (cheap — unlike the pool
+// backward, whose EVERY block paid a barrier tail). counter is a
+// persistent zeroed int the tail resets for the next launch.
+__global__ void bn_partial_fused_kernel(const unsigned short* __restrict__ x,
+                                        float* __restrict__ slab, int64_t M,
+                                        int C, int rows_per_block,
+                                        float* __restrict__ mean,
+                                        float* __restrict__ invstd,
+                                        float* __restrict__ running_mean,
+                                        float* __restrict__ running_var,
+                                        float eps, float momentum,
+                                        int* __restrict__ counter) {
+    float* gsum = slab + (int64_t)blockIdx.x * 2 * C;
+    float* gsq = gsum + C;
+    __shared__ float red8[2][2048];
+    const int64_t r0 = (int64_t)blockIdx.x * rows_per_block;
+    const int64_t r1 = min(r0 + rows_per_block, M);
+    const int noct = C >> 3;
+    int lanes = (int)blockDim.x / noct;
+    lanes = lanes ? (1 << (31 - __clz(lanes))) : 1;
+    const int oct = threadIdx.x % noct;
+    const int rl = threadIdx.x / noct;
+    float acc[8] = {0.f}, acc2[8] = {0.f};
+    if (rl < lanes) {
+        for (int64_t r = r0 + rl; r < r1; r += lanes) {
+            u16x8 v8 = *reinterpret_cast<const u16x8*>(&x[r * C + oct * 8]);
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                float v = bf2f(v8[j]);
+                acc[j] += v;
+                acc2[j] += v * v;
+            }
+        }
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+        red8[0][threadIdx.x * 8 + j] = acc[j];
+        red8[1][threadIdx.x * 8 + j] = acc2[j];
+    }
+    __syncthreads();
+    for (int off = lanes >> 1; off >= 1; off >>= 1) {
+        if (rl < off) {
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                red8[0][threadIdx.x * 8 + j] +=
+                    red8[0][(threadIdx.x + off * noct) * 8 + j];
+                red8[1][threadIdx.x * 8 + j] +=
+                    red8[1][(threadIdx.x + off * noct) * 8 + j];
+            }
+        }
+        __syncthreads();
+    }
+    if (rl == 0) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            gsum[oct * 8 + j] = red8[0][threadIdx.x * 8 + j];
+            gsq[oct * 8 + j] = red8[1][threadIdx.x * 8 + j];
+        }
+    }
+    // last arriving block finalizes (slab rows are visible via the fence)
+    __shared__ int is_last;
+    __threadfence();
+    if (threadIdx.x == 0)
+        is_last = (atomicAdd(counter, 1) == (int)gridDim.x - 1) ? 1 : 0;
+    __syncthreads();
+    if (!is_last) return;
+    for (int c = threadIdx.x; c < C; c += blockDim.x) {
+        float s1 = 0.f, s2 = 0.f;
+        for (int b = 0; b < (int)gridDim.x; ++b) {
+            s1 += slab[(int64_t)b * 2 * C + c];
+            s2 += slab[(int64_t)b * 2 * C + C + c];
+        }
+        const float mu = s1 / (float)M;
+        const float var = fmaxf(s2 / (float)M - mu * mu, 0.f);
+        mean[c] = mu;
+        invstd[c] = rsqrtf(var + eps);
+        if (running_mean) {
+            float unb = var * ((float)M / (float)max(M - 1, (int64_t)1));
+            running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mu;
+            running_var[c] = (1.f - momentum) * running_var[c] + momentum * unb;
+        }
+    }
+    if (threadIdx.x == 0) *counter = 0;  // self-reset for the next launch
+}
+
 __global__ void bn_finalize_kernel(const float* __restrict__ slab, int nblk,
                                    float* __restrict__ mean,
                                    float* __restrict__ invstd,
@@ -3898,14 +3986,32 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
     int rpb = (int)std::max<int64_t>(64, (M + 127) / 128);
     int nblk = (int)((M + rpb - 1) / rpb);
     auto slab = torch::empty({nblk, 2, C}, f32);
-    hipLaunchKernelGGL(bn_partial_kernel, dim3(nblk), dim3(256), 0, stream,
-                       bf_ptr(x), slab.data_ptr<float>(), M, C, rpb);
-    hipLaunchKernelGGL(bn_finalize_kernel, dim3(C), dim3(256), 0,
-                       stream, slab.data_ptr<float>(), nblk,
-                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                       rmean.numel() ? rmean.data_ptr<float>() : nullptr,
-                       rvar.numel() ? rvar.data_ptr<float>() : nullptr, M, C,
-                       (float)eps, (float)momentum);
+    static const bool bn_fuse = [] {
+        const char* e = getenv("HEFL_BN_FUSE");
+        return e && e[0] == '1';  // probe: last-block finalize
+    }();
+    if (bn_fuse && (C & 7) == 0 && C <= 2048) {
+        static torch::Tensor ctr;
+        if (!ctr.defined() || ctr.device() != x.device())
+            ctr = torch::zeros({1}, x.options().dtype(torch::kInt32));
+        hipLaunchKernelGGL(bn_partial_fused_kernel, dim3(nblk), dim3(256), 0,
+                           stream, bf_ptr(x), slab.data_ptr<float>(), M, C,
+                           rpb, mean.data_ptr<float>(),
+                           invstd.data_ptr<float>(),
+                           rmean.numel() ? rmean.data_ptr<float>() : nullptr,
+                           rvar.numel() ? rvar.data_ptr<float>() : nullptr,
+                           (float)eps, (float)momentum,
+                           ctr.data_ptr<int>());
+    } else {
+        hipLaunchKernelGGL(bn_partial_kernel, dim3(nblk), dim3(256), 0, stream,
+                           bf_ptr(x), slab.data_ptr<float>(), M, C, rpb);
+        hipLaunchKernelGGL(bn_finalize_kernel, dim3(C), dim3(256), 0,
+                           stream, slab.data_ptr<float>(), nblk,
+                           mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                           rmean.numel() ? rmean.data_ptr<float>() : nullptr,
+                           rvar.numel() ? rvar.data_ptr<float>() : nullptr, M,
+                           C, (float)eps, (float)momentum);
+    }
     int64_t total = x.numel();
     int blocks = std::min<int64_t>(ceildiv(total, 256), 2048);
     hipLaunchKernelGGL(bn_apply_kernel, dim3(blocks), dim3(256), 0, stream,
