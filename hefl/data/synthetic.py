@@ -96,31 +96,39 @@ class SyntheticMedicalImages:
         """Per-sample zoom/shear/h-flip via border-clamped bilinear sampling
         (torch reference path; the GPU bf16 path fuses the same transform
         into the synth kernel)."""
-        import torch.nn.functional as F
-        n = tmpl.shape[0]
         self._aug_ctr = getattr(self, "_aug_ctr", 0) + 1
         g = torch.Generator(device="cpu").manual_seed(
             self.seed * 31 + self._aug_ctr)
-        zoom = 1 + zr * (2 * torch.rand(n, generator=g) - 1)
-        shear = sr * (2 * torch.rand(n, generator=g) - 1)
-        sf = torch.where(torch.rand(n, generator=g) < 0.5, -1.0, 1.0) \
-            if fl else torch.ones(n)
-        # inverse map (output -> source), about the image center; grid_sample
-        # normalized coords with align_corners=True match the (dim-1) scaling
-        H, W = self.H, self.W
-        hh = torch.arange(H, dtype=torch.float32)
-        ww = torch.arange(W, dtype=torch.float32)
-        cy, cx = 0.5 * (H - 1), 0.5 * (W - 1)
-        dy = (hh - cy).view(1, H, 1)
-        dx = (ww - cx).view(1, 1, W)
-        sy = cy + dy / zoom.view(-1, 1, 1) + torch.zeros(1, 1, W)
-        sx = (cx + dx * (sf / zoom).view(-1, 1, 1)
-              + shear.view(-1, 1, 1) * dy)
-        grid = torch.stack([2 * sx / (W - 1) - 1, 2 * sy / (H - 1) - 1],
-                           dim=-1).to(tmpl.device, tmpl.dtype)
-        out = F.grid_sample(tmpl.permute(0, 3, 1, 2), grid, mode="bilinear",
-                            padding_mode="border", align_corners=True)
-        return out.permute(0, 2, 3, 1)
+        return affine_sample(tmpl, zr, sr, fl, g)
+
+
+def affine_sample(x: torch.Tensor, zr: float, sr: float, fl: bool,
+                  g: torch.Generator) -> torch.Tensor:
+    """Per-sample random zoom/shear/h-flip of NHWC images via inverse-map
+    border-clamped bilinear sampling (the reference ImageDataGenerator
+    transform set, FLPyfhelin.py:80-86). Shared by the synthetic dataset's
+    CPU path and the file-backed dataset."""
+    import torch.nn.functional as F
+    n, H, W = x.shape[0], x.shape[1], x.shape[2]
+    zoom = 1 + zr * (2 * torch.rand(n, generator=g) - 1)
+    shear = sr * (2 * torch.rand(n, generator=g) - 1)
+    sf = torch.where(torch.rand(n, generator=g) < 0.5, -1.0, 1.0) \
+        if fl else torch.ones(n)
+    # inverse map (output -> source), about the image center; grid_sample
+    # normalized coords with align_corners=True match the (dim-1) scaling
+    hh = torch.arange(H, dtype=torch.float32)
+    ww = torch.arange(W, dtype=torch.float32)
+    cy, cx = 0.5 * (H - 1), 0.5 * (W - 1)
+    dy = (hh - cy).view(1, H, 1)
+    dx = (ww - cx).view(1, 1, W)
+    sy = cy + dy / zoom.view(-1, 1, 1) + torch.zeros(1, 1, W)
+    sx = (cx + dx * (sf / zoom).view(-1, 1, 1)
+          + shear.view(-1, 1, 1) * dy)
+    grid = torch.stack([2 * sx / (W - 1) - 1, 2 * sy / (H - 1) - 1],
+                       dim=-1).to(x.device, torch.float32)
+    out = F.grid_sample(x.float().permute(0, 3, 1, 2), grid, mode="bilinear",
+                        padding_mode="border", align_corners=True)
+    return out.permute(0, 2, 3, 1).to(x.dtype)
 
 
 def make_client_loader(ds: SyntheticMedicalImages, client: int, n_clients: int,
