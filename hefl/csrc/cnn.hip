@@ -2131,6 +2131,162 @@ __global__ void pool_relu_bias_bwd_scalar_kernel(
 
 
 // ---------------------------------------------------------------------------
+// Fused backward of the 2-layer dense head (Dense-relu -> Dense-logits) at
+// M <= 32: dW2, db2, dh1 (relu-gated), dW1, db1 and dx in ONE single-
+// workgroup launch. The six separate launches it replaces (2 gemm dgrads,
+// 2 gemm wgrads, relu+bias bwd, bias grad) ran at the ~5-7 us replay floor
+// each = ~34 us of the ~146 us config #2 step; every GEMM here has
+// K = 32 (the batch), so each 16x16 MFMA tile is a single instruction and
+// results stream straight to HBM with no accumulator carry.
+// ---------------------------------------------------------------------------
+
+__global__ void __launch_bounds__(TPB)
+dense_head2_bwd_kernel(const unsigned short* __restrict__ dl_g,  // [M, N2]
+                       const unsigned short* __restrict__ x,     // [M, K]
+                       const unsigned short* __restrict__ h1g,   // [M, N1]
+                       const unsigned short* __restrict__ w1,    // [N1, K]
+                       const unsigned short* __restrict__ w2,    // [N2, N1]
+                       unsigned short* __restrict__ dx,          // [M, K]
+                       float* __restrict__ dw1,                  // [N1, K]
+                       float* __restrict__ db1,                  // [N1]
+                       float* __restrict__ dw2,                  // [N2, N1]
+                       float* __restrict__ db2,                  // [N2]
+                       int M, int K, int N1, int N2) {
+    constexpr int MP = 32;   // padded batch rows (mfma K-depth)
+    constexpr int NP = 32;   // padded N2 (k-depth of the dh1 product)
+    extern __shared__ unsigned short xs[];        // [MP][K]
+    __shared__ unsigned short dl[MP][NP];         // dlogits, zero-padded
+    __shared__ unsigned short h1s[MP][128];       // post-relu activations
+    __shared__ unsigned short dh[MP][128];        // gated dh1
+    __shared__ unsigned short w2s[NP][128];       // w2, zero-padded rows
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+    const int half = lane >> 4, sub = lane & 15;
+    // ---- stage (zero-padding all tails) ----
+    for (int i = tid; i < MP * NP; i += TPB) {
+        const int m = i / NP, c = i % NP;
+        dl[m][c] = (m < M && c < N2) ? dl_g[m * N2 + c] : (unsigned short)0;
+    }
+    for (int i = tid; i < MP * N1; i += TPB) {
+        const int m = i / N1, c = i % N1;
+        h1s[m][c] = m < M ? h1g[m * N1 + c] : (unsigned short)0;
+    }
+    for (int i = tid; i < NP * N1; i += TPB) {
+        const int r = i / N1, c = i % N1;
+        w2s[r][c] = r < N2 ? w2[r * N1 + c] : (unsigned short)0;
+    }
+    for (int i = tid; i < MP * (K / 8); i += TPB) {
+        const int m = i / (K / 8), c8 = (i % (K / 8)) * 8;
+        u16x8 v = {};
+        if (m < M) v = *reinterpret_cast<const u16x8*>(&x[(int64_t)m * K + c8]);
+        *reinterpret_cast<u16x8*>(&xs[m * K + c8]) = v;
+    }
+    __syncthreads();
+    // ---- phase A: dh1 = relu-gate(dlogits @ w2), into LDS ----
+    {
+        const int ntile = (MP / 16) * (N1 / 16);
+        for (int t = wave; t < ntile; t += TPB / 64) {
+            const int rb = (t / (N1 / 16)) * 16;   // m rows
+            const int cb = (t % (N1 / 16)) * 16;   // n1 cols
+            bf16x8 a, b;
+#pragma unroll
+            for (int j = 0; j < 8; ++j)
+                a[j] = *reinterpret_cast<const bf16_t*>(
+                    &dl[rb + sub][half * 8 + j]);
+#pragma unroll
+            for (int j = 0; j < 8; ++j)
+                b[j] = *reinterpret_cast<const bf16_t*>(
+                    &w2s[half * 8 + j][cb + sub]);
+            f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+            acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int m = rb + half * 4 + r, c = cb + sub;
+                const unsigned short hv = h1s[m][c];
+                const bool act = (hv & 0x7fffu) != 0 && !(hv & 0x8000u);
+                dh[m][c] = act ? f2bf(acc[r]) : (unsigned short)0;
+            }
+        }
+    }
+    __syncthreads();
+    // ---- phase B: independent products, wave-strided tile list ----
+    const int t_dw2 = N1 / 16;                    // dW2 tiles (N2 fits 16)
+    const int t_dw1 = (N1 / 16) * (K / 16);
+    const int t_dx = (MP / 16) * (K / 16);
+    const int total = t_dw2 + t_dw1 + t_dx;
+    for (int t = wave; t < total; t += TPB / 64) {
+        bf16x8 a, b;
+        f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+        if (t < t_dw2) {
+            // dW2[n2, n1] = sum_m dl[m][n2] * h1[m][n1]
+            const int cb = t * 16;
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                a[j] = *reinterpret_cast<const bf16_t*>(
+                    &dl[half * 8 + j][sub]);
+                b[j] = *reinterpret_cast<const bf16_t*>(
+                    &h1s[half * 8 + j][cb + sub]);
+            }
+            acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int n2 = half * 4 + r;
+                if (n2 < N2) dw2[n2 * N1 + cb + sub] = acc[r];
+            }
+        } else if (t < t_dw2 + t_dw1) {
+            // dW1[n1, k] = sum_m dh[m][n1] * x[m][k]
+            const int tt = t - t_dw2;
+            const int rb = (tt / (K / 16)) * 16;
+            const int cb = (tt % (K / 16)) * 16;
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                a[j] = *reinterpret_cast<const bf16_t*>(
+                    &dh[half * 8 + j][rb + sub]);
+                b[j] = *reinterpret_cast<const bf16_t*>(
+                    &xs[(half * 8 + j) * K + cb + sub]);
+            }
+            acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+#pragma unroll
+            for (int r = 0; r < 4; ++r)
+                dw1[(int64_t)(rb + half * 4 + r) * K + cb + sub] = acc[r];
+        } else {
+            // dx[m, k] = sum_n1 dh[m][n1] * w1[n1][k]  (N1/32 k-steps)
+            const int tt = t - t_dw2 - t_dw1;
+            const int rb = (tt / (K / 16)) * 16;
+            const int cb = (tt % (K / 16)) * 16;
+            for (int kk = 0; kk < N1; kk += 32) {
+#pragma unroll
+                for (int j = 0; j < 8; ++j) {
+                    a[j] = *reinterpret_cast<const bf16_t*>(
+                        &dh[rb + sub][kk + half * 8 + j]);
+                    b[j] = *reinterpret_cast<const bf16_t*>(
+                        &w1[(int64_t)(kk + half * 8 + j) * K + cb + sub]);
+                }
+                acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc,
+                                                              0, 0, 0);
+            }
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int m = rb + half * 4 + r;
+                if (m < M) dx[(int64_t)m * K + cb + sub] = f2bf(acc[r]);
+            }
+        }
+    }
+    // ---- bias grads (tiny column sums from LDS) ----
+    if (tid < N1) {
+        float s = 0.f;
+        for (int m = 0; m < M; ++m) s += bf2f(dh[m][tid]);
+        db1[tid] = s;
+    } else if (tid >= 128 && tid - 128 < N2) {
+        const int c = tid - 128;
+        float s = 0.f;
+        for (int m = 0; m < M; ++m) s += bf2f(dl[m][c]);
+        db2[c] = s;
+    }
+}
+
+// ---------------------------------------------------------------------------
 // Fused softmax + categorical cross-entropy (mean), one wave per row.
 // logits bf16 [M, C]; probs f32 out; loss = sum(-log p[label]) / M.
 // ---------------------------------------------------------------------------
@@ -3874,6 +4030,32 @@ std::vector<torch::Tensor> pool_relu_bias_bwd(torch::Tensor dy,
                        db.data_ptr<float>(), M, K, rpb, (int)H, (int)W, OH,
                        OW, fdiv_make((unsigned)W), fdiv_make((unsigned)H));
     return {dym, db};
+}
+
+std::vector<torch::Tensor> dense_head2_bwd(torch::Tensor dlogits,
+                                           torch::Tensor x, torch::Tensor h1,
+                                           torch::Tensor w1,
+                                           torch::Tensor w2) {
+    CHECK_GPU(dlogits);
+    const int M = (int)dlogits.size(0), N2 = (int)dlogits.size(1);
+    const int N1 = (int)w1.size(0), K = (int)w1.size(1);
+    TORCH_CHECK(M <= 32 && N2 <= 16 && N1 <= 128 && N1 % 16 == 0 &&
+                K % 16 == 0 && K <= 1024,
+                "dense_head2_bwd shape envelope");
+    auto f32 = x.options().dtype(torch::kFloat32);
+    auto dx = torch::empty({M, (int64_t)K}, x.options());
+    auto dw1 = torch::empty({N1, (int64_t)K}, f32);
+    auto db1 = torch::empty({N1}, f32);
+    auto dw2 = torch::empty({N2, (int64_t)N1}, f32);
+    auto db2 = torch::empty({N2}, f32);
+    hipLaunchKernelGGL(dense_head2_bwd_kernel, dim3(1), dim3(TPB),
+                       32 * K * sizeof(unsigned short),
+                       at::cuda::getCurrentCUDAStream(), bf_ptr(dlogits),
+                       bf_ptr(x), bf_ptr(h1), bf_ptr(w1), bf_ptr(w2),
+                       bf_ptr_mut(dx), dw1.data_ptr<float>(),
+                       db1.data_ptr<float>(), dw2.data_ptr<float>(),
+                       db2.data_ptr<float>(), M, K, N1, N2);
+    return {dx, dw1, db1, dw2, db2};
 }
 
 std::vector<torch::Tensor> softmax_xent_fwd(torch::Tensor logits,

@@ -46,11 +46,24 @@ class _SeqCNN(nn.Module):
             din = u
         head.append(Dense(din, n_classes, relu=False, gen=gen))
         self.head = nn.ModuleList(head)
+        # single-hidden heads within the fused-backward envelope dispatch
+        # dense_head2 on GPU (one kernel instead of six at the replay floor)
+        self._head2_ok = (len(head) == 2 and head[0].relu
+                          and not head[1].relu and head[0].bias is not None
+                          and head[1].bias is not None
+                          and n_classes <= 16
+                          and head[0].weight.shape[0] <= 128
+                          and head[0].weight.shape[0] % 16 == 0
+                          and feat % 16 == 0 and feat <= 1024)
 
     def forward(self, x):
         for m in self.trunk:
             x = m(x)
         x = self.flatten(x)
+        if x.is_cuda and self._head2_ok and x.shape[0] <= 32:
+            from ..ops import functional as Fx
+            return Fx.dense_head2(x, self.head[0].weight, self.head[0].bias,
+                                  self.head[1].weight, self.head[1].bias)
         for m in self.head:
             x = m(x)
         return x

@@ -261,6 +261,35 @@ def linear(x, w, b=None, relu: bool = False):
     return _LinearFn.apply(x, w, b, relu)
 
 
+class _DenseHead2Fn(torch.autograd.Function):
+    """Dense(relu) -> Dense(logits) with a SINGLE-LAUNCH fused backward
+    (dW2/db2, relu-gated dh1, dW1/db1, dx in one workgroup) — the six
+    separate small-GEMM/bias launches it replaces each ran at the graph
+    replay floor and summed to ~23% of the config #2 step. GPU only; the
+    CPU path composes the plain linear ops."""
+
+    @staticmethod
+    def forward(ctx, x, w1, b1, w2, b2):
+        w1b, w2b = _gpu_dtype(w1), _gpu_dtype(w2)
+        xc = x.contiguous()
+        h1 = _C().linear_fwd(xc, w1b.contiguous(), b1.detach().float(), True)
+        logits = _C().linear_fwd(h1, w2b.contiguous(), b2.detach().float(),
+                                 False)
+        ctx.save_for_backward(xc, h1, w1b, w2b)
+        return logits
+
+    @staticmethod
+    def backward(ctx, dlogits):
+        x, h1, w1b, w2b = ctx.saved_tensors
+        dx, dw1, db1, dw2, db2 = _C().dense_head2_bwd(
+            dlogits.contiguous(), x, h1, w1b.contiguous(), w2b.contiguous())
+        return dx, dw1, db1, dw2, db2
+
+
+def dense_head2(x, w1, b1, w2, b2):
+    return _DenseHead2Fn.apply(x, w1, b1, w2, b2)
+
+
 # ---------------------------------------------------------------------------
 # Fused softmax + categorical cross-entropy (mean over batch) — the
 # reference's loss (FLPyfhelin.py:141).

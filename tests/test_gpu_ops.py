@@ -307,3 +307,35 @@ def test_conv_tile3_matches_glds(shape):
     os.environ.pop("HEFL_TILE3", None)
     assert torch.equal(y_tile, y_ref) or \
         (y_tile.float() - y_ref.float()).abs().max().item() < 1e-2
+
+
+def test_dense_head2_fused_matches_composed():
+    """Single-launch dense-head backward vs the composed linear ops on
+    identical bf16 inputs/weights (cnn2's exact head shape M=32/partial
+    M=16, K=800, N1=64, N2=10)."""
+    torch.manual_seed(7)
+    for M in (32, 16):
+        K, N1, N2 = 800, 64, 10
+        x = torch.randn(M, K, device="cuda", dtype=torch.bfloat16)
+        w1 = (torch.randn(N1, K, device="cuda") * 0.05)
+        b1 = torch.randn(N1, device="cuda") * 0.1
+        w2 = (torch.randn(N2, N1, device="cuda") * 0.1)
+        b2 = torch.randn(N2, device="cuda") * 0.1
+
+        def run(fused):
+            xd = x.clone().requires_grad_(True)
+            p = [torch.nn.Parameter(t.clone()) for t in (w1, b1, w2, b2)]
+            if fused:
+                y = Fx.dense_head2(xd, *p)
+            else:
+                y = Fx.linear(Fx.linear(xd, p[0], p[1], relu=True),
+                              p[2], p[3], relu=False)
+            y.float().pow(2).sum().backward()
+            return y, xd.grad, [q.grad for q in p]
+
+        yf, dxf, gf = run(True)
+        yc, dxc, gc = run(False)
+        _close(yf, yc, rel=1e-5)
+        _close(dxf, dxc, rel=1e-3)
+        for a, b in zip(gf, gc):
+            _close(a, b, rel=1e-3)
