@@ -46,9 +46,16 @@ class _SeqCNN(nn.Module):
             din = u
         head.append(Dense(din, n_classes, relu=False, gen=gen))
         self.head = nn.ModuleList(head)
-        # single-hidden heads within the fused-backward envelope dispatch
-        # dense_head2 on GPU (one kernel instead of six at the replay floor)
-        self._head2_ok = (len(head) == 2 and head[0].relu
+        # single-hidden heads within the fused-backward envelope CAN
+        # dispatch dense_head2 on GPU (one kernel instead of six at the
+        # replay floor) — measured SLOWER end-to-end (config #2 36.7 ->
+        # 32.6 rounds/s: the single-workgroup kernel serializes ~270 MFMAs
+        # + strided w1 reads on ONE CU, worse than six well-overlapped
+        # launches). Numerics-validated and kept behind HEFL_HEAD2=1 as the
+        # documented probe.
+        import os
+        self._head2_ok = (os.environ.get("HEFL_HEAD2", "0") == "1"
+                          and len(head) == 2 and head[0].relu
                           and not head[1].relu and head[0].bias is not None
                           and head[1].bias is not None
                           and n_classes <= 16
