@@ -339,3 +339,31 @@ def test_dense_head2_fused_matches_composed():
         _close(dxf, dxc, rel=1e-3)
         for a, b in zip(gf, gc):
             _close(a, b, rel=1e-3)
+
+
+@pytest.mark.parametrize("shape", [
+    # (N, H, W, C, K, pad): dgrad via the tiled kernel (rotated weights)
+    (4, 20, 20, 32, 32, 0),
+    (4, 21, 21, 32, 64, 1),
+    (2, 18, 18, 32, 96, 0),
+])
+def test_conv_tile3_dgrad_matches_glds(shape):
+    """Tiled dgrad (dy conv 180-rotated transposed weights, pad'=2-pad) vs
+    the implicit-GEMM dgrad on identical inputs."""
+    import os
+    import hefl
+    C_ = hefl.load_extension()
+    N, H, W, C, K, pad = shape
+    torch.manual_seed(1)
+    OH, OW = H + 2 * pad - 2, W + 2 * pad - 2
+    dy = torch.randn(N, OH, OW, K, device="cuda", dtype=torch.bfloat16)
+    w = (torch.randn(K, 3, 3, C, device="cuda") * 0.1).to(torch.bfloat16)
+    os.environ["HEFL_TILE3"] = "2"
+    try:
+        dx_tile = C_.conv2d_dgrad(dy, w, 1, H, W, pad)
+    finally:
+        os.environ["HEFL_TILE3"] = "0"
+    dx_ref = C_.conv2d_dgrad(dy, w, 1, H, W, pad)
+    os.environ.pop("HEFL_TILE3", None)
+    err = (dx_tile.float() - dx_ref.float()).abs().max().item()
+    assert err < 1e-2, err
