@@ -8,7 +8,6 @@ the reference gets this only implicitly from its scattered pickle files.
 """
 from __future__ import annotations
 
-import io
 import os
 from typing import Optional
 

@@ -18,7 +18,6 @@ required).
 from __future__ import annotations
 
 import io
-import pickle
 import struct
 from typing import Optional
 
