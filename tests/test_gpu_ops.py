@@ -365,5 +365,7 @@ def test_conv_tile3_dgrad_matches_glds(shape):
         os.environ["HEFL_TILE3"] = "0"
     dx_ref = C_.conv2d_dgrad(dy, w, 1, H, W, pad)
     os.environ.pop("HEFL_TILE3", None)
-    err = (dx_tile.float() - dx_ref.float()).abs().max().item()
-    assert err < 1e-2, err
+    # both paths accumulate f32 but in different orders and round to bf16,
+    # so one-ulp flips at |dx| >= 1.28 are expected — relative check like
+    # the rest of the suite (a real indexing bug is O(max), far outside it)
+    _close(dx_tile, dx_ref, rel=1e-2)
