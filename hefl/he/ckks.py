@@ -105,8 +105,17 @@ class GpuBackend:
         self.ratio_words = torch.from_numpy(
             np.array(words, dtype=np.int64)).to(device)
 
+    @staticmethod
+    def _fresh(x: torch.Tensor) -> torch.Tensor:
+        """Materialize a tensor the in-place NTT kernels may own: one copy,
+        not two — .contiguous() already copies non-contiguous slices, so
+        clone() only when it was a no-op (round-2 profile: the redundant
+        clone showed up as ~2.5% of config #5 in hipMemcpy DtoD)."""
+        c = x.contiguous()
+        return c.clone() if c.data_ptr() == x.data_ptr() else c
+
     def ntt(self, x: torch.Tensor, limb: int, inverse: bool = False) -> torch.Tensor:
-        out = x.contiguous().clone()
+        out = self._fresh(x)
         flat = out.reshape(-1, self.n)
         if inverse:
             self._C.intt_batch(flat, self.winv[limb], self.winv_shoup[limb],
@@ -128,7 +137,7 @@ class GpuBackend:
     def ntt_all(self, x: torch.Tensor, inverse: bool = False) -> torch.Tensor:
         """x [..., L, n] -> NTT per limb in ONE fused launch set."""
         L = x.shape[-2]
-        out = x.contiguous().clone()
+        out = self._fresh(x)
         flat = out.reshape(-1, L, self.n)
         if inverse:
             self._C.intt_limbs(flat, self.winv, self.winv_shoup, self.qs,
@@ -414,7 +423,7 @@ class CKKSContext:
             # fast path (the FedAvg pipeline decrypts at level 1): center on
             # device, f64 is exact for plaintext magnitudes << 2^53
             q = self._q(0)
-            c = self.backend.ntt(data[..., 0, :].contiguous(), 0, inverse=True)
+            c = self.backend.ntt(data[..., 0, :], 0, inverse=True)
             cent = torch.where(c > q // 2, c - q, c)
             if self.device.type == "cuda":
                 return self.encoder.decode_torch(cent, pt.scale, k)
@@ -424,11 +433,11 @@ class CKKSContext:
         # centered limb-0 residue IS the integer value — checked against
         # limb 1 on a sample, falling back to exact big-int CRT on mismatch.
         q0 = self._q(0)
-        c0 = self.backend.ntt(data[..., 0, :].contiguous(), 0, inverse=True)
+        c0 = self.backend.ntt(data[..., 0, :], 0, inverse=True)
         cent0 = torch.where(c0 > q0 // 2, c0 - q0, c0)
         if cent0.abs().max().item() < q0 // 4:
             q1 = self._q(1)
-            c1 = self.backend.ntt(data[..., 1, :].contiguous(), 1, inverse=True)
+            c1 = self.backend.ntt(data[..., 1, :], 1, inverse=True)
             cent1 = torch.where(c1 > q1 // 2, c1 - q1, c1)
             if torch.equal(torch.remainder(cent0, q1), torch.remainder(cent1, q1)):
                 if self.device.type == "cuda":
@@ -437,7 +446,7 @@ class CKKSContext:
         # exact big-int CRT path (plaintext too large for limb-0 shortcut)
         coeff_limbs = []
         for i in range(nlimbs):
-            c = self.backend.ntt(data[..., i, :].contiguous(), i, inverse=True)
+            c = self.backend.ntt(data[..., i, :], i, inverse=True)
             coeff_limbs.append(c.cpu().numpy().astype(object))
         qs = [self._q(i) for i in range(nlimbs)]
         Q = math.prod(qs)
@@ -553,7 +562,7 @@ class CKKSContext:
             # launch, one fused NTT set, one strided-slice modsub (no
             # materialized slice), one scalar-limbs multiply — zero
             # at::native kernels on this path
-            cl = self.backend.ntt(data[..., last, :].contiguous(), last,
+            cl = self.backend.ntt(data[..., last, :], last,
                                   inverse=True)
             r = self.backend.bcast_center_mod(cl, qL, last)
             r_ntt = self.backend.ntt_all(r)
@@ -561,7 +570,7 @@ class CKKSContext:
             inv = [pow(qL % self._q(i), -1, self._q(i)) for i in range(last)]
             return self.backend.modmul_scalar_limbs(diff, inv)
         # coefficient-domain last limb, centered for round-to-nearest
-        cl = self.backend.ntt(data[..., last, :].contiguous(), last,
+        cl = self.backend.ntt(data[..., last, :], last,
                               inverse=True)
         cl_c = torch.where(cl > qL_half, cl - qL, cl)
         out = data[..., :last, :].clone()
@@ -682,7 +691,7 @@ class CKKSContext:
             inv = [pow(P % self._q(i), -1, self._q(i)) for i in range(L)]
             outs = []
             for acc in (acc0, acc1):
-                cl = be.ntt(acc[..., L, :].contiguous(), L, inverse=True)
+                cl = be.ntt(acc[..., L, :], L, inverse=True)
                 r_ntt = be.ntt_all(be.bcast_center_mod(cl, P, L))
                 diff = be.modsub_limbs(acc, r_ntt, L)
                 outs.append(be.modmul_scalar_limbs(diff, inv))
