@@ -35,6 +35,12 @@ def main():
                          "shear 0.2 / zoom 0.2 / h-flip set)")
     ap.add_argument("--callbacks", action="store_true",
                     help="enable EarlyStopping/ReduceLROnPlateau per client")
+    ap.add_argument("--checkpoint", default=None, metavar="PATH",
+                    help="save round state (model+optimizer+public HE "
+                         "material; sk to PATH.private) after every round "
+                         "and RESUME from PATH if it exists — the round-"
+                         "granularity recovery the reference gets only "
+                         "implicitly from its pickle files (SURVEY.md §5)")
     ap.add_argument("--json", action="store_true", help="JSON line output")
     args = ap.parse_args()
 
@@ -69,12 +75,32 @@ def main():
 
     from hefl.fl.sequential import SequentialFL
     fl = SequentialFL(cfg, device=device, verbose=True)
+    start_round = 0
+    if args.checkpoint:
+        import os
+
+        from hefl.fl.checkpoint import load_round_state, save_round_state
+        if os.path.exists(args.checkpoint):
+            # resume: restore the GLOBAL model + round counter (clients
+            # re-derive their state from the global weights each round)
+            ref_client = fl.clients[0]
+            start_round, _ = load_round_state(args.checkpoint,
+                                              fl.global_model,
+                                              ref_client.opt)
+            print(f"resumed from {args.checkpoint} at round {start_round}")
     reports = []
-    for r in range(args.rounds):
+    for r in range(start_round, args.rounds):
         rep = fl.run_round(epochs=args.epochs, use_callbacks=args.callbacks)
         reports.append(rep)
         print(f"round {r}: metrics={rep.metrics} round_s={rep.round_seconds:.2f}")
+        if args.checkpoint:
+            save_round_state(args.checkpoint, fl.global_model,
+                             fl.clients[0].opt, r + 1)
     total = time.time() - t_all
+
+    if not reports:  # resumed past the requested round count
+        print("nothing to do: checkpoint already at round", start_round)
+        return
 
     last = reports[-1]
     if args.json:

@@ -84,3 +84,27 @@ def test_bench_two_rank_driver_invocation():
     assert d["n_gpus"] == 2 and d["scaling"] == "weak"
     assert d["config"]["global_batch"] == 64
     assert d["config"]["parallelism"].startswith("fl-dp2")
+
+
+def test_cli_checkpoint_resume(tmp_path):
+    """--checkpoint saves per-round state and resumes past completed rounds
+    (the round-granularity recovery story)."""
+    import subprocess
+    import sys
+    ck = str(tmp_path / "round.pt")
+    base = [sys.executable, "-m", "hefl", "--preset", "config1", "--rounds",
+            "1", "--epochs", "1", "--clients", "2", "--checkpoint", ck,
+            "--json"]
+    r1 = subprocess.run(base, capture_output=True, text=True, timeout=240)
+    assert r1.returncode == 0, r1.stderr[-800:]
+    import os
+    assert os.path.exists(ck)
+    # second invocation with the same target round count resumes and exits
+    r2 = subprocess.run(base, capture_output=True, text=True, timeout=240)
+    assert r2.returncode == 0, r2.stderr[-800:]
+    assert "resumed from" in r2.stdout and "nothing to do" in r2.stdout
+    # raising --rounds continues from the checkpoint
+    r3 = subprocess.run(base[:6] + ["2"] + base[7:], capture_output=True,
+                        text=True, timeout=240)
+    assert r3.returncode == 0, r3.stderr[-800:]
+    assert "round 1:" in r3.stdout and "round 0:" not in r3.stdout
