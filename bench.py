@@ -112,6 +112,7 @@ def main() -> None:
                 "local_epochs": epochs,
                 "samples_per_client": cfg.fl.samples_per_client,
                 "samples_per_sec_per_client": samples_per_sec_per_client,
+                "augment": cfg.fl.augment,
                 "he": {"scheme": "CKKS", "m": cfg.he.m,
                        "q_bits": list(cfg.he.q_bits),
                        "scale_bits": cfg.he.scale_bits,
