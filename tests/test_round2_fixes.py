@@ -287,3 +287,28 @@ def test_four_rank_unseeded_bucketed_fedavg():
     from conftest import free_port
     for rank, err in _run(_worker_four, 4, free_port()):
         assert err < 1e-3, (rank, err)
+
+
+def test_round_checkpoint_custom_private_path(tmp_path):
+    """save/load_round_state honor an explicit private-key artifact path
+    (the decrypting party may keep sk on separate storage)."""
+    from hefl.fl.checkpoint import load_round_state, save_round_state
+    from hefl.models import build_model
+    from hefl.ops.adam import FusedAdam
+    he = Pyfhel()
+    he.contextGen(m=64, scale_bits=30, q_bits=(50, 30), seed=8)
+    he.keyGen()
+    ct = he.encryptFrac(2.5)
+    model = build_model(ModelConfig("cnn2", (28, 28, 1), 10))
+    opt = FusedAdam(model.parameters(), lr=1e-3)
+    rp = str(tmp_path / "round.pt")
+    sp = str(tmp_path / "keys" / "sk.pt")
+    os.makedirs(os.path.dirname(sp))
+    save_round_state(rp, model, opt, 1, he=he, private_path=sp)
+    assert os.path.exists(sp) and not os.path.exists(rp + ".private")
+    he2 = Pyfhel()
+    m2 = build_model(ModelConfig("cnn2", (28, 28, 1), 10), seed=2)
+    o2 = FusedAdam(m2.parameters(), lr=1e-3)
+    rnd, _ = load_round_state(rp, m2, o2, he=he2, private_path=sp)
+    assert rnd == 1
+    assert abs(he2.decryptFrac(ct) - 2.5) < 1e-4
