@@ -3813,7 +3813,11 @@ torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
     const float* bias = b.numel() ? b.data_ptr<float>() : nullptr;
     auto y = torch::empty({M, N}, x.options());
     if (M <= 64) {
-        int kc = std::max(256, K / 4);
+        // slab count targets >= 256 blocks: K/4 left the cnn4 flatten
+        // layer (K = 18432) at 64 blocks / 25% of the CUs (~440 GB/s)
+        const int gmn = ceildiv(M, 16) * ceildiv(N, 16);
+        const int target = std::max(4, 256 / std::max(gmn, 1));
+        int kc = std::max(256, ceildiv(K, target));
         kc = ((kc + 31) / 32) * 32;
         if (K <= 1024) kc = ((K + 31) / 32) * 32;  // one slab: direct write
         const int slabs = ceildiv(K, kc);
