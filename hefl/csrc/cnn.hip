@@ -4169,7 +4169,7 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
     auto invstd = torch::empty({C}, f32);
     auto y = torch::empty_like(x);
     auto stream = at::cuda::getCurrentCUDAStream();
-    int rpb = (int)std::max<int64_t>(64, (M + 127) / 128);
+    int rpb = (int)std::max<int64_t>(64, (M + 511) / 512);  // fill 2 blocks/CU
     int nblk = (int)((M + rpb - 1) / rpb);
     auto slab = torch::empty({nblk, 2, C}, f32);
     static const bool bn_fuse = [] {
@@ -4239,7 +4239,7 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
     auto dbeta = torch::empty({C}, f32);
     auto dx = torch::empty_like(x);
     auto stream = at::cuda::getCurrentCUDAStream();
-    int rpb = (int)std::max<int64_t>(64, (M + 127) / 128);
+    int rpb = (int)std::max<int64_t>(64, (M + 511) / 512);  // fill 2 blocks/CU
     int nblk = (int)((M + rpb - 1) / rpb);
     auto slab = torch::empty({nblk, 2, C}, f32);
     hipLaunchKernelGGL(bn_bwd_partial_kernel, dim3(nblk), dim3(256), 0, stream,
