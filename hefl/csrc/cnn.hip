@@ -3713,7 +3713,7 @@ torch::Tensor conv2d_wgrad(torch::Tensor dy, torch::Tensor x, int64_t stride,
     if (small_fast) {
         auto dw = torch::empty({s.Kout, R, S, s.C},
                                x.options().dtype(torch::kFloat32));
-        int kc = std::max(1, std::min(ceildiv(KK, 256), 512));
+        int kc = std::max(1, std::min(ceildiv(KK, 64), 512));  // fill: KK/256 left cnn2 conv1 wgrad at 84 blocks
         if (kc == 1) {
             hipLaunchKernelGGL(conv_wgrad_small_kernel, dim3(1, 1, 1),
                                dim3(TPB), 0, stream, bf_ptr(dy), bf_ptr(x),
