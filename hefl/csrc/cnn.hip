@@ -3713,7 +3713,10 @@ torch::Tensor conv2d_wgrad(torch::Tensor dy, torch::Tensor x, int64_t stride,
     if (small_fast) {
         auto dw = torch::empty({s.Kout, R, S, s.C},
                                x.options().dtype(torch::kFloat32));
-        int kc = std::max(1, std::min(ceildiv(KK, 64), 512));  // fill: KK/256 left cnn2 conv1 wgrad at 84 blocks
+        // measured: KK/64 slabs (338 blocks) LOST to KK/256 (84) — the
+        // slab reduce outgrows the parallel gain; this path is latency-
+        // not occupancy-bound
+        int kc = std::max(1, std::min(ceildiv(KK, 256), 512));
         if (kc == 1) {
             hipLaunchKernelGGL(conv_wgrad_small_kernel, dim3(1, 1, 1),
                                dim3(TPB), 0, stream, bf_ptr(dy), bf_ptr(x),
