@@ -107,5 +107,5 @@ def test_training_reduces_loss():
         opt.zero_grad()
         loss.backward()
         opt.step()
-        losses.append(float(loss))
+        losses.append(float(loss.detach()))
     assert losses[-1] < losses[0] * 0.5, losses[:3] + losses[-3:]
