@@ -23,11 +23,25 @@
 //   conv_wgrad_small_kernel        Kout<=64 & RSC<=16 first-layer wgrad:
 //                                  waves split over pixels, LDS reduce
 //   sum_slabs_f32_kernel           lane-parallel split-K slab reduction
-//   maxpool_fwd_kernel (2x2)       pool fwd + argmax byte
+//   xcd_remap                      consecutive conv M-tiles -> one XCD's L2
+//   conv_fwd_glds64_kernel         BK=64 two-buffer fwd (measured slower at
+//                                  these short-K shapes; HEFL_GLDS64 A/B)
+//   conv_fwd_tile3_kernel          direct tiled 3x3 s1 conv: input tile
+//                                  staged once, all nine taps from LDS
+//                                  (~144 MFMAs per barrier); also drives
+//                                  dgrad via rot180_transpose_w_kernel
+//   conv_wgrad_glds_k32_kernel     Kout<=32 wgrad tile, 128-pixel K-steps
+//   pad_channels_kernel            C%8!=0 stem -> glds MFMA path
+//   maxpool_fwd_kernel (2x2) / maxpool_fwd_oct_kernel
+//                                  pool fwd + argmax byte (octet: 16-B
+//                                  loads + packed 8-B argmax store)
 //   maxpool2x2_bwd_gather_kernel   gather-form pool bwd (no scatter/fill)
-//   pool_relu_gate / pool_relu_bias_bwd_kernel
-//                                  FUSED conv->relu->pool backward: pool
-//                                  gather + ReLU mask + bias grad, bf16x8
+//   pool_relu_bias_bwd2_kernel     FUSED conv->relu->pool backward: one
+//                                  thread per 2x2 block x 8 channels,
+//                                  ReLU gate on the POOLED output
+//   pool_relu_bias_bwd_scalar_kernel  K%8!=0 fallback (LeNet), lane-reduce
+//   dense_head2_bwd_kernel         single-launch 2-layer head backward
+//                                  (measured slower end-to-end; HEFL_HEAD2)
 //   relu_bias_bwd_kernel           fused ReLU-mask + bias grad (fc layers)
 //   linear_splitk_kernel           skinny-M wave GEMM, direct-from-global,
 //                                  unrolled wave-uniform fast path
@@ -41,8 +55,12 @@
 //   pack_mt/unpack_mt_kernel       FedAvg weight vector pack/unpack
 //   bn_partial/bn_finalize/bn_apply/bn_bwd_partial/bn_bwd_finalize/bn_dx
 //                                  two-stage BatchNorm, bf16x8 octet paths,
-//                                  ReLU gate fused into backward
-//   synth_batch_kernel             one-pass synthetic data generation
+//                                  ReLU gate fused into backward;
+//                                  bn_partial_fused_kernel = last-block-
+//                                  finalize probe (HEFL_BN_FUSE, -15%)
+//   synth_batch_kernel<AUG>        one-pass synthetic data generation,
+//                                  AUG = in-kernel zoom/shear/flip via
+//                                  inverse-affine template sampling
 //   maxpool_gen_fwd/bwd_kernel     generic k/s/p pooling (ResNet stem)
 //   avgpool_global_fwd/bwd_kernel  global average pool
 //   add_relu_kernel                residual add + ReLU (bf16x8)
