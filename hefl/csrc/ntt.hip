@@ -15,6 +15,15 @@
 //                                            levels per barrier)
 //   ntt/intt_global_radix4_ml_kernel         fused radix-4 global stage pair
 //   modmul/modadd/modsub/modmul_scalar       Barrett/Shoup pointwise ops
+//   modadd_limbs/modadd3_limbs/modsub_limbs  fused per-limb pointwise glue
+//                                            (limb = (i/n) % L; modsub takes
+//                                            a strided source for rescale)
+//   bcast_center_mod_kernel                  [.., n] -> [.., L, n] center +
+//                                            per-limb Barrett reduce
+//   ct_mul_kernel                            ct x ct tensor product
+//                                            (d0, d1, d2 in one launch)
+//   ks_inner_kernel                          whole key-switch digit inner
+//                                            product, both accumulators
 //   modreduce_ kernel                        lazy-sum -> [0,q) (post
 //                                            all-reduce), branchless
 //   cbd21_kernel                             centered-binomial noise from
