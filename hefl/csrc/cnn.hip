@@ -1200,13 +1200,19 @@ conv_dgrad_glds_kernel(const unsigned short* __restrict__ dy,
     }
 }
 
-// f32 accumulator -> bf16 (no bias/relu): split-K epilogue for dgrad
-__global__ void cast_f32_bf16_kernel(const float* __restrict__ src,
+// f32 accumulator -> bf16 (no bias/relu): split-K epilogue for dgrad.
+// clear=1: consume-and-clear — zero each element after reading so a pooled
+// accumulator (acc_pool below) is zero again for its next user without a
+// separate fill launch (the FillFunctor fills were 11.5% of config #2
+// kernel time, profiles/r02_config2_final_kernel_stats.csv).
+__global__ void cast_f32_bf16_kernel(float* __restrict__ src,
                                      unsigned short* __restrict__ dst,
-                                     int64_t total) {
+                                     int64_t total, int clear) {
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
-         i += (int64_t)gridDim.x * blockDim.x)
+         i += (int64_t)gridDim.x * blockDim.x) {
         dst[i] = f2bf(src[i]);
+        if (clear) src[i] = 0.f;
+    }
 }
 
 // Zero-pad the channel (last) dim C -> C8: routes C % 8 != 0 stem convs
@@ -2552,15 +2558,19 @@ __global__ void linear_splitk_kernel(const unsigned short* __restrict__ x,
     }
 }
 
-__global__ void linear_epilogue_kernel(const float* __restrict__ y32,
+// clear=1: consume-and-clear of a pooled split-K accumulator (see
+// cast_f32_bf16_kernel / acc_pool).
+__global__ void linear_epilogue_kernel(float* __restrict__ y32,
                                        const float* __restrict__ bias,
                                        unsigned short* __restrict__ y,
                                        int64_t total, int N, int relu,
-                                       int slabs) {
+                                       int slabs, int clear) {
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
          i += (int64_t)gridDim.x * blockDim.x) {
         float v = bias ? bias[i % N] : 0.f;
         for (int sl = 0; sl < slabs; ++sl) v += y32[sl * total + i];
+        if (clear)
+            for (int sl = 0; sl < slabs; ++sl) y32[sl * total + i] = 0.f;
         if (relu) v = v > 0.f ? v : 0.f;
         y[i] = f2bf(v);
     }
@@ -3430,6 +3440,36 @@ unsigned short* bf_ptr_mut(torch::Tensor& t) {
     return reinterpret_cast<unsigned short*>(t.data_ptr<at::BFloat16>());
 }
 
+// Persistent fp32 split-K accumulator pool. Invariant: every element is
+// zero BETWEEN ops on the compute stream — each producer atomically
+// accumulates into it and its consumer kernel (linear_epilogue /
+// cast_f32_bf16 with clear=1) zeroes what it read, the same
+// consume-and-clear idiom fused_adam_mt_kernel uses for grads. This
+// replaces a per-call torch::zeros whose fill launches were 11.5% of
+// config #2 kernel time (4 us launch-bound fills,
+// profiles/r02_config2_final_kernel_stats.csv). Valid because the engine
+// runs all conv/pool work on one stream per rank, and each host wrapper
+// enqueues accumulate + consume before returning. Growth while a hipGraph
+// capture is active would allocate from the capture mempool (whose blocks
+// die with the graph), so it falls back to a per-call zeros tensor then —
+// eager warmup before capture normally sizes the pool first.
+float* acc_pool(int64_t n, const torch::TensorOptions& opts,
+                hipStream_t stream, torch::Tensor& holder) {
+    static torch::Tensor pool;
+    auto f32 = opts.dtype(torch::kFloat32);
+    if (!pool.defined() || pool.device() != f32.device_opt().value() ||
+        pool.numel() < n) {
+        hipStreamCaptureStatus st = hipStreamCaptureStatusNone;
+        (void)hipStreamIsCapturing(stream, &st);
+        if (st != hipStreamCaptureStatusNone) {
+            holder = torch::zeros({n}, f32);
+            return holder.data_ptr<float>();
+        }
+        pool = torch::zeros({n}, f32);
+    }
+    return pool.data_ptr<float>();
+}
+
 }  // namespace
 
 // ---------------------------------------------------------------------------
@@ -3513,10 +3553,8 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
         dim3 grid(ceildiv(M, BM), ceildiv(s.Kout, BN), k_chunks);
         torch::Tensor y32;
         float* y32p = nullptr;
-        if (k_chunks > 1) {
-            y32 = torch::zeros({M, s.Kout}, x.options().dtype(torch::kFloat32));
-            y32p = y32.data_ptr<float>();
-        }
+        if (k_chunks > 1)
+            y32p = acc_pool((int64_t)M * s.Kout, x.options(), stream, y32);
         #define LAUNCH_F64(BM_, BN_, WM_, WN_, FM_, FN_, P0)                  \
             hipLaunchKernelGGL((conv_fwd_glds64_kernel<BM_, BN_, WM_, WN_,    \
                                                        FM_, FN_, P0>),       \
@@ -3536,7 +3574,7 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
             hipLaunchKernelGGL(linear_epilogue_kernel,
                                dim3((int)std::min<int64_t>(ceildiv(total, 256), 2048)),
                                dim3(256), 0, stream, y32p, bias,
-                               bf_ptr_mut(y), total, s.Kout, relu ? 1 : 0, 1);
+                               bf_ptr_mut(y), total, s.Kout, relu ? 1 : 0, 1, 1);
         }
         return y;
     }
@@ -3553,17 +3591,18 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
             k_chunks = std::max(1, std::min(ksteps / 4, 512 / tiles));
         dim3 grid(ceildiv(M, 64), ceildiv(s.Kout, 64), k_chunks);
         if (k_chunks > 1) {
-            auto y32 = torch::zeros({M, s.Kout},
-                                    x.options().dtype(torch::kFloat32));
+            torch::Tensor y32h;
+            float* y32p = acc_pool((int64_t)M * s.Kout, x.options(), stream,
+                                   y32h);
             hipLaunchKernelGGL((conv_fwd_glds_kernel<64, 64, 2, 2, 2, 2, 32>),
                                grid, dim3(TPB), 0, stream, bf_ptr(x), bf_ptr(w),
-                               bias, bf_ptr_mut(y), y32.data_ptr<float>(),
+                               bias, bf_ptr_mut(y), y32p,
                                bf_ptr(zbuf), s, relu ? 1 : 0, k_chunks);
             int64_t total = (int64_t)M * s.Kout;
             hipLaunchKernelGGL(linear_epilogue_kernel,
                                dim3((int)std::min<int64_t>(ceildiv(total, 256), 2048)),
-                               dim3(256), 0, stream, y32.data_ptr<float>(), bias,
-                               bf_ptr_mut(y), total, s.Kout, relu ? 1 : 0, 1);
+                               dim3(256), 0, stream, y32p, bias,
+                               bf_ptr_mut(y), total, s.Kout, relu ? 1 : 0, 1, 1);
         } else {
             hipLaunchKernelGGL((conv_fwd_glds_kernel<64, 64, 2, 2, 2, 2, 32>),
                                grid, dim3(TPB), 0, stream, bf_ptr(x), bf_ptr(w),
@@ -3651,10 +3690,8 @@ torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor w, int64_t stride,
         dim3 grid(ceildiv(M, 64), ceildiv(C, 64), k_chunks);
         torch::Tensor dx32;
         float* dx32p = nullptr;
-        if (k_chunks > 1) {
-            dx32 = torch::zeros({M, C}, dy.options().dtype(torch::kFloat32));
-            dx32p = dx32.data_ptr<float>();
-        }
+        if (k_chunks > 1)
+            dx32p = acc_pool((int64_t)M * C, dy.options(), stream, dx32);
         if (stride == 1)
             hipLaunchKernelGGL((conv_dgrad_glds_kernel<64, 64, 2, 2, 2, 2, true>),
                                grid, dim3(TPB), 0, stream, bf_ptr(dy),
@@ -3670,7 +3707,7 @@ torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor w, int64_t stride,
             hipLaunchKernelGGL(cast_f32_bf16_kernel,
                                dim3((int)std::min<int64_t>(ceildiv(total, 256), 2048)),
                                dim3(256), 0, stream, dx32p, bf_ptr_mut(dx),
-                               total);
+                               total, 1);
         }
     } else if (false) {  // 128-tile: measured neutral-to-worse (barrier-bound)
         dim3 grid(ceildiv(M, 128), ceildiv(C, 64));
@@ -3707,7 +3744,12 @@ torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor w, int64_t stride,
 }
 
 torch::Tensor conv2d_wgrad(torch::Tensor dy, torch::Tensor x, int64_t stride,
-                           int64_t R, int64_t S, int64_t pad) {
+                           int64_t R, int64_t S, int64_t pad,
+                           bool zero_init) {
+    // zero_init=false: epoch-graph capture contract (ops/functional.py
+    // _graph_no_zero) — the returned grad is stolen into p.grad and the
+    // multi-tensor Adam clears it in-graph each step, so the split-K
+    // atomics land on an already-zero buffer without a fill launch.
     CHECK_GPU(dy);
     TORCH_CHECK(dy.is_contiguous() && x.is_contiguous());
     ConvShape s;
@@ -3768,10 +3810,11 @@ torch::Tensor conv2d_wgrad(torch::Tensor dy, torch::Tensor x, int64_t stride,
         int tiles32 = ceildiv(s.Kout, 32) * ceildiv(NN, 64);
         int kc = std::max(1, std::min(ceildiv(KK, 128),
                                       512 / std::max(tiles32, 1)));
-        auto dw32 = kc > 1 ? torch::zeros({s.Kout, R, S, s.C},
-                                          x.options().dtype(torch::kFloat32))
-                           : torch::empty({s.Kout, R, S, s.C},
-                                          x.options().dtype(torch::kFloat32));
+        auto dw32 = (kc > 1 && zero_init)
+                        ? torch::zeros({s.Kout, R, S, s.C},
+                                       x.options().dtype(torch::kFloat32))
+                        : torch::empty({s.Kout, R, S, s.C},
+                                       x.options().dtype(torch::kFloat32));
         dim3 g32(ceildiv(s.Kout, 32), ceildiv(NN, 64), kc);
         if (bkp128)
             hipLaunchKernelGGL((conv_wgrad_glds_k32_kernel<64, 2, 2, 1, 2, 128>),
@@ -3788,7 +3831,7 @@ torch::Tensor conv2d_wgrad(torch::Tensor dy, torch::Tensor x, int64_t stride,
     // measured: slab-rows + reduce LOSES to fp32 atomics here (CDNA4 L2
     // atomics absorb the z-chunk contention; the slab variant paid extra
     // write+read traffic) — profiles/r02_bench_conv logs
-    auto dw = k_chunks > 1
+    auto dw = (k_chunks > 1 && zero_init)
                   ? torch::zeros({s.Kout, R, S, s.C},
                                  x.options().dtype(torch::kFloat32))
                   : torch::empty({s.Kout, R, S, s.C},
@@ -3859,7 +3902,7 @@ torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
         hipLaunchKernelGGL(linear_epilogue_kernel,
                            dim3((int)std::min<int64_t>(ceildiv(total, 256), 2048)),
                            dim3(256), 0, stream, y32.data_ptr<float>(), bias,
-                           bf_ptr_mut(y), total, N, relu ? 1 : 0, slabs);
+                           bf_ptr_mut(y), total, N, relu ? 1 : 0, slabs, 0);
         return y;
     }
     dim3 grid(ceildiv(M, BM), ceildiv(N, BN));
@@ -3991,7 +4034,8 @@ torch::Tensor maxpool2x2_bwd(torch::Tensor dy, torch::Tensor idx, int64_t H,
     return dx;
 }
 
-std::vector<torch::Tensor> relu_bias_bwd(torch::Tensor dy, torch::Tensor y) {
+std::vector<torch::Tensor> relu_bias_bwd(torch::Tensor dy, torch::Tensor y,
+                                         bool zero_init) {
     CHECK_GPU(dy);
     auto dyc = dy.contiguous();
     const int K = (int)dyc.size(-1);
@@ -3999,7 +4043,7 @@ std::vector<torch::Tensor> relu_bias_bwd(torch::Tensor dy, torch::Tensor y) {
     auto dym = torch::empty_like(dyc);
     int rpb = (int)std::max<int64_t>(64, (M + 511) / 512);
     int nblk = (int)((M + rpb - 1) / rpb);
-    auto db = nblk == 1
+    auto db = (nblk == 1 || !zero_init)
                   ? torch::empty({K}, dyc.options().dtype(torch::kFloat32))
                   : torch::zeros({K}, dyc.options().dtype(torch::kFloat32));
     hipLaunchKernelGGL(relu_bias_bwd_kernel, dim3(nblk), dim3(256), 0,
@@ -4011,7 +4055,7 @@ std::vector<torch::Tensor> relu_bias_bwd(torch::Tensor dy, torch::Tensor y) {
 std::vector<torch::Tensor> pool_relu_bias_bwd(torch::Tensor dy,
                                               torch::Tensor idx,
                                               torch::Tensor p, int64_t H,
-                                              int64_t W) {
+                                              int64_t W, bool zero_init) {
     CHECK_GPU(dy);
     auto dyc = dy.contiguous();
     const int N = (int)dyc.size(0), OH = (int)dyc.size(1),
@@ -4027,7 +4071,7 @@ std::vector<torch::Tensor> pool_relu_bias_bwd(torch::Tensor dy,
         // zero-fill serialized the whole layer onto one CU (-30% on the
         // headline config) — keep the parallel grid and pay the 4.7 us fill
         int blocks = (int)std::min<int64_t>(ceildiv(total8, 256), 4096);
-        auto db = blocks == 1
+        auto db = (blocks == 1 || !zero_init)
                       ? torch::empty({K}, dyc.options().dtype(torch::kFloat32))
                       : torch::zeros({K}, dyc.options().dtype(torch::kFloat32));
         hipLaunchKernelGGL(pool_relu_bias_bwd2_kernel, dim3(blocks),
@@ -4046,7 +4090,7 @@ std::vector<torch::Tensor> pool_relu_bias_bwd(torch::Tensor dy,
     lanes = 1 << (31 - __builtin_clz(lanes));
     rpb = std::max(rpb, 2 * lanes);
     int nblk = (int)((M + rpb - 1) / rpb);
-    auto db = nblk == 1
+    auto db = (nblk == 1 || !zero_init)
                   ? torch::empty({K}, dyc.options().dtype(torch::kFloat32))
                   : torch::zeros({K}, dyc.options().dtype(torch::kFloat32));
     hipLaunchKernelGGL(pool_relu_bias_bwd_scalar_kernel, dim3(nblk),
@@ -4306,16 +4350,24 @@ torch::Tensor maxpool_bwd(torch::Tensor dy, torch::Tensor idx, int64_t H,
     CHECK_GPU(dy);
     const int N = (int)dy.size(0), OH = (int)dy.size(1), OW = (int)dy.size(2),
               C = (int)dy.size(3);
-    auto dx32 = torch::zeros({N, H, W, C}, dy.options().dtype(torch::kFloat32));
+    auto stream = at::cuda::getCurrentCUDAStream();
+    const int64_t in_total = (int64_t)N * H * W * C;
+    torch::Tensor dx32h;
+    float* dx32p = acc_pool(in_total, dy.options(), stream, dx32h);
     int64_t total = (int64_t)N * OH * OW * C;
     int blocks = std::min<int64_t>(ceildiv(total, 256), 2048);
     hipLaunchKernelGGL(maxpool_gen_bwd_kernel, dim3(blocks), dim3(256), 0,
-                       at::cuda::getCurrentCUDAStream(), bf_ptr(dy),
-                       idx.data_ptr<uint8_t>(), dx32.data_ptr<float>(), N,
+                       stream, bf_ptr(dy),
+                       idx.data_ptr<uint8_t>(), dx32p, N,
                        (int)H, (int)W, C, OH, OW, (int)k, (int)s, (int)p,
                        fdiv_make((unsigned)C), fdiv_make((unsigned)OW),
                        fdiv_make((unsigned)OH), fdiv_make((unsigned)k));
-    return dx32.to(torch::kBFloat16);
+    auto dx = torch::empty({N, H, W, C}, dy.options());
+    hipLaunchKernelGGL(cast_f32_bf16_kernel,
+                       dim3((int)std::min<int64_t>(ceildiv(in_total, 256), 2048)),
+                       dim3(256), 0, stream, dx32p, bf_ptr_mut(dx), in_total,
+                       1);
+    return dx;
 }
 
 torch::Tensor avgpool_global_fwd(torch::Tensor x) {
@@ -4373,14 +4425,14 @@ torch::Tensor add_relu(torch::Tensor a, torch::Tensor b) {
     return y;
 }
 
-torch::Tensor bias_grad(torch::Tensor dy) {
+torch::Tensor bias_grad(torch::Tensor dy, bool zero_init) {
     CHECK_GPU(dy);
     auto dyc = dy.contiguous();
     const int K = (int)dyc.size(-1);
     const int64_t M = dyc.numel() / K;
     int rpb = (int)std::max<int64_t>(2048, (M + 63) / 64);
     dim3 grid(K, (unsigned)((M + rpb - 1) / rpb));
-    auto db = grid.y == 1
+    auto db = (grid.y == 1 || !zero_init)
                   ? torch::empty({K}, dyc.options().dtype(torch::kFloat32))
                   : torch::zeros({K}, dyc.options().dtype(torch::kFloat32));
     hipLaunchKernelGGL(bias_grad_kernel, grid, dim3(256), 0,

@@ -217,22 +217,35 @@ class LocalClient:
         g = torch.cuda.CUDAGraph()
         import hefl
         C = hefl.load_extension()
-        with torch.cuda.graph(g, stream=side):
-            if in_graph_data:
-                Y = self.dataset.labels.index_select(0, order_buf)
-                zr, sr, fl = self.affine or (0.0, 0.0, False)
-                X = C.synth_batch_g(self.dataset.templates, Y, seed_buf, 0,
-                                    zoom=zr, shear=sr, flip=int(bool(fl)))
-            self.opt.prep_epoch(steps)
-            for s, i in enumerate(range(0, n, B)):
-                self.opt.zero_grad()  # grads=None -> backward steals
-                logits = self.model(X[i:i + B])
-                loss = softmax_xent(logits, Y[i:i + B], self._acc_loss,
-                                    self._acc_correct)
-                loss.backward(gradient=self._one)
-                rows_per_step.append(self.opt.current_ptr_rows())
-                self.opt.step_mt_at(shells[s], s)
-                grads_hold.append([p.grad for p in self.opt.params])
+        from ..ops import functional as Fx
+        try:
+            # capture contract: stolen grads skip their zero-init fills;
+            # the captured Adam clears each grad as it consumes it
+            # (zero_grad=True below) and the eager zero_() after capture
+            # covers the first replay (capture-time contents are stale).
+            Fx.GRAPH_NO_ZERO = True
+            with torch.cuda.graph(g, stream=side):
+                if in_graph_data:
+                    Y = self.dataset.labels.index_select(0, order_buf)
+                    zr, sr, fl = self.affine or (0.0, 0.0, False)
+                    X = C.synth_batch_g(self.dataset.templates, Y, seed_buf, 0,
+                                        zoom=zr, shear=sr, flip=int(bool(fl)))
+                self.opt.prep_epoch(steps)
+                for s, i in enumerate(range(0, n, B)):
+                    self.opt.zero_grad()  # grads=None -> backward steals
+                    logits = self.model(X[i:i + B])
+                    loss = softmax_xent(logits, Y[i:i + B], self._acc_loss,
+                                        self._acc_correct)
+                    loss.backward(gradient=self._one)
+                    rows_per_step.append(self.opt.current_ptr_rows())
+                    self.opt.step_mt_at(shells[s], s, zero_grad=True)
+                    grads_hold.append([p.grad for p in self.opt.params])
+        finally:
+            Fx.GRAPH_NO_ZERO = False
+        with torch.no_grad():
+            for step_grads in grads_hold:
+                for t in step_grads:
+                    t.zero_()
         for shell, rows in zip(shells, rows_per_step):
             self.opt.fill_mt_shell(shell, rows)
         ent = {"graph": g, "X": X, "Y": Y, "mt": shells,

@@ -180,13 +180,15 @@ class FusedAdam:
         shell["ptrs"].copy_(torch.tensor(rows, dtype=torch.int64))
 
     @torch.no_grad()
-    def step_mt_at(self, table, sched_off: int):
+    def step_mt_at(self, table, sched_off: int, zero_grad: bool = False):
         """Adam-only launch (no prep) against the epoch schedule row
-        `sched_off` — captured once per step inside the epoch graph."""
+        `sched_off` — captured once per step inside the epoch graph.
+        zero_grad: consume-and-clear each grad (required when the backward
+        skipped its zero-init fills — functional.GRAPH_NO_ZERO)."""
         C = hefl.load_extension()
         C.fused_adam_mt(table["meta"], table["ptrs"], table["sizes"],
                         table["n"], self._sched_ep, self.beta1, self.beta2,
-                        self.eps, 0, sched_off)
+                        self.eps, 1 if zero_grad else 0, sched_off)
 
     def zero_grad(self):
         for p in self.params:

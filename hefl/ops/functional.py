@@ -45,6 +45,18 @@ def _pad8(t: torch.Tensor) -> torch.Tensor:
     return _C().pad_channels(t.contiguous(), ((C + 7) // 8) * 8)
 
 
+# Epoch-graph capture contract (fl/client.py epoch capture): while True,
+# grad outputs that autograd steals straight into p.grad skip their
+# zero-init fill launch — the captured multi-tensor Adam clears every
+# stolen grad in-graph (zero_g=1) and the client zeroes the held buffers
+# once right after capture, so the buffer is provably zero when the
+# split-K/scatter atomics of the next replay land. Channel-padded conv dw
+# is SLICED before the steal (a fresh tensor becomes p.grad, not the
+# accumulation buffer), so it keeps its fill. The fills this removes were
+# 11.5% of config #2 kernel time (r02_config2_final_kernel_stats.csv).
+GRAPH_NO_ZERO = False
+
+
 # ---------------------------------------------------------------------------
 # Conv2d (valid padding or explicit pre-pad, square stride), NHWC.
 # ---------------------------------------------------------------------------
@@ -88,16 +100,18 @@ class _Conv2dFn(torch.autograd.Function):
         if dy.is_cuda:
             dy = dy.contiguous()
             db = None
+            zi = not GRAPH_NO_ZERO
             if relu and ctx.has_bias:
-                dy, db = _C().relu_bias_bwd(dy, y)  # one fused pass
+                dy, db = _C().relu_bias_bwd(dy, y, zero_init=zi)  # one fused pass
             elif relu:
                 dy = _C().relu_bwd(dy, y)
             elif ctx.has_bias:
-                db = _C().bias_grad(dy)
+                db = _C().bias_grad(dy, zero_init=zi)
             dx = _C().conv2d_dgrad(dy, w, stride, x.shape[1], x.shape[2], pad) \
                 if ctx.needs_input_grad[0] else None
             dw = _C().conv2d_wgrad(dy, x.contiguous(), stride, w.shape[1],
-                                   w.shape[2], pad)
+                                   w.shape[2], pad,
+                                   zero_init=zi or ctx.cpad)
             if ctx.cpad:  # drop the zero-padded channel lanes
                 dw = dw[..., :ctx.in_C].contiguous()
                 if dx is not None:
@@ -192,11 +206,13 @@ class _ConvReluPoolFn(torch.autograd.Function):
     def backward(ctx, dy):
         x, w, p, idx = ctx.saved_tensors
         ch, cw = ctx.conv_hw
-        dym, db = _C().pool_relu_bias_bwd(dy.contiguous(), idx, p, ch, cw)
+        zi = not GRAPH_NO_ZERO
+        dym, db = _C().pool_relu_bias_bwd(dy.contiguous(), idx, p, ch, cw,
+                                          zero_init=zi)
         dx = _C().conv2d_dgrad(dym, w, 1, x.shape[1], x.shape[2], ctx.pad) \
             if ctx.needs_input_grad[0] else None
         dw = _C().conv2d_wgrad(dym, x.contiguous(), 1, w.shape[1], w.shape[2],
-                               ctx.pad)
+                               ctx.pad, zero_init=zi or ctx.cpad)
         if ctx.cpad:
             dw = dw[..., :ctx.in_C].contiguous()
             if dx is not None:
@@ -239,12 +255,13 @@ class _LinearFn(torch.autograd.Function):
         if dy.is_cuda:
             dy = dy.contiguous()
             db = None
+            zi = not GRAPH_NO_ZERO
             if ctx.relu and ctx.has_bias:
-                dy, db = _C().relu_bias_bwd(dy, y)
+                dy, db = _C().relu_bias_bwd(dy, y, zero_init=zi)
             elif ctx.relu:
                 dy = _C().relu_bwd(dy, y)
             elif ctx.has_bias:
-                db = _C().bias_grad(dy.view(-1, dy.shape[-1]))
+                db = _C().bias_grad(dy.view(-1, dy.shape[-1]), zero_init=zi)
             dx = _C().linear_dgrad(dy, w) if ctx.needs_input_grad[0] else None
             dw = _C().linear_wgrad(dy, x).to(torch.float32)
         else:

@@ -369,3 +369,69 @@ def test_conv_tile3_dgrad_matches_glds(shape):
     # so one-ulp flips at |dx| >= 1.28 are expected — relative check like
     # the rest of the suite (a real indexing bug is O(max), far outside it)
     _close(dx_tile, dx_ref, rel=1e-2)
+
+
+def test_splitk_accumulator_pool_stays_clean():
+    """The split-K fp32 accumulator is a pooled buffer that consumer kernels
+    consume-and-clear (cnn.hip acc_pool): repeated calls must give identical
+    results (a broken clear would leak the previous call's partial sums into
+    the next), and match the fp32 reference."""
+    torch.manual_seed(3)
+    # small-M, deep-K shape: tiles < 256 and ksteps >= 16 force k_chunks > 1
+    # in both the fwd and dgrad glds paths
+    N, H, W, C, K = 2, 8, 8, 64, 64
+    x = torch.randn(N, H, W, C)
+    w = torch.randn(K, 3, 3, C) * 0.1
+    b = torch.randn(K) * 0.1
+
+    xc = x.clone().requires_grad_(True)
+    wc = w.clone().requires_grad_(True)
+    bc = b.clone().requires_grad_(True)
+    yc = Fx.conv2d(xc, wc, bc, stride=1, relu=False, pad=0)
+    g = torch.randn_like(yc)
+    yc.backward(g)
+
+    outs = []
+    for _ in range(3):
+        xg = x.to("cuda", torch.bfloat16).requires_grad_(True)
+        wg = w.cuda().requires_grad_(True)
+        bg = b.cuda().requires_grad_(True)
+        yg = Fx.conv2d(xg, wg, bg, stride=1, relu=False, pad=0)
+        yg.backward(g.to("cuda", torch.bfloat16))
+        outs.append((yg.detach(), xg.grad))
+    # repeat-stable: fp32 atomics reorder (ulp wiggle) but a broken clear
+    # would leak the whole previous sum (O(1) relative error)
+    for i in (1, 2):
+        _close(outs[i][0], outs[0][0], rel=1e-3)
+        _close(outs[i][1], outs[0][1], rel=1e-3)
+    _close(outs[0][0], yc)
+    _close(outs[0][1], xc.grad, rel=5e-2)
+
+
+def test_generic_maxpool_bwd_pooled_accumulator():
+    """maxpool_bwd (generic k/s/p path) scatters into the pooled accumulator;
+    repeated calls must match each other and the fp32 reference (3x3 stride-2
+    pad-1 pool: overlapping windows + padding exercise untouched elements)."""
+    torch.manual_seed(4)
+    N, H, W, C = 2, 16, 16, 32
+    # quantize to bf16 first: CPU and GPU must see identical values or
+    # near-ties argmax differently and the scatter lands elsewhere (same
+    # caveat as test_maxpool_gpu)
+    x = torch.randn(N, H, W, C).to(torch.bfloat16).float()
+    xc = x.clone().requires_grad_(True)
+    # NHWC -> NCHW for the torch reference
+    yc = F.max_pool2d(xc.permute(0, 3, 1, 2), 3, 2, 1)
+    g = torch.randn_like(yc)
+    yc.backward(g)
+
+    outs = []
+    for _ in range(3):
+        xg = x.to("cuda", torch.bfloat16).requires_grad_(True)
+        yg = Fx.maxpool(xg, k=3, s=2, p=1)
+        yg.backward(g.permute(0, 2, 3, 1).contiguous().to("cuda", torch.bfloat16))
+        outs.append((yg.detach(), xg.grad))
+    for i in (1, 2):
+        _close(outs[i][0], outs[0][0], rel=1e-3)
+        _close(outs[i][1], outs[0][1], rel=1e-3)
+    _close(outs[0][0], yc.permute(0, 2, 3, 1))
+    _close(outs[0][1], xc.grad)  # xc is NHWC, so its grad already is too
