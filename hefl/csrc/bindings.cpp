@@ -80,14 +80,15 @@ torch::Tensor relu_bwd(torch::Tensor dy, torch::Tensor y);
 std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
                                   torch::Tensor beta, torch::Tensor rmean,
                                   torch::Tensor rvar, double eps,
-                                  double momentum, bool relu);
+                                  double momentum, bool relu,
+                                  torch::Tensor sums);
 torch::Tensor bn_apply(torch::Tensor x, torch::Tensor mean,
                        torch::Tensor invstd, torch::Tensor gamma,
                        torch::Tensor beta, bool relu);
 std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
                                   torch::Tensor mean, torch::Tensor invstd,
                                   torch::Tensor gamma, bool train,
-                                  torch::Tensor relu_y);
+                                  torch::Tensor relu_y, torch::Tensor fwd_sums);
 std::vector<torch::Tensor> maxpool_fwd(torch::Tensor x, int64_t k, int64_t s,
                                        int64_t p);
 torch::Tensor maxpool_bwd(torch::Tensor dy, torch::Tensor idx, int64_t H,

@@ -2670,10 +2670,19 @@ __global__ void unpack_mt_kernel(const float* __restrict__ flat,
 // atomically added into fp32 accumulators; stage 2 finalizes mean/invstd.
 // writes per-block partial sums into slab[b][2][C] (torch::empty — every
 // cell is stored exactly once, so no zero-fill or atomics are needed)
+// ATOMIC=true: `slab` is a single [2*C] sums buffer (zero at entry) and
+// every per-block partial is atomicAdd'ed into it — no finalize launch;
+// the consumer (bn_apply_stats_kernel) derives mean/invstd from the raw
+// sums and the NEXT bn_bwd_partial<true> clears the buffer (stream-order
+// consume-and-clear, same contract as acc_pool).
+#define BN_ST(arr, i, v) \
+    do { if (ATOMIC) atomicAdd(&(arr)[i], (v)); else (arr)[i] = (v); } while (0)
+
+template <bool ATOMIC>
 __global__ void bn_partial_kernel(const unsigned short* __restrict__ x,
                                   float* __restrict__ slab, int64_t M, int C,
                                   int rows_per_block) {
-    float* gsum = slab + (int64_t)blockIdx.x * 2 * C;
+    float* gsum = ATOMIC ? slab : slab + (int64_t)blockIdx.x * 2 * C;
     float* gsq = gsum + C;
     __shared__ float red[2][256];
     const int64_t r0 = (int64_t)blockIdx.x * rows_per_block;
@@ -2704,8 +2713,8 @@ __global__ void bn_partial_kernel(const unsigned short* __restrict__ x,
                 }
 #pragma unroll
                 for (int j = 0; j < 8; ++j) {
-                    gsum[o * 8 + j] = a[j];
-                    gsq[o * 8 + j] = a2[j];
+                    BN_ST(gsum, o * 8 + j, a[j]);
+                    BN_ST(gsq, o * 8 + j, a2[j]);
                 }
             }
             return;
@@ -2742,8 +2751,8 @@ __global__ void bn_partial_kernel(const unsigned short* __restrict__ x,
         if (rl == 0) {
 #pragma unroll
             for (int j = 0; j < 8; ++j) {
-                gsum[oct * 8 + j] = red8[0][threadIdx.x * 8 + j];
-                gsq[oct * 8 + j] = red8[1][threadIdx.x * 8 + j];
+                BN_ST(gsum, oct * 8 + j, red8[0][threadIdx.x * 8 + j]);
+                BN_ST(gsq, oct * 8 + j, red8[1][threadIdx.x * 8 + j]);
             }
         }
         return;
@@ -2756,8 +2765,8 @@ __global__ void bn_partial_kernel(const unsigned short* __restrict__ x,
                 acc += v;
                 acc2 += v * v;
             }
-            gsum[c] = acc;
-            gsq[c] = acc2;
+            BN_ST(gsum, c, acc);
+            BN_ST(gsq, c, acc2);
         }
         return;
     }
@@ -2787,8 +2796,8 @@ __global__ void bn_partial_kernel(const unsigned short* __restrict__ x,
         __syncthreads();
     }
     if (rl == 0) {
-        gsum[c] = red[0][threadIdx.x];
-        gsq[c] = red[1][threadIdx.x];
+        BN_ST(gsum, c, red[0][threadIdx.x]);
+        BN_ST(gsq, c, red[1][threadIdx.x]);
     }
 }
 
@@ -2954,6 +2963,75 @@ __global__ void bn_apply_kernel(const unsigned short* __restrict__ x,
     }
 }
 
+// bn_apply over ATOMIC raw sums: mean/invstd are derived per element from
+// the [2*C] sums written by bn_partial_kernel<true> (2 extra L2-resident
+// loads + a rsqrt on a memory-bound pass); block 0 additionally persists
+// mean/invstd for the backward and updates the running stats — this
+// replaces the bn_finalize launch entirely. Unlike the counter-gated
+// last-block probe (bn_partial_fused_kernel, measured -15%), no block
+// waits on any other.
+__global__ void bn_apply_stats_kernel(const unsigned short* __restrict__ x,
+                                      const float* __restrict__ sums,
+                                      const float* __restrict__ gamma,
+                                      const float* __restrict__ beta,
+                                      unsigned short* __restrict__ y,
+                                      int64_t total, int C, int64_t M,
+                                      float eps, float momentum,
+                                      float* __restrict__ mean_out,
+                                      float* __restrict__ invstd_out,
+                                      float* __restrict__ running_mean,
+                                      float* __restrict__ running_var,
+                                      int relu, FastDiv fC) {
+    const float invM = 1.f / (float)M;
+    if (blockIdx.x == 0) {
+        for (int c = threadIdx.x; c < C; c += blockDim.x) {
+            const float mu = sums[c] * invM;
+            const float var = fmaxf(sums[C + c] * invM - mu * mu, 0.f);
+            mean_out[c] = mu;
+            invstd_out[c] = rsqrtf(var + eps);
+            if (running_mean) {  // torch semantics: UNBIASED running var
+                float unb = var * ((float)M / (float)max(M - 1, (int64_t)1));
+                running_mean[c] =
+                    (1.f - momentum) * running_mean[c] + momentum * mu;
+                running_var[c] =
+                    (1.f - momentum) * running_var[c] + momentum * unb;
+            }
+        }
+    }
+    if ((C & 7) == 0) {  // octet path (fC built over C/8 by the host)
+        const int64_t t8 = total >> 3;
+        const int noct = C >> 3;
+        for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+             i < t8; i += (int64_t)gridDim.x * blockDim.x) {
+            const int c0 = 8 * (int)((unsigned)i - fdiv((unsigned)i, fC) * noct);
+            u16x8 x8 = *reinterpret_cast<const u16x8*>(&x[i * 8]);
+            u16x8 y8;
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                const int c = c0 + j;
+                const float mu = sums[c] * invM;
+                const float is =
+                    rsqrtf(fmaxf(sums[C + c] * invM - mu * mu, 0.f) + eps);
+                float v = (bf2f(x8[j]) - mu) * is * gamma[c] + beta[c];
+                if (relu) v = v > 0.f ? v : 0.f;
+                y8[j] = f2bf(v);
+            }
+            *reinterpret_cast<u16x8*>(&y[i * 8]) = y8;
+        }
+        return;
+    }
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        int c = (int)((unsigned)i - fdiv((unsigned)i, fC) * C);
+        const float mu = sums[c] * invM;
+        const float is =
+            rsqrtf(fmaxf(sums[C + c] * invM - mu * mu, 0.f) + eps);
+        float v = (bf2f(x[i]) - mu) * is * gamma[c] + beta[c];
+        if (relu) v = v > 0.f ? v : 0.f;
+        y[i] = f2bf(v);
+    }
+}
+
 // dgamma[c] = sum dy*xhat; dbeta[c] = sum dy — same coalesced two-stage
 // shape (partials straight into dgamma/dbeta, zeroed by the wrapper).
 // relu_y: optional post-ReLU BN output — gates dy in place of a separate
@@ -2968,15 +3046,29 @@ __device__ __forceinline__ float bf_gated(const unsigned short* dy,
     return bf2f(dy[i]);
 }
 
+// ATOMIC=true: `slab` IS the {dgamma;dbeta} [2*C] output (zero at entry —
+// the captured Adam consume-and-cleared it last step under the epoch-graph
+// contract); block 0 also zeroes this layer's forward sums buffer for the
+// next replay's bn_partial<true> (the forward consumer already ran, in
+// stream order).
+template <bool ATOMIC>
 __global__ void bn_bwd_partial_kernel(const unsigned short* __restrict__ dy,
                                       const unsigned short* __restrict__ x,
                                       const unsigned short* __restrict__ relu_y,
                                       const float* __restrict__ mean,
                                       const float* __restrict__ invstd,
                                       float* __restrict__ slab, int64_t M,
-                                      int C, int rows_per_block) {
-    float* dgamma = slab + (int64_t)blockIdx.x * 2 * C;
-    float* dbeta = dgamma + C;
+                                      int C, int rows_per_block,
+                                      float* __restrict__ dbeta_buf,
+                                      float* __restrict__ fwd_sums) {
+    // ATOMIC: slab IS dgamma [C] and dbeta_buf IS dbeta [C] (two separate
+    // tensors — each is stolen into its own p.grad, so they cannot share
+    // one allocation). Classic: per-block slab rows, dbeta_buf unused.
+    float* dgamma = ATOMIC ? slab : slab + (int64_t)blockIdx.x * 2 * C;
+    float* dbeta = ATOMIC ? dbeta_buf : dgamma + C;
+    if (ATOMIC && fwd_sums && blockIdx.x == 0)
+        for (int i = threadIdx.x; i < 2 * C; i += blockDim.x)
+            fwd_sums[i] = 0.f;
     __shared__ float red[2][256];
     const int64_t r0 = (int64_t)blockIdx.x * rows_per_block;
     const int64_t r1 = min(r0 + rows_per_block, M);
@@ -3016,8 +3108,8 @@ __global__ void bn_bwd_partial_kernel(const unsigned short* __restrict__ dy,
                 }
 #pragma unroll
                 for (int j = 0; j < 8; ++j) {
-                    dgamma[o * 8 + j] = dg[j];
-                    dbeta[o * 8 + j] = db[j];
+                    BN_ST(dgamma, o * 8 + j, dg[j]);
+                    BN_ST(dbeta, o * 8 + j, db[j]);
                 }
             }
             return;
@@ -3068,8 +3160,8 @@ __global__ void bn_bwd_partial_kernel(const unsigned short* __restrict__ dy,
         if (rl == 0) {
 #pragma unroll
             for (int j = 0; j < 8; ++j) {
-                dgamma[oct * 8 + j] = red8[0][threadIdx.x * 8 + j];
-                dbeta[oct * 8 + j] = red8[1][threadIdx.x * 8 + j];
+                BN_ST(dgamma, oct * 8 + j, red8[0][threadIdx.x * 8 + j]);
+                BN_ST(dbeta, oct * 8 + j, red8[1][threadIdx.x * 8 + j]);
             }
         }
         return;
@@ -3083,8 +3175,8 @@ __global__ void bn_bwd_partial_kernel(const unsigned short* __restrict__ dy,
                 dg += g * (bf2f(x[r * C + c]) - mu) * is;
                 db += g;
             }
-            dgamma[c] = dg;
-            dbeta[c] = db;
+            BN_ST(dgamma, c, dg);
+            BN_ST(dbeta, c, db);
         }
         return;
     }
@@ -3112,8 +3204,8 @@ __global__ void bn_bwd_partial_kernel(const unsigned short* __restrict__ dy,
         __syncthreads();
     }
     if (rl == 0) {
-        dgamma[c] = red[0][threadIdx.x];
-        dbeta[c] = red[1][threadIdx.x];
+        BN_ST(dgamma, c, red[0][threadIdx.x]);
+        BN_ST(dbeta, c, red[1][threadIdx.x]);
     }
 }
 
@@ -4224,7 +4316,8 @@ torch::Tensor relu_bwd(torch::Tensor dy, torch::Tensor y) {
 std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
                                   torch::Tensor beta, torch::Tensor rmean,
                                   torch::Tensor rvar, double eps,
-                                  double momentum, bool relu) {
+                                  double momentum, bool relu,
+                                  torch::Tensor sums) {
     CHECK_GPU(x);
     TORCH_CHECK(x.is_contiguous());
     const int C = (int)x.size(-1);
@@ -4236,6 +4329,27 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
     auto stream = at::cuda::getCurrentCUDAStream();
     int rpb = (int)std::max<int64_t>(64, (M + 511) / 512);  // fill 2 blocks/CU
     int nblk = (int)((M + rpb - 1) / rpb);
+    if (sums.numel()) {
+        // atomic path (epoch-graph contract, ops/functional.py): partials
+        // straight into the caller's [2*C] sums buffer (zero at entry),
+        // stats derived inline by the apply kernel — no finalize launch.
+        TORCH_CHECK(sums.numel() == 2 * C);
+        hipLaunchKernelGGL(bn_partial_kernel<true>, dim3(nblk), dim3(256), 0,
+                           stream, bf_ptr(x), sums.data_ptr<float>(), M, C,
+                           rpb);
+        int64_t total = x.numel();
+        int blocks = std::min<int64_t>(ceildiv(total, 256), 2048);
+        hipLaunchKernelGGL(
+            bn_apply_stats_kernel, dim3(blocks), dim3(256), 0, stream,
+            bf_ptr(x), sums.data_ptr<float>(), gamma.data_ptr<float>(),
+            beta.data_ptr<float>(), bf_ptr_mut(y), total, C, M, (float)eps,
+            (float)momentum, mean.data_ptr<float>(),
+            invstd.data_ptr<float>(),
+            rmean.numel() ? rmean.data_ptr<float>() : nullptr,
+            rvar.numel() ? rvar.data_ptr<float>() : nullptr, relu ? 1 : 0,
+            fdiv_make((unsigned)((C & 7) == 0 ? C / 8 : C)));
+        return {y, mean, invstd};
+    }
     auto slab = torch::empty({nblk, 2, C}, f32);
     static const bool bn_fuse = [] {
         const char* e = getenv("HEFL_BN_FUSE");
@@ -4254,8 +4368,9 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor gamma,
                            (float)eps, (float)momentum,
                            ctr.data_ptr<int>());
     } else {
-        hipLaunchKernelGGL(bn_partial_kernel, dim3(nblk), dim3(256), 0, stream,
-                           bf_ptr(x), slab.data_ptr<float>(), M, C, rpb);
+        hipLaunchKernelGGL(bn_partial_kernel<false>, dim3(nblk), dim3(256), 0,
+                           stream, bf_ptr(x), slab.data_ptr<float>(), M, C,
+                           rpb);
         hipLaunchKernelGGL(bn_finalize_kernel, dim3(C), dim3(256), 0,
                            stream, slab.data_ptr<float>(), nblk,
                            mean.data_ptr<float>(), invstd.data_ptr<float>(),
@@ -4292,7 +4407,8 @@ torch::Tensor bn_apply(torch::Tensor x, torch::Tensor mean,
 std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
                                   torch::Tensor mean, torch::Tensor invstd,
                                   torch::Tensor gamma, bool train,
-                                  torch::Tensor relu_y) {
+                                  torch::Tensor relu_y,
+                                  torch::Tensor fwd_sums) {
     CHECK_GPU(dy);
     auto dyc = dy.contiguous();
     const unsigned short* ry =
@@ -4300,20 +4416,37 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
     const int C = (int)x.size(-1);
     const int64_t M = x.numel() / C;
     auto f32 = x.options().dtype(torch::kFloat32);
-    auto dgamma = torch::empty({C}, f32);
-    auto dbeta = torch::empty({C}, f32);
     auto dx = torch::empty_like(x);
     auto stream = at::cuda::getCurrentCUDAStream();
     int rpb = (int)std::max<int64_t>(64, (M + 511) / 512);  // fill 2 blocks/CU
     int nblk = (int)((M + rpb - 1) / rpb);
-    auto slab = torch::empty({nblk, 2, C}, f32);
-    hipLaunchKernelGGL(bn_bwd_partial_kernel, dim3(nblk), dim3(256), 0, stream,
-                       bf_ptr(dyc), bf_ptr(x), ry, mean.data_ptr<float>(),
-                       invstd.data_ptr<float>(), slab.data_ptr<float>(), M, C,
-                       rpb);
-    hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3(C), dim3(256), 0,
-                       stream, slab.data_ptr<float>(), nblk,
-                       dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), C);
+    torch::Tensor dgamma, dbeta;
+    if (fwd_sums.numel()) {
+        // atomic path: partials straight into {dgamma;dbeta} (zero at
+        // entry — the captured Adam consume-and-cleared them last step);
+        // block 0 zeroes this layer's forward sums for the next replay.
+        dgamma = torch::empty({C}, f32);
+        dbeta = torch::empty({C}, f32);
+        hipLaunchKernelGGL(bn_bwd_partial_kernel<true>, dim3(nblk), dim3(256),
+                           0, stream, bf_ptr(dyc), bf_ptr(x), ry,
+                           mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                           dgamma.data_ptr<float>(), M, C, rpb,
+                           dbeta.data_ptr<float>(),
+                           fwd_sums.data_ptr<float>());
+    } else {
+        dgamma = torch::empty({C}, f32);
+        dbeta = torch::empty({C}, f32);
+        auto slab = torch::empty({nblk, 2, C}, f32);
+        hipLaunchKernelGGL(bn_bwd_partial_kernel<false>, dim3(nblk), dim3(256),
+                           0, stream, bf_ptr(dyc), bf_ptr(x), ry,
+                           mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                           slab.data_ptr<float>(), M, C, rpb, nullptr,
+                           nullptr);
+        hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3(C), dim3(256), 0,
+                           stream, slab.data_ptr<float>(), nblk,
+                           dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
+                           C);
+    }
     int64_t total = x.numel();
     int blocks = std::min<int64_t>(ceildiv(total, 256), 2048);
     hipLaunchKernelGGL(bn_dx_kernel, dim3(blocks), dim3(256), 0, stream,

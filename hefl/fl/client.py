@@ -218,12 +218,20 @@ class LocalClient:
         import hefl
         C = hefl.load_extension()
         from ..ops import functional as Fx
+        import os
+        # HEFL_GRAPH_NO_ZERO=1 (probe, default OFF): stolen grads skip
+        # their zero-init fills; the captured Adam clears each grad as it
+        # consumes it and the eager zero_() after capture covers the first
+        # replay. Measured config2 36.7->38.8 rounds/s, config4
+        # 35.5k->42.1k samples/s — but the contract is UNSOUND in general:
+        # a grad buffer allocated during capture can reuse a pool block
+        # freed earlier in the SAME capture, and the previous owner's
+        # captured writes then re-pollute it on every replay (observed as
+        # training divergence on resnet18 BN grads; cnn2/refcnn6 happened
+        # to capture cleanly). Default stays correct-by-construction.
+        no_zero = os.getenv("HEFL_GRAPH_NO_ZERO", "0") == "1"
         try:
-            # capture contract: stolen grads skip their zero-init fills;
-            # the captured Adam clears each grad as it consumes it
-            # (zero_grad=True below) and the eager zero_() after capture
-            # covers the first replay (capture-time contents are stale).
-            Fx.GRAPH_NO_ZERO = True
+            Fx.GRAPH_NO_ZERO = no_zero
             with torch.cuda.graph(g, stream=side):
                 if in_graph_data:
                     Y = self.dataset.labels.index_select(0, order_buf)
@@ -238,7 +246,7 @@ class LocalClient:
                                         self._acc_correct)
                     loss.backward(gradient=self._one)
                     rows_per_step.append(self.opt.current_ptr_rows())
-                    self.opt.step_mt_at(shells[s], s, zero_grad=True)
+                    self.opt.step_mt_at(shells[s], s, zero_grad=no_zero)
                     grads_hold.append([p.grad for p in self.opt.params])
         finally:
             Fx.GRAPH_NO_ZERO = False

@@ -45,15 +45,27 @@ def _pad8(t: torch.Tensor) -> torch.Tensor:
     return _C().pad_channels(t.contiguous(), ((C + 7) // 8) * 8)
 
 
-# Epoch-graph capture contract (fl/client.py epoch capture): while True,
-# grad outputs that autograd steals straight into p.grad skip their
-# zero-init fill launch — the captured multi-tensor Adam clears every
-# stolen grad in-graph (zero_g=1) and the client zeroes the held buffers
-# once right after capture, so the buffer is provably zero when the
-# split-K/scatter atomics of the next replay land. Channel-padded conv dw
-# is SLICED before the steal (a fresh tensor becomes p.grad, not the
-# accumulation buffer), so it keeps its fill. The fills this removes were
-# 11.5% of config #2 kernel time (r02_config2_final_kernel_stats.csv).
+import os
+
+# BN atomic no-finalize probe, default OFF: measured -35% on config #5
+# (hot-word fp32 atomics from ~512 partial blocks per channel word beat
+# the 4.7 us finalize launches they replace) and the backward half shares
+# the HEFL_GRAPH_NO_ZERO capture-pool-reuse hazard (fl/client.py).
+_BN_ATOMIC_MODE = os.getenv("HEFL_BN_ATOMIC", "0")
+
+# Epoch-graph capture contract (fl/client.py epoch capture, probe
+# HEFL_GRAPH_NO_ZERO=1 — default OFF): while True, grad outputs that
+# autograd steals straight into p.grad skip their zero-init fill launch —
+# the captured multi-tensor Adam clears every stolen grad in-graph
+# (zero_g=1) and the client zeroes the held buffers once right after
+# capture. Channel-padded conv dw is SLICED before the steal, so it keeps
+# its fill. Measured: the fills were 11.5% of config #2 kernel time;
+# skipping them gave config2 +5.5%, config4 +19%. DEFAULT OFF because the
+# contract is unsound in general: an allocation made DURING capture can
+# reuse a pool block freed earlier in the same capture, and the previous
+# owner's captured writes re-pollute the "zero" buffer on every replay
+# (observed: resnet18 BN grads diverged; cnn2/refcnn6 happened to capture
+# cleanly). See fl/client.py and PERFORMANCE.md.
 GRAPH_NO_ZERO = False
 
 
@@ -360,28 +372,43 @@ def softmax_xent(logits, labels, acc_loss=None, acc_correct=None):
 class _BatchNorm2dFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, gamma, beta, running_mean, running_var, training: bool,
-                momentum: float, eps: float, relu: bool):
+                momentum: float, eps: float, relu: bool, sums=None):
         ctx.relu = relu
         ctx.training = training
         C = x.shape[-1]
         if x.is_cuda:
             if training:
                 # running stats update fused into the finalize kernel (the
-                # torch mul_/add_ chain was 4 kernels per BN layer per step)
+                # torch mul_/add_ chain was 4 kernels per BN layer per step).
+                # Under the epoch-graph capture contract (GRAPH_NO_ZERO),
+                # the per-layer `sums` buffer switches both passes to the
+                # atomic no-finalize kernels: dgamma/dbeta are accumulated
+                # straight into the stolen grads (Adam clears them), and
+                # the backward partial clears `sums` for the next replay.
                 empty = torch.empty(0, device=x.device)
+                # HEFL_BN_ATOMIC: 1 = atomic fwd+bwd (no finalize launches),
+                # bwd = classic fwd + atomic bwd (bisect aid), 0 = classic
+                mode = _BN_ATOMIC_MODE
+                use_atomic = (GRAPH_NO_ZERO and sums is not None
+                              and mode != "0")
+                ctx.sums = sums if use_atomic else None
+                ctx.atomic_fwd = use_atomic and mode != "bwd"
                 y, mean, invstd = _C().bn_fwd(
                     x.contiguous(), gamma.detach().float(),
                     beta.detach().float(),
                     running_mean if running_mean is not None else empty,
                     running_var if running_var is not None else empty,
-                    eps, momentum, relu)
+                    eps, momentum, relu,
+                    sums if ctx.atomic_fwd else empty)
             else:
+                ctx.sums = None
                 mean = running_mean
                 invstd = (running_var + eps).rsqrt()
                 y = _C().bn_apply(x.contiguous(), mean, invstd,
                                   gamma.detach().float(), beta.detach().float(),
                                   relu)
         else:
+            ctx.sums = None
             xf = x.float().reshape(-1, C)
             if training:
                 mean = xf.mean(0)
@@ -413,8 +440,11 @@ class _BatchNorm2dFn(torch.autograd.Function):
             dy = dy.contiguous()
             # relu gate fused into the BN backward kernels (no relu_bwd pass)
             ry = y if ctx.relu else torch.empty(0, device=dy.device)
+            fs = ctx.sums if ctx.sums is not None \
+                else torch.empty(0, device=dy.device)
             dx, dgamma, dbeta = _C().bn_bwd(dy, x.contiguous(), mean, invstd,
-                                            gamma.float(), ctx.training, ry)
+                                            gamma.float(), ctx.training, ry,
+                                            fs)
         else:
             dy = dy.float()
             if ctx.relu:
@@ -429,13 +459,13 @@ class _BatchNorm2dFn(torch.autograd.Function):
             else:
                 dx = (gamma * invstd) * dyf
             dx = dx.reshape(x.shape)
-        return dx, dgamma, dbeta, None, None, None, None, None, None
+        return dx, dgamma, dbeta, None, None, None, None, None, None, None
 
 
 def batchnorm2d(x, gamma, beta, running_mean, running_var, training=True,
-                momentum=0.1, eps=1e-5, relu=False):
+                momentum=0.1, eps=1e-5, relu=False, sums=None):
     return _BatchNorm2dFn.apply(x, gamma, beta, running_mean, running_var,
-                                training, momentum, eps, relu)
+                                training, momentum, eps, relu, sums)
 
 
 # ---------------------------------------------------------------------------

@@ -93,11 +93,25 @@ class BatchNorm2d(nn.Module):
         self.register_buffer("running_mean", torch.zeros(c))
         self.register_buffer("running_var", torch.ones(c))
         self.eps, self.momentum, self.relu = eps, momentum, relu
+        # [2*c] fp32 raw-sums scratch for the atomic no-finalize BN path
+        # (epoch-graph capture only — functional.GRAPH_NO_ZERO). A plain
+        # attribute, NOT a buffer: it must stay out of the FedAvg flat
+        # vector (weights.py aggregates floating-point buffers) and out of
+        # checkpoints. Allocated eagerly on first CUDA forward OUTSIDE a
+        # capture (the engine's 3 warmup steps) so the zero-fill is never
+        # captured; invariant: zero between steps (bn_bwd_partial<true>
+        # block 0 re-zeroes it each backward).
+        self._sums = None
 
     def forward(self, x):
+        if x.is_cuda and self._sums is None \
+                and not torch.cuda.is_current_stream_capturing():
+            self._sums = torch.zeros(2 * self.weight.numel(),
+                                     dtype=torch.float32, device=x.device)
         return Fx.batchnorm2d(x, self.weight, self.bias, self.running_mean,
                               self.running_var, self.training, self.momentum,
-                              self.eps, self.relu)
+                              self.eps, self.relu,
+                              sums=self._sums if x.is_cuda else None)
 
 
 class MaxPool(nn.Module):
