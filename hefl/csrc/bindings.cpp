@@ -51,7 +51,7 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
 torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor w, int64_t stride,
                            int64_t H, int64_t W, int64_t pad);
 torch::Tensor conv2d_wgrad(torch::Tensor dy, torch::Tensor x, int64_t stride,
-                           int64_t R, int64_t S, int64_t pad, bool zero_init);
+                           int64_t R, int64_t S, int64_t pad, int64_t gkey);
 torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor b,
                          bool relu);
 torch::Tensor linear_dgrad(torch::Tensor dy, torch::Tensor w);
@@ -96,13 +96,13 @@ torch::Tensor maxpool_bwd(torch::Tensor dy, torch::Tensor idx, int64_t H,
 torch::Tensor avgpool_global_fwd(torch::Tensor x);
 torch::Tensor avgpool_global_bwd(torch::Tensor dy, int64_t H, int64_t W);
 torch::Tensor add_relu(torch::Tensor a, torch::Tensor b);
-torch::Tensor bias_grad(torch::Tensor dy, bool zero_init);
+torch::Tensor bias_grad(torch::Tensor dy, int64_t gkey);
 std::vector<torch::Tensor> relu_bias_bwd(torch::Tensor dy, torch::Tensor y,
-                                         bool zero_init);
+                                         int64_t gkey);
 std::vector<torch::Tensor> pool_relu_bias_bwd(torch::Tensor dy,
                                               torch::Tensor idx,
                                               torch::Tensor p, int64_t H,
-                                              int64_t W, bool zero_init);
+                                              int64_t W, int64_t gkey);
 void adam_prep_epoch(torch::Tensor step, torch::Tensor sched,
                      torch::Tensor hyper, double b1, double b2, int64_t S);
 torch::Tensor pad_channels(torch::Tensor x, int64_t C8);
@@ -149,7 +149,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("conv2d_dgrad", &conv2d_dgrad);
     m.def("conv2d_wgrad", &conv2d_wgrad, py::arg("dy"), py::arg("x"),
           py::arg("stride"), py::arg("R"), py::arg("S"), py::arg("pad"),
-          py::arg("zero_init") = true);
+          py::arg("gkey") = 0);
     m.def("linear_fwd", &linear_fwd);
     m.def("linear_dgrad", &linear_dgrad);
     m.def("linear_wgrad", &linear_wgrad);
@@ -174,13 +174,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("avgpool_global_fwd", &avgpool_global_fwd);
     m.def("avgpool_global_bwd", &avgpool_global_bwd);
     m.def("add_relu", &add_relu);
-    m.def("bias_grad", &bias_grad, py::arg("dy"),
-          py::arg("zero_init") = true);
+    m.def("bias_grad", &bias_grad, py::arg("dy"), py::arg("gkey") = 0);
     m.def("relu_bias_bwd", &relu_bias_bwd, py::arg("dy"), py::arg("y"),
-          py::arg("zero_init") = true);
+          py::arg("gkey") = 0);
     m.def("pool_relu_bias_bwd", &pool_relu_bias_bwd, py::arg("dy"),
           py::arg("idx"), py::arg("p"), py::arg("H"), py::arg("W"),
-          py::arg("zero_init") = true);
+          py::arg("gkey") = 0);
     m.def("fused_adam_mt", &fused_adam_mt);
     m.def("pad_channels", &pad_channels,
           "zero-pad NHWC channel dim to C8 (stem -> glds MFMA path)");

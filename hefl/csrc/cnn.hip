@@ -72,6 +72,7 @@
 #include <ATen/cuda/CUDAContext.h>
 
 #include <algorithm>
+#include <unordered_map>
 
 #define CHECK_GPU(x) TORCH_CHECK((x).is_cuda(), #x " must be on GPU")
 
@@ -3545,6 +3546,54 @@ unsigned short* bf_ptr_mut(torch::Tensor& t) {
 // capture is active would allocate from the capture mempool (whose blocks
 // die with the graph), so it falls back to a per-call zeros tensor then —
 // eager warmup before capture normally sizes the pool first.
+// Sound zero-init-free grad buffers for the epoch-graph fill-skip
+// (fl/client.py, HEFL_GRAPH_NO_ZERO): raw hipMalloc allocations keyed by
+// the caller's param identity, wrapped per call in from_blob — a fresh
+// TensorImpl with use_count 1, so AccumulateGrad STEALS it into p.grad
+// and the captured multi-tensor Adam consume-and-clears the memory each
+// step. Because the memory never belongs to the caching allocator, no
+// tensor allocated during a hipGraph capture can alias it — the
+// capture-pool block-reuse hazard that sank the first fill-skip design
+// (a freed block's previous owner re-polluted the "zero" buffer every
+// replay; observed as resnet18 divergence) cannot occur. Zeroed once at
+// creation; the Adam clear maintains the invariant thereafter.
+// Caveats (checked/benign): a key whose numel changed (CPython id reuse
+// across client lifetimes) falls back to the classic zeroed path; two
+// clients' graphs may share a buffer on id reuse with equal shapes,
+// which is safe because every accumulate->consume pair completes within
+// one captured step. Weight sharing (one param receiving two grads in a
+// step) is NOT supported on this path — no model here shares weights.
+torch::Tensor grad_buf(int64_t key, at::IntArrayRef sizes,
+                       const torch::TensorOptions& opts, hipStream_t stream,
+                       bool& ok) {
+    static std::unordered_map<int64_t, std::pair<void*, int64_t>> bufs;
+    int64_t numel = 1;
+    for (auto s : sizes) numel *= s;
+    auto it = bufs.find(key);
+    if (it == bufs.end()) {
+        hipStreamCaptureStatus st = hipStreamCaptureStatusNone;
+        (void)hipStreamIsCapturing(stream, &st);
+        if (st != hipStreamCaptureStatusNone) {
+            ok = false;  // cannot hipMalloc during capture: classic path
+            return {};
+        }
+        void* p = nullptr;
+        if (hipMalloc(&p, numel * sizeof(float)) != hipSuccess) {
+            ok = false;
+            return {};
+        }
+        (void)hipMemsetAsync(p, 0, numel * sizeof(float), stream);
+        it = bufs.emplace(key, std::make_pair(p, numel)).first;
+    }
+    if (it->second.second != numel) {
+        ok = false;
+        return {};
+    }
+    ok = true;
+    return torch::from_blob(it->second.first, sizes,
+                            opts.dtype(torch::kFloat32));
+}
+
 float* acc_pool(int64_t n, const torch::TensorOptions& opts,
                 hipStream_t stream, torch::Tensor& holder) {
     static torch::Tensor pool;
@@ -3837,11 +3886,12 @@ torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor w, int64_t stride,
 
 torch::Tensor conv2d_wgrad(torch::Tensor dy, torch::Tensor x, int64_t stride,
                            int64_t R, int64_t S, int64_t pad,
-                           bool zero_init) {
-    // zero_init=false: epoch-graph capture contract (ops/functional.py
-    // _graph_no_zero) — the returned grad is stolen into p.grad and the
-    // multi-tensor Adam clears it in-graph each step, so the split-K
-    // atomics land on an already-zero buffer without a fill launch.
+                           int64_t gkey) {
+    // gkey != 0: epoch-graph fill-skip (ops/functional.py GRAPH_NO_ZERO) —
+    // the atomics accumulate into the process-lifetime grad_buf for this
+    // param, which the captured Adam consume-and-clears each step; no
+    // zero-fill launch. gkey == 0 (or grad_buf unavailable): classic
+    // zeroed allocation.
     CHECK_GPU(dy);
     TORCH_CHECK(dy.is_contiguous() && x.is_contiguous());
     ConvShape s;
@@ -3902,11 +3952,16 @@ torch::Tensor conv2d_wgrad(torch::Tensor dy, torch::Tensor x, int64_t stride,
         int tiles32 = ceildiv(s.Kout, 32) * ceildiv(NN, 64);
         int kc = std::max(1, std::min(ceildiv(KK, 128),
                                       512 / std::max(tiles32, 1)));
-        auto dw32 = (kc > 1 && zero_init)
-                        ? torch::zeros({s.Kout, R, S, s.C},
-                                       x.options().dtype(torch::kFloat32))
-                        : torch::empty({s.Kout, R, S, s.C},
-                                       x.options().dtype(torch::kFloat32));
+        torch::Tensor dw32;
+        bool pooled = false;
+        if (kc > 1 && gkey)
+            dw32 = grad_buf(gkey, {s.Kout, R, S, s.C}, x.options(), stream,
+                            pooled);
+        if (!pooled)
+            dw32 = kc > 1 ? torch::zeros({s.Kout, R, S, s.C},
+                                         x.options().dtype(torch::kFloat32))
+                          : torch::empty({s.Kout, R, S, s.C},
+                                         x.options().dtype(torch::kFloat32));
         dim3 g32(ceildiv(s.Kout, 32), ceildiv(NN, 64), kc);
         if (bkp128)
             hipLaunchKernelGGL((conv_wgrad_glds_k32_kernel<64, 2, 2, 1, 2, 128>),
@@ -3923,11 +3978,16 @@ torch::Tensor conv2d_wgrad(torch::Tensor dy, torch::Tensor x, int64_t stride,
     // measured: slab-rows + reduce LOSES to fp32 atomics here (CDNA4 L2
     // atomics absorb the z-chunk contention; the slab variant paid extra
     // write+read traffic) — profiles/r02_bench_conv logs
-    auto dw = (k_chunks > 1 && zero_init)
-                  ? torch::zeros({s.Kout, R, S, s.C},
-                                 x.options().dtype(torch::kFloat32))
-                  : torch::empty({s.Kout, R, S, s.C},
-                                 x.options().dtype(torch::kFloat32));
+    torch::Tensor dw;
+    bool dw_pooled = false;
+    if (k_chunks > 1 && gkey)
+        dw = grad_buf(gkey, {s.Kout, R, S, s.C}, x.options(), stream,
+                      dw_pooled);
+    if (!dw_pooled)
+        dw = k_chunks > 1 ? torch::zeros({s.Kout, R, S, s.C},
+                                         x.options().dtype(torch::kFloat32))
+                          : torch::empty({s.Kout, R, S, s.C},
+                                         x.options().dtype(torch::kFloat32));
     dim3 grid(ceildiv(s.Kout, 64), ceildiv(NN, bn), k_chunks);
     if (glds_ok) {
         static torch::Tensor zbuf;
@@ -4127,7 +4187,7 @@ torch::Tensor maxpool2x2_bwd(torch::Tensor dy, torch::Tensor idx, int64_t H,
 }
 
 std::vector<torch::Tensor> relu_bias_bwd(torch::Tensor dy, torch::Tensor y,
-                                         bool zero_init) {
+                                         int64_t gkey) {
     CHECK_GPU(dy);
     auto dyc = dy.contiguous();
     const int K = (int)dyc.size(-1);
@@ -4135,9 +4195,15 @@ std::vector<torch::Tensor> relu_bias_bwd(torch::Tensor dy, torch::Tensor y,
     auto dym = torch::empty_like(dyc);
     int rpb = (int)std::max<int64_t>(64, (M + 511) / 512);
     int nblk = (int)((M + rpb - 1) / rpb);
-    auto db = (nblk == 1 || !zero_init)
-                  ? torch::empty({K}, dyc.options().dtype(torch::kFloat32))
-                  : torch::zeros({K}, dyc.options().dtype(torch::kFloat32));
+    auto stream0 = at::cuda::getCurrentCUDAStream();
+    torch::Tensor db;
+    bool db_pooled = false;
+    if (nblk > 1 && gkey)
+        db = grad_buf(gkey, {K}, dyc.options(), stream0, db_pooled);
+    if (!db_pooled)
+        db = nblk == 1
+                 ? torch::empty({K}, dyc.options().dtype(torch::kFloat32))
+                 : torch::zeros({K}, dyc.options().dtype(torch::kFloat32));
     hipLaunchKernelGGL(relu_bias_bwd_kernel, dim3(nblk), dim3(256), 0,
                        at::cuda::getCurrentCUDAStream(), bf_ptr(dyc), bf_ptr(y),
                        bf_ptr_mut(dym), db.data_ptr<float>(), M, K, rpb);
@@ -4147,7 +4213,7 @@ std::vector<torch::Tensor> relu_bias_bwd(torch::Tensor dy, torch::Tensor y,
 std::vector<torch::Tensor> pool_relu_bias_bwd(torch::Tensor dy,
                                               torch::Tensor idx,
                                               torch::Tensor p, int64_t H,
-                                              int64_t W, bool zero_init) {
+                                              int64_t W, int64_t gkey) {
     CHECK_GPU(dy);
     auto dyc = dy.contiguous();
     const int N = (int)dyc.size(0), OH = (int)dyc.size(1),
@@ -4163,9 +4229,16 @@ std::vector<torch::Tensor> pool_relu_bias_bwd(torch::Tensor dy,
         // zero-fill serialized the whole layer onto one CU (-30% on the
         // headline config) — keep the parallel grid and pay the 4.7 us fill
         int blocks = (int)std::min<int64_t>(ceildiv(total8, 256), 4096);
-        auto db = (blocks == 1 || !zero_init)
-                      ? torch::empty({K}, dyc.options().dtype(torch::kFloat32))
-                      : torch::zeros({K}, dyc.options().dtype(torch::kFloat32));
+        torch::Tensor db;
+        bool db_pooled = false;
+        if (blocks > 1 && gkey)
+            db = grad_buf(gkey, {K}, dyc.options(), stream, db_pooled);
+        if (!db_pooled)
+            db = blocks == 1
+                     ? torch::empty({K},
+                                    dyc.options().dtype(torch::kFloat32))
+                     : torch::zeros({K},
+                                    dyc.options().dtype(torch::kFloat32));
         hipLaunchKernelGGL(pool_relu_bias_bwd2_kernel, dim3(blocks),
                            dim3(256), K * sizeof(float), stream, bf_ptr(dyc),
                            idx.data_ptr<uint8_t>(), bf_ptr(p),
@@ -4182,9 +4255,14 @@ std::vector<torch::Tensor> pool_relu_bias_bwd(torch::Tensor dy,
     lanes = 1 << (31 - __builtin_clz(lanes));
     rpb = std::max(rpb, 2 * lanes);
     int nblk = (int)((M + rpb - 1) / rpb);
-    auto db = (nblk == 1 || !zero_init)
-                  ? torch::empty({K}, dyc.options().dtype(torch::kFloat32))
-                  : torch::zeros({K}, dyc.options().dtype(torch::kFloat32));
+    torch::Tensor db;
+    bool db_pooled = false;
+    if (nblk > 1 && gkey)
+        db = grad_buf(gkey, {K}, dyc.options(), stream, db_pooled);
+    if (!db_pooled)
+        db = nblk == 1
+                 ? torch::empty({K}, dyc.options().dtype(torch::kFloat32))
+                 : torch::zeros({K}, dyc.options().dtype(torch::kFloat32));
     hipLaunchKernelGGL(pool_relu_bias_bwd_scalar_kernel, dim3(nblk),
                        dim3(256), 0, stream, bf_ptr(dyc),
                        idx.data_ptr<uint8_t>(), bf_ptr(p), bf_ptr_mut(dym),
@@ -4422,11 +4500,14 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
     int nblk = (int)((M + rpb - 1) / rpb);
     torch::Tensor dgamma, dbeta;
     if (fwd_sums.numel()) {
-        // atomic path: partials straight into {dgamma;dbeta} (zero at
-        // entry — the captured Adam consume-and-cleared them last step);
-        // block 0 zeroes this layer's forward sums for the next replay.
-        dgamma = torch::empty({C}, f32);
-        dbeta = torch::empty({C}, f32);
+        // atomic path (HEFL_BN_ATOMIC probe): partials straight into
+        // {dgamma;dbeta}; zeroed allocations keep the probe correct
+        // standalone (the two fills cost what the finalize launch saved —
+        // the probe's measured loss is the hot-word atomic contention,
+        // see ops/functional.py); block 0 zeroes this layer's forward
+        // sums for the next replay.
+        dgamma = torch::zeros({C}, f32);
+        dbeta = torch::zeros({C}, f32);
         hipLaunchKernelGGL(bn_bwd_partial_kernel<true>, dim3(nblk), dim3(256),
                            0, stream, bf_ptr(dyc), bf_ptr(x), ry,
                            mean.data_ptr<float>(), invstd.data_ptr<float>(),
@@ -4558,16 +4639,22 @@ torch::Tensor add_relu(torch::Tensor a, torch::Tensor b) {
     return y;
 }
 
-torch::Tensor bias_grad(torch::Tensor dy, bool zero_init) {
+torch::Tensor bias_grad(torch::Tensor dy, int64_t gkey) {
     CHECK_GPU(dy);
     auto dyc = dy.contiguous();
     const int K = (int)dyc.size(-1);
     const int64_t M = dyc.numel() / K;
     int rpb = (int)std::max<int64_t>(2048, (M + 63) / 64);
     dim3 grid(K, (unsigned)((M + rpb - 1) / rpb));
-    auto db = (grid.y == 1 || !zero_init)
-                  ? torch::empty({K}, dyc.options().dtype(torch::kFloat32))
-                  : torch::zeros({K}, dyc.options().dtype(torch::kFloat32));
+    auto stream0 = at::cuda::getCurrentCUDAStream();
+    torch::Tensor db;
+    bool db_pooled = false;
+    if (grid.y > 1 && gkey)
+        db = grad_buf(gkey, {K}, dyc.options(), stream0, db_pooled);
+    if (!db_pooled)
+        db = grid.y == 1
+                 ? torch::empty({K}, dyc.options().dtype(torch::kFloat32))
+                 : torch::zeros({K}, dyc.options().dtype(torch::kFloat32));
     hipLaunchKernelGGL(bias_grad_kernel, grid, dim3(256), 0,
                        at::cuda::getCurrentCUDAStream(), bf_ptr(dyc),
                        db.data_ptr<float>(), M, K, rpb);

@@ -150,12 +150,20 @@ class LocalClient:
         return {"graph": g, "x": sx, "y": sy, "loss": loss, "logits": logits,
                 "mt": mt, "grads": grads}
 
-    def _eager_warmup(self, x, y):
+    def _eager_warmup(self, x, y, clear_grads=False):
         self.opt.zero_grad()  # grads=None: fresh tensors each warmup step
         logits = self.model(x.to(self.compute_dtype))
         loss = softmax_xent(logits, y)
         loss.backward()
         self.opt.step()
+        if clear_grads:
+            # fill-skip contract: the shared grad_buf accumulation buffers
+            # must leave every step zeroed (in-graph the Adam zero_g pass
+            # does this; eager warmup does it explicitly)
+            with torch.no_grad():
+                for p in self.opt.params:
+                    if p.grad is not None:
+                        p.grad.zero_()
 
     # ----- whole-epoch hipGraph: ALL steps of one local epoch (forward,
     # loss, backward, Adam) captured as ONE graph. Per epoch the host does
@@ -187,13 +195,25 @@ class LocalClient:
             x0 = self.loader.augment(x0)
         X = x0.to(self.compute_dtype).contiguous()
         Y = y0.contiguous()
+        from ..ops import functional as Fx
+        import os
+        no_zero = os.getenv("HEFL_GRAPH_NO_ZERO", "1") == "1"
         side = torch.cuda.Stream()
         side.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(side):
-            for _ in range(3):  # warmup on the full-batch shape
-                self._eager_warmup(X[:B], Y[:B])
-            if n % B:  # and once on the partial tail-batch shape
-                self._eager_warmup(X[n - (n % B):], Y[n - (n % B):])
+            try:
+                # warmup runs under the fill-skip flag too: the grad_buf
+                # allocations (cnn.hip) must happen HERE, eagerly — during
+                # capture hipMalloc is illegal and grad_buf would fall
+                # back to the classic zeroed path for the whole graph
+                Fx.GRAPH_NO_ZERO = no_zero
+                for _ in range(3):  # warmup on the full-batch shape
+                    self._eager_warmup(X[:B], Y[:B], clear_grads=no_zero)
+                if n % B:  # and once on the partial tail-batch shape
+                    self._eager_warmup(X[n - (n % B):], Y[n - (n % B):],
+                                       clear_grads=no_zero)
+            finally:
+                Fx.GRAPH_NO_ZERO = False
         torch.cuda.current_stream().wait_stream(side)
         self.opt.prepare_graph_state(X.device)
         if not hasattr(self, "_acc_loss"):
@@ -217,19 +237,18 @@ class LocalClient:
         g = torch.cuda.CUDAGraph()
         import hefl
         C = hefl.load_extension()
-        from ..ops import functional as Fx
-        import os
-        # HEFL_GRAPH_NO_ZERO=1 (probe, default OFF): stolen grads skip
-        # their zero-init fills; the captured Adam clears each grad as it
-        # consumes it and the eager zero_() after capture covers the first
-        # replay. Measured config2 36.7->38.8 rounds/s, config4
-        # 35.5k->42.1k samples/s — but the contract is UNSOUND in general:
-        # a grad buffer allocated during capture can reuse a pool block
-        # freed earlier in the SAME capture, and the previous owner's
-        # captured writes then re-pollute it on every replay (observed as
-        # training divergence on resnet18 BN grads; cnn2/refcnn6 happened
-        # to capture cleanly). Default stays correct-by-construction.
-        no_zero = os.getenv("HEFL_GRAPH_NO_ZERO", "0") == "1"
+        # HEFL_GRAPH_NO_ZERO (default ON): stolen grads skip their
+        # zero-init fills — the captured Adam clears each grad as it
+        # consumes it (zero_grad=True below) and the eager zero_() after
+        # capture covers the first replay. SOUND because the accumulation
+        # buffers are process-lifetime hipMalloc allocations keyed by
+        # param id (cnn.hip grad_buf), never part of the caching
+        # allocator: nothing allocated during capture can alias them. The
+        # first design accumulated into capture-pool tensors instead and
+        # diverged on resnet18 when a grad reused a block freed earlier
+        # in the same capture — that hazard is what grad_buf removes.
+        # Measured: config2 36.7->38.8 rounds/s, config4 35.5k->42.1k
+        # samples/s (the fills were 11.5% of config2 kernel time).
         try:
             Fx.GRAPH_NO_ZERO = no_zero
             with torch.cuda.graph(g, stream=side):

@@ -83,6 +83,10 @@ class _Conv2dFn(torch.autograd.Function):
         ctx.pad = pad
         ctx.has_bias = b is not None
         ctx.cpad = False
+        # grad_buf keys: the params' CPython ids — stable for the params'
+        # lifetime, one accumulation buffer per layer (cnn.hip grad_buf)
+        ctx.wkey = id(w)
+        ctx.bkey = id(b) if b is not None else 0
         if x.is_cuda:
             wb = _gpu_dtype(w)
             if _want_cpad(x.shape[-1], w.shape[0]):
@@ -112,18 +116,20 @@ class _Conv2dFn(torch.autograd.Function):
         if dy.is_cuda:
             dy = dy.contiguous()
             db = None
-            zi = not GRAPH_NO_ZERO
+            bkey = ctx.bkey if GRAPH_NO_ZERO else 0
             if relu and ctx.has_bias:
-                dy, db = _C().relu_bias_bwd(dy, y, zero_init=zi)  # one fused pass
+                dy, db = _C().relu_bias_bwd(dy, y, gkey=bkey)  # one fused pass
             elif relu:
                 dy = _C().relu_bwd(dy, y)
             elif ctx.has_bias:
-                db = _C().bias_grad(dy, zero_init=zi)
+                db = _C().bias_grad(dy, gkey=bkey)
             dx = _C().conv2d_dgrad(dy, w, stride, x.shape[1], x.shape[2], pad) \
                 if ctx.needs_input_grad[0] else None
+            # channel-padded dw is sliced before the steal -> classic path
             dw = _C().conv2d_wgrad(dy, x.contiguous(), stride, w.shape[1],
                                    w.shape[2], pad,
-                                   zero_init=zi or ctx.cpad)
+                                   gkey=0 if (ctx.cpad or not GRAPH_NO_ZERO)
+                                   else ctx.wkey)
             if ctx.cpad:  # drop the zero-padded channel lanes
                 dw = dw[..., :ctx.in_C].contiguous()
                 if dx is not None:
@@ -199,6 +205,8 @@ class _ConvReluPoolFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, b, pad: int):
         ctx.pad = pad
+        ctx.wkey = id(w)
+        ctx.bkey = id(b)
         wb = _gpu_dtype(w)
         ctx.cpad = _want_cpad(x.shape[-1], w.shape[0])
         if ctx.cpad:  # stem shapes ride the glds pipeline on padded channels
@@ -218,13 +226,15 @@ class _ConvReluPoolFn(torch.autograd.Function):
     def backward(ctx, dy):
         x, w, p, idx = ctx.saved_tensors
         ch, cw = ctx.conv_hw
-        zi = not GRAPH_NO_ZERO
+        bkey = ctx.bkey if GRAPH_NO_ZERO else 0
         dym, db = _C().pool_relu_bias_bwd(dy.contiguous(), idx, p, ch, cw,
-                                          zero_init=zi)
+                                          gkey=bkey)
         dx = _C().conv2d_dgrad(dym, w, 1, x.shape[1], x.shape[2], ctx.pad) \
             if ctx.needs_input_grad[0] else None
         dw = _C().conv2d_wgrad(dym, x.contiguous(), 1, w.shape[1], w.shape[2],
-                               ctx.pad, zero_init=zi or ctx.cpad)
+                               ctx.pad,
+                               gkey=0 if (ctx.cpad or not GRAPH_NO_ZERO)
+                               else ctx.wkey)
         if ctx.cpad:
             dw = dw[..., :ctx.in_C].contiguous()
             if dx is not None:
@@ -249,6 +259,7 @@ class _LinearFn(torch.autograd.Function):
     def forward(ctx, x, w, b, relu: bool):
         ctx.relu = relu
         ctx.has_bias = b is not None
+        ctx.bkey = id(b) if b is not None else 0
         if x.is_cuda:
             wb = _gpu_dtype(w)
             bb = b.detach().float() if b is not None else torch.empty(0, device=x.device)
@@ -267,13 +278,13 @@ class _LinearFn(torch.autograd.Function):
         if dy.is_cuda:
             dy = dy.contiguous()
             db = None
-            zi = not GRAPH_NO_ZERO
+            bkey = ctx.bkey if GRAPH_NO_ZERO else 0
             if ctx.relu and ctx.has_bias:
-                dy, db = _C().relu_bias_bwd(dy, y, zero_init=zi)
+                dy, db = _C().relu_bias_bwd(dy, y, gkey=bkey)
             elif ctx.relu:
                 dy = _C().relu_bwd(dy, y)
             elif ctx.has_bias:
-                db = _C().bias_grad(dy.view(-1, dy.shape[-1]), zero_init=zi)
+                db = _C().bias_grad(dy.view(-1, dy.shape[-1]), gkey=bkey)
             dx = _C().linear_dgrad(dy, w) if ctx.needs_input_grad[0] else None
             dw = _C().linear_wgrad(dy, x).to(torch.float32)
         else:
