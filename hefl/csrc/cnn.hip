@@ -3818,7 +3818,11 @@ torch::Tensor conv2d_dgrad(torch::Tensor dy, torch::Tensor w, int64_t stride,
             return dx;
         }
     }
-    if (C > 16 && C % 8 == 0 && s.Kout % 8 == 0) {
+    // C == 16 included (cnn2 conv2 dgrad): measured a WASH end-to-end on
+    // config #2 (38.45 vs 38.44 rounds/s) — the BN=64 tile wastes 3/4 of
+    // its columns, which cancels the 16-B-vector advantage over the
+    // generic gather at this width. Kept for the uniform path.
+    if (C >= 16 && C % 8 == 0 && s.Kout % 8 == 0) {
         static torch::Tensor zbuf;
         if (!zbuf.defined() || zbuf.device() != dy.device())
             zbuf = torch::zeros({8}, dy.options());
