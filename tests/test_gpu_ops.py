@@ -435,3 +435,28 @@ def test_generic_maxpool_bwd_pooled_accumulator():
         _close(outs[i][1], outs[0][1], rel=1e-3)
     _close(outs[0][0], yc.permute(0, 2, 3, 1))
     _close(outs[0][1], xc.grad)  # xc is NHWC, so its grad already is too
+
+
+@pytest.mark.parametrize("preset_name", ["config3", "config4"])
+def test_graphed_client_converges_other_models(preset_name):
+    """Graphed-epoch convergence for the remaining model families (lenet5,
+    cnn4). Together with the cnn2 and resnet18 graphed tests this guards
+    the fill-skip contract (functional.GRAPH_NO_ZERO + cnn.hip grad_buf)
+    on every model the presets name — the capture-pool-reuse hazard that
+    motivated grad_buf was allocation-layout (i.e. model) dependent."""
+    from hefl.config import preset
+    from hefl.fl.client import LocalClient
+
+    cfg = preset(preset_name)
+    cfg.fl.n_clients = 1
+    cfg.fl.samples_per_client = 64
+    cfg.train.batch_size = 16
+    cfg.train.hip_graphs = True
+    c = LocalClient(cfg, 0, device="cuda:0")
+    assert c.use_graphs
+    first = c.local_train(epochs=1)
+    losses = [c.local_train(epochs=1).train_loss for _ in range(25)]
+    tail = sum(losses[-5:]) / 5
+    assert tail < first.train_loss * 0.7, (first.train_loss, losses[-5:])
+    w = c.get_weights()
+    assert not torch.isnan(w).any()
