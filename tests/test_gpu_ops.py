@@ -437,7 +437,7 @@ def test_generic_maxpool_bwd_pooled_accumulator():
     _close(outs[0][1], xc.grad)  # xc is NHWC, so its grad already is too
 
 
-@pytest.mark.parametrize("preset_name", ["config3", "config4"])
+@pytest.mark.parametrize("preset_name", ["config3", "config4", "reference"])
 def test_graphed_client_converges_other_models(preset_name):
     """Graphed-epoch convergence for the remaining model families (lenet5,
     cnn4). Together with the cnn2 and resnet18 graphed tests this guards
