@@ -247,8 +247,8 @@ class LocalClient:
         # first design accumulated into capture-pool tensors instead and
         # diverged on resnet18 when a grad reused a block freed earlier
         # in the same capture — that hazard is what grad_buf removes.
-        # Measured: config2 36.7->38.8 rounds/s, config4 35.5k->42.1k
-        # samples/s (the fills were 11.5% of config2 kernel time).
+        # Measured (sound build): config2 36.7->38.4 rounds/s, reference
+        # 4.8->5.03 (the fills were 11.5% of config2 kernel time).
         try:
             Fx.GRAPH_NO_ZERO = no_zero
             with torch.cuda.graph(g, stream=side):
