@@ -49,9 +49,28 @@ import os
 
 # BN atomic no-finalize probe, default OFF: measured -35% on config #5
 # (hot-word fp32 atomics from ~512 partial blocks per channel word beat
-# the 4.7 us finalize launches they replace) and the backward half shares
-# the HEFL_GRAPH_NO_ZERO capture-pool-reuse hazard (fl/client.py).
+# the 4.7 us finalize launches they replace).
 _BN_ATOMIC_MODE = os.getenv("HEFL_BN_ATOMIC", "0")
+
+# PROBE (default OFF, HEFL_BWD_FORK=1 to enable): fork each conv's wgrad
+# onto a side stream DURING GRAPH CAPTURE so the recorded fork/join
+# become parallel hipGraph branches and wgrad overlaps the dgrad chain.
+# Measured a clear LOSS: config2 38.9 -> 29.1 rounds/s (-25%), reference
+# 5.0 -> 4.6 (-8%), config4 flat — the per-conv fork/join event nodes
+# (cross-stream graph dependencies) cost more at replay than the overlap
+# recovers at the ~5 us per-kernel floor. Numerics-correct (all GPU
+# tests pass with it on); kept as the documented A/B for the
+# "multi-stream backward" idea from the round-1 roadmap.
+_BWD_FORK = os.getenv("HEFL_BWD_FORK", "0") == "1"
+_SIDE_STREAMS = {}
+
+
+def _fork_side(dev):
+    s = _SIDE_STREAMS.get(dev)
+    if s is None:
+        s = torch.cuda.Stream(device=dev)
+        _SIDE_STREAMS[dev] = s
+    return s
 
 # Epoch-graph capture contract (fl/client.py epoch capture, probe
 # HEFL_GRAPH_NO_ZERO=1 — default OFF): while True, grad outputs that
@@ -123,17 +142,37 @@ class _Conv2dFn(torch.autograd.Function):
                 dy = _C().relu_bwd(dy, y)
             elif ctx.has_bias:
                 db = _C().bias_grad(dy, gkey=bkey)
-            dx = _C().conv2d_dgrad(dy, w, stride, x.shape[1], x.shape[2], pad) \
-                if ctx.needs_input_grad[0] else None
-            # channel-padded dw is sliced before the steal -> classic path
-            dw = _C().conv2d_wgrad(dy, x.contiguous(), stride, w.shape[1],
-                                   w.shape[2], pad,
-                                   gkey=0 if (ctx.cpad or not GRAPH_NO_ZERO)
-                                   else ctx.wkey)
-            if ctx.cpad:  # drop the zero-padded channel lanes
-                dw = dw[..., :ctx.in_C].contiguous()
-                if dx is not None:
-                    dx = dx[..., :ctx.in_C].contiguous()
+            wg = ctx.wkey if (GRAPH_NO_ZERO and not ctx.cpad) else 0
+            fork = (_BWD_FORK and ctx.needs_input_grad[0]
+                    and torch.cuda.is_current_stream_capturing())
+            if fork:
+                cur = torch.cuda.current_stream()
+                side = _fork_side(dy.device)
+                ev = torch.cuda.Event()
+                ev.record(cur)
+                with torch.cuda.stream(side):
+                    side.wait_event(ev)
+                    dw = _C().conv2d_wgrad(dy, x.contiguous(), stride,
+                                           w.shape[1], w.shape[2], pad,
+                                           gkey=wg)
+                    if ctx.cpad:  # drop the zero-padded channel lanes
+                        dw = dw[..., :ctx.in_C].contiguous()
+                    ev2 = torch.cuda.Event()
+                    ev2.record(side)
+                dx = _C().conv2d_dgrad(dy, w, stride, x.shape[1], x.shape[2],
+                                       pad)
+                cur.wait_event(ev2)
+            else:
+                dx = _C().conv2d_dgrad(dy, w, stride, x.shape[1],
+                                       x.shape[2], pad) \
+                    if ctx.needs_input_grad[0] else None
+                # channel-padded dw sliced before the steal -> classic path
+                dw = _C().conv2d_wgrad(dy, x.contiguous(), stride, w.shape[1],
+                                       w.shape[2], pad, gkey=wg)
+                if ctx.cpad:
+                    dw = dw[..., :ctx.in_C].contiguous()
+            if ctx.cpad and dx is not None:
+                dx = dx[..., :ctx.in_C].contiguous()
         else:
             dy = dy.float()
             if relu:
@@ -229,16 +268,34 @@ class _ConvReluPoolFn(torch.autograd.Function):
         bkey = ctx.bkey if GRAPH_NO_ZERO else 0
         dym, db = _C().pool_relu_bias_bwd(dy.contiguous(), idx, p, ch, cw,
                                           gkey=bkey)
-        dx = _C().conv2d_dgrad(dym, w, 1, x.shape[1], x.shape[2], ctx.pad) \
-            if ctx.needs_input_grad[0] else None
-        dw = _C().conv2d_wgrad(dym, x.contiguous(), 1, w.shape[1], w.shape[2],
-                               ctx.pad,
-                               gkey=0 if (ctx.cpad or not GRAPH_NO_ZERO)
-                               else ctx.wkey)
-        if ctx.cpad:
-            dw = dw[..., :ctx.in_C].contiguous()
-            if dx is not None:
-                dx = dx[..., :ctx.in_C].contiguous()
+        wg = ctx.wkey if (GRAPH_NO_ZERO and not ctx.cpad) else 0
+        fork = (_BWD_FORK and ctx.needs_input_grad[0]
+                and torch.cuda.is_current_stream_capturing())
+        if fork:
+            cur = torch.cuda.current_stream()
+            side = _fork_side(dy.device)
+            ev = torch.cuda.Event()
+            ev.record(cur)
+            with torch.cuda.stream(side):
+                side.wait_event(ev)
+                dw = _C().conv2d_wgrad(dym, x.contiguous(), 1, w.shape[1],
+                                       w.shape[2], ctx.pad, gkey=wg)
+                if ctx.cpad:
+                    dw = dw[..., :ctx.in_C].contiguous()
+                ev2 = torch.cuda.Event()
+                ev2.record(side)
+            dx = _C().conv2d_dgrad(dym, w, 1, x.shape[1], x.shape[2], ctx.pad)
+            cur.wait_event(ev2)
+        else:
+            dx = _C().conv2d_dgrad(dym, w, 1, x.shape[1], x.shape[2],
+                                   ctx.pad) \
+                if ctx.needs_input_grad[0] else None
+            dw = _C().conv2d_wgrad(dym, x.contiguous(), 1, w.shape[1],
+                                   w.shape[2], ctx.pad, gkey=wg)
+            if ctx.cpad:
+                dw = dw[..., :ctx.in_C].contiguous()
+        if ctx.cpad and dx is not None:
+            dx = dx[..., :ctx.in_C].contiguous()
         return dx, dw, db, None
 
 
