@@ -72,19 +72,20 @@ def _fork_side(dev):
         _SIDE_STREAMS[dev] = s
     return s
 
-# Epoch-graph capture contract (fl/client.py epoch capture, probe
-# HEFL_GRAPH_NO_ZERO=1 — default OFF): while True, grad outputs that
-# autograd steals straight into p.grad skip their zero-init fill launch —
-# the captured multi-tensor Adam clears every stolen grad in-graph
-# (zero_g=1) and the client zeroes the held buffers once right after
-# capture. Channel-padded conv dw is SLICED before the steal, so it keeps
-# its fill. Measured: the fills were 11.5% of config #2 kernel time;
-# skipping them gave config2 +5.5%, config4 +19%. DEFAULT OFF because the
-# contract is unsound in general: an allocation made DURING capture can
-# reuse a pool block freed earlier in the same capture, and the previous
-# owner's captured writes re-pollute the "zero" buffer on every replay
-# (observed: resnet18 BN grads diverged; cnn2/refcnn6 happened to capture
-# cleanly). See fl/client.py and PERFORMANCE.md.
+# Epoch-graph capture contract (fl/client.py sets this around warmup +
+# capture; HEFL_GRAPH_NO_ZERO=0 disables): while True, conv dw / bias db
+# outputs skip their zero-init fill launch — the gkey passed to the C++
+# wrappers routes the split-K/scatter atomics into a process-lifetime
+# hipMalloc buffer (cnn.hip grad_buf) that autograd steals into p.grad
+# and the captured multi-tensor Adam consume-and-clears each step (the
+# client zeroes the held buffers once right after capture for the first
+# replay). Sound because grad_buf memory never belongs to the caching
+# allocator, so nothing allocated during a capture can alias it — an
+# earlier design that accumulated into capture-pool tensors diverged on
+# resnet18 when a grad reused a block freed earlier in the same capture.
+# Channel-padded conv dw is SLICED before the steal, so it keeps its
+# fill (gkey=0). Measured: fills were 11.5% of config #2 kernel time;
+# config2 36.7 -> 38.4-38.9 rounds/s, reference 4.8 -> 4.9-5.0.
 GRAPH_NO_ZERO = False
 
 
