@@ -106,6 +106,9 @@ std::vector<torch::Tensor> pool_relu_bias_bwd(torch::Tensor dy,
 void adam_prep_epoch(torch::Tensor step, torch::Tensor sched,
                      torch::Tensor hyper, double b1, double b2, int64_t S);
 torch::Tensor pad_channels(torch::Tensor x, int64_t C8);
+std::vector<torch::Tensor> dense_head2_fwd(torch::Tensor x,
+                                           torch::Tensor w1, torch::Tensor b1,
+                                           torch::Tensor w2, torch::Tensor b2);
 std::vector<torch::Tensor> dense_head2_bwd(torch::Tensor dlogits,
                                            torch::Tensor x, torch::Tensor h1,
                                            torch::Tensor w1, torch::Tensor w2);
@@ -183,6 +186,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("fused_adam_mt", &fused_adam_mt);
     m.def("pad_channels", &pad_channels,
           "zero-pad NHWC channel dim to C8 (stem -> glds MFMA path)");
+    m.def("dense_head2_fwd", &dense_head2_fwd);
     m.def("dense_head2_bwd", &dense_head2_bwd,
           "single-launch backward of the 2-layer dense head (M <= 32)");
     m.def("pack_mt", &pack_mt);

@@ -62,15 +62,32 @@ class _SeqCNN(nn.Module):
                           and head[0].weight.shape[0] <= 128
                           and head[0].weight.shape[0] % 16 == 0
                           and feat % 16 == 0 and feat <= 1024)
+        # fused FORWARD only (row-parallel, one block per sample, backward
+        # composed from the standard kernels): measured -3% on config #2
+        # (37.5 vs 38.8 rounds/s) — at M=32 the kernel is 32 blocks of
+        # serial 800-element VALU dot chains, slower than the two MFMA
+        # linear_splitk launches it replaces. Numerics-validated
+        # (test_dense_head2_fused_forward_parity); probe HEFL_HEAD2F=1.
+        self._head2f_ok = (os.environ.get("HEFL_HEAD2F", "0") == "1"
+                           and len(head) == 2 and head[0].relu
+                           and not head[1].relu and head[0].bias is not None
+                           and head[1].bias is not None
+                           and n_classes <= 256
+                           and head[0].weight.shape[0] <= 256
+                           and feat % 8 == 0 and feat <= 4096)
 
     def forward(self, x):
         for m in self.trunk:
             x = m(x)
         x = self.flatten(x)
+        from ..ops import functional as Fx
         if x.is_cuda and self._head2_ok and x.shape[0] <= 32:
-            from ..ops import functional as Fx
             return Fx.dense_head2(x, self.head[0].weight, self.head[0].bias,
                                   self.head[1].weight, self.head[1].bias)
+        if x.is_cuda and self._head2f_ok and x.shape[0] <= 4096:
+            return Fx.dense_head2_fwdfused(
+                x, self.head[0].weight, self.head[0].bias,
+                self.head[1].weight, self.head[1].bias)
         for m in self.head:
             x = m(x)
         return x

@@ -460,3 +460,30 @@ def test_graphed_client_converges_other_models(preset_name):
     assert tail < first.train_loss * 0.7, (first.train_loss, losses[-5:])
     w = c.get_weights()
     assert not torch.isnan(w).any()
+
+
+def test_dense_head2_fused_forward_parity():
+    """dense_head2_fwdfused (row-parallel fused fwd + composed backward)
+    must match the composed linear(relu) -> linear pipeline in forward
+    values AND all five gradients (its backward runs the same kernels)."""
+    torch.manual_seed(9)
+    M, K, N1, N2 = 32, 800, 128, 10
+    x = torch.randn(M, K)
+    w1 = torch.randn(N1, K) * 0.05
+    b1 = torch.randn(N1) * 0.1
+    w2 = torch.randn(N2, N1) * 0.1
+    b2 = torch.randn(N2) * 0.1
+    g = torch.randn(M, N2)
+
+    def run(fn):
+        xg = x.to("cuda", torch.bfloat16).requires_grad_(True)
+        ps = [t.cuda().requires_grad_(True) for t in (w1, b1, w2, b2)]
+        y = fn(xg, *ps)
+        y.backward(g.to("cuda", torch.bfloat16))
+        return (y.detach(), xg.grad, *[p.grad for p in ps])
+
+    ref = run(lambda xg, a, b, c, d: Fx.linear(Fx.linear(xg, a, b, relu=True),
+                                               c, d, relu=False))
+    out = run(Fx.dense_head2_fwdfused)
+    for a, b_ in zip(out, ref):
+        _close(a, b_, rel=2e-2, atol=1e-3)
